@@ -1,0 +1,328 @@
+"""Stream: the per-stream task graph (THE hot loop).
+
+MI355X-native re-design of reference crates/arkflow-core/src/stream/mod.rs:
+  input → [WAL] → [buffer] → N processor workers → sequence-ordered output
+with bounded queues, 1024-in-flight backpressure (stream/mod.rs:37,388-395),
+at-least-once ack chain, EOF/reconnect semantics (:282-306), WAL replay before
+ingest (:190-210) and the close order input→buffer→pipeline→output→
+error_output→WAL (:542-591).
+
+Host orchestration is asyncio (the Tokio analog); per-batch compute runs in
+HIP kernels / C++ that release the GIL, so `thread_num` workers here give
+overlap of GPU compute with ingest + output exactly like the reference's
+worker tasks.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+from typing import List, Optional, Tuple
+
+from .batch import MessageBatch
+from .config import StreamConfig
+from .errors import DisconnectionError, EOFError_
+from .metrics import RuntimeMetrics
+from .pipeline import Pipeline
+from .registry import build_component
+from .spi import Ack, Buffer, Input, NoopAck, Output, Resource
+
+log = logging.getLogger("arkflow_amd.stream")
+
+MAX_IN_FLIGHT = 1024       # reference stream/mod.rs:37
+RECONNECT_SECS = 5.0       # reference stream/mod.rs:300 (patchable in tests)
+
+_SENTINEL = object()
+
+
+async def _race(coro, cancel: asyncio.Event):
+    """Run *coro* but abort (returning _SENTINEL) if *cancel* fires first."""
+    task = asyncio.ensure_future(coro)
+    waiter = asyncio.ensure_future(cancel.wait())
+    done, _ = await asyncio.wait(
+        {task, waiter}, return_when=asyncio.FIRST_COMPLETED
+    )
+    if task in done:
+        waiter.cancel()
+        return task.result()
+    task.cancel()
+    try:
+        await task
+    except (asyncio.CancelledError, Exception):  # noqa: BLE001
+        pass
+    return _SENTINEL
+
+
+class Stream:
+    def __init__(
+        self,
+        config: StreamConfig,
+        input_: Input,
+        pipeline: Pipeline,
+        output: Output,
+        error_output: Optional[Output] = None,
+        buffer: Optional[Buffer] = None,
+        wal=None,
+        temporaries: Optional[dict] = None,
+        metrics: Optional[RuntimeMetrics] = None,
+    ):
+        self.config = config
+        self.input = input_
+        self.pipeline = pipeline
+        self.output = output
+        self.error_output = error_output
+        self.buffer = buffer
+        self.wal = wal
+        self.temporaries = temporaries or {}
+        self.metrics = metrics or RuntimeMetrics()
+        self.thread_num = config.pipeline.resolved_thread_num()
+        # sequencing / backpressure state
+        self._seq = 0
+        self._next_seq = 0
+        self._bp_event = asyncio.Event()
+        self._fatal: Optional[BaseException] = None
+
+    # ------------------------------------------------------------------- run
+    async def run(self, cancel: asyncio.Event) -> None:
+        await self.input.connect()
+        await self.output.connect()
+        if self.error_output is not None:
+            await self.error_output.connect()
+        for t in self.temporaries.values():
+            await t.connect()
+
+        qsize = self.thread_num * 4
+        input_q: asyncio.Queue = asyncio.Queue(maxsize=qsize)
+        output_q: asyncio.Queue = asyncio.Queue(maxsize=qsize)
+
+        tasks: List[asyncio.Task] = []
+        if self.buffer is not None:
+            tasks.append(asyncio.ensure_future(
+                self._do_buffer(input_q)))
+        for i in range(self.thread_num):
+            tasks.append(asyncio.ensure_future(
+                self._do_processor(input_q, output_q, cancel)))
+        out_task = asyncio.ensure_future(
+            self._do_output(output_q))
+        tasks.append(out_task)
+
+        # WAL recovery replay BEFORE live ingest (stream/mod.rs:190-210):
+        # consumers are already running so the bounded queue drains.
+        if self.wal is not None:
+            async for seq, batch in self.wal.read_after_cursor():
+                from .wal.wal import WalAck
+                await self._forward(input_q, batch, WalAck(self.wal, seq, NoopAck()))
+
+        in_task = asyncio.ensure_future(
+            self._do_input(input_q, cancel))
+
+        try:
+            await in_task
+            # input done (EOF or cancel): drain the buffer's final window
+            if self.buffer is not None:
+                await self.buffer.flush()
+                await tasks[0]  # buffer task exits after drain
+            # stop processors
+            for _ in range(self.thread_num):
+                await input_q.put(_SENTINEL)
+            await asyncio.gather(
+                *[t for t in tasks if t is not out_task and not t.done()])
+            await out_task
+        finally:
+            for t in tasks + [in_task]:
+                if not t.done():
+                    t.cancel()
+            await asyncio.gather(*tasks, in_task, return_exceptions=True)
+            await self._close_all()
+        if self._fatal is not None:
+            raise self._fatal
+
+    async def _close_all(self) -> None:
+        # close order per reference stream/mod.rs:542-591
+        for closer in (
+            self.input.close,
+            (self.buffer.close if self.buffer else None),
+            self.pipeline.close,
+            self.output.close,
+            (self.error_output.close if self.error_output else None),
+            (self.wal.close if self.wal else None),
+        ):
+            if closer is None:
+                continue
+            try:
+                await closer()
+            except Exception:  # noqa: BLE001
+                log.exception("close failure in stream %s", self.config.id)
+        for t in self.temporaries.values():
+            try:
+                await t.close()
+            except Exception:  # noqa: BLE001
+                pass
+
+    # ----------------------------------------------------------------- input
+    async def _forward(self, input_q: asyncio.Queue, batch: MessageBatch,
+                       ack: Ack) -> None:
+        """Route to buffer or straight to the processor queue
+        (stream/mod.rs:229-243)."""
+        if self.buffer is not None:
+            await self.buffer.write(batch, ack)
+        else:
+            await input_q.put((batch, ack))
+
+    async def _do_input(self, input_q: asyncio.Queue, cancel: asyncio.Event
+                        ) -> None:
+        while not cancel.is_set():
+            try:
+                item = await _race(self.input.read(), cancel)
+                if item is _SENTINEL:
+                    break
+                batch, ack = item
+            except EOFError_:
+                break
+            except DisconnectionError:
+                self.metrics.input_reconnects += 1
+                slept = await _race(asyncio.sleep(RECONNECT_SECS), cancel)
+                if slept is _SENTINEL:
+                    break
+                try:
+                    await self.input.connect()
+                except Exception:  # noqa: BLE001
+                    self.metrics.input_errors += 1
+                continue
+            except Exception:  # noqa: BLE001
+                self.metrics.input_errors += 1
+                log.exception("input read error in stream %s", self.config.id)
+                continue
+            self.metrics.input_batches += 1
+            self.metrics.input_messages += batch.num_rows
+            if self.wal is not None:
+                from .wal.wal import WalAck
+                seq = await self.wal.append(batch)
+                ack = WalAck(self.wal, seq, ack)
+            await self._forward(input_q, batch, ack)
+
+    # ---------------------------------------------------------------- buffer
+    async def _do_buffer(self, input_q: asyncio.Queue) -> None:
+        while True:
+            item = await self.buffer.read()
+            if item is None:
+                break
+            batch, ack = item
+            await input_q.put((batch, ack))
+
+    # ------------------------------------------------------------- processor
+    async def _do_processor(self, input_q: asyncio.Queue,
+                            output_q: asyncio.Queue,
+                            cancel: asyncio.Event) -> None:
+        while True:
+            # backpressure (stream/mod.rs:388-395): bound un-acked in-flight
+            while self._seq - self._next_seq > MAX_IN_FLIGHT:
+                self._bp_event.clear()
+                await self._bp_event.wait()
+            item = await input_q.get()
+            if item is _SENTINEL:
+                await output_q.put(_SENTINEL)
+                return
+            batch, ack = item
+            seq = self._seq
+            self._seq += 1
+            try:
+                results = await self.pipeline.process(batch)
+                await output_q.put((seq, results, None, ack))
+            except Exception as e:  # noqa: BLE001
+                self.metrics.processing_errors += 1
+                await output_q.put((seq, None, (batch, e), ack))
+
+    # ---------------------------------------------------------------- output
+    async def _do_output(self, output_q: asyncio.Queue) -> None:
+        pending = {}
+        sentinels = 0
+        while sentinels < self.thread_num or pending:
+            item = await output_q.get()
+            if item is _SENTINEL:
+                sentinels += 1
+                continue
+            seq, results, err, ack = item
+            pending[seq] = (results, err, ack)
+            while self._next_seq in pending:
+                results, err, ack = pending.pop(self._next_seq)
+                await self._emit(results, err, ack)
+                self._next_seq += 1
+                self._bp_event.set()  # wake backpressured processors
+
+    async def _emit(self, results, err, ack: Ack) -> None:
+        if err is not None:
+            batch, exc = err
+            if self.error_output is not None:
+                try:
+                    await self.error_output.write_batch([batch])
+                    await ack.ack()
+                except Exception:  # noqa: BLE001
+                    self.metrics.output_errors += 1
+                    log.exception("error-output failure in %s", self.config.id)
+            else:
+                log.error("stream %s processing error (no error_output): %s",
+                          self.config.id, exc)
+                await ack.ack()
+            return
+        if not results:
+            await ack.ack()
+            return
+        try:
+            await self.output.write_batch(results)
+            self.metrics.output_batches += len(results)
+            self.metrics.output_messages += sum(b.num_rows for b in results)
+            await ack.ack()
+        except Exception:  # noqa: BLE001
+            # ack withheld → WAL / source offset replay (stream/mod.rs:517-537)
+            self.metrics.output_errors += 1
+            log.exception("output failure in stream %s (ack withheld)",
+                          self.config.id)
+
+
+def build_stream(config: StreamConfig) -> Stream:
+    """StreamConfig → wired Stream (reference stream/mod.rs:1495 build()).
+
+    Build order: temporaries → input → pipeline → output → error_output →
+    buffer → WAL.
+    """
+    resource = Resource()
+    resource.device = _resolve_device(config.device)
+
+    temporaries = {}
+    for t_spec in config.temporary:
+        spec = dict(t_spec)
+        name = spec.pop("name", spec.get("type"))
+        temporaries[name] = build_component("temporary", spec, resource)
+    resource.temporaries = temporaries
+
+    input_ = build_component("input", config.input, resource)
+    processors = [
+        build_component("processor", p, resource)
+        for p in config.pipeline.processors
+    ]
+    output = build_component("output", config.output, resource)
+    error_output = (
+        build_component("output", config.error_output, resource)
+        if config.error_output else None
+    )
+    buffer = (
+        build_component("buffer", config.buffer, resource)
+        if config.buffer else None
+    )
+    wal = None
+    if config.durability and config.durability.enabled:
+        from .wal.wal import Wal
+        wal = Wal.open(config.durability, stream_id=config.id)
+    return Stream(
+        config, input_, Pipeline(processors), output, error_output,
+        buffer, wal, temporaries,
+    )
+
+
+def _resolve_device(device: Optional[str]):
+    import torch
+    if device is not None:
+        return torch.device(device)
+    if torch.cuda.is_available():
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")

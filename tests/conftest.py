@@ -1,0 +1,30 @@
+import asyncio
+
+import pytest
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run with -m gpu)"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    """Skip gpu-marked tests automatically when no GPU is visible."""
+    import torch
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture
+def run():
+    """Run a coroutine to completion on a fresh event loop."""
+    def _run(coro, timeout=30.0):
+        return asyncio.get_event_loop_policy().new_event_loop().run_until_complete(
+            asyncio.wait_for(coro, timeout)
+        )
+    return _run

@@ -1,0 +1,239 @@
+"""Engine-level tests with stub components — the reference's dominant test
+pattern (stream/mod.rs:594-1448: StubInput, StubOutput/always-fails,
+CountingOutput, ordering, backpressure)."""
+import asyncio
+
+import pytest
+
+import arkflow_amd as af
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.config import EngineConfig, PipelineConfig, StreamConfig
+from arkflow_amd.errors import EOFError_
+from arkflow_amd.pipeline import Pipeline
+from arkflow_amd.spi import Ack, Input, NoopAck, Output, Processor
+from arkflow_amd.stream import Stream, build_stream
+
+
+class StubInput(Input):
+    """VecDeque-backed input, EOF at end (reference stream/mod.rs:594+)."""
+
+    def __init__(self, batches):
+        self.batches = list(batches)
+        self.acked = []
+
+    async def read(self):
+        if not self.batches:
+            raise EOFError_()
+        b = self.batches.pop(0)
+
+        class A(Ack):
+            def __init__(self, sink, tag):
+                self.sink, self.tag = sink, tag
+
+            async def ack(self):
+                self.sink.append(self.tag)
+
+        return b, A(self.acked, b.num_rows)
+
+
+class CountingOutput(Output):
+    def __init__(self):
+        self.rows = 0
+        self.batches = []
+
+    async def write(self, batch):
+        self.rows += batch.num_rows
+        self.batches.append(batch)
+
+
+class FailingOutput(Output):
+    """Always fails → acks withheld (reference StubOutput)."""
+
+    async def write(self, batch):
+        raise RuntimeError("sink down")
+
+
+class AddOneProcessor(Processor):
+    async def process(self, batch):
+        import torch
+        col = batch.column("v")
+        from arkflow_amd.batch import Column
+        return [batch.with_columns({"v": Column("numeric", col.data + 1)})]
+
+
+class ExplodeProcessor(Processor):
+    """Returns multiple batches (ProcessResult::Multiple)."""
+
+    async def process(self, batch):
+        return [batch, batch]
+
+
+class DropAllProcessor(Processor):
+    async def process(self, batch):
+        return []
+
+
+def _mk(v):
+    return MessageBatch.from_dict({"v": v})
+
+
+def _stream(inp, procs, out, thread_num=2, error_output=None, buffer=None):
+    cfg = StreamConfig(
+        id="t", input={"type": "memory"}, output={"type": "drop"},
+        pipeline=PipelineConfig(thread_num=thread_num),
+    )
+    return Stream(cfg, inp, Pipeline(procs), out,
+                  error_output=error_output, buffer=buffer)
+
+
+def test_end_to_end_order_and_acks(run):
+    inp = StubInput([_mk([i]) for i in range(50)])
+    out = CountingOutput()
+    s = _stream(inp, [AddOneProcessor()], out, thread_num=4)
+    run(s.run(asyncio.Event()))
+    assert out.rows == 50
+    # ordered output: values arrive in input order despite 4 workers
+    vals = [b.column("v").to_pylist()[0] for b in out.batches]
+    assert vals == [i + 1 for i in range(50)]
+    assert len(inp.acked) == 50
+
+
+def test_multiple_fanout(run):
+    inp = StubInput([_mk([1, 2])])
+    out = CountingOutput()
+    s = _stream(inp, [ExplodeProcessor(), AddOneProcessor()], out)
+    run(s.run(asyncio.Event()))
+    assert out.rows == 4  # exploded to 2 batches × 2 rows
+
+
+def test_none_result_still_acks(run):
+    inp = StubInput([_mk([1]), _mk([2])])
+    out = CountingOutput()
+    s = _stream(inp, [DropAllProcessor()], out)
+    run(s.run(asyncio.Event()))
+    assert out.rows == 0
+    assert len(inp.acked) == 2
+
+
+def test_failing_output_withholds_acks(run):
+    inp = StubInput([_mk([1]), _mk([2])])
+    out = FailingOutput()
+    s = _stream(inp, [], out)
+    run(s.run(asyncio.Event()))
+    assert inp.acked == []  # ack withheld on sink failure
+    assert s.metrics.output_errors == 2
+
+
+def test_error_routed_to_error_output(run):
+    class Boom(Processor):
+        async def process(self, batch):
+            raise ValueError("boom")
+
+    inp = StubInput([_mk([7])])
+    out = CountingOutput()
+    err_out = CountingOutput()
+    s = _stream(inp, [Boom()], out, error_output=err_out)
+    run(s.run(asyncio.Event()))
+    assert out.rows == 0
+    assert err_out.rows == 1
+    assert s.metrics.processing_errors == 1
+    assert len(inp.acked) == 1  # error path still acks after error_output
+
+
+def test_cancellation_stops_stream(run):
+    class Endless(Input):
+        async def read(self):
+            await asyncio.sleep(0.01)
+            return _mk([1]), NoopAck()
+
+    out = CountingOutput()
+    s = _stream(Endless(), [], out)
+
+    async def main():
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(s.run(cancel))
+        await asyncio.sleep(0.15)
+        cancel.set()
+        await asyncio.wait_for(task, 5)
+
+    run(main())
+    assert out.rows > 0
+
+
+def test_backpressure_bounded(run):
+    """1024-in-flight cap: a slow output must not let in-flight grow
+    unboundedly (reference stream/mod.rs:1345 notify wakeup test)."""
+    N = 300
+    inp = StubInput([_mk([i]) for i in range(N)])
+
+    class SlowOutput(CountingOutput):
+        async def write(self, batch):
+            await asyncio.sleep(0)
+            await super().write(batch)
+
+    out = SlowOutput()
+    s = _stream(inp, [], out, thread_num=8)
+    run(s.run(asyncio.Event()))
+    assert out.rows == N
+    vals = [b.column("v").to_pylist()[0] for b in out.batches]
+    assert vals == list(range(N))
+
+
+def test_engine_runs_config_to_eof(run):
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "s1",
+            "input": {"type": "generate", "count": 100, "batch_size": 10,
+                      "interval": "0ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "pipeline": {"thread_num": 2, "processors": []},
+            "output": {"type": "memory"},
+        }]
+    })
+    eng = af.Engine(cfg)
+    run(eng.run_with_cancellation())
+    entry = eng.runtime.entries["s1"]
+    assert entry.metrics.input_messages == 100
+    assert entry.metrics.output_messages == 100
+    assert entry.state.value == "stopped"
+
+
+def test_config_validation():
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "bad",
+            "input": {"type": "nope"},
+            "output": {"type": "drop"},
+        }]
+    })
+    errs = cfg.validate()
+    assert any("unknown input type" in e for e in errs)
+
+
+def test_runtime_lifecycle(run):
+    async def main():
+        cfg = EngineConfig.from_dict({
+            "streams": [{
+                "id": "s1",
+                "input": {"type": "generate", "batch_size": 5,
+                          "interval": "5ms",
+                          "fields": {"v": {"dtype": "float32"}}},
+                "output": {"type": "drop"},
+            }]
+        })
+        eng = af.Engine(cfg)
+        for sc in cfg.streams:
+            eng.runtime.register(sc)
+        await eng.runtime.start("s1")
+        assert eng.runtime.get("s1").state.value == "running"
+        await asyncio.sleep(0.05)
+        await eng.runtime.stop("s1")
+        assert eng.runtime.get("s1").state.value == "stopped"
+        await eng.runtime.restart("s1")
+        assert eng.runtime.get("s1").state.value == "running"
+        assert eng.runtime.get("s1").metrics.restarts == 1
+        await eng.runtime.stop("s1")
+        ops_kinds = {e.kind for e in eng.runtime.events.list()}
+        assert {"registered", "starting", "running", "stopping"} <= ops_kinds
+
+    run(main())
