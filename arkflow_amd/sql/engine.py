@@ -1,0 +1,397 @@
+"""SQL planner + columnar executor.
+
+Replaces DataFusion's logical/physical plan for the engine's SQL subset
+(reference processor/sql.rs:107-146 execute_query). Execution order:
+FROM/JOIN → WHERE → GROUP BY + aggregates → HAVING → SELECT projection →
+DISTINCT → ORDER BY → LIMIT. All row-level work is tensor ops routed through
+:mod:`arkflow_amd.ops` (HIP kernels on GPU, torch on CPU).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from ..batch import Column, MessageBatch
+from .eval import (
+    Env,
+    as_tensor,
+    collect_aggregates,
+    contains_aggregate,
+    eval_expr,
+    expr_name,
+)
+from .parser import (
+    BinaryOp,
+    ColumnRef,
+    FuncCall,
+    Literal,
+    Select,
+    SqlError,
+    Star,
+    parse_sql,
+)
+
+DEFAULT_TABLE = "flow"  # reference processor/sql.rs registers the batch as `flow`
+
+
+class SqlExecutor:
+    """Pre-parsed, reusable statement (reference pre-parses once, sql.rs:189)."""
+
+    def __init__(self, sql: str):
+        self.sql = sql
+        self.select: Select = parse_sql(sql)
+        self._aggs: List[FuncCall] = []
+        for e, _ in self.select.projections:
+            collect_aggregates(e, self._aggs)
+        if self.select.having is not None:
+            collect_aggregates(self.select.having, self._aggs)
+        # dedup by expression text
+        seen = {}
+        for a in self._aggs:
+            seen.setdefault(expr_name(a), a)
+        self._aggs = list(seen.values())
+        self.is_aggregate = bool(self._aggs or self.select.group_by)
+
+    # ------------------------------------------------------------------- run
+    def execute(self, tables: Dict[str, MessageBatch]) -> MessageBatch:
+        sel = self.select
+        from_name = sel.from_table or DEFAULT_TABLE
+        if from_name not in tables:
+            raise SqlError(f"unknown table {from_name!r}; have {sorted(tables)}")
+        base = tables[from_name]
+        device = base.device
+
+        columns: Dict[str, Column] = dict(base.columns)
+        alias = sel.from_alias or from_name
+        for k, v in base.columns.items():
+            columns[f"{alias}.{k}"] = v
+        order: List[str] = list(base.columns.keys())
+
+        # ------------------------------------------------------------- joins
+        for j in sel.joins:
+            right = tables.get(j.table)
+            if right is None:
+                raise SqlError(f"unknown join table {j.table!r}")
+            columns, order = self._apply_join(
+                columns, order, alias, j, right, device)
+
+        n_rows = len(columns[order[0]]) if order else 0
+        env = Env(columns, n_rows, device)
+
+        # ------------------------------------------------------------- where
+        if sel.where is not None:
+            idx = self._filter_indices(sel.where, env)
+            columns = {k: c.take(idx) for k, c in columns.items()}
+            n_rows = int(idx.shape[0])
+            env = Env(columns, n_rows, device)
+
+        # --------------------------------------------------------- aggregate
+        if self.is_aggregate:
+            env, order = self._aggregate(env, order)
+
+        # ------------------------------------------------------------ having
+        if sel.having is not None:
+            mask = as_tensor(eval_expr(sel.having, env), env).bool()
+            idx = ops.mask_to_indices(mask).long()
+            cols = {k: c.take(idx) for k, c in env.columns.items()}
+            aggs = {k: v[idx] for k, v in env.agg_results.items()}
+            env = Env(cols, int(idx.shape[0]), device, aggs)
+
+        # -------------------------------------------------------- projection
+        out_cols: Dict[str, Column] = {}
+        for e, alias_name in sel.projections:
+            if isinstance(e, Star):
+                for name in order:
+                    if e.table and not name.startswith(f"{e.table}."):
+                        continue
+                    out_name = name.split(".", 1)[1] if e.table else name
+                    if out_name not in out_cols:
+                        out_cols[out_name] = env.columns[name]
+                continue
+            v = eval_expr(e, env)
+            name = alias_name or expr_name(e)
+            out_cols[name] = _to_column(v, env)
+        result = MessageBatch(out_cols, input_name=None)
+
+        # ---------------------------------------------------------- distinct
+        if sel.distinct:
+            gid, _, g = _encode_keys(
+                [c if c.kind == "binary" else c.data
+                 for c in result.columns.values()], device)
+            rep = _first_index_per_group(gid, g)
+            result = result.take(rep)
+
+        # ---------------------------------------------------------- order by
+        if sel.order_by:
+            final_env = Env(
+                {**env.columns, **result.columns},
+                result.num_rows, device, env.agg_results)
+            # multi-key sort: stable sorts applied last-key-first
+            for e, asc in reversed(sel.order_by):
+                v = eval_expr(e, final_env)
+                key = v.data if isinstance(v, Column) and v.kind == "numeric" \
+                    else v
+                if isinstance(key, Column):
+                    # binary sort key: CPU fallback
+                    import numpy as np
+                    vals = key.to_pylist()
+                    idx = torch.tensor(
+                        sorted(range(len(vals)), key=lambda i: vals[i],
+                               reverse=not asc),
+                        dtype=torch.int64, device=device)
+                else:
+                    idx = ops.sort_indices(key, ascending=asc)
+                result = result.take(idx)
+                final_env = Env(
+                    {**{k: c.take(idx) for k, c in env.columns.items()},
+                     **result.columns},
+                    result.num_rows, device, env.agg_results)
+
+        # ------------------------------------------------------------- limit
+        if sel.limit is not None and result.num_rows > sel.limit:
+            result = result.slice(0, sel.limit)
+        return result
+
+    # --------------------------------------------------------------- filters
+    def _filter_indices(self, pred, env: Env) -> torch.Tensor:
+        """WHERE → row indices. Fast path: `col OP numeric-literal` fuses
+        compare+compact in one HIP kernel (csrc/filter.hip)."""
+        if isinstance(pred, BinaryOp) and pred.op in (
+                "<", "<=", ">", ">=", "=", "!="):
+            l, r = pred.left, pred.right
+            if isinstance(l, ColumnRef) and isinstance(r, Literal) \
+                    and isinstance(r.value, (int, float)):
+                col = env.lookup(l)
+                if col.kind == "numeric" and col.data.dtype in (
+                        torch.float32, torch.int64, torch.int32):
+                    return ops.filter_cmp_scalar(
+                        col.data, pred.op, float(r.value)).long()
+        mask = as_tensor(eval_expr(pred, env), env).bool()
+        return ops.mask_to_indices(mask).long()
+
+    # ------------------------------------------------------------------ join
+    def _apply_join(self, columns, order, left_alias, j, right: MessageBatch,
+                    device):
+        r_alias = j.alias or j.table
+        conjuncts = _split_and(j.on)
+        key_pairs = []
+        residual = []
+        for c in conjuncts:
+            if (isinstance(c, BinaryOp) and c.op == "=" and
+                    isinstance(c.left, ColumnRef) and
+                    isinstance(c.right, ColumnRef)):
+                key_pairs.append((c.left, c.right))
+            else:
+                residual.append(c)
+        if not key_pairs:
+            raise SqlError("JOIN requires at least one equality condition")
+
+        def side_of(ref: ColumnRef):
+            if ref.table == r_alias or (
+                    ref.table is None and ref.name in right.columns
+                    and ref.name not in columns):
+                return "right"
+            return "left"
+
+        l_keys, r_keys = [], []
+        n_left = len(columns[order[0]])
+        l_env = Env(columns, n_left, device)
+        r_cols_q = {f"{r_alias}.{k}": v for k, v in right.columns.items()}
+        r_env = Env({**right.columns, **r_cols_q}, right.num_rows, device)
+        for a, b in key_pairs:
+            if side_of(a) == "right":
+                a, b = b, a
+            l_keys.append(as_tensor(eval_expr(a, l_env), l_env))
+            r_keys.append(as_tensor(eval_expr(b, r_env), r_env))
+        lk, _, _ = (_encode_keys(l_keys, device) if len(l_keys) > 1
+                    else (l_keys[0], None, 0))
+        rk = r_keys[0] if len(r_keys) == 1 else None
+        if len(l_keys) > 1:
+            # co-encode both sides so hashes align
+            lk, rk = _co_encode(l_keys, r_keys, device)
+        lk = lk.to(torch.int64) if lk.dtype not in (
+            torch.int64, torch.float32) else lk
+        rk = rk.to(lk.dtype) if rk.dtype != lk.dtype else rk
+        if j.kind == "left":
+            l_idx, r_idx = ops.join_left(lk, rk)
+        else:
+            l_idx, r_idx = ops.join_inner(lk, rk)
+
+        new_cols: Dict[str, Column] = {}
+        new_order: List[str] = []
+        for name in order:
+            c = columns[name].take(l_idx)
+            new_cols[name] = c
+            new_cols[f"{left_alias}.{name.split('.', 1)[-1]}"] = c
+            new_order.append(name)
+        safe_r = r_idx.clamp(min=0)
+        null_rows = r_idx < 0
+        for name, col in right.columns.items():
+            taken = col.take(safe_r)
+            if bool(null_rows.any()):
+                validity = ~null_rows
+                taken = Column(taken.kind, taken.data, taken.offsets, validity)
+            qual = f"{r_alias}.{name}"
+            new_cols[qual] = taken
+            if name not in new_cols:  # unqualified only when unambiguous
+                new_cols[name] = taken
+                new_order.append(name)
+            else:
+                new_order.append(qual)
+        if residual:
+            n = int(l_idx.shape[0])
+            env = Env(new_cols, n, device)
+            mask = torch.ones(n, dtype=torch.bool, device=device)
+            for c in residual:
+                mask &= as_tensor(eval_expr(c, env), env).bool()
+            idx = ops.mask_to_indices(mask).long()
+            new_cols = {k: c.take(idx) for k, c in new_cols.items()}
+        return new_cols, new_order
+
+    # ------------------------------------------------------------- aggregate
+    def _aggregate(self, env: Env, order: List[str]
+                   ) -> Tuple[Env, List[str]]:
+        sel = self.select
+        device = env.device
+        n = env.n_rows
+        if sel.group_by:
+            key_vals = [eval_expr(e, env) for e in sel.group_by]
+            keys = [v if isinstance(v, Column) else as_tensor(v, env)
+                    for v in key_vals]
+            single = keys[0]
+            if len(keys) == 1 and isinstance(single, torch.Tensor) \
+                    and single.dtype in (torch.int64, torch.int32):
+                gid, _, g = ops.hash_group(single)
+                gid = gid.long()
+            else:
+                gid, _, g = _encode_keys(keys, device)
+        else:
+            gid = torch.zeros(n, dtype=torch.int64, device=device)
+            g = 1 if n > 0 else 1  # global aggregate yields one row even if empty
+
+        agg_results: Dict[str, torch.Tensor] = {}
+        for a in self._aggs:
+            agg_results[expr_name(a)] = self._compute_agg(a, env, gid, g)
+
+        # representative row per group for key/non-agg column access
+        if n > 0:
+            rep = _first_index_per_group(gid, g)
+            cols = {k: c.take(rep) for k, c in env.columns.items()}
+        else:
+            cols = {k: c.slice(0, 0) for k, c in env.columns.items()}
+            if not sel.group_by:
+                # 1-row global aggregate over empty input: null keys
+                cols = {k: _null_like(c, 1) for k, c in env.columns.items()}
+        return Env(cols, g, device, agg_results), order
+
+    def _compute_agg(self, a: FuncCall, env: Env, gid: torch.Tensor, g: int
+                     ) -> torch.Tensor:
+        name = a.name
+        device = env.device
+        if env.n_rows == 0:
+            if name == "count":
+                return torch.zeros(g, dtype=torch.int64, device=device)
+            return torch.full((g,), float("nan"), device=device)
+        if name == "count" and (not a.args or isinstance(a.args[0], Star)):
+            return ops.segment_reduce(
+                torch.ones(env.n_rows, device=device), gid, g, "count")
+        arg = eval_expr(a.args[0], env)
+        if name == "count":
+            if a.distinct:
+                key = arg.data if isinstance(arg, Column) else arg
+                pair_gid, _, _ = _encode_keys([gid, key], device)
+                # distinct (gid, val) pairs per gid
+                uniq_pair = _first_index_per_group(
+                    pair_gid, int(pair_gid.max().item()) + 1)
+                sub_gid = gid[uniq_pair]
+                return ops.segment_reduce(
+                    torch.ones(sub_gid.shape[0], device=device),
+                    sub_gid, g, "count")
+            if isinstance(arg, Column) and arg.validity is not None:
+                return ops.segment_reduce(
+                    arg.validity.to(torch.float32), gid, g, "sum"
+                ).to(torch.int64)
+            return ops.segment_reduce(
+                torch.ones(env.n_rows, device=device), gid, g, "count")
+        vals = as_tensor(arg, env)
+        if name == "avg":
+            return ops.segment_reduce(vals, gid, g, "mean")
+        out = ops.segment_reduce(vals, gid, g, name)  # sum|min|max
+        if not vals.dtype.is_floating_point and name in ("sum", "min", "max"):
+            out = out.to(torch.int64)
+        elif vals.dtype == torch.float32 and out.dtype == torch.float64:
+            out = out  # keep f64 accumulations for numeric fidelity
+        return out
+
+
+# -------------------------------------------------------------------- helpers
+def _split_and(e) -> list:
+    if isinstance(e, BinaryOp) and e.op == "and":
+        return _split_and(e.left) + _split_and(e.right)
+    return [e]
+
+
+def _to_column(v, env: Env) -> Column:
+    if isinstance(v, Column):
+        return v
+    if isinstance(v, torch.Tensor):
+        return Column("numeric", v)
+    if isinstance(v, str):
+        return Column.from_strings([v] * env.n_rows)
+    # scalar broadcast
+    return Column("numeric", as_tensor(v, env))
+
+
+def _encode_keys(keys: list, device) -> Tuple[torch.Tensor, None, int]:
+    """Encode 1..k key columns (tensors or binary Columns) into group ids."""
+    if any(isinstance(k, Column) and k.kind == "binary" for k in keys):
+        # dictionary-encode binary keys host-side (GPU-native path: round 2)
+        lists = []
+        for k in keys:
+            if isinstance(k, Column):
+                lists.append(k.to_pylist())
+            else:
+                lists.append(k.detach().cpu().tolist())
+        seen: dict = {}
+        gids = []
+        for row in zip(*lists):
+            gids.append(seen.setdefault(row, len(seen)))
+        return (torch.tensor(gids, dtype=torch.int64, device=device), None,
+                len(seen))
+    ts = []
+    all_int = all(not k.dtype.is_floating_point for k in keys)
+    for k in keys:
+        ts.append(k.to(torch.int64) if all_int else k.to(torch.float64))
+    stacked = torch.stack(ts, dim=1)
+    uniq, inverse = torch.unique(stacked, dim=0, return_inverse=True)
+    return inverse.to(torch.int64), None, int(uniq.shape[0])
+
+
+def _co_encode(l_keys, r_keys, device):
+    n_l = l_keys[0].shape[0]
+    combined = [torch.cat([l.to(torch.float64), r.to(torch.float64)])
+                for l, r in zip(l_keys, r_keys)]
+    gid, _, _ = _encode_keys(combined, device)
+    return gid[:n_l], gid[n_l:]
+
+
+def _first_index_per_group(gid: torch.Tensor, g: int) -> torch.Tensor:
+    n = gid.shape[0]
+    idx = torch.arange(n, dtype=torch.int64, device=gid.device)
+    first = torch.full((g,), n, dtype=torch.int64, device=gid.device)
+    first.scatter_reduce_(0, gid.long(), idx, reduce="amin", include_self=True)
+    return first
+
+
+def _null_like(c: Column, n: int) -> Column:
+    import torch as _t
+    validity = _t.zeros(n, dtype=_t.bool)
+    if c.kind == "numeric":
+        return Column("numeric",
+                      _t.zeros(n, dtype=c.data.dtype, device=c.data.device),
+                      validity=validity.to(c.data.device))
+    off = _t.zeros(n + 1, dtype=_t.int64, device=c.data.device)
+    return Column("binary", _t.empty(0, dtype=_t.uint8, device=c.data.device),
+                  off, validity.to(c.data.device))

@@ -1,0 +1,456 @@
+"""Columnar expression evaluator.
+
+Evaluates parsed SQL expressions over an environment of named columns.
+Numeric/bool expressions are torch tensors (device-resident on GPU — each
+operator is a kernel on the batch's device); string/binary values stay
+:class:`~arkflow_amd.batch.Column` (binary) with vectorized compare paths.
+
+Replaces DataFusion's PhysicalExpr evaluation (reference expr/mod.rs:29-210
+cached-PhysicalExpr + processor/sql.rs execution).
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional, Union
+
+import numpy as np
+import torch
+
+from ..batch import Column
+from .parser import (
+    AGGREGATE_FUNCS,
+    Between,
+    BinaryOp,
+    Case,
+    Cast,
+    ColumnRef,
+    FuncCall,
+    InList,
+    IsNull,
+    Like,
+    Literal,
+    SqlError,
+    Star,
+    UnaryOp,
+)
+
+Value = Union[torch.Tensor, Column, int, float, str, bool, None]
+
+
+class Env:
+    """Column environment: unqualified + qualified names, plus precomputed
+    aggregate results keyed by expression text."""
+
+    def __init__(self, columns: Dict[str, Column], n_rows: int,
+                 device: torch.device,
+                 agg_results: Optional[Dict[str, torch.Tensor]] = None):
+        self.columns = columns
+        self.n_rows = n_rows
+        self.device = device
+        self.agg_results = agg_results or {}
+
+    def lookup(self, ref: ColumnRef) -> Column:
+        if ref.table:
+            qual = f"{ref.table}.{ref.name}"
+            if qual in self.columns:
+                return self.columns[qual]
+        if ref.name in self.columns:
+            return self.columns[ref.name]
+        raise SqlError(f"unknown column {ref.table + '.' if ref.table else ''}"
+                       f"{ref.name}; have {sorted(self.columns)}")
+
+
+def expr_name(e) -> str:
+    """Readable output-column name for an unaliased projection."""
+    if isinstance(e, ColumnRef):
+        return e.name
+    if isinstance(e, Literal):
+        return repr(e.value)
+    if isinstance(e, FuncCall):
+        args = ", ".join(expr_name(a) for a in e.args)
+        return f"{e.name}({'DISTINCT ' if e.distinct else ''}{args})"
+    if isinstance(e, Star):
+        return "*"
+    if isinstance(e, BinaryOp):
+        return f"{expr_name(e.left)} {e.op} {expr_name(e.right)}"
+    if isinstance(e, UnaryOp):
+        return f"{e.op} {expr_name(e.operand)}"
+    if isinstance(e, Cast):
+        return f"cast({expr_name(e.expr)} as {e.to_type})"
+    if isinstance(e, Case):
+        return "case"
+    if isinstance(e, Between):
+        return f"{expr_name(e.expr)} between"
+    if isinstance(e, InList):
+        return f"{expr_name(e.expr)} in"
+    if isinstance(e, IsNull):
+        return f"{expr_name(e.expr)} is null"
+    if isinstance(e, Like):
+        return f"{expr_name(e.expr)} like"
+    return "expr"
+
+
+def contains_aggregate(e) -> bool:
+    if isinstance(e, FuncCall):
+        if e.name in AGGREGATE_FUNCS:
+            return True
+        return any(contains_aggregate(a) for a in e.args)
+    if isinstance(e, BinaryOp):
+        return contains_aggregate(e.left) or contains_aggregate(e.right)
+    if isinstance(e, UnaryOp):
+        return contains_aggregate(e.operand)
+    if isinstance(e, Cast):
+        return contains_aggregate(e.expr)
+    if isinstance(e, Case):
+        return any(contains_aggregate(c) or contains_aggregate(v)
+                   for c, v in e.whens) or (
+            e.else_ is not None and contains_aggregate(e.else_))
+    if isinstance(e, (Between,)):
+        return any(contains_aggregate(x) for x in (e.expr, e.low, e.high))
+    if isinstance(e, InList):
+        return contains_aggregate(e.expr)
+    if isinstance(e, (IsNull, Like)):
+        return contains_aggregate(e.expr)
+    return False
+
+
+def collect_aggregates(e, out: list) -> None:
+    """Find aggregate FuncCall nodes (outermost) in an expression tree."""
+    if isinstance(e, FuncCall):
+        if e.name in AGGREGATE_FUNCS:
+            out.append(e)
+            return
+        for a in e.args:
+            collect_aggregates(a, out)
+        return
+    if isinstance(e, BinaryOp):
+        collect_aggregates(e.left, out)
+        collect_aggregates(e.right, out)
+    elif isinstance(e, UnaryOp):
+        collect_aggregates(e.operand, out)
+    elif isinstance(e, Cast):
+        collect_aggregates(e.expr, out)
+    elif isinstance(e, Case):
+        for c, v in e.whens:
+            collect_aggregates(c, out)
+            collect_aggregates(v, out)
+        if e.else_ is not None:
+            collect_aggregates(e.else_, out)
+    elif isinstance(e, Between):
+        for x in (e.expr, e.low, e.high):
+            collect_aggregates(x, out)
+    elif isinstance(e, (InList, IsNull, Like)):
+        collect_aggregates(e.expr, out)
+
+
+# ------------------------------------------------------------------ broadcast
+def as_tensor(v: Value, env: Env) -> torch.Tensor:
+    if isinstance(v, torch.Tensor):
+        return v
+    if isinstance(v, Column):
+        if v.kind == "numeric":
+            return v.data
+        raise SqlError("binary column used in numeric context")
+    if v is None:
+        return torch.full((env.n_rows,), float("nan"), device=env.device)
+    if isinstance(v, bool):
+        return torch.full((env.n_rows,), v, dtype=torch.bool, device=env.device)
+    if isinstance(v, int):
+        return torch.full((env.n_rows,), v, dtype=torch.int64, device=env.device)
+    if isinstance(v, float):
+        return torch.full((env.n_rows,), v, dtype=torch.float64,
+                          device=env.device)
+    raise SqlError(f"cannot use {type(v).__name__} in numeric context")
+
+
+def _binary_bytes(col: Column) -> list:
+    return col.to_pylist()
+
+
+def _bin_eq_literal(col: Column, lit: bytes, device) -> torch.Tensor:
+    """Vectorized bytes == literal: length check + content gather compare."""
+    lengths = col.offsets[1:] - col.offsets[:-1]
+    L = len(lit)
+    mask = lengths == L
+    if L == 0 or not bool(mask.any()):
+        return mask
+    cand = torch.nonzero(mask).flatten()
+    starts = col.offsets[:-1][cand]
+    pos = starts.unsqueeze(1) + torch.arange(L, device=col.data.device)
+    litt = torch.from_numpy(
+        np.frombuffer(lit, dtype=np.uint8).copy()).to(col.data.device)
+    eq = (col.data[pos] == litt).all(dim=1)
+    out = torch.zeros_like(mask)
+    out[cand] = eq
+    return out
+
+
+def _bin_eq_col(a: Column, b: Column) -> torch.Tensor:
+    la = a.offsets[1:] - a.offsets[:-1]
+    lb = b.offsets[1:] - b.offsets[:-1]
+    mask = la == lb
+    if not bool(mask.any()):
+        return mask
+    # fallback per-row compare for candidates (vectorize later if hot)
+    av, bv = a.to_pylist(), b.to_pylist()
+    res = [x == y for x, y in zip(av, bv)]
+    return torch.tensor(res, dtype=torch.bool, device=a.data.device)
+
+
+def eval_expr(e, env: Env) -> Value:
+    if isinstance(e, Literal):
+        return e.value
+    if isinstance(e, ColumnRef):
+        col = env.lookup(e)
+        return col if col.kind == "binary" else col.data
+    if isinstance(e, Star):
+        raise SqlError("* only allowed in COUNT(*) or projection list")
+    if isinstance(e, FuncCall):
+        if e.name in AGGREGATE_FUNCS:
+            key = expr_name(e)
+            if key in env.agg_results:
+                return env.agg_results[key]
+            raise SqlError(f"aggregate {key} outside aggregate context")
+        return _eval_func(e, env)
+    if isinstance(e, UnaryOp):
+        if e.op == "not":
+            v = as_tensor(eval_expr(e.operand, env), env)
+            return ~v.bool()
+        v = as_tensor(eval_expr(e.operand, env), env)
+        return -v
+    if isinstance(e, BinaryOp):
+        return _eval_binop(e, env)
+    if isinstance(e, Cast):
+        v = eval_expr(e.expr, env)
+        return _eval_cast(v, e.to_type, env)
+    if isinstance(e, Case):
+        result = None
+        for cond, val in reversed(list(e.whens)):
+            c = as_tensor(eval_expr(cond, env), env).bool()
+            v = as_tensor(eval_expr(val, env), env)
+            if result is None:
+                if e.else_ is not None:
+                    result = as_tensor(eval_expr(e.else_, env), env)
+                else:
+                    result = torch.full(
+                        (env.n_rows,), float("nan"), device=env.device)
+            result, v = _promote(result, v)
+            result = torch.where(c, v, result)
+        return result
+    if isinstance(e, Between):
+        v = as_tensor(eval_expr(e.expr, env), env)
+        lo = as_tensor(eval_expr(e.low, env), env)
+        hi = as_tensor(eval_expr(e.high, env), env)
+        r = (v >= lo) & (v <= hi)
+        return ~r if e.negated else r
+    if isinstance(e, InList):
+        v = eval_expr(e.expr, env)
+        if isinstance(v, Column) and v.kind == "binary":
+            r = torch.zeros(env.n_rows, dtype=torch.bool, device=env.device)
+            for item in e.items:
+                lit = eval_expr(item, env)
+                if isinstance(lit, str):
+                    lit = lit.encode()
+                r |= _bin_eq_literal(v, lit, env.device)
+        else:
+            t = as_tensor(v, env)
+            r = torch.zeros_like(t, dtype=torch.bool)
+            for item in e.items:
+                r |= (t == as_tensor(eval_expr(item, env), env))
+        return ~r if e.negated else r
+    if isinstance(e, IsNull):
+        v = eval_expr(e.expr, env)
+        if isinstance(v, Column) and v.validity is not None:
+            r = ~v.validity
+        elif isinstance(v, torch.Tensor) and v.is_floating_point():
+            r = torch.isnan(v)
+        else:
+            r = torch.zeros(env.n_rows, dtype=torch.bool, device=env.device)
+        return ~r if e.negated else r
+    if isinstance(e, Like):
+        v = eval_expr(e.expr, env)
+        if not isinstance(v, Column) or v.kind != "binary":
+            raise SqlError("LIKE requires a string column")
+        pat = e.pattern
+        vals = v.to_pylist()
+        if pat.startswith("%") and pat.endswith("%") and len(pat) > 1:
+            needle = pat[1:-1].encode()
+            res = [needle in x for x in vals]
+        elif pat.endswith("%"):
+            p = pat[:-1].encode()
+            res = [x.startswith(p) for x in vals]
+        elif pat.startswith("%"):
+            p = pat[1:].encode()
+            res = [x.endswith(p) for x in vals]
+        else:
+            res = [x == pat.encode() for x in vals]
+        r = torch.tensor(res, dtype=torch.bool, device=env.device)
+        return ~r if e.negated else r
+    raise SqlError(f"cannot evaluate {type(e).__name__}")
+
+
+def _promote(a: torch.Tensor, b: torch.Tensor):
+    dt = torch.promote_types(a.dtype, b.dtype)
+    return a.to(dt), b.to(dt)
+
+
+def _eval_binop(e: BinaryOp, env: Env) -> Value:
+    if e.op in ("and", "or"):
+        l = as_tensor(eval_expr(e.left, env), env).bool()
+        r = as_tensor(eval_expr(e.right, env), env).bool()
+        return (l & r) if e.op == "and" else (l | r)
+    lv = eval_expr(e.left, env)
+    rv = eval_expr(e.right, env)
+    # string comparison / concat paths
+    l_bin = isinstance(lv, Column) and lv.kind == "binary"
+    r_bin = isinstance(rv, Column) and rv.kind == "binary"
+    if e.op == "||":
+        if l_bin or r_bin or isinstance(lv, str) or isinstance(rv, str):
+            ls = lv.to_strlist() if l_bin else [str(lv)] * env.n_rows
+            rs = rv.to_strlist() if r_bin else [str(rv)] * env.n_rows
+            return Column.from_strings([a + b for a, b in zip(ls, rs)])
+        raise SqlError("|| requires strings")
+    if l_bin or r_bin or isinstance(lv, str) or isinstance(rv, str):
+        if e.op not in ("=", "!="):
+            # ordered string compare: CPU fallback
+            ls = lv.to_pylist() if l_bin else [
+                (lv.encode() if isinstance(lv, str) else lv)] * env.n_rows
+            rs = rv.to_pylist() if r_bin else [
+                (rv.encode() if isinstance(rv, str) else rv)] * env.n_rows
+            fn = {"<": lambda a, b: a < b, "<=": lambda a, b: a <= b,
+                  ">": lambda a, b: a > b, ">=": lambda a, b: a >= b}[e.op]
+            return torch.tensor([fn(a, b) for a, b in zip(ls, rs)],
+                                dtype=torch.bool, device=env.device)
+        if l_bin and isinstance(rv, str):
+            r = _bin_eq_literal(lv, rv.encode(), env.device)
+        elif r_bin and isinstance(lv, str):
+            r = _bin_eq_literal(rv, lv.encode(), env.device)
+        elif l_bin and r_bin:
+            r = _bin_eq_col(lv, rv)
+        else:
+            raise SqlError("string compared with non-string")
+        return ~r if e.op == "!=" else r
+    l = as_tensor(lv, env)
+    r = as_tensor(rv, env)
+    if e.op in ("=", "!=", "<", "<=", ">", ">="):
+        l, r = _promote(l, r)
+        return {
+            "=": l == r, "!=": l != r, "<": l < r,
+            "<=": l <= r, ">": l > r, ">=": l >= r,
+        }[e.op]
+    l, r = _promote(l, r)
+    if e.op == "+":
+        return l + r
+    if e.op == "-":
+        return l - r
+    if e.op == "*":
+        return l * r
+    if e.op == "/":
+        if l.dtype.is_floating_point or r.dtype.is_floating_point:
+            return l / r
+        return torch.div(l, r, rounding_mode="trunc")
+    if e.op == "%":
+        return torch.remainder(l, r)
+    raise SqlError(f"unknown operator {e.op}")
+
+
+_CAST_TYPES = {
+    "int": torch.int64, "integer": torch.int64, "bigint": torch.int64,
+    "smallint": torch.int16, "tinyint": torch.int8,
+    "float": torch.float32, "real": torch.float32,
+    "double": torch.float64, "bool": torch.bool, "boolean": torch.bool,
+    "bf16": torch.bfloat16, "bfloat16": torch.bfloat16,
+}
+
+
+def _eval_cast(v: Value, to_type: str, env: Env) -> Value:
+    ty = to_type.lower()
+    if ty in ("varchar", "text", "string"):
+        if isinstance(v, Column) and v.kind == "binary":
+            return v
+        t = as_tensor(v, env)
+        vals = t.detach().cpu().tolist()
+        return Column.from_strings([
+            str(int(x)) if isinstance(x, float) and x.is_integer()
+            and not t.dtype.is_floating_point else str(x) for x in vals
+        ])
+    if ty not in _CAST_TYPES:
+        raise SqlError(f"unknown cast type {to_type!r}")
+    dt = _CAST_TYPES[ty]
+    if isinstance(v, Column) and v.kind == "binary":
+        vals = v.to_strlist()
+        if dt.is_floating_point:
+            return torch.tensor([float(x) for x in vals], dtype=dt,
+                                device=env.device)
+        return torch.tensor([int(float(x)) for x in vals], dtype=torch.int64,
+                            device=env.device).to(dt)
+    t = as_tensor(v, env)
+    if not dt.is_floating_point and t.dtype.is_floating_point:
+        t = t.trunc()
+    return t.to(dt)
+
+
+_UNARY_MATH = {
+    "abs": torch.abs, "sqrt": torch.sqrt, "floor": torch.floor,
+    "ceil": torch.ceil, "exp": torch.exp, "ln": torch.log,
+    "log10": torch.log10, "log2": torch.log2, "sin": torch.sin,
+    "cos": torch.cos, "tan": torch.tan, "sign": torch.sign,
+}
+
+
+def _eval_func(e: FuncCall, env: Env) -> Value:
+    name = e.name
+    if name in _UNARY_MATH:
+        v = as_tensor(eval_expr(e.args[0], env), env)
+        return _UNARY_MATH[name](v.double() if not v.dtype.is_floating_point
+                                 else v)
+    if name == "round":
+        v = as_tensor(eval_expr(e.args[0], env), env).double()
+        digits = 0
+        if len(e.args) > 1:
+            digits = int(eval_expr(e.args[1], env))
+        return torch.round(v, decimals=digits)
+    if name in ("pow", "power"):
+        b = as_tensor(eval_expr(e.args[0], env), env).double()
+        p = as_tensor(eval_expr(e.args[1], env), env).double()
+        return torch.pow(b, p)
+    if name == "coalesce":
+        result = None
+        for a in e.args:
+            v = as_tensor(eval_expr(a, env), env)
+            if result is None:
+                result = v
+            else:
+                result, v = _promote(result, v)
+                result = torch.where(torch.isnan(result) if
+                                     result.is_floating_point() else
+                                     torch.zeros_like(result, dtype=torch.bool),
+                                     v, result)
+        return result
+    if name in ("length", "char_length", "octet_length"):
+        v = eval_expr(e.args[0], env)
+        if isinstance(v, Column) and v.kind == "binary":
+            return (v.offsets[1:] - v.offsets[:-1]).to(torch.int64)
+        raise SqlError("length() requires a string column")
+    if name in ("upper", "lower"):
+        v = eval_expr(e.args[0], env)
+        if isinstance(v, Column) and v.kind == "binary":
+            vals = v.to_strlist()
+            f = str.upper if name == "upper" else str.lower
+            return Column.from_strings([f(x) for x in vals])
+        if isinstance(v, str):
+            return v.upper() if name == "upper" else v.lower()
+        raise SqlError(f"{name}() requires a string")
+    if name == "concat":
+        parts = [eval_expr(a, env) for a in e.args]
+        lists = []
+        for p in parts:
+            if isinstance(p, Column) and p.kind == "binary":
+                lists.append(p.to_strlist())
+            else:
+                lists.append([str(p)] * env.n_rows)
+        return Column.from_strings(["".join(t) for t in zip(*lists)])
+    if name == "nullif":
+        a = as_tensor(eval_expr(e.args[0], env), env).double()
+        b = as_tensor(eval_expr(e.args[1], env), env).double()
+        return torch.where(a == b, torch.full_like(a, float("nan")), a)
+    raise SqlError(f"unknown function {name}()")

@@ -1,0 +1,180 @@
+"""SQL engine tests: parser + executor semantics (CPU; the same executor runs
+HIP kernels on GPU — see test_gpu_kernels.py for device parity)."""
+import math
+
+import pytest
+import torch
+
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.sql.engine import SqlExecutor
+from arkflow_amd.sql.parser import SqlError, parse_sql
+
+
+def q(sql, **tables):
+    return SqlExecutor(sql).execute(tables)
+
+
+@pytest.fixture
+def flow():
+    return MessageBatch.from_dict({
+        "id": [1, 2, 3, 4, 5, 6],
+        "value": [10.0, 20.0, 5.0, 40.0, 15.0, 20.0],
+        "grp": [1, 1, 2, 2, 3, 3],
+        "name": ["alpha", "beta", "gamma", "alpha", "delta", "beta"],
+    })
+
+
+def test_select_star_where(flow):
+    r = q("SELECT * FROM flow WHERE value >= 15", flow=flow)
+    assert r.column("id").to_pylist() == [2, 4, 5, 6]
+    assert r.column_names == ["id", "value", "grp", "name"]
+
+
+def test_projection_arithmetic_alias(flow):
+    r = q("SELECT id, value * 2 + 1 AS v2 FROM flow WHERE id <= 2", flow=flow)
+    assert r.column("v2").to_pylist() == [21.0, 41.0]
+
+
+def test_where_and_or_not_between_in(flow):
+    r = q("SELECT id FROM flow WHERE (value > 10 AND grp = 1) OR id = 5",
+          flow=flow)
+    assert r.column("id").to_pylist() == [2, 5]
+    r = q("SELECT id FROM flow WHERE value BETWEEN 10 AND 20", flow=flow)
+    assert r.column("id").to_pylist() == [1, 2, 5, 6]
+    r = q("SELECT id FROM flow WHERE id IN (1, 3, 6)", flow=flow)
+    assert r.column("id").to_pylist() == [1, 3, 6]
+    r = q("SELECT id FROM flow WHERE NOT value < 15", flow=flow)
+    assert r.column("id").to_pylist() == [2, 4, 5, 6]
+
+
+def test_string_predicates(flow):
+    r = q("SELECT id FROM flow WHERE name = 'alpha'", flow=flow)
+    assert r.column("id").to_pylist() == [1, 4]
+    r = q("SELECT id FROM flow WHERE name LIKE '%et%'", flow=flow)
+    assert r.column("id").to_pylist() == [2, 6]
+    r = q("SELECT id, length(name) AS l FROM flow WHERE name != 'beta'",
+          flow=flow)
+    assert r.column("l").to_pylist() == [5, 5, 5, 5]
+
+
+def test_global_aggregates(flow):
+    r = q("SELECT count(*) AS c, sum(value) AS s, avg(value) AS a, "
+          "min(value) AS lo, max(value) AS hi FROM flow", flow=flow)
+    assert r.num_rows == 1
+    assert r.column("c").to_pylist() == [6]
+    assert r.column("s").to_pylist() == [110.0]
+    assert abs(r.column("a").to_pylist()[0] - 110.0 / 6) < 1e-9
+    assert r.column("lo").to_pylist() == [5.0]
+    assert r.column("hi").to_pylist() == [40.0]
+
+
+def test_group_by(flow):
+    r = q("SELECT grp, sum(value) AS s, count(*) AS c FROM flow "
+          "GROUP BY grp ORDER BY grp", flow=flow)
+    assert r.column("grp").to_pylist() == [1, 2, 3]
+    assert r.column("s").to_pylist() == [30.0, 45.0, 35.0]
+    assert r.column("c").to_pylist() == [2, 2, 2]
+
+
+def test_group_by_string_key(flow):
+    r = q("SELECT name, count(*) AS c FROM flow GROUP BY name "
+          "ORDER BY c DESC, name", flow=flow)
+    names = r.column("name").to_strlist()
+    assert names[:2] == ["alpha", "beta"]
+    assert r.column("c").to_pylist() == [2, 2, 1, 1]
+
+
+def test_having(flow):
+    r = q("SELECT grp, sum(value) AS s FROM flow GROUP BY grp "
+          "HAVING sum(value) > 30 ORDER BY grp", flow=flow)
+    assert r.column("grp").to_pylist() == [2, 3]
+
+
+def test_count_distinct(flow):
+    r = q("SELECT count(DISTINCT grp) AS g FROM flow", flow=flow)
+    assert r.column("g").to_pylist() == [3]
+
+
+def test_order_limit(flow):
+    r = q("SELECT id, value FROM flow ORDER BY value DESC, id ASC LIMIT 3",
+          flow=flow)
+    assert r.column("id").to_pylist() == [4, 2, 6]
+
+
+def test_distinct(flow):
+    r = q("SELECT DISTINCT grp FROM flow ORDER BY grp", flow=flow)
+    assert r.column("grp").to_pylist() == [1, 2, 3]
+
+
+def test_case_when(flow):
+    r = q("SELECT id, CASE WHEN value >= 20 THEN 1 ELSE 0 END AS big "
+          "FROM flow ORDER BY id", flow=flow)
+    assert r.column("big").to_pylist() == [0.0, 1.0, 0.0, 1.0, 0.0, 1.0]
+
+
+def test_cast_and_functions(flow):
+    r = q("SELECT cast(value AS int) AS vi, sqrt(value) AS sq FROM flow "
+          "WHERE id = 4", flow=flow)
+    assert r.column("vi").to_pylist() == [40]
+    assert abs(r.column("sq").to_pylist()[0] - math.sqrt(40)) < 1e-9
+
+
+def test_join_inner(flow):
+    dims = MessageBatch.from_dict({
+        "grp": [1, 2],
+        "label": ["one", "two"],
+    })
+    r = q("SELECT f.id, d.label FROM flow f JOIN dims d ON f.grp = d.grp "
+          "ORDER BY f.id", flow=flow, dims=dims)
+    assert r.column("id").to_pylist() == [1, 2, 3, 4]
+    assert r.column("label").to_strlist() == ["one", "one", "two", "two"]
+
+
+def test_join_left(flow):
+    dims = MessageBatch.from_dict({"grp": [1], "label": ["one"]})
+    r = q("SELECT f.id, d.label FROM flow f LEFT JOIN dims d ON f.grp = d.grp "
+          "ORDER BY f.id", flow=flow, dims=dims)
+    assert r.num_rows == 6
+    labels = r.column("label").to_pylist()
+    assert labels[0] == b"one" and labels[2] is None
+
+
+def test_empty_aggregate_returns_row():
+    empty = MessageBatch.from_dict({"v": []})
+    r = q("SELECT count(*) AS c FROM flow", flow=empty)
+    assert r.column("c").to_pylist() == [0]
+
+
+def test_ddl_rejected():
+    with pytest.raises(SqlError):
+        parse_sql("DROP TABLE flow")
+    with pytest.raises(SqlError):
+        parse_sql("INSERT INTO flow VALUES (1)")
+
+
+def test_unknown_column_errors(flow):
+    with pytest.raises(SqlError):
+        q("SELECT nope FROM flow", flow=flow)
+
+
+def test_meta_columns_queryable():
+    """Reference lib.rs metadata tests: __meta_* columns are SQL-visible."""
+    b = MessageBatch.from_dict({
+        "v": [1, 2, 3],
+        "__meta_offset": [100, 101, 102],
+        "__meta_source": ["k1", "k1", "k2"],
+    })
+    r = q("SELECT v FROM flow WHERE __meta_offset > 100 "
+          "AND __meta_source = 'k1'", flow=b)
+    assert r.column("v").to_pylist() == [2]
+
+
+def test_sql_processor_component(run):
+    import asyncio
+    from arkflow_amd.processors.sql import SqlProcessor
+    p = SqlProcessor({"query": "SELECT id FROM flow WHERE id > 1"})
+    b = MessageBatch.from_dict({"id": [1, 2, 3]})
+    out = run(p.process(b))
+    assert out[0].column("id").to_pylist() == [2, 3]
+    # empty in → ProcessResult::None
+    assert run(p.process(MessageBatch.from_dict({"id": []}))) == []
