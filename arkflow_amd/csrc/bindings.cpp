@@ -1,0 +1,355 @@
+// Torch extension bindings: arkflow_amd._native.
+// Thin host layer over the gfx950 kernels in *.hip — tensor checks, workspace
+// allocation, the tiny device cumsum between two-pass kernels, pybind exports.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+#include <tuple>
+
+// ---- extern kernel launchers (defined in the .hip TUs) ----------------------
+extern "C" {
+int filter_grid(int64_t n);
+void launch_filter_count_f32(const float*, int64_t, int, float, int32_t*,
+                             hipStream_t);
+void launch_filter_scatter_f32(const float*, int64_t, int, float,
+                               const int32_t*, int32_t*, hipStream_t);
+void launch_filter_count_i64(const int64_t*, int64_t, int, int64_t, int32_t*,
+                             hipStream_t);
+void launch_filter_scatter_i64(const int64_t*, int64_t, int, int64_t,
+                               const int32_t*, int32_t*, hipStream_t);
+void launch_mask_count(const bool*, int64_t, int32_t*, hipStream_t);
+void launch_mask_scatter(const bool*, int64_t, const int32_t*, int32_t*,
+                         hipStream_t);
+void launch_gather(const void*, const int32_t*, int64_t, void*, int,
+                   hipStream_t);
+void launch_hash_build(const int64_t*, int64_t, int64_t*, int32_t*, uint32_t,
+                       int32_t*, int32_t*, hipStream_t);
+void launch_hash_export(const int64_t*, const int32_t*, uint32_t, int64_t*,
+                        hipStream_t);
+void launch_segment_reduce_f32(const float*, const int32_t*, int64_t, int, int,
+                               float*, int32_t*, hipStream_t);
+void launch_join_build(const int64_t*, int64_t, int64_t*, int32_t*, int32_t*,
+                       uint32_t, hipStream_t);
+void launch_join_probe_count(const int64_t*, int64_t, const int64_t*,
+                             const int32_t*, const int32_t*, const int64_t*,
+                             uint32_t, int32_t*, hipStream_t);
+void launch_join_probe_emit(const int64_t*, int64_t, const int64_t*,
+                            const int32_t*, const int32_t*, const int64_t*,
+                            uint32_t, const int32_t*, int64_t*, int64_t*,
+                            hipStream_t);
+void launch_gemm_bf16(const void*, const void*, const float*, void*, int, int,
+                      int, int, hipStream_t);
+void launch_layernorm_bf16(const void*, const void*, const float*,
+                           const float*, void*, void*, int64_t, int, float,
+                           hipStream_t);
+void launch_softmax_bf16(const void*, void*, int64_t, int, float, hipStream_t);
+void launch_bias_act_bf16(const void*, const float*, void*, int64_t, int, int,
+                          hipStream_t);
+int launch_attention_bf16(const void*, const void*, const void*, void*, int,
+                          int, int, float, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_cuda(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+uint32_t next_pow2(uint32_t v) {
+  v--;
+  v |= v >> 1; v |= v >> 2; v |= v >> 4; v |= v >> 8; v |= v >> 16;
+  return v + 1;
+}
+
+// exclusive scan of int32 block counts → (offsets, total tensor on device)
+std::tuple<torch::Tensor, torch::Tensor> exscan(const torch::Tensor& counts) {
+  auto inc = counts.cumsum(0, torch::kInt32);
+  auto offs = inc - counts;
+  auto total = inc.numel() > 0
+                   ? inc.slice(0, inc.numel() - 1, inc.numel())
+                   : torch::zeros({1}, counts.options());
+  return {offs, total};
+}
+
+// ---------------------------------------------------------------------- filter
+torch::Tensor mask_to_indices(torch::Tensor mask) {
+  check_cuda(mask, "mask");
+  TORCH_CHECK(mask.scalar_type() == torch::kBool, "mask must be bool");
+  int64_t n = mask.numel();
+  auto st = cur_stream();
+  int nblocks = filter_grid(n);
+  auto counts = torch::empty({std::max(nblocks, 1)},
+                             mask.options().dtype(torch::kInt32));
+  if (n == 0) return torch::empty({0}, mask.options().dtype(torch::kInt32));
+  launch_mask_count(mask.data_ptr<bool>(), n, counts.data_ptr<int32_t>(), st);
+  auto [offs, total_t] = exscan(counts);
+  int64_t total = total_t.item<int32_t>();  // one device sync per filter op
+  auto out = torch::empty({total}, mask.options().dtype(torch::kInt32));
+  if (total)
+    launch_mask_scatter(mask.data_ptr<bool>(), n, offs.data_ptr<int32_t>(),
+                        out.data_ptr<int32_t>(), st);
+  return out;
+}
+
+torch::Tensor filter_cmp_scalar(torch::Tensor col, int64_t op, double scalar) {
+  check_cuda(col, "col");
+  int64_t n = col.numel();
+  if (n == 0) return torch::empty({0}, col.options().dtype(torch::kInt32));
+  auto st = cur_stream();
+  int nblocks = filter_grid(n);
+  auto counts = torch::empty({nblocks}, col.options().dtype(torch::kInt32));
+  if (col.scalar_type() == torch::kInt32) col = col.to(torch::kInt64);
+  if (col.scalar_type() == torch::kFloat32) {
+    launch_filter_count_f32(col.data_ptr<float>(), n, (int)op, (float)scalar,
+                            counts.data_ptr<int32_t>(), st);
+  } else if (col.scalar_type() == torch::kInt64) {
+    launch_filter_count_i64(col.data_ptr<int64_t>(), n, (int)op,
+                            (int64_t)scalar, counts.data_ptr<int32_t>(), st);
+  } else {
+    TORCH_CHECK(false, "filter_cmp_scalar supports f32/i64/i32 cols");
+  }
+  auto [offs, total_t] = exscan(counts);
+  int64_t total = total_t.item<int32_t>();
+  auto out = torch::empty({total}, col.options().dtype(torch::kInt32));
+  if (total) {
+    if (col.scalar_type() == torch::kFloat32)
+      launch_filter_scatter_f32(col.data_ptr<float>(), n, (int)op,
+                                (float)scalar, offs.data_ptr<int32_t>(),
+                                out.data_ptr<int32_t>(), st);
+    else
+      launch_filter_scatter_i64(col.data_ptr<int64_t>(), n, (int)op,
+                                (int64_t)scalar, offs.data_ptr<int32_t>(),
+                                out.data_ptr<int32_t>(), st);
+  }
+  return out;
+}
+
+torch::Tensor gather(torch::Tensor col, torch::Tensor idx) {
+  check_cuda(col, "col");
+  check_cuda(idx, "idx");
+  auto idx32 = idx.scalar_type() == torch::kInt32 ? idx
+                                                  : idx.to(torch::kInt32);
+  int64_t m = idx32.numel();
+  auto out = torch::empty({m}, col.options());
+  int esz = (int)col.element_size();
+  TORCH_CHECK(esz == 1 || esz == 2 || esz == 4 || esz == 8,
+              "unsupported element size");
+  if (m)
+    launch_gather(col.data_ptr(), idx32.data_ptr<int32_t>(), m,
+                  out.data_ptr(), esz, cur_stream());
+  return out;
+}
+
+// ------------------------------------------------------------------- group-by
+std::tuple<torch::Tensor, torch::Tensor> hash_group_i64(torch::Tensor keys) {
+  check_cuda(keys, "keys");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  int64_t n = keys.numel();
+  auto opts32 = keys.options().dtype(torch::kInt32);
+  if (n == 0)
+    return {torch::empty({0}, opts32), torch::empty({0}, keys.options())};
+  uint32_t tsize = next_pow2((uint32_t)std::max<int64_t>(n * 2, 64));
+  auto table_keys = torch::empty({(int64_t)tsize}, keys.options());
+  auto table_gids = torch::empty({(int64_t)tsize}, opts32);
+  auto counter = torch::zeros({1}, opts32);
+  auto gids = torch::empty({n}, opts32);
+  auto st = cur_stream();
+  launch_hash_build(keys.data_ptr<int64_t>(), n,
+                    table_keys.data_ptr<int64_t>(),
+                    table_gids.data_ptr<int32_t>(), tsize,
+                    counter.data_ptr<int32_t>(), gids.data_ptr<int32_t>(), st);
+  int64_t g = counter.item<int32_t>();  // sync to size the uniq output
+  auto uniq = torch::empty({g}, keys.options());
+  launch_hash_export(table_keys.data_ptr<int64_t>(),
+                     table_gids.data_ptr<int32_t>(), tsize,
+                     uniq.data_ptr<int64_t>(), st);
+  return {gids, uniq};
+}
+
+torch::Tensor segment_reduce_f32(torch::Tensor vals, torch::Tensor gids,
+                                 int64_t g, int64_t op) {
+  check_cuda(vals, "vals");
+  check_cuda(gids, "gids");
+  TORCH_CHECK(vals.scalar_type() == torch::kFloat32, "vals must be f32");
+  auto g32 = gids.scalar_type() == torch::kInt32 ? gids
+                                                 : gids.to(torch::kInt32);
+  auto out = torch::zeros({g}, vals.options());
+  if (vals.numel() == 0) return out;
+  torch::Tensor scratch;
+  int32_t* scratch_ptr = nullptr;
+  if (op != 0) {
+    scratch = torch::empty({g}, vals.options().dtype(torch::kInt32));
+    scratch_ptr = scratch.data_ptr<int32_t>();
+  }
+  launch_segment_reduce_f32(vals.data_ptr<float>(), g32.data_ptr<int32_t>(),
+                            vals.numel(), (int)g, (int)op,
+                            out.data_ptr<float>(), scratch_ptr, cur_stream());
+  return out;
+}
+
+// ----------------------------------------------------------------------- join
+std::tuple<torch::Tensor, torch::Tensor> join_inner_i64(torch::Tensor lk,
+                                                        torch::Tensor rk) {
+  check_cuda(lk, "left_keys");
+  check_cuda(rk, "right_keys");
+  TORCH_CHECK(lk.scalar_type() == torch::kInt64 &&
+                  rk.scalar_type() == torch::kInt64,
+              "join keys must be int64");
+  int64_t nl = lk.numel(), nr = rk.numel();
+  auto opts64 = lk.options();
+  auto opts32 = lk.options().dtype(torch::kInt32);
+  if (nl == 0 || nr == 0)
+    return {torch::empty({0}, opts64), torch::empty({0}, opts64)};
+  uint32_t tsize = next_pow2((uint32_t)std::max<int64_t>(nr * 2, 64));
+  auto table_keys = torch::empty({(int64_t)tsize}, opts64);
+  auto table_head = torch::empty({(int64_t)tsize}, opts32);
+  auto next = torch::empty({nr}, opts32);
+  auto st = cur_stream();
+  launch_join_build(rk.data_ptr<int64_t>(), nr, table_keys.data_ptr<int64_t>(),
+                    table_head.data_ptr<int32_t>(), next.data_ptr<int32_t>(),
+                    tsize, st);
+  auto counts = torch::empty({nl}, opts32);
+  launch_join_probe_count(lk.data_ptr<int64_t>(), nl,
+                          table_keys.data_ptr<int64_t>(),
+                          table_head.data_ptr<int32_t>(),
+                          next.data_ptr<int32_t>(), rk.data_ptr<int64_t>(),
+                          tsize, counts.data_ptr<int32_t>(), st);
+  auto [offs, total_t] = exscan(counts);
+  int64_t total = total_t.item<int32_t>();
+  auto l_out = torch::empty({total}, opts64);
+  auto r_out = torch::empty({total}, opts64);
+  if (total)
+    launch_join_probe_emit(lk.data_ptr<int64_t>(), nl,
+                           table_keys.data_ptr<int64_t>(),
+                           table_head.data_ptr<int32_t>(),
+                           next.data_ptr<int32_t>(), rk.data_ptr<int64_t>(),
+                           tsize, offs.data_ptr<int32_t>(),
+                           l_out.data_ptr<int64_t>(),
+                           r_out.data_ptr<int64_t>(), st);
+  return {l_out, r_out};
+}
+
+// ------------------------------------------------------------------ inference
+torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
+                        c10::optional<torch::Tensor> bias, int64_t act) {
+  check_cuda(A, "A");
+  check_cuda(Bt, "Bt");
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+                  Bt.scalar_type() == torch::kBFloat16,
+              "gemm_bf16 requires bf16 inputs");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2, "A [M,K], Bt [N,K]");
+  int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 32 == 0 && K >= 32, "K must be a multiple of 32");
+  auto C = torch::empty({M, N}, A.options());
+  const float* bias_ptr = nullptr;
+  torch::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->scalar_type() == torch::kFloat32
+                 ? bias->contiguous()
+                 : bias->to(torch::kFloat32).contiguous();
+    TORCH_CHECK(bias_f.numel() == N, "bias must be [N]");
+    bias_ptr = bias_f.data_ptr<float>();
+  }
+  launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
+                   (int)M, (int)N, (int)K, (int)act, cur_stream());
+  return C;
+}
+
+torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor gamma,
+                             torch::Tensor beta, double eps,
+                             c10::optional<torch::Tensor> residual) {
+  check_cuda(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  int64_t n = x.size(-1);
+  TORCH_CHECK(n % 8 == 0, "inner dim must be a multiple of 8");
+  int64_t rows = x.numel() / n;
+  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto b = beta.to(torch::kFloat32).contiguous();
+  auto out = torch::empty_like(x);
+  const void* res_ptr = nullptr;
+  if (residual.has_value() && residual->defined()) {
+    TORCH_CHECK(residual->is_contiguous() &&
+                residual->scalar_type() == torch::kBFloat16);
+    res_ptr = residual->data_ptr();
+  }
+  launch_layernorm_bf16(x.data_ptr(), res_ptr, g.data_ptr<float>(),
+                        b.data_ptr<float>(), out.data_ptr(), nullptr, rows,
+                        (int)n, (float)eps, cur_stream());
+  return out;
+}
+
+torch::Tensor softmax_bf16(torch::Tensor x, double scale) {
+  check_cuda(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  int64_t n = x.size(-1);
+  TORCH_CHECK(n % 8 == 0, "inner dim must be a multiple of 8");
+  int64_t rows = x.numel() / n;
+  auto out = torch::empty_like(x);
+  launch_softmax_bf16(x.data_ptr(), out.data_ptr(), rows, (int)n,
+                      (float)scale, cur_stream());
+  return out;
+}
+
+torch::Tensor bias_act_bf16(torch::Tensor x,
+                            c10::optional<torch::Tensor> bias, int64_t act) {
+  check_cuda(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  int64_t n = x.size(-1);
+  int64_t rows = x.numel() / n;
+  auto out = torch::empty_like(x);
+  const float* bias_ptr = nullptr;
+  torch::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->to(torch::kFloat32).contiguous();
+    bias_ptr = bias_f.data_ptr<float>();
+  }
+  launch_bias_act_bf16(x.data_ptr(), bias_ptr, out.data_ptr(), rows, (int)n,
+                       (int)act, cur_stream());
+  return out;
+}
+
+torch::Tensor attention_bf16(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                             double scale) {
+  check_cuda(q, "q");
+  check_cuda(k, "k");
+  check_cuda(v, "v");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attention requires bf16");
+  TORCH_CHECK(q.dim() == 4, "q must be [B,H,S,D]");
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  auto out = torch::empty_like(q);
+  int rc = launch_attention_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                                 out.data_ptr(), (int)(B * H), (int)S, (int)D,
+                                 (float)scale, cur_stream());
+  TORCH_CHECK(rc == 0, "attention_bf16: unsupported shape S=", S, " D=", D);
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "arkflow_amd gfx950 native kernels";
+  m.def("mask_to_indices", &mask_to_indices);
+  m.def("filter_cmp_scalar", &filter_cmp_scalar);
+  m.def("gather", &gather);
+  m.def("hash_group_i64", &hash_group_i64);
+  m.def("segment_reduce_f32", &segment_reduce_f32);
+  m.def("join_inner_i64", &join_inner_i64);
+  m.def("gemm_bf16", &gemm_bf16, py::arg("A"), py::arg("Bt"),
+        py::arg("bias") = py::none(), py::arg("act") = 0);
+  m.def("layernorm_bf16", &layernorm_bf16, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("eps") = 1e-5,
+        py::arg("residual") = py::none());
+  m.def("softmax_bf16", &softmax_bf16, py::arg("x"), py::arg("scale") = 1.0);
+  m.def("bias_act_bf16", &bias_act_bf16, py::arg("x"),
+        py::arg("bias") = py::none(), py::arg("act") = 0);
+  m.def("attention_bf16", &attention_bf16, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("scale"));
+}

@@ -1,0 +1,205 @@
+// Order-preserving stream compaction + fused compare-and-compact + gather.
+//
+// MI355X-native replacement for DataFusion's FilterExec on the `sql`
+// processor's hot path (reference processor/sql.rs:107-146). Two-pass
+// block-count → scan → scatter keeps row order without global atomics;
+// the scan between passes is one tiny device cumsum done by the caller.
+#include "common.h"
+
+// items per block: 256 threads × 4 = 1024 rows — small batches (8192 rows)
+// still fill ≥8 blocks; big batches scale to thousands of workgroups.
+#define FILTER_BLOCK 256
+#define FILTER_IPT 4
+#define FILTER_TILE (FILTER_BLOCK * FILTER_IPT)
+
+enum CmpOp { LT = 0, LE = 1, GT = 2, GE = 3, EQ = 4, NE = 5 };
+
+template <typename T>
+DEV_INLINE bool cmp_apply(T v, int op, T s) {
+  switch (op) {
+    case LT: return v < s;
+    case LE: return v <= s;
+    case GT: return v > s;
+    case GE: return v >= s;
+    case EQ: return v == s;
+    default: return v != s;
+  }
+}
+
+// ---- pass 1: per-block count of matching rows -------------------------------
+template <typename T>
+__global__ void filter_count_kernel(const T* __restrict__ col, int64_t n,
+                                    int op, T scalar,
+                                    int32_t* __restrict__ block_counts) {
+  int64_t base = (int64_t)blockIdx.x * FILTER_TILE;
+  int cnt = 0;
+#pragma unroll
+  for (int i = 0; i < FILTER_IPT; ++i) {
+    int64_t idx = base + threadIdx.x + i * FILTER_BLOCK;
+    if (idx < n && cmp_apply(col[idx], op, scalar)) ++cnt;
+  }
+  // wave reduce then LDS across the 4 waves
+  __shared__ int wsum[FILTER_BLOCK / WAVE];
+  int v = cnt;
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  if ((threadIdx.x & 63) == 0) wsum[threadIdx.x >> 6] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int total = 0;
+#pragma unroll
+    for (int w = 0; w < FILTER_BLOCK / WAVE; ++w) total += wsum[w];
+    block_counts[blockIdx.x] = total;
+  }
+}
+
+// ---- pass 2: scatter matching row indices at block offsets ------------------
+template <typename T>
+__global__ void filter_scatter_kernel(const T* __restrict__ col, int64_t n,
+                                      int op, T scalar,
+                                      const int32_t* __restrict__ block_offsets,
+                                      int32_t* __restrict__ out_idx) {
+  int64_t base = (int64_t)blockIdx.x * FILTER_TILE;
+  __shared__ int wave_base[FILTER_BLOCK / WAVE + 1];
+  int write = block_offsets[blockIdx.x];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  // process the tile in FILTER_IPT ordered strips so output stays row-ordered
+#pragma unroll
+  for (int i = 0; i < FILTER_IPT; ++i) {
+    int64_t idx = base + i * FILTER_BLOCK + threadIdx.x;
+    bool pred = (idx < n) && cmp_apply(col[idx], op, scalar);
+    uint64_t ballot = __ballot(pred);
+    int rank = __popcll(ballot & lanemask_lt());
+    int wave_total = __popcll(ballot);
+    if (lane == 0) wave_base[wid + 1] = wave_total;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      wave_base[0] = write;
+      for (int w = 1; w <= FILTER_BLOCK / WAVE; ++w)
+        wave_base[w] += wave_base[w - 1];
+    }
+    __syncthreads();
+    if (pred) out_idx[wave_base[wid] + rank] = (int32_t)idx;
+    write = wave_base[FILTER_BLOCK / WAVE];
+    __syncthreads();
+  }
+}
+
+// ---- mask (bool) variants ---------------------------------------------------
+__global__ void mask_count_kernel(const bool* __restrict__ mask, int64_t n,
+                                  int32_t* __restrict__ block_counts) {
+  int64_t base = (int64_t)blockIdx.x * FILTER_TILE;
+  int cnt = 0;
+#pragma unroll
+  for (int i = 0; i < FILTER_IPT; ++i) {
+    int64_t idx = base + threadIdx.x + i * FILTER_BLOCK;
+    if (idx < n && mask[idx]) ++cnt;
+  }
+  __shared__ int wsum[FILTER_BLOCK / WAVE];
+  int v = cnt;
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  if ((threadIdx.x & 63) == 0) wsum[threadIdx.x >> 6] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int total = 0;
+#pragma unroll
+    for (int w = 0; w < FILTER_BLOCK / WAVE; ++w) total += wsum[w];
+    block_counts[blockIdx.x] = total;
+  }
+}
+
+__global__ void mask_scatter_kernel(const bool* __restrict__ mask, int64_t n,
+                                    const int32_t* __restrict__ block_offsets,
+                                    int32_t* __restrict__ out_idx) {
+  int64_t base = (int64_t)blockIdx.x * FILTER_TILE;
+  __shared__ int wave_base[FILTER_BLOCK / WAVE + 1];
+  int write = block_offsets[blockIdx.x];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+#pragma unroll
+  for (int i = 0; i < FILTER_IPT; ++i) {
+    int64_t idx = base + i * FILTER_BLOCK + threadIdx.x;
+    bool pred = (idx < n) && mask[idx];
+    uint64_t ballot = __ballot(pred);
+    int rank = __popcll(ballot & lanemask_lt());
+    int wave_total = __popcll(ballot);
+    if (lane == 0) wave_base[wid + 1] = wave_total;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      wave_base[0] = write;
+      for (int w = 1; w <= FILTER_BLOCK / WAVE; ++w)
+        wave_base[w] += wave_base[w - 1];
+    }
+    __syncthreads();
+    if (pred) out_idx[wave_base[wid] + rank] = (int32_t)idx;
+    write = wave_base[FILTER_BLOCK / WAVE];
+    __syncthreads();
+  }
+}
+
+// ---- gather -----------------------------------------------------------------
+// Coalesced row gather by element size; one thread per output row × chunk.
+template <typename T>
+__global__ void gather_kernel(const T* __restrict__ src,
+                              const int32_t* __restrict__ idx, int64_t m,
+                              T* __restrict__ dst) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < m; i += stride) dst[i] = src[idx[i]];
+}
+
+// ---- host-visible launchers (called from bindings.cpp) ----------------------
+extern "C" {
+
+int filter_grid(int64_t n) { return (int)((n + FILTER_TILE - 1) / FILTER_TILE); }
+
+void launch_filter_count_f32(const float* col, int64_t n, int op, float s,
+                             int32_t* counts, hipStream_t st) {
+  filter_count_kernel<float><<<filter_grid(n), FILTER_BLOCK, 0, st>>>(
+      col, n, op, s, counts);
+}
+void launch_filter_scatter_f32(const float* col, int64_t n, int op, float s,
+                               const int32_t* offs, int32_t* out,
+                               hipStream_t st) {
+  filter_scatter_kernel<float><<<filter_grid(n), FILTER_BLOCK, 0, st>>>(
+      col, n, op, s, offs, out);
+}
+void launch_filter_count_i64(const int64_t* col, int64_t n, int op, int64_t s,
+                             int32_t* counts, hipStream_t st) {
+  filter_count_kernel<int64_t><<<filter_grid(n), FILTER_BLOCK, 0, st>>>(
+      col, n, op, s, counts);
+}
+void launch_filter_scatter_i64(const int64_t* col, int64_t n, int op,
+                               int64_t s, const int32_t* offs, int32_t* out,
+                               hipStream_t st) {
+  filter_scatter_kernel<int64_t><<<filter_grid(n), FILTER_BLOCK, 0, st>>>(
+      col, n, op, s, offs, out);
+}
+void launch_mask_count(const bool* mask, int64_t n, int32_t* counts,
+                       hipStream_t st) {
+  mask_count_kernel<<<filter_grid(n), FILTER_BLOCK, 0, st>>>(mask, n, counts);
+}
+void launch_mask_scatter(const bool* mask, int64_t n, const int32_t* offs,
+                         int32_t* out, hipStream_t st) {
+  mask_scatter_kernel<<<filter_grid(n), FILTER_BLOCK, 0, st>>>(
+      mask, n, offs, out);
+}
+
+void launch_gather(const void* src, const int32_t* idx, int64_t m,
+                   void* dst, int elem_size, hipStream_t st) {
+  int grid = (int)((m + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid == 0) return;
+  switch (elem_size) {
+    case 1: gather_kernel<uint8_t><<<grid, 256, 0, st>>>(
+        (const uint8_t*)src, idx, m, (uint8_t*)dst); break;
+    case 2: gather_kernel<uint16_t><<<grid, 256, 0, st>>>(
+        (const uint16_t*)src, idx, m, (uint16_t*)dst); break;
+    case 4: gather_kernel<uint32_t><<<grid, 256, 0, st>>>(
+        (const uint32_t*)src, idx, m, (uint32_t*)dst); break;
+    default: gather_kernel<uint64_t><<<grid, 256, 0, st>>>(
+        (const uint64_t*)src, idx, m, (uint64_t*)dst); break;
+  }
+}
+
+}  // extern "C"

@@ -1,0 +1,169 @@
+// bf16 MFMA GEMM with fused bias+activation epilogue — the inference
+// processor's matmul (replaces the reference's python-processor escape hatch
+// for ML, reference processor/python.rs:47-98, with a native CDNA4 path).
+//
+// Structure: 128×128 tile, BK=32, 4 waves (256 threads), each wave owns a
+// 64×64 sub-tile = 4×4 fragments of mfma_f32_16x16x32_bf16; staging is
+// direct global→LDS via __builtin_amdgcn_global_load_lds width=16 (the
+// CDNA4 cp.async analog). B is supplied transposed ([N,K] row-major — the
+// torch Linear weight layout), so both A and B^T fragments read LDS with the
+// same contiguous-16B pattern (ds_read_b128).
+//
+// C/D fragment mapping (gfx950, 16x16x32): col = lane&15,
+// row = (lane>>4)*4 + reg. Verified against torch.matmul in
+// tests/test_gpu_kernels.py with random asymmetric inputs (transpose-detecting).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define GEMM_THREADS 256
+
+enum Act { ACT_NONE = 0, ACT_RELU = 1, ACT_GELU = 2, ACT_SILU = 3 };
+
+DEV_INLINE float apply_act(float x, int act) {
+  switch (act) {
+    case ACT_RELU: return fmaxf(x, 0.f);
+    case ACT_GELU: {
+      // tanh approximation (matches torch.nn.GELU(approximate="tanh"))
+      float c = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+      return 0.5f * x * (1.f + tanhf(c));
+    }
+    case ACT_SILU: return x / (1.f + __expf(-x));
+    default: return x;
+  }
+}
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 2)
+void gemm_bf16_kernel(const __bf16* __restrict__ A,   // [M,K]
+                      const __bf16* __restrict__ Bt,  // [N,K]
+                      const float* __restrict__ bias, // [N] or null
+                      __bf16* __restrict__ C,         // [M,N]
+                      int M, int N, int K, int tiles_n) {
+  // XCD-aware bijective block swizzle (8 XCDs; guide §5 m204 variant)
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * BM, col0 = bn * BN;
+
+  __shared__ __bf16 Asm[BM * BK];  // [128][32] linear
+  __shared__ __bf16 Bsm[BN * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 4 waves: 2×2 over the 128×128 tile
+  const int wm = (wid >> 1) * 64;     // wave row offset
+  const int wn = (wid & 1) * 64;     // wave col offset
+
+  f32x4 acc[4][4] = {};
+
+  // staging geometry: each global_load_lds call moves 64 lanes × 16 B = 1 KiB
+  // per wave; a 8 KiB tile needs 2 calls per wave (iter 0/1).
+  // linear LDS byte offset for this lane/iter: wid*1024 + lane*16 + iter*4096
+  const int lin0 = wid * 1024 + lane * 16;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int lin = lin0 + it * 4096;           // byte offset in the 8 KiB tile
+      int trow = lin >> 6;                  // tile row (64 B per row)
+      int tcol = lin & 63;                  // byte within row
+      // A: clamp source row (out-of-range rows produce garbage that only
+      // lands in out-of-range C rows, which the epilogue never writes)
+      int ga_row = row0 + trow;
+      ga_row = ga_row < M ? ga_row : M - 1;
+      const char* a_src = (const char*)(A + (int64_t)ga_row * K + k0) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)a_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Asm + (wid * 1024 + it * 4096)),
+          16, 0, 0);
+      int gb_row = col0 + trow;
+      gb_row = gb_row < N ? gb_row : N - 1;
+      const char* b_src = (const char*)(Bt + (int64_t)gb_row * K + k0) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)b_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Bsm + (wid * 1024 + it * 4096)),
+          16, 0, 0);
+    }
+    __syncthreads();  // drains vmcnt → LDS tiles ready
+
+    // fragment reads: lane holds 8 contiguous bf16 at
+    // [row = sub*16 + (lane&15)][(lane>>4)*8]
+    bf16x8 a_frag[4], b_frag[4];
+    const int fr = lane & 15;
+    const int fk = (lane >> 4) * 8;
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      a_frag[m] = *(const bf16x8*)&Asm[(wm + m * 16 + fr) * BK + fk];
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      b_frag[n] = *(const bf16x8*)&Bsm[(wn + n * 16 + fr) * BK + fk];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // epilogue: bias + activation, bf16 store with bounds check
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" {
+
+void launch_gemm_bf16(const void* A, const void* Bt, const float* bias,
+                      void* C, int M, int N, int K, int act, hipStream_t st) {
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define DISPATCH(ACT)                                                        \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_kernel<ACT, true><<<grid, block, 0, st>>>(                   \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_kernel<ACT, false><<<grid, block, 0, st>>>(                  \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: DISPATCH(ACT_RELU); break;
+    case ACT_GELU: DISPATCH(ACT_GELU); break;
+    case ACT_SILU: DISPATCH(ACT_SILU); break;
+    default: DISPATCH(ACT_NONE); break;
+  }
+#undef DISPATCH
+}
+
+}  // extern "C"

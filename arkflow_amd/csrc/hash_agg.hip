@@ -1,0 +1,194 @@
+// GPU hash GROUP BY: open-addressing device hash table + segment reductions.
+//
+// Replaces DataFusion's hash-aggregate physical operator on the `sql`
+// processor path (reference processor/sql.rs execute_query → AggregateExec).
+// Build: per-row atomicCAS claim into a power-of-two table; first claimant
+// assigns a dense group id. Reduce: LDS-tiled per-block partials when the
+// group count is small (the common GROUP BY case), global atomics otherwise.
+#include "common.h"
+
+#define EMPTY_KEY 0x8000000000000000ll
+#define AGG_BLOCK 256
+#define LDS_GROUPS 2048
+
+enum RedOp { SUM = 0, MIN = 1, MAX = 2 };
+
+static int32_t float_flip_host(float f) {
+  union { float f; int32_t i; } u{f};
+  return u.i < 0 ? ~u.i : (int32_t)(u.i | 0x80000000);
+}
+
+// ---- group-id assignment ----------------------------------------------------
+__global__ void hash_build_kernel(const int64_t* __restrict__ keys, int64_t n,
+                                  int64_t* __restrict__ table_keys,
+                                  int32_t* __restrict__ table_gids,
+                                  uint32_t table_mask,
+                                  int32_t* __restrict__ counter,
+                                  int32_t* __restrict__ gids_out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    uint32_t h = (uint32_t)(mix64((uint64_t)k) & table_mask);
+    for (;;) {
+      int64_t cur = __hip_atomic_load(&table_keys[h], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == k) break;
+      if (cur == EMPTY_KEY) {
+        int64_t prev = (int64_t)atomicCAS(
+            (unsigned long long*)&table_keys[h], (unsigned long long)EMPTY_KEY,
+            (unsigned long long)k);
+        if (prev == EMPTY_KEY) {
+          int32_t gid = atomicAdd(counter, 1);
+          __hip_atomic_store(&table_gids[h], gid, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          break;
+        }
+        if (prev == k) break;
+      }
+      h = (h + 1) & table_mask;
+    }
+    int32_t g;
+    while ((g = __hip_atomic_load(&table_gids[h], __ATOMIC_ACQUIRE,
+                                  __HIP_MEMORY_SCOPE_AGENT)) < 0) {}
+    gids_out[i] = g;
+  }
+}
+
+// uniq[gid] = key
+__global__ void hash_export_kernel(const int64_t* __restrict__ table_keys,
+                                   const int32_t* __restrict__ table_gids,
+                                   uint32_t table_size,
+                                   int64_t* __restrict__ uniq) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= table_size) return;
+  int64_t k = table_keys[i];
+  if (k != EMPTY_KEY) uniq[table_gids[i]] = k;
+}
+
+// ---- segment reductions ------------------------------------------------------
+// SUM, LDS-tiled: per-block float partials, one global atomic per group/block.
+__global__ void segment_sum_lds_kernel(const float* __restrict__ vals,
+                                       const int32_t* __restrict__ gids,
+                                       int64_t n, int g,
+                                       float* __restrict__ out) {
+  __shared__ float part[LDS_GROUPS];
+  for (int j = threadIdx.x; j < g; j += blockDim.x) part[j] = 0.f;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) atomicAdd(&part[gids[i]], vals[i]);
+  __syncthreads();
+  for (int j = threadIdx.x; j < g; j += blockDim.x)
+    if (part[j] != 0.f) atomicAdd(&out[j], part[j]);
+}
+
+__global__ void segment_sum_global_kernel(const float* __restrict__ vals,
+                                          const int32_t* __restrict__ gids,
+                                          int64_t n, float* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) atomicAdd(&out[gids[i]], vals[i]);
+}
+
+// MIN/MAX on monotone-flipped int32 (handles negative floats correctly).
+__global__ void segment_mm_lds_kernel(const float* __restrict__ vals,
+                                      const int32_t* __restrict__ gids,
+                                      int64_t n, int g, int op,
+                                      int32_t* __restrict__ out_flipped,
+                                      int32_t init) {
+  __shared__ int32_t part[LDS_GROUPS];
+  for (int j = threadIdx.x; j < g; j += blockDim.x) part[j] = init;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int32_t v = float_flip(vals[i]);
+    if (op == MIN) atomicMin(&part[gids[i]], v);
+    else atomicMax(&part[gids[i]], v);
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < g; j += blockDim.x) {
+    if (op == MIN) atomicMin(&out_flipped[j], part[j]);
+    else atomicMax(&out_flipped[j], part[j]);
+  }
+}
+
+__global__ void segment_mm_global_kernel(const float* __restrict__ vals,
+                                         const int32_t* __restrict__ gids,
+                                         int64_t n, int op,
+                                         int32_t* __restrict__ out_flipped) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int32_t v = float_flip(vals[i]);
+    if (op == MIN) atomicMin(&out_flipped[gids[i]], v);
+    else atomicMax(&out_flipped[gids[i]], v);
+  }
+}
+
+__global__ void unflip_kernel(const int32_t* __restrict__ in,
+                              float* __restrict__ out, int g) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < g) out[i] = float_unflip(in[i]);
+}
+
+__global__ void fill_i64_kernel(int64_t* p, int64_t v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = v;
+}
+__global__ void fill_i32_kernel(int32_t* p, int32_t v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = v;
+}
+
+// ---- host launchers ----------------------------------------------------------
+extern "C" {
+
+static int grid_for(int64_t n, int block) {
+  int64_t g = (n + block - 1) / block;
+  return (int)(g > 2048 ? 2048 : (g < 1 ? 1 : g));
+}
+
+void launch_hash_build(const int64_t* keys, int64_t n, int64_t* table_keys,
+                       int32_t* table_gids, uint32_t table_size,
+                       int32_t* counter, int32_t* gids_out, hipStream_t st) {
+  fill_i64_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_keys, EMPTY_KEY, table_size);
+  fill_i32_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_gids, -1, table_size);
+  hash_build_kernel<<<grid_for(n, AGG_BLOCK), AGG_BLOCK, 0, st>>>(
+      keys, n, table_keys, table_gids, table_size - 1, counter, gids_out);
+}
+
+void launch_hash_export(const int64_t* table_keys, const int32_t* table_gids,
+                        uint32_t table_size, int64_t* uniq, hipStream_t st) {
+  hash_export_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_keys, table_gids, table_size, uniq);
+}
+
+// op: 0=sum 1=min 2=max; out must be zero-filled for sum.
+// scratch_flipped: int32[g] workspace for min/max.
+void launch_segment_reduce_f32(const float* vals, const int32_t* gids,
+                               int64_t n, int g, int op, float* out,
+                               int32_t* scratch_flipped, hipStream_t st) {
+  int grid = grid_for(n, AGG_BLOCK);
+  if (op == SUM) {
+    if (g <= LDS_GROUPS)
+      segment_sum_lds_kernel<<<grid, AGG_BLOCK, 0, st>>>(vals, gids, n, g, out);
+    else
+      segment_sum_global_kernel<<<grid, AGG_BLOCK, 0, st>>>(vals, gids, n, out);
+    return;
+  }
+  int32_t init = float_flip_host(op == MIN ? INFINITY : -INFINITY);
+  fill_i32_kernel<<<grid_for(g, 256), 256, 0, st>>>(scratch_flipped, init, g);
+  if (g <= LDS_GROUPS)
+    segment_mm_lds_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+        vals, gids, n, g, op, scratch_flipped, init);
+  else
+    segment_mm_global_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+        vals, gids, n, op, scratch_flipped);
+  unflip_kernel<<<grid_for(g, 256), 256, 0, st>>>(scratch_flipped, out, g);
+}
+
+}  // extern "C"

@@ -1,0 +1,72 @@
+"""Neural-net ops for the inference processor: bf16 MFMA GEMM (fused
+bias+activation), LayerNorm, softmax. GPU = hand-written gfx950 kernels
+(csrc/gemm_bf16.hip, csrc/rowops.hip); CPU = fp32 torch reference (also the
+numerics oracle in tests)."""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import require_native
+
+ACT_NONE, ACT_RELU, ACT_GELU, ACT_SILU = 0, 1, 2, 3
+_ACTS = {"none": ACT_NONE, "relu": ACT_RELU, "gelu": ACT_GELU,
+         "silu": ACT_SILU}
+
+
+def linear_bf16(x: torch.Tensor, weight: torch.Tensor,
+                bias: Optional[torch.Tensor] = None,
+                act: str = "none") -> torch.Tensor:
+    """act(x @ weight.T + bias). x:[M,K] bf16, weight:[N,K] (torch Linear
+    layout — consumed directly as the GEMM's B^T operand)."""
+    a = _ACTS[act]
+    if x.is_cuda:
+        nat = require_native()
+        x2 = x.reshape(-1, x.shape[-1]).contiguous()
+        out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
+        return out.reshape(*x.shape[:-1], weight.shape[0])
+    y = torch.nn.functional.linear(x.float(), weight.float(),
+                                   bias.float() if bias is not None else None)
+    if act == "relu":
+        y = torch.relu(y)
+    elif act == "gelu":
+        y = torch.nn.functional.gelu(y, approximate="tanh")
+    elif act == "silu":
+        y = torch.nn.functional.silu(y)
+    return y.to(x.dtype)
+
+
+def layernorm_bf16(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                   eps: float = 1e-5,
+                   residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if x.is_cuda:
+        nat = require_native()
+        return nat.layernorm_bf16(x.contiguous(), gamma, beta, eps,
+                                  residual.contiguous()
+                                  if residual is not None else None)
+    xf = x.float()
+    if residual is not None:
+        xf = xf + residual.float()
+    y = torch.nn.functional.layer_norm(
+        xf, (x.shape[-1],), gamma.float(), beta.float(), eps)
+    return y.to(x.dtype)
+
+
+def softmax_bf16(x: torch.Tensor, scale: float = 1.0) -> torch.Tensor:
+    if x.is_cuda:
+        nat = require_native()
+        return nat.softmax_bf16(x.contiguous(), scale)
+    return torch.softmax(x.float() * scale, dim=-1).to(x.dtype)
+
+
+def attention_bf16(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   scale: float) -> torch.Tensor:
+    """Fused attention for [B,H,S,D] bf16. GPU: one-workgroup-per-(b,h)
+    MFMA kernel (csrc/attention.hip); CPU: fp32 reference."""
+    if q.is_cuda:
+        nat = require_native()
+        return nat.attention_bf16(q.contiguous(), k.contiguous(),
+                                  v.contiguous(), scale)
+    p = torch.softmax(q.float() @ k.float().transpose(-1, -2) * scale, dim=-1)
+    return (p @ v.float()).to(q.dtype)

@@ -1,0 +1,179 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark: generate → sql(filter) → ML inference → drop.
+
+Measures the BASELINE.json headline metric — rows/sec (+ p50 pipeline
+latency) for the generate→sql→infer pipeline — on N GPUs of one node, one
+rank per GPU over RCCL (weak scaling: each rank runs an independent stream
+shard, matching the engine's stream-per-GPU sharding model).
+
+  python bench.py --gpus N --steps K --warmup W [--model mlp|bert]
+                  [--batch-size 8192]
+
+One step = one batch of --batch-size rows through the full pipeline
+(synthetic generator data, random-init weights; bf16 inference compute).
+Rank 0 prints ONE JSON line per the driver contract.
+"""
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import time
+
+import torch
+
+
+def build_pipeline(args, device):
+    import arkflow_amd  # noqa: F401  (registers components)
+    from arkflow_amd.inputs.generate import GenerateInput
+    from arkflow_amd.processors.sql import SqlProcessor
+    from arkflow_amd.processors.inference import InferenceProcessor
+    from arkflow_amd.pipeline import Pipeline
+
+    n_features = args.features
+    fields = {f"f{i}": {"dtype": "float32", "low": 0.0, "high": 1.0}
+              for i in range(n_features)}
+    fields["key"] = {"dtype": "int64", "low": 0, "high": 1024}
+    gen = GenerateInput({
+        "batch_size": args.batch_size,
+        "interval": "0ms",
+        "fields": fields,
+        "device": str(device),
+        "seed": 7 + args.rank,
+    })
+    # filter keeps ~80% of rows, then per-row MLP scoring (or BERT per-seq)
+    sql = SqlProcessor({"query": "SELECT * FROM flow WHERE f0 >= 0.2"})
+    if args.model == "bert":
+        gen_tok = GenerateInput({
+            "batch_size": args.batch_size,
+            "interval": "0ms",
+            "fields": {"token": {"dtype": "int64", "low": 0, "high": 30000},
+                       "f0": {"dtype": "float32", "low": 0.0, "high": 1.0}},
+            "device": str(device),
+            "seed": 7 + args.rank,
+        })
+        infer = InferenceProcessor({
+            "model": "bert_base", "seq_len": 128, "device": str(device),
+        })
+        # BERT consumes whole batches (no filter — sequences must stay full)
+        return gen_tok, Pipeline([infer])
+    infer = InferenceProcessor({
+        "model": "mlp_anomaly",
+        "columns": [f"f{i}" for i in range(n_features)],
+        "hidden": [args.hidden, args.hidden],
+        "device": str(device),
+    })
+    return gen, Pipeline([sql, infer])
+
+
+async def run_steps(gen, pipeline, n_steps):
+    rows = 0
+    times = []
+    for _ in range(n_steps):
+        t0 = time.perf_counter()
+        batch, ack = await gen.read()
+        outs = await pipeline.process(batch)
+        for b in outs:
+            rows += b.num_rows
+        if batch.device.type == "cuda":
+            torch.cuda.synchronize()
+        times.append(time.perf_counter() - t0)
+        await ack.ack()
+    return rows, times
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch-size", type=int, default=8192)
+    p.add_argument("--model", choices=["mlp", "bert"], default="mlp")
+    p.add_argument("--features", type=int, default=16)
+    p.add_argument("--hidden", type=int, default=256)
+    args = p.parse_args()
+
+    args.rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", args.rank))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group(backend=backend)
+    if torch.cuda.is_available():
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    gen, pipeline = build_pipeline(args, device)
+    loop = asyncio.new_event_loop()
+
+    # warmup (untimed)
+    loop.run_until_complete(run_steps(gen, pipeline, args.warmup))
+
+    # timed region: barrier + sync on both sides
+    if dist:
+        dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    rows, times = loop.run_until_complete(run_steps(gen, pipeline, args.steps))
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # aggregate across ranks: MAX elapsed, SUM rows
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device.type == "cuda" else "cpu")
+        r = torch.tensor([float(rows)], dtype=torch.float64, device=t.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(r, op=dist.ReduceOp.SUM)
+        elapsed = float(t.item())
+        rows = int(r.item())
+
+    # rows/sec counts INPUT rows processed per wall-second across the job
+    input_rows = args.steps * args.batch_size * world
+    value = input_rows / elapsed
+    p50_ms = statistics.median(times) * 1000.0
+
+    if args.rank == 0:
+        print(json.dumps({
+            "metric": "rows/sec",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world if world > 1 else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "p50_ms": p50_ms,
+            "config": {
+                "model": ("generate→sql(filter)→mlp_anomaly[16→256→256→1]"
+                          if args.model == "mlp"
+                          else "generate→bert_base(12L,768H,seq128)"),
+                "global_batch": args.batch_size * max(world, args.gpus),
+                "seq_len": 128 if args.model == "bert" else 1,
+                "parallelism": f"dp{world if world > 1 else args.gpus}",
+                "batch_size_per_gpu": args.batch_size,
+                "p50_ms": p50_ms,
+                "filter": "WHERE f0 >= 0.2" if args.model == "mlp" else None,
+            },
+        }))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
