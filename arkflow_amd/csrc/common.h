@@ -30,13 +30,15 @@ DEV_INLINE float wave_reduce_max(float v) {
   return v;
 }
 
-// atomic float max/min via ordered-int trick (handles negatives)
-DEV_INLINE int32_t float_flip(float f) {
-  int32_t i = __float_as_int(f);
-  return i < 0 ? ~i : (i | 0x80000000);
+// atomic float max/min via monotone flip to UNSIGNED int order
+// (the flipped values must be compared unsigned — top-bit-set patterns)
+DEV_INLINE uint32_t float_flip(float f) {
+  uint32_t u = (uint32_t)__float_as_int(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
 }
-DEV_INLINE float float_unflip(int32_t i) {
-  return __int_as_float(i < 0 ? (i & 0x7fffffff) : ~i);
+DEV_INLINE float float_unflip(uint32_t u) {
+  return __int_as_float(
+      (int32_t)((u & 0x80000000u) ? (u & 0x7fffffffu) : ~u));
 }
 
 constexpr int cdiv(int a, int b) { return (a + b - 1) / b; }

@@ -14,8 +14,9 @@
 enum RedOp { SUM = 0, MIN = 1, MAX = 2 };
 
 static int32_t float_flip_host(float f) {
-  union { float f; int32_t i; } u{f};
-  return u.i < 0 ? ~u.i : (int32_t)(u.i | 0x80000000);
+  union { float f; uint32_t u; } x{f};
+  uint32_t v = (x.u & 0x80000000u) ? ~x.u : (x.u | 0x80000000u);
+  return (int32_t)v;  // bit pattern; device compares as unsigned
 }
 
 // ---- group-id assignment ----------------------------------------------------
@@ -103,14 +104,14 @@ __global__ void segment_mm_lds_kernel(const float* __restrict__ vals,
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
-    int32_t v = float_flip(vals[i]);
-    if (op == MIN) atomicMin(&part[gids[i]], v);
-    else atomicMax(&part[gids[i]], v);
+    uint32_t v = float_flip(vals[i]);
+    if (op == MIN) atomicMin((uint32_t*)&part[gids[i]], v);
+    else atomicMax((uint32_t*)&part[gids[i]], v);
   }
   __syncthreads();
   for (int j = threadIdx.x; j < g; j += blockDim.x) {
-    if (op == MIN) atomicMin(&out_flipped[j], part[j]);
-    else atomicMax(&out_flipped[j], part[j]);
+    if (op == MIN) atomicMin((uint32_t*)&out_flipped[j], (uint32_t)part[j]);
+    else atomicMax((uint32_t*)&out_flipped[j], (uint32_t)part[j]);
   }
 }
 
@@ -121,16 +122,16 @@ __global__ void segment_mm_global_kernel(const float* __restrict__ vals,
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
-    int32_t v = float_flip(vals[i]);
-    if (op == MIN) atomicMin(&out_flipped[gids[i]], v);
-    else atomicMax(&out_flipped[gids[i]], v);
+    uint32_t v = float_flip(vals[i]);
+    if (op == MIN) atomicMin((uint32_t*)&out_flipped[gids[i]], v);
+    else atomicMax((uint32_t*)&out_flipped[gids[i]], v);
   }
 }
 
 __global__ void unflip_kernel(const int32_t* __restrict__ in,
                               float* __restrict__ out, int g) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < g) out[i] = float_unflip(in[i]);
+  if (i < g) out[i] = float_unflip((uint32_t)in[i]);
 }
 
 __global__ void fill_i64_kernel(int64_t* p, int64_t v, int64_t n) {
