@@ -70,6 +70,8 @@ def init() -> None:
         "arkflow_amd.outputs.file",
         "arkflow_amd.outputs.http",
         "arkflow_amd.temporary.memory_table",
+        "arkflow_amd.wal.store",
+        "arkflow_amd.wal.segment_store",
     ):
         try:
             __import__(mod)

@@ -1,0 +1,136 @@
+"""BaseWindow: shared machinery for the windowing buffers.
+
+Mirrors reference crates/arkflow-plugin/src/buffer/window.rs (per-input-name
+queues, per-input concat on emit, optional SQL join across named inputs) and
+buffer/join.rs (join emitted only when all expected inputs present).
+
+Window contents are device-resident MessageBatches held by reference
+(zero-copy); emit concatenation is a device kernel. Ack semantics follow the
+reference: acks are released when their rows leave the window.
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+from collections import defaultdict, deque
+from typing import Deque, Dict, List, Optional, Tuple
+
+from ..batch import MessageBatch, concat_batches
+from ..spi import Ack, Buffer, VecAck
+
+
+class BaseWindowBuffer(Buffer):
+    def __init__(self, config: dict, resource=None):
+        # per-input-name FIFO of (batch, ack) (reference window.rs DashMap)
+        self.queues: Dict[str, Deque[Tuple[MessageBatch, Ack]]] = \
+            defaultdict(deque)
+        self._notify = asyncio.Event()
+        self._draining = False
+        self._closed = False
+        self.resource = resource
+        join_cfg = config.get("join")
+        self.join_query: Optional[str] = None
+        self.expected_inputs: List[str] = []
+        if join_cfg:
+            self.join_query = join_cfg.get("query")
+            self.expected_inputs = list(
+                join_cfg.get("inputs")
+                or (resource.input_names if resource else [])
+            )
+
+    # ------------------------------------------------------------------ write
+    async def write(self, batch: MessageBatch, ack: Ack) -> None:
+        name = batch.input_name or "default"
+        self.queues[name].append((batch, ack))
+        self.on_write(batch)
+        self._notify.set()
+
+    def on_write(self, batch: MessageBatch) -> None:
+        """Subclass hook (e.g. session gap tracking)."""
+
+    # ------------------------------------------------------------------- read
+    async def read(self) -> Optional[Tuple[MessageBatch, Ack]]:
+        while True:
+            emitted = self.try_emit(draining=self._draining)
+            if emitted is not None:
+                return emitted
+            if self._draining:
+                final = self.drain_remaining()
+                if final is not None:
+                    return final
+                return None
+            self._notify.clear()
+            timeout = self.next_deadline()
+            try:
+                if timeout is None:
+                    await self._notify.wait()
+                else:
+                    await asyncio.wait_for(self._notify.wait(),
+                                           max(timeout, 1e-4))
+            except asyncio.TimeoutError:
+                pass
+
+    async def flush(self) -> None:
+        """End-of-input: emit everything left, then read() returns None."""
+        self._draining = True
+        self._notify.set()
+
+    # --------------------------------------------------------------- triggers
+    def try_emit(self, draining: bool = False
+                 ) -> Optional[Tuple[MessageBatch, Ack]]:
+        """Subclass: emit a window if its trigger fired, else None."""
+        raise NotImplementedError
+
+    def drain_remaining(self) -> Optional[Tuple[MessageBatch, Ack]]:
+        """Default final drain: one combined window of whatever is left."""
+        return self._emit_all()
+
+    def next_deadline(self) -> Optional[float]:
+        """Seconds until the next timer trigger (None = no timer)."""
+        return None
+
+    # ------------------------------------------------------------------- emit
+    def _total_buffered(self) -> int:
+        return sum(len(q) for q in self.queues.values())
+
+    def _emit_all(self) -> Optional[Tuple[MessageBatch, Ack]]:
+        """Pop everything; single input → concat; multi-input → join SQL
+        (reference window.rs:100-178 process_window + join.rs)."""
+        items: Dict[str, List[Tuple[MessageBatch, Ack]]] = {}
+        for name, q in self.queues.items():
+            if q:
+                items[name] = list(q)
+                q.clear()
+        if not items:
+            return None
+        acks = [a for lst in items.values() for _, a in lst]
+        if self.join_query and len(self.expected_inputs) > 1:
+            # join across named inputs; skip (re-buffer) until all present
+            missing = [n for n in self.expected_inputs if n not in items]
+            if missing and not self._draining:
+                for name, lst in items.items():
+                    self.queues[name].extendleft(reversed(lst))
+                return None
+            tables = {
+                name: concat_batches([b for b, _ in lst])
+                for name, lst in items.items()
+            }
+            from ..sql.engine import SqlExecutor
+            first = self.expected_inputs[0]
+            if first in tables:
+                tables.setdefault("flow", tables[first])
+            result = SqlExecutor(self.join_query).execute(tables)
+            return result, VecAck(acks)
+        batches = [b for lst in items.values() for b, _ in lst]
+        try:
+            combined = concat_batches(batches)
+        except ValueError:
+            # heterogeneous schemas: emit the first input's batches only,
+            # re-buffer the rest? Reference concats per input; emit per-input
+            # sequentially instead.
+            name, lst = next(iter(items.items()))
+            for other, olst in list(items.items())[1:]:
+                self.queues[other].extendleft(reversed(olst))
+            combined = concat_batches([b for b, _ in lst])
+            acks = [a for _, a in lst]
+        return combined, VecAck(acks)

@@ -1,0 +1,182 @@
+"""Segmented WAL store — the S3-backend design on a directory object store.
+
+Mirrors reference crates/arkflow-plugin/src/wal/s3.rs (2,203 LoC, the
+reference's most perf-engineered subsystem): in-memory active segment sealed
+on max_entries/max_bytes/flush_interval, parallel PUT workers, batched
+manifest, CRC torn-tail truncation, recovery = manifest ∪ LIST, sealed
+segments ≤ cursor reclaimed. The "object store" here is a directory (same
+interface an S3/object client would implement — no network in this env);
+parallel PUTs use a thread pool like the reference's 1-8 PUT workers.
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+import time
+import zlib
+from concurrent.futures import ThreadPoolExecutor
+from typing import Iterator, List, Optional, Tuple
+
+from ..registry import register
+from .store import decode_frames, encode_frame
+
+
+class SegmentWalStore:
+    def __init__(self, path: str, stream_id: str = "stream",
+                 max_entries: int = 1024, max_bytes: int = 4 << 20,
+                 flush_interval_secs: float = 0.5, put_workers: int = 2,
+                 compress: bool = False):
+        self.dir = os.path.join(path, stream_id)
+        os.makedirs(self.dir, exist_ok=True)
+        self.max_entries = max_entries
+        self.max_bytes = max_bytes
+        self.flush_interval = flush_interval_secs
+        self.compress = compress
+        self.manifest_path = os.path.join(self.dir, "manifest.json")
+        self._active: List[Tuple[int, bytes]] = []
+        self._active_bytes = 0
+        self._seg_counter = 0
+        self._cursor = 0
+        self._lock = threading.Lock()
+        self._pool = ThreadPoolExecutor(max_workers=max(1, put_workers))
+        self._pending_puts = []
+        self._last_seal = time.monotonic()
+        self._manifest = {"watermark": 0, "segments": [], "cursor": 0}
+        self._load_manifest()
+        self._cursor = int(self._manifest.get("cursor", 0))
+        self.max_seq = self._recover_max_seq()
+
+    # ---------------------------------------------------------------- manifest
+    def _load_manifest(self) -> None:
+        try:
+            with open(self.manifest_path) as f:
+                self._manifest = json.load(f)
+            self._seg_counter = self._manifest.get("next_seg", 0)
+        except (OSError, json.JSONDecodeError):
+            pass
+
+    def _store_manifest(self) -> None:
+        self._manifest["next_seg"] = self._seg_counter
+        self._manifest["cursor"] = self._cursor
+        tmp = self.manifest_path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(self._manifest, f)
+        os.replace(tmp, self.manifest_path)
+
+    # ------------------------------------------------------------------ append
+    def append_batch(self, entries: List[Tuple[int, bytes]],
+                     sync: bool = True) -> None:
+        with self._lock:
+            for seq, payload in entries:
+                self._active.append((seq, payload))
+                self._active_bytes += len(payload) + 16
+            if (len(self._active) >= self.max_entries
+                    or self._active_bytes >= self.max_bytes
+                    or (sync and self._active)
+                    or time.monotonic() - self._last_seal
+                    > self.flush_interval):
+                self._seal_locked()
+        if sync:
+            self.wait_puts()
+
+    def _seal_locked(self) -> None:
+        if not self._active:
+            return
+        entries = self._active
+        self._active = []
+        self._active_bytes = 0
+        self._last_seal = time.monotonic()
+        name = f"seg-{self._seg_counter:08d}.wal"
+        self._seg_counter += 1
+        first, last = entries[0][0], entries[-1][0]
+        data = b"".join(
+            encode_frame(s, p, self.compress) for s, p in entries)
+        fut = self._pool.submit(self._put_segment, name, data)
+        self._pending_puts.append(fut)
+        self._manifest["segments"].append(
+            {"name": name, "first": first, "last": last})
+        self._store_manifest()
+
+    def _put_segment(self, name: str, data: bytes) -> None:
+        p = os.path.join(self.dir, name)
+        with open(p + ".tmp", "wb") as f:
+            f.write(data)
+            os.fsync(f.fileno())
+        os.replace(p + ".tmp", p)
+
+    def wait_puts(self) -> None:
+        puts, self._pending_puts = self._pending_puts, []
+        for f in puts:
+            f.result()
+
+    # ------------------------------------------------------------------- read
+    def read_after(self, cursor: int) -> Iterator[Tuple[int, bytes]]:
+        with self._lock:
+            self._seal_locked()
+        self.wait_puts()
+        # recovery = manifest ∪ LIST (reference s3.rs:680+)
+        names = {s["name"] for s in self._manifest.get("segments", [])}
+        names |= {f for f in os.listdir(self.dir)
+                  if f.startswith("seg-") and f.endswith(".wal")}
+        for name in sorted(names):
+            p = os.path.join(self.dir, name)
+            try:
+                with open(p, "rb") as f:
+                    buf = f.read()
+            except OSError:
+                continue
+            for seq, payload in decode_frames(buf):
+                if seq > cursor:
+                    yield seq, payload
+
+    def _recover_max_seq(self) -> int:
+        mx = 0
+        for seq, _ in self.read_after(0):
+            mx = max(mx, seq)
+        return mx
+
+    # ------------------------------------------------------------------ cursor
+    @property
+    def cursor(self) -> int:
+        return self._cursor
+
+    def write_cursor(self, seq: int) -> None:
+        self._cursor = max(self._cursor, seq)
+        self._store_manifest()
+        self._reclaim()
+
+    def _reclaim(self) -> None:
+        """Delete sealed segments entirely ≤ cursor (s3.rs reclamation)."""
+        keep = []
+        for seg in self._manifest.get("segments", []):
+            if seg["last"] <= self._cursor:
+                try:
+                    os.remove(os.path.join(self.dir, seg["name"]))
+                except OSError:
+                    pass
+            else:
+                keep.append(seg)
+        self._manifest["segments"] = keep
+
+    def close(self) -> None:
+        with self._lock:
+            self._seal_locked()
+        self.wait_puts()
+        self._store_manifest()
+        self._pool.shutdown(wait=True)
+
+
+@register("wal_store", "segment",
+          description="Segmented WAL store (sealed segments + manifest + "
+                      "parallel PUT workers; the S3-backend design)")
+def _build_segment_store(config: dict, resource=None) -> SegmentWalStore:
+    return SegmentWalStore(
+        config.get("path", "./wal"),
+        stream_id=config.get("stream_id", "stream"),
+        max_entries=int(config.get("max_entries", 1024)),
+        max_bytes=int(config.get("max_bytes", 4 << 20)),
+        flush_interval_secs=float(config.get("flush_interval_secs", 0.5)),
+        put_workers=int(config.get("put_workers", 2)),
+        compress=bool(config.get("compress", False)),
+    )

@@ -1,0 +1,261 @@
+"""WAL storage: batch serialization + local append-only log store.
+
+Mirrors reference crates/arkflow-core/src/wal/store.rs: a WalStore interface
+(trait :131), frame codec shared by all backends, and a local single-file
+store standing in for redb (:248). Frames use the reference segment codec
+(wal/segment.rs:52-120): ``[seq u64 BE | len u32 BE | payload | crc32 u32
+BE]`` with torn-tail truncation on recovery. Batch payloads are a compact
+columnar encoding (input-name prefix + per-column buffers), the analog of the
+reference's Arrow-IPC-plus-input-name serialization (store.rs:58-117). GPU
+batches are copied to host before serialization (D2H), back on replay.
+"""
+from __future__ import annotations
+
+import os
+import struct
+import zlib
+from typing import Iterator, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..batch import Column, MessageBatch
+from ..registry import register
+
+_MAGIC = b"AWAL"
+_DTYPES = {
+    "float32": torch.float32, "float64": torch.float64,
+    "float16": torch.float16, "bfloat16": torch.bfloat16,
+    "int64": torch.int64, "int32": torch.int32, "int16": torch.int16,
+    "int8": torch.int8, "uint8": torch.uint8, "bool": torch.bool,
+}
+
+
+def _tensor_bytes(t: torch.Tensor) -> bytes:
+    t = t.detach().to("cpu").contiguous()
+    if t.dtype == torch.bfloat16:
+        return t.view(torch.int16).numpy().tobytes()
+    return t.numpy().tobytes()
+
+
+def _tensor_from(data: bytes, dtype: torch.dtype, n: int) -> torch.Tensor:
+    if dtype == torch.bfloat16:
+        return torch.frombuffer(bytearray(data),
+                                dtype=torch.int16).view(torch.bfloat16)[:n]
+    return torch.frombuffer(bytearray(data), dtype=dtype)[:n]
+
+
+def serialize_batch(batch: MessageBatch) -> bytes:
+    parts: List[bytes] = [_MAGIC]
+    name = (batch.input_name or "").encode()
+    parts.append(struct.pack(">H", len(name)))
+    parts.append(name)
+    parts.append(struct.pack(">I", len(batch.columns)))
+    for cname, col in batch.columns.items():
+        nb = cname.encode()
+        parts.append(struct.pack(">H", len(nb)))
+        parts.append(nb)
+        kind = 0 if col.kind == "numeric" else 1
+        has_validity = col.validity is not None
+        parts.append(struct.pack(">BB", kind, int(has_validity)))
+        if kind == 0:
+            dt = str(col.data.dtype).replace("torch.", "").encode()
+            data = _tensor_bytes(col.data)
+            parts.append(struct.pack(">HqI", len(dt), len(col), len(data)))
+            parts.append(dt)
+            parts.append(data)
+        else:
+            data = _tensor_bytes(col.data)
+            offs = _tensor_bytes(col.offsets)
+            parts.append(struct.pack(">qII", len(col), len(data), len(offs)))
+            parts.append(data)
+            parts.append(offs)
+        if has_validity:
+            v = _tensor_bytes(col.validity)
+            parts.append(struct.pack(">I", len(v)))
+            parts.append(v)
+    return b"".join(parts)
+
+
+def deserialize_batch(buf: bytes) -> MessageBatch:
+    if buf[:4] != _MAGIC:
+        raise ValueError("bad WAL payload magic")
+    pos = 4
+    (nlen,) = struct.unpack_from(">H", buf, pos)
+    pos += 2
+    input_name = buf[pos:pos + nlen].decode() or None
+    pos += nlen
+    (ncols,) = struct.unpack_from(">I", buf, pos)
+    pos += 4
+    cols = {}
+    for _ in range(ncols):
+        (clen,) = struct.unpack_from(">H", buf, pos)
+        pos += 2
+        cname = buf[pos:pos + clen].decode()
+        pos += clen
+        kind, has_validity = struct.unpack_from(">BB", buf, pos)
+        pos += 2
+        if kind == 0:
+            dtlen, n, dlen = struct.unpack_from(">HqI", buf, pos)
+            pos += 14
+            dt = _DTYPES[buf[pos:pos + dtlen].decode()]
+            pos += dtlen
+            data = _tensor_from(buf[pos:pos + dlen], dt, n)
+            pos += dlen
+            col = Column("numeric", data)
+        else:
+            n, dlen, olen = struct.unpack_from(">qII", buf, pos)
+            pos += 16
+            data = _tensor_from(buf[pos:pos + dlen], torch.uint8, dlen)
+            pos += dlen
+            offs = _tensor_from(buf[pos:pos + olen], torch.int64, n + 1)
+            pos += olen
+            col = Column("binary", data, offs)
+        if has_validity:
+            (vlen,) = struct.unpack_from(">I", buf, pos)
+            pos += 4
+            col.validity = _tensor_from(buf[pos:pos + vlen], torch.bool, -1)[
+                : len(col)]
+            pos += vlen
+        cols[cname] = col
+    return MessageBatch(cols, input_name)
+
+
+# --------------------------------------------------------------------- frames
+def encode_frame(seq: int, payload: bytes, compress: bool = False) -> bytes:
+    if compress:
+        payload = b"Z" + zlib.compress(payload, 1)
+    else:
+        payload = b"R" + payload
+    crc = zlib.crc32(payload)
+    return struct.pack(">QI", seq, len(payload)) + payload + \
+        struct.pack(">I", crc)
+
+
+def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:
+    """Yields (seq, payload); stops at a torn/corrupt tail
+    (reference segment.rs:89 CRC truncation)."""
+    pos = 0
+    n = len(buf)
+    while pos + 12 <= n:
+        seq, ln = struct.unpack_from(">QI", buf, pos)
+        if pos + 12 + ln + 4 > n:
+            return
+        payload = buf[pos + 12: pos + 12 + ln]
+        (crc,) = struct.unpack_from(">I", buf, pos + 12 + ln)
+        if zlib.crc32(payload) != crc:
+            return
+        if payload[:1] == b"Z":
+            yield seq, zlib.decompress(payload[1:])
+        else:
+            yield seq, payload[1:]
+        pos += 12 + ln + 4
+
+
+# ---------------------------------------------------------------- local store
+class LocalWalStore:
+    """Single-file append-only log + cursor file (the redb analog)."""
+
+    def __init__(self, path: str, stream_id: str = "stream",
+                 compress: bool = False, fsync: bool = True):
+        self.dir = path
+        os.makedirs(path, exist_ok=True)
+        self.log_path = os.path.join(path, f"{stream_id}.wal")
+        self.cursor_path = os.path.join(path, f"{stream_id}.cursor")
+        self.compress = compress
+        self.fsync = fsync
+        self._cursor = self._read_cursor()
+        self._compact_on_open()
+        self._f = open(self.log_path, "ab")
+        self.max_seq = self._scan_max_seq()
+
+    # cursor -------------------------------------------------------------
+    def _read_cursor(self) -> int:
+        try:
+            with open(self.cursor_path, "rb") as f:
+                raw = f.read(12)
+            seq, crc = struct.unpack(">QI", raw)
+            if zlib.crc32(raw[:8]) != crc:
+                return 0
+            return seq
+        except (OSError, struct.error):
+            return 0
+
+    def write_cursor(self, seq: int) -> None:
+        self._cursor = max(self._cursor, seq)
+        raw = struct.pack(">Q", self._cursor)
+        tmp = self.cursor_path + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(raw + struct.pack(">I", zlib.crc32(raw)))
+            if self.fsync:
+                os.fsync(f.fileno())
+        os.replace(tmp, self.cursor_path)
+
+    @property
+    def cursor(self) -> int:
+        return self._cursor
+
+    # log ----------------------------------------------------------------
+    def append_batch(self, entries: List[Tuple[int, bytes]],
+                     sync: bool = True) -> None:
+        for seq, payload in entries:
+            self._f.write(encode_frame(seq, payload, self.compress))
+        self._f.flush()
+        if sync and self.fsync:
+            os.fsync(self._f.fileno())
+
+    def read_after(self, cursor: int) -> Iterator[Tuple[int, bytes]]:
+        self._f.flush()
+        with open(self.log_path, "rb") as f:
+            buf = f.read()
+        for seq, payload in decode_frames(buf):
+            if seq > cursor:
+                yield seq, payload
+
+    def _scan_max_seq(self) -> int:
+        mx = 0
+        try:
+            with open(self.log_path, "rb") as f:
+                buf = f.read()
+            for seq, _ in decode_frames(buf):
+                mx = max(mx, seq)
+        except OSError:
+            pass
+        return mx
+
+    def _compact_on_open(self) -> None:
+        """Drop fully-acked prefix (reference reclaims sealed segments ≤
+        cursor, s3.rs)."""
+        if self._cursor == 0 or not os.path.exists(self.log_path):
+            return
+        keep = []
+        with open(self.log_path, "rb") as f:
+            buf = f.read()
+        for seq, payload in decode_frames(buf):
+            if seq > self._cursor:
+                keep.append((seq, payload))
+        tmp = self.log_path + ".tmp"
+        with open(tmp, "wb") as f:
+            for seq, payload in keep:
+                f.write(encode_frame(seq, payload, self.compress))
+        os.replace(tmp, self.log_path)
+
+    def close(self) -> None:
+        try:
+            self._f.flush()
+            if self.fsync:
+                os.fsync(self._f.fileno())
+            self._f.close()
+        except OSError:
+            pass
+
+
+@register("wal_store", "local",
+          description="Single-file append-only WAL store with CRC frames")
+def _build_local_store(config: dict, resource=None) -> LocalWalStore:
+    return LocalWalStore(
+        config.get("path", "./wal"),
+        stream_id=config.get("stream_id", "stream"),
+        compress=bool(config.get("compress", False)),
+        fsync=bool(config.get("fsync", True)),
+    )

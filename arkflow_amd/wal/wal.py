@@ -1,0 +1,142 @@
+"""Wal coordinator: seq assignment, sync policies, group-commit flusher,
+cursor advance, replay.
+
+Mirrors reference crates/arkflow-core/src/wal/mod.rs: `append` (:352) assigns
+a monotonically increasing seq and either commits synchronously (per_entry)
+or stages + notifies a background flusher (group_commit / periodic,
+:312-338,393-402); `WalAck` (:432) advances the cursor THEN acks the inner
+source ack; recovery replays everything after the cursor (§3.4).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import AsyncIterator, List, Optional, Tuple
+
+from ..config import DurabilityConfig
+from ..registry import build_component
+from ..spi import Ack
+from .store import deserialize_batch, serialize_batch
+
+
+class WalAck(Ack):
+    """Ack decorator: advance WAL cursor, then the source ack
+    (reference wal/mod.rs:432-454)."""
+
+    def __init__(self, wal: "Wal", seq: int, inner: Ack):
+        self.wal = wal
+        self.seq = seq
+        self.inner = inner
+
+    async def ack(self) -> None:
+        await self.wal.advance(self.seq)
+        await self.inner.ack()
+
+
+class Wal:
+    def __init__(self, store, sync_policy: str = "group_commit",
+                 group_window_ms: int = 5, periodic_interval_ms: int = 200,
+                 cursor_flush_every: int = 64):
+        self.store = store
+        self.sync_policy = sync_policy
+        self.group_window = group_window_ms / 1000.0
+        self.periodic_interval = periodic_interval_ms / 1000.0
+        self._seq = store.max_seq
+        self._pending: List[Tuple[int, bytes]] = []
+        self._pending_event: Optional[asyncio.Event] = None
+        self._flusher_task: Optional[asyncio.Task] = None
+        self._closed = False
+        self._cursor_dirty = 0
+        self.cursor_flush_every = cursor_flush_every
+
+    @staticmethod
+    def open(config: DurabilityConfig, stream_id: str = "stream") -> "Wal":
+        store = build_component("wal_store", {
+            "type": config.backend,
+            "path": config.path,
+            "stream_id": stream_id,
+            **config.extra,
+        })
+        return Wal(store, config.sync_policy, config.group_window_ms,
+                   config.periodic_interval_ms)
+
+    # ------------------------------------------------------------------ append
+    async def append(self, batch) -> int:
+        self._seq += 1
+        seq = self._seq
+        payload = serialize_batch(batch)
+        if self.sync_policy == "per_entry":
+            loop = asyncio.get_running_loop()
+            await loop.run_in_executor(
+                None, self.store.append_batch, [(seq, payload)], True)
+            return seq
+        self._pending.append((seq, payload))
+        self._ensure_flusher()
+        self._pending_event.set()
+        return seq
+
+    def _ensure_flusher(self) -> None:
+        if self._flusher_task is None or self._flusher_task.done():
+            self._pending_event = asyncio.Event()
+            self._flusher_task = asyncio.ensure_future(self._flusher())
+
+    async def _flusher(self) -> None:
+        """Background group-commit/periodic flusher
+        (reference wal/mod.rs flush_pending)."""
+        while not self._closed:
+            if self.sync_policy == "group_commit":
+                await self._pending_event.wait()
+                await asyncio.sleep(self.group_window)
+            else:  # periodic
+                await asyncio.sleep(self.periodic_interval)
+            await self.flush_pending()
+            self._pending_event.clear()
+            if self._closed:
+                return
+
+    async def flush_pending(self) -> None:
+        if not self._pending:
+            return
+        batch, self._pending = self._pending, []
+        loop = asyncio.get_running_loop()
+        await loop.run_in_executor(
+            None, self.store.append_batch, batch, True)
+
+    # ----------------------------------------------------------------- advance
+    async def advance(self, seq: int) -> None:
+        """Monotonic cursor advance; flushed every N acks + on close."""
+        if seq <= self.store.cursor:
+            return
+        self._cursor_dirty += 1
+        if self._cursor_dirty >= self.cursor_flush_every:
+            self._cursor_dirty = 0
+            loop = asyncio.get_running_loop()
+            await loop.run_in_executor(None, self.store.write_cursor, seq)
+        else:
+            # in-memory advance only; persisted on the next flush/close
+            self.store._cursor = max(self.store._cursor, seq)
+
+    # ------------------------------------------------------------------ replay
+    async def read_after_cursor(self) -> AsyncIterator:
+        await self.flush_pending()
+        loop = asyncio.get_running_loop()
+        entries = await loop.run_in_executor(
+            None, lambda: list(self.store.read_after(self.store.cursor)))
+        for seq, payload in entries:
+            yield seq, deserialize_batch(payload)
+
+    # ------------------------------------------------------------------- close
+    async def close(self) -> None:
+        self._closed = True
+        await self.flush_pending()
+        if self._flusher_task is not None:
+            if self._pending_event:
+                self._pending_event.set()
+            self._flusher_task.cancel()
+            try:
+                await self._flusher_task
+            except (asyncio.CancelledError, Exception):  # noqa: BLE001
+                pass
+        loop = asyncio.get_running_loop()
+        await loop.run_in_executor(
+            None, self.store.write_cursor, self.store.cursor)
+        await loop.run_in_executor(None, self.store.close)

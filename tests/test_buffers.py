@@ -1,0 +1,180 @@
+"""Window/buffer semantics tests (reference buffer/{memory,tumbling_window,
+sliding_window,session_window}.rs inline tests)."""
+import asyncio
+
+import pytest
+
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.buffers.memory_buffer import MemoryBuffer
+from arkflow_amd.buffers.windows import (
+    SessionWindowBuffer,
+    SlidingWindowBuffer,
+    TumblingWindowBuffer,
+)
+from arkflow_amd.spi import Ack
+
+
+class TAck(Ack):
+    def __init__(self, log, tag):
+        self.log, self.tag = log, tag
+
+    async def ack(self):
+        self.log.append(self.tag)
+
+
+def _mk(vals):
+    return MessageBatch.from_dict({"v": vals}, input_name="in")
+
+
+def test_memory_buffer_capacity(run):
+    async def main():
+        buf = MemoryBuffer({"capacity": 5})
+        log = []
+        for i in range(5):
+            await buf.write(_mk([i]), TAck(log, i))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.num_rows == 5
+        assert batch.column("v").to_pylist() == [0, 1, 2, 3, 4]
+        await ack.ack()
+        assert log == [0, 1, 2, 3, 4]
+
+    run(main())
+
+
+def test_memory_buffer_timeout(run):
+    async def main():
+        buf = MemoryBuffer({"capacity": 1000, "timeout": "50ms"})
+        log = []
+        await buf.write(_mk([1, 2]), TAck(log, "a"))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.num_rows == 2
+        await ack.ack()
+        assert log == ["a"]
+
+    run(main())
+
+
+def test_memory_buffer_drain_on_flush(run):
+    async def main():
+        buf = MemoryBuffer({"capacity": 1000})
+        log = []
+        await buf.write(_mk([1]), TAck(log, "a"))
+        await buf.flush()
+        item = await asyncio.wait_for(buf.read(), 2)
+        assert item is not None and item[0].num_rows == 1
+        assert await asyncio.wait_for(buf.read(), 2) is None
+
+    run(main())
+
+
+def test_tumbling_window(run):
+    async def main():
+        buf = TumblingWindowBuffer({"interval": "60ms"})
+        log = []
+        await buf.write(_mk([1]), TAck(log, 1))
+        await buf.write(_mk([2]), TAck(log, 2))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.column("v").to_pylist() == [1, 2]
+        await ack.ack()
+        assert sorted(log) == [1, 2]
+
+    run(main())
+
+
+def test_sliding_window_ack_on_leave(run):
+    """sliding_window.rs:148-163 — acks released only when batches leave."""
+    async def main():
+        buf = SlidingWindowBuffer({"window_size": 3, "slide_size": 2})
+        log = []
+        for i in range(2):
+            await buf.write(_mk([i]), TAck(log, i))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.column("v").to_pylist() == [0, 1]
+        await ack.ack()
+        assert log == []  # nothing left the window yet
+        for i in range(2, 4):
+            await buf.write(_mk([i]), TAck(log, i))
+        batch2, ack2 = await asyncio.wait_for(buf.read(), 2)
+        # after 4 writes window holds last 3 → batch 0 left
+        await ack2.ack()
+        assert log == [0]
+        assert batch2.column("v").to_pylist() == [1, 2, 3]
+        # drain: remaining window acked
+        await buf.flush()
+        item = await asyncio.wait_for(buf.read(), 2)
+        _, ack3 = item
+        await ack3.ack()
+        assert sorted(log) == [0, 1, 2, 3]
+        assert await asyncio.wait_for(buf.read(), 2) is None
+
+    run(main())
+
+
+def test_session_window_gap(run):
+    async def main():
+        buf = SessionWindowBuffer({"gap": "40ms"})
+        log = []
+        await buf.write(_mk([1]), TAck(log, 1))
+        await asyncio.sleep(0.01)
+        await buf.write(_mk([2]), TAck(log, 2))
+        t0 = asyncio.get_event_loop().time()
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        waited = asyncio.get_event_loop().time() - t0
+        assert batch.column("v").to_pylist() == [1, 2]
+        assert waited >= 0.02  # closed only after the gap
+        await ack.ack()
+        assert sorted(log) == [1, 2]
+
+    run(main())
+
+
+def test_window_join_two_inputs(run):
+    """BaseWindow SQL join across named inputs (reference buffer/join.rs)."""
+    async def main():
+        buf = TumblingWindowBuffer({
+            "interval": "40ms",
+            "join": {
+                "query": "SELECT orders.id, orders.amount, users.name "
+                         "FROM orders JOIN users ON orders.uid = users.uid",
+                "inputs": ["orders", "users"],
+            },
+        })
+        log = []
+        orders = MessageBatch.from_dict(
+            {"id": [1, 2], "amount": [10.0, 20.0], "uid": [7, 8]},
+            input_name="orders")
+        users = MessageBatch.from_dict(
+            {"uid": [7, 8], "name": ["ann", "bob"]}, input_name="users")
+        await buf.write(orders, TAck(log, "o"))
+        await buf.write(users, TAck(log, "u"))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.num_rows == 2
+        assert batch.column("name").to_strlist() == ["ann", "bob"]
+        await ack.ack()
+        assert sorted(log) == ["o", "u"]
+
+    run(main())
+
+
+def test_window_join_waits_for_all_inputs(run):
+    """join.rs:62-130 — no emit until all expected inputs are present."""
+    async def main():
+        buf = TumblingWindowBuffer({
+            "interval": "30ms",
+            "join": {
+                "query": "SELECT a.v FROM a JOIN b ON a.v = b.v",
+                "inputs": ["a", "b"],
+            },
+        })
+        log = []
+        await buf.write(MessageBatch.from_dict({"v": [1]}, input_name="a"),
+                        TAck(log, "a"))
+        # only input a present: first window tick must NOT emit
+        with pytest.raises(asyncio.TimeoutError):
+            await asyncio.wait_for(buf.read(), 0.15)
+        await buf.write(MessageBatch.from_dict({"v": [1]}, input_name="b"),
+                        TAck(log, "b"))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.num_rows == 1
+
+    run(main())
