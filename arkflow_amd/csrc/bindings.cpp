@@ -49,6 +49,9 @@ void launch_bias_act_bf16(const void*, const float*, void*, int64_t, int, int,
                           hipStream_t);
 int launch_attention_bf16(const void*, const void*, const void*, void*, int,
                           int, int, float, hipStream_t);
+void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
+                         const int*, const int*, const int*, const int*,
+                         int64_t*, double*, int32_t*, hipStream_t);
 }
 
 namespace {
@@ -332,6 +335,35 @@ torch::Tensor attention_bf16(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> proto_decode(
+    torch::Tensor data, torch::Tensor offsets,
+    std::vector<int64_t> fno, std::vector<int64_t> kind,
+    std::vector<int64_t> is_float, std::vector<int64_t> slot,
+    int64_t n_int, int64_t n_float) {
+  check_cuda(data, "data");
+  check_cuda(offsets, "offsets");
+  TORCH_CHECK(data.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt64);
+  int64_t n = offsets.numel() - 1;
+  int nf = (int)fno.size();
+  std::vector<int> f(nf), k(nf), isf(nf), sl(nf);
+  for (int i = 0; i < nf; ++i) {
+    f[i] = (int)fno[i]; k[i] = (int)kind[i];
+    isf[i] = (int)is_float[i]; sl[i] = (int)slot[i];
+  }
+  auto out_i = torch::zeros({std::max<int64_t>(n_int, 1), std::max<int64_t>(n, 1)},
+                            data.options().dtype(torch::kInt64));
+  auto out_f = torch::zeros({std::max<int64_t>(n_float, 1), std::max<int64_t>(n, 1)},
+                            data.options().dtype(torch::kFloat64));
+  auto err = torch::zeros({1}, data.options().dtype(torch::kInt32));
+  if (n > 0)
+    launch_proto_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
+                        n, nf, f.data(), k.data(), isf.data(), sl.data(),
+                        out_i.data_ptr<int64_t>(), out_f.data_ptr<double>(),
+                        err.data_ptr<int32_t>(), cur_stream());
+  return {out_i, out_f, err};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -352,4 +384,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = py::none(), py::arg("act") = 0);
   m.def("attention_bf16", &attention_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("scale"));
+  m.def("proto_decode", &proto_decode);
 }

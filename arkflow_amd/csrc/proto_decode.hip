@@ -1,0 +1,137 @@
+// GPU protobuf scalar decode: one thread per message, parsing varint /
+// fixed-width fields straight out of the device-resident binary column
+// (data + offsets) into columnar outputs — the survey's "warp-per-message
+// varint decode kernel writing columnar" mapping for the reference's
+// protobuf_to_arrow processor (crates/arkflow-plugin/src/processor/
+// protobuf.rs — scalar proto3 only). String fields are skipped here (host
+// path handles them); numeric-only schemas stay fully on-device.
+#include "common.h"
+
+#define PROTO_MAX_FIELDS 16
+
+// kind: 0 varint, 1 zigzag, 2 f64, 3 f32, 4 u64(fixed), 5 i64(sfixed),
+//       6 u32(fixed), 7 i32(sfixed), 8 bool
+struct ProtoSpec {
+  int nf;
+  int fno[PROTO_MAX_FIELDS];
+  int kind[PROTO_MAX_FIELDS];
+  int is_float[PROTO_MAX_FIELDS];  // 1 → out_f64 slot, 0 → out_i64 slot
+  int slot[PROTO_MAX_FIELDS];
+};
+
+__global__ void proto_decode_kernel(const uint8_t* __restrict__ data,
+                                    const int64_t* __restrict__ offsets,
+                                    int64_t n_msgs, ProtoSpec spec,
+                                    int64_t* __restrict__ out_i64,  // [ni][n]
+                                    double* __restrict__ out_f64,   // [nd][n]
+                                    int32_t* __restrict__ err_flags) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n_msgs; i += stride) {
+    int64_t pos = offsets[i];
+    const int64_t end = offsets[i + 1];
+    // proto3 defaults
+    while (pos < end) {
+      // read tag varint
+      uint64_t tag = 0;
+      int shift = 0;
+      while (pos < end) {
+        uint8_t b = data[pos++];
+        tag |= (uint64_t)(b & 0x7F) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+        if (shift > 63) { err_flags[0] = 1; return; }
+      }
+      int fno = (int)(tag >> 3);
+      int wt = (int)(tag & 7);
+      // find field slot
+      int fi = -1;
+#pragma unroll
+      for (int f = 0; f < PROTO_MAX_FIELDS; ++f)
+        if (f < spec.nf && spec.fno[f] == fno) fi = f;
+      uint64_t raw = 0;
+      if (wt == 0) {
+        int s2 = 0;
+        while (pos < end) {
+          uint8_t b = data[pos++];
+          raw |= (uint64_t)(b & 0x7F) << s2;
+          if (!(b & 0x80)) break;
+          s2 += 7;
+          if (s2 > 63) { err_flags[0] = 1; return; }
+        }
+      } else if (wt == 1) {
+        if (pos + 8 > end) { err_flags[0] = 1; return; }
+#pragma unroll
+        for (int b = 0; b < 8; ++b) raw |= (uint64_t)data[pos + b] << (8 * b);
+        pos += 8;
+      } else if (wt == 5) {
+        if (pos + 4 > end) { err_flags[0] = 1; return; }
+#pragma unroll
+        for (int b = 0; b < 4; ++b) raw |= (uint64_t)data[pos + b] << (8 * b);
+        pos += 4;
+      } else if (wt == 2) {
+        // length-delimited: skip (strings handled host-side)
+        uint64_t ln = 0;
+        int s2 = 0;
+        while (pos < end) {
+          uint8_t b = data[pos++];
+          ln |= (uint64_t)(b & 0x7F) << s2;
+          if (!(b & 0x80)) break;
+          s2 += 7;
+        }
+        pos += (int64_t)ln;
+        continue;
+      } else {
+        err_flags[0] = 1;
+        return;
+      }
+      if (fi < 0) continue;
+      int kind = spec.kind[fi];
+      if (spec.is_float[fi]) {
+        double v;
+        if (kind == 2) {
+          v = __longlong_as_double((long long)raw);
+        } else if (kind == 3) {
+          v = (double)__uint_as_float((uint32_t)raw);
+        } else {
+          v = (double)raw;
+        }
+        out_f64[(int64_t)spec.slot[fi] * n_msgs + i] = v;
+      } else {
+        int64_t v;
+        switch (kind) {
+          case 1: v = (int64_t)((raw >> 1) ^ (~(raw & 1) + 1)); break;
+          case 5: v = (int64_t)raw; break;
+          case 7: v = (int64_t)(int32_t)(uint32_t)raw; break;
+          case 8: v = raw ? 1 : 0; break;
+          case 0: default: v = (int64_t)raw; break;
+        }
+        out_i64[(int64_t)spec.slot[fi] * n_msgs + i] = v;
+      }
+    }
+  }
+}
+
+extern "C" {
+
+void launch_proto_decode(const uint8_t* data, const int64_t* offsets,
+                         int64_t n_msgs, int nf, const int* fno,
+                         const int* kind, const int* is_float,
+                         const int* slot, int64_t* out_i64, double* out_f64,
+                         int32_t* err_flags, hipStream_t st) {
+  ProtoSpec spec{};
+  spec.nf = nf > PROTO_MAX_FIELDS ? PROTO_MAX_FIELDS : nf;
+  for (int i = 0; i < spec.nf; ++i) {
+    spec.fno[i] = fno[i];
+    spec.kind[i] = kind[i];
+    spec.is_float[i] = is_float[i];
+    spec.slot[i] = slot[i];
+  }
+  int grid = (int)((n_msgs + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  proto_decode_kernel<<<grid, 256, 0, st>>>(data, offsets, n_msgs, spec,
+                                            out_i64, out_f64, err_flags);
+}
+
+}  // extern "C"

@@ -1,0 +1,116 @@
+"""`file` input: CSV/JSON/Parquet/Arrow files → batches.
+
+Mirrors reference crates/arkflow-plugin/src/input/file.rs (:46-90): format
+selection, streaming record batches (8192 rows), optional SQL query applied
+to each batch. Object-store URLs / Ballista offload are out of scope in this
+offline environment (RCCL sharding replaces remote offload);
+local paths and globs are supported.
+"""
+from __future__ import annotations
+
+import glob as globmod
+import os
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..batch import Column, DEFAULT_RECORD_BATCH, MessageBatch
+from ..errors import ConfigError, EOFError_
+from ..registry import register
+from ..spi import Ack, Input, NoopAck
+
+
+def _table_to_batch(table, device=None) -> MessageBatch:
+    import pyarrow as pa
+    cols = {}
+    for name in table.column_names:
+        arr = table.column(name).combine_chunks()
+        if pa.types.is_integer(arr.type):
+            cols[name] = Column.from_numeric(torch.from_numpy(
+                arr.cast(pa.int64()).to_numpy(zero_copy_only=False).copy()))
+        elif pa.types.is_floating(arr.type):
+            cols[name] = Column.from_numeric(torch.from_numpy(
+                arr.cast(pa.float64()).to_numpy(zero_copy_only=False).copy()))
+        elif pa.types.is_boolean(arr.type):
+            cols[name] = Column.from_numeric(torch.from_numpy(
+                arr.to_numpy(zero_copy_only=False).astype(np.bool_)))
+        elif pa.types.is_binary(arr.type) or pa.types.is_large_binary(arr.type):
+            cols[name] = Column.from_bytes(
+                [v if v is not None else b"" for v in arr.to_pylist()])
+        else:
+            cols[name] = Column.from_strings(
+                [str(v) if v is not None else "" for v in arr.to_pylist()])
+    b = MessageBatch(cols, input_name="file")
+    if device is not None:
+        b = b.to(device)
+    return b
+
+
+class FileInput(Input):
+    def __init__(self, config: dict, resource=None):
+        path = config.get("path")
+        if not path:
+            raise ConfigError("file input requires 'path'")
+        self.paths = sorted(globmod.glob(path)) if any(
+            c in path for c in "*?[") else [path]
+        self.format = config.get("format")  # csv|json|parquet|arrow (by ext)
+        self.batch_size = int(config.get("batch_size", DEFAULT_RECORD_BATCH))
+        self.query = config.get("query")
+        self._executor = None
+        if self.query:
+            from ..sql.engine import SqlExecutor
+            self._executor = SqlExecutor(self.query)
+        self.device = getattr(resource, "device", None)
+        self._chunks: List[MessageBatch] = []
+        self._loaded = False
+
+    def _fmt(self, path: str) -> str:
+        if self.format:
+            return self.format
+        ext = os.path.splitext(path)[1].lower().lstrip(".")
+        return {"jsonl": "json", "ndjson": "json", "pq": "parquet",
+                "feather": "arrow", "ipc": "arrow"}.get(ext, ext or "csv")
+
+    def _load_all(self) -> None:
+        import pyarrow as pa
+        for path in self.paths:
+            fmt = self._fmt(path)
+            if fmt == "csv":
+                import pyarrow.csv as pacsv
+                table = pacsv.read_csv(path)
+            elif fmt == "json":
+                import pyarrow.json as pajson
+                table = pajson.read_json(path)
+            elif fmt == "parquet":
+                import pyarrow.parquet as pq
+                table = pq.read_table(path)
+            elif fmt == "arrow":
+                with pa.memory_map(path) as src:
+                    table = pa.ipc.open_file(src).read_all()
+            else:
+                raise ConfigError(f"unknown file format {fmt!r}")
+            for start in range(0, table.num_rows, self.batch_size):
+                chunk = table.slice(start, self.batch_size)
+                batch = _table_to_batch(chunk, self.device)
+                if self._executor is not None:
+                    batch = self._executor.execute({"flow": batch})
+                    batch.input_name = "file"
+                if batch.num_rows:
+                    self._chunks.append(batch)
+        self._loaded = True
+
+    async def read(self) -> Tuple[MessageBatch, Ack]:
+        if not self._loaded:
+            self._load_all()
+        if not self._chunks:
+            raise EOFError_("file input exhausted")
+        return self._chunks.pop(0), NoopAck()
+
+
+@register("input", "file",
+          description="CSV/JSON/Parquet/Arrow file reader (glob paths, "
+                      "optional SQL query per chunk)",
+          example={"type": "file", "path": "data/*.parquet"})
+def _build_file(config: dict, resource=None) -> FileInput:
+    return FileInput(config, resource)

@@ -1,0 +1,103 @@
+"""`http` input: HTTP server endpoint ingesting request bodies.
+
+Mirrors reference crates/arkflow-plugin/src/input/http.rs (:455): an embedded
+server receives POST bodies into a bounded queue; optional bearer-token auth
+(auth_middleware.rs) and a token-bucket rate limiter (rate_limiter.rs).
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+from typing import Optional, Tuple
+
+from ..batch import MessageBatch
+from ..errors import ConfigError, EOFError_
+from ..registry import register
+from ..spi import Ack, Input, NoopAck
+
+
+class TokenBucket:
+    """Simple token-bucket limiter (reference rate_limiter.rs, governor)."""
+
+    def __init__(self, rate: float, burst: int):
+        self.rate = rate
+        self.burst = burst
+        self.tokens = float(burst)
+        self.last = time.monotonic()
+
+    def allow(self) -> bool:
+        now = time.monotonic()
+        self.tokens = min(self.burst, self.tokens + (now - self.last) * self.rate)
+        self.last = now
+        if self.tokens >= 1.0:
+            self.tokens -= 1.0
+            return True
+        return False
+
+
+class HttpInput(Input):
+    def __init__(self, config: dict, resource=None):
+        self.address = config.get("address", "127.0.0.1:0")
+        self.path = config.get("path", "/ingest")
+        self.token = config.get("token")
+        self.queue_size = int(config.get("queue_size", 1024))
+        rate = config.get("rate_limit")
+        self.bucket = TokenBucket(float(rate), int(config.get("burst", rate)))\
+            if rate else None
+        self._q: asyncio.Queue = asyncio.Queue(maxsize=self.queue_size)
+        self._runner = None
+        self._site = None
+        self.port: Optional[int] = None
+        self._closed = False
+
+    async def connect(self) -> None:
+        from aiohttp import web
+
+        async def handler(request):
+            if self.token:
+                auth = request.headers.get("Authorization", "")
+                if auth != f"Bearer {self.token}":
+                    return web.Response(status=401, text="unauthorized")
+            if self.bucket and not self.bucket.allow():
+                return web.Response(status=429, text="rate limited")
+            body = await request.read()
+            if not body:
+                return web.Response(status=400, text="empty body")
+            batch = MessageBatch.from_binary([body], input_name="http")
+            try:
+                self._q.put_nowait(batch)
+            except asyncio.QueueFull:
+                return web.Response(status=503, text="queue full")
+            return web.Response(status=200, text="ok")
+
+        app = web.Application()
+        app.router.add_post(self.path, handler)
+        self._runner = web.AppRunner(app)
+        await self._runner.setup()
+        host, _, port = self.address.partition(":")
+        self._site = web.TCPSite(self._runner, host, int(port or 0))
+        await self._site.start()
+        self.port = self._runner.addresses[0][1] if self._runner.addresses \
+            else None
+
+    async def read(self) -> Tuple[MessageBatch, Ack]:
+        if self._closed:
+            raise EOFError_("http input closed")
+        batch = await self._q.get()
+        if batch is None:
+            raise EOFError_("http input closed")
+        return batch, NoopAck()
+
+    async def close(self) -> None:
+        self._closed = True
+        self._q.put_nowait(None)
+        if self._runner is not None:
+            await self._runner.cleanup()
+
+
+@register("input", "http",
+          description="HTTP server ingest endpoint (bearer auth + rate limit)",
+          example={"type": "http", "address": "127.0.0.1:8085",
+                   "path": "/ingest"})
+def _build_http(config: dict, resource=None) -> HttpInput:
+    return HttpInput(config, resource)

@@ -1,0 +1,128 @@
+"""JSON ↔ columnar processors.
+
+Mirrors reference crates/arkflow-plugin/src/processor/json.rs +
+component/json.rs: `json_to_arrow` parses the ``__value__`` JSON payloads into
+typed columns with the schema inferred from the FIRST record
+(infer_json_schema(..., Some(1)), component/json.rs:27) and optional column
+projection; `arrow_to_json` renders rows as line-delimited JSON bytes in
+``__value__``.
+
+Parse path is the survey's "host-side simdjson-style parse → staging →
+device" mapping: pyarrow's C++ line-JSON reader does the parse, columns are
+then moved to the stream's device in one copy per column.
+"""
+from __future__ import annotations
+
+import json
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from ..batch import Column, DEFAULT_BINARY_VALUE_FIELD, MessageBatch
+from ..errors import ProcessError
+from ..registry import register
+from ..spi import Processor
+
+
+def json_payloads_to_columns(payloads: List[bytes],
+                             projection: Optional[List[str]] = None,
+                             device=None) -> MessageBatch:
+    if not payloads:
+        return MessageBatch({})
+    try:
+        import pyarrow as pa
+        import pyarrow.json as pajson
+        blob = b"\n".join(p.strip() for p in payloads if p.strip())
+        table = pajson.read_json(
+            pa.BufferReader(blob),
+            parse_options=pajson.ParseOptions(newlines_in_values=False),
+        )
+    except Exception as e:  # noqa: BLE001
+        raise ProcessError(f"JSON parse failed: {e}") from e
+    cols = {}
+    names = projection or table.column_names
+    for name in names:
+        if name not in table.column_names:
+            continue
+        arr = table.column(name).combine_chunks()
+        import pyarrow as pa
+        if pa.types.is_integer(arr.type):
+            t = torch.from_numpy(
+                arr.cast(pa.int64()).to_numpy(zero_copy_only=False).copy())
+            col = Column("numeric", t)
+        elif pa.types.is_floating(arr.type):
+            t = torch.from_numpy(
+                arr.cast(pa.float64()).to_numpy(zero_copy_only=False).copy())
+            col = Column("numeric", t)
+        elif pa.types.is_boolean(arr.type):
+            t = torch.from_numpy(
+                arr.to_numpy(zero_copy_only=False).astype(np.bool_))
+            col = Column("numeric", t)
+        elif pa.types.is_string(arr.type) or pa.types.is_large_string(arr.type):
+            col = Column.from_strings(arr.to_pylist())
+        else:
+            # nested / other types → JSON-encoded strings
+            col = Column.from_strings(
+                [json.dumps(v, separators=(",", ":")) if v is not None else ""
+                 for v in arr.to_pylist()])
+        if device is not None:
+            col = col.to(device)
+        cols[name] = col
+    return MessageBatch(cols)
+
+
+class JsonToArrowProcessor(Processor):
+    def __init__(self, config: dict, resource=None):
+        self.projection = config.get("columns") or config.get("projection")
+        self.value_field = config.get("value_field",
+                                      DEFAULT_BINARY_VALUE_FIELD)
+        dev = config.get("device")
+        self.device = torch.device(dev) if dev else getattr(
+            resource, "device", None)
+        self.keep_meta = bool(config.get("keep_meta", True))
+
+    async def process(self, batch: MessageBatch) -> List[MessageBatch]:
+        if batch.num_rows == 0:
+            return []
+        col = batch.columns.get(self.value_field)
+        if col is None or col.kind != "binary":
+            raise ProcessError(
+                f"json_to_arrow: no binary column {self.value_field!r}")
+        out = json_payloads_to_columns(col.to_pylist(), self.projection,
+                                       self.device)
+        if self.keep_meta:
+            meta = {k: (v.to(self.device) if self.device is not None else v)
+                    for k, v in batch.columns.items()
+                    if k.startswith("__meta_") and len(v) == out.num_rows}
+            out = out.with_columns(meta)
+        out.input_name = batch.input_name
+        return [out]
+
+
+class ArrowToJsonProcessor(Processor):
+    def __init__(self, config: dict, resource=None):
+        self.drop_meta = bool(config.get("drop_meta", False))
+
+    async def process(self, batch: MessageBatch) -> List[MessageBatch]:
+        if batch.num_rows == 0:
+            return []
+        b = batch.drop_meta() if self.drop_meta else batch
+        lines = b.to_json_lines()
+        out = MessageBatch.from_binary(lines, input_name=batch.input_name)
+        return [out]
+
+
+@register("processor", "json_to_arrow",
+          description="Parse __value__ JSON payloads into typed device "
+                      "columns (schema inferred from the first record)",
+          example={"type": "json_to_arrow"})
+def _build_j2a(config: dict, resource=None) -> JsonToArrowProcessor:
+    return JsonToArrowProcessor(config, resource)
+
+
+@register("processor", "arrow_to_json",
+          description="Render rows as line-delimited JSON into __value__",
+          example={"type": "arrow_to_json"})
+def _build_a2j(config: dict, resource=None) -> ArrowToJsonProcessor:
+    return ArrowToJsonProcessor(config, resource)

@@ -1,0 +1,299 @@
+"""Processor/codec/component-library tests (reference arkflow-plugin inline
+tests: processor/json.rs, protobuf.rs, batch.rs, python.rs, vrl.rs round-trip,
+codec tests, temporary joins)."""
+import asyncio
+import json
+
+import pytest
+import torch
+
+from arkflow_amd.batch import DEFAULT_BINARY_VALUE_FIELD, MessageBatch
+
+
+def test_json_to_arrow_roundtrip(run):
+    from arkflow_amd.processors.json_proc import (
+        ArrowToJsonProcessor,
+        JsonToArrowProcessor,
+    )
+    payloads = [json.dumps({"a": i, "b": i * 1.5, "s": f"x{i}"}).encode()
+                for i in range(100)]
+    b = MessageBatch.from_binary(payloads, input_name="t")
+    out = run(JsonToArrowProcessor({}, None).process(b))[0]
+    assert out.num_rows == 100
+    assert out.column("a").to_pylist()[:3] == [0, 1, 2]
+    assert out.column("b").to_pylist()[2] == 3.0
+    assert out.column("s").to_strlist()[5] == "x5"
+    back = run(ArrowToJsonProcessor({}, None).process(out))[0]
+    row0 = json.loads(back.binary_values()[0])
+    assert row0 == {"a": 0, "b": 0.0, "s": "x0"}
+
+
+def test_json_projection(run):
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+    b = MessageBatch.from_binary([b'{"a": 1, "b": 2}'])
+    out = run(JsonToArrowProcessor({"columns": ["b"]}, None).process(b))[0]
+    assert out.column_names == ["b"]
+
+
+def test_batch_processor(run):
+    from arkflow_amd.processors.batch_proc import BatchProcessor
+    p = BatchProcessor({"count": 3, "timeout_ms": 60_000})
+    b = MessageBatch.from_dict({"v": [1]})
+    assert run(p.process(b)) == []
+    assert run(p.process(b)) == []
+    out = run(p.process(b))
+    assert len(out) == 1 and out[0].num_rows == 3
+
+
+def test_python_processor_inline_code(run):
+    from arkflow_amd.processors.python_proc import PythonProcessor
+    p = PythonProcessor({
+        "code": "def process(batch):\n"
+                "    import torch\n"
+                "    from arkflow_amd.batch import Column\n"
+                "    return batch.with_columns("
+                "{'v2': Column('numeric', batch.column('v').data * 2)})",
+    })
+    b = MessageBatch.from_dict({"v": [1, 2, 3]})
+    out = run(p.process(b))[0]
+    assert out.column("v2").to_pylist() == [2, 4, 6]
+
+
+def test_python_processor_pyarrow(run):
+    from arkflow_amd.processors.python_proc import PythonProcessor
+    p = PythonProcessor({
+        "convert": "pyarrow",
+        "code": "def process(table):\n"
+                "    return table.select(['v'])",
+    })
+    b = MessageBatch.from_dict({"v": [1.0, 2.0], "w": [0.0, 0.0]})
+    out = run(p.process(b))[0]
+    assert out.column_names == ["v"]
+
+
+def test_expr_processor(run):
+    from arkflow_amd.processors.expr_proc import ExprProcessor
+    p = ExprProcessor({
+        "assignments": {"total": "price * qty",
+                        "flag": "CASE WHEN price > 10 THEN 1 ELSE 0 END"},
+        "drop": ["qty"],
+    })
+    b = MessageBatch.from_dict({"price": [5.0, 20.0], "qty": [2, 3]})
+    out = run(p.process(b))[0]
+    assert out.column("total").to_pylist() == [10.0, 60.0]
+    assert out.column("flag").to_pylist() == [0.0, 1.0]
+    assert "qty" not in out.columns
+
+
+PROTO = """
+syntax = "proto3";
+message Telemetry {
+  double temp = 1;
+  float press = 2;
+  int64 ts = 3;
+  sint32 delta = 4;
+  bool ok = 5;
+  string tag = 6;
+  fixed32 fx = 7;
+}
+"""
+
+
+def test_proto_wire_roundtrip():
+    from arkflow_amd.processors.proto_wire import (
+        ProtoSchema,
+        decode_message,
+        encode_message,
+    )
+    schema = ProtoSchema.parse(PROTO)
+    row = {"temp": 21.5, "press": 1.25, "ts": 171234, "delta": -7,
+           "ok": True, "tag": "dev1", "fx": 42}
+    buf = encode_message(row, schema)
+    back = decode_message(buf, schema)
+    assert back["temp"] == 21.5
+    assert abs(back["press"] - 1.25) < 1e-6
+    assert back["ts"] == 171234
+    assert back["delta"] == -7
+    assert back["ok"] is True
+    assert back["tag"] == "dev1"
+    assert back["fx"] == 42
+    # absent fields → proto3 defaults
+    empty = decode_message(b"", schema)
+    assert empty["temp"] == 0.0 and empty["tag"] == ""
+
+
+def test_protobuf_processors_cpu(run):
+    from arkflow_amd.processors.proto_wire import ProtoSchema, encode_message
+    from arkflow_amd.processors.protobuf_proc import (
+        ArrowToProtobufProcessor,
+        ProtobufToArrowProcessor,
+    )
+    schema = ProtoSchema.parse(PROTO)
+    payloads = [
+        encode_message({"temp": float(i), "press": 0.5 * i, "ts": i,
+                        "delta": -i, "ok": i % 2 == 0, "tag": f"t{i}",
+                        "fx": i}, schema)
+        for i in range(50)
+    ]
+    b = MessageBatch.from_binary(payloads)
+    p2a = ProtobufToArrowProcessor({"proto": PROTO}, None)
+    out = run(p2a.process(b))[0]
+    assert out.column("temp").to_pylist()[:3] == [0.0, 1.0, 2.0]
+    assert out.column("delta").to_pylist()[3] == -3
+    assert out.column("tag").to_strlist()[7] == "t7"
+    a2p = ArrowToProtobufProcessor({"proto": PROTO}, None)
+    back = run(a2p.process(out))[0]
+    assert back.binary_values()[5] == payloads[5]
+
+
+def test_json_codec():
+    from arkflow_amd.codecs.json_codec import JsonCodec
+    c = JsonCodec({}, None)
+    b = MessageBatch.from_dict({"v": [1, 2], "s": ["a", "b"]})
+    enc = c.encode(b)
+    dec = c.decode(enc)
+    assert dec.column("v").to_pylist() == [1, 2]
+    assert dec.column("s").to_strlist() == ["a", "b"]
+
+
+def test_protobuf_codec():
+    from arkflow_amd.codecs.protobuf_codec import ProtobufCodec
+    c = ProtobufCodec({"proto": PROTO}, None)
+    b = MessageBatch.from_dict({
+        "temp": [1.0], "press": [2.0], "ts": [3], "delta": [-1],
+        "ok": [True], "tag": ["z"], "fx": [9],
+    })
+    dec = c.decode(c.encode(b))
+    assert dec.column("temp").to_pylist() == [1.0]
+    assert dec.column("tag").to_strlist() == ["z"]
+
+
+def test_file_input_output(tmp_path, run):
+    from arkflow_amd.inputs.file import FileInput
+    from arkflow_amd.outputs.file import FileOutput
+    # write parquet then read back
+    out = FileOutput({"path": str(tmp_path / "o.parquet")})
+    b = MessageBatch.from_dict({"v": list(range(10)), "s": [f"r{i}" for i in
+                                                            range(10)]})
+    run(out.connect())
+    run(out.write(b))
+    run(out.close())
+    inp = FileInput({"path": str(tmp_path / "o.parquet"), "batch_size": 4})
+    batches = []
+    from arkflow_amd.errors import EOFError_
+    async def drain():
+        while True:
+            try:
+                batch, _ = await inp.read()
+            except EOFError_:
+                return
+            batches.append(batch)
+    run(drain())
+    assert sum(x.num_rows for x in batches) == 10
+    assert batches[0].column("v").to_pylist() == [0, 1, 2, 3]
+    # csv path
+    out2 = FileOutput({"path": str(tmp_path / "o.csv")})
+    run(out2.connect())
+    run(out2.write(b))
+    run(out2.close())
+    inp2 = FileInput({"path": str(tmp_path / "o.csv")})
+    async def one():
+        return await inp2.read()
+    batch, _ = run(one())
+    assert batch.num_rows == 10
+
+
+def test_file_input_with_query(tmp_path, run):
+    from arkflow_amd.inputs.file import FileInput
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    pq.write_table(pa.table({"v": list(range(100))}),
+                   str(tmp_path / "d.parquet"))
+    inp = FileInput({"path": str(tmp_path / "d.parquet"),
+                     "query": "SELECT v FROM flow WHERE v >= 95"})
+    async def one():
+        return await inp.read()
+    batch, _ = run(one())
+    assert batch.column("v").to_pylist() == [95, 96, 97, 98, 99]
+
+
+def test_http_input_output_loopback(run):
+    """HTTP input server + HTTP output client, full loopback."""
+    from arkflow_amd.inputs.http import HttpInput
+    from arkflow_amd.outputs.http import HttpOutput
+
+    async def main():
+        inp = HttpInput({"address": "127.0.0.1:0", "token": "sek"})
+        await inp.connect()
+        out = HttpOutput({
+            "url": f"http://127.0.0.1:{inp.port}/ingest",
+            "token": "sek", "raw_value": True,
+        })
+        await out.connect()
+        b = MessageBatch.from_binary([b'{"x": 1}'])
+        await out.write(b)
+        got, _ = await asyncio.wait_for(inp.read(), 5)
+        assert got.binary_values() == [b'{"x": 1}']
+        # wrong token rejected
+        import aiohttp
+        async with aiohttp.ClientSession() as s:
+            async with s.post(f"http://127.0.0.1:{inp.port}/ingest",
+                              data=b"x") as resp:
+                assert resp.status == 401
+        await out.close()
+        await inp.close()
+
+    run(main())
+
+
+def test_multiple_inputs_fanin(run):
+    from arkflow_amd.inputs.multiple import MultipleInputs
+    from arkflow_amd.spi import Resource
+    from arkflow_amd.errors import EOFError_
+    res = Resource()
+    mi = MultipleInputs({
+        "inputs": {
+            "a": {"type": "generate", "count": 3, "batch_size": 1,
+                  "fields": {"v": {"dtype": "float32"}}},
+            "b": {"type": "generate", "count": 2, "batch_size": 1,
+                  "fields": {"v": {"dtype": "float32"}}},
+        },
+    }, res)
+    assert res.input_names == ["a", "b"]
+
+    async def main():
+        await mi.connect()
+        names = []
+        try:
+            while True:
+                batch, _ = await asyncio.wait_for(mi.read(), 5)
+                names.append(batch.input_name)
+        except EOFError_:
+            pass
+        await mi.close()
+        return names
+
+    names = run(main())
+    assert sorted(names) == ["a", "a", "a", "b", "b"]
+
+
+def test_sql_temporary_join(run):
+    """SQL processor joining a temporary lookup table
+    (reference processor/sql.rs:148-183)."""
+    from arkflow_amd.processors.sql import SqlProcessor
+    from arkflow_amd.spi import Resource
+    from arkflow_amd.temporary.memory_table import MemoryTemporary
+    res = Resource()
+    temp = MemoryTemporary({"key_column": "uid",
+                            "rows": [{"uid": 1, "name": "ann"},
+                                     {"uid": 2, "name": "bob"}]})
+    res.temporaries = {"users": temp}
+    p = SqlProcessor({
+        "query": "SELECT flow.uid, users.name FROM flow "
+                 "JOIN users ON flow.uid = users.uid ORDER BY flow.uid",
+        "temporaries": [{"name": "users", "key": "uid"}],
+    }, res)
+    b = MessageBatch.from_dict({"uid": [2, 1, 2]})
+    out = run(p.process(b))[0]
+    assert out.column("uid").to_pylist() == [1, 2, 2]
+    assert out.column("name").to_strlist() == ["ann", "bob", "bob"]
