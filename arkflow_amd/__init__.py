@@ -60,6 +60,7 @@ def init() -> None:
         "arkflow_amd.processors.python_proc",
         "arkflow_amd.processors.protobuf_proc",
         "arkflow_amd.processors.expr_proc",
+        "arkflow_amd.processors.repartition",
         "arkflow_amd.buffers.memory_buffer",
         "arkflow_amd.buffers.windows",
         "arkflow_amd.codecs.json_codec",

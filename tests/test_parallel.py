@@ -1,0 +1,76 @@
+"""Multi-process distributed tests over gloo (world_size 2) — the CPU stand-in
+for the RCCL path, as the reference gates broker tests behind containers
+(SURVEY §4.6)."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _run_repartition(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.parallel.dist import repartition_by_key
+
+    n = 100
+    batch = MessageBatch.from_dict({
+        "k": torch.arange(rank * n, (rank + 1) * n, dtype=torch.int64),
+        "v": torch.arange(n, dtype=torch.float32) + rank * 1000,
+        "s": [f"r{rank}-{i}" for i in range(n)],
+    })
+    out = repartition_by_key(batch, "k")
+    # every key must land on exactly the rank selected by the hash
+    z = out.column("k").data * 0x9E3779B97F4A7C15
+    z = torch.bitwise_xor(z, z >> 30) * -0x40A7B892E31B1A47
+    z = torch.bitwise_xor(z, z >> 27)
+    dest = torch.remainder(z, world).abs()
+    assert bool((dest == rank).all()), "row landed on wrong rank"
+    # string column survives with matching k suffixes
+    ks = out.column("k").to_pylist()
+    ss = out.column("s").to_strlist()
+    for k, s in zip(ks, ss):
+        assert s.endswith(f"-{k % n}")
+    results[rank] = out.num_rows
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_repartition_two_ranks():
+    port = 29531
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_run_repartition, args=(2, port, results), nprocs=2,
+                 join=True)
+        total = results[0] + results[1]
+        assert total == 200  # no rows lost or duplicated
+
+
+def _run_bench_style(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from arkflow_amd.parallel.dist import all_reduce_scalar
+    total = all_reduce_scalar(float(rank + 1), "sum")
+    assert total == 3.0
+    mx = all_reduce_scalar(float(rank), "max")
+    assert mx == 1.0
+    results[rank] = total
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_all_reduce_scalar_two_ranks():
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_run_bench_style, args=(2, 29532, results), nprocs=2,
+                 join=True)
+        assert results[0] == results[1] == 3.0
