@@ -1,0 +1,156 @@
+"""Node HTTP API tests via ASGI transport (the reference tests its axum
+Router with tower::oneshot the same way — no network)."""
+import asyncio
+
+import httpx
+import pytest
+
+import arkflow_amd as af
+from arkflow_amd.config import EngineConfig
+from arkflow_amd.server.api import create_app
+
+
+def _engine():
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "s1",
+            "input": {"type": "generate", "batch_size": 5, "interval": "10ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "pipeline": {"thread_num": 1, "processors": [
+                {"type": "sql", "query": "SELECT * FROM flow WHERE v >= 0"},
+            ]},
+            "output": {"type": "drop"},
+        }],
+        "server": {"enabled": True, "address": "127.0.0.1:0"},
+    })
+    return af.Engine(cfg)
+
+
+def test_api_surface(run):
+    async def main():
+        eng = _engine()
+        for sc in eng.config.streams:
+            eng.runtime.register(sc)
+        await eng.runtime.start("s1")
+        eng.ready = True
+        eng.running = True
+        app = create_app(eng)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://node") as c:
+            assert (await c.get("/health")).json() == {"status": "ok"}
+            assert (await c.get("/readiness")).status_code == 200
+            assert (await c.get("/liveness")).status_code == 200
+
+            r = await c.get("/api/v1/system/status")
+            assert r.json()["streams"] == 1
+
+            streams = (await c.get("/api/v1/streams")).json()
+            assert streams[0]["id"] == "s1"
+            assert streams[0]["state"] == "running"
+
+            one = (await c.get("/api/v1/streams/s1")).json()
+            assert one["convergence"] == "converged"
+            assert (await c.get("/api/v1/streams/nope")).status_code == 404
+
+            # lifecycle: stop → snapshot stopped → restart
+            op = (await c.post("/api/v1/streams/s1/stop")).json()
+            assert op["state"] in ("succeeded", "running")
+            one = (await c.get("/api/v1/streams/s1")).json()
+            assert one["state"] == "stopped"
+            ops = (await c.get("/api/v1/operations")).json()
+            assert len(ops) >= 1
+            opd = (await c.get(f"/api/v1/operations/{ops[-1]['id']}")).json()
+            assert opd["stream_id"] == "s1"
+
+            evs = (await c.get("/api/v1/events")).json()
+            kinds = {e["kind"] for e in evs}
+            assert {"registered", "running", "stopping"} <= kinds
+
+            comps = (await c.get("/api/v1/components")).json()
+            names = {(m["kind"], m["name"]) for m in comps}
+            assert ("processor", "sql") in names
+            assert ("input", "generate") in names
+            md = (await c.get("/api/v1/components/processor/sql")).json()
+            assert "query" in str(md["example"])
+
+            schema = (await c.get("/api/v1/schema")).json()
+            assert "streams" in schema["properties"]
+
+            # config validate / apply
+            bad = {"streams": [{"id": "x", "input": {"type": "nope"},
+                                "output": {"type": "drop"}}]}
+            v = (await c.post("/api/v1/configuration/validate",
+                              json=bad)).json()
+            assert not v["valid"]
+
+            good = {"streams": [{
+                "id": "s2",
+                "input": {"type": "generate", "batch_size": 1,
+                          "interval": "50ms",
+                          "fields": {"v": {"dtype": "float32"}}},
+                "output": {"type": "drop"},
+            }]}
+            ap = (await c.post("/api/v1/configuration/apply",
+                               json=good)).json()
+            assert ap["applied"] and ap["version"] == 1
+            streams = (await c.get("/api/v1/streams")).json()
+            ids = {s["id"] for s in streams}
+            assert ids == {"s2"}  # s1 removed, s2 added
+            vs = (await c.get("/api/v1/configuration/versions")).json()
+            assert len(vs) == 1
+
+            # prometheus text
+            m = (await c.get("/metrics")).text
+            assert "arkflow_input_messages_total" in m
+            assert 'stream="s2"' in m
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)
+
+
+def test_api_token_auth(run):
+    async def main():
+        eng = _engine()
+        eng.config.server.token = "sekret"
+        app = create_app(eng)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://node") as c:
+            assert (await c.get("/api/v1/streams")).status_code == 401
+            ok = await c.get("/api/v1/streams",
+                             headers={"Authorization": "Bearer sekret"})
+            assert ok.status_code == 200
+            # health stays open
+            assert (await c.get("/health")).status_code == 200
+
+    run(main())
+
+
+def test_cli_components_and_schema(capsys):
+    from arkflow_amd.cli import main
+    assert main(["components", "list"]) == 0
+    out = capsys.readouterr().out
+    assert "generate" in out and "sql" in out
+    assert main(["components", "show", "processor", "sql"]) == 0
+    assert main(["schema"]) == 0
+
+
+def test_cli_validate(tmp_path, capsys):
+    from arkflow_amd.cli import main
+    p = tmp_path / "c.yaml"
+    p.write_text("""
+streams:
+  - id: s1
+    input: {type: generate, batch_size: 1, count: 1}
+    output: {type: drop}
+""")
+    assert main(["--config", str(p), "--validate"]) == 0
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("""
+streams:
+  - id: s1
+    input: {type: nonexistent}
+    output: {type: drop}
+""")
+    assert main(["--config", str(bad), "--validate"]) == 1
