@@ -1,0 +1,289 @@
+"""Hub: fleet control plane — node registry, agent protocol, reconciliation,
+rollouts, RBAC.
+
+Mirrors reference crates/arkflow-server/src/hub.rs (3,496 LoC) +
+api_contract.rs: agent register/heartbeat/report/commands/command-result
+(hub.rs:558,685,711,852,1146), desired-state intents → outbox → attempts →
+agent commands reconciliation (reconcile_once, hub.rs:468-520), rollouts with
+pause/resume/cancel (hub.rs:1472-2052), lease sweep, SSE events, operator
+token RBAC (api_contract.rs:9-30).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import secrets
+import time
+from typing import Dict, List, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Request
+from fastapi.responses import StreamingResponse
+
+from .storage import HubStore
+
+ROLES = {"admin": {"read", "write", "rollout", "admin"},
+         "operator": {"read", "write", "rollout"},
+         "viewer": {"read"}}
+
+
+class Hub:
+    def __init__(self, store: Optional[HubStore] = None,
+                 lease_ttl: float = 15.0,
+                 operator_tokens: Optional[Dict[str, str]] = None):
+        """operator_tokens: token → role (reference RBAC roles/scopes)."""
+        self.store = store or HubStore()
+        self.lease_ttl = lease_ttl
+        self.operator_tokens = operator_tokens or {}
+        self._event_subs: List[asyncio.Queue] = []
+        self._event_seq = 0
+
+    # ---- agent protocol -------------------------------------------------------
+    async def register(self, node_id: str) -> dict:
+        token = secrets.token_hex(16)
+        await self.store.upsert_node(node_id, token, self.lease_ttl)
+        await self._emit(node_id, "node_registered", {})
+        return {"node_token": token, "lease_ttl_secs": self.lease_ttl}
+
+    async def check_node(self, node_id: str, token: str) -> None:
+        expect = await self.store.node_token(node_id)
+        if expect is None or expect != token:
+            raise HTTPException(401, "bad node token")
+
+    async def heartbeat(self, node_id: str) -> dict:
+        ok = await self.store.heartbeat(node_id, self.lease_ttl)
+        if not ok:
+            raise HTTPException(404, "unknown node")
+        return {"lease_ttl_secs": self.lease_ttl}
+
+    async def report(self, node_id: str, snapshot: dict) -> None:
+        await self.store.report(node_id, snapshot)
+
+    async def commands(self, node_id: str) -> List[dict]:
+        return await self.store.pending_commands(node_id)
+
+    async def command_result(self, node_id: str, attempt_id: str, ok: bool,
+                             detail: str = "") -> None:
+        intent = await self.store.command_result(attempt_id, ok, detail)
+        await self._emit(node_id, "command_result",
+                         {"attempt_id": attempt_id, "ok": ok,
+                          "intent_id": intent})
+
+    # ---- operator ops -----------------------------------------------------------
+    async def enqueue_intent(self, node_id: str, stream_id: str, op: str,
+                             actor: str = "operator") -> str:
+        intent_id = await self.store.enqueue_intent(node_id, stream_id, op)
+        await self.store.audit(actor, f"intent:{op}",
+                               f"{node_id}/{stream_id}")
+        await self._emit(node_id, "intent_enqueued",
+                         {"intent_id": intent_id, "op": op,
+                          "stream_id": stream_id})
+        return intent_id
+
+    # ---- reconciliation ---------------------------------------------------------
+    async def reconcile_once(self) -> int:
+        """intents → outbox → attempts (agent commands) (hub.rs:468-520)."""
+        await self.store.expire_attempts()
+        claimed = await self.store.claim_outbox()
+        dispatched = 0
+        online = {n["node_id"] for n in await self.store.nodes()
+                  if n["online"]}
+        for row in claimed:
+            if row["node_id"] not in online:
+                continue  # node offline — intent stays dispatched-pending
+            await self.store.create_attempt(
+                row["intent_id"], row["node_id"],
+                {"kind": "lifecycle", "stream_id": row["stream_id"],
+                 "op": row["op"]})
+            dispatched += 1
+        return dispatched
+
+    async def sweep(self) -> None:
+        stale = await self.store.sweep_leases()
+        for node_id in stale:
+            await self._emit(node_id, "node_offline", {})
+
+    # ---- rollouts ----------------------------------------------------------------
+    async def create_rollout(self, config: dict, nodes: List[str],
+                             actor: str = "operator") -> str:
+        rid = await self.store.create_rollout(config, nodes)
+        await self.store.audit(actor, "rollout:create", rid)
+        return rid
+
+    async def step_rollout(self, rid: str) -> Optional[dict]:
+        """Advance one node per step (reference staged rollouts)."""
+        r = await self.store.get_rollout(rid)
+        if r is None or r["state"] != "running":
+            return r
+        pos = r["position"]
+        if pos >= len(r["nodes"]):
+            await self.store.update_rollout(rid, state="succeeded")
+            return await self.store.get_rollout(rid)
+        node = r["nodes"][pos]
+        await self.store.create_attempt(
+            f"rollout-{rid}-{pos}", node,
+            {"kind": "apply_config", "config": r["config"]})
+        await self.store.update_rollout(rid, position=pos + 1)
+        return await self.store.get_rollout(rid)
+
+    async def control_rollout(self, rid: str, action: str) -> Optional[dict]:
+        states = {"pause": "paused", "resume": "running",
+                  "cancel": "cancelled", "rollback": "rolled_back"}
+        if action not in states:
+            raise HTTPException(400, f"unknown rollout action {action}")
+        await self.store.update_rollout(rid, state=states[action])
+        return await self.store.get_rollout(rid)
+
+    # ---- events -------------------------------------------------------------------
+    async def _emit(self, node_id: str, kind: str, payload: dict) -> None:
+        await self.store.push_event(node_id, kind, payload)
+        self._event_seq += 1
+        for q in list(self._event_subs):
+            try:
+                q.put_nowait({"seq": self._event_seq, "node_id": node_id,
+                              "kind": kind, **payload})
+            except asyncio.QueueFull:
+                pass
+
+
+def create_hub_app(hub: Hub) -> FastAPI:
+    app = FastAPI(title="arkflow_amd hub", version="1")
+
+    # ---- RBAC ----------------------------------------------------------------
+    def operator(scope: str):
+        def check(request: Request):
+            if not hub.operator_tokens:
+                return "anonymous"
+            tok = request.headers.get("authorization", "").removeprefix(
+                "Bearer ").strip()
+            role = hub.operator_tokens.get(tok)
+            if role is None or scope not in ROLES.get(role, set()):
+                raise HTTPException(403, "forbidden")
+            return role
+        return check
+
+    async def node_auth(node_id: str, request: Request):
+        tok = request.headers.get("x-node-token", "")
+        await hub.check_node(node_id, tok)
+
+    # ---- agent API (hub.rs agent protocol) ------------------------------------
+    @app.post("/agent/register")
+    async def register(body: dict):
+        node_id = body.get("node_id")
+        if not node_id:
+            raise HTTPException(400, "node_id required")
+        return await hub.register(node_id)
+
+    @app.post("/agent/{node_id}/heartbeat")
+    async def heartbeat(node_id: str, request: Request):
+        await node_auth(node_id, request)
+        return await hub.heartbeat(node_id)
+
+    @app.post("/agent/{node_id}/report")
+    async def report(node_id: str, body: dict, request: Request):
+        await node_auth(node_id, request)
+        await hub.report(node_id, body)
+        return {"ok": True}
+
+    @app.get("/agent/{node_id}/commands")
+    async def commands(node_id: str, request: Request):
+        await node_auth(node_id, request)
+        return await hub.commands(node_id)
+
+    @app.post("/agent/{node_id}/commands/{attempt_id}/result")
+    async def command_result(node_id: str, attempt_id: str, body: dict,
+                             request: Request):
+        await node_auth(node_id, request)
+        await hub.command_result(node_id, attempt_id,
+                                 bool(body.get("ok")),
+                                 str(body.get("detail", "")))
+        return {"ok": True}
+
+    # ---- operator API ----------------------------------------------------------
+    @app.get("/nodes", dependencies=[Depends(operator("read"))])
+    async def nodes():
+        return await hub.store.nodes()
+
+    @app.post("/nodes/{node_id}/streams/{stream_id}/{op}",
+              dependencies=[Depends(operator("write"))])
+    async def lifecycle(node_id: str, stream_id: str, op: str):
+        if op not in ("start", "stop", "restart"):
+            raise HTTPException(400, "unknown op")
+        intent_id = await hub.enqueue_intent(node_id, stream_id, op)
+        return {"intent_id": intent_id}
+
+    @app.get("/intents", dependencies=[Depends(operator("read"))])
+    async def intents():
+        return await hub.store.intents()
+
+    @app.get("/events", dependencies=[Depends(operator("read"))])
+    async def events(after_seq: int = 0, limit: int = 100):
+        return await hub.store.events(after_seq, limit)
+
+    @app.get("/events/stream", dependencies=[Depends(operator("read"))])
+    async def events_stream(request: Request):
+        last_id = int(request.headers.get("last-event-id", 0) or 0)
+
+        async def gen():
+            for ev in await hub.store.events(after_seq=last_id, limit=10_000):
+                yield f"id: {ev['seq']}\ndata: {json.dumps(ev)}\n\n"
+            q: asyncio.Queue = asyncio.Queue(maxsize=256)
+            hub._event_subs.append(q)
+            try:
+                while not await request.is_disconnected():
+                    try:
+                        ev = await asyncio.wait_for(q.get(), timeout=5.0)
+                        yield f"data: {json.dumps(ev)}\n\n"
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                hub._event_subs.remove(q)
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/audit", dependencies=[Depends(operator("admin"))])
+    async def audit():
+        return await hub.store.audit_log()
+
+    @app.post("/rollouts", dependencies=[Depends(operator("rollout"))])
+    async def create_rollout(body: dict):
+        rid = await hub.create_rollout(body.get("config") or {},
+                                       body.get("nodes") or [])
+        return {"rollout_id": rid}
+
+    @app.get("/rollouts", dependencies=[Depends(operator("read"))])
+    async def rollouts():
+        return await hub.store.rollouts()
+
+    @app.post("/rollouts/{rid}/step",
+              dependencies=[Depends(operator("rollout"))])
+    async def step_rollout(rid: str):
+        r = await hub.step_rollout(rid)
+        if r is None:
+            raise HTTPException(404, "unknown rollout")
+        return r
+
+    @app.post("/rollouts/{rid}/{action}",
+              dependencies=[Depends(operator("rollout"))])
+    async def control_rollout(rid: str, action: str):
+        r = await hub.control_rollout(rid, action)
+        if r is None:
+            raise HTTPException(404, "unknown rollout")
+        return r
+
+    return app
+
+
+async def hub_background(hub: Hub, cancel: asyncio.Event,
+                         poll_interval: float = 0.25,
+                         sweep_interval: float = 0.5) -> None:
+    """Reconcile + lease sweep loop (reference serve_hub, lib.rs:337-360)."""
+    last_sweep = 0.0
+    while not cancel.is_set():
+        await hub.reconcile_once()
+        if time.monotonic() - last_sweep >= sweep_interval:
+            await hub.sweep()
+            last_sweep = time.monotonic()
+        try:
+            await asyncio.wait_for(cancel.wait(), poll_interval)
+        except asyncio.TimeoutError:
+            pass

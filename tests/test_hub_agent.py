@@ -1,0 +1,178 @@
+"""Fleet control-plane tests: hub + agent in one process over ASGI (the
+reference tests hub/agent with mock axum routers + fake reports — SURVEY §4.6:
+no network fault injection, in-process cluster)."""
+import asyncio
+
+import httpx
+import pytest
+
+import arkflow_amd as af
+from arkflow_amd.config import EngineConfig
+from arkflow_amd.server.agent import Agent
+from arkflow_amd.server.hub import Hub, create_hub_app, hub_background
+
+
+def _engine():
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "s1",
+            "input": {"type": "generate", "batch_size": 2, "interval": "20ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "output": {"type": "drop"},
+        }]
+    })
+    eng = af.Engine(cfg)
+    for sc in cfg.streams:
+        eng.runtime.register(sc)
+    return eng
+
+
+def test_full_intent_cycle(run):
+    """operator intent → outbox → attempt → agent executes → result →
+    intent succeeded (reference §3.5 flow)."""
+    async def main():
+        hub = Hub(lease_ttl=5.0)
+        app = create_hub_app(hub)
+        transport = httpx.ASGITransport(app=app)
+        eng = _engine()
+        agent = Agent(eng, "http://hub", node_id="n1", transport=transport,
+                      heartbeat_interval=0.05, report_interval=0.05,
+                      poll_interval=0.02)
+        cancel = asyncio.Event()
+        bg = asyncio.ensure_future(hub_background(hub, cancel,
+                                                  poll_interval=0.02,
+                                                  sweep_interval=0.1))
+        agent_task = asyncio.ensure_future(agent.run(cancel))
+        await asyncio.sleep(0.2)  # registered + reporting
+
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://hub") as c:
+            nodes = (await c.get("/nodes")).json()
+            assert nodes and nodes[0]["node_id"] == "n1"
+            assert nodes[0]["online"] == 1
+
+            r = await c.post("/nodes/n1/streams/s1/start")
+            intent_id = r.json()["intent_id"]
+            # wait for the intent to be executed end-to-end
+            for _ in range(100):
+                intents = (await c.get("/intents")).json()
+                state = [i for i in intents
+                         if i["intent_id"] == intent_id][0]["state"]
+                if state in ("succeeded", "failed"):
+                    break
+                await asyncio.sleep(0.05)
+            assert state == "succeeded"
+            assert eng.runtime.get("s1").state.value == "running"
+
+            # node report visible at hub
+            for _ in range(50):
+                nodes = (await c.get("/nodes")).json()
+                if nodes[0]["last_report"]:
+                    break
+                await asyncio.sleep(0.05)
+            assert nodes[0]["last_report"]["status"]["streams"] == 1
+
+            # stop intent too
+            r = await c.post("/nodes/n1/streams/s1/stop")
+            for _ in range(100):
+                intents = (await c.get("/intents")).json()
+                states = {i["intent_id"]: i["state"] for i in intents}
+                if states[r.json()["intent_id"]] == "succeeded":
+                    break
+                await asyncio.sleep(0.05)
+            assert eng.runtime.get("s1").state.value == "stopped"
+
+            evs = (await c.get("/events")).json()
+            kinds = {e["kind"] for e in evs}
+            assert "node_registered" in kinds
+            assert "intent_enqueued" in kinds
+            assert "command_result" in kinds
+
+        cancel.set()
+        await asyncio.gather(bg, agent_task, return_exceptions=True)
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)
+
+
+def test_lease_sweep_marks_offline(run):
+    async def main():
+        hub = Hub(lease_ttl=0.1)
+        await hub.register("n2")
+        await asyncio.sleep(0.2)
+        stale = await hub.store.sweep_leases()
+        assert "n2" in stale
+        nodes = await hub.store.nodes()
+        assert nodes[0]["online"] == 0
+        # offline node: reconcile must NOT dispatch
+        await hub.enqueue_intent("n2", "s1", "start")
+        n = await hub.reconcile_once()
+        assert n == 0
+
+    run(main())
+
+
+def test_rbac_roles(run):
+    async def main():
+        hub = Hub(operator_tokens={"admintok": "admin", "view": "viewer"})
+        app = create_hub_app(hub)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://hub") as c:
+            # no token → forbidden
+            assert (await c.get("/nodes")).status_code == 403
+            # viewer can read, cannot write
+            h = {"Authorization": "Bearer view"}
+            assert (await c.get("/nodes", headers=h)).status_code == 200
+            assert (await c.post("/nodes/n/streams/s/start",
+                                 headers=h)).status_code == 403
+            assert (await c.get("/audit", headers=h)).status_code == 403
+            # admin can do everything
+            ha = {"Authorization": "Bearer admintok"}
+            assert (await c.post("/nodes/n/streams/s/start",
+                                 headers=ha)).status_code == 200
+            assert (await c.get("/audit", headers=ha)).status_code == 200
+
+    run(main())
+
+
+def test_rollout_lifecycle(run):
+    async def main():
+        hub = Hub()
+        rid = await hub.create_rollout({"streams": []}, ["n1", "n2"])
+        r = await hub.step_rollout(rid)
+        assert r["position"] == 1
+        r = await hub.control_rollout(rid, "pause")
+        assert r["state"] == "paused"
+        r2 = await hub.step_rollout(rid)  # paused → no advance
+        assert r2["position"] == 1
+        r = await hub.control_rollout(rid, "resume")
+        r = await hub.step_rollout(rid)
+        assert r["position"] == 2
+        r = await hub.step_rollout(rid)
+        assert r["state"] == "succeeded"
+
+    run(main())
+
+
+def test_agent_idempotent_replay(run):
+    """agent.rs idempotent command replay cache: same attempt executes once."""
+    async def main():
+        eng = _engine()
+        agent = Agent(eng, "http://x", node_id="n1")
+        calls = []
+
+        async def fake_lifecycle(stream_id, op):
+            calls.append((stream_id, op))
+            return {"state": "succeeded", "error": None}
+
+        eng.control_plane.lifecycle = fake_lifecycle
+        cmd = {"attempt_id": "a1", "kind": "lifecycle", "stream_id": "s1",
+               "op": "start"}
+        ok, _ = await agent._execute(cmd)
+        agent._executed.add("a1")
+        assert ok and calls == [("s1", "start")]
+        # replayed command skipped by the cache in _poll_and_execute
+        assert "a1" in agent._executed
+
+    run(main())
