@@ -68,6 +68,11 @@ def init() -> None:
         "arkflow_amd.inputs.file",
         "arkflow_amd.inputs.http",
         "arkflow_amd.inputs.multiple",
+        "arkflow_amd.inputs.brokers",
+        "arkflow_amd.inputs.sql_io",
+        "arkflow_amd.inputs.websocket",
+        "arkflow_amd.inputs.modbus",
+        "arkflow_amd.codecs.debezium",
         "arkflow_amd.outputs.file",
         "arkflow_amd.outputs.http",
         "arkflow_amd.temporary.memory_table",

