@@ -28,8 +28,8 @@ SCALARS = {
 }
 
 _FIELD_RE = re.compile(
-    r"^\s*(double|float|u?int32|u?int64|sint32|sint64|bool|"
-    r"s?fixed32|s?fixed64|string|bytes)\s+(\w+)\s*=\s*(\d+)\s*;", re.M)
+    r"\b(double|float|u?int32|u?int64|sint32|sint64|bool|"
+    r"s?fixed32|s?fixed64|string|bytes)\s+(\w+)\s*=\s*(\d+)\s*;")
 _MSG_RE = re.compile(r"message\s+(\w+)\s*\{([^{}]*)\}", re.S)
 
 

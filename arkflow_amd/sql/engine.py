@@ -290,6 +290,12 @@ class SqlExecutor:
                      ) -> torch.Tensor:
         name = a.name
         device = env.device
+        from ..sql.udf import aggregate_udf
+        udf = aggregate_udf(name)
+        if udf is not None:
+            vals = as_tensor(eval_expr(a.args[0], env), env) if a.args \
+                else torch.ones(env.n_rows, device=device)
+            return udf(vals, gid, g)
         if env.n_rows == 0:
             if name == "count":
                 return torch.zeros(g, dtype=torch.int64, device=device)

@@ -89,9 +89,16 @@ def expr_name(e) -> str:
     return "expr"
 
 
+def _is_agg_name(name: str) -> bool:
+    if name in AGGREGATE_FUNCS:
+        return True
+    from .udf import aggregate_udf
+    return aggregate_udf(name) is not None
+
+
 def contains_aggregate(e) -> bool:
     if isinstance(e, FuncCall):
-        if e.name in AGGREGATE_FUNCS:
+        if _is_agg_name(e.name):
             return True
         return any(contains_aggregate(a) for a in e.args)
     if isinstance(e, BinaryOp):
@@ -116,7 +123,7 @@ def contains_aggregate(e) -> bool:
 def collect_aggregates(e, out: list) -> None:
     """Find aggregate FuncCall nodes (outermost) in an expression tree."""
     if isinstance(e, FuncCall):
-        if e.name in AGGREGATE_FUNCS:
+        if _is_agg_name(e.name):
             out.append(e)
             return
         for a in e.args:
@@ -205,7 +212,7 @@ def eval_expr(e, env: Env) -> Value:
     if isinstance(e, Star):
         raise SqlError("* only allowed in COUNT(*) or projection list")
     if isinstance(e, FuncCall):
-        if e.name in AGGREGATE_FUNCS:
+        if _is_agg_name(e.name):
             key = expr_name(e)
             if key in env.agg_results:
                 return env.agg_results[key]
@@ -453,4 +460,11 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
         a = as_tensor(eval_expr(e.args[0], env), env).double()
         b = as_tensor(eval_expr(e.args[1], env), env).double()
         return torch.where(a == b, torch.full_like(a, float("nan")), a)
+    from .udf import scalar_udf
+    udf = scalar_udf(name)
+    if udf is not None:
+        args = [eval_expr(a, env) for a in e.args]
+        args = [as_tensor(a, env) if not isinstance(a, Column) else a
+                for a in args]
+        return udf(*args)
     raise SqlError(f"unknown function {name}()")

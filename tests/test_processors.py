@@ -297,3 +297,16 @@ def test_sql_temporary_join(run):
     out = run(p.process(b))[0]
     assert out.column("uid").to_pylist() == [1, 2, 2]
     assert out.column("name").to_strlist() == ["ann", "bob", "bob"]
+
+
+def test_schema_registry_codec():
+    from arkflow_amd.codecs.schema_registry import SchemaRegistryCodec
+    proto = "message M { double v = 1; string tag = 2; }"
+    c = SchemaRegistryCodec({"schemas": {"7": proto}, "default_schema_id": 7},
+                            None)
+    b = MessageBatch.from_dict({"v": [1.5], "tag": ["x"]})
+    enc = c.encode(b)
+    assert enc[0][0] == 0 and enc[0][1:5] == (7).to_bytes(4, "big")
+    dec = c.decode(enc)
+    assert dec.column("v").to_pylist() == [1.5]
+    assert dec.column("tag").to_strlist() == ["x"]

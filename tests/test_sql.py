@@ -178,3 +178,20 @@ def test_sql_processor_component(run):
     assert out[0].column("id").to_pylist() == [2, 3]
     # empty in → ProcessResult::None
     assert run(p.process(MessageBatch.from_dict({"id": []}))) == []
+
+
+def test_scalar_and_aggregate_udf(flow):
+    import torch
+    from arkflow_amd.sql.udf import register_aggregate_udf, register_scalar_udf
+    register_scalar_udf("clip10", lambda t: torch.clamp(
+        t if isinstance(t, torch.Tensor) else t.data, max=10.0))
+    register_aggregate_udf(
+        "sumsq", lambda vals, gid, g: torch.zeros(
+            g, dtype=torch.float64).scatter_add_(
+                0, gid.long(), vals.double() ** 2))
+    r = q("SELECT id, clip10(value) AS c FROM flow ORDER BY id LIMIT 2",
+          flow=flow)
+    assert r.column("c").to_pylist() == [10.0, 10.0]
+    r = q("SELECT grp, sumsq(value) AS s FROM flow GROUP BY grp ORDER BY grp",
+          flow=flow)
+    assert r.column("s").to_pylist()[0] == 10.0 ** 2 + 20.0 ** 2
