@@ -225,3 +225,37 @@ def test_bert_forward_gpu(dev):
     logits_cpu = enc_cpu.forward(ids)
     assert (logits.cpu() - logits_cpu).abs().max() < 0.35, \
         (logits, logits_cpu)
+
+
+def test_proto_decode_gpu_matches_cpu(nat, dev):
+    """GPU varint decode kernel vs the host wire codec (numeric schema)."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.proto_wire import ProtoSchema, encode_message
+    from arkflow_amd.processors.protobuf_proc import ProtobufToArrowProcessor
+    import asyncio
+    proto = """
+    message T {
+      double a = 1; float b = 2; int64 c = 3; sint32 d = 4;
+      bool e = 5; fixed32 f = 6; sfixed64 g = 7;
+    }
+    """
+    schema = ProtoSchema.parse(proto)
+    torch.manual_seed(20)
+    rows = []
+    payloads = []
+    for i in range(10_000):
+        r = {"a": float(i) * 0.5 - 100, "b": float(i % 7), "c": i * 3 - 5000,
+             "d": (-1) ** i * i, "e": i % 3 == 0, "f": i % 100,
+             "g": -i * 7}
+        rows.append(r)
+        payloads.append(encode_message(r, schema))
+    batch = MessageBatch.from_binary(payloads).to(dev)
+    proc = ProtobufToArrowProcessor({"proto": proto}, None)
+    out = asyncio.new_event_loop().run_until_complete(proc.process(batch))[0]
+    assert out.column("a").data.is_cuda
+    a = out.column("a").to_pylist()
+    assert a[:3] == [rows[0]["a"], rows[1]["a"], rows[2]["a"]]
+    assert out.column("c").to_pylist()[999] == rows[999]["c"]
+    assert out.column("d").to_pylist()[7] == rows[7]["d"]
+    assert out.column("e").to_pylist()[9] == rows[9]["e"]
+    assert out.column("g").to_pylist()[123] == rows[123]["g"]
