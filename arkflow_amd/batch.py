@@ -349,6 +349,26 @@ class MessageBatch:
         )
 
     def take(self, indices: torch.Tensor) -> "MessageBatch":
+        # GPU fast path: all plain numeric columns gather in ONE kernel
+        # launch (csrc gather_multi) instead of one launch per column.
+        if indices.is_cuda:
+            simple = [
+                (k, c) for k, c in self.columns.items()
+                if c.kind == "numeric" and c.validity is None
+            ]
+            if len(simple) > 1:
+                from .ops import native_available, require_native
+                if native_available():
+                    nat = require_native()
+                    outs = nat.gather_columns([c.data for _, c in simple],
+                                              indices)
+                    cols = {k: Column("numeric", t)
+                            for (k, _), t in zip(simple, outs)}
+                    for k, c in self.columns.items():
+                        if k not in cols:
+                            cols[k] = c.take(indices)
+                    return MessageBatch(
+                        {k: cols[k] for k in self.columns}, self.input_name)
         return MessageBatch(
             {k: c.take(indices) for k, c in self.columns.items()},
             self.input_name,

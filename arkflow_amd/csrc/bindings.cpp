@@ -7,6 +7,7 @@
 
 #include <cstdint>
 #include <tuple>
+#include <vector>
 
 // ---- extern kernel launchers (defined in the .hip TUs) ----------------------
 extern "C" {
@@ -52,6 +53,8 @@ int launch_attention_bf16(const void*, const void*, const void*, void*, int,
 void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
                          const int*, const int*, const int*, const int*,
                          int64_t*, double*, int32_t*, hipStream_t);
+void launch_gather_multi(int, const void**, void**, const int*,
+                         const int32_t*, int64_t, hipStream_t);
 }
 
 namespace {
@@ -335,6 +338,35 @@ torch::Tensor attention_bf16(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return out;
 }
 
+std::vector<torch::Tensor> gather_columns(std::vector<torch::Tensor> cols,
+                                          torch::Tensor idx) {
+  check_cuda(idx, "idx");
+  auto idx32 = idx.scalar_type() == torch::kInt32 ? idx : idx.to(torch::kInt32);
+  int64_t m = idx32.numel();
+  std::vector<torch::Tensor> outs;
+  outs.reserve(cols.size());
+  std::vector<const void*> src;
+  std::vector<void*> dst;
+  std::vector<int> esz;
+  auto flush = [&]() {
+    if (!src.empty() && m > 0)
+      launch_gather_multi((int)src.size(), src.data(), dst.data(), esz.data(),
+                          idx32.data_ptr<int32_t>(), m, cur_stream());
+    src.clear(); dst.clear(); esz.clear();
+  };
+  for (auto& c : cols) {
+    check_cuda(c, "col");
+    auto out = torch::empty({m}, c.options());
+    outs.push_back(out);
+    src.push_back(c.data_ptr());
+    dst.push_back(out.data_ptr());
+    esz.push_back((int)c.element_size());
+    if ((int)src.size() == 32) flush();
+  }
+  flush();
+  return outs;
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> proto_decode(
     torch::Tensor data, torch::Tensor offsets,
     std::vector<int64_t> fno, std::vector<int64_t> kind,
@@ -385,4 +417,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_bf16", &attention_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("scale"));
   m.def("proto_decode", &proto_decode);
+  m.def("gather_columns", &gather_columns);
 }

@@ -203,3 +203,52 @@ void launch_gather(const void* src, const int32_t* idx, int64_t m,
 }
 
 }  // extern "C"
+
+// ---- multi-column gather: one launch for a whole batch ----------------------
+#define GATHER_MAX_COLS 32
+struct GatherSpec {
+  int ncols;
+  const void* src[GATHER_MAX_COLS];
+  void* dst[GATHER_MAX_COLS];
+  int esz[GATHER_MAX_COLS];
+};
+
+__global__ void gather_multi_kernel(GatherSpec spec,
+                                    const int32_t* __restrict__ idx,
+                                    int64_t m) {
+  int64_t total = (int64_t)spec.ncols * m;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int c = (int)(i / m);
+    int64_t j = i - (int64_t)c * m;
+    int64_t s = idx[j];
+    switch (spec.esz[c]) {
+      case 1: ((uint8_t*)spec.dst[c])[j] = ((const uint8_t*)spec.src[c])[s];
+              break;
+      case 2: ((uint16_t*)spec.dst[c])[j] = ((const uint16_t*)spec.src[c])[s];
+              break;
+      case 4: ((uint32_t*)spec.dst[c])[j] = ((const uint32_t*)spec.src[c])[s];
+              break;
+      default: ((uint64_t*)spec.dst[c])[j] = ((const uint64_t*)spec.src[c])[s];
+               break;
+    }
+  }
+}
+
+extern "C" void launch_gather_multi(int ncols, const void** src, void** dst,
+                                    const int* esz, const int32_t* idx,
+                                    int64_t m, hipStream_t st) {
+  GatherSpec spec{};
+  spec.ncols = ncols;
+  for (int c = 0; c < ncols; ++c) {
+    spec.src[c] = src[c];
+    spec.dst[c] = dst[c];
+    spec.esz[c] = esz[c];
+  }
+  int64_t total = (int64_t)ncols * m;
+  int grid = (int)((total + 255) / 256);
+  if (grid > 4096) grid = 4096;
+  if (grid < 1) return;
+  gather_multi_kernel<<<grid, 256, 0, st>>>(spec, idx, m);
+}

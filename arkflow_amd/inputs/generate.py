@@ -66,25 +66,46 @@ class GenerateInput(Input):
             if self._gen is None:
                 self._gen = torch.Generator(device=self.device)
                 self._gen.manual_seed(self.seed)
+                self._float_fields = [
+                    (name, spec) for name, spec in self.fields.items()
+                    if str(spec.get("dtype", "float32")) not in
+                    ("int32", "int64")
+                ]
+                # one fused rand*scale+offset launch for ALL float fields:
+                # rows of a [nf, n] tensor are contiguous zero-copy columns
+                self._scale = torch.tensor(
+                    [[float(s.get("high", 100.0)) - float(s.get("low", 0.0))]
+                     for _, s in self._float_fields],
+                    device=self.device, dtype=torch.float32)
+                self._offset = torch.tensor(
+                    [[float(s.get("low", 0.0))] for _, s in
+                     self._float_fields],
+                    device=self.device, dtype=torch.float32)
             cols = {}
-            for name, spec in self.fields.items():
-                dtype = str(spec.get("dtype", "float32"))
-                low = float(spec.get("low", 0.0))
-                high = float(spec.get("high", 100.0))
-                if dtype in ("int32", "int64"):
-                    t = torch.randint(
-                        int(low), max(int(high), int(low) + 1), (n,),
-                        generator=self._gen, device=self.device,
-                        dtype=getattr(torch, dtype),
-                    )
-                else:
-                    t = torch.rand(
-                        (n,), generator=self._gen, device=self.device,
-                        dtype=torch.float32,
-                    ) * (high - low) + low
+            nf = len(self._float_fields)
+            if nf:
+                block = torch.rand((nf, n), generator=self._gen,
+                                   device=self.device, dtype=torch.float32)
+                block = torch.addcmul(self._offset, block, self._scale)
+                for i, (name, spec) in enumerate(self._float_fields):
+                    t = block[i]
+                    dtype = str(spec.get("dtype", "float32"))
                     if dtype != "float32":
                         t = t.to(getattr(torch, dtype))
+                    cols[name] = Column("numeric", t)
+            for name, spec in self.fields.items():
+                dtype = str(spec.get("dtype", "float32"))
+                if dtype not in ("int32", "int64"):
+                    continue
+                low = float(spec.get("low", 0.0))
+                high = float(spec.get("high", 100.0))
+                t = torch.randint(
+                    int(low), max(int(high), int(low) + 1), (n,),
+                    generator=self._gen, device=self.device,
+                    dtype=getattr(torch, dtype),
+                )
                 cols[name] = Column("numeric", t)
+            cols = {name: cols[name] for name in self.fields}  # declared order
             return MessageBatch(cols, input_name="generate")
         payload = (self.context or '{"timestamp": 0, "value": 1}').encode()
         if self._payload_batch is None or self._payload_batch.num_rows != n:
