@@ -42,6 +42,8 @@ void launch_join_probe_emit(const int64_t*, int64_t, const int64_t*,
                             hipStream_t);
 void launch_gemm_bf16(const void*, const void*, const float*, void*, int, int,
                       int, int, hipStream_t);
+int launch_gemm_bf16_8p(const void*, const void*, const float*, void*, int,
+                        int, int, int, int, hipStream_t);
 void launch_layernorm_bf16(const void*, const void*, const float*,
                            const float*, void*, void*, int64_t, int, float,
                            hipStream_t);
@@ -264,8 +266,43 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
     TORCH_CHECK(bias_f.numel() == N, "bias must be [N]");
     bias_ptr = bias_f.data_ptr<float>();
   }
-  launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
-                   (int)M, (int)N, (int)K, (int)act, cur_stream());
+  // prefer the 8-phase 256² pipelined kernel where the shape fills the chip
+  int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
+  int rc = tiles >= 128
+               ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                     C.data_ptr(), (int)M, (int)N, (int)K,
+                                     (int)act, /*swz=*/0, cur_stream())
+               : -1;
+  if (rc != 0)
+    launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
+                     (int)M, (int)N, (int)K, (int)act, cur_stream());
+  return C;
+}
+
+torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
+                                c10::optional<torch::Tensor> bias,
+                                int64_t act, int64_t variant) {
+  // variant: 0 = 128² tile, 1 = 8-phase linear LDS, 2 = 8-phase + swizzle
+  check_cuda(A, "A");
+  check_cuda(Bt, "Bt");
+  int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  auto C = torch::empty({M, N}, A.options());
+  const float* bias_ptr = nullptr;
+  torch::Tensor bias_f;
+  if (bias.has_value() && bias->defined()) {
+    bias_f = bias->to(torch::kFloat32).contiguous();
+    bias_ptr = bias_f.data_ptr<float>();
+  }
+  if (variant == 0) {
+    launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
+                     (int)M, (int)N, (int)K, (int)act, cur_stream());
+  } else {
+    int rc = launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                 C.data_ptr(), (int)M, (int)N, (int)K,
+                                 (int)act, variant == 2 ? 1 : 0,
+                                 cur_stream());
+    TORCH_CHECK(rc == 0, "shape not supported by 8-phase kernel");
+  }
   return C;
 }
 
@@ -418,4 +455,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("v"), py::arg("scale"));
   m.def("proto_decode", &proto_decode);
   m.def("gather_columns", &gather_columns);
+  m.def("gemm_bf16_variant", &gemm_bf16_variant);
 }

@@ -168,6 +168,8 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
     async def metrics():
         return render_prometheus(engine)
 
+    from .console import mount_console
+    mount_console(app)
     return app
 
 

@@ -270,6 +270,8 @@ def create_hub_app(hub: Hub) -> FastAPI:
             raise HTTPException(404, "unknown rollout")
         return r
 
+    from .console import mount_console
+    mount_console(app)
     return app
 
 

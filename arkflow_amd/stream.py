@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import time
 from typing import List, Optional, Tuple
 
 from .batch import MessageBatch
@@ -172,6 +173,7 @@ class Stream:
                         ) -> None:
         while not cancel.is_set():
             try:
+                self._t_read0 = time.perf_counter_ns()
                 item = await _race(self.input.read(), cancel)
                 if item is _SENTINEL:
                     break
@@ -194,6 +196,8 @@ class Stream:
                 continue
             self.metrics.input_batches += 1
             self.metrics.input_messages += batch.num_rows
+            self.metrics.stage_ns["input"] += time.perf_counter_ns() - (
+                getattr(self, "_t_read0", time.perf_counter_ns()))
             if self.wal is not None:
                 from .wal.wal import WalAck
                 seq = await self.wal.append(batch)
@@ -226,7 +230,9 @@ class Stream:
             seq = self._seq
             self._seq += 1
             try:
+                t0 = time.perf_counter_ns()
                 results = await self.pipeline.process(batch)
+                self.metrics.stage_ns["process"] += time.perf_counter_ns() - t0
                 await output_q.put((seq, results, None, ack))
             except Exception as e:  # noqa: BLE001
                 self.metrics.processing_errors += 1
@@ -268,7 +274,9 @@ class Stream:
             await ack.ack()
             return
         try:
+            t0 = time.perf_counter_ns()
             await self.output.write_batch(results)
+            self.metrics.stage_ns["output"] += time.perf_counter_ns() - t0
             self.metrics.output_batches += len(results)
             self.metrics.output_messages += sum(b.num_rows for b in results)
             await ack.ack()

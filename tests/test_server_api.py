@@ -154,3 +154,17 @@ streams:
     output: {type: drop}
 """)
     assert main(["--config", str(bad), "--validate"]) == 1
+
+
+def test_console_served(run):
+    async def main():
+        eng = _engine()
+        app = create_app(eng)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://node") as c:
+            r = await c.get("/")
+            assert r.status_code == 200
+            assert "arkflow_amd console" in r.text
+
+    run(main())
