@@ -266,13 +266,15 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
     TORCH_CHECK(bias_f.numel() == N, "bias must be [N]");
     bias_ptr = bias_f.data_ptr<float>();
   }
-  // prefer the 8-phase 256² pipelined kernel where the shape fills the chip
+  // prefer the 8-phase 256² pipelined kernel (with LDS swizzle) where it
+  // measured faster than the 128² tile: chip-filling grids with enough K
+  // to amortize the pipeline (see profiles/ r02 GEMM table)
   int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
-  int rc = tiles >= 128
-               ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
-                                     C.data_ptr(), (int)M, (int)N, (int)K,
-                                     (int)act, /*swz=*/0, cur_stream())
-               : -1;
+  bool use8p = tiles >= 128 && (K >= 1536 || N >= 3072);
+  int rc = use8p ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                       C.data_ptr(), (int)M, (int)N, (int)K,
+                                       (int)act, /*swz=*/1, cur_stream())
+                 : -1;
   if (rc != 0)
     launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
                      (int)M, (int)N, (int)K, (int)act, cur_stream());
