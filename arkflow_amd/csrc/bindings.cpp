@@ -270,7 +270,7 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
   // measured faster than the 128² tile: chip-filling grids with enough K
   // to amortize the pipeline (see profiles/ r02 GEMM table)
   int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
-  bool use8p = tiles >= 128 && (K >= 1536 || N >= 3072);
+  bool use8p = tiles >= 128 && K >= 1536;
   int rc = use8p ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                        C.data_ptr(), (int)M, (int)N, (int)K,
                                        (int)act, /*swz=*/1, cur_stream())

@@ -22,9 +22,15 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor,
     layout — consumed directly as the GEMM's B^T operand)."""
     a = _ACTS[act]
     if x.is_cuda:
+        import os
         nat = require_native()
         x2 = x.reshape(-1, x.shape[-1]).contiguous()
-        out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
+        force = os.environ.get("ARKFLOW_GEMM_VARIANT")
+        if force is not None:
+            out = nat.gemm_bf16_variant(x2, weight.contiguous(), bias, a,
+                                        int(force))
+        else:
+            out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
         return out.reshape(*x.shape[:-1], weight.shape[0])
     y = torch.nn.functional.linear(x.float(), weight.float(),
                                    bias.float() if bias is not None else None)
