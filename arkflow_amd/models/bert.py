@@ -67,10 +67,41 @@ class BertEncoder:
         self.cls_w = w(c.num_labels, c.hidden)
         self.cls_b = bias(c.num_labels)
         self._scale = 1.0 / math.sqrt(c.head_dim)
+        # hipGraph capture: the encoder is shape-static per (B, S), so the
+        # whole forward (~80 kernel launches) replays as ONE graph launch
+        self.use_graph = True
+        self._graphs = {}  # (B, S) → (graph, static_ids, static_logits)
 
     # -------------------------------------------------------------- forward
     def forward(self, token_ids: torch.Tensor) -> torch.Tensor:
         """token_ids: [B, S] int64 → logits [B, num_labels] float32."""
+        if self.use_graph and self.device.type == "cuda":
+            return self._forward_graphed(token_ids)
+        return self._forward_eager(token_ids)
+
+    def _forward_graphed(self, token_ids: torch.Tensor) -> torch.Tensor:
+        key = tuple(token_ids.shape)
+        entry = self._graphs.get(key)
+        if entry is None:
+            static_ids = token_ids.to(self.device).clone()
+            # warmup on a side stream (allocator settles), then capture
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self._forward_eager(static_ids)
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_logits = self._forward_eager(static_ids)
+            entry = (graph, static_ids, static_logits)
+            self._graphs[key] = entry
+        graph, static_ids, static_logits = entry
+        static_ids.copy_(token_ids.to(self.device))
+        graph.replay()
+        return static_logits
+
+    def _forward_eager(self, token_ids: torch.Tensor) -> torch.Tensor:
         c = self.cfg
         B, S = token_ids.shape
         ids = token_ids.to(self.device)
