@@ -57,6 +57,47 @@ def build_pipeline(args, device):
         })
         # BERT consumes whole batches (no filter — sequences must stay full)
         return gen_tok, Pipeline([infer])
+    if args.model == "sqlagg":
+        # BASELINE config 2: sql filter + hash aggregate, GPU-resident columns
+        agg = SqlProcessor({
+            "query": "SELECT key, count(*) AS c, sum(f0) AS s FROM flow "
+                     "WHERE f0 >= 0.2 GROUP BY key"})
+        return gen, Pipeline([agg])
+    if args.model == "proto_mlp":
+        # BASELINE config 3: kafka-shaped protobuf payloads → GPU varint
+        # decode → MLP anomaly scoring
+        from arkflow_amd.batch import MessageBatch
+        from arkflow_amd.processors.proto_wire import (
+            ProtoSchema, encode_message)
+        from arkflow_amd.processors.protobuf_proc import (
+            ProtobufToArrowProcessor)
+        import random
+        proto = ("message T { double f0 = 1; double f1 = 2; double f2 = 3; "
+                 "double f3 = 4; int64 key = 5; }")
+        schema = ProtoSchema.parse(proto)
+        rng = random.Random(42 + args.rank)
+        payloads = [
+            encode_message({"f0": rng.random(), "f1": rng.random(),
+                            "f2": rng.random(), "f3": rng.random(),
+                            "key": rng.randrange(1024)}, schema)
+            for _ in range(args.batch_size)
+        ]
+        device_batch = MessageBatch.from_binary(
+            payloads, input_name="kafka").to(device)
+
+        class _ProtoGen:
+            """Kafka-shaped source: device-resident wire-format payloads."""
+
+            async def read(self):
+                from arkflow_amd.spi import NoopAck
+                return device_batch, NoopAck()
+
+        decode = ProtobufToArrowProcessor({"proto": proto}, None)
+        infer = InferenceProcessor({
+            "model": "mlp_anomaly", "columns": ["f0", "f1", "f2", "f3"],
+            "hidden": [args.hidden, args.hidden], "device": str(device),
+        })
+        return _ProtoGen(), Pipeline([decode, infer])
     infer = InferenceProcessor({
         "model": "mlp_anomaly",
         "columns": [f"f{i}" for i in range(n_features)],
@@ -88,7 +129,8 @@ def main():
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-size", type=int, default=8192)
-    p.add_argument("--model", choices=["mlp", "bert"], default="mlp")
+    p.add_argument("--model", choices=["mlp", "bert", "sqlagg", "proto_mlp"],
+                   default="mlp")
     p.add_argument("--features", type=int, default=16)
     p.add_argument("--hidden", type=int, default=256)
     args = p.parse_args()
@@ -160,9 +202,12 @@ def main():
             "data": "synthetic",
             "p50_ms": p50_ms,
             "config": {
-                "model": ("generate→sql(filter)→mlp_anomaly[16→256→256→1]"
-                          if args.model == "mlp"
-                          else "generate→bert_base(12L,768H,seq128)"),
+                "model": {
+                    "mlp": "generate→sql(filter)→mlp_anomaly[16→256→256→1]",
+                    "bert": "generate→bert_base(12L,768H,seq128)",
+                    "sqlagg": "generate→sql(filter+hash_agg by 1024 keys)",
+                    "proto_mlp": "kafka-shaped→protobuf_decode→mlp_anomaly",
+                }[args.model],
                 "global_batch": args.batch_size * max(world, args.gpus),
                 "seq_len": 128 if args.model == "bert" else 1,
                 "parallelism": f"dp{world if world > 1 else args.gpus}",
