@@ -19,13 +19,15 @@ static int32_t float_flip_host(float f) {
   return (int32_t)v;  // bit pattern; device compares as unsigned
 }
 
-// ---- group-id assignment ----------------------------------------------------
-__global__ void hash_build_kernel(const int64_t* __restrict__ keys, int64_t n,
-                                  int64_t* __restrict__ table_keys,
-                                  int32_t* __restrict__ table_gids,
-                                  uint32_t table_mask,
-                                  int32_t* __restrict__ counter,
-                                  int32_t* __restrict__ gids_out) {
+// ---- group-id assignment ------------------------------------------------
+// Two kernels: INSERT claims keys and assigns dense gids; LOOKUP probes with
+// plain loads after the kernel boundary (which orders all insert writes) —
+// no per-row acquire/invalidate in the hot path.
+__global__ void hash_insert_kernel(const int64_t* __restrict__ keys, int64_t n,
+                                   int64_t* __restrict__ table_keys,
+                                   int32_t* __restrict__ table_gids,
+                                   uint32_t table_mask,
+                                   int32_t* __restrict__ counter) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -49,10 +51,21 @@ __global__ void hash_build_kernel(const int64_t* __restrict__ keys, int64_t n,
       }
       h = (h + 1) & table_mask;
     }
-    int32_t g;
-    while ((g = __hip_atomic_load(&table_gids[h], __ATOMIC_ACQUIRE,
-                                  __HIP_MEMORY_SCOPE_AGENT)) < 0) {}
-    gids_out[i] = g;
+  }
+}
+
+__global__ void hash_lookup_kernel(const int64_t* __restrict__ keys, int64_t n,
+                                   const int64_t* __restrict__ table_keys,
+                                   const int32_t* __restrict__ table_gids,
+                                   uint32_t table_mask,
+                                   int32_t* __restrict__ gids_out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    uint32_t h = (uint32_t)(mix64((uint64_t)k) & table_mask);
+    while (table_keys[h] != k) h = (h + 1) & table_mask;
+    gids_out[i] = table_gids[h];
   }
 }
 
@@ -158,8 +171,10 @@ void launch_hash_build(const int64_t* keys, int64_t n, int64_t* table_keys,
       table_keys, EMPTY_KEY, table_size);
   fill_i32_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
       table_gids, -1, table_size);
-  hash_build_kernel<<<grid_for(n, AGG_BLOCK), AGG_BLOCK, 0, st>>>(
-      keys, n, table_keys, table_gids, table_size - 1, counter, gids_out);
+  hash_insert_kernel<<<grid_for(n, AGG_BLOCK), AGG_BLOCK, 0, st>>>(
+      keys, n, table_keys, table_gids, table_size - 1, counter);
+  hash_lookup_kernel<<<grid_for(n, AGG_BLOCK), AGG_BLOCK, 0, st>>>(
+      keys, n, table_keys, table_gids, table_size - 1, gids_out);
 }
 
 void launch_hash_export(const int64_t* table_keys, const int32_t* table_gids,

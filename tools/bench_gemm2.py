@@ -4,7 +4,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.join(__import__("os").path.dirname(__file__), ".."))
 from arkflow_amd import ops
 
 nat = ops.require_native()
