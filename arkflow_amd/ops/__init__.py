@@ -105,6 +105,14 @@ def segment_reduce(values: torch.Tensor, group_ids: torch.Tensor,
     """Per-group reduction: op in sum|min|max|count|mean."""
     gid = group_ids.long()
     if op == "count":
+        if values.is_cuda:
+            # int64 scatter_add on GPU is a CAS loop — catastrophic under
+            # high key contention; the native f32 LDS-buffered sum is exact
+            # for counts < 2^24 per group and orders of magnitude faster.
+            ones = torch.ones(values.shape[0], dtype=torch.float32,
+                              device=values.device)
+            return segment_reduce(ones, group_ids, num_groups,
+                                  "sum").to(torch.int64)
         out = torch.zeros(num_groups, dtype=torch.int64, device=values.device)
         out.scatter_add_(0, gid, torch.ones_like(gid))
         return out

@@ -385,6 +385,12 @@ def _co_encode(l_keys, r_keys, device):
 
 def _first_index_per_group(gid: torch.Tensor, g: int) -> torch.Tensor:
     n = gid.shape[0]
+    if gid.is_cuda and n < (1 << 24):
+        # int64 scatter_reduce on GPU is CAS-loop atomics — use the native
+        # flipped-f32 segment min (exact for indices < 2^24)
+        idx_f = torch.arange(n, dtype=torch.float32, device=gid.device)
+        return ops.segment_reduce(idx_f, gid.to(torch.int32), g,
+                                  "min").to(torch.int64)
     idx = torch.arange(n, dtype=torch.int64, device=gid.device)
     first = torch.full((g,), n, dtype=torch.int64, device=gid.device)
     first.scatter_reduce_(0, gid.long(), idx, reduce="amin", include_self=True)
