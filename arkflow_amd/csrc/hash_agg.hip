@@ -75,9 +75,11 @@ __global__ void hash_export_kernel(const int64_t* __restrict__ table_keys,
                                    uint32_t table_size,
                                    int64_t* __restrict__ uniq) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= table_size) return;
-  int64_t k = table_keys[i];
-  if (k != EMPTY_KEY) uniq[table_gids[i]] = k;
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < table_size; i += stride) {
+    int64_t k = table_keys[i];
+    if (k != EMPTY_KEY) uniq[table_gids[i]] = k;
+  }
 }
 
 // ---- segment reductions ------------------------------------------------------
@@ -149,11 +151,13 @@ __global__ void unflip_kernel(const int32_t* __restrict__ in,
 
 __global__ void fill_i64_kernel(int64_t* p, int64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = v;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;  // grid-stride: grid is capped at 2048
 }
 __global__ void fill_i32_kernel(int32_t* p, int32_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = v;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;
 }
 
 // ---- host launchers ----------------------------------------------------------

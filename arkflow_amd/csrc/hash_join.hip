@@ -101,11 +101,13 @@ __global__ void join_probe_emit_kernel(const int64_t* __restrict__ lkeys,
 
 __global__ void join_fill_kernel(int64_t* p, int64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = v;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;  // grid-stride: grid capped at 2048
 }
 __global__ void join_fill32_kernel(int32_t* p, int32_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = v;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;
 }
 
 extern "C" {

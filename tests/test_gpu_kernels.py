@@ -259,3 +259,27 @@ def test_proto_decode_gpu_matches_cpu(nat, dev):
     assert out.column("d").to_pylist()[7] == rows[7]["d"]
     assert out.column("e").to_pylist()[9] == rows[9]["e"]
     assert out.column("g").to_pylist()[123] == rows[123]["g"]
+
+
+def test_hash_group_large(nat, dev):
+    """Regression: tables > 512K entries must be fully initialized
+    (fill kernels are grid-stride; grid is capped at 2048 blocks)."""
+    torch.manual_seed(30)
+    keys = torch.randint(0, 1024, (1_000_000,), device=dev,
+                         dtype=torch.int64)
+    import time
+    t0 = time.time()
+    gid, uniq = nat.hash_group_i64(keys)
+    torch.cuda.synchronize()
+    took = time.time() - t0
+    assert torch.equal(uniq[gid.long()], keys)
+    assert uniq.shape[0] == torch.unique(keys).shape[0]
+    assert took < 1.0, f"hash_group at 1M rows took {took:.2f}s"
+
+
+def test_join_large(nat, dev):
+    torch.manual_seed(31)
+    lk = torch.randint(0, 1 << 20, (400_000,), device=dev, dtype=torch.int64)
+    rk = torch.randint(0, 1 << 20, (600_000,), device=dev, dtype=torch.int64)
+    l_idx, r_idx = nat.join_inner_i64(lk, rk)
+    assert torch.equal(lk[l_idx], rk[r_idx])
