@@ -114,6 +114,8 @@ class KafkaInput(Input):
         self.driver = config.get("driver") or (
             "memory" if str(self.brokers[0]).startswith("memory://")
             else "real")
+        from ..codecs.helper import build_codec
+        self.codec = build_codec(config, resource)
         self.bus: Optional[FakeBus] = None
         self._closed = False
         # in-memory read positions (per consumer instance): a read advances
@@ -167,6 +169,9 @@ class KafkaInput(Input):
                         },
                         input_name=topic,
                     )
+                    if self.codec is not None:
+                        from ..codecs.helper import apply_codec
+                        batch = apply_codec(batch, self.codec)
                     return batch, KafkaAck(self.bus, self.group, topic, p, off)
             self.bus.notify.clear()
             try:
@@ -257,6 +262,8 @@ class _PubSubInput(Input):
         self.url = str(config.get("url", "memory://default"))
         self.driver = config.get("driver") or (
             "memory" if self.url.startswith("memory://") else "real")
+        from ..codecs.helper import build_codec
+        self.codec = build_codec(config, resource)
         self.bus: Optional[FakeBus] = None
         self._q: Optional[asyncio.Queue] = None
         self._closed = False
@@ -285,6 +292,9 @@ class _PubSubInput(Input):
             },
             input_name=self.topic,
         )
+        if self.codec is not None:
+            from ..codecs.helper import apply_codec
+            batch = apply_codec(batch, self.codec)
         return batch, NoopAck()
 
     async def close(self) -> None:

@@ -44,6 +44,8 @@ class HttpInput(Input):
         rate = config.get("rate_limit")
         self.bucket = TokenBucket(float(rate), int(config.get("burst", rate)))\
             if rate else None
+        from ..codecs.helper import build_codec
+        self.codec = build_codec(config, resource)
         self._q: asyncio.Queue = asyncio.Queue(maxsize=self.queue_size)
         self._runner = None
         self._site = None
@@ -86,6 +88,9 @@ class HttpInput(Input):
         batch = await self._q.get()
         if batch is None:
             raise EOFError_("http input closed")
+        if self.codec is not None:
+            from ..codecs.helper import apply_codec
+            batch = apply_codec(batch, self.codec)
         return batch, NoopAck()
 
     async def close(self) -> None:

@@ -29,6 +29,10 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
     cp = engine.control_plane
     prefix = cfg.api_prefix.rstrip("/")
     app = FastAPI(title="arkflow_amd node API", version="1")
+    if cfg.cors:
+        from fastapi.middleware.cors import CORSMiddleware
+        app.add_middleware(CORSMiddleware, allow_origins=["*"],
+                           allow_methods=["*"], allow_headers=["*"])
 
     def auth(request: Request):
         if cfg.token:

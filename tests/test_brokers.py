@@ -245,3 +245,21 @@ def test_debezium_codec():
     back = c.encode(b)
     import json
     assert json.loads(back[0])["payload"]["after"]["id"] == 1
+
+
+def test_kafka_input_with_codec(run):
+    """codec-on-input (reference codec_helper): payloads decode as they
+    enter the stream, __meta_* preserved."""
+    async def main():
+        bus = FakeBus.get("t1")
+        bus.produce("enc", None, b'{"v": 7, "s": "x"}')
+        inp = KafkaInput({"brokers": "memory://t1", "topic": "enc",
+                          "consumer_group": "g",
+                          "codec": {"type": "json"}})
+        await inp.connect()
+        b, _ = await asyncio.wait_for(inp.read(), 2)
+        assert b.column("v").to_pylist() == [7]
+        assert b.column("s").to_strlist() == ["x"]
+        assert b.column("__meta_source").to_strlist() == ["enc"]
+
+    run(main())
