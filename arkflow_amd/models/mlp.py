@@ -30,17 +30,62 @@ class MlpAnomalyDetector:
             self.weights.append(w.to(self.device, torch.bfloat16))
             self.biases.append(b.to(self.device, torch.float32))
         self.dims = dims
+        self.use_graph = True
+        self._graphs = {}  # padded-cap → (graph, static_in, static_out)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        """x: [n, in_features] float → anomaly score [n] float32."""
+        """x: [n, in_features] float → anomaly score [n] float32.
+
+        On GPU the whole forward is hipGraph-captured over a padded static
+        batch (row count varies after upstream filters; scores are per-row so
+        stale pad rows are harmless and sliced off)."""
+        if self.use_graph and self.device.type == "cuda":
+            return self._forward_graphed(x)
+        return self._forward_eager(x)
+
+    def _forward_eager(self, x: torch.Tensor) -> torch.Tensor:
         h = x.to(self.device, torch.bfloat16)
         pad = _pad32(self.in_features) - self.in_features
         if pad:
             h = torch.nn.functional.pad(h, (0, pad))
+        return self._net(h)
+
+    def _net(self, h: torch.Tensor) -> torch.Tensor:
         for i, (w, b) in enumerate(zip(self.weights, self.biases)):
             act = "relu" if i < len(self.weights) - 1 else "none"
             h = opsnn.linear_bf16(h, w, b, act=act)
         return h.reshape(-1).to(torch.float32)
+
+    def _forward_graphed(self, x: torch.Tensor) -> torch.Tensor:
+        n = x.shape[0]
+        cap = self._cap_for(n)
+        entry = self._graphs.get(cap)
+        if entry is None:
+            static_in = torch.zeros(cap, _pad32(self.in_features),
+                                    device=self.device, dtype=torch.bfloat16)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self._net(static_in)
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_out = self._net(static_in)
+            entry = (graph, static_in, static_out)
+            self._graphs[cap] = entry
+        graph, static_in, static_out = entry
+        static_in[:n, : self.in_features].copy_(
+            x.to(self.device, torch.bfloat16))
+        graph.replay()
+        return static_out[:n].clone()
+
+    @staticmethod
+    def _cap_for(n: int) -> int:
+        cap = 4096
+        while cap < n:
+            cap <<= 1
+        return cap
 
 
 def _pad32(k: int) -> int:
