@@ -75,6 +75,7 @@ def init() -> None:
         "arkflow_amd.codecs.debezium",
         "arkflow_amd.outputs.file",
         "arkflow_amd.outputs.http",
+        "arkflow_amd.outputs.influxdb",
         "arkflow_amd.temporary.memory_table",
         "arkflow_amd.wal.store",
         "arkflow_amd.wal.segment_store",

@@ -352,7 +352,48 @@ def _mk_pubsub(kind: str):
 MqttInput, MqttOutput = _mk_pubsub("mqtt")
 NatsInput, NatsOutput = _mk_pubsub("nats")
 PulsarInput, PulsarOutput = _mk_pubsub("pulsar")
-RedisInput, RedisOutput = _mk_pubsub("redis")
+_RedisPubSubIn, RedisOutput = _mk_pubsub("redis")
+
+
+class RedisInput(_RedisPubSubIn):
+    """redis input modes (reference input/redis.rs): pubsub (default),
+    list (BRPOP-style: consume from the log, at-least-once like kafka) and
+    stream (offset-tracked) — over the fake bus, list/stream use the
+    append-only topic log with a consumer position."""
+
+    def __init__(self, config: dict, resource=None):
+        super().__init__(config, resource)
+        self.mode = config.get("mode", "pubsub")
+        self._pos = 0
+
+    async def read(self):
+        if self.mode == "pubsub":
+            return await super().read()
+        # list/stream: consume the topic log in order
+        import time as _t
+        while True:
+            if self._closed:
+                raise EOFError_("redis input closed")
+            self.bus.ensure_topic(self.topic)
+            log = self.bus.topics[self.topic][0]
+            if self._pos < len(log):
+                key, value, _ts = log[self._pos]
+                self._pos += 1
+                batch = MessageBatch(
+                    {"__value__": Column.from_bytes([value]),
+                     "__meta_source": Column.from_strings([self.topic]),
+                     "__meta_offset": Column.from_numeric([self._pos - 1]),
+                     "__meta_timestamp": Column.from_numeric([_t.time()])},
+                    input_name=self.topic)
+                if self.codec is not None:
+                    from ..codecs.helper import apply_codec
+                    batch = apply_codec(batch, self.codec)
+                return batch, NoopAck()
+            self.bus.notify.clear()
+            try:
+                await asyncio.wait_for(self.bus.notify.wait(), 0.5)
+            except asyncio.TimeoutError:
+                pass
 
 
 # ---- registrations ------------------------------------------------------------
@@ -376,8 +417,7 @@ def _build_kafka_out(config, resource=None):
 
 for _name, _in, _out in (("mqtt", MqttInput, MqttOutput),
                          ("nats", NatsInput, NatsOutput),
-                         ("pulsar", PulsarInput, PulsarOutput),
-                         ("redis", RedisInput, RedisOutput)):
+                         ("pulsar", PulsarInput, PulsarOutput)):
     register("input", _name,
              description=f"{_name} subscriber (driver: memory offline)",
              example={"type": _name, "url": "memory://default",
@@ -388,3 +428,19 @@ for _name, _in, _out in (("mqtt", MqttInput, MqttOutput),
              example={"type": _name, "url": "memory://default",
                       "topic": "t"})(
         (lambda cls: lambda config, resource=None: cls(config, resource))(_out))
+
+
+@register("input", "redis",
+          description="redis subscriber: pubsub/list/stream modes "
+                      "(driver: memory offline)",
+          example={"type": "redis", "url": "memory://default", "topic": "t",
+                   "mode": "list"})
+def _build_redis_in(config, resource=None):
+    return RedisInput(config, resource)
+
+
+@register("output", "redis",
+          description="redis publisher (driver: memory offline)",
+          example={"type": "redis", "url": "memory://default", "topic": "t"})
+def _build_redis_out(config, resource=None):
+    return RedisOutput(config, resource)

@@ -22,6 +22,19 @@ from ..registry import register
 from .store import decode_frames, encode_frame
 
 
+# segment sealing presets (reference wal/config.rs:31 SegmentStrategy)
+SEGMENT_STRATEGIES = {
+    # seal often → small crash window, more PUTs
+    "low_latency": {"max_entries": 128, "max_bytes": 1 << 20,
+                    "flush_interval_secs": 0.1},
+    "balanced": {"max_entries": 1024, "max_bytes": 4 << 20,
+                 "flush_interval_secs": 0.5},
+    # big segments → max PUT throughput, larger crash window
+    "aggressive": {"max_entries": 8192, "max_bytes": 32 << 20,
+                   "flush_interval_secs": 2.0},
+}
+
+
 class SegmentWalStore:
     def __init__(self, path: str, stream_id: str = "stream",
                  max_entries: int = 1024, max_bytes: int = 4 << 20,
@@ -171,6 +184,13 @@ class SegmentWalStore:
           description="Segmented WAL store (sealed segments + manifest + "
                       "parallel PUT workers; the S3-backend design)")
 def _build_segment_store(config: dict, resource=None) -> SegmentWalStore:
+    preset = dict(SEGMENT_STRATEGIES.get(
+        config.get("segment_strategy", "balanced"),
+        SEGMENT_STRATEGIES["balanced"]))
+    preset.update({k: v for k, v in config.items()
+                   if k in ("max_entries", "max_bytes",
+                            "flush_interval_secs")})
+    config = {**config, **preset}
     return SegmentWalStore(
         config.get("path", "./wal"),
         stream_id=config.get("stream_id", "stream"),

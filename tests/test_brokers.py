@@ -263,3 +263,63 @@ def test_kafka_input_with_codec(run):
         assert b.column("__meta_source").to_strlist() == ["enc"]
 
     run(main())
+
+
+def test_influx_line_protocol():
+    from arkflow_amd.outputs.influxdb import format_line_protocol
+    b = MessageBatch.from_dict({
+        "sensor": ["a b", "c,d"], "temp": [21.5, 22.0], "n": [3, 4],
+        "ok": [True, False], "ts": [1.0, 2.0],
+    })
+    lines = format_line_protocol(b, "m x", ["sensor"], None, "ts")
+    assert lines[0] == b"m\\ x,sensor=a\\ b temp=21.5,n=3i,ok=t 1000000000"
+    assert lines[1].startswith(b"m\\ x,sensor=c\\,d temp=22.0,n=4i,ok=f")
+
+
+def test_influx_output_http_loopback(run):
+    from aiohttp import web
+    from arkflow_amd.outputs.influxdb import InfluxDbOutput
+
+    async def main():
+        got = []
+
+        async def handler(request):
+            got.append(await request.read())
+            return web.Response(status=204)
+
+        app = web.Application()
+        app.router.add_post("/api/v2/write", handler)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        port = runner.addresses[0][1]
+        out = InfluxDbOutput({"url": f"http://127.0.0.1:{port}",
+                              "org": "o", "bucket": "b",
+                              "measurement": "t", "tags": ["s"],
+                              "token": "tok"})
+        await out.connect()
+        await out.write(MessageBatch.from_dict({"s": ["x"], "v": [1.5]}))
+        await out.close()
+        await runner.cleanup()
+        assert got and b"t,s=x v=1.5" in got[0]
+
+    run(main(), timeout=30)
+
+
+def test_redis_list_mode(run):
+    async def main():
+        from arkflow_amd.inputs.brokers import RedisInput, RedisOutput
+        out = RedisOutput({"url": "memory://t1", "topic": "q"})
+        await out.connect()
+        await out.write(MessageBatch.from_binary([b"a"]))
+        await out.write(MessageBatch.from_binary([b"b"]))
+        inp = RedisInput({"url": "memory://t1", "topic": "q", "mode": "list"})
+        await inp.connect()
+        b1, _ = await asyncio.wait_for(inp.read(), 2)
+        b2, _ = await asyncio.wait_for(inp.read(), 2)
+        assert b1.binary_values() == [b"a"]
+        assert b2.binary_values() == [b"b"]
+        assert b2.column("__meta_offset").to_pylist() == [1]
+
+    run(main())
