@@ -18,6 +18,7 @@ from .eval import (
     Env,
     as_tensor,
     collect_aggregates,
+    collect_window_calls,
     contains_aggregate,
     eval_expr,
     expr_name,
@@ -53,6 +54,37 @@ class SqlExecutor:
             seen.setdefault(expr_name(a), a)
         self._aggs = list(seen.values())
         self.is_aggregate = bool(self._aggs or self.select.group_by)
+        self._windows: List[FuncCall] = []
+        for e, _ in self.select.projections:
+            collect_window_calls(e, self._windows)
+        wseen = {}
+        for w_ in self._windows:
+            wseen.setdefault(expr_name(w_), w_)
+        self._windows = list(wseen.values())
+        self._union = SqlExecutor._from_select(self.select.union_all) \
+            if self.select.union_all else None
+
+    @staticmethod
+    def _from_select(select: Select) -> "SqlExecutor":
+        ex = SqlExecutor.__new__(SqlExecutor)
+        ex.sql = ""
+        ex.select = select
+        ex._aggs = []
+        for e, _ in select.projections:
+            collect_aggregates(e, ex._aggs)
+        if select.having is not None:
+            collect_aggregates(select.having, ex._aggs)
+        seen = {}
+        for a in ex._aggs:
+            seen.setdefault(expr_name(a), a)
+        ex._aggs = list(seen.values())
+        ex.is_aggregate = bool(ex._aggs or select.group_by)
+        ex._windows = []
+        for e, _ in select.projections:
+            collect_window_calls(e, ex._windows)
+        ex._union = SqlExecutor._from_select(select.union_all) \
+            if select.union_all else None
+        return ex
 
     # ------------------------------------------------------------------- run
     def execute(self, tables: Dict[str, MessageBatch]) -> MessageBatch:
@@ -86,6 +118,12 @@ class SqlExecutor:
             columns = {k: c.take(idx) for k, c in columns.items()}
             n_rows = int(idx.shape[0])
             env = Env(columns, n_rows, device)
+
+        # ----------------------------------------------------------- windows
+        if self._windows and not self.is_aggregate:
+            wres = {expr_name(w_): self._compute_window(w_, env)
+                    for w_ in self._windows}
+            env = Env(env.columns, env.n_rows, device, wres)
 
         # --------------------------------------------------------- aggregate
         if self.is_aggregate:
@@ -149,10 +187,93 @@ class SqlExecutor:
                      **result.columns},
                     result.num_rows, device, env.agg_results)
 
-        # ------------------------------------------------------------- limit
+        # ---------------------------------------------------- offset / limit
+        if sel.offset:
+            result = result.slice(min(sel.offset, result.num_rows),
+                                  max(result.num_rows - sel.offset, 0))
         if sel.limit is not None and result.num_rows > sel.limit:
             result = result.slice(0, sel.limit)
+
+        # --------------------------------------------------------- union all
+        if self._union is not None:
+            from ..batch import concat_batches
+            other = self._union.execute(tables)
+            # positional union: rename other's columns to ours
+            if other.column_names != result.column_names and \
+                    len(other.column_names) == len(result.column_names):
+                other = MessageBatch(
+                    dict(zip(result.column_names, other.columns.values())),
+                    other.input_name)
+            result = concat_batches([result, other])
         return result
+
+    # --------------------------------------------------------------- windows
+    def _compute_window(self, w_: FuncCall, env: Env) -> torch.Tensor:
+        """Window functions over PARTITION BY (full-partition frame):
+        row_number/rank/dense_rank and partition-wide aggregates."""
+        device = env.device
+        n = env.n_rows
+        if w_.over.partition_by:
+            keys = [eval_expr(p, env) for p in w_.over.partition_by]
+            keys = [k if isinstance(k, Column) else as_tensor(k, env)
+                    for k in keys]
+            single = keys[0]
+            if len(keys) == 1 and isinstance(single, torch.Tensor) \
+                    and single.dtype in (torch.int64, torch.int32):
+                gid, _, g = ops.hash_group(single)
+                gid = gid.long()
+            else:
+                gid, _, g = _encode_keys(keys, device)
+        else:
+            gid = torch.zeros(n, dtype=torch.int64, device=device)
+            g = 1
+        name = w_.name
+        if name in ("sum", "count", "avg", "min", "max"):
+            if name == "count" and (not w_.args or isinstance(w_.args[0],
+                                                              Star)):
+                vals = torch.ones(n, device=device)
+            else:
+                vals = as_tensor(eval_expr(w_.args[0], env), env)
+            per_group = ops.segment_reduce(
+                vals.float() if name != "count" else vals, gid.to(torch.int32),
+                g, "mean" if name == "avg" else name)
+            return per_group[gid]
+        # rank family: stable sort by order keys (last-first), then by gid
+        perm = torch.arange(n, dtype=torch.int64, device=device)
+        sort_keys = []
+        for e, asc in reversed(w_.over.order_by):
+            v = eval_expr(e, env)
+            key = v.data if isinstance(v, Column) and v.kind == "numeric" \
+                else as_tensor(v, env)
+            sort_keys.append((key, asc))
+            idx = ops.sort_indices(key[perm], ascending=asc)
+            perm = perm[idx]
+        idx = ops.sort_indices(gid[perm], ascending=True)
+        perm = perm[idx]
+        sorted_gid = gid[perm]
+        ar = torch.arange(n, dtype=torch.int64, device=device)
+        part_start = torch.zeros(n, dtype=torch.bool, device=device)
+        part_start[0] = True
+        part_start[1:] = sorted_gid[1:] != sorted_gid[:-1]
+        group_start = torch.cummax(ar * part_start, 0).values
+        out_sorted = ar - group_start + 1  # row_number
+        if name in ("rank", "dense_rank"):
+            tie_change = part_start.clone()
+            for key, asc in sort_keys:
+                sk = key[perm]
+                tie_change[1:] |= sk[1:] != sk[:-1]
+            if name == "rank":
+                rank_start = torch.cummax(ar * tie_change, 0).values
+                out_sorted = rank_start - group_start + 1
+            else:
+                cum = torch.cumsum(tie_change.long(), 0)
+                base = cum[group_start]
+                out_sorted = cum - base + 1
+        elif name != "row_number":
+            raise SqlError(f"unsupported window function {name}()")
+        out = torch.empty(n, dtype=torch.int64, device=device)
+        out[perm] = out_sorted
+        return out
 
     # --------------------------------------------------------------- filters
     def _filter_indices(self, pred, env: Env) -> torch.Tensor:

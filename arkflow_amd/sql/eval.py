@@ -67,7 +67,13 @@ def expr_name(e) -> str:
         return repr(e.value)
     if isinstance(e, FuncCall):
         args = ", ".join(expr_name(a) for a in e.args)
-        return f"{e.name}({'DISTINCT ' if e.distinct else ''}{args})"
+        base = f"{e.name}({'DISTINCT ' if e.distinct else ''}{args})"
+        if e.over is not None:
+            part = ", ".join(expr_name(p) for p in e.over.partition_by)
+            ob = ", ".join(expr_name(o) + ("" if asc else " desc")
+                           for o, asc in e.over.order_by)
+            base += f" over(partition by {part} order by {ob})"
+        return base
     if isinstance(e, Star):
         return "*"
     if isinstance(e, BinaryOp):
@@ -98,6 +104,8 @@ def _is_agg_name(name: str) -> bool:
 
 def contains_aggregate(e) -> bool:
     if isinstance(e, FuncCall):
+        if e.over is not None:
+            return False  # window functions are row-level, not GROUP BY
         if _is_agg_name(e.name):
             return True
         return any(contains_aggregate(a) for a in e.args)
@@ -123,6 +131,8 @@ def contains_aggregate(e) -> bool:
 def collect_aggregates(e, out: list) -> None:
     """Find aggregate FuncCall nodes (outermost) in an expression tree."""
     if isinstance(e, FuncCall):
+        if e.over is not None:
+            return  # window functions handled separately
         if _is_agg_name(e.name):
             out.append(e)
             return
@@ -212,10 +222,12 @@ def eval_expr(e, env: Env) -> Value:
     if isinstance(e, Star):
         raise SqlError("* only allowed in COUNT(*) or projection list")
     if isinstance(e, FuncCall):
-        if _is_agg_name(e.name):
+        if e.over is not None or _is_agg_name(e.name):
             key = expr_name(e)
             if key in env.agg_results:
                 return env.agg_results[key]
+            if e.over is not None:
+                raise SqlError(f"window fn {key} not computed")
             raise SqlError(f"aggregate {key} outside aggregate context")
         return _eval_func(e, env)
     if isinstance(e, UnaryOp):
@@ -468,3 +480,27 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
                 for a in args]
         return udf(*args)
     raise SqlError(f"unknown function {name}()")
+
+
+def collect_window_calls(e, out: list) -> None:
+    """Find FuncCall nodes with an OVER clause."""
+    if isinstance(e, FuncCall):
+        if e.over is not None:
+            out.append(e)
+            return
+        for a in e.args:
+            collect_window_calls(a, out)
+        return
+    if isinstance(e, BinaryOp):
+        collect_window_calls(e.left, out)
+        collect_window_calls(e.right, out)
+    elif isinstance(e, UnaryOp):
+        collect_window_calls(e.operand, out)
+    elif isinstance(e, Cast):
+        collect_window_calls(e.expr, out)
+    elif isinstance(e, Case):
+        for c, v in e.whens:
+            collect_window_calls(c, out)
+            collect_window_calls(v, out)
+        if e.else_ is not None:
+            collect_window_calls(e.else_, out)

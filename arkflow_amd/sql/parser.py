@@ -53,10 +53,17 @@ class UnaryOp:
 
 
 @dataclass
+class WindowSpec:
+    partition_by: list
+    order_by: list  # [(expr, asc)]
+
+
+@dataclass
 class FuncCall:
     name: str
     args: list
     distinct: bool = False
+    over: "WindowSpec" = None
 
 
 @dataclass
@@ -124,6 +131,8 @@ class Select:
     having: Optional[object] = None
     order_by: List[Tuple[object, bool]] = field(default_factory=list)  # (expr, asc)
     limit: Optional[int] = None
+    offset: int = 0
+    union_all: Optional["Select"] = None
 
 
 AGGREGATE_FUNCS = {"count", "sum", "avg", "min", "max"}
@@ -134,6 +143,8 @@ _KEYWORDS = {
     "limit", "as", "and", "or", "not", "in", "is", "null", "between", "like",
     "case", "when", "then", "else", "end", "cast", "join", "inner", "left",
     "right", "full", "outer", "on", "asc", "desc", "true", "false",
+    "offset", "union", "all", "over", "partition", "row_number", "rank",
+    "dense_rank",
     # rejected verbs (detected for a clear error)
     "insert", "update", "delete", "create", "drop", "alter", "truncate",
 }
@@ -256,6 +267,11 @@ class Parser:
 
     # entry
     def parse(self) -> Select:
+        sel = self._parse_select()
+        self.expect("end")
+        return sel
+
+    def _parse_select(self) -> Select:
         t = self.peek()
         if t.kind == "kw" and t.value in (
             "insert", "update", "delete", "create", "drop", "alter", "truncate"
@@ -311,7 +327,11 @@ class Parser:
                     break
         if self.accept("kw", "limit"):
             sel.limit = int(self.expect("number").value)
-        self.expect("end")
+        if self.accept("kw", "offset"):
+            sel.offset = int(self.expect("number").value)
+        if self.accept("kw", "union"):
+            self.expect("kw", "all")  # UNION (distinct) unsupported; ALL only
+            sel.union_all = self._parse_select()
         return sel
 
     def _table_name(self) -> str:
@@ -470,7 +490,8 @@ class Parser:
             e = self._expr()
             self.expect("op", ")")
             return e
-        if t.kind == "ident" or (t.kind == "kw" and t.value in ("left", "right")):
+        if t.kind == "ident" or (t.kind == "kw" and t.value in (
+                "left", "right", "row_number", "rank", "dense_rank")):
             name = t.value
             # function call
             if self.accept("op", "("):
@@ -483,7 +504,30 @@ class Parser:
                     while self.accept("op", ","):
                         args.append(self._expr())
                 self.expect("op", ")")
-                return FuncCall(name.lower(), args, distinct)
+                over = None
+                if self.accept("kw", "over"):
+                    self.expect("op", "(")
+                    part, order = [], []
+                    if self.accept("kw", "partition"):
+                        self.expect("kw", "by")
+                        part.append(self._expr())
+                        while self.accept("op", ","):
+                            part.append(self._expr())
+                    if self.accept("kw", "order"):
+                        self.expect("kw", "by")
+                        while True:
+                            e = self._expr()
+                            asc = True
+                            if self.accept("kw", "desc"):
+                                asc = False
+                            else:
+                                self.accept("kw", "asc")
+                            order.append((e, asc))
+                            if not self.accept("op", ","):
+                                break
+                    self.expect("op", ")")
+                    over = WindowSpec(part, order)
+                return FuncCall(name.lower(), args, distinct, over)
             # qualified column
             if self.accept("op", "."):
                 col = self.expect("ident").value

@@ -195,3 +195,30 @@ def test_scalar_and_aggregate_udf(flow):
     r = q("SELECT grp, sumsq(value) AS s FROM flow GROUP BY grp ORDER BY grp",
           flow=flow)
     assert r.column("s").to_pylist()[0] == 10.0 ** 2 + 20.0 ** 2
+
+
+def test_offset_and_union_all(flow):
+    r = q("SELECT id FROM flow ORDER BY id LIMIT 2 OFFSET 3", flow=flow)
+    assert r.column("id").to_pylist() == [4, 5]
+    other = MessageBatch.from_dict({"x": [100, 200]})
+    r = q("SELECT id FROM flow WHERE id <= 2 UNION ALL SELECT x FROM other",
+          flow=flow, other=other)
+    assert r.column("id").to_pylist() == [1, 2, 100, 200]
+
+
+def test_window_functions(flow):
+    r = q("SELECT id, row_number() OVER (PARTITION BY grp ORDER BY value) "
+          "AS rn FROM flow ORDER BY id", flow=flow)
+    # grp1: ids 1(v10),2(v20) → rn 1,2; grp2: 3(v5),4(v40) → 1,2;
+    # grp3: 5(v15),6(v20) → 1,2
+    assert r.column("rn").to_pylist() == [1, 2, 1, 2, 1, 2]
+    r = q("SELECT id, sum(value) OVER (PARTITION BY grp) AS s FROM flow "
+          "ORDER BY id", flow=flow)
+    assert r.column("s").to_pylist() == [30.0, 30.0, 45.0, 45.0, 35.0, 35.0]
+    r = q("SELECT id, rank() OVER (ORDER BY value) AS rk FROM flow "
+          "ORDER BY id", flow=flow)
+    # values: 10,20,5,40,15,20 → ranks 2,4,1,6,3,4
+    assert r.column("rk").to_pylist() == [2, 4, 1, 6, 3, 4]
+    r = q("SELECT id, dense_rank() OVER (ORDER BY value) AS dr FROM flow "
+          "ORDER BY id", flow=flow)
+    assert r.column("dr").to_pylist() == [2, 4, 1, 5, 3, 4]
