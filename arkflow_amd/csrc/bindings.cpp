@@ -57,6 +57,8 @@ void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
                          int64_t*, double*, int32_t*, hipStream_t);
 void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
+void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
+                       hipStream_t);
 }
 
 namespace {
@@ -406,6 +408,17 @@ std::vector<torch::Tensor> gather_columns(std::vector<torch::Tensor> cols,
   return outs;
 }
 
+torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
+  check_cuda(data, "data");
+  check_cuda(offsets, "offsets");
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, offsets.options().dtype(torch::kInt64));
+  if (n > 0)
+    launch_bytes_hash(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
+                      n, out.data_ptr<int64_t>(), cur_stream());
+  return out;
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> proto_decode(
     torch::Tensor data, torch::Tensor offsets,
     std::vector<int64_t> fno, std::vector<int64_t> kind,
@@ -458,4 +471,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("proto_decode", &proto_decode);
   m.def("gather_columns", &gather_columns);
   m.def("gemm_bf16_variant", &gemm_bf16_variant);
+  m.def("bytes_hash", &bytes_hash);
 }

@@ -252,3 +252,30 @@ extern "C" void launch_gather_multi(int ncols, const void** src, void** dst,
   if (grid < 1) return;
   gather_multi_kernel<<<grid, 256, 0, st>>>(spec, idx, m);
 }
+
+// ---- per-row bytes hash (binary columns: group-by / repartition keys) -------
+// 64-bit FNV-1a over each row's byte slice; one thread per row. Used to
+// dictionary-encode string keys on-device (collision probability ~n²/2^65 —
+// negligible at stream batch sizes; exact verification is the CPU path).
+__global__ void bytes_hash_kernel(const uint8_t* __restrict__ data,
+                                  const int64_t* __restrict__ offsets,
+                                  int64_t n, int64_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t h = 0xcbf29ce484222325ull;
+    for (int64_t p = offsets[i]; p < offsets[i + 1]; ++p) {
+      h ^= data[p];
+      h *= 0x100000001b3ull;
+    }
+    out[i] = (int64_t)h;
+  }
+}
+
+extern "C" void launch_bytes_hash(const uint8_t* data, const int64_t* offsets,
+                                  int64_t n, int64_t* out, hipStream_t st) {
+  int grid = (int)((n + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  bytes_hash_kernel<<<grid, 256, 0, st>>>(data, offsets, n, out);
+}

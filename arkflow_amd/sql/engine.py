@@ -474,7 +474,19 @@ def _to_column(v, env: Env) -> Column:
 def _encode_keys(keys: list, device) -> Tuple[torch.Tensor, None, int]:
     """Encode 1..k key columns (tensors or binary Columns) into group ids."""
     if any(isinstance(k, Column) and k.kind == "binary" for k in keys):
-        # dictionary-encode binary keys host-side (GPU-native path: round 2)
+        if all((k.data.is_cuda if isinstance(k, Column) else k.is_cuda)
+               for k in keys):
+            # device path: FNV-1a row hashes for binary keys, then the same
+            # numeric encoding (csrc bytes_hash kernel)
+            from ..ops import native_available, require_native
+            if native_available():
+                nat = require_native()
+                ts = [nat.bytes_hash(k.data, k.offsets)
+                      if isinstance(k, Column) and k.kind == "binary"
+                      else (k.data if isinstance(k, Column) else k)
+                      for k in keys]
+                return _encode_keys(ts, device)
+        # dictionary-encode binary keys host-side (CPU / fallback)
         lists = []
         for k in keys:
             if isinstance(k, Column):

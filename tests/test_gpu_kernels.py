@@ -283,3 +283,23 @@ def test_join_large(nat, dev):
     rk = torch.randint(0, 1 << 20, (600_000,), device=dev, dtype=torch.int64)
     l_idx, r_idx = nat.join_inner_i64(lk, rk)
     assert torch.equal(lk[l_idx], rk[r_idx])
+
+
+def test_bytes_hash_and_string_groupby(nat, dev):
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+    col = Column.from_strings(["ab", "cd", "ab", "", "cd", "ab"]).to(dev)
+    h = nat.bytes_hash(col.data, col.offsets)
+    hl = h.cpu().tolist()
+    assert hl[0] == hl[2] == hl[5] and hl[1] == hl[4]
+    assert len({hl[0], hl[1], hl[3]}) == 3
+    flow = MessageBatch({
+        "name": col,
+        "v": Column.from_numeric(
+            torch.tensor([1., 2., 3., 4., 5., 6.], device=dev)),
+    })
+    r = SqlExecutor("SELECT name, sum(v) AS s FROM flow GROUP BY name "
+                    "ORDER BY s").execute({"flow": flow})
+    assert r.column("s").to_pylist() == [4.0, 7.0, 10.0]
+    names = r.column("name").to_strlist()
+    assert names == ["", "cd", "ab"]
