@@ -237,3 +237,83 @@ def test_runtime_lifecycle(run):
         assert {"registered", "starting", "running", "stopping"} <= ops_kinds
 
     run(main())
+
+
+def test_replace_config_and_rollback(run):
+    """reference runtime.rs:554-632 replace_config with rollback-on-failure."""
+    async def main():
+        cfg = EngineConfig.from_dict({
+            "streams": [{
+                "id": "s1",
+                "input": {"type": "generate", "batch_size": 2,
+                          "interval": "10ms",
+                          "fields": {"v": {"dtype": "float32"}}},
+                "output": {"type": "drop"},
+            }]
+        })
+        eng = af.Engine(cfg)
+        eng.runtime.register(cfg.streams[0])
+        await eng.runtime.start("s1")
+        # replace with a valid config
+        from arkflow_amd.config import _parse_stream
+        new_sc = _parse_stream({
+            "id": "s1",
+            "input": {"type": "generate", "batch_size": 5, "interval": "10ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "output": {"type": "memory"},
+        }, 0)
+        await eng.runtime.replace_config("s1", new_sc)
+        assert eng.runtime.get("s1").state.value == "running"
+        assert eng.runtime.get("s1").config.output == {"type": "memory"}
+        # replace with a BROKEN config → rollback to previous, still running
+        bad_sc = _parse_stream({
+            "id": "s1",
+            "input": {"type": "generate", "batch_size": 1},
+            "output": {"type": "sql"},  # missing 'table' → build error
+        }, 0)
+        import pytest as _pt
+        with _pt.raises(Exception):
+            await eng.runtime.replace_config("s1", bad_sc)
+        assert eng.runtime.get("s1").state.value == "running"
+        assert eng.runtime.get("s1").config.output == {"type": "memory"}
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=30)
+
+
+def test_control_plane_apply_and_rollback(run):
+    async def main():
+        cfg = EngineConfig.from_dict({
+            "streams": [{
+                "id": "s1",
+                "input": {"type": "generate", "batch_size": 1,
+                          "interval": "20ms",
+                          "fields": {"v": {"dtype": "float32"}}},
+                "output": {"type": "drop"},
+            }]
+        })
+        eng = af.Engine(cfg)
+        for sc in cfg.streams:
+            eng.runtime.register(sc)
+        await eng.runtime.start("s1")
+        cp = eng.control_plane
+        v1 = {"streams": [{"id": "s1",
+                           "input": {"type": "generate", "batch_size": 2,
+                                     "interval": "20ms",
+                                     "fields": {"v": {"dtype": "float32"}}},
+                           "output": {"type": "drop"}}]}
+        r1 = await cp.apply_configuration(v1, note="v1")
+        assert r1["applied"] and r1["version"] == 1
+        v2 = {"streams": [{"id": "s2",
+                           "input": {"type": "generate", "batch_size": 1,
+                                     "interval": "20ms",
+                                     "fields": {"v": {"dtype": "float32"}}},
+                           "output": {"type": "drop"}}]}
+        r2 = await cp.apply_configuration(v2, note="v2")
+        assert r2["applied"] and set(eng.runtime.entries) == {"s2"}
+        rb = await cp.rollback(1)
+        assert rb["applied"] and set(eng.runtime.entries) == {"s1"}
+        assert len(cp.versions.list()) == 3  # v1, v2, rollback-apply
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=30)
