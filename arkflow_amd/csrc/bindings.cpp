@@ -44,6 +44,8 @@ void launch_gemm_bf16(const void*, const void*, const float*, void*, int, int,
                       int, int, hipStream_t);
 int launch_gemm_bf16_8p(const void*, const void*, const float*, void*, int,
                         int, int, int, int, hipStream_t);
+int launch_gemm_bf16_2p(const void*, const void*, const float*, void*, int,
+                        int, int, int, hipStream_t);
 void launch_layernorm_bf16(const void*, const void*, const float*,
                            const float*, void*, void*, int64_t, int, float,
                            hipStream_t);
@@ -300,6 +302,11 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
   if (variant == 0) {
     launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
                      (int)M, (int)N, (int)K, (int)act, cur_stream());
+  } else if (variant == 3) {
+    int rc = launch_gemm_bf16_2p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                 C.data_ptr(), (int)M, (int)N, (int)K,
+                                 (int)act, cur_stream());
+    TORCH_CHECK(rc == 0, "shape not supported by 2-phase kernel");
   } else {
     int rc = launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                  C.data_ptr(), (int)M, (int)N, (int)K,

@@ -138,7 +138,158 @@ void gemm_bf16_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- 2-phase double-buffered variant -----------------------------------------
+// T3-minimal prefetch (guide §5.5): stage K-tile t+1 into the other LDS
+// buffer while computing tile t; ONE vmcnt(0)+barrier per K-tile AFTER the
+// MFMAs so the staging loads overlap compute. Pays at short-K / latency-bound
+// shapes (BERT: K=768) where block occupancy can't hide the HBM latency.
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 2)
+void gemm_bf16_2p_kernel(const __bf16* __restrict__ A,
+                         const __bf16* __restrict__ Bt,
+                         const float* __restrict__ bias,
+                         __bf16* __restrict__ C,
+                         int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * BM, col0 = bn * BN;
+
+  __shared__ __bf16 Asm[2][BM * BK];
+  __shared__ __bf16 Bsm[2][BN * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+  const int lin0 = wid * 1024 + lane * 16;
+  const int NT = K / BK;
+
+  auto stage = [&](int buf, int t) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int lin = lin0 + it * 4096;
+      int trow = lin >> 6;
+      int tcol = lin & 63;
+      int ga_row = row0 + trow;
+      ga_row = ga_row < M ? ga_row : M - 1;
+      const char* a_src =
+          (const char*)(A + (int64_t)ga_row * K + t * BK) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)a_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Asm[buf] +
+                                                        lin),
+          16, 0, 0);
+      int gb_row = col0 + trow;
+      gb_row = gb_row < N ? gb_row : N - 1;
+      const char* b_src =
+          (const char*)(Bt + (int64_t)gb_row * K + t * BK) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)b_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Bsm[buf] +
+                                                        lin),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __builtin_amdgcn_sched_barrier(0);
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (int t = 0; t < NT; ++t) {
+    if (t + 1 < NT) stage(cur ^ 1, t + 1);  // prefetch next tile
+    bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      a_frag[m] = *(const bf16x8*)&Asm[cur][(wm + m * 16 + fr) * BK + fk];
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+      b_frag[n] = *(const bf16x8*)&Bsm[cur][(wn + n * 16 + fr) * BK + fk];
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_sched_barrier(0);
+    // next tile must be resident before anyone reads buf cur^1
+    asm volatile("s_waitcnt vmcnt(0)");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+template <int ACT>
+static void dispatch_2p(const void* A, const void* Bt, const float* bias,
+                        void* C, int M, int N, int K, int tiles_n, dim3 grid,
+                        dim3 block, hipStream_t st) {
+  if (bias)
+    gemm_bf16_2p_kernel<ACT, true><<<grid, block, 0, st>>>(
+        (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,
+        tiles_n);
+  else
+    gemm_bf16_2p_kernel<ACT, false><<<grid, block, 0, st>>>(
+        (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,
+        tiles_n);
+}
+
 extern "C" {
+
+int launch_gemm_bf16_2p(const void* A, const void* Bt, const float* bias,
+                        void* C, int M, int N, int K, int act,
+                        hipStream_t st) {
+  if (K % BK != 0 || K / BK < 2) return -1;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n), block(GEMM_THREADS);
+  switch (act) {
+    case ACT_RELU: dispatch_2p<ACT_RELU>(A, Bt, bias, C, M, N, K, tiles_n,
+                                         grid, block, st); break;
+    case ACT_GELU: dispatch_2p<ACT_GELU>(A, Bt, bias, C, M, N, K, tiles_n,
+                                         grid, block, st); break;
+    case ACT_SILU: dispatch_2p<ACT_SILU>(A, Bt, bias, C, M, N, K, tiles_n,
+                                         grid, block, st); break;
+    default: dispatch_2p<ACT_NONE>(A, Bt, bias, C, M, N, K, tiles_n, grid,
+                                   block, st); break;
+  }
+  return 0;
+}
 
 void launch_gemm_bf16(const void* A, const void* Bt, const float* bias,
                       void* C, int M, int N, int K, int act, hipStream_t st) {
