@@ -21,9 +21,13 @@ from ..spi import Ack, Buffer, VecAck
 
 class BaseWindowBuffer(Buffer):
     def __init__(self, config: dict, resource=None):
-        # per-input-name FIFO of (batch, ack) (reference window.rs DashMap)
-        self.queues: Dict[str, Deque[Tuple[MessageBatch, Ack]]] = \
+        # per-input-name FIFO of (item, ack) where item is a MessageBatch or
+        # an absolute (start, end) range into that input's device ring
+        # (reference window.rs DashMap of batches; SURVEY §2.9 ring mapping)
+        self.queues: Dict[str, Deque[Tuple[object, Ack]]] = \
             defaultdict(deque)
+        self.use_ring = bool(config.get("device_ring", True))
+        self.rings: Dict[str, "DeviceRingBuffer"] = {}
         self._notify = asyncio.Event()
         self._draining = False
         self._closed = False
@@ -41,9 +45,60 @@ class BaseWindowBuffer(Buffer):
     # ------------------------------------------------------------------ write
     async def write(self, batch: MessageBatch, ack: Ack) -> None:
         name = batch.input_name or "default"
-        self.queues[name].append((batch, ack))
+        item: object = batch
+        if self.use_ring:
+            from .ring import DeviceRingBuffer
+            ring = self.rings.get(name)
+            if ring is None:
+                ring = self.rings[name] = DeviceRingBuffer()
+            rng = ring.append(batch)
+            if rng is not None:
+                item = rng  # the window holds a view range, not a copy
+        self.queues[name].append((item, ack))
         self.on_write(batch)
         self._notify.set()
+
+    def _materialize(self, name: str, items: list) -> MessageBatch:
+        """items: ordered [(item, ack)] for one input → one batch.
+        Consecutive ring ranges coalesce into zero-copy slices."""
+        ring = self.rings.get(name)
+        parts: List[MessageBatch] = []
+        pending_ranges: List[Tuple[int, int]] = []
+
+        def flush_ranges():
+            if pending_ranges:
+                parts.append(ring.slice_many(list(pending_ranges)))
+                pending_ranges.clear()
+
+        for it, _ in items:
+            if isinstance(it, tuple):
+                pending_ranges.append(it)
+            else:
+                flush_ranges()
+                parts.append(it)
+        flush_ranges()
+        out = parts[0] if len(parts) == 1 else concat_batches(parts)
+        return out
+
+    def _release_ack(self, inner: Ack, items_by_name: dict) -> Ack:
+        """Ring rows may be overwritten only after downstream ACKS the emitted
+        window (the emitted batch is a VIEW over the ring). Wrap the combined
+        ack so release happens post-ack; un-acked windows keep their rows
+        (the ring grows instead of overwriting)."""
+        rings = self.rings
+
+        class _RingReleaseAck(Ack):
+            async def ack(self_inner) -> None:
+                await inner.ack()
+                for name, lst in items_by_name.items():
+                    ring = rings.get(name)
+                    if ring is None:
+                        continue
+                    ends = [it[1] for it, _ in lst if isinstance(it, tuple)]
+                    if ends:
+                        ring.release_before(max(ends))
+
+        return _RingReleaseAck()
 
     def on_write(self, batch: MessageBatch) -> None:
         """Subclass hook (e.g. session gap tracking)."""
@@ -112,7 +167,7 @@ class BaseWindowBuffer(Buffer):
                     self.queues[name].extendleft(reversed(lst))
                 return None
             tables = {
-                name: concat_batches([b for b, _ in lst])
+                name: self._materialize(name, lst)
                 for name, lst in items.items()
             }
             from ..sql.engine import SqlExecutor
@@ -120,17 +175,19 @@ class BaseWindowBuffer(Buffer):
             if first in tables:
                 tables.setdefault("flow", tables[first])
             result = SqlExecutor(self.join_query).execute(tables)
-            return result, VecAck(acks)
-        batches = [b for lst in items.values() for b, _ in lst]
+            return result, self._release_ack(VecAck(acks), items)
         try:
-            combined = concat_batches(batches)
+            per_input = [self._materialize(name, lst)
+                         for name, lst in items.items()]
+            combined = per_input[0] if len(per_input) == 1 \
+                else concat_batches(per_input)
         except ValueError:
-            # heterogeneous schemas: emit the first input's batches only,
-            # re-buffer the rest? Reference concats per input; emit per-input
-            # sequentially instead.
+            # heterogeneous schemas across inputs: emit the first input now,
+            # re-buffer the rest (reference concats per input)
             name, lst = next(iter(items.items()))
             for other, olst in list(items.items())[1:]:
                 self.queues[other].extendleft(reversed(olst))
-            combined = concat_batches([b for b, _ in lst])
+            combined = self._materialize(name, lst)
             acks = [a for _, a in lst]
-        return combined, VecAck(acks)
+            return combined, self._release_ack(VecAck(acks), {name: lst})
+        return combined, self._release_ack(VecAck(acks), items)

@@ -178,3 +178,51 @@ def test_window_join_waits_for_all_inputs(run):
         assert batch.num_rows == 1
 
     run(main())
+
+
+def test_device_ring_buffer_views():
+    import torch
+    from arkflow_amd.buffers.ring import DeviceRingBuffer
+    ring = DeviceRingBuffer(capacity=8)
+    b1 = MessageBatch.from_dict({"v": torch.tensor([1., 2., 3.])})
+    b2 = MessageBatch.from_dict({"v": torch.tensor([4., 5.])})
+    r1 = ring.append(b1)
+    r2 = ring.append(b2)
+    assert r1 == (0, 3) and r2 == (3, 5)
+    s = ring.slice_many([r1, r2])
+    assert s.column("v").to_pylist() == [1., 2., 3., 4., 5.]
+    # zero-copy: contiguous range shares storage with the ring
+    assert s.column("v").data.data_ptr() == ring.cols["v"].data_ptr()
+    # wrap-around after release
+    ring.release_before(5)
+    r3 = ring.append(MessageBatch.from_dict(
+        {"v": torch.tensor([6., 7., 8., 9., 10.])}))
+    assert r3 == (5, 10)
+    assert ring.slice(*r3).column("v").to_pylist() == [6., 7., 8., 9., 10.]
+    # growth with live (unreleased) data preserved
+    big = MessageBatch.from_dict({"v": torch.arange(20, dtype=torch.float32)})
+    r4 = ring.append(big)
+    assert ring.slice(*r4).column("v").to_pylist() == list(range(20))
+    assert ring.slice(*r3).column("v").to_pylist() == [6., 7., 8., 9., 10.]
+    # binary columns are not ring-able
+    assert ring.append(MessageBatch.from_dict({"s": ["x"]})) is None
+
+
+def test_window_ring_release_on_ack(run):
+    """Ring rows are reusable only after the emitted window is ACKED
+    (the emitted batch is a view over the ring)."""
+    async def main():
+        buf = MemoryBuffer({"capacity": 2, "device_ring": True})
+        log = []
+        import torch
+        await buf.write(MessageBatch.from_dict(
+            {"v": torch.tensor([1., 2.])}, input_name="in"), TAck(log, 1))
+        batch, ack = await asyncio.wait_for(buf.read(), 2)
+        assert batch.column("v").to_pylist() == [1., 2.]
+        ring = buf.rings["in"]
+        assert ring.head == 0  # not yet released
+        await ack.ack()
+        assert ring.head == 2  # released after ack
+        assert log == [1]
+
+    run(main())
