@@ -61,6 +61,19 @@ void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
                        hipStream_t);
+int radix_sort_nblocks(int64_t);
+void launch_radix_hist_u32(const uint32_t*, const int32_t*, int64_t, int,
+                           int32_t*, int, hipStream_t);
+void launch_radix_scatter_u32(const uint32_t*, const int32_t*, int64_t, int,
+                              const int32_t*, int, int32_t*, hipStream_t);
+void launch_radix_hist_u64(const uint64_t*, const int32_t*, int64_t, int,
+                           int32_t*, int, hipStream_t);
+void launch_radix_scatter_u64(const uint64_t*, const int32_t*, int64_t, int,
+                              const int32_t*, int, int32_t*, hipStream_t);
+void launch_f32_to_ordered(const float*, uint32_t*, int64_t, int,
+                           hipStream_t);
+void launch_i64_to_ordered(const int64_t*, uint64_t*, int64_t, int,
+                           hipStream_t);
 }
 
 namespace {
@@ -415,6 +428,63 @@ std::vector<torch::Tensor> gather_columns(std::vector<torch::Tensor> cols,
   return outs;
 }
 
+torch::Tensor radix_argsort(torch::Tensor keys, bool descending) {
+  check_cuda(keys, "keys");
+  int64_t n = keys.numel();
+  auto opts32 = keys.options().dtype(torch::kInt32);
+  if (n == 0) return torch::empty({0}, opts32);
+  auto st = cur_stream();
+  bool wide;  // 64-bit key path
+  torch::Tensor ordered;
+  if (keys.scalar_type() == torch::kFloat32) {
+    wide = false;
+    ordered = torch::empty({n}, opts32);  // uint32 bits in an int32 tensor
+    launch_f32_to_ordered(keys.data_ptr<float>(),
+                          (uint32_t*)ordered.data_ptr<int32_t>(), n,
+                          descending ? 1 : 0, st);
+  } else if (keys.scalar_type() == torch::kInt64 ||
+             keys.scalar_type() == torch::kInt32) {
+    wide = true;
+    auto k64 = keys.scalar_type() == torch::kInt64 ? keys
+                                                   : keys.to(torch::kInt64);
+    ordered = torch::empty({n}, keys.options().dtype(torch::kInt64));
+    launch_i64_to_ordered(k64.data_ptr<int64_t>(),
+                          (uint64_t*)ordered.data_ptr<int64_t>(), n,
+                          descending ? 1 : 0, st);
+  } else {
+    TORCH_CHECK(false, "radix_argsort supports f32/i64/i32 keys");
+  }
+  int nblocks = radix_sort_nblocks(n);
+  auto idx_a = torch::arange(n, opts32);
+  auto idx_b = torch::empty({n}, opts32);
+  auto hist = torch::empty({(int64_t)16 * nblocks}, opts32);
+  int total_bits = wide ? 64 : 32;
+  for (int shift = 0; shift < total_bits; shift += 4) {
+    if (wide)
+      launch_radix_hist_u64((const uint64_t*)ordered.data_ptr<int64_t>(),
+                            idx_a.data_ptr<int32_t>(), n, shift,
+                            hist.data_ptr<int32_t>(), nblocks, st);
+    else
+      launch_radix_hist_u32((const uint32_t*)ordered.data_ptr<int32_t>(),
+                            idx_a.data_ptr<int32_t>(), n, shift,
+                            hist.data_ptr<int32_t>(), nblocks, st);
+    auto inc = hist.cumsum(0, torch::kInt32);
+    auto offs = (inc - hist).contiguous();
+    if (wide)
+      launch_radix_scatter_u64((const uint64_t*)ordered.data_ptr<int64_t>(),
+                               idx_a.data_ptr<int32_t>(), n, shift,
+                               offs.data_ptr<int32_t>(), nblocks,
+                               idx_b.data_ptr<int32_t>(), st);
+    else
+      launch_radix_scatter_u32((const uint32_t*)ordered.data_ptr<int32_t>(),
+                               idx_a.data_ptr<int32_t>(), n, shift,
+                               offs.data_ptr<int32_t>(), nblocks,
+                               idx_b.data_ptr<int32_t>(), st);
+    std::swap(idx_a, idx_b);
+  }
+  return idx_a;
+}
+
 torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
@@ -479,4 +549,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_columns", &gather_columns);
   m.def("gemm_bf16_variant", &gemm_bf16_variant);
   m.def("bytes_hash", &bytes_hash);
+  m.def("radix_argsort", &radix_argsort, py::arg("keys"), py::arg("descending") = false);
 }

@@ -143,7 +143,12 @@ def segment_reduce(values: torch.Tensor, group_ids: torch.Tensor,
 
 # ----------------------------------------------------------------------- sort
 def sort_indices(key: torch.Tensor, ascending: bool = True) -> torch.Tensor:
-    """argsort (stable). GPU fast path: device radix sort (later round)."""
+    """Stable argsort. GPU: native LSD radix sort (csrc/radix_sort.hip) for
+    f32/i64/i32 keys; torch elsewhere."""
+    if _use_native(key) and key.dtype in (torch.float32, torch.int64,
+                                          torch.int32):
+        return require_native().radix_argsort(
+            key.contiguous(), not ascending).long()
     return torch.argsort(key, stable=True, descending=not ascending)
 
 

@@ -303,3 +303,22 @@ def test_bytes_hash_and_string_groupby(nat, dev):
     assert r.column("s").to_pylist() == [4.0, 7.0, 10.0]
     names = r.column("name").to_strlist()
     assert names == ["", "cd", "ab"]
+
+
+def test_radix_argsort(nat, dev):
+    torch.manual_seed(40)
+    for n in (1, 63, 1000, 300_000):
+        f = torch.randn(n, device=dev) * 1000
+        got = nat.radix_argsort(f, False).long()
+        exp = torch.argsort(f, stable=True)
+        assert torch.equal(f[got], f[exp]), f"f32 asc n={n}"
+        got_d = nat.radix_argsort(f, True).long()
+        assert torch.equal(f[got_d], torch.sort(f, descending=True).values)
+        i = torch.randint(-10**12, 10**12, (n,), device=dev,
+                          dtype=torch.int64)
+        got = nat.radix_argsort(i, False).long()
+        assert torch.equal(i[got], torch.sort(i).values), f"i64 n={n}"
+    # stability: equal keys keep input order
+    k = torch.tensor([2., 1., 2., 1., 2.], device=dev)
+    got = nat.radix_argsort(k, False).long().cpu().tolist()
+    assert got == [1, 3, 0, 2, 4]
