@@ -201,6 +201,10 @@ def render_prometheus(engine) -> str:
         lines.append(
             f'arkflow_stream_uptime_seconds{{stream="{sid}"}} '
             f'{snap["uptime_secs"]:.3f}')
+        for stage, ms in snap["stage_ms"].items():
+            lines.append(
+                f'arkflow_stage_ms_total{{stream="{sid}",stage="{stage}"}} '
+                f'{ms:.3f}')
     return "\n".join(lines) + "\n"
 
 

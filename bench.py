@@ -107,10 +107,16 @@ def build_pipeline(args, device):
     return gen, Pipeline([sql, infer])
 
 
-async def run_steps(gen, pipeline, n_steps):
+async def run_steps(gen, pipeline, n_steps, workers=1):
+    """Process n_steps batches. workers>1 pipelines steps concurrently the
+    way the engine's thread_num processor workers do (stream.py), overlapping
+    host dispatch of step k+1 with device execution of step k; per-step
+    latency (p50) is still measured per batch with a sync."""
     rows = 0
     times = []
-    for _ in range(n_steps):
+
+    async def one_step():
+        nonlocal rows
         t0 = time.perf_counter()
         batch, ack = await gen.read()
         outs = await pipeline.process(batch)
@@ -120,6 +126,19 @@ async def run_steps(gen, pipeline, n_steps):
             torch.cuda.synchronize()
         times.append(time.perf_counter() - t0)
         await ack.ack()
+
+    if workers <= 1:
+        for _ in range(n_steps):
+            await one_step()
+        return rows, times
+
+    sem = asyncio.Semaphore(workers)
+
+    async def guarded():
+        async with sem:
+            await one_step()
+
+    await asyncio.gather(*[guarded() for _ in range(n_steps)])
     return rows, times
 
 
@@ -133,6 +152,8 @@ def main():
                    default="mlp")
     p.add_argument("--features", type=int, default=16)
     p.add_argument("--hidden", type=int, default=256)
+    p.add_argument("--workers", type=int, default=2,
+                   help="concurrent in-flight steps (engine thread_num analog)")
     args = p.parse_args()
 
     args.rank = int(os.environ.get("RANK", 0))
@@ -156,7 +177,8 @@ def main():
     loop = asyncio.new_event_loop()
 
     # warmup (untimed)
-    loop.run_until_complete(run_steps(gen, pipeline, args.warmup))
+    loop.run_until_complete(run_steps(gen, pipeline, args.warmup,
+                                      args.workers))
 
     # timed region: barrier + sync on both sides
     if dist:
@@ -164,7 +186,8 @@ def main():
     if device.type == "cuda":
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    rows, times = loop.run_until_complete(run_steps(gen, pipeline, args.steps))
+    rows, times = loop.run_until_complete(
+        run_steps(gen, pipeline, args.steps, args.workers))
     if device.type == "cuda":
         torch.cuda.synchronize()
     if dist:
@@ -212,6 +235,7 @@ def main():
                 "seq_len": 128 if args.model == "bert" else 1,
                 "parallelism": f"dp{world if world > 1 else args.gpus}",
                 "batch_size_per_gpu": args.batch_size,
+                "workers": args.workers,
                 "p50_ms": p50_ms,
                 "filter": "WHERE f0 >= 0.2" if args.model == "mlp" else None,
             },
