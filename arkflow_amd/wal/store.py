@@ -31,11 +31,14 @@ _DTYPES = {
 }
 
 
-def _tensor_bytes(t: torch.Tensor) -> bytes:
+def _tensor_bytes(t: torch.Tensor):
+    """Zero-copy buffer view of a CPU tensor (memoryview joins fine)."""
     t = t.detach().to("cpu").contiguous()
     if t.dtype == torch.bfloat16:
-        return t.view(torch.int16).numpy().tobytes()
-    return t.numpy().tobytes()
+        t = t.view(torch.int16)
+    if t.dtype == torch.bool:
+        t = t.to(torch.uint8)
+    return memoryview(t.numpy()).cast("B")
 
 
 def _tensor_from(data: bytes, dtype: torch.dtype, n: int) -> torch.Tensor:
@@ -61,18 +64,19 @@ def serialize_batch(batch: MessageBatch) -> bytes:
         if kind == 0:
             dt = str(col.data.dtype).replace("torch.", "").encode()
             data = _tensor_bytes(col.data)
-            parts.append(struct.pack(">HqI", len(dt), len(col), len(data)))
+            parts.append(struct.pack(">HqI", len(dt), len(col), data.nbytes))
             parts.append(dt)
             parts.append(data)
         else:
             data = _tensor_bytes(col.data)
             offs = _tensor_bytes(col.offsets)
-            parts.append(struct.pack(">qII", len(col), len(data), len(offs)))
+            parts.append(struct.pack(">qII", len(col), data.nbytes,
+                                     offs.nbytes))
             parts.append(data)
             parts.append(offs)
         if has_validity:
             v = _tensor_bytes(col.validity)
-            parts.append(struct.pack(">I", len(v)))
+            parts.append(struct.pack(">I", v.nbytes))
             parts.append(v)
     return b"".join(parts)
 
@@ -124,12 +128,14 @@ def deserialize_batch(buf: bytes) -> MessageBatch:
 # --------------------------------------------------------------------- frames
 def encode_frame(seq: int, payload: bytes, compress: bool = False) -> bytes:
     if compress:
-        payload = b"Z" + zlib.compress(payload, 1)
+        body = zlib.compress(payload, 1)
+        tag = b"Z"
     else:
-        payload = b"R" + payload
-    crc = zlib.crc32(payload)
-    return struct.pack(">QI", seq, len(payload)) + payload + \
-        struct.pack(">I", crc)
+        body = payload
+        tag = b"R"
+    crc = zlib.crc32(body, zlib.crc32(tag))
+    return b"".join([struct.pack(">QI", seq, len(body) + 1), tag, body,
+                     struct.pack(">I", crc)])
 
 
 def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:

@@ -63,13 +63,16 @@ class Wal:
     async def append(self, batch) -> int:
         self._seq += 1
         seq = self._seq
-        payload = serialize_batch(batch)
         if self.sync_policy == "per_entry":
+            payload = serialize_batch(batch)
             loop = asyncio.get_running_loop()
             await loop.run_in_executor(
                 None, self.store.append_batch, [(seq, payload)], True)
             return seq
-        self._pending.append((seq, payload))
+        # staged: serialization (incl. any D2H) happens in the flusher
+        # thread, keeping the ingest loop free (reference stages + notifies,
+        # wal/mod.rs:312-338 — its ~1-50 µs staged append defers the work)
+        self._pending.append((seq, batch))
         self._ensure_flusher()
         self._pending_event.set()
         return seq
@@ -96,10 +99,14 @@ class Wal:
     async def flush_pending(self) -> None:
         if not self._pending:
             return
-        batch, self._pending = self._pending, []
+        staged, self._pending = self._pending, []
         loop = asyncio.get_running_loop()
-        await loop.run_in_executor(
-            None, self.store.append_batch, batch, True)
+
+        def work():
+            entries = [(seq, serialize_batch(b)) for seq, b in staged]
+            self.store.append_batch(entries, True)
+
+        await loop.run_in_executor(None, work)
 
     # ----------------------------------------------------------------- advance
     async def advance(self, seq: int) -> None:

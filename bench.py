@@ -115,6 +115,8 @@ async def run_steps(gen, pipeline, n_steps, workers=1):
     rows = 0
     times = []
 
+    loop = asyncio.get_running_loop()
+
     async def one_step():
         nonlocal rows
         t0 = time.perf_counter()
@@ -123,7 +125,12 @@ async def run_steps(gen, pipeline, n_steps, workers=1):
         for b in outs:
             rows += b.num_rows
         if batch.device.type == "cuda":
-            torch.cuda.synchronize()
+            # event-based wait in an executor thread so OTHER in-flight
+            # steps keep dispatching while this one drains (a blocking
+            # torch.cuda.synchronize would stall the whole event loop)
+            ev = torch.cuda.Event()
+            ev.record()
+            await loop.run_in_executor(None, ev.synchronize)
         times.append(time.perf_counter() - t0)
         await ack.ack()
 
@@ -152,7 +159,7 @@ def main():
                    default="mlp")
     p.add_argument("--features", type=int, default=16)
     p.add_argument("--hidden", type=int, default=256)
-    p.add_argument("--workers", type=int, default=2,
+    p.add_argument("--workers", type=int, default=1,
                    help="concurrent in-flight steps (engine thread_num analog)")
     args = p.parse_args()
 
