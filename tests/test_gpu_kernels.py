@@ -322,3 +322,44 @@ def test_radix_argsort(nat, dev):
     k = torch.tensor([2., 1., 2., 1., 2.], device=dev)
     got = nat.radix_argsort(k, False).long().cpu().tolist()
     assert got == [1, 3, 0, 2, 4]
+
+
+def test_full_engine_on_gpu(dev):
+    """The actual async engine (stream task graph, backpressure, acks) with
+    GPU-resident batches: generate → sql filter+agg → inference → memory."""
+    import asyncio
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "gpu-e2e",
+            "device": "cuda:0",
+            "input": {"type": "generate", "batch_size": 8192, "count": 8192 * 20,
+                      "interval": "0ms",
+                      "fields": {"f0": {"dtype": "float32", "low": 0,
+                                        "high": 1},
+                                 "f1": {"dtype": "float32", "low": 0,
+                                        "high": 1},
+                                 "key": {"dtype": "int64", "low": 0,
+                                         "high": 64}}},
+            "pipeline": {"thread_num": 2, "processors": [
+                {"type": "sql",
+                 "query": "SELECT * FROM flow WHERE f0 >= 0.5"},
+                {"type": "inference", "model": "mlp_anomaly",
+                 "columns": ["f0", "f1"], "device": "cuda:0"},
+                {"type": "sql",
+                 "query": "SELECT key, count(*) AS c, max(score) AS mx "
+                          "FROM flow GROUP BY key"},
+            ]},
+            "output": {"type": "memory"},
+        }]
+    })
+    eng = af.Engine(cfg)
+    asyncio.new_event_loop().run_until_complete(
+        asyncio.wait_for(eng.run_with_cancellation(), 120))
+    e = eng.runtime.entries["gpu-e2e"]
+    assert e.state.value == "stopped"
+    assert e.metrics.input_messages == 8192 * 20
+    assert e.metrics.output_batches == 20  # one agg result per batch
+    assert e.metrics.processing_errors == 0
