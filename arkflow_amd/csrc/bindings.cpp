@@ -283,15 +283,23 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
     TORCH_CHECK(bias_f.numel() == N, "bias must be [N]");
     bias_ptr = bias_f.data_ptr<float>();
   }
-  // prefer the 8-phase 256² pipelined kernel (with LDS swizzle) where it
-  // measured faster than the 128² tile: chip-filling grids with enough K
-  // to amortize the pipeline (see profiles/ r02 GEMM table)
-  int64_t tiles = ((M + 255) / 256) * ((N + 255) / 256);
-  bool use8p = tiles >= 128 && K >= 1536;
-  int rc = use8p ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
-                                       C.data_ptr(), (int)M, (int)N, (int)K,
-                                       (int)act, /*swz=*/1, cur_stream())
-                 : -1;
+  // measured dispatch (profiles/ r02/r06 GEMM tables): big square shapes →
+  // 8-phase BM=256+swizzle; chip-filling mid shapes → 8-phase BM=128+swizzle;
+  // everything else → the 128² single-buffer tile.
+  int64_t t256 = ((M + 255) / 256) * ((N + 255) / 256);
+  int64_t t128 = ((M + 127) / 128) * ((N + 255) / 256);
+  // NOTE: BM128+swz wins STANDALONE at BERT shapes (+8-16%) but measured
+  // ~2% slower in-context (L2-warm inputs; see profiles/ r06) — so only the
+  // big-shape case dispatches to the pipelined kernel.
+  (void)t128;
+  int swz_code = -1;
+  if (t256 >= 128 && K >= 1536 && N >= 2048)
+    swz_code = 1;  // BM256 + swizzle
+  int rc = swz_code >= 0
+               ? launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                     C.data_ptr(), (int)M, (int)N, (int)K,
+                                     (int)act, swz_code, cur_stream())
+               : -1;
   if (rc != 0)
     launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
                      (int)M, (int)N, (int)K, (int)act, cur_stream());
@@ -321,10 +329,11 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
                                  (int)act, cur_stream());
     TORCH_CHECK(rc == 0, "shape not supported by 2-phase kernel");
   } else {
+    // 8-phase variants: 1=BM256, 2=BM256+swz, 4=BM128, 5=BM128+swz
+    int swz_code = variant == 2 ? 1 : variant == 4 ? 2 : variant == 5 ? 3 : 0;
     int rc = launch_gemm_bf16_8p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                  C.data_ptr(), (int)M, (int)N, (int)K,
-                                 (int)act, variant == 2 ? 1 : 0,
-                                 cur_stream());
+                                 (int)act, swz_code, cur_stream());
     TORCH_CHECK(rc == 0, "shape not supported by 8-phase kernel");
   }
   return C;

@@ -24,11 +24,11 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define P8_THREADS 512
-#define P8_BM 256
 #define P8_BN 256
 #define P8_BK 64
-// one half-tile = 128 rows × 64 cols bf16 = 16 KiB = 2 global_load_lds/thread
-#define HALF_BYTES (128 * 64 * 2)
+// B half-tile = 128 rows × 64 cols bf16 = 16 KiB = 2 global_load_lds/thread;
+// A half-tile = (BM/2) rows — 16 KiB at BM=256, 8 KiB at BM=128 (1 load).
+#define B_HALF_BYTES (128 * 64 * 2)
 
 enum Act8 { A8_NONE = 0, A8_RELU = 1, A8_GELU = 2, A8_SILU = 3 };
 
@@ -52,19 +52,24 @@ DEV_INLINE int swz16x32(int byte_off) {
   return SWZ ? (byte_off ^ (((byte_off >> 7) & 7) << 4)) : byte_off;
 }
 
-template <int ACT, bool HAS_BIAS, bool SWZ>
+template <int BMT, int ACT, bool HAS_BIAS, bool SWZ>
 __global__ __launch_bounds__(P8_THREADS, 1)
 void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
                          const __bf16* __restrict__ Bt,  // [N,K]
                          const float* __restrict__ bias,
                          __bf16* __restrict__ C, int M, int N, int K,
                          int tiles_n) {
-  // LDS: 2 buffers × (A[2 halves] + B[2 halves]) × 16 KiB = 128 KiB
+  constexpr int RH = BMT / 2;            // rows per A half
+  constexpr int A_HALF = RH * 64 * 2;    // bytes
+  constexpr int A_LOADS = A_HALF / (P8_THREADS * 16);  // insts per thread
+  constexpr int QF = BMT / 64;           // m-frags per quadrant phase
+  // LDS: 2 buffers × (2 A-halves + 2 B-halves)
+  constexpr int BUF_BYTES = 2 * A_HALF + 2 * B_HALF_BYTES;
   extern __shared__ char lds[];
-  // half slot base: buf*64KiB + (is_b*2 + half)*16KiB
   auto slot = [&](int buf, int is_b, int half) -> char* {
-    return lds + (size_t)buf * (4 * HALF_BYTES) +
-           (size_t)(is_b * 2 + half) * HALF_BYTES;
+    return lds + (size_t)buf * BUF_BYTES +
+           (is_b ? (size_t)(2 * A_HALF + half * B_HALF_BYTES)
+                 : (size_t)half * A_HALF);
   };
 
   int nwg = gridDim.x;
@@ -75,7 +80,7 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
   }
   const int bm = bid / tiles_n, bn = bid % tiles_n;
-  const int row0 = bm * P8_BM, col0 = bn * P8_BN;
+  const int row0 = bm * BMT, col0 = bn * P8_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -93,9 +98,11 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
   auto stage_half = [&](int buf, int is_b, int half, int t) {
     const __bf16* src_base = is_b ? Bt : A;
     int rdim = is_b ? N : M;
-    int grow0 = (is_b ? col0 : row0) + half * 128;
+    int grow0 = (is_b ? col0 : row0) + half * (is_b ? 128 : RH);
+    const int nloads = is_b ? 2 : A_LOADS;
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
+      if (c >= nloads) break;
       int p = p_base + c * 8192;
       int l = swz16x32<SWZ>(p);
       int lrow = l >> 7;          // logical row in the 128×64 half
@@ -125,7 +132,7 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
     }
   };
 
-  f32x4 acc[8][4] = {};
+  f32x4 acc[2 * QF][4] = {};
 
   // ---- prologue: A0(0) B0(0) B1(0) A1(0) A0(1) B0(1) B1(1) → vmcnt(6) ------
   // (issue order per tile: A0, B0, B1, A1 — j = 0,3,1,2 in stage_seq terms)
@@ -136,7 +143,11 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
   stage_seq(1, 0);  // A0(1)
   stage_seq(1, 3);  // B0(1)
   stage_seq(1, 1);  // B1(1)
-  asm volatile("s_waitcnt vmcnt(6)");  // tile 0 fully resident
+  if constexpr (BMT == 256) {          // tile 0 fully resident
+    asm volatile("s_waitcnt vmcnt(6)");
+  } else {
+    asm volatile("s_waitcnt vmcnt(5)");
+  }
   __builtin_amdgcn_sched_barrier(0);
   __builtin_amdgcn_s_barrier();
 
@@ -198,9 +209,13 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
     P8_MFMA(1, 0, b0_frag);
 
     // K-tile boundary: tile t+1's last half (A1) issued at P0 above is
-    // followed by exactly 3 halves (6 loads) ⇒ counted wait, no drain.
+    // followed by exactly 3 halves (A0+B0+B1 of t+2) ⇒ counted wait.
     if (t + 2 < NT) {
-      asm volatile("s_waitcnt vmcnt(6)");
+      if constexpr (BMT == 256) {
+        asm volatile("s_waitcnt vmcnt(6)");
+      } else {
+        asm volatile("s_waitcnt vmcnt(5)");
+      }
     } else {
       asm volatile("s_waitcnt vmcnt(0)");
     }
@@ -212,8 +227,8 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
   // ---- epilogue: bias + activation, bf16 stores (bounds-checked) -----------
   const int c_row_in_frag = (lane >> 4) * 4;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    int asel = i >> 2, q = i & 3;
+  for (int i = 0; i < 2 * QF; ++i) {
+    int asel = i / QF, q = i % QF;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       int bsel = j >> 1, r2 = j & 1;
@@ -222,7 +237,7 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
       float b = HAS_BIAS ? bias[col] : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        int row = row0 + asel * 128 + q * 32 + wm * 16 + c_row_in_frag + rr;
+        int row = row0 + asel * RH + q * 32 + wm * 16 + c_row_in_frag + rr;
         if (row >= M) continue;
         float v = acc[i][j][rr] + b;
         C[(int64_t)row * N + col] = (__bf16)apply_act8(v, ACT);
@@ -231,42 +246,71 @@ void gemm_bf16_8p_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
-template <int ACT, bool HAS_BIAS, bool SWZ>
+template <int BMT, int ACT, bool HAS_BIAS, bool SWZ>
 static void launch_one_8p(const void* A, const void* Bt, const float* bias,
                           void* C, int M, int N, int K, int tiles_n,
                           dim3 grid, dim3 block, size_t lds_bytes,
                           hipStream_t st) {
   static bool attr_done = false;
   if (!attr_done) {
-    hipFuncSetAttribute((const void*)gemm_bf16_8p_kernel<ACT, HAS_BIAS, SWZ>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        (int)lds_bytes);
+    hipFuncSetAttribute(
+        (const void*)gemm_bf16_8p_kernel<BMT, ACT, HAS_BIAS, SWZ>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);
     attr_done = true;
   }
-  gemm_bf16_8p_kernel<ACT, HAS_BIAS, SWZ><<<grid, block, lds_bytes, st>>>(
+  gemm_bf16_8p_kernel<BMT, ACT, HAS_BIAS, SWZ><<<grid, block, lds_bytes,
+                                                 st>>>(
       (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,
       tiles_n);
 }
 
-template <int ACT>
+template <int BMT, int ACT>
 static void dispatch_bias_swz(const void* A, const void* Bt,
                               const float* bias, void* C, int M, int N,
                               int K, int tiles_n, dim3 grid, dim3 block,
                               size_t lds, int swz, hipStream_t st) {
   if (bias) {
     if (swz)
-      launch_one_8p<ACT, true, true>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                     block, lds, st);
+      launch_one_8p<BMT, ACT, true, true>(A, Bt, bias, C, M, N, K, tiles_n,
+                                          grid, block, lds, st);
     else
-      launch_one_8p<ACT, true, false>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                      block, lds, st);
+      launch_one_8p<BMT, ACT, true, false>(A, Bt, bias, C, M, N, K, tiles_n,
+                                           grid, block, lds, st);
   } else {
     if (swz)
-      launch_one_8p<ACT, false, true>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                      block, lds, st);
+      launch_one_8p<BMT, ACT, false, true>(A, Bt, bias, C, M, N, K, tiles_n,
+                                           grid, block, lds, st);
     else
-      launch_one_8p<ACT, false, false>(A, Bt, bias, C, M, N, K, tiles_n,
-                                       grid, block, lds, st);
+      launch_one_8p<BMT, ACT, false, false>(A, Bt, bias, C, M, N, K,
+                                            tiles_n, grid, block, lds, st);
+  }
+}
+
+template <int BMT>
+static void dispatch_act(const void* A, const void* Bt, const float* bias,
+                         void* C, int M, int N, int K, int act, int swz,
+                         hipStream_t st) {
+  int tiles_m = (M + BMT - 1) / BMT;
+  int tiles_n = (N + P8_BN - 1) / P8_BN;
+  size_t lds = 2 * (2 * (size_t)(BMT / 2) * 64 * 2 + 2 * B_HALF_BYTES);
+  dim3 grid(tiles_m * tiles_n), block(P8_THREADS);
+  switch (act) {
+    case A8_RELU:
+      dispatch_bias_swz<BMT, A8_RELU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
+                                      block, lds, swz, st);
+      break;
+    case A8_GELU:
+      dispatch_bias_swz<BMT, A8_GELU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
+                                      block, lds, swz, st);
+      break;
+    case A8_SILU:
+      dispatch_bias_swz<BMT, A8_SILU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
+                                      block, lds, swz, st);
+      break;
+    default:
+      dispatch_bias_swz<BMT, A8_NONE>(A, Bt, bias, C, M, N, K, tiles_n, grid,
+                                      block, lds, swz, st);
+      break;
   }
 }
 
@@ -277,27 +321,11 @@ int launch_gemm_bf16_8p(const void* A, const void* Bt, const float* bias,
                         void* C, int M, int N, int K, int act, int swz,
                         hipStream_t st) {
   if (K % P8_BK != 0 || K / P8_BK < 3) return -1;
-  int tiles_m = (M + P8_BM - 1) / P8_BM;
-  int tiles_n = (N + P8_BN - 1) / P8_BN;
-  size_t lds_bytes = 2 * 4 * HALF_BYTES;  // 128 KiB
-  dim3 grid(tiles_m * tiles_n), block(P8_THREADS);
-  switch (act) {
-    case A8_RELU:
-      dispatch_bias_swz<A8_RELU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                 block, lds_bytes, swz, st);
-      break;
-    case A8_GELU:
-      dispatch_bias_swz<A8_GELU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                 block, lds_bytes, swz, st);
-      break;
-    case A8_SILU:
-      dispatch_bias_swz<A8_SILU>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                 block, lds_bytes, swz, st);
-      break;
-    default:
-      dispatch_bias_swz<A8_NONE>(A, Bt, bias, C, M, N, K, tiles_n, grid,
-                                 block, lds_bytes, swz, st);
-      break;
+  // swz>=2 forces the BM=128 tile (chip-filling at narrow N)
+  if (swz >= 2) {
+    dispatch_act<128>(A, Bt, bias, C, M, N, K, act, swz == 3 ? 1 : 0, st);
+  } else {
+    dispatch_act<256>(A, Bt, bias, C, M, N, K, act, swz, st);
   }
   return 0;
 }

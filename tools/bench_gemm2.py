@@ -51,7 +51,7 @@ def bench(fn, iters=30):
 
 print("=== refcheck ===", flush=True)
 ok = True
-for variant in (1, 2):
+for variant in (1, 2, 4, 5):
     ok &= refcheck(512, 512, 192, variant)     # NT=3 edge
     ok &= refcheck(8192, 2304, 768, variant)
     ok &= refcheck(8192, 3072, 768, variant, act=2, bias=True)
@@ -61,16 +61,16 @@ for variant in (1, 2):
 print("ALL_REFCHECK_OK" if ok else "REFCHECK_FAILED", flush=True)
 
 print("=== perf ===", flush=True)
-for (M, N, K) in [(8192, 2304, 768), (8192, 3072, 768), (4096, 4096, 4096),
-                  (8192, 8192, 8192)]:
+for (M, N, K) in [(8192, 2304, 768), (8192, 3072, 768), (8192, 768, 768),
+                  (8192, 768, 3072), (4096, 4096, 4096)]:
     A = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
     Bt = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
     B = Bt.T.contiguous()
     fl = 2.0 * M * N * K
     t0v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 0))
-    t1v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 1))
     t2v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 2))
+    t5v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 5))
     tt = bench(lambda: A @ B)
     print(f"M{M} N{N} K{K}: 128²={fl/t0v/1e12:6.1f}TF  "
-          f"8p={fl/t1v/1e12:6.1f}TF  8p+swz={fl/t2v/1e12:6.1f}TF  "
+          f"8p256+swz={fl/t2v/1e12:6.1f}TF  8p128+swz={fl/t5v/1e12:6.1f}TF  "
           f"torch={fl/tt/1e12:6.1f}TF", flush=True)
