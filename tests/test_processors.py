@@ -310,3 +310,31 @@ def test_schema_registry_codec():
     dec = c.decode(enc)
     assert dec.column("v").to_pylist() == [1.5]
     assert dec.column("tag").to_strlist() == ["x"]
+
+
+def test_inference_mlp_cpu(run):
+    from arkflow_amd.processors.inference import InferenceProcessor
+    p = InferenceProcessor({"model": "mlp_anomaly", "columns": ["a", "b"],
+                            "hidden": [32], "device": "cpu"})
+    b = MessageBatch.from_dict({"a": [0.1, 0.9], "b": [0.2, 0.8],
+                                "k": [1, 2]})
+    out = run(p.process(b))[0]
+    assert "score" in out.columns
+    assert out.num_rows == 2
+    assert out.column("k").to_pylist() == [1, 2]  # passthrough cols kept
+    # deterministic per seed
+    p2 = InferenceProcessor({"model": "mlp_anomaly", "columns": ["a", "b"],
+                             "hidden": [32], "device": "cpu"})
+    out2 = run(p2.process(b))[0]
+    assert out.column("score").to_pylist() == out2.column("score").to_pylist()
+
+
+def test_inference_bert_cpu_tiny(run):
+    from arkflow_amd.processors.inference import InferenceProcessor
+    p = InferenceProcessor({"model": "bert_base", "layers": 1,
+                            "hidden_size": 64, "heads": 2, "ff": 128,
+                            "seq_len": 8, "device": "cpu"})
+    b = MessageBatch.from_dict({"token": list(range(16))})
+    out = run(p.process(b))[0]
+    assert out.num_rows == 2  # 16 tokens / seq_len 8 → 2 sequences
+    assert "logit_0" in out.columns and "score" in out.columns
