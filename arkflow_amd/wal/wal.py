@@ -32,6 +32,10 @@ class WalAck(Ack):
         await self.inner.ack()
 
 
+MAX_PENDING_BATCHES = 256  # staged-append backpressure (reference bounds its
+                           # PUT channel at 16 segments, s3.rs flume channel)
+
+
 class Wal:
     def __init__(self, store, sync_policy: str = "group_commit",
                  group_window_ms: int = 5, periodic_interval_ms: int = 200,
@@ -75,6 +79,10 @@ class Wal:
         self._pending.append((seq, batch))
         self._ensure_flusher()
         self._pending_event.set()
+        if len(self._pending) >= MAX_PENDING_BATCHES:
+            # backpressure: the disk can't keep up — make the ingest path
+            # pay for the flush instead of growing the staged queue
+            await self.flush_pending()
         return seq
 
     def _ensure_flusher(self) -> None:
