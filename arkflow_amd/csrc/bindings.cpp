@@ -74,6 +74,9 @@ void launch_f32_to_ordered(const float*, uint32_t*, int64_t, int,
                            hipStream_t);
 void launch_i64_to_ordered(const int64_t*, uint64_t*, int64_t, int,
                            hipStream_t);
+void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
+                        const char*, const int*, const int*, const int*,
+                        double*, int64_t*, uint8_t*, int32_t*, hipStream_t);
 }
 
 namespace {
@@ -494,6 +497,41 @@ torch::Tensor radix_argsort(torch::Tensor keys, bool descending) {
   return idx_a;
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+json_decode(torch::Tensor data, torch::Tensor offsets,
+            std::vector<std::string> names, std::vector<int64_t> is_float,
+            std::vector<int64_t> slot, int64_t n_int, int64_t n_float) {
+  check_cuda(data, "data");
+  check_cuda(offsets, "offsets");
+  int64_t n = offsets.numel() - 1;
+  int nf = (int)names.size();
+  std::vector<char> packed(nf * 24, 0);
+  std::vector<int> nl(nf), isf(nf), sl(nf);
+  for (int f = 0; f < nf; ++f) {
+    TORCH_CHECK(names[f].size() < 24, "json field name too long");
+    memcpy(&packed[f * 24], names[f].data(), names[f].size());
+    nl[f] = (int)names[f].size();
+    isf[f] = (int)is_float[f];
+    sl[f] = (int)slot[f];
+  }
+  auto out_f = torch::zeros({std::max<int64_t>(n_float, 1),
+                             std::max<int64_t>(n, 1)},
+                            data.options().dtype(torch::kFloat64));
+  auto out_i = torch::zeros({std::max<int64_t>(n_int, 1),
+                             std::max<int64_t>(n, 1)},
+                            data.options().dtype(torch::kInt64));
+  auto found = torch::zeros({std::max(nf, 1), std::max<int64_t>(n, 1)},
+                            data.options().dtype(torch::kUInt8));
+  auto err = torch::zeros({1}, data.options().dtype(torch::kInt32));
+  if (n > 0)
+    launch_json_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
+                       n, nf, packed.data(), nl.data(), isf.data(), sl.data(),
+                       out_f.data_ptr<double>(), out_i.data_ptr<int64_t>(),
+                       found.data_ptr<uint8_t>(), err.data_ptr<int32_t>(),
+                       cur_stream());
+  return {out_f, out_i, found, err};
+}
+
 torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
@@ -558,5 +596,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_columns", &gather_columns);
   m.def("gemm_bf16_variant", &gemm_bf16_variant);
   m.def("bytes_hash", &bytes_hash);
+  m.def("json_decode", &json_decode);
   m.def("radix_argsort", &radix_argsort, py::arg("keys"), py::arg("descending") = false);
 }

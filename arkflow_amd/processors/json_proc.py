@@ -81,6 +81,47 @@ class JsonToArrowProcessor(Processor):
         self.device = torch.device(dev) if dev else getattr(
             resource, "device", None)
         self.keep_meta = bool(config.get("keep_meta", True))
+        # fixed-schema GPU fast path: {name: float|int|bool} decodes on-device
+        # (csrc/json_decode.hip); no schema → host pyarrow parse + inference
+        self.schema = config.get("schema")
+        if self.schema:
+            for t in self.schema.values():
+                if t not in ("float", "int", "bool"):
+                    from ..errors import ConfigError
+                    raise ConfigError(f"json schema type {t!r} not supported "
+                                      "(float|int|bool)")
+
+    def _decode_gpu(self, col) -> MessageBatch:
+        from .. import ops
+        nat = ops.require_native()
+        names, isf, slot = [], [], []
+        fcols, icols = [], []
+        for name, t in self.schema.items():
+            names.append(name)
+            if t == "float":
+                isf.append(1)
+                slot.append(len(fcols))
+                fcols.append(name)
+            else:
+                isf.append(0)
+                slot.append(len(icols))
+                icols.append(name)
+        out_f, out_i, found, err = nat.json_decode(
+            col.data, col.offsets, names, isf, slot, len(icols), len(fcols))
+        if int(err.item()) != 0:
+            raise ProcessError("json decode error (malformed document)")
+        cols = {}
+        for f, name in enumerate(names):
+            validity = found[f].to(torch.bool)
+            if self.schema[name] == "float":
+                data = out_f[fcols.index(name)].contiguous()
+            elif self.schema[name] == "bool":
+                data = out_i[icols.index(name)].to(torch.bool)
+            else:
+                data = out_i[icols.index(name)].contiguous()
+            v = None if bool(validity.all()) else validity
+            cols[name] = Column("numeric", data, validity=v)
+        return MessageBatch(cols)
 
     async def process(self, batch: MessageBatch) -> List[MessageBatch]:
         if batch.num_rows == 0:
@@ -89,8 +130,11 @@ class JsonToArrowProcessor(Processor):
         if col is None or col.kind != "binary":
             raise ProcessError(
                 f"json_to_arrow: no binary column {self.value_field!r}")
-        out = json_payloads_to_columns(col.to_pylist(), self.projection,
-                                       self.device)
+        if self.schema and col.data.is_cuda:
+            out = self._decode_gpu(col)
+        else:
+            out = json_payloads_to_columns(col.to_pylist(), self.projection,
+                                           self.device)
         if self.keep_meta:
             meta = {k: (v.to(self.device) if self.device is not None else v)
                     for k, v in batch.columns.items()

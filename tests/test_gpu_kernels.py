@@ -363,3 +363,39 @@ def test_full_engine_on_gpu(dev):
     assert e.metrics.input_messages == 8192 * 20
     assert e.metrics.output_batches == 20  # one agg result per batch
     assert e.metrics.processing_errors == 0
+
+
+def test_json_decode_gpu_matches_host(nat, dev):
+    """GPU fixed-schema JSON decode vs the host pyarrow path."""
+    import asyncio
+    import json as _json
+    import random
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+    rng = random.Random(7)
+    payloads = []
+    for i in range(20_000):
+        doc = {"a": rng.randint(-10**9, 10**9),
+               "b": round(rng.uniform(-1e6, 1e6), 4),
+               "ok": rng.random() < 0.5,
+               "skipme": {"nested": [1, 2, "x\\"y"]},
+               "s": "text, with: tricky {chars}"}
+        if i % 7 == 0:
+            del doc["b"]  # absent field → null
+        payloads.append(_json.dumps(doc).encode())
+    batch = MessageBatch.from_binary(payloads).to(dev)
+    proc = JsonToArrowProcessor({"schema": {"a": "int", "b": "float",
+                                            "ok": "bool"}}, None)
+    out = asyncio.new_event_loop().run_until_complete(proc.process(batch))[0]
+    assert out.column("a").data.is_cuda
+    a = out.column("a").to_pylist()
+    b = out.column("b").to_pylist()
+    ok = out.column("ok").to_pylist()
+    for i in (0, 1, 7, 1234, 19999):
+        doc = _json.loads(payloads[i])
+        assert a[i] == doc["a"], i
+        assert ok[i] == doc["ok"], i
+        if "b" in doc:
+            assert abs(b[i] - doc["b"]) < max(1e-6 * abs(doc["b"]), 1e-6), i
+        else:
+            assert b[i] is None, i
