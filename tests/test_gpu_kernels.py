@@ -378,7 +378,7 @@ def test_json_decode_gpu_matches_host(nat, dev):
         doc = {"a": rng.randint(-10**9, 10**9),
                "b": round(rng.uniform(-1e6, 1e6), 4),
                "ok": rng.random() < 0.5,
-               "skipme": {"nested": [1, 2, "x\\"y"]},
+               "skipme": {"nested": [1, 2, 'x"y']},
                "s": "text, with: tricky {chars}"}
         if i % 7 == 0:
             del doc["b"]  # absent field → null
