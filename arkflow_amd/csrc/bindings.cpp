@@ -497,6 +497,75 @@ torch::Tensor radix_argsort(torch::Tensor keys, bool descending) {
   return idx_a;
 }
 
+// one-call C++ execution of `SELECT * FROM flow WHERE col OP literal`:
+// fused compare+compact then one multi-column gather — a single host→device
+// round trip instead of per-op Python dispatch (the native pipeline-executor
+// path; reference's Pipeline is compiled Rust end to end).
+std::tuple<std::vector<torch::Tensor>, int64_t> fused_filter_gather(
+    std::vector<torch::Tensor> cols, int64_t filter_idx, int64_t op,
+    double scalar) {
+  TORCH_CHECK(filter_idx >= 0 && filter_idx < (int64_t)cols.size());
+  auto col = cols[filter_idx];
+  check_cuda(col, "filter col");
+  int64_t n = col.numel();
+  auto st = cur_stream();
+  auto opts32 = col.options().dtype(torch::kInt32);
+  if (n == 0) {
+    std::vector<torch::Tensor> outs;
+    for (auto& c : cols) outs.push_back(torch::empty({0}, c.options()));
+    return {outs, 0};
+  }
+  int nblocks = filter_grid(n);
+  auto counts = torch::empty({nblocks}, opts32);
+  bool is_f32 = col.scalar_type() == torch::kFloat32;
+  torch::Tensor col64;
+  if (!is_f32) {
+    col64 = col.scalar_type() == torch::kInt64 ? col : col.to(torch::kInt64);
+  }
+  if (is_f32)
+    launch_filter_count_f32(col.data_ptr<float>(), n, (int)op, (float)scalar,
+                            counts.data_ptr<int32_t>(), st);
+  else
+    launch_filter_count_i64(col64.data_ptr<int64_t>(), n, (int)op,
+                            (int64_t)scalar, counts.data_ptr<int32_t>(), st);
+  auto [offs, total_t] = exscan(counts);
+  int64_t total = total_t.item<int32_t>();  // the ONE host sync
+  auto idx = torch::empty({total}, opts32);
+  std::vector<torch::Tensor> outs;
+  outs.reserve(cols.size());
+  if (total) {
+    if (is_f32)
+      launch_filter_scatter_f32(col.data_ptr<float>(), n, (int)op,
+                                (float)scalar, offs.data_ptr<int32_t>(),
+                                idx.data_ptr<int32_t>(), st);
+    else
+      launch_filter_scatter_i64(col64.data_ptr<int64_t>(), n, (int)op,
+                                (int64_t)scalar, offs.data_ptr<int32_t>(),
+                                idx.data_ptr<int32_t>(), st);
+    std::vector<const void*> src;
+    std::vector<void*> dst;
+    std::vector<int> esz;
+    auto flush = [&]() {
+      if (!src.empty())
+        launch_gather_multi((int)src.size(), src.data(), dst.data(),
+                            esz.data(), idx.data_ptr<int32_t>(), total, st);
+      src.clear(); dst.clear(); esz.clear();
+    };
+    for (auto& c : cols) {
+      auto out = torch::empty({total}, c.options());
+      outs.push_back(out);
+      src.push_back(c.data_ptr());
+      dst.push_back(out.data_ptr());
+      esz.push_back((int)c.element_size());
+      if ((int)src.size() == 32) flush();
+    }
+    flush();
+  } else {
+    for (auto& c : cols) outs.push_back(torch::empty({0}, c.options()));
+  }
+  return {outs, total};
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 json_decode(torch::Tensor data, torch::Tensor offsets,
             std::vector<std::string> names, std::vector<int64_t> is_float,
@@ -597,5 +666,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_variant", &gemm_bf16_variant);
   m.def("bytes_hash", &bytes_hash);
   m.def("json_decode", &json_decode);
+  m.def("fused_filter_gather", &fused_filter_gather);
   m.def("radix_argsort", &radix_argsort, py::arg("keys"), py::arg("descending") = false);
 }

@@ -399,3 +399,27 @@ def test_json_decode_gpu_matches_host(nat, dev):
             assert abs(b[i] - doc["b"]) < max(1e-6 * abs(doc["b"]), 1e-6), i
         else:
             assert b[i] is None, i
+
+
+def test_fused_filter_gather_matches_slow_path(nat, dev):
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.sql import SqlProcessor
+    import asyncio
+    torch.manual_seed(50)
+    n = 200_000
+    batch = MessageBatch.from_dict({
+        "f0": torch.rand(n, device=dev),
+        "f1": torch.rand(n, device=dev),
+        "k": torch.randint(0, 100, (n,), device=dev, dtype=torch.int64),
+    })
+    p = SqlProcessor({"query": "SELECT * FROM flow WHERE f0 >= 0.5"})
+    assert p._fast_filter is not None
+    loop = asyncio.new_event_loop()
+    fast = loop.run_until_complete(p.process(batch))[0]
+    # oracle: full executor path
+    from arkflow_amd.sql.engine import SqlExecutor
+    slow = SqlExecutor("SELECT * FROM flow WHERE f0 >= 0.5").execute(
+        {"flow": batch})
+    assert fast.num_rows == slow.num_rows
+    assert torch.equal(fast.column("k").data, slow.column("k").data)
+    assert torch.equal(fast.column("f1").data, slow.column("f1").data)
