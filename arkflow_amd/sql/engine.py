@@ -114,9 +114,13 @@ class SqlExecutor:
 
         # ------------------------------------------------------------- where
         if sel.where is not None:
-            idx = self._filter_indices(sel.where, env)
-            columns = {k: c.take(idx) for k, c in columns.items()}
-            n_rows = int(idx.shape[0])
+            fused = self._try_fused_filter(sel.where, columns, env)
+            if fused is not None:
+                columns, n_rows = fused
+            else:
+                idx = self._filter_indices(sel.where, env)
+                columns = {k: c.take(idx) for k, c in columns.items()}
+                n_rows = int(idx.shape[0])
             env = Env(columns, n_rows, device)
 
         # ----------------------------------------------------------- windows
@@ -276,6 +280,46 @@ class SqlExecutor:
         return out
 
     # --------------------------------------------------------------- filters
+    def _try_fused_filter(self, pred, columns, env: Env):
+        """`col OP literal` over all-numeric device columns → ONE native call
+        (fused compare+compact+multi-gather, csrc bindings)."""
+        if not (isinstance(pred, BinaryOp) and pred.op in (
+                "<", "<=", ">", ">=", "=", "!=")):
+            return None
+        l, r = pred.left, pred.right
+        if not (isinstance(l, ColumnRef) and isinstance(r, Literal)
+                and isinstance(r.value, (int, float))):
+            return None
+        try:
+            fcol = env.lookup(l)
+        except SqlError:
+            return None
+        if fcol.kind != "numeric" or not fcol.data.is_cuda \
+                or fcol.data.dtype not in (torch.float32, torch.int64,
+                                           torch.int32):
+            return None
+        names, tensors = [], []
+        for n, c in columns.items():
+            if c.kind != "numeric" or c.validity is not None:
+                return None
+            names.append(n)
+            tensors.append(c.data)
+        fidx = None
+        for i, t in enumerate(tensors):
+            if t.data_ptr() == fcol.data.data_ptr():
+                fidx = i
+                break
+        if fidx is None:
+            return None
+        from ..ops import native_available, require_native
+        if not native_available():
+            return None
+        opi = {"<": 0, "<=": 1, ">": 2, ">=": 3, "=": 4, "!=": 5}[pred.op]
+        outs, total = require_native().fused_filter_gather(
+            tensors, fidx, opi, float(r.value))
+        return ({n: Column("numeric", t) for n, t in zip(names, outs)},
+                int(total))
+
     def _filter_indices(self, pred, env: Env) -> torch.Tensor:
         """WHERE → row indices. Fast path: `col OP numeric-literal` fuses
         compare+compact in one HIP kernel (csrc/filter.hip)."""
