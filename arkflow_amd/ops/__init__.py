@@ -8,8 +8,6 @@ GPU test can never pass on a non-native path. On CPU, torch reference
 implementations keep CI green and serve as the numerics oracle.
 """
 from __future__ import annotations
-
-import os
 from typing import Optional, Tuple
 
 import torch

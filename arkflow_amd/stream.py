@@ -80,7 +80,6 @@ class Stream:
         self._seq = 0
         self._next_seq = 0
         self._bp_event = asyncio.Event()
-        self._fatal: Optional[BaseException] = None
 
     # ------------------------------------------------------------------- run
     async def run(self, cancel: asyncio.Event) -> None:
@@ -134,8 +133,6 @@ class Stream:
                     t.cancel()
             await asyncio.gather(*tasks, in_task, return_exceptions=True)
             await self._close_all()
-        if self._fatal is not None:
-            raise self._fatal
 
     async def _close_all(self) -> None:
         # close order per reference stream/mod.rs:542-591
