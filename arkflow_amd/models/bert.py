@@ -69,7 +69,8 @@ class BertEncoder:
         self._scale = 1.0 / math.sqrt(c.head_dim)
         # hipGraph capture: the encoder is shape-static per (B, S), so the
         # whole forward (~80 kernel launches) replays as ONE graph launch
-        self.use_graph = True
+        import os as _os
+        self.use_graph = _os.environ.get("ARKFLOW_BERT_GRAPH", "1") != "0"
         self._graphs = {}  # (B, S) → (graph, static_ids, static_logits)
 
     # -------------------------------------------------------------- forward
