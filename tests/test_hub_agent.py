@@ -176,3 +176,41 @@ def test_agent_idempotent_replay(run):
         assert "a1" in agent._executed
 
     run(main())
+
+
+def test_rollout_config_applied_via_agent(run):
+    """Rollout step → apply_config attempt → agent executes →
+    engine reconfigured (reference rollout + agent.rs execute_command)."""
+    async def main():
+        hub = Hub(lease_ttl=5.0)
+        app = create_hub_app(hub)
+        transport = httpx.ASGITransport(app=app)
+        eng = _engine()
+        agent = Agent(eng, "http://hub", node_id="n1", transport=transport,
+                      heartbeat_interval=0.05, report_interval=0.5,
+                      poll_interval=0.02)
+        cancel = asyncio.Event()
+        agent_task = asyncio.ensure_future(agent.run(cancel))
+        await asyncio.sleep(0.15)
+
+        new_cfg = {"streams": [{
+            "id": "rolled",
+            "input": {"type": "generate", "batch_size": 1, "interval": "50ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "output": {"type": "drop"},
+        }]}
+        rid = await hub.create_rollout(new_cfg, ["n1"])
+        await hub.step_rollout(rid)  # creates the apply_config attempt
+        for _ in range(100):
+            if "rolled" in eng.runtime.entries:
+                break
+            await asyncio.sleep(0.05)
+        assert "rolled" in eng.runtime.entries
+        assert eng.runtime.get("rolled").state.value == "running"
+        r = await hub.step_rollout(rid)
+        assert r["state"] == "succeeded"
+        cancel.set()
+        await asyncio.gather(agent_task, return_exceptions=True)
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)
