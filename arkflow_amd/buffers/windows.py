@@ -51,23 +51,38 @@ class TumblingWindowBuffer(BaseWindowBuffer):
 
 
 class SlidingWindowBuffer(BaseWindowBuffer):
-    """Count-based overlapping window over batches."""
+    """Count-based overlapping window over batches. On device, batches land
+    in the shared ring once and every overlapping emit is slice views —
+    the reference re-concats the whole window per slide
+    (sliding_window.rs process_slide)."""
 
     def __init__(self, config: dict, resource=None):
         super().__init__(config, resource)
         self.window_size = int(config.get("window_size", 10))
         self.slide_size = int(config.get("slide_size", 5))
         self.interval = _parse_duration(config.get("interval", "0s"))
-        self.window: Deque[Tuple[MessageBatch, Ack]] = deque()
+        self.window: Deque[Tuple[object, Ack]] = deque()  # batch | ring range
         self._new_since_emit = 0
         self._deadline = None
 
     async def write(self, batch: MessageBatch, ack: Ack) -> None:
-        self.window.append((batch, ack))
+        item: object = batch
+        if self.use_ring:
+            from .ring import DeviceRingBuffer
+            ring = self.rings.get("window")
+            if ring is None:
+                ring = self.rings["window"] = DeviceRingBuffer()
+            rng = ring.append(batch)
+            if rng is not None:
+                item = rng
+        self.window.append((item, ack))
         self._new_since_emit += 1
         if self.interval > 0 and self._deadline is None:
             self._deadline = time.monotonic() + self.interval
         self._notify.set()
+
+    def _window_batch(self) -> MessageBatch:
+        return self._materialize("window", list(self.window))
 
     def try_emit(self, draining: bool = False):
         timer_fired = (self._deadline is not None
@@ -80,21 +95,28 @@ class SlidingWindowBuffer(BaseWindowBuffer):
         self._new_since_emit = 0
         self._deadline = (time.monotonic() + self.interval
                           if self.interval > 0 else None)
-        # batches leaving the window are the only ones acked now
+        # batches leaving the window are the only ones acked now; their ring
+        # rows free on ack (the emitted views no longer cover them)
         leaving = []
+        left_items = []
         while len(self.window) > self.window_size:
-            _, a = self.window.popleft()
+            it, a = self.window.popleft()
             leaving.append(a)
-        combined = concat_batches([b for b, _ in self.window])
-        return combined, (VecAck(leaving) if leaving else NoopAck())
+            left_items.append((it, a))
+        combined = self._window_batch()
+        if leaving:
+            return combined, self._release_ack(VecAck(leaving),
+                                               {"window": left_items})
+        return combined, NoopAck()
 
     def drain_remaining(self):
         if not self.window:
             return None
-        acks = [a for _, a in self.window]
-        combined = concat_batches([b for b, _ in self.window])
+        items = list(self.window)
+        acks = [a for _, a in items]
+        combined = self._window_batch()
         self.window.clear()
-        return combined, VecAck(acks)
+        return combined, self._release_ack(VecAck(acks), {"window": items})
 
     def next_deadline(self) -> Optional[float]:
         if self._deadline is None:

@@ -423,3 +423,38 @@ def test_fused_filter_gather_matches_slow_path(nat, dev):
     assert fast.num_rows == slow.num_rows
     assert torch.equal(fast.column("k").data, slow.column("k").data)
     assert torch.equal(fast.column("f1").data, slow.column("f1").data)
+
+
+def test_engine_window_ring_on_gpu(dev):
+    """Windowed engine on device: tumbling window emits zero-copy ring views
+    feeding a GROUP BY — full async engine."""
+    import asyncio
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "win-gpu",
+            "device": "cuda:0",
+            "input": {"type": "generate", "batch_size": 4096,
+                      "count": 4096 * 12, "interval": "2ms",
+                      "fields": {"v": {"dtype": "float32"},
+                                 "k": {"dtype": "int64", "low": 0,
+                                       "high": 32}}},
+            "buffer": {"type": "tumbling_window", "interval": "30ms"},
+            "pipeline": {"thread_num": 1, "processors": [
+                {"type": "sql",
+                 "query": "SELECT k, count(*) AS c, sum(v) AS s FROM flow "
+                          "GROUP BY k"},
+            ]},
+            "output": {"type": "memory"},
+        }]
+    })
+    eng = af.Engine(cfg)
+    asyncio.new_event_loop().run_until_complete(
+        asyncio.wait_for(eng.run_with_cancellation(), 120))
+    e = eng.runtime.entries["win-gpu"]
+    assert e.state.value == "stopped"
+    assert e.metrics.input_messages == 4096 * 12
+    assert e.metrics.processing_errors == 0
+    assert e.metrics.output_batches >= 1
