@@ -1,0 +1,99 @@
+"""Differential SQL testing: our columnar executor vs SQLite on randomized
+tables and a grid of query shapes (the oracle role DataFusion's own test
+suite plays for the reference)."""
+import math
+import random
+import sqlite3
+import zlib
+
+import pytest
+
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.sql.engine import SqlExecutor
+
+QUERIES = [
+    "SELECT a, b FROM flow WHERE a >= 50",
+    "SELECT a, b, c FROM flow WHERE b < 0.5 AND a % 3 = 0",
+    "SELECT a + b AS s, a * 2 AS d FROM flow WHERE NOT a < 10",
+    "SELECT k, count(*) AS c, sum(a) AS s, min(b) AS lo, max(b) AS hi "
+    "FROM flow GROUP BY k",
+    "SELECT k, avg(a) AS m FROM flow GROUP BY k HAVING count(*) > 3",
+    "SELECT a FROM flow WHERE a BETWEEN 20 AND 60 ORDER BY a DESC LIMIT 7",
+    "SELECT DISTINCT k FROM flow",
+    "SELECT count(*) AS c FROM flow WHERE b >= 0.25 AND b < 0.75",
+    "SELECT k, count(DISTINCT a) AS u FROM flow GROUP BY k",
+    "SELECT a, CASE WHEN a >= 50 THEN 1 ELSE 0 END AS hi FROM flow "
+    "WHERE c != 0 ORDER BY a, b LIMIT 20",
+    "SELECT a FROM flow WHERE a IN (1, 2, 3, 5, 8, 13, 21, 34, 55, 89)",
+    "SELECT sum(a * b) AS dot FROM flow",
+    "SELECT a FROM flow ORDER BY a LIMIT 5 OFFSET 3",
+    "SELECT f.a, d.label FROM flow f JOIN dims d ON f.k = d.k ORDER BY f.a",
+]
+
+
+def _random_table(rng, n):
+    return {
+        "a": [rng.randrange(100) for _ in range(n)],
+        "b": [round(rng.random(), 6) for _ in range(n)],
+        "c": [rng.randrange(-5, 6) for _ in range(n)],
+        "k": [rng.randrange(8) for _ in range(n)],
+    }
+
+
+def _sqlite_exec(data, dims, sql):
+    conn = sqlite3.connect(":memory:")
+    conn.execute("CREATE TABLE flow (a INTEGER, b REAL, c INTEGER, "
+                 "k INTEGER)")
+    conn.executemany("INSERT INTO flow VALUES (?,?,?,?)",
+                     list(zip(data["a"], data["b"], data["c"], data["k"])))
+    conn.execute("CREATE TABLE dims (k INTEGER, label TEXT)")
+    conn.executemany("INSERT INTO dims VALUES (?,?)",
+                     list(zip(dims["k"], dims["label"])))
+    # sqlite aliases: our engine registers `flow f` via FROM alias — sqlite
+    # accepts the same SQL directly.
+    rows = conn.execute(sql).fetchall()
+    conn.close()
+    return rows
+
+
+def _normalize(rows):
+    out = []
+    for r in rows:
+        norm = []
+        for v in r:
+            if isinstance(v, (bytes, bytearray)):
+                v = v.decode()
+            if isinstance(v, bool):
+                v = int(v)
+            if isinstance(v, float):
+                v = None if math.isnan(v) else round(v, 4)
+            norm.append(v)
+        out.append(tuple(norm))
+    return out
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+@pytest.mark.parametrize("sql", QUERIES)
+def test_differential_vs_sqlite(seed, sql):
+    rng = random.Random(seed * 1000 + zlib.crc32(sql.encode()) % 997)
+    n = rng.choice([0, 1, 17, 200])
+    if n == 0 and ("JOIN" in sql or "GROUP BY" in sql):
+        n = 17  # sqlite group-by-on-empty differs from DataFusion semantics
+    data = _random_table(rng, n)
+    dims = {"k": list(range(8)), "label": [f"L{i}" for i in range(8)]}
+    flow = MessageBatch.from_dict(data) if n else MessageBatch.from_dict(
+        {k: [] for k in data})
+    dims_b = MessageBatch.from_dict(dims)
+    ours_b = SqlExecutor(sql).execute({"flow": flow, "dims": dims_b})
+    ours = _normalize([tuple(r.values()) for r in ours_b.to_rows()])
+    theirs = _normalize(_sqlite_exec(data, dims, sql))
+    has_order = "ORDER BY" in sql
+    if not has_order:
+        ours = sorted(ours)
+        theirs = sorted(theirs)
+    if "count(*)" in sql and "GROUP BY" not in sql and n == 0:
+        # global aggregate over empty input: both return one row
+        assert ours == theirs
+        return
+    assert ours == theirs, f"{sql}\nseed={seed} n={n}\n{ours[:5]} vs " \
+                           f"{theirs[:5]}"
