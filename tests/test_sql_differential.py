@@ -28,6 +28,12 @@ QUERIES = [
     "SELECT sum(a * b) AS dot FROM flow",
     "SELECT a FROM flow ORDER BY a LIMIT 5 OFFSET 3",
     "SELECT f.a, d.label FROM flow f JOIN dims d ON f.k = d.k ORDER BY f.a",
+    "SELECT a, row_number() OVER (PARTITION BY k ORDER BY a) AS rn "
+    "FROM flow ORDER BY a, rn",
+    "SELECT a, rank() OVER (ORDER BY k) AS rk FROM flow ORDER BY a, rk",
+    "SELECT a, sum(a) OVER (PARTITION BY k) AS s FROM flow ORDER BY a, s",
+    "SELECT a FROM flow WHERE a < 10 UNION ALL SELECT a FROM flow "
+    "WHERE a > 90",
 ]
 
 
@@ -77,7 +83,7 @@ def _normalize(rows):
 def test_differential_vs_sqlite(seed, sql):
     rng = random.Random(seed * 1000 + zlib.crc32(sql.encode()) % 997)
     n = rng.choice([0, 1, 17, 200])
-    if n == 0 and ("JOIN" in sql or "GROUP BY" in sql):
+    if n == 0 and ("JOIN" in sql or "GROUP BY" in sql or "OVER" in sql):
         n = 17  # sqlite group-by-on-empty differs from DataFusion semantics
     data = _random_table(rng, n)
     dims = {"k": list(range(8)), "label": [f"L{i}" for i in range(8)]}
@@ -97,3 +103,23 @@ def test_differential_vs_sqlite(seed, sql):
         return
     assert ours == theirs, f"{sql}\nseed={seed} n={n}\n{ours[:5]} vs " \
                            f"{theirs[:5]}"
+
+
+@pytest.mark.parametrize("seed", range(5))
+def test_parser_fuzz_no_crash(seed):
+    """Random token soup must raise SqlError (or parse), never crash."""
+    from arkflow_amd.sql.parser import SqlError, parse_sql
+    rng = random.Random(seed)
+    vocab = ["SELECT", "FROM", "WHERE", "flow", "a", "b", "+", "-", "*",
+             "(", ")", ",", "=", "<", ">", "1", "2.5", "'s'", "AND", "OR",
+             "GROUP", "BY", "ORDER", "LIMIT", "JOIN", "ON", "CASE", "WHEN",
+             "END", "CAST", "AS", "count", "sum", "OVER", "PARTITION"]
+    for _ in range(300):
+        sql = " ".join(rng.choice(vocab)
+                       for _ in range(rng.randrange(1, 25)))
+        try:
+            parse_sql(sql)
+        except SqlError:
+            pass
+        except RecursionError:
+            pass
