@@ -75,6 +75,7 @@ class Stream:
         self.wal = wal
         self.temporaries = temporaries or {}
         self.metrics = metrics or RuntimeMetrics()
+        self.device = _resolve_device(config.device)
         self.thread_num = config.pipeline.resolved_thread_num()
         # sequencing / backpressure state
         self._seq = 0
@@ -110,7 +111,10 @@ class Stream:
         if self.wal is not None:
             async for seq, batch in self.wal.read_after_cursor():
                 from .wal.wal import WalAck
-                await self._forward(input_q, batch, WalAck(self.wal, seq, NoopAck()))
+                if self.device.type == "cuda":
+                    batch = batch.to(self.device)  # H2D: restore residency
+                await self._forward(input_q, batch,
+                                    WalAck(self.wal, seq, NoopAck()))
 
         in_task = asyncio.ensure_future(
             self._do_input(input_q, cancel))

@@ -458,3 +458,44 @@ def test_engine_window_ring_on_gpu(dev):
     assert e.metrics.input_messages == 4096 * 12
     assert e.metrics.processing_errors == 0
     assert e.metrics.output_batches >= 1
+
+
+def test_wal_replay_restores_device(dev, tmp_path):
+    """Crash-replayed batches must come back GPU-resident when the stream is
+    pinned to a device (WAL serializes via D2H; replay is H2D)."""
+    import asyncio
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.config import (DurabilityConfig, PipelineConfig,
+                                    StreamConfig)
+    from arkflow_amd.pipeline import Pipeline
+    from arkflow_amd.spi import Processor
+    from arkflow_amd.stream import Stream
+    from arkflow_amd.wal.wal import Wal
+    from tests.test_stream_engine import CountingOutput, StubInput
+
+    async def main():
+        cfg_d = DurabilityConfig(enabled=True, path=str(tmp_path),
+                                 sync_policy="per_entry")
+        wal0 = Wal.open(cfg_d, "g")
+        await wal0.append(MessageBatch.from_dict(
+            {"v": torch.rand(64, device=dev)}))
+        await wal0.close()
+
+        seen_devices = []
+
+        class Probe(Processor):
+            async def process(self, batch):
+                seen_devices.append(batch.device.type)
+                return [batch]
+
+        sc = StreamConfig(id="g", input={"type": "memory"},
+                          output={"type": "drop"}, device="cuda:0",
+                          pipeline=PipelineConfig(thread_num=1))
+        out = CountingOutput()
+        s = Stream(sc, StubInput([]), Pipeline([Probe()]), out,
+                   wal=Wal.open(cfg_d, "g"))
+        await asyncio.wait_for(s.run(asyncio.Event()), 60)
+        assert out.rows == 64
+        assert seen_devices == ["cuda"]
+
+    asyncio.new_event_loop().run_until_complete(main())
