@@ -265,3 +265,16 @@ def test_event_ring_bounded():
     assert len(evs) == 4
     assert evs[0].kind == "k6"
     assert es.list(after_seq=evs[-1].seq - 1)[0].kind == "k9"
+
+
+def test_all_example_configs_validate():
+    import glob
+    import os
+    from arkflow_amd.config import EngineConfig
+    examples = glob.glob(os.path.join(
+        os.path.dirname(__file__), "..", "examples", "*.yaml"))
+    assert len(examples) >= 12
+    for path in examples:
+        cfg = EngineConfig.from_file(path)
+        errs = cfg.validate()
+        assert not errs, f"{path}: {errs}"
