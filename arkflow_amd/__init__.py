@@ -73,6 +73,7 @@ def init() -> None:
         "arkflow_amd.inputs.websocket",
         "arkflow_amd.inputs.modbus",
         "arkflow_amd.codecs.debezium",
+        "arkflow_amd.codecs.schema_registry",
         "arkflow_amd.outputs.file",
         "arkflow_amd.outputs.http",
         "arkflow_amd.outputs.influxdb",

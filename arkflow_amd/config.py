@@ -150,6 +150,15 @@ class EngineConfig:
                     errors.append(f"{where}: {kind} missing 'type'")
                 elif t not in _registry(kind).builders:
                     errors.append(f"{where}: unknown {kind} type {t!r}")
+                codec_spec = spec.get("codec") if isinstance(spec, dict) \
+                    else None
+                if codec_spec is not None:
+                    ct = codec_spec.get("type") \
+                        if isinstance(codec_spec, dict) else None
+                    if not ct:
+                        errors.append(f"{where}: codec missing 'type'")
+                    elif ct not in _registry("codec").builders:
+                        errors.append(f"{where}: unknown codec type {ct!r}")
             for p in s.pipeline.processors:
                 t = p.get("type") if isinstance(p, dict) else None
                 if not t:

@@ -278,3 +278,16 @@ def test_all_example_configs_validate():
         cfg = EngineConfig.from_file(path)
         errs = cfg.validate()
         assert not errs, f"{path}: {errs}"
+
+
+def test_codec_registry_complete_and_validated():
+    from arkflow_amd.registry import registry
+    assert set(registry("codec").names()) == {
+        "json", "protobuf", "debezium_json", "schema_registry"}
+    from arkflow_amd.config import EngineConfig
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "c", "input": {"type": "kafka", "topic": "t",
+                             "codec": {"type": "nope"}},
+        "output": {"type": "drop"}}]})
+    errs = cfg.validate()
+    assert any("unknown codec" in e for e in errs)
