@@ -215,6 +215,8 @@ def main():
     input_rows = args.steps * args.batch_size * world
     value = input_rows / elapsed
     p50_ms = statistics.median(times) * 1000.0
+    p99_ms = (sorted(times)[max(0, int(len(times) * 0.99) - 1)] * 1000.0
+              if times else 0.0)
 
     if args.rank == 0:
         print(json.dumps({
@@ -231,6 +233,7 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "p50_ms": p50_ms,
+            "p99_ms": p99_ms,
             "config": {
                 "model": {
                     "mlp": "generate→sql(filter)→mlp_anomaly[16→256→256→1]",
