@@ -19,16 +19,29 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor,
                 bias: Optional[torch.Tensor] = None,
                 act: str = "none") -> torch.Tensor:
     """act(x @ weight.T + bias). x:[M,K] bf16, weight:[N,K] (torch Linear
-    layout — consumed directly as the GEMM's B^T operand)."""
+    layout — consumed directly as the GEMM's B^T operand).
+
+    FUSED linears (activation epilogues) run our gfx950 MFMA kernels;
+    PLAIN linears (no activation) go to hipBLASLt via torch.addmm — the
+    division of labor the design brief prescribes (hand-written kernels for
+    fused hot ops, the vendor library for plain GEMMs).
+    Env overrides: ARKFLOW_GEMM_VARIANT forces a specific native kernel;
+    ARKFLOW_PLAIN_GEMM=native keeps plain linears on our kernel.
+    """
+    import os
     a = _ACTS[act]
     if x.is_cuda:
-        import os
         nat = require_native()
         x2 = x.reshape(-1, x.shape[-1]).contiguous()
         force = os.environ.get("ARKFLOW_GEMM_VARIANT")
         if force is not None:
             out = nat.gemm_bf16_variant(x2, weight.contiguous(), bias, a,
                                         int(force))
+        elif a == ACT_NONE and \
+                os.environ.get("ARKFLOW_PLAIN_GEMM", "blaslt") != "native":
+            out = torch.nn.functional.linear(
+                x2, weight,
+                bias.to(torch.bfloat16) if bias is not None else None)
         else:
             out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
         return out.reshape(*x.shape[:-1], weight.shape[0])
