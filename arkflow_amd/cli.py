@@ -61,6 +61,14 @@ def main(argv: Optional[List[str]] = None) -> int:
 
     sub.add_parser("schema", help="print the engine config JSON schema")
 
+    hub_p = sub.add_parser("hub", help="serve the fleet hub")
+    hub_p.add_argument("--address", default="127.0.0.1:9000")
+    hub_p.add_argument("--store", default=":memory:",
+                       help="SQLite path for durable hub state")
+    hub_p.add_argument("--lease-ttl", type=float, default=15.0)
+    hub_p.add_argument("--operator-token", action="append", default=[],
+                       help="token:role (role in admin|operator|viewer)")
+
     args = parser.parse_args(argv)
 
     if args.command == "components":
@@ -74,6 +82,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     if args.command == "schema":
         print(json.dumps(_bcs(), indent=2))
         return 0
+    if args.command == "hub":
+        return _run_hub(args)
 
     if not args.config:
         parser.error("--config is required to run the engine")
@@ -123,3 +133,35 @@ def main(argv: Optional[List[str]] = None) -> int:
 
 if __name__ == "__main__":
     sys.exit(main())
+
+
+def _run_hub(args) -> int:
+    """Serve the fleet hub (reference arkflow-server hub mode)."""
+    import uvicorn
+
+    from .server.hub import Hub, create_hub_app, hub_background
+    from .server.storage import HubStore
+
+    tokens = {}
+    for spec in args.operator_token:
+        token, _, role = spec.partition(":")
+        tokens[token] = role or "operator"
+    hub = Hub(HubStore(args.store), lease_ttl=args.lease_ttl,
+              operator_tokens=tokens)
+    app = create_hub_app(hub)
+    host, _, port = args.address.partition(":")
+
+    async def serve():
+        cancel = asyncio.Event()
+        bg = asyncio.ensure_future(hub_background(hub, cancel))
+        config = uvicorn.Config(app, host=host or "127.0.0.1",
+                                port=int(port or 9000), log_level="info")
+        server = uvicorn.Server(config)
+        try:
+            await server.serve()
+        finally:
+            cancel.set()
+            await bg
+
+    asyncio.run(serve())
+    return 0

@@ -73,6 +73,7 @@ class ServerConfig:
     node_id: Optional[str] = None
     node_token: Optional[str] = None
     lease_ttl_secs: float = 15.0
+    config_store: Optional[str] = None  # persist config versions for rollback
 
 
 @dataclass
@@ -129,6 +130,7 @@ class EngineConfig:
             node_id=srv_raw.get("node_id"),
             node_token=srv_raw.get("node_token"),
             lease_ttl_secs=float(srv_raw.get("lease_ttl_secs", 15.0)),
+            config_store=srv_raw.get("config_store"),
         )
         return EngineConfig(streams, logging_cfg, server_cfg)
 

@@ -20,7 +20,8 @@ class Engine:
         self.ready = False
         self.running = False
         from .control_plane import ControlPlane
-        self.control_plane = ControlPlane(self)
+        self.control_plane = ControlPlane(
+            self, version_store_path=config.server.config_store)
 
     async def run_with_cancellation(self, cancel: Optional[asyncio.Event] = None,
                                     install_signal_handlers: bool = False) -> None:

@@ -168,3 +168,31 @@ def test_console_served(run):
             assert "arkflow_amd console" in r.text
 
     run(main())
+
+
+def test_config_versions_persist(tmp_path, run):
+    async def main():
+        cfg = EngineConfig.from_dict({
+            "streams": [{"id": "s1",
+                         "input": {"type": "generate", "batch_size": 1,
+                                   "interval": "50ms",
+                                   "fields": {"v": {"dtype": "float32"}}},
+                         "output": {"type": "drop"}}],
+            "server": {"config_store": str(tmp_path / "versions.json")},
+        })
+        eng = af.Engine(cfg)
+        for sc in cfg.streams:
+            eng.runtime.register(sc)
+        r = await eng.control_plane.apply_configuration({
+            "streams": [{"id": "s2",
+                         "input": {"type": "generate", "batch_size": 1,
+                                   "interval": "50ms",
+                                   "fields": {"v": {"dtype": "float32"}}},
+                         "output": {"type": "drop"}}]}, note="v1")
+        assert r["applied"]
+        await eng.runtime.stop_all()
+        # a fresh engine sees the persisted version store
+        eng2 = af.Engine(cfg)
+        assert len(eng2.control_plane.versions.list()) == 1
+
+    run(main(), timeout=30)
