@@ -156,6 +156,29 @@ class ControlPlane:
         self.engine.config = new_cfg
         return {"applied": True, "version": version, "errors": []}
 
+    def diff_config(self, raw: dict) -> dict:
+        """Structured diff of a candidate config against the running one
+        (reference node API config diff route, server/src/lib.rs router)."""
+        from dataclasses import asdict
+        current = {s.id: asdict(s) for s in self.engine.config.streams}
+        candidate = {s.get("id", f"?{i}"): s
+                     for i, s in enumerate(raw.get("streams", []))}
+        added = sorted(set(candidate) - set(current))
+        removed = sorted(set(current) - set(candidate))
+        changed = []
+        for sid in sorted(set(current) & set(candidate)):
+            cur = redact_secrets(current[sid])
+            cand = redact_secrets(candidate[sid])
+            fields = sorted(k for k in set(cur) | set(cand)
+                            if cur.get(k) != cand.get(k)
+                            and (cur.get(k) or cand.get(k)))
+            if fields:
+                changed.append({"stream_id": sid, "fields": fields})
+        return {"added": added, "removed": removed, "changed": changed,
+                "unchanged": sorted(
+                    sid for sid in set(current) & set(candidate)
+                    if not any(c["stream_id"] == sid for c in changed))}
+
     async def rollback(self, version: int) -> dict:
         entry = self.versions.get(version)
         if entry is None:

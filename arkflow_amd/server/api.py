@@ -137,6 +137,10 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
     async def validate_configuration(body: dict):
         return cp.validate_config(body)
 
+    @app.post(f"{prefix}/configuration/diff", dependencies=[Depends(auth)])
+    async def diff_configuration(body: dict):
+        return cp.diff_config(body)
+
     @app.post(f"{prefix}/configuration/apply", dependencies=[Depends(auth)])
     async def apply_configuration(body: dict):
         return await cp.apply_configuration(body)

@@ -196,3 +196,22 @@ def test_config_versions_persist(tmp_path, run):
         assert len(eng2.control_plane.versions.list()) == 1
 
     run(main(), timeout=30)
+
+
+def test_config_diff(run):
+    async def main():
+        eng = _engine()
+        d = eng.control_plane.diff_config({"streams": [
+            {"id": "s1", "input": {"type": "generate", "batch_size": 9,
+                                   "interval": "50ms",
+                                   "fields": {"v": {"dtype": "float32"}}},
+             "output": {"type": "drop"}},
+            {"id": "brand_new", "input": {"type": "generate"},
+             "output": {"type": "drop"}},
+        ]})
+        assert d["added"] == ["brand_new"]
+        assert d["removed"] == []
+        assert d["changed"] and d["changed"][0]["stream_id"] == "s1"
+        assert "input" in d["changed"][0]["fields"]
+
+    run(main())
