@@ -17,7 +17,7 @@ import time
 from collections import deque
 from typing import Deque, Optional, Tuple
 
-from ..batch import MessageBatch, concat_batches
+from ..batch import MessageBatch
 from ..inputs.generate import _parse_duration
 from ..registry import register
 from ..spi import Ack, NoopAck, VecAck

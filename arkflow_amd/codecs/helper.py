@@ -3,7 +3,7 @@
 enters the stream, preserving __meta_* columns when row counts align."""
 from __future__ import annotations
 
-from typing import Optional
+
 
 from ..batch import DEFAULT_BINARY_VALUE_FIELD, MessageBatch
 from ..registry import build_component

@@ -7,7 +7,7 @@ from typing import List, Sequence
 import torch
 
 from ..batch import Column, MessageBatch
-from ..processors.proto_wire import ProtoSchema, decode_message, encode_message
+from ..processors.proto_wire import decode_message, encode_message
 from ..processors.protobuf_proc import _load_schema
 from ..registry import register
 from ..spi import Codec

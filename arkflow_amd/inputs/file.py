@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import glob as globmod
 import os
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import torch

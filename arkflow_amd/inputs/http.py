@@ -11,7 +11,7 @@ import time
 from typing import Optional, Tuple
 
 from ..batch import MessageBatch
-from ..errors import ConfigError, EOFError_
+from ..errors import EOFError_
 from ..registry import register
 from ..spi import Ack, Input, NoopAck
 

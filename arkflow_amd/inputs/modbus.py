@@ -8,7 +8,7 @@ import time
 from typing import Dict, List, Tuple
 
 from ..batch import Column, MessageBatch
-from ..errors import ConfigError, ConnectionError_, EOFError_
+from ..errors import ConnectionError_, EOFError_
 from ..registry import register
 from ..spi import Ack, Input, NoopAck
 

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import re
 import struct
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 from ..errors import ConfigError
 

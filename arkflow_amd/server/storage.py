@@ -13,7 +13,7 @@ import json
 import sqlite3
 import time
 import uuid
-from typing import Any, Dict, List, Optional
+from typing import List, Optional
 
 _SCHEMA = """
 CREATE TABLE IF NOT EXISTS nodes (

@@ -8,7 +8,7 @@ DISTINCT → ORDER BY → LIMIT. All row-level work is tensor ops routed through
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 

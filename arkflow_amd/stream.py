@@ -17,7 +17,7 @@ from __future__ import annotations
 import asyncio
 import logging
 import time
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from .batch import MessageBatch
 from .config import StreamConfig

@@ -6,10 +6,10 @@ client library is importable (offline env: connect() raises).
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
-from ..batch import MessageBatch, concat_batches
-from ..errors import ConfigError, ConnectionError_
+from ..batch import MessageBatch
+from ..errors import ConnectionError_
 from ..registry import register
 from ..spi import Temporary
 

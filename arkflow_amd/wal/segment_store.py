@@ -14,9 +14,8 @@ import json
 import os
 import threading
 import time
-import zlib
 from concurrent.futures import ThreadPoolExecutor
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, Tuple
 
 from ..registry import register
 from .store import decode_frames, encode_frame

@@ -14,7 +14,7 @@ from __future__ import annotations
 import os
 import struct
 import zlib
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, Tuple
 
 import numpy as np
 import torch
