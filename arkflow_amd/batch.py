@@ -86,7 +86,10 @@ class Column:
         if isinstance(values, torch.Tensor):
             t = values
         elif isinstance(values, np.ndarray):
-            t = torch.from_numpy(np.ascontiguousarray(values))
+            arr = np.ascontiguousarray(values)
+            if not arr.flags.writeable:  # e.g. zero-copy pyarrow buffers
+                arr = arr.copy()
+            t = torch.from_numpy(arr)
         else:
             vals = list(values)
             if vals and isinstance(vals[0], bool):

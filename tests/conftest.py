@@ -24,7 +24,9 @@ def pytest_collection_modifyitems(config, items):
 def run():
     """Run a coroutine to completion on a fresh event loop."""
     def _run(coro, timeout=30.0):
-        return asyncio.get_event_loop_policy().new_event_loop().run_until_complete(
-            asyncio.wait_for(coro, timeout)
-        )
+        loop = asyncio.get_event_loop_policy().new_event_loop()
+        try:
+            return loop.run_until_complete(asyncio.wait_for(coro, timeout))
+        finally:
+            loop.close()
     return _run
