@@ -18,7 +18,7 @@ from concurrent.futures import ThreadPoolExecutor
 from typing import Iterator, List, Tuple
 
 from ..registry import register
-from .store import decode_frames, encode_frame
+from .store import _nwal, decode_frames, encode_frame
 
 
 # segment sealing presets (reference wal/config.rs:31 SegmentStrategy)
@@ -102,8 +102,11 @@ class SegmentWalStore:
         name = f"seg-{self._seg_counter:08d}.wal"
         self._seg_counter += 1
         first, last = entries[0][0], entries[-1][0]
-        data = b"".join(
-            encode_frame(s, p, self.compress) for s, p in entries)
+        if _nwal is not None and not self.compress:
+            data = _nwal.encode_frames(entries)
+        else:
+            data = b"".join(
+                encode_frame(s, p, self.compress) for s, p in entries)
         fut = self._pool.submit(self._put_segment, name, data)
         self._pending_puts.append(fut)
         self._manifest["segments"].append(

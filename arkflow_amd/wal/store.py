@@ -22,6 +22,11 @@ import torch
 from ..batch import Column, MessageBatch
 from ..registry import register
 
+try:  # native frame codec (csrc/wal_codec.cpp) — Python framing is the
+    from .. import _wal_native as _nwal  # fallback when the ext isn't built
+except ImportError:  # pragma: no cover
+    _nwal = None
+
 _MAGIC = b"AWAL"
 _DTYPES = {
     "float32": torch.float32, "float64": torch.float64,
@@ -140,7 +145,12 @@ def encode_frame(seq: int, payload: bytes, compress: bool = False) -> bytes:
 
 def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:
     """Yields (seq, payload); stops at a torn/corrupt tail
-    (reference segment.rs:89 CRC truncation)."""
+    (reference segment.rs:89 CRC truncation). Uses the native codec's
+    one-pass parser when built."""
+    if _nwal is not None:
+        for seq, tag, body in _nwal.decode_frames(buf):
+            yield seq, (zlib.decompress(body) if tag == 0x5A else bytes(body))
+        return
     pos = 0
     n = len(buf)
     while pos + 12 <= n:
@@ -204,8 +214,11 @@ class LocalWalStore:
     # log ----------------------------------------------------------------
     def append_batch(self, entries: List[Tuple[int, bytes]],
                      sync: bool = True) -> None:
-        for seq, payload in entries:
-            self._f.write(encode_frame(seq, payload, self.compress))
+        if _nwal is not None and not self.compress:
+            self._f.write(_nwal.encode_frames(entries))
+        else:
+            for seq, payload in entries:
+                self._f.write(encode_frame(seq, payload, self.compress))
         self._f.flush()
         if sync and self.fsync:
             os.fsync(self._f.fileno())

@@ -7,7 +7,7 @@ to GPU boxes (no JIT cache dependence).
 """
 import os
 
-from setuptools import setup
+from setuptools import Extension, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -32,7 +32,13 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
-        )
+        ),
+        # CPU-only native WAL frame codec (plain CPython extension, no torch)
+        Extension(
+            name="arkflow_amd._wal_native",
+            sources=[os.path.join(CSRC, "wal_codec.cpp")],
+            extra_compile_args=["-O3", "-std=c++17"],
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

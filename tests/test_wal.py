@@ -179,3 +179,23 @@ def test_segment_store_reclaim(tmp_path):
     assert len(segs2) < len(segs)
     assert [s for s, _ in st.read_after(st.cursor)] == [3, 4]
     st.close()
+
+
+def test_native_frame_codec_interop():
+    """Native (csrc/wal_codec.cpp) and Python framing must be bit-identical
+    and cross-readable, including torn-tail truncation."""
+    nwal = pytest.importorskip("arkflow_amd._wal_native")
+    import zlib
+    from arkflow_amd.wal.store import decode_frames, encode_frame
+    entries = [(1, b"alpha"), (2, b""), (3, bytes(range(256)) * 100)]
+    blob = nwal.encode_frames(entries)
+    assert blob == b"".join(encode_frame(s, p) for s, p in entries)
+    assert list(decode_frames(blob)) == entries
+    # torn tail: drop 3 bytes → last frame discarded by both decoders
+    assert [s for s, _ in decode_frames(blob[:-3])] == [1, 2]
+    # corrupt a body byte of frame 2 → truncates from there
+    bad = bytearray(blob)
+    bad[len(blob) - 50] ^= 0xFF
+    assert [s for s, _ in decode_frames(bytes(bad))] == [1, 2]
+    # crc32 matches zlib for arbitrary inits
+    assert nwal.crc32(b"data", 1234) == zlib.crc32(b"data", 1234)
