@@ -54,9 +54,12 @@ void launch_bias_act_bf16(const void*, const float*, void*, int64_t, int, int,
                           hipStream_t);
 int launch_attention_bf16(const void*, const void*, const void*, void*, int,
                           int, int, float, hipStream_t);
+void launch_proto_copy_bytes(const uint8_t*, const int64_t*, const int64_t*,
+                             int64_t, uint8_t*, hipStream_t);
 void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
                          const int*, const int*, const int*, const int*,
-                         int64_t*, double*, int32_t*, hipStream_t);
+                         int64_t*, double*, int64_t*, int32_t*, int32_t*,
+                         hipStream_t);
 void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
@@ -76,7 +79,11 @@ void launch_i64_to_ordered(const int64_t*, uint64_t*, int64_t, int,
                            hipStream_t);
 void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
                         const char*, const int*, const int*, const int*,
-                        double*, int64_t*, uint8_t*, int32_t*, hipStream_t);
+                        double*, int64_t*, int64_t*, int32_t*, uint8_t*,
+                        int32_t*, hipStream_t);
+void launch_json_copy_strings(const uint8_t*, const int64_t*, const int64_t*,
+                              const uint8_t*, int64_t, int64_t, uint8_t*,
+                              hipStream_t);
 }
 
 namespace {
@@ -566,21 +573,23 @@ std::tuple<std::vector<torch::Tensor>, int64_t> fused_filter_gather(
   return {outs, total};
 }
 
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
+           std::vector<std::tuple<torch::Tensor, torch::Tensor>>>
 json_decode(torch::Tensor data, torch::Tensor offsets,
-            std::vector<std::string> names, std::vector<int64_t> is_float,
-            std::vector<int64_t> slot, int64_t n_int, int64_t n_float) {
+            std::vector<std::string> names, std::vector<int64_t> kind,
+            std::vector<int64_t> slot, int64_t n_int, int64_t n_float,
+            int64_t n_str) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
   int64_t n = offsets.numel() - 1;
   int nf = (int)names.size();
   std::vector<char> packed(nf * 24, 0);
-  std::vector<int> nl(nf), isf(nf), sl(nf);
+  std::vector<int> nl(nf), kd(nf), sl(nf);
   for (int f = 0; f < nf; ++f) {
     TORCH_CHECK(names[f].size() < 24, "json field name too long");
     memcpy(&packed[f * 24], names[f].data(), names[f].size());
     nl[f] = (int)names[f].size();
-    isf[f] = (int)is_float[f];
+    kd[f] = (int)kind[f];
     sl[f] = (int)slot[f];
   }
   auto out_f = torch::zeros({std::max<int64_t>(n_float, 1),
@@ -589,16 +598,50 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   auto out_i = torch::zeros({std::max<int64_t>(n_int, 1),
                              std::max<int64_t>(n, 1)},
                             data.options().dtype(torch::kInt64));
+  auto str_start = torch::zeros({std::max<int64_t>(n_str, 1),
+                                 std::max<int64_t>(n, 1)},
+                                data.options().dtype(torch::kInt64));
+  auto str_ulen = torch::zeros({std::max<int64_t>(n_str, 1),
+                                std::max<int64_t>(n, 1)},
+                               data.options().dtype(torch::kInt32));
   auto found = torch::zeros({std::max(nf, 1), std::max<int64_t>(n, 1)},
                             data.options().dtype(torch::kUInt8));
   auto err = torch::zeros({1}, data.options().dtype(torch::kInt32));
   if (n > 0)
     launch_json_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
-                       n, nf, packed.data(), nl.data(), isf.data(), sl.data(),
+                       n, nf, packed.data(), nl.data(), kd.data(), sl.data(),
                        out_f.data_ptr<double>(), out_i.data_ptr<int64_t>(),
+                       str_start.data_ptr<int64_t>(),
+                       str_ulen.data_ptr<int32_t>(),
                        found.data_ptr<uint8_t>(), err.data_ptr<int32_t>(),
                        cur_stream());
-  return {out_f, out_i, found, err};
+  // string copy-out: per field, offsets = exclusive cumsum of unescaped
+  // lengths; one host sync for ALL totals at once, then one copy kernel per
+  // string field (ulen of absent docs is 0 → empty strings, validity=found)
+  std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
+  if (n_str > 0 && n > 0) {
+    auto offs64 = torch::zeros({n_str, n + 1},
+                               data.options().dtype(torch::kInt64));
+    offs64.slice(1, 1, n + 1).copy_(str_ulen.to(torch::kInt64).cumsum(1));
+    auto totals = offs64.select(1, n).to(torch::kCPU);  // one device sync
+    auto* tot = totals.data_ptr<int64_t>();
+    for (int64_t s = 0; s < n_str; ++s) {
+      auto so = offs64[s].contiguous();
+      auto out = torch::empty({std::max<int64_t>(tot[s], 1)},
+                              data.options().dtype(torch::kUInt8));
+      // found row for this slot: find the field index with this slot
+      int frow = 0;
+      for (int f = 0; f < nf; ++f)
+        if (kd[f] == 2 && sl[f] == (int)s) frow = f;
+      if (tot[s] > 0)
+        launch_json_copy_strings(
+            data.data_ptr<uint8_t>(), str_start[s].data_ptr<int64_t>(),
+            so.data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(), n,
+            data.numel(), out.data_ptr<uint8_t>(), cur_stream());
+      strings.emplace_back(out.slice(0, 0, tot[s]), so);
+    }
+  }
+  return {out_f, out_i, found, err, strings};
 }
 
 torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
@@ -612,11 +655,13 @@ torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
   return out;
 }
 
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> proto_decode(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
+           std::vector<std::tuple<torch::Tensor, torch::Tensor>>>
+proto_decode(
     torch::Tensor data, torch::Tensor offsets,
     std::vector<int64_t> fno, std::vector<int64_t> kind,
     std::vector<int64_t> is_float, std::vector<int64_t> slot,
-    int64_t n_int, int64_t n_float) {
+    int64_t n_int, int64_t n_float, int64_t n_str) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
   TORCH_CHECK(data.scalar_type() == torch::kUInt8);
@@ -632,13 +677,41 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> proto_decode(
                             data.options().dtype(torch::kInt64));
   auto out_f = torch::zeros({std::max<int64_t>(n_float, 1), std::max<int64_t>(n, 1)},
                             data.options().dtype(torch::kFloat64));
+  auto str_start = torch::zeros({std::max<int64_t>(n_str, 1),
+                                 std::max<int64_t>(n, 1)},
+                                data.options().dtype(torch::kInt64));
+  auto str_len = torch::zeros({std::max<int64_t>(n_str, 1),
+                               std::max<int64_t>(n, 1)},
+                              data.options().dtype(torch::kInt32));
   auto err = torch::zeros({1}, data.options().dtype(torch::kInt32));
   if (n > 0)
     launch_proto_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
                         n, nf, f.data(), k.data(), isf.data(), sl.data(),
                         out_i.data_ptr<int64_t>(), out_f.data_ptr<double>(),
-                        err.data_ptr<int32_t>(), cur_stream());
-  return {out_i, out_f, err};
+                        str_start.data_ptr<int64_t>(),
+                        str_len.data_ptr<int32_t>(), err.data_ptr<int32_t>(),
+                        cur_stream());
+  // string/bytes copy-out (proto3 missing field → len 0 → empty value)
+  std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
+  if (n_str > 0 && n > 0) {
+    auto offs64 = torch::zeros({n_str, n + 1},
+                               data.options().dtype(torch::kInt64));
+    offs64.slice(1, 1, n + 1).copy_(str_len.to(torch::kInt64).cumsum(1));
+    auto totals = offs64.select(1, n).to(torch::kCPU);  // one device sync
+    auto* tot = totals.data_ptr<int64_t>();
+    for (int64_t s = 0; s < n_str; ++s) {
+      auto so = offs64[s].contiguous();
+      auto out = torch::empty({std::max<int64_t>(tot[s], 1)},
+                              data.options().dtype(torch::kUInt8));
+      if (tot[s] > 0)
+        launch_proto_copy_bytes(data.data_ptr<uint8_t>(),
+                                str_start[s].data_ptr<int64_t>(),
+                                so.data_ptr<int64_t>(), n,
+                                out.data_ptr<uint8_t>(), cur_stream());
+      strings.emplace_back(out.slice(0, 0, tot[s]), so);
+    }
+  }
+  return {out_i, out_f, err, strings};
 }
 
 }  // namespace

@@ -13,7 +13,7 @@ struct JsonSpec {
   int nf;
   char names[JSON_MAX_FIELDS][JSON_MAX_NAME];
   int name_len[JSON_MAX_FIELDS];
-  int is_float[JSON_MAX_FIELDS];  // 1 → out_f64, 0 → out_i64 (ints + bools)
+  int kind[JSON_MAX_FIELDS];  // 0 → out_i64 (ints+bools), 1 → out_f64, 2 → str
   int slot[JSON_MAX_FIELDS];
 };
 
@@ -110,11 +110,111 @@ DEV_INLINE int64_t parse_number(const uint8_t* d, int64_t p, int64_t end,
   return p;
 }
 
+DEV_INLINE uint32_t hex4(const uint8_t* d, int64_t p) {
+  uint32_t v = 0;
+  for (int i = 0; i < 4; ++i) {
+    uint8_t c = d[p + i];
+    uint32_t x = (c >= '0' && c <= '9')   ? c - '0'
+                 : (c >= 'a' && c <= 'f') ? c - 'a' + 10
+                 : (c >= 'A' && c <= 'F') ? c - 'A' + 10
+                                          : 0;
+    v = (v << 4) | x;
+  }
+  return v;
+}
+
+// Walk a JSON string body (p = first char after the opening quote), computing
+// the UNESCAPED byte length (simple escapes → 1 byte, \uXXXX → UTF-8 length,
+// surrogate pairs → 4). Returns the position of the closing quote. The copy
+// kernel below must mirror this walk byte-for-byte.
+DEV_INLINE int64_t scan_string(const uint8_t* __restrict__ d, int64_t p,
+                               int64_t end, int32_t* ulen) {
+  int32_t u = 0;
+  while (p < end && d[p] != '"') {
+    if (d[p] == '\\' && p + 1 < end) {
+      uint8_t e = d[p + 1];
+      if (e == 'u' && p + 5 < end) {
+        uint32_t cp = hex4(d, p + 2);
+        p += 6;
+        if (cp >= 0xD800 && cp <= 0xDBFF && p + 5 < end && d[p] == '\\' &&
+            d[p + 1] == 'u') {
+          uint32_t lo = hex4(d, p + 2);
+          if (lo >= 0xDC00 && lo <= 0xDFFF) {
+            p += 6;
+            u += 4;
+            continue;
+          }
+        }
+        u += cp < 0x80 ? 1 : (cp < 0x800 ? 2 : 3);
+        continue;
+      }
+      p += 2;
+      u += 1;
+      continue;
+    }
+    ++p;
+    ++u;
+  }
+  *ulen = u;
+  return p;
+}
+
+DEV_INLINE int64_t unescape_copy(const uint8_t* __restrict__ d, int64_t p,
+                                 int64_t end, uint8_t* __restrict__ out,
+                                 int64_t w, int64_t wend) {
+  while (w < wend && p < end && d[p] != '"') {
+    if (d[p] == '\\' && p + 1 < end) {
+      uint8_t e = d[p + 1];
+      if (e == 'u' && p + 5 < end) {
+        uint32_t cp = hex4(d, p + 2);
+        p += 6;
+        if (cp >= 0xD800 && cp <= 0xDBFF && p + 5 < end && d[p] == '\\' &&
+            d[p + 1] == 'u') {
+          uint32_t lo = hex4(d, p + 2);
+          if (lo >= 0xDC00 && lo <= 0xDFFF) {
+            p += 6;
+            cp = 0x10000 + (((cp - 0xD800) << 10) | (lo - 0xDC00));
+            out[w++] = 0xF0 | (cp >> 18);
+            out[w++] = 0x80 | ((cp >> 12) & 0x3F);
+            out[w++] = 0x80 | ((cp >> 6) & 0x3F);
+            out[w++] = 0x80 | (cp & 0x3F);
+            continue;
+          }
+        }
+        if (cp < 0x80) {
+          out[w++] = (uint8_t)cp;
+        } else if (cp < 0x800) {
+          out[w++] = 0xC0 | (cp >> 6);
+          out[w++] = 0x80 | (cp & 0x3F);
+        } else {
+          out[w++] = 0xE0 | (cp >> 12);
+          out[w++] = 0x80 | ((cp >> 6) & 0x3F);
+          out[w++] = 0x80 | (cp & 0x3F);
+        }
+        continue;
+      }
+      uint8_t v = e == 'n'   ? '\n'
+                  : e == 't' ? '\t'
+                  : e == 'r' ? '\r'
+                  : e == 'b' ? '\b'
+                  : e == 'f' ? '\f'
+                             : e;  // \" \\ \/ and unknown escapes → literal
+      out[w++] = v;
+      p += 2;
+      continue;
+    }
+    out[w++] = d[p++];
+  }
+  return w;
+}
+
 __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
                                    const int64_t* __restrict__ offsets,
                                    int64_t n_docs, JsonSpec spec,
                                    double* __restrict__ out_f64,  // [nd][n]
                                    int64_t* __restrict__ out_i64,
+                                   int64_t* __restrict__ str_start,  // [ns][n]
+                                   int32_t* __restrict__ str_ulen,   // [ns][n]
                                    uint8_t* __restrict__ found,  // [nf][n]
                                    int32_t* __restrict__ err) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -159,6 +259,20 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
         continue;
       }
       uint8_t c = p < end ? data[p] : 0;
+      if (spec.kind[fi] == 2) {  // string field
+        if (c != '"') {  // non-string value under a string schema → absent
+          p = skip_value(data, p, end);
+          continue;
+        }
+        int32_t ulen = 0;
+        int64_t s0 = p + 1;
+        p = scan_string(data, s0, end, &ulen);
+        if (p < end) ++p;  // closing quote
+        str_start[(int64_t)spec.slot[fi] * n_docs + i] = s0;
+        str_ulen[(int64_t)spec.slot[fi] * n_docs + i] = ulen;
+        found[(int64_t)fi * n_docs + i] = 1;
+        continue;
+      }
       double v = 0.0;
       bool is_int = true;
       if (c == 't') { v = 1.0; p += 4; }
@@ -166,7 +280,7 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
       else if (c == 'n') { p += 4; continue; }  // null → absent
       else if (c == '"') { p = skip_value(data, p, end); continue; }
       else p = parse_number(data, p, end, &v, &is_int);
-      if (spec.is_float[fi])
+      if (spec.kind[fi] == 1)
         out_f64[(int64_t)spec.slot[fi] * n_docs + i] = v;
       else
         out_i64[(int64_t)spec.slot[fi] * n_docs + i] = (int64_t)v;
@@ -175,17 +289,34 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
   }
 }
 
+// copy-out pass for ONE string field: thread per doc, unescaping into the
+// contiguous output at out_offs[i] (out_offs = exclusive cumsum of ulen)
+__global__ void json_copy_strings_kernel(const uint8_t* __restrict__ data,
+                                         const int64_t* __restrict__ start,
+                                         const int64_t* __restrict__ out_offs,
+                                         const uint8_t* __restrict__ found,
+                                         int64_t n_docs, int64_t data_len,
+                                         uint8_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n_docs; i += stride) {
+    if (!found[i]) continue;
+    unescape_copy(data, start[i], data_len, out, out_offs[i], out_offs[i + 1]);
+  }
+}
+
 extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
                                    int64_t n_docs, int nf, const char* names,
-                                   const int* name_len, const int* is_float,
+                                   const int* name_len, const int* kind,
                                    const int* slot, double* out_f64,
-                                   int64_t* out_i64, uint8_t* found,
+                                   int64_t* out_i64, int64_t* str_start,
+                                   int32_t* str_ulen, uint8_t* found,
                                    int32_t* err, hipStream_t st) {
   JsonSpec spec{};
   spec.nf = nf > JSON_MAX_FIELDS ? JSON_MAX_FIELDS : nf;
   for (int f = 0; f < spec.nf; ++f) {
     spec.name_len[f] = name_len[f];
-    spec.is_float[f] = is_float[f];
+    spec.kind[f] = kind[f];
     spec.slot[f] = slot[f];
     for (int c = 0; c < name_len[f] && c < JSON_MAX_NAME; ++c)
       spec.names[f][c] = names[f * JSON_MAX_NAME + c];
@@ -194,5 +325,19 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
   if (grid > 2048) grid = 2048;
   if (grid < 1) return;
   json_decode_kernel<<<grid, 256, 0, st>>>(data, offsets, n_docs, spec,
-                                           out_f64, out_i64, found, err);
+                                           out_f64, out_i64, str_start,
+                                           str_ulen, found, err);
+}
+
+extern "C" void launch_json_copy_strings(const uint8_t* data,
+                                         const int64_t* start,
+                                         const int64_t* out_offs,
+                                         const uint8_t* found, int64_t n_docs,
+                                         int64_t data_len, uint8_t* out,
+                                         hipStream_t st) {
+  int grid = (int)((n_docs + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  json_copy_strings_kernel<<<grid, 256, 0, st>>>(data, start, out_offs, found,
+                                                 n_docs, data_len, out);
 }

@@ -10,7 +10,7 @@
 #define PROTO_MAX_FIELDS 16
 
 // kind: 0 varint, 1 zigzag, 2 f64, 3 f32, 4 u64(fixed), 5 i64(sfixed),
-//       6 u32(fixed), 7 i32(sfixed), 8 bool
+//       6 u32(fixed), 7 i32(sfixed), 8 bool, 9 string/bytes (span record)
 struct ProtoSpec {
   int nf;
   int fno[PROTO_MAX_FIELDS];
@@ -24,6 +24,8 @@ __global__ void proto_decode_kernel(const uint8_t* __restrict__ data,
                                     int64_t n_msgs, ProtoSpec spec,
                                     int64_t* __restrict__ out_i64,  // [ni][n]
                                     double* __restrict__ out_f64,   // [nd][n]
+                                    int64_t* __restrict__ str_start,  // [ns][n]
+                                    int32_t* __restrict__ str_len,
                                     int32_t* __restrict__ err_flags) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -70,7 +72,8 @@ __global__ void proto_decode_kernel(const uint8_t* __restrict__ data,
         for (int b = 0; b < 4; ++b) raw |= (uint64_t)data[pos + b] << (8 * b);
         pos += 4;
       } else if (wt == 2) {
-        // length-delimited: skip (strings handled host-side)
+        // length-delimited: string/bytes fields record their span for the
+        // copy-out pass; other wt==2 payloads are skipped
         uint64_t ln = 0;
         int s2 = 0;
         while (pos < end) {
@@ -78,6 +81,11 @@ __global__ void proto_decode_kernel(const uint8_t* __restrict__ data,
           ln |= (uint64_t)(b & 0x7F) << s2;
           if (!(b & 0x80)) break;
           s2 += 7;
+        }
+        if (pos + (int64_t)ln > end) { err_flags[0] = 1; return; }
+        if (fi >= 0 && spec.kind[fi] == 9) {
+          str_start[(int64_t)spec.slot[fi] * n_msgs + i] = pos;
+          str_len[(int64_t)spec.slot[fi] * n_msgs + i] = (int32_t)ln;
         }
         pos += (int64_t)ln;
         continue;
@@ -112,12 +120,39 @@ __global__ void proto_decode_kernel(const uint8_t* __restrict__ data,
   }
 }
 
+// copy-out pass for ONE string/bytes field: raw span copy, no transform
+__global__ void proto_copy_bytes_kernel(const uint8_t* __restrict__ data,
+                                        const int64_t* __restrict__ start,
+                                        const int64_t* __restrict__ out_offs,
+                                        int64_t n_msgs,
+                                        uint8_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n_msgs; i += stride) {
+    int64_t w = out_offs[i];
+    const int64_t wend = out_offs[i + 1];
+    int64_t p = start[i];
+    for (; w < wend; ++w, ++p) out[w] = data[p];
+  }
+}
+
 extern "C" {
+
+void launch_proto_copy_bytes(const uint8_t* data, const int64_t* start,
+                             const int64_t* out_offs, int64_t n_msgs,
+                             uint8_t* out, hipStream_t st) {
+  int grid = (int)((n_msgs + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  proto_copy_bytes_kernel<<<grid, 256, 0, st>>>(data, start, out_offs, n_msgs,
+                                                out);
+}
 
 void launch_proto_decode(const uint8_t* data, const int64_t* offsets,
                          int64_t n_msgs, int nf, const int* fno,
                          const int* kind, const int* is_float,
                          const int* slot, int64_t* out_i64, double* out_f64,
+                         int64_t* str_start, int32_t* str_len,
                          int32_t* err_flags, hipStream_t st) {
   ProtoSpec spec{};
   spec.nf = nf > PROTO_MAX_FIELDS ? PROTO_MAX_FIELDS : nf;
@@ -131,7 +166,8 @@ void launch_proto_decode(const uint8_t* data, const int64_t* offsets,
   if (grid > 2048) grid = 2048;
   if (grid < 1) return;
   proto_decode_kernel<<<grid, 256, 0, st>>>(data, offsets, n_msgs, spec,
-                                            out_i64, out_f64, err_flags);
+                                            out_i64, out_f64, str_start,
+                                            str_len, err_flags);
 }
 
 }  // extern "C"

@@ -81,45 +81,57 @@ class JsonToArrowProcessor(Processor):
         self.device = torch.device(dev) if dev else getattr(
             resource, "device", None)
         self.keep_meta = bool(config.get("keep_meta", True))
-        # fixed-schema GPU fast path: {name: float|int|bool} decodes on-device
-        # (csrc/json_decode.hip); no schema → host pyarrow parse + inference
+        # fixed-schema GPU fast path: {name: float|int|bool|str} decodes
+        # on-device (csrc/json_decode.hip: scalar extraction + two-pass string
+        # copy-out with escape/\uXXXX handling); no schema → host pyarrow
+        # parse + inference
         self.schema = config.get("schema")
         if self.schema:
             for t in self.schema.values():
-                if t not in ("float", "int", "bool"):
+                if t not in ("float", "int", "bool", "str", "string"):
                     from ..errors import ConfigError
                     raise ConfigError(f"json schema type {t!r} not supported "
-                                      "(float|int|bool)")
+                                      "(float|int|bool|str)")
 
     def _decode_gpu(self, col) -> MessageBatch:
         from .. import ops
         nat = ops.require_native()
-        names, isf, slot = [], [], []
-        fcols, icols = [], []
+        names, kinds, slot = [], [], []
+        fcols, icols, scols = [], [], []
         for name, t in self.schema.items():
             names.append(name)
             if t == "float":
-                isf.append(1)
+                kinds.append(1)
                 slot.append(len(fcols))
                 fcols.append(name)
+            elif t in ("str", "string"):
+                kinds.append(2)
+                slot.append(len(scols))
+                scols.append(name)
             else:
-                isf.append(0)
+                kinds.append(0)
                 slot.append(len(icols))
                 icols.append(name)
-        out_f, out_i, found, err = nat.json_decode(
-            col.data, col.offsets, names, isf, slot, len(icols), len(fcols))
+        out_f, out_i, found, err, strings = nat.json_decode(
+            col.data, col.offsets, names, kinds, slot,
+            len(icols), len(fcols), len(scols))
         if int(err.item()) != 0:
             raise ProcessError("json decode error (malformed document)")
         cols = {}
         for f, name in enumerate(names):
             validity = found[f].to(torch.bool)
+            v = None if bool(validity.all()) else validity
             if self.schema[name] == "float":
                 data = out_f[fcols.index(name)].contiguous()
             elif self.schema[name] == "bool":
                 data = out_i[icols.index(name)].to(torch.bool)
+            elif self.schema[name] in ("str", "string"):
+                sdata, soffs = strings[scols.index(name)]
+                cols[name] = Column("binary", sdata.contiguous(),
+                                    soffs.contiguous(), validity=v)
+                continue
             else:
                 data = out_i[icols.index(name)].contiguous()
-            v = None if bool(validity.all()) else validity
             cols[name] = Column("numeric", data, validity=v)
         return MessageBatch(cols)
 

@@ -3,9 +3,9 @@
 Mirrors reference crates/arkflow-plugin/src/processor/protobuf.rs +
 component/protobuf.rs: dynamic decode/encode of scalar proto3 fields from a
 .proto source (no nested/repeated/map/oneof — header :14-25). The decode hot
-path for numeric-only schemas is a GPU kernel parsing the device-resident
-binary column directly (csrc/proto_decode.hip); schemas with string/bytes
-fields use the host codec.
+path is a GPU kernel parsing the device-resident binary column directly
+(csrc/proto_decode.hip) — numeric fields into typed columns, string/bytes
+fields via a two-pass span-record + copy-out into binary columns.
 """
 from __future__ import annotations
 
@@ -17,7 +17,7 @@ from ..batch import Column, DEFAULT_BINARY_VALUE_FIELD, MessageBatch
 from ..errors import ConfigError, ProcessError
 from ..registry import register
 from ..spi import Processor
-from .proto_wire import ProtoSchema, SCALARS, decode_message, encode_message
+from .proto_wire import ProtoSchema, decode_message, encode_message
 
 _KIND_ENUM = {
     "varint": 0, "zigzag": 1, "f64": 2, "f32": 3,
@@ -52,7 +52,7 @@ class ProtobufToArrowProcessor(Processor):
         if col is None or col.kind != "binary":
             raise ProcessError(
                 f"protobuf_to_arrow: no binary column {self.value_field!r}")
-        if col.data.is_cuda and self.schema.numeric_only():
+        if col.data.is_cuda:
             out = self._decode_gpu(col)
         else:
             out = self._decode_cpu(col)
@@ -64,32 +64,32 @@ class ProtobufToArrowProcessor(Processor):
         from .. import ops
         nat = ops.require_native()
         fno, kind, isf, slot = [], [], [], []
-        int_fields, float_fields = [], []
+        int_fields, float_fields, str_fields = [], [], []
         for no in sorted(self.schema.fields):
             name, t = self.schema.fields[no]
-            wire_kind, _ = SCALARS[t]
             fno.append(no)
-            if t in _FLOAT_TYPES or t in ("fixed32", "fixed64", "uint32",
-                                          "uint64"):
-                # unsigned 64-bit may exceed int64 → f64 slot for fixed/uint?
-                # keep uint32/uint64 in int64 (values < 2^63 in practice);
-                # only true floats go to the f64 output.
-                pass
-            if t in _FLOAT_TYPES:
+            if t in ("string", "bytes"):
+                kind.append(9)
+                isf.append(0)
+                slot.append(len(str_fields))
+                str_fields.append(name)
+            elif t in _FLOAT_TYPES:
                 kind.append(_KIND_ENUM["f64" if t == "double" else "f32"])
                 isf.append(1)
                 slot.append(len(float_fields))
                 float_fields.append(name)
             else:
+                # uint32/uint64/fixed stay in int64 (values < 2^63 in
+                # practice); only true floats go to the f64 output
                 k = {"sint32": 1, "sint64": 1, "fixed64": 4, "sfixed64": 5,
                      "fixed32": 6, "sfixed32": 7}.get(t, 0)
                 kind.append(k)
                 isf.append(0)
                 slot.append(len(int_fields))
                 int_fields.append(name)
-        out_i, out_f, err = nat.proto_decode(
+        out_i, out_f, err, strings = nat.proto_decode(
             col.data, col.offsets, fno, kind, isf, slot,
-            len(int_fields), len(float_fields))
+            len(int_fields), len(float_fields), len(str_fields))
         if int(err.item()) != 0:
             raise ProcessError("protobuf decode error (malformed message)")
         cols = {}
@@ -101,6 +101,10 @@ class ProtobufToArrowProcessor(Processor):
             cols[name] = Column("numeric", data.contiguous())
         for i, name in enumerate(float_fields):
             cols[name] = Column("numeric", out_f[i].contiguous())
+        for i, name in enumerate(str_fields):
+            sdata, soffs = strings[i]
+            cols[name] = Column("binary", sdata.contiguous(),
+                                soffs.contiguous())
         # preserve declared field order
         ordered = {self.schema.fields[no][0]: cols[self.schema.fields[no][0]]
                    for no in sorted(self.schema.fields)

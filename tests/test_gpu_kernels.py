@@ -261,6 +261,48 @@ def test_proto_decode_gpu_matches_cpu(nat, dev):
     assert out.column("g").to_pylist()[123] == rows[123]["g"]
 
 
+def test_proto_decode_gpu_strings(nat, dev):
+    """GPU string/bytes span decode vs the host wire codec, incl. unicode,
+    empty and missing fields (proto3 default → empty string)."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.proto_wire import ProtoSchema, encode_message
+    from arkflow_amd.processors.protobuf_proc import ProtobufToArrowProcessor
+    import asyncio
+    proto = """
+    message U {
+      string name = 1; int64 v = 2; bytes blob = 3; string note = 4;
+    }
+    """
+    schema = ProtoSchema.parse(proto)
+    rows = []
+    payloads = []
+    for i in range(8_000):
+        r = {"name": f"user-{i}-caf\u00e9" * (i % 4), "v": i,
+             "blob": bytes([i % 256]) * (i % 30),
+             "note": "" if i % 5 == 0 else f"note {i} \u4e2d\u6587"}
+        rows.append(r)
+        payloads.append(encode_message(r, schema))
+    batch = MessageBatch.from_binary(payloads).to(dev)
+    proc = ProtobufToArrowProcessor({"proto": proto}, None)
+    out = asyncio.new_event_loop().run_until_complete(proc.process(batch))[0]
+    assert out.column("name").data.is_cuda
+
+    def s(v):
+        return v.decode() if isinstance(v, (bytes, bytearray)) else (v or "")
+
+    names = out.column("name").to_pylist()
+    notes = out.column("note").to_pylist()
+    blobs = out.column("blob").to_pylist()
+    vs = out.column("v").to_pylist()
+    for i in (0, 1, 5, 99, 4321, 7999):
+        assert s(names[i]) == rows[i]["name"], i
+        assert s(notes[i]) == rows[i]["note"], i
+        raw = blobs[i] if isinstance(blobs[i], (bytes, bytearray)) \
+            else (blobs[i] or "").encode("latin1")
+        assert bytes(raw) == rows[i]["blob"], i
+        assert vs[i] == rows[i]["v"]
+
+
 def test_hash_group_large(nat, dev):
     """Regression: tables > 512K entries must be fully initialized
     (fill kernels are grid-stride; grid is capped at 2048 blocks)."""
@@ -399,6 +441,52 @@ def test_json_decode_gpu_matches_host(nat, dev):
             assert abs(b[i] - doc["b"]) < max(1e-6 * abs(doc["b"]), 1e-6), i
         else:
             assert b[i] is None, i
+
+
+def test_json_decode_gpu_strings(nat, dev):
+    """GPU string-field extraction: escapes, \\uXXXX, surrogate pairs,
+    unicode, absent fields — values must match Python's json.loads."""
+    import asyncio
+    import json as _json
+    import random
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+    rng = random.Random(11)
+    corpus = ["plain", "", "with \"quotes\" and \\backslash\\",
+              "tabs\tnewlines\n\r", "unicode: caf\u00e9 \u4e2d\u6587",
+              "emoji \U0001F600 pair", "slash\/ok", "ctrl \b\f end",
+              "x" * 500]
+    payloads = []
+    docs = []
+    for i in range(5_000):
+        doc = {"name": rng.choice(corpus) + str(i % 100),
+               "v": i,
+               "tag": rng.choice(corpus)}
+        if i % 11 == 0:
+            del doc["tag"]
+        if i % 13 == 0:
+            doc["name"] = 12345  # wrong type under string schema → absent
+        docs.append(doc)
+        # ensure_ascii True → \uXXXX escapes (incl. surrogate pairs) on wire
+        payloads.append(_json.dumps(doc, ensure_ascii=(i % 2 == 0)).encode())
+    batch = MessageBatch.from_binary(payloads).to(dev)
+    proc = JsonToArrowProcessor({"schema": {"name": "str", "v": "int",
+                                            "tag": "str"}}, None)
+    out = asyncio.new_event_loop().run_until_complete(proc.process(batch))[0]
+    assert out.column("name").data.is_cuda
+    names = out.column("name").to_pylist()
+    tags = out.column("tag").to_pylist()
+    vs = out.column("v").to_pylist()
+    for i, doc in enumerate(docs):
+        expect_name = doc["name"] if isinstance(doc["name"], str) else None
+        got = names[i].decode() if isinstance(names[i], (bytes, bytearray)) \
+            else names[i]
+        assert got == expect_name, f"row {i}: {got!r} != {expect_name!r}"
+        expect_tag = doc.get("tag")
+        got_t = tags[i].decode() if isinstance(tags[i], (bytes, bytearray)) \
+            else tags[i]
+        assert got_t == expect_tag, f"row {i}"
+        assert vs[i] == doc["v"]
 
 
 def test_fused_filter_gather_matches_slow_path(nat, dev):
