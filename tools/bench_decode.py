@@ -1,0 +1,59 @@
+"""Throughput microbench for the GPU JSON / protobuf decoders (scalar and
+string schemas). Run on a GPU box:  python tools/bench_decode.py
+"""
+import asyncio
+import json
+import time
+
+import torch
+
+from arkflow_amd.batch import MessageBatch
+from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+from arkflow_amd.processors.proto_wire import ProtoSchema, encode_message
+from arkflow_amd.processors.protobuf_proc import ProtobufToArrowProcessor
+
+N = 262_144
+REPS = 20
+dev = torch.device("cuda:0")
+loop = asyncio.new_event_loop()
+
+
+def run(name, proc, batch):
+    for _ in range(3):
+        loop.run_until_complete(proc.process(batch))
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(REPS):
+        loop.run_until_complete(proc.process(batch))
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / REPS
+    nbytes = int(batch.column("__value__").data.numel())
+    print(f"{name:34s} {N/dt/1e9:7.2f} B docs/s   "
+          f"{nbytes/dt/1e9:7.1f} GB/s   {dt*1e3:6.2f} ms/batch")
+
+
+payloads = [json.dumps({
+    "a": i, "b": i * 0.5, "ok": i % 2 == 0,
+    "name": f"user-{i}-café", "tag": f"segment {i % 32} 中"
+}).encode() for i in range(N)]
+batch = MessageBatch.from_binary(payloads).to(dev)
+run("json scalars (a,b,ok)",
+    JsonToArrowProcessor({"schema": {"a": "int", "b": "float",
+                                     "ok": "bool"}}, None), batch)
+run("json scalars + 2 strings",
+    JsonToArrowProcessor({"schema": {"a": "int", "b": "float", "ok": "bool",
+                                     "name": "str", "tag": "str"}}, None),
+    batch)
+
+proto = """
+message T { double a = 1; int64 b = 2; bool ok = 3;
+            string name = 4; string tag = 5; }
+"""
+schema = ProtoSchema.parse(proto)
+payloads = [encode_message(
+    {"a": i * 0.5, "b": i, "ok": i % 2 == 0,
+     "name": f"user-{i}-café", "tag": f"segment {i % 32}"}, schema)
+    for i in range(N)]
+pbatch = MessageBatch.from_binary(payloads).to(dev)
+run("proto scalars+strings", ProtobufToArrowProcessor({"proto": proto}, None),
+    pbatch)
