@@ -416,6 +416,14 @@ _UNARY_MATH = {
 }
 
 
+def _fmt_num(x) -> str:
+    if isinstance(x, bool):
+        return "true" if x else "false"
+    if isinstance(x, float) and x == int(x) and abs(x) < 1e15:
+        return str(int(x))
+    return str(x)
+
+
 def _eval_func(e: FuncCall, env: Env) -> Value:
     name = e.name
     if name in _UNARY_MATH:
@@ -445,6 +453,26 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
                                      torch.zeros_like(result, dtype=torch.bool),
                                      v, result)
         return result
+    if name in ("to_int", "to_float", "to_string"):
+        # VRL-style conversions (processor/expr_proc.py translate_vrl)
+        v = eval_expr(e.args[0], env)
+        if name == "to_string":
+            if isinstance(v, Column) and v.kind == "binary":
+                return v
+            t = as_tensor(v, env)
+            return Column.from_strings([_fmt_num(x) for x in t.tolist()])
+        if isinstance(v, Column) and v.kind == "binary":
+            vals = v.to_strlist()
+            if name == "to_int":
+                return torch.tensor([int(float(x)) if x else 0 for x in vals],
+                                    dtype=torch.int64, device=env.device)
+            return torch.tensor([float(x) if x else 0.0 for x in vals],
+                                dtype=torch.float64, device=env.device)
+        t = as_tensor(v, env)
+        if name == "to_int":
+            return t.trunc().to(torch.int64) if t.dtype.is_floating_point \
+                else t.to(torch.int64)
+        return t.to(torch.float64)
     if name in ("length", "char_length", "octet_length"):
         v = eval_expr(e.args[0], env)
         if isinstance(v, Column) and v.kind == "binary":

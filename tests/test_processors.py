@@ -338,3 +338,37 @@ def test_inference_bert_cpu_tiny(run):
     out = run(p.process(b))[0]
     assert out.num_rows == 2  # 16 tokens / seq_len 8 → 2 sequences
     assert "logit_0" in out.columns and "score" in out.columns
+
+
+def test_vrl_statement_translation(run):
+    """VRL-style source (vrl.rs config shape) translated to columnar ops."""
+    from arkflow_amd.processors.expr_proc import ExprProcessor
+
+    async def main():
+        proc = ExprProcessor({"statement": """
+            .total = .price * .qty       # arithmetic on fields
+            .name_up = upcase(.name)
+            .price_i = to_int(.price)
+            .note = to_string(.qty)
+            del(.tmp)
+        """})
+        b = MessageBatch.from_dict({
+            "price": [1.5, 2.0], "qty": [2, 3],
+            "name": ["ab", "cd"], "tmp": [0, 0]})
+        out = (await proc.process(b))[0]
+        assert out.column("total").to_pylist() == [3.0, 6.0]
+        assert out.column("name_up").to_strlist() == ["AB", "CD"]
+        assert out.column("price_i").to_pylist() == [1, 2]
+        assert out.column("note").to_strlist() == ["2", "3"]
+        assert "tmp" not in out.columns
+
+    run(main())
+
+
+def test_vrl_statement_coalesce_and_errors():
+    from arkflow_amd.errors import ConfigError
+    from arkflow_amd.processors.expr_proc import ExprProcessor, translate_vrl
+    a, _ = translate_vrl(".v = .maybe ?? 0")
+    assert a == [("v", "coalesce(maybe, 0)")]
+    with pytest.raises(ConfigError):
+        ExprProcessor({"statement": "if .a > 1 { .b = 2 }"})
