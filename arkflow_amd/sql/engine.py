@@ -242,7 +242,10 @@ class SqlExecutor:
                 vals.float() if name != "count" else vals, gid.to(torch.int32),
                 g, "mean" if name == "avg" else name)
             return per_group[gid]
-        # rank family: stable sort by order keys (last-first), then by gid
+        from .udf import window_udf
+        wudf = window_udf(name)
+        # rank family / window UDFs: stable sort by order keys (last-first),
+        # then by gid
         perm = torch.arange(n, dtype=torch.int64, device=device)
         sort_keys = []
         for e, asc in reversed(w_.over.order_by):
@@ -273,6 +276,12 @@ class SqlExecutor:
                 cum = torch.cumsum(tie_change.long(), 0)
                 base = cum[group_start]
                 out_sorted = cum - base + 1
+        elif wudf is not None:
+            vals = None
+            if w_.args and not isinstance(w_.args[0], Star):
+                vals = as_tensor(eval_expr(w_.args[0], env), env)
+            res = wudf(vals, gid, g, perm)
+            return res if res.dtype != torch.bool else res.long()
         elif name != "row_number":
             raise SqlError(f"unsupported window function {name}()")
         out = torch.empty(n, dtype=torch.int64, device=device)

@@ -222,3 +222,22 @@ def test_window_functions(flow):
     r = q("SELECT id, dense_rank() OVER (ORDER BY value) AS dr FROM flow "
           "ORDER BY id", flow=flow)
     assert r.column("dr").to_pylist() == [2, 4, 1, 5, 3, 4]
+
+
+def test_window_udf():
+    """Custom window UDF (reference udf/window_udf.rs): per-partition
+    normalized value via the registered callable."""
+    import torch
+    from arkflow_amd.sql.udf import register_window_udf
+
+    def frac_of_partition(vals, gid, g, perm):
+        from arkflow_amd import ops
+        totals = ops.segment_reduce(vals.float(), gid.to(torch.int32), g,
+                                    "sum")
+        return vals.float() / totals[gid]
+
+    register_window_udf("frac_of_partition", frac_of_partition)
+    flow = MessageBatch.from_dict({"k": [1, 1, 2], "v": [1.0, 3.0, 10.0]})
+    r = q("SELECT v, frac_of_partition(v) OVER (PARTITION BY k) AS f "
+          "FROM flow ORDER BY v", flow=flow)
+    assert r.column("f").to_pylist() == [0.25, 0.75, 1.0]
