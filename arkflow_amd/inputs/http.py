@@ -35,6 +35,36 @@ class TokenBucket:
         return False
 
 
+class Lockout:
+    """Failed-auth lockout per client (reference auth_middleware.rs account
+    lockout): after `max_failures` bad tokens, the client is rejected with
+    429 for `lockout_secs`; a success clears the counter."""
+
+    def __init__(self, max_failures: int = 5, lockout_secs: float = 60.0):
+        self.max_failures = max_failures
+        self.lockout_secs = lockout_secs
+        self._failures = {}
+        self._locked_until = {}
+
+    def locked(self, client: str) -> bool:
+        until = self._locked_until.get(client, 0.0)
+        if until and time.monotonic() < until:
+            return True
+        if until:
+            del self._locked_until[client]
+            self._failures.pop(client, None)
+        return False
+
+    def failure(self, client: str) -> None:
+        n = self._failures.get(client, 0) + 1
+        self._failures[client] = n
+        if n >= self.max_failures:
+            self._locked_until[client] = time.monotonic() + self.lockout_secs
+
+    def success(self, client: str) -> None:
+        self._failures.pop(client, None)
+
+
 class HttpInput(Input):
     def __init__(self, config: dict, resource=None):
         self.address = config.get("address", "127.0.0.1:0")
@@ -44,6 +74,9 @@ class HttpInput(Input):
         rate = config.get("rate_limit")
         self.bucket = TokenBucket(float(rate), int(config.get("burst", rate)))\
             if rate else None
+        self.lockout = Lockout(
+            int(config.get("max_auth_failures", 5)),
+            float(config.get("lockout_secs", 60.0))) if self.token else None
         from ..codecs.helper import build_codec
         self.codec = build_codec(config, resource)
         self._q: asyncio.Queue = asyncio.Queue(maxsize=self.queue_size)
@@ -57,9 +90,14 @@ class HttpInput(Input):
 
         async def handler(request):
             if self.token:
+                client = request.remote or "?"
+                if self.lockout.locked(client):
+                    return web.Response(status=429, text="locked out")
                 auth = request.headers.get("Authorization", "")
                 if auth != f"Bearer {self.token}":
+                    self.lockout.failure(client)
                     return web.Response(status=401, text="unauthorized")
+                self.lockout.success(client)
             if self.bucket and not self.bucket.allow():
                 return web.Response(status=429, text="rate limited")
             body = await request.read()

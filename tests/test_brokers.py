@@ -323,3 +323,32 @@ def test_redis_list_mode(run):
         assert b2.column("__meta_offset").to_pylist() == [1]
 
     run(main())
+
+
+def test_http_auth_lockout(run):
+    """Repeated bad tokens lock the client out (429) until the window
+    expires; a good token clears the failure count."""
+    import aiohttp
+    from arkflow_amd.inputs.http import HttpInput
+
+    async def main():
+        inp = HttpInput({"address": "127.0.0.1:0", "token": "secret",
+                         "max_auth_failures": 3, "lockout_secs": 0.3})
+        await inp.connect()
+        url = f"http://127.0.0.1:{inp.port}/ingest"
+        async with aiohttp.ClientSession() as s:
+            bad = {"Authorization": "Bearer wrong"}
+            good = {"Authorization": "Bearer secret"}
+            for _ in range(3):
+                r = await s.post(url, data=b"x", headers=bad)
+                assert r.status == 401
+            r = await s.post(url, data=b"x", headers=bad)
+            assert r.status == 429  # locked
+            r = await s.post(url, data=b"x", headers=good)
+            assert r.status == 429  # still locked even with the right token
+            await asyncio.sleep(0.35)
+            r = await s.post(url, data=b"x", headers=good)
+            assert r.status == 200  # lockout expired, success clears state
+        await inp.close()
+
+    run(main(), timeout=30)
