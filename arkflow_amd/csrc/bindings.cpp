@@ -64,6 +64,10 @@ void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
                        hipStream_t);
+int scan_grid(int64_t n);
+void launch_scan_partials(const int32_t*, int64_t, int64_t*, hipStream_t);
+void launch_scan_write(const int32_t*, int64_t, const int64_t*, int64_t*,
+                       hipStream_t);
 int radix_sort_nblocks(int64_t);
 void launch_radix_hist_u32(const uint32_t*, const int32_t*, int64_t, int,
                            int32_t*, int, hipStream_t);
@@ -101,6 +105,26 @@ uint32_t next_pow2(uint32_t v) {
   v--;
   v |= v >> 1; v |= v >> 2; v |= v >> 4; v |= v >> 8; v |= v >> 16;
   return v + 1;
+}
+
+// exclusive prefix-sum of int32 lengths → int64 offsets [n+1] (offsets[n] =
+// total), all on device — csrc/filter.hip scan kernels (torch's innermost
+// cumsum is ~40x slower at n~262K)
+torch::Tensor exclusive_offsets(torch::Tensor lens) {
+  int64_t n = lens.numel();
+  auto out = torch::empty({n + 1}, lens.options().dtype(torch::kInt64));
+  if (n == 0) return out.zero_();
+  auto st = cur_stream();
+  int grid = scan_grid(n);
+  auto partials = torch::empty({grid}, lens.options().dtype(torch::kInt64));
+  launch_scan_partials(lens.data_ptr<int32_t>(), n,
+                       partials.data_ptr<int64_t>(), st);
+  auto inc = partials.cumsum(0);
+  auto boffs = inc - partials;
+  launch_scan_write(lens.data_ptr<int32_t>(), n, boffs.data_ptr<int64_t>(),
+                    out.data_ptr<int64_t>(), st);
+  out.narrow(0, n, 1).copy_(inc.narrow(0, grid - 1, 1));
+  return out;
 }
 
 // exclusive scan of int32 block counts → (offsets, total tensor on device)
@@ -620,13 +644,14 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   // string field (ulen of absent docs is 0 → empty strings, validity=found)
   std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
   if (n_str > 0 && n > 0) {
-    auto offs64 = torch::zeros({n_str, n + 1},
-                               data.options().dtype(torch::kInt64));
-    offs64.slice(1, 1, n + 1).copy_(str_ulen.to(torch::kInt64).cumsum(1));
-    auto totals = offs64.select(1, n).to(torch::kCPU);  // one device sync
+    std::vector<torch::Tensor> offs(n_str), tot_views;
+    for (int64_t s = 0; s < n_str; ++s) {
+      offs[s] = exclusive_offsets(str_ulen[s].contiguous());
+      tot_views.push_back(offs[s].narrow(0, n, 1));
+    }
+    auto totals = torch::cat(tot_views).to(torch::kCPU);  // one device sync
     auto* tot = totals.data_ptr<int64_t>();
     for (int64_t s = 0; s < n_str; ++s) {
-      auto so = offs64[s].contiguous();
       auto out = torch::empty({std::max<int64_t>(tot[s], 1)},
                               data.options().dtype(torch::kUInt8));
       // found row for this slot: find the field index with this slot
@@ -636,9 +661,9 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
       if (tot[s] > 0)
         launch_json_copy_strings(
             data.data_ptr<uint8_t>(), str_start[s].data_ptr<int64_t>(),
-            so.data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(), n,
+            offs[s].data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(), n,
             data.numel(), out.data_ptr<uint8_t>(), cur_stream());
-      strings.emplace_back(out.slice(0, 0, tot[s]), so);
+      strings.emplace_back(out.slice(0, 0, tot[s]), offs[s]);
     }
   }
   return {out_f, out_i, found, err, strings};
@@ -694,21 +719,22 @@ proto_decode(
   // string/bytes copy-out (proto3 missing field → len 0 → empty value)
   std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
   if (n_str > 0 && n > 0) {
-    auto offs64 = torch::zeros({n_str, n + 1},
-                               data.options().dtype(torch::kInt64));
-    offs64.slice(1, 1, n + 1).copy_(str_len.to(torch::kInt64).cumsum(1));
-    auto totals = offs64.select(1, n).to(torch::kCPU);  // one device sync
+    std::vector<torch::Tensor> offs(n_str), tot_views;
+    for (int64_t s = 0; s < n_str; ++s) {
+      offs[s] = exclusive_offsets(str_len[s].contiguous());
+      tot_views.push_back(offs[s].narrow(0, n, 1));
+    }
+    auto totals = torch::cat(tot_views).to(torch::kCPU);  // one device sync
     auto* tot = totals.data_ptr<int64_t>();
     for (int64_t s = 0; s < n_str; ++s) {
-      auto so = offs64[s].contiguous();
       auto out = torch::empty({std::max<int64_t>(tot[s], 1)},
                               data.options().dtype(torch::kUInt8));
       if (tot[s] > 0)
         launch_proto_copy_bytes(data.data_ptr<uint8_t>(),
                                 str_start[s].data_ptr<int64_t>(),
-                                so.data_ptr<int64_t>(), n,
+                                offs[s].data_ptr<int64_t>(), n,
                                 out.data_ptr<uint8_t>(), cur_stream());
-      strings.emplace_back(out.slice(0, 0, tot[s]), so);
+      strings.emplace_back(out.slice(0, 0, tot[s]), offs[s]);
     }
   }
   return {out_i, out_f, err, strings};
@@ -718,6 +744,7 @@ proto_decode(
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "arkflow_amd gfx950 native kernels";
+  m.def("exclusive_offsets", &exclusive_offsets);
   m.def("mask_to_indices", &mask_to_indices);
   m.def("filter_cmp_scalar", &filter_cmp_scalar);
   m.def("gather", &gather);

@@ -279,3 +279,84 @@ extern "C" void launch_bytes_hash(const uint8_t* data, const int64_t* offsets,
   if (grid < 1) return;
   bytes_hash_kernel<<<grid, 256, 0, st>>>(data, offsets, n, out);
 }
+
+// ------------------------------------------------------------------ prefix sum
+// Exclusive prefix sum of an int32 array into int64 offsets — the device-side
+// cumsum for string-column copy-out (json/proto decoders) and any
+// offsets-from-lengths step. Two passes: per-block sums, then (after a tiny
+// torch cumsum over <=2048 block sums) an in-block scan + add. Replaces
+// torch's innermost-dim scan kernel, which is ~600 µs at n=262K vs ~15 µs
+// for this pair.
+#define SCAN_BLOCK 256
+#define SCAN_IPT 8
+#define SCAN_TILE (SCAN_BLOCK * SCAN_IPT)
+
+__global__ void scan_partials_kernel(const int32_t* __restrict__ in, int64_t n,
+                                     int64_t* __restrict__ partials) {
+  int64_t base = (int64_t)blockIdx.x * SCAN_TILE;
+  int64_t local = 0;
+#pragma unroll
+  for (int k = 0; k < SCAN_IPT; ++k) {
+    int64_t i = base + threadIdx.x * SCAN_IPT + k;
+    if (i < n) local += in[i];
+  }
+  __shared__ int64_t lds[SCAN_BLOCK];
+  lds[threadIdx.x] = local;
+  __syncthreads();
+  // tree reduce
+  for (int s = SCAN_BLOCK / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) lds[threadIdx.x] += lds[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
+}
+
+__global__ void scan_write_kernel(const int32_t* __restrict__ in, int64_t n,
+                                  const int64_t* __restrict__ block_offs,
+                                  int64_t* __restrict__ out) {
+  int64_t base = (int64_t)blockIdx.x * SCAN_TILE;
+  int32_t vals[SCAN_IPT];
+  int64_t local = 0;
+#pragma unroll
+  for (int k = 0; k < SCAN_IPT; ++k) {
+    int64_t i = base + threadIdx.x * SCAN_IPT + k;
+    vals[k] = i < n ? in[i] : 0;
+    local += vals[k];
+  }
+  __shared__ int64_t lds[SCAN_BLOCK];
+  lds[threadIdx.x] = local;
+  __syncthreads();
+  // Hillis-Steele inclusive scan over per-thread sums
+  for (int s = 1; s < SCAN_BLOCK; s <<= 1) {
+    int64_t add = threadIdx.x >= s ? lds[threadIdx.x - s] : 0;
+    __syncthreads();
+    lds[threadIdx.x] += add;
+    __syncthreads();
+  }
+  int64_t run = block_offs[blockIdx.x] +
+                (threadIdx.x ? lds[threadIdx.x - 1] : 0);
+#pragma unroll
+  for (int k = 0; k < SCAN_IPT; ++k) {
+    int64_t i = base + threadIdx.x * SCAN_IPT + k;
+    if (i < n) {
+      out[i] = run;  // exclusive
+      run += vals[k];
+    }
+  }
+}
+
+extern "C" int scan_grid(int64_t n) {
+  return (int)((n + SCAN_TILE - 1) / SCAN_TILE);
+}
+
+extern "C" void launch_scan_partials(const int32_t* in, int64_t n,
+                                     int64_t* partials, hipStream_t st) {
+  scan_partials_kernel<<<scan_grid(n), SCAN_BLOCK, 0, st>>>(in, n, partials);
+}
+
+extern "C" void launch_scan_write(const int32_t* in, int64_t n,
+                                  const int64_t* block_offs, int64_t* out,
+                                  hipStream_t st) {
+  scan_write_kernel<<<scan_grid(n), SCAN_BLOCK, 0, st>>>(in, n, block_offs,
+                                                         out);
+}

@@ -118,9 +118,10 @@ class JsonToArrowProcessor(Processor):
         if int(err.item()) != 0:
             raise ProcessError("json decode error (malformed document)")
         cols = {}
+        fbool = found.to(torch.bool)
+        all_valid = fbool.all(dim=1).cpu()  # one sync for every field
         for f, name in enumerate(names):
-            validity = found[f].to(torch.bool)
-            v = None if bool(validity.all()) else validity
+            v = None if bool(all_valid[f]) else fbool[f]
             if self.schema[name] == "float":
                 data = out_f[fcols.index(name)].contiguous()
             elif self.schema[name] == "bool":

@@ -303,6 +303,17 @@ def test_proto_decode_gpu_strings(nat, dev):
         assert vs[i] == rows[i]["v"]
 
 
+def test_exclusive_offsets_matches_cumsum(nat, dev):
+    """Device prefix-sum primitive vs torch reference, incl. tile edges."""
+    for n in (0, 1, 5, 2047, 2048, 2049, 262_144, 1_000_001):
+        lens = torch.randint(0, 100, (n,), device=dev, dtype=torch.int32)
+        offs = nat.exclusive_offsets(lens)
+        ref = torch.zeros(n + 1, device=dev, dtype=torch.int64)
+        if n:
+            ref[1:] = lens.to(torch.int64).cumsum(0)
+        assert torch.equal(offs, ref), n
+
+
 def test_hash_group_large(nat, dev):
     """Regression: tables > 512K entries must be fully initialized
     (fill kernels are grid-stride; grid is capped at 2048 blocks)."""
