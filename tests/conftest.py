@@ -28,5 +28,11 @@ def run():
         try:
             return loop.run_until_complete(asyncio.wait_for(coro, timeout))
         finally:
+            pending = asyncio.all_tasks(loop)
+            for t in pending:
+                t.cancel()
+            if pending:
+                loop.run_until_complete(
+                    asyncio.gather(*pending, return_exceptions=True))
             loop.close()
     return _run
