@@ -179,6 +179,13 @@ class Column:
                 validity=self.validity[indices] if self.validity is not None else None,
             )
         idx = indices.to(torch.int64)
+        if self.data.is_cuda:
+            from . import ops
+            out, new_off = ops.require_native().take_binary(
+                self.data, self.offsets, idx.to(self.data.device))
+            return Column("binary", out.contiguous(), new_off,
+                          self.validity[idx]
+                          if self.validity is not None else None)
         lengths_all = self.offsets[1:] - self.offsets[:-1]
         lengths = lengths_all[idx]
         new_off = torch.zeros(idx.shape[0] + 1, dtype=torch.int64, device=idx.device)

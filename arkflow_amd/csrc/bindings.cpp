@@ -669,6 +669,33 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   return {out_f, out_i, found, err, strings};
 }
 
+// gather rows of a binary column: (data, offsets, idx) -> (out_data,
+// out_offsets). Device prefix-sum for the new offsets + span-copy kernel;
+// ONE host sync (the total, needed for allocation).
+std::tuple<torch::Tensor, torch::Tensor> take_binary(torch::Tensor data,
+                                                     torch::Tensor offsets,
+                                                     torch::Tensor idx) {
+  check_cuda(data, "data");
+  check_cuda(offsets, "offsets");
+  check_cuda(idx, "idx");
+  int64_t n = offsets.numel() - 1;
+  auto idx64 = idx.scalar_type() == torch::kInt64 ? idx : idx.to(torch::kInt64);
+  int64_t m = idx64.numel();
+  auto lens_all = offsets.slice(0, 1, n + 1) - offsets.slice(0, 0, n);
+  auto lens = lens_all.index_select(0, idx64).to(torch::kInt32);
+  auto starts = offsets.slice(0, 0, n).index_select(0, idx64).contiguous();
+  auto new_off = exclusive_offsets(lens.contiguous());
+  int64_t total = m ? new_off.narrow(0, m, 1).item<int64_t>() : 0;
+  auto out = torch::empty({std::max<int64_t>(total, 1)},
+                          data.options().dtype(torch::kUInt8));
+  if (total > 0)
+    launch_proto_copy_bytes(data.data_ptr<uint8_t>(),
+                            starts.data_ptr<int64_t>(),
+                            new_off.data_ptr<int64_t>(), m,
+                            out.data_ptr<uint8_t>(), cur_stream());
+  return {out.slice(0, 0, total), new_off};
+}
+
 torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
@@ -745,6 +772,7 @@ proto_decode(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "arkflow_amd gfx950 native kernels";
   m.def("exclusive_offsets", &exclusive_offsets);
+  m.def("take_binary", &take_binary);
   m.def("mask_to_indices", &mask_to_indices);
   m.def("filter_cmp_scalar", &filter_cmp_scalar);
   m.def("gather", &gather);

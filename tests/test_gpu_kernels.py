@@ -314,6 +314,22 @@ def test_exclusive_offsets_matches_cumsum(nat, dev):
         assert torch.equal(offs, ref), n
 
 
+def test_take_binary_matches_cpu(nat, dev):
+    """Binary-column gather (take_binary kernel) vs the host path."""
+    from arkflow_amd.batch import MessageBatch
+    strs = [f"row-{i}-" + "x" * (i % 50) for i in range(100_000)]
+    b = MessageBatch.from_dict({"s": strs, "v": list(range(100_000))})
+    idx = torch.randint(0, 100_000, (30_000,), dtype=torch.int64)
+    cpu = b.column("s").take(idx).to_pylist()
+    gpu_col = b.to(dev).column("s").take(idx.to(dev))
+    assert gpu_col.data.is_cuda
+    assert gpu_col.to_pylist() == cpu
+    # empty gather
+    e = b.to(dev).column("s").take(torch.empty(0, dtype=torch.int64,
+                                               device=dev))
+    assert len(e) == 0
+
+
 def test_hash_group_large(nat, dev):
     """Regression: tables > 512K entries must be fully initialized
     (fill kernels are grid-stride; grid is capped at 2048 blocks)."""
