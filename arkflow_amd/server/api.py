@@ -82,6 +82,16 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
         return await cp.lifecycle(stream_id, op)
 
     # ---- operations ---------------------------------------------------------
+    @app.get(f"{prefix}/streams/{{stream_id}}/metrics",
+             dependencies=[Depends(auth)])
+    async def stream_metrics(stream_id: str):
+        from ..errors import ArkError
+        try:
+            e = engine.runtime.get(stream_id)
+        except ArkError:
+            raise HTTPException(404, f"unknown stream {stream_id!r}")
+        return e.metrics.snapshot()
+
     @app.get(f"{prefix}/operations", dependencies=[Depends(auth)])
     async def operations(limit: int = 100):
         return [o.to_dict() for o in engine.runtime.operations.list(limit)]
@@ -153,6 +163,15 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
     @app.get(f"{prefix}/configuration/versions", dependencies=[Depends(auth)])
     async def versions():
         return cp.versions.list()
+
+    @app.get(f"{prefix}/configuration/versions/{{version}}",
+             dependencies=[Depends(auth)])
+    async def version_detail(version: int):
+        from ..control_plane import redact_secrets
+        v = cp.versions.get(version)
+        if v is None:
+            raise HTTPException(404, f"no config version {version}")
+        return redact_secrets(v)
 
     # ---- components ----------------------------------------------------------
     @app.get(f"{prefix}/components", dependencies=[Depends(auth)])

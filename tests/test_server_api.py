@@ -215,3 +215,32 @@ def test_config_diff(run):
         assert "input" in d["changed"][0]["fields"]
 
     run(main())
+
+
+def test_stream_metrics_and_version_detail(run):
+    async def main():
+        eng = _engine()
+        for sc in eng.config.streams:
+            eng.runtime.register(sc)
+        app = create_app(eng)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://n") as c:
+            r = await c.get("/api/v1/streams/s1/metrics")
+            assert r.status_code == 200
+            assert "input_messages" in r.json()
+            assert (await c.get("/api/v1/streams/nope/metrics")
+                    ).status_code == 404
+            await eng.control_plane.apply_configuration(
+                {"streams": [{"id": "s2",
+                              "input": {"type": "generate", "batch_size": 1,
+                                        "interval": "50ms",
+                                        "fields": {"v": {"dtype": "float32"}}},
+                              "output": {"type": "drop"}}]}, note="v1")
+            r = await c.get("/api/v1/configuration/versions/1")
+            assert r.status_code == 200
+            assert (await c.get("/api/v1/configuration/versions/99")
+                    ).status_code == 404
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=30)
