@@ -10,6 +10,7 @@ source ack; recovery replays everything after the cursor (§3.4).
 from __future__ import annotations
 
 import asyncio
+import heapq
 from typing import AsyncIterator, List, Optional, Tuple
 
 from ..config import DurabilityConfig
@@ -51,6 +52,13 @@ class Wal:
         self._closed = False
         self._cursor_dirty = 0
         self.cursor_flush_every = cursor_flush_every
+        # low-watermark ack tracking: the cursor only advances past a seq when
+        # EVERY issued seq at or below it is acked — out-of-order acks (e.g. a
+        # failed output write for batch k while k+1 succeeded) must not leap
+        # the cursor over the un-acked entry or it would be lost on replay
+        self._out_heap: List[int] = []  # issued (appended/replayed) seqs
+        self._acked: set = set()
+        self._max_issued = store.max_seq
 
     @staticmethod
     def open(config: DurabilityConfig, stream_id: str = "stream") -> "Wal":
@@ -67,6 +75,7 @@ class Wal:
     async def append(self, batch) -> int:
         self._seq += 1
         seq = self._seq
+        self._issue(seq)
         if self.sync_policy == "per_entry":
             payload = serialize_batch(batch)
             loop = asyncio.get_running_loop()
@@ -117,18 +126,33 @@ class Wal:
         await loop.run_in_executor(None, work)
 
     # ----------------------------------------------------------------- advance
+    def _issue(self, seq: int) -> None:
+        heapq.heappush(self._out_heap, seq)
+        if seq > self._max_issued:
+            self._max_issued = seq
+
+    def _frontier(self) -> int:
+        while self._out_heap and self._out_heap[0] in self._acked:
+            self._acked.discard(heapq.heappop(self._out_heap))
+        if self._out_heap:
+            return self._out_heap[0] - 1  # smallest un-acked outstanding
+        return self._max_issued
+
     async def advance(self, seq: int) -> None:
-        """Monotonic cursor advance; flushed every N acks + on close."""
-        if seq <= self.store.cursor:
+        """Low-watermark cursor advance (contiguous-ack frontier); flushed
+        every N acks + on close."""
+        self._acked.add(seq)
+        frontier = self._frontier()
+        if frontier <= self.store.cursor:
             return
         self._cursor_dirty += 1
         if self._cursor_dirty >= self.cursor_flush_every:
             self._cursor_dirty = 0
             loop = asyncio.get_running_loop()
-            await loop.run_in_executor(None, self.store.write_cursor, seq)
+            await loop.run_in_executor(None, self.store.write_cursor, frontier)
         else:
             # in-memory advance only; persisted on the next flush/close
-            self.store._cursor = max(self.store._cursor, seq)
+            self.store._cursor = max(self.store._cursor, frontier)
 
     # ------------------------------------------------------------------ replay
     async def read_after_cursor(self) -> AsyncIterator:
@@ -137,6 +161,7 @@ class Wal:
         entries = await loop.run_in_executor(
             None, lambda: list(self.store.read_after(self.store.cursor)))
         for seq, payload in entries:
+            self._issue(seq)
             yield seq, deserialize_batch(payload)
 
     # ------------------------------------------------------------------- close
