@@ -91,7 +91,12 @@ class FakeBus:
 
 # ---------------------------------------------------------------------- kafka
 class KafkaAck(Ack):
-    """Commit the consumed offset on ack (input/kafka.rs:277-296)."""
+    """Commit the consumed offset on ack (input/kafka.rs:277-296).
+
+    Kafka's offset model cannot express ack gaps (commit N ⇒ everything
+    below N is consumed) — identical to the reference's store_offset
+    semantics. For gap-safe recovery enable the WAL, whose cursor is a
+    contiguous-ack low watermark (wal/wal.py advance())."""
 
     def __init__(self, bus: FakeBus, group: str, topic: str, partition: int,
                  offset: int):
