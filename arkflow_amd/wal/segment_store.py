@@ -157,9 +157,12 @@ class SegmentWalStore:
         return self._cursor
 
     def write_cursor(self, seq: int) -> None:
-        self._cursor = max(self._cursor, seq)
-        self._store_manifest()
-        self._reclaim()
+        # called from executor threads concurrently with seals — the manifest
+        # tmp-file replace must be serialized (soak-caught race)
+        with self._lock:
+            self._cursor = max(self._cursor, seq)
+            self._store_manifest()
+            self._reclaim()
 
     def _reclaim(self) -> None:
         """Delete sealed segments entirely ≤ cursor (s3.rs reclamation)."""

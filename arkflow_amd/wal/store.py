@@ -174,6 +174,8 @@ class LocalWalStore:
 
     def __init__(self, path: str, stream_id: str = "stream",
                  compress: bool = False, fsync: bool = True):
+        import threading
+        self._lock = threading.Lock()
         self.dir = path
         os.makedirs(path, exist_ok=True)
         self.log_path = os.path.join(path, f"{stream_id}.wal")
@@ -198,14 +200,15 @@ class LocalWalStore:
             return 0
 
     def write_cursor(self, seq: int) -> None:
-        self._cursor = max(self._cursor, seq)
-        raw = struct.pack(">Q", self._cursor)
-        tmp = self.cursor_path + ".tmp"
-        with open(tmp, "wb") as f:
-            f.write(raw + struct.pack(">I", zlib.crc32(raw)))
-            if self.fsync:
-                os.fsync(f.fileno())
-        os.replace(tmp, self.cursor_path)
+        with self._lock:  # concurrent executor-thread acks (soak-caught race)
+            self._cursor = max(self._cursor, seq)
+            raw = struct.pack(">Q", self._cursor)
+            tmp = self.cursor_path + ".tmp"
+            with open(tmp, "wb") as f:
+                f.write(raw + struct.pack(">I", zlib.crc32(raw)))
+                if self.fsync:
+                    os.fsync(f.fileno())
+            os.replace(tmp, self.cursor_path)
 
     @property
     def cursor(self) -> int:
@@ -214,6 +217,11 @@ class LocalWalStore:
     # log ----------------------------------------------------------------
     def append_batch(self, entries: List[Tuple[int, bytes]],
                      sync: bool = True) -> None:
+        with self._lock:
+            self._append_locked(entries, sync)
+
+    def _append_locked(self, entries: List[Tuple[int, bytes]],
+                       sync: bool) -> None:
         if _nwal is not None and not self.compress:
             self._f.write(_nwal.encode_frames(entries))
         else:

@@ -257,3 +257,19 @@ def test_chaos_flaky_output_at_least_once(tmp_path, run):
         assert not missing, f"lost rows {missing}"
 
     run(main())
+
+
+def test_concurrent_cursor_writes_race_free(tmp_path):
+    """write_cursor from many threads (executor acks) must not collide on
+    the tmp-file replace (GPU-soak-caught race) for either store."""
+    from concurrent.futures import ThreadPoolExecutor
+    from arkflow_amd.registry import build_component
+
+    for spec in ({"type": "local", "path": str(tmp_path / "l")},
+                 {"type": "segment", "path": str(tmp_path / "s")}):
+        st = build_component("wal_store", {**spec, "stream_id": "r"})
+        st.append_batch([(i, b"x" * 100) for i in range(1, 65)], True)
+        with ThreadPoolExecutor(16) as pool:
+            list(pool.map(st.write_cursor, list(range(1, 600)) * 4))
+        assert st.cursor == 599
+        st.close()
