@@ -82,10 +82,20 @@ def softmax_bf16(x: torch.Tensor, scale: float = 1.0) -> torch.Tensor:
 def attention_bf16(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                    scale: float) -> torch.Tensor:
     """Fused attention for [B,H,S,D] bf16. GPU: one-workgroup-per-(b,h)
-    MFMA kernel (csrc/attention.hip); CPU: fp32 reference."""
+    MFMA kernel (csrc/attention.hip) for its supported tile (S=128, D=64 —
+    the BERT-base shape); other shapes compose batched hipBLASLt GEMMs with
+    our softmax kernel. CPU: fp32 reference."""
     if q.is_cuda:
         nat = require_native()
-        return nat.attention_bf16(q.contiguous(), k.contiguous(),
-                                  v.contiguous(), scale)
+        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        try:
+            return nat.attention_bf16(qc, kc, vc, scale)
+        except RuntimeError as e:
+            if "unsupported shape" not in str(e):
+                raise
+        scores = torch.matmul(qc, kc.transpose(-1, -2))
+        p = nat.softmax_bf16(scores.reshape(-1, scores.shape[-1]).contiguous(),
+                             scale).reshape(scores.shape)
+        return torch.matmul(p, vc)
     p = torch.softmax(q.float() @ k.float().transpose(-1, -2) * scale, dim=-1)
     return (p @ v.float()).to(q.dtype)

@@ -314,6 +314,22 @@ def test_exclusive_offsets_matches_cumsum(nat, dev):
         assert torch.equal(offs, ref), n
 
 
+def test_attention_other_shapes(nat, dev):
+    """Shapes outside the fused kernel's tile compose GEMM + our softmax;
+    values must match the CPU fp32 oracle."""
+    from arkflow_amd.ops.nn import attention_bf16
+    torch.manual_seed(3)
+    for B, H, S, D in ((2, 4, 64, 64), (1, 2, 256, 32), (2, 3, 96, 128)):
+        q = torch.randn(B, H, S, D, dtype=torch.bfloat16)
+        k = torch.randn(B, H, S, D, dtype=torch.bfloat16)
+        v = torch.randn(B, H, S, D, dtype=torch.bfloat16)
+        ref = attention_bf16(q, k, v, 1.0 / D ** 0.5)
+        got = attention_bf16(q.to(dev), k.to(dev), v.to(dev),
+                             1.0 / D ** 0.5).cpu()
+        err = (got.float() - ref.float()).abs().max().item()
+        assert err < 0.05, (B, H, S, D, err)
+
+
 def test_take_binary_matches_cpu(nat, dev):
     """Binary-column gather (take_binary kernel) vs the host path."""
     from arkflow_amd.batch import MessageBatch
