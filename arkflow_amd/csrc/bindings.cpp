@@ -64,6 +64,8 @@ void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
                        hipStream_t);
+void launch_bytes_match(const uint8_t*, const int64_t*, int64_t,
+                        const uint8_t*, int, int, bool*, hipStream_t);
 int scan_grid(int64_t n);
 void launch_scan_partials(const int32_t*, int64_t, int64_t*, hipStream_t);
 void launch_scan_write(const int32_t*, int64_t, const int64_t*, int64_t*,
@@ -696,6 +698,20 @@ std::tuple<torch::Tensor, torch::Tensor> take_binary(torch::Tensor data,
   return {out.slice(0, 0, total), new_off};
 }
 
+torch::Tensor bytes_match(torch::Tensor data, torch::Tensor offsets,
+                          std::string needle, int64_t mode) {
+  check_cuda(data, "data");
+  check_cuda(offsets, "offsets");
+  TORCH_CHECK(needle.size() <= 64, "LIKE needle too long for device match");
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, data.options().dtype(torch::kBool));
+  if (n > 0)
+    launch_bytes_match(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
+                       n, (const uint8_t*)needle.data(), (int)needle.size(),
+                       (int)mode, out.data_ptr<bool>(), cur_stream());
+  return out;
+}
+
 torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
@@ -773,6 +789,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "arkflow_amd gfx950 native kernels";
   m.def("exclusive_offsets", &exclusive_offsets);
   m.def("take_binary", &take_binary);
+  m.def("bytes_match", &bytes_match);
   m.def("mask_to_indices", &mask_to_indices);
   m.def("filter_cmp_scalar", &filter_cmp_scalar);
   m.def("gather", &gather);

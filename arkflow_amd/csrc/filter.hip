@@ -360,3 +360,62 @@ extern "C" void launch_scan_write(const int32_t* in, int64_t n,
   scan_write_kernel<<<scan_grid(n), SCAN_BLOCK, 0, st>>>(in, n, block_offs,
                                                          out);
 }
+
+// ------------------------------------------------------------------ LIKE match
+// SQL LIKE on a binary column entirely on-device: mode 0 = contains (%x%),
+// 1 = prefix (x%), 2 = suffix (%x), 3 = equals. Needle passed by value in
+// the kernarg segment (bounded, like JsonSpec).
+#define MATCH_MAX_NEEDLE 64
+
+struct MatchNeedle {
+  uint8_t bytes[MATCH_MAX_NEEDLE];
+  int len;
+  int mode;
+};
+
+__global__ void bytes_match_kernel(const uint8_t* __restrict__ data,
+                                   const int64_t* __restrict__ offsets,
+                                   int64_t n, MatchNeedle nd,
+                                   bool* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t s = offsets[i], e = offsets[i + 1];
+    const int64_t len = e - s;
+    bool m = false;
+    if (nd.len == 0) {
+      m = (nd.mode == 3) ? (len == 0) : true;
+    } else if (len >= nd.len) {
+      if (nd.mode == 1 || nd.mode == 3) {        // prefix / equals
+        m = (nd.mode == 1 || len == nd.len);
+        for (int k = 0; m && k < nd.len; ++k)
+          m = data[s + k] == nd.bytes[k];
+      } else if (nd.mode == 2) {                 // suffix
+        m = true;
+        for (int k = 0; m && k < nd.len; ++k)
+          m = data[e - nd.len + k] == nd.bytes[k];
+      } else {                                   // contains
+        for (int64_t p = s; !m && p + nd.len <= e; ++p) {
+          bool eq = true;
+          for (int k = 0; eq && k < nd.len; ++k)
+            eq = data[p + k] == nd.bytes[k];
+          m = eq;
+        }
+      }
+    }
+    out[i] = m;
+  }
+}
+
+extern "C" void launch_bytes_match(const uint8_t* data, const int64_t* offsets,
+                                   int64_t n, const uint8_t* needle, int nl,
+                                   int mode, bool* out, hipStream_t st) {
+  MatchNeedle nd{};
+  nd.len = nl > MATCH_MAX_NEEDLE ? MATCH_MAX_NEEDLE : nl;
+  nd.mode = mode;
+  for (int k = 0; k < nd.len; ++k) nd.bytes[k] = needle[k];
+  int grid = (int)((n + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  bytes_match_kernel<<<grid, 256, 0, st>>>(data, offsets, n, nd, out);
+}

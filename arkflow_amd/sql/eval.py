@@ -290,6 +290,22 @@ def eval_expr(e, env: Env) -> Value:
         if not isinstance(v, Column) or v.kind != "binary":
             raise SqlError("LIKE requires a string column")
         pat = e.pattern
+        if v.data.is_cuda and len(pat) <= 66 and "_" not in pat \
+                and "%" not in pat.strip("%"):
+            from .. import ops
+            body = pat.strip("%").encode()
+            if pat.startswith("%") and pat.endswith("%") and len(pat) > 1:
+                mode = 0
+            elif pat.endswith("%") and not pat.startswith("%"):
+                mode = 1
+            elif pat.startswith("%") and not pat.endswith("%"):
+                mode = 2
+            else:
+                mode = 3
+            # pybind converts Python bytes → std::string raw (no re-encode)
+            r = ops.require_native().bytes_match(
+                v.data.contiguous(), v.offsets.contiguous(), body, mode)
+            return ~r if e.negated else r
         vals = v.to_pylist()
         if pat.startswith("%") and pat.endswith("%") and len(pat) > 1:
             needle = pat[1:-1].encode()

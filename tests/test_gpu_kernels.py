@@ -330,6 +330,26 @@ def test_attention_other_shapes(nat, dev):
         assert err < 0.05, (B, H, S, D, err)
 
 
+def test_like_on_gpu_matches_cpu(nat, dev):
+    """LIKE over device strings (bytes_match kernel) vs the host path,
+    all four modes + NOT LIKE + unicode needles."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+    vals = ["apple pie", "apple", "pineapple", "grape", "", "café au lait",
+            "PIE apple"] * 2000
+    b = MessageBatch.from_dict({"s": vals, "v": list(range(len(vals)))})
+    for sql in ("SELECT v FROM flow WHERE s LIKE 'apple%'",
+                "SELECT v FROM flow WHERE s LIKE '%apple'",
+                "SELECT v FROM flow WHERE s LIKE '%apple%'",
+                "SELECT v FROM flow WHERE s LIKE 'grape'",
+                "SELECT v FROM flow WHERE s NOT LIKE '%apple%'",
+                "SELECT v FROM flow WHERE s LIKE '%café%'"):
+        cpu = SqlExecutor(sql).execute({"flow": b}).column("v").to_pylist()
+        gpu = SqlExecutor(sql).execute(
+            {"flow": b.to(dev)}).column("v").to_pylist()
+        assert gpu == cpu, sql
+
+
 def test_take_binary_matches_cpu(nat, dev):
     """Binary-column gather (take_binary kernel) vs the host path."""
     from arkflow_amd.batch import MessageBatch
