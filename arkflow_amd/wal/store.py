@@ -47,6 +47,9 @@ def _tensor_bytes(t: torch.Tensor):
 
 
 def _tensor_from(data: bytes, dtype: torch.dtype, n: int) -> torch.Tensor:
+    if len(data) == 0:  # torch.frombuffer rejects zero-length buffers
+        return torch.empty(0, dtype=torch.int16).view(torch.bfloat16) \
+            if dtype == torch.bfloat16 else torch.empty(0, dtype=dtype)
     if dtype == torch.bfloat16:
         return torch.frombuffer(bytearray(data),
                                 dtype=torch.int16).view(torch.bfloat16)[:n]
