@@ -12,7 +12,7 @@ class RuntimeMetrics:
         "input_batches", "input_messages", "processing_errors",
         "output_batches", "output_messages", "input_errors",
         "input_reconnects", "output_errors", "restarts",
-        "stage_ns", "started_at",
+        "stage_ns", "started_at", "wal_lag",
     )
 
     def __init__(self):
@@ -27,6 +27,7 @@ class RuntimeMetrics:
         self.restarts = 0
         # per-stage wall-time accumulators (ns): input/process/output
         self.stage_ns: Dict[str, int] = {"input": 0, "process": 0, "output": 0}
+        self.wal_lag = 0  # entries appended but not yet cursor-acked
         self.started_at = time.time()
 
     def snapshot(self) -> dict:
@@ -43,6 +44,7 @@ class RuntimeMetrics:
             "restarts": self.restarts,
             "uptime_secs": time.time() - self.started_at,
             "stage_ms": {k: v / 1e6 for k, v in self.stage_ns.items()},
+            "wal_lag": self.wal_lag,
         }
 
 

@@ -224,6 +224,8 @@ def render_prometheus(engine) -> str:
         lines.append(
             f'arkflow_stream_uptime_seconds{{stream="{sid}"}} '
             f'{snap["uptime_secs"]:.3f}')
+        lines.append(
+            f'arkflow_wal_lag{{stream="{sid}"}} {snap.get("wal_lag", 0)}')
         for stage, ms in snap["stage_ms"].items():
             lines.append(
                 f'arkflow_stage_ms_total{{stream="{sid}",stage="{stage}"}} '

@@ -202,6 +202,7 @@ class Stream:
             if self.wal is not None:
                 from .wal.wal import WalAck
                 seq = await self.wal.append(batch)
+                self.metrics.wal_lag = max(0, seq - self.wal.store.cursor)
                 ack = WalAck(self.wal, seq, ack)
             await self._forward(input_q, batch, ack)
 

@@ -189,6 +189,9 @@ class LocalWalStore:
         self._compact_on_open()
         self._f = open(self.log_path, "ab")
         self.max_seq = self._scan_max_seq()
+        # online compaction trigger: acked prefix larger than this is dropped
+        # without waiting for a restart (segment store reclaims online too)
+        self.compact_bytes = 64 * 1024 * 1024
 
     # cursor -------------------------------------------------------------
     def _read_cursor(self) -> int:
@@ -212,6 +215,29 @@ class LocalWalStore:
                 if self.fsync:
                     os.fsync(f.fileno())
             os.replace(tmp, self.cursor_path)
+            try:
+                if self._f.tell() >= self.compact_bytes:
+                    self._compact_locked()
+            except OSError:
+                pass
+
+    def _compact_locked(self) -> None:
+        """Rewrite the log keeping only entries past the cursor; called with
+        the lock held (online analog of _compact_on_open)."""
+        self._f.flush()
+        with open(self.log_path, "rb") as f:
+            buf = f.read()
+        keep = [(seq, payload) for seq, payload in decode_frames(buf)
+                if seq > self._cursor]
+        tmp = self.log_path + ".tmp"
+        with open(tmp, "wb") as f:
+            for seq, payload in keep:
+                f.write(encode_frame(seq, payload, self.compress))
+            if self.fsync:
+                os.fsync(f.fileno())
+        os.replace(tmp, self.log_path)
+        self._f.close()
+        self._f = open(self.log_path, "ab")
 
     @property
     def cursor(self) -> int:

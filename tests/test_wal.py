@@ -274,3 +274,26 @@ def test_concurrent_cursor_writes_race_free(tmp_path):
             list(pool.map(st.write_cursor, list(range(1, 200)) * 4))
         assert st.cursor == 199
         st.close()
+
+
+def test_local_store_online_compaction(tmp_path):
+    """The acked prefix is dropped online once the log exceeds the
+    compaction threshold, without a restart."""
+    import os
+    from arkflow_amd.wal.store import LocalWalStore
+    st = LocalWalStore(str(tmp_path), "oc", fsync=False)
+    st.compact_bytes = 50_000
+    payload = b"z" * 1000
+    st.append_batch([(i, payload) for i in range(1, 101)], True)
+    before = os.path.getsize(st.log_path)
+    assert before > 50_000
+    st.write_cursor(90)  # 90% acked → compaction triggers
+    after = os.path.getsize(st.log_path)
+    assert after < before / 5
+    # the live tail survives and replays
+    live = [s for s, _ in st.read_after(st.cursor)]
+    assert live == list(range(91, 101))
+    # appends continue cleanly on the reopened handle
+    st.append_batch([(101, payload)], True)
+    assert [s for s, _ in st.read_after(100)] == [101]
+    st.close()
