@@ -63,13 +63,45 @@ def registry(kind: str) -> _Registry:
         raise ConfigError(f"unknown component kind {kind!r}") from None
 
 
+def _json_type(v) -> dict:
+    if isinstance(v, bool):
+        return {"type": "boolean"}
+    if isinstance(v, int):
+        return {"type": "integer"}
+    if isinstance(v, float):
+        return {"type": "number"}
+    if isinstance(v, str):
+        return {"type": "string"}
+    if isinstance(v, list):
+        item = _json_type(v[0]) if v else {}
+        return {"type": "array", "items": item}
+    if isinstance(v, dict):
+        return {"type": "object",
+                "properties": {k: _json_type(x) for k, x in v.items()}}
+    return {}
+
+
+def schema_from_example(example: dict) -> dict:
+    """Minimal JSON Schema derived from a component's example config
+    (reference components publish hand-written schemas,
+    component/mod.rs:96; the derived form covers discovery + docs and is
+    refined per component over time)."""
+    props = {k: _json_type(v) for k, v in example.items() if k != "type"}
+    props["type"] = {"type": "string", "const": example.get("type", "")}
+    return {"type": "object", "properties": props,
+            "additionalProperties": True}
+
+
 def register(kind: str, name: str, *, description: str = "",
              config_schema: Optional[dict] = None, example: Optional[dict] = None):
     """Decorator: ``@register("input", "generate")`` on a builder callable."""
     def deco(builder):
+        schema = config_schema
+        if schema is None and example:
+            schema = schema_from_example({**example, "type": name})
         registry(kind).register(
             name, builder,
-            ComponentMetadata(kind, name, description, config_schema, example),
+            ComponentMetadata(kind, name, description, schema, example),
         )
         return builder
     return deco

@@ -186,6 +186,9 @@ class SegmentWalStore:
 
 
 @register("wal_store", "segment",
+          example={"type": "segment", "path": "./wal",
+                   "segment_strategy": "balanced", "max_entries": 1024,
+                   "put_workers": 4},
           description="Segmented WAL store (sealed segments + manifest + "
                       "parallel PUT workers; the S3-backend design)")
 def _build_segment_store(config: dict, resource=None) -> SegmentWalStore:

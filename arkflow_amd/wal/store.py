@@ -307,7 +307,9 @@ class LocalWalStore:
 
 
 @register("wal_store", "local",
-          description="Single-file append-only WAL store with CRC frames")
+          description="Single-file append-only WAL store with CRC frames",
+          example={"type": "local", "path": "./wal", "compress": False,
+                   "fsync": True})
 def _build_local_store(config: dict, resource=None) -> LocalWalStore:
     return LocalWalStore(
         config.get("path", "./wal"),
