@@ -138,6 +138,25 @@ class EngineConfig:
         """Structural validation against the registries; returns error list
         (reference configuration.rs:176 validate_config)."""
         from .registry import registry as _registry
+
+        def check_types(where: str, kind: str, t: str, spec: dict) -> None:
+            """Primitive-type check against the component's JSON Schema
+            (additionalProperties stays open; only declared keys checked)."""
+            md = _registry(kind).metadata.get(t)
+            schema = (md.config_schema or {}) if md else {}
+            checks = {"string": str, "integer": int, "number": (int, float),
+                      "boolean": bool, "array": list, "object": dict}
+            for key, sub in (schema.get("properties") or {}).items():
+                if key == "type" or key not in spec:
+                    continue
+                expect = checks.get(sub.get("type"))
+                val = spec[key]
+                if expect and not isinstance(val, expect) \
+                        or (expect is int and isinstance(val, bool)):
+                    errors.append(
+                        f"{where}: {kind} {t!r} field {key!r} expects "
+                        f"{sub.get('type')}, got {type(val).__name__}")
+
         errors: List[str] = []
         for s in self.streams:
             where = f"stream {s.id!r}"
@@ -152,6 +171,8 @@ class EngineConfig:
                     errors.append(f"{where}: {kind} missing 'type'")
                 elif t not in _registry(kind).builders:
                     errors.append(f"{where}: unknown {kind} type {t!r}")
+                else:
+                    check_types(where, kind, t, spec)
                 codec_spec = spec.get("codec") if isinstance(spec, dict) \
                     else None
                 if codec_spec is not None:
@@ -167,6 +188,8 @@ class EngineConfig:
                     errors.append(f"{where}: processor missing 'type'")
                 elif t not in _registry("processor").builders:
                     errors.append(f"{where}: unknown processor type {t!r}")
+                else:
+                    check_types(where, "processor", t, p)
             for t_ in s.temporary:
                 tt = t_.get("type") if isinstance(t_, dict) else None
                 if tt and tt not in _registry("temporary").builders:

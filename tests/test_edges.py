@@ -318,3 +318,18 @@ def test_every_component_example_builds(tmp_path):
                 comp.close()
             built += 1
     assert built >= 48
+
+
+def test_config_schema_type_checking():
+    from arkflow_amd.config import EngineConfig
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "s",
+        "input": {"type": "generate", "batch_size": "eight"},  # wrong type
+        "output": {"type": "drop"}}]})
+    errs = cfg.validate()
+    assert any("batch_size" in e and "integer" in e for e in errs)
+    # bool is not an acceptable integer
+    cfg2 = EngineConfig.from_dict({"streams": [{
+        "id": "s", "input": {"type": "generate", "batch_size": True},
+        "output": {"type": "drop"}}]})
+    assert any("batch_size" in e for e in cfg2.validate())
