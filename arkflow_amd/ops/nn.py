@@ -42,6 +42,14 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor,
             out = torch.nn.functional.linear(
                 x2, weight,
                 bias.to(torch.bfloat16) if bias is not None else None)
+        elif a != ACT_NONE and \
+                os.environ.get("ARKFLOW_FUSED_GEMM", "native") == "split":
+            # A/B path: hipBLASLt GEMM + our activation kernel (measured
+            # per-shape; see profiles r15)
+            out = torch.nn.functional.linear(
+                x2, weight,
+                bias.to(torch.bfloat16) if bias is not None else None)
+            out = nat.bias_act_bf16(out, None, a)
         else:
             out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
         return out.reshape(*x.shape[:-1], weight.shape[0])
