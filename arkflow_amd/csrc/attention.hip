@@ -13,12 +13,18 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define ATTN_THREADS 256
 
+// Strided layout: row s of head h in batch b lives at
+// base + ((int64_t)b*S + s)*ld + h*D. With Hd=1/ld=D this is the packed
+// [BH, S, D] layout; with Hd=H/ld=3*H*D the kernel reads q/k/v DIRECTLY out
+// of the [B,S,3,H,D] QKV-projection tensor (no transpose copies), and with
+// ldo=H*D writes O back in [B,S,H*D] so the next linear consumes it as-is.
 template <int S, int D>
 __global__ __launch_bounds__(ATTN_THREADS, 1)
-void attention_kernel(const __bf16* __restrict__ Q,  // [BH, S, D]
+void attention_kernel(const __bf16* __restrict__ Q,
                       const __bf16* __restrict__ K,
                       const __bf16* __restrict__ V,
-                      __bf16* __restrict__ O, float scale) {
+                      __bf16* __restrict__ O, float scale,
+                      int ldq, int ldo, int Hd) {
   constexpr int DP = D + 4;   // padded row strides (bank spread)
   constexpr int SP = S + 4;
   constexpr int QROWS = 32;   // q rows per wave
@@ -34,18 +40,19 @@ void attention_kernel(const __bf16* __restrict__ Q,  // [BH, S, D]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const __bf16* Qb = Q + (int64_t)bh * S * D;
-  const __bf16* Kb = K + (int64_t)bh * S * D;
-  const __bf16* Vb = V + (int64_t)bh * S * D;
-  __bf16* Ob = O + (int64_t)bh * S * D;
+  const int b = bh / Hd, h = bh % Hd;
+  const __bf16* Qb = Q + (int64_t)b * S * ldq + h * D;
+  const __bf16* Kb = K + (int64_t)b * S * ldq + h * D;
+  const __bf16* Vb = V + (int64_t)b * S * ldq + h * D;
+  __bf16* Ob = O + (int64_t)b * S * ldo + h * D;
 
   // ---- stage K and V^T (coalesced global reads) -----------------------------
   for (int i = tid; i < S * D / 2; i += ATTN_THREADS) {
     // 2 elements per thread via 32-bit loads
     int s = (i * 2) / D, d = (i * 2) % D;
-    uint32_t kv = *(const uint32_t*)(Kb + s * D + d);
+    uint32_t kv = *(const uint32_t*)(Kb + (int64_t)s * ldq + d);
     *(uint32_t*)(&K_lds[s * DP + d]) = kv;
-    uint32_t vv = *(const uint32_t*)(Vb + s * D + d);
+    uint32_t vv = *(const uint32_t*)(Vb + (int64_t)s * ldq + d);
     __bf16 v0 = ((const __bf16*)&vv)[0], v1 = ((const __bf16*)&vv)[1];
     Vt_lds[d * SP + s] = v0;
     Vt_lds[(d + 1) * SP + s] = v1;
@@ -60,8 +67,8 @@ void attention_kernel(const __bf16* __restrict__ Q,  // [BH, S, D]
   for (int am = 0; am < 2; ++am)
 #pragma unroll
     for (int ks = 0; ks < D / 32; ++ks)
-      q_frag[am][ks] =
-          *(const bf16x8*)(Qb + (q0 + am * 16 + fr) * D + ks * 32 + fk);
+      q_frag[am][ks] = *(const bf16x8*)(
+          Qb + (int64_t)(q0 + am * 16 + fr) * ldq + ks * 32 + fk);
 
   __syncthreads();  // K/Vt staged
 
@@ -144,7 +151,7 @@ void attention_kernel(const __bf16* __restrict__ Q,  // [BH, S, D]
       for (int r = 0; r < 4; ++r) {
         int row = q0 + am * 16 + (lane >> 4) * 4 + r;
         int col = nd * 16 + fr;
-        Ob[row * D + col] = (__bf16)acc2[am][nd][r];
+        Ob[(int64_t)row * ldo + col] = (__bf16)acc2[am][nd][r];
       }
 }
 
@@ -167,7 +174,32 @@ int launch_attention_bf16(const void* Q, const void* K, const void* V,
     }
     attention_kernel<SS, DD><<<BH, ATTN_THREADS, lds, st>>>(
         (const __bf16*)Q, (const __bf16*)K, (const __bf16*)V, (__bf16*)O,
-        scale);
+        scale, DD, DD, 1);
+    return 0;
+  }
+  return -1;
+}
+
+// qkv: [B, S, 3, H, D] contiguous (the QKV linear's natural output);
+// O: [B, S, H*D]. No transpose copies on either side.
+int launch_attention_qkv_bf16(const void* QKV, void* O, int B, int H, int S,
+                              int D, float scale, hipStream_t st) {
+  if (S == 128 && D == 64) {
+    constexpr int SS = 128, DD = 64;
+    size_t lds = (SS * (DD + 4) + DD * (SS + 4) + 4 * 32 * (SS + 4)) *
+                 sizeof(__bf16);
+    static bool attr_set2 = false;
+    if (!attr_set2) {
+      hipFuncSetAttribute(
+          (const void*)attention_kernel<SS, DD>,
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+      attr_set2 = true;
+    }
+    const __bf16* base = (const __bf16*)QKV;
+    int ld = 3 * H * D;
+    attention_kernel<SS, DD><<<B * H, ATTN_THREADS, lds, st>>>(
+        base, base + (int64_t)H * D, base + (int64_t)2 * H * D, (__bf16*)O,
+        scale, ld, H * D, H);
     return 0;
   }
   return -1;

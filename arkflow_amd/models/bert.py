@@ -123,11 +123,8 @@ class BertEncoder:
         x2 = x.reshape(B * S, h)
         qkv = opsnn.linear_bf16(x2, ly["wqkv"], ly["bqkv"])  # [B*S, 3h]
         qkv = qkv.reshape(B, S, 3, c.heads, c.head_dim)
-        q = qkv[:, :, 0].permute(0, 2, 1, 3).contiguous()  # [B,H,S,D]
-        k = qkv[:, :, 1].permute(0, 2, 1, 3).contiguous()
-        v = qkv[:, :, 2].permute(0, 2, 1, 3).contiguous()
-        attn = opsnn.attention_bf16(q, k, v, self._scale)   # [B,H,S,D]
-        attn = attn.permute(0, 2, 1, 3).reshape(B * S, h).contiguous()
+        # strided attention straight off the QKV tensor (no transposes)
+        attn = opsnn.attention_qkv_bf16(qkv, self._scale).reshape(B * S, h)
         proj = opsnn.linear_bf16(attn, ly["wo"], ly["bo"])
         x = opsnn.layernorm_bf16(proj.reshape(B, S, h), ly["ln1_g"],
                                  ly["ln1_b"], residual=x)

@@ -107,3 +107,22 @@ def attention_bf16(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         return torch.matmul(p, vc)
     p = torch.softmax(q.float() @ k.float().transpose(-1, -2) * scale, dim=-1)
     return (p @ v.float()).to(q.dtype)
+
+
+def attention_qkv_bf16(qkv: torch.Tensor, scale: float) -> torch.Tensor:
+    """Attention straight from the QKV projection: qkv [B,S,3,H,D] bf16 →
+    [B,S,H*D]. GPU: strided-load kernel, zero transpose copies; shapes the
+    kernel doesn't cover (or CPU) go through the permuted path."""
+    B, S, three, H, D = qkv.shape
+    if qkv.is_cuda:
+        nat = require_native()
+        try:
+            return nat.attention_qkv_bf16(qkv.contiguous(), scale)
+        except RuntimeError as e:
+            if "unsupported shape" not in str(e):
+                raise
+    q = qkv[:, :, 0].permute(0, 2, 1, 3).contiguous()
+    k = qkv[:, :, 1].permute(0, 2, 1, 3).contiguous()
+    v = qkv[:, :, 2].permute(0, 2, 1, 3).contiguous()
+    attn = attention_bf16(q, k, v, scale)
+    return attn.permute(0, 2, 1, 3).reshape(B, S, H * D)

@@ -314,6 +314,27 @@ def test_exclusive_offsets_matches_cumsum(nat, dev):
         assert torch.equal(offs, ref), n
 
 
+def test_attention_qkv_strided_matches_permuted(nat, dev):
+    """QKV-direct strided attention vs the permute+packed kernel and the
+    CPU fp32 oracle."""
+    from arkflow_amd.ops.nn import attention_bf16, attention_qkv_bf16
+    torch.manual_seed(9)
+    B, S, H, D = 4, 128, 12, 64
+    qkv = torch.randn(B, S, 3, H, D, dtype=torch.bfloat16)
+    scale = 1.0 / D ** 0.5
+    cpu = attention_qkv_bf16(qkv, scale)           # CPU permuted reference
+    gpu = attention_qkv_bf16(qkv.to(dev), scale).cpu()
+    err = (gpu.float() - cpu.float()).abs().max().item()
+    assert err < 0.05, err
+    # and vs the packed [B,H,S,D] kernel on device
+    q = qkv[:, :, 0].permute(0, 2, 1, 3).contiguous().to(dev)
+    k = qkv[:, :, 1].permute(0, 2, 1, 3).contiguous().to(dev)
+    v = qkv[:, :, 2].permute(0, 2, 1, 3).contiguous().to(dev)
+    packed = attention_bf16(q, k, v, scale).permute(0, 2, 1, 3).reshape(
+        B, S, H * D).cpu()
+    assert torch.equal(gpu, packed)
+
+
 def test_attention_other_shapes(nat, dev):
     """Shapes outside the fused kernel's tile compose GEMM + our softmax;
     values must match the CPU fp32 oracle."""
