@@ -3,7 +3,7 @@
 from __future__ import annotations
 
 import asyncio
-from typing import Optional, Tuple
+from typing import Tuple
 
 from ..batch import MessageBatch
 from ..errors import EOFError_
