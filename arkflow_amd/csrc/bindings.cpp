@@ -87,6 +87,7 @@ void launch_i64_to_ordered(const int64_t*, uint64_t*, int64_t, int,
                            hipStream_t);
 void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
                         const char*, const int*, const int*, const int*,
+                        const int*, int, const char*, const int*,
                         double*, int64_t*, int64_t*, int32_t*, uint8_t*,
                         int32_t*, hipStream_t);
 void launch_json_copy_strings(const uint8_t*, const int64_t*, const int64_t*,
@@ -628,13 +629,42 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   int64_t n = offsets.numel() - 1;
   int nf = (int)names.size();
   std::vector<char> packed(nf * 24, 0);
-  std::vector<int> nl(nf), kd(nf), sl(nf);
+  std::vector<int> nl(nf), kd(nf), sl(nf), par(nf);
+  // dotted names ("user.id") address one level of nesting: collect the
+  // distinct parent segments, scope each field to its parent index
+  std::vector<std::string> parents;
   for (int f = 0; f < nf; ++f) {
-    TORCH_CHECK(names[f].size() < 24, "json field name too long");
-    memcpy(&packed[f * 24], names[f].data(), names[f].size());
-    nl[f] = (int)names[f].size();
+    std::string leaf = names[f];
+    par[f] = -1;
+    auto dot = leaf.find('.');
+    if (dot != std::string::npos) {
+      std::string parent = leaf.substr(0, dot);
+      leaf = leaf.substr(dot + 1);
+      TORCH_CHECK(leaf.find('.') == std::string::npos,
+                  "json paths support one nesting level: ", names[f]);
+      int pi = -1;
+      for (size_t q = 0; q < parents.size(); ++q)
+        if (parents[q] == parent) pi = (int)q;
+      if (pi < 0) {
+        parents.push_back(parent);
+        pi = (int)parents.size() - 1;
+      }
+      par[f] = pi;
+    }
+    TORCH_CHECK(leaf.size() < 24, "json field name too long");
+    memcpy(&packed[f * 24], leaf.data(), leaf.size());
+    nl[f] = (int)leaf.size();
     kd[f] = (int)kind[f];
     sl[f] = (int)slot[f];
+  }
+  TORCH_CHECK(parents.size() <= 8, "too many nested parents (max 8)");
+  int np = (int)parents.size();
+  std::vector<char> ppacked(std::max(np, 1) * 24, 0);
+  std::vector<int> pl(std::max(np, 1), 0);
+  for (int q = 0; q < np; ++q) {
+    TORCH_CHECK(parents[q].size() < 24, "json parent name too long");
+    memcpy(&ppacked[q * 24], parents[q].data(), parents[q].size());
+    pl[q] = (int)parents[q].size();
   }
   auto out_f = torch::zeros({std::max<int64_t>(n_float, 1),
                              std::max<int64_t>(n, 1)},
@@ -654,6 +684,7 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   if (n > 0)
     launch_json_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
                        n, nf, packed.data(), nl.data(), kd.data(), sl.data(),
+                       par.data(), np, ppacked.data(), pl.data(),
                        out_f.data_ptr<double>(), out_i.data_ptr<int64_t>(),
                        str_start.data_ptr<int64_t>(),
                        str_ulen.data_ptr<int32_t>(),

@@ -9,12 +9,18 @@
 #define JSON_MAX_FIELDS 16
 #define JSON_MAX_NAME 24
 
+#define JSON_MAX_PARENTS 8
+
 struct JsonSpec {
   int nf;
   char names[JSON_MAX_FIELDS][JSON_MAX_NAME];
   int name_len[JSON_MAX_FIELDS];
   int kind[JSON_MAX_FIELDS];  // 0 → out_i64 (ints+bools), 1 → out_f64, 2 → str
   int slot[JSON_MAX_FIELDS];
+  int parent[JSON_MAX_FIELDS];  // -1 = top level, else index into parents
+  int np;
+  char parents[JSON_MAX_PARENTS][JSON_MAX_NAME];
+  int parent_len[JSON_MAX_PARENTS];
 };
 
 DEV_INLINE bool is_ws(uint8_t c) {
@@ -228,9 +234,16 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
       continue;
     }
     ++p;
+    int ctx = -1;  // current parent-object context (-1 = top level)
     for (;;) {
       while (p < end && (is_ws(data[p]) || data[p] == ',')) ++p;
-      if (p >= end || data[p] == '}') break;
+      if (p >= end) break;
+      if (data[p] == '}') {
+        if (ctx < 0) break;  // end of document
+        ctx = -1;            // pop out of the nested object
+        ++p;
+        continue;
+      }
       if (data[p] != '"') { err[0] = 1; break; }
       // key
       int64_t k0 = ++p;
@@ -243,11 +256,12 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
       while (p < end && is_ws(data[p])) ++p;
       if (p < end && data[p] == ':') ++p;
       while (p < end && is_ws(data[p])) ++p;
-      // match against spec
+      // match against spec (fields are scoped to the current context)
       int fi = -1;
 #pragma unroll
       for (int f = 0; f < JSON_MAX_FIELDS; ++f) {
-        if (f < spec.nf && spec.name_len[f] == klen) {
+        if (f < spec.nf && spec.name_len[f] == klen
+            && spec.parent[f] == ctx) {
           bool eq = true;
           for (int c = 0; c < klen; ++c)
             if (spec.names[f][c] != (char)data[k0 + c]) { eq = false; break; }
@@ -255,6 +269,26 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
         }
       }
       if (fi < 0) {
+        // at top level, a key naming a parent object descends one level
+        if (ctx < 0 && p < end && data[p] == '{') {
+          int pi = -1;
+          for (int q = 0; q < spec.np; ++q) {
+            if (spec.parent_len[q] == klen) {
+              bool eq = true;
+              for (int c = 0; c < klen; ++c)
+                if (spec.parents[q][c] != (char)data[k0 + c]) {
+                  eq = false;
+                  break;
+                }
+              if (eq) { pi = q; break; }
+            }
+          }
+          if (pi >= 0) {
+            ctx = pi;
+            ++p;  // enter the object
+            continue;
+          }
+        }
         p = skip_value(data, p, end);
         continue;
       }
@@ -308,7 +342,9 @@ __global__ void json_copy_strings_kernel(const uint8_t* __restrict__ data,
 extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
                                    int64_t n_docs, int nf, const char* names,
                                    const int* name_len, const int* kind,
-                                   const int* slot, double* out_f64,
+                                   const int* slot, const int* parent, int np,
+                                   const char* parents, const int* parent_len,
+                                   double* out_f64,
                                    int64_t* out_i64, int64_t* str_start,
                                    int32_t* str_ulen, uint8_t* found,
                                    int32_t* err, hipStream_t st) {
@@ -318,8 +354,15 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
     spec.name_len[f] = name_len[f];
     spec.kind[f] = kind[f];
     spec.slot[f] = slot[f];
+    spec.parent[f] = parent[f];
     for (int c = 0; c < name_len[f] && c < JSON_MAX_NAME; ++c)
       spec.names[f][c] = names[f * JSON_MAX_NAME + c];
+  }
+  spec.np = np > JSON_MAX_PARENTS ? JSON_MAX_PARENTS : np;
+  for (int q = 0; q < spec.np; ++q) {
+    spec.parent_len[q] = parent_len[q];
+    for (int c = 0; c < parent_len[q] && c < JSON_MAX_NAME; ++c)
+      spec.parents[q][c] = parents[q * JSON_MAX_NAME + c];
   }
   int grid = (int)((n_docs + 255) / 256);
   if (grid > 2048) grid = 2048;

@@ -81,7 +81,8 @@ class JsonToArrowProcessor(Processor):
         self.device = torch.device(dev) if dev else getattr(
             resource, "device", None)
         self.keep_meta = bool(config.get("keep_meta", True))
-        # fixed-schema GPU fast path: {name: float|int|bool|str} decodes
+        # fixed-schema GPU fast path: {name: float|int|bool|str} decodes;
+        # dotted names ("user.id") extract one level of nesting on-device
         # on-device (csrc/json_decode.hip: scalar extraction + two-pass string
         # copy-out with escape/\uXXXX handling); no schema → host pyarrow
         # parse + inference

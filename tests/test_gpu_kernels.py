@@ -573,6 +573,43 @@ def test_json_decode_gpu_strings(nat, dev):
         assert vs[i] == doc["v"]
 
 
+def test_json_decode_gpu_nested_paths(nat, dev):
+    """Dotted schema names extract one level of nesting on-device."""
+    import asyncio
+    import json as _json
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+    payloads = []
+    docs = []
+    for i in range(4_000):
+        doc = {"id": i,
+               "user": {"id": i * 7, "name": f"u{i}", "skip": [1, 2]},
+               "metrics": {"score": i * 0.5},
+               "other": {"id": -1}}
+        if i % 9 == 0:
+            del doc["user"]
+        docs.append(doc)
+        payloads.append(_json.dumps(doc).encode())
+    batch = MessageBatch.from_binary(payloads).to(dev)
+    proc = JsonToArrowProcessor({"schema": {
+        "id": "int", "user.id": "int", "user.name": "str",
+        "metrics.score": "float"}}, None)
+    out = asyncio.new_event_loop().run_until_complete(proc.process(batch))[0]
+    uid = out.column("user.id").to_pylist()
+    uname = out.column("user.name").to_pylist()
+    score = out.column("metrics.score").to_pylist()
+    for i in (0, 1, 9, 3999):
+        assert out.column("id").to_pylist()[i] == i
+        if "user" in docs[i]:
+            assert uid[i] == docs[i]["user"]["id"], i
+            got = uname[i].decode() if isinstance(uname[i], bytes) \
+                else uname[i]
+            assert got == docs[i]["user"]["name"], i
+        else:
+            assert uid[i] is None and uname[i] is None, i
+        assert abs(score[i] - docs[i]["metrics"]["score"]) < 1e-6
+
+
 def test_fused_filter_gather_matches_slow_path(nat, dev):
     from arkflow_amd.batch import MessageBatch
     from arkflow_amd.processors.sql import SqlProcessor
