@@ -114,13 +114,56 @@ def repartition_by_key(batch: MessageBatch, key_column: str
     dist.all_gather(size_mat, in_t)
     out_splits = [int(size_mat[src][rank()].item()) for src in range(w)]
 
+    # column-type consensus: a rank with an EMPTY shard cannot know a
+    # column's kind/dtype (an empty list types as numeric float64), and gloo/
+    # RCCL collectives hang or kill peers if ranks disagree on the exchange
+    # protocol. One small all_gather settles (kind, dtype) per column.
+    DT = [torch.float64, torch.float32, torch.bfloat16, torch.float16,
+          torch.int64, torch.int32, torch.int16, torch.int8, torch.uint8,
+          torch.bool]
+    names = list(reordered.columns)
+    local = []
+    for name in names:
+        col = reordered.columns[name]
+        kind = 1 if col.kind == "binary" else 0
+        dtc = -1 if kind else DT.index(col.data.dtype)
+        local += [kind, dtc, 1 if len(col) else 0]
+    lt = torch.tensor(local, dtype=torch.int64)
+    gathered_meta = [torch.zeros_like(lt) for _ in range(w)]
+    dist.all_gather(gathered_meta, lt)
+
     cols: Dict[str, Column] = {}
-    for name, col in reordered.columns.items():
-        if col.kind == "numeric":
-            data = _exchange_numeric(col.data, in_splits, out_splits)
-            cols[name] = Column("numeric", data)
-        else:
+    for ci, name in enumerate(names):
+        col = reordered.columns[name]
+        nonempty = [g for g in gathered_meta if int(g[3 * ci + 2])]
+        is_binary = any(int(g[3 * ci]) == 1 for g in nonempty) or (
+            not nonempty and col.kind == "binary")
+        if is_binary:
+            if col.kind != "binary":
+                if len(col) != 0:
+                    raise ValueError(f"repartition: column {name!r} kind "
+                                     "differs across ranks")
+                col = Column("binary", torch.empty(0, dtype=torch.uint8),
+                             torch.zeros(1, dtype=torch.int64))
             cols[name] = _exchange_binary(col, in_splits, out_splits, w)
+            continue
+        # symmetric fold over NON-EMPTY ranks' dtypes (identical on every
+        # rank) so all ranks pick the same wire dtype
+        dts = [DT[int(g[3 * ci + 1])] for g in nonempty
+               if int(g[3 * ci]) == 0]
+        if dts:
+            target = dts[0]
+            for dt in dts[1:]:
+                target = torch.promote_types(target, dt)
+        else:
+            target = col.data.dtype if col.kind == "numeric" \
+                else torch.float64
+        if col.kind != "numeric":  # binary-empty vs numeric consensus
+            data = torch.empty(0, dtype=target)
+        else:
+            data = col.data.to(target)
+        data = _exchange_numeric(data, in_splits, out_splits)
+        cols[name] = Column("numeric", data)
     return MessageBatch(cols, batch.input_name)
 
 

@@ -129,3 +129,36 @@ def test_engine_sharded_session_agg():
                  join=True)
         # both ranks produced grouped outputs; union of sessions ≤ 64 per batch
         assert results[0] > 0 and results[1] > 0
+
+
+def _run_repartition_w4(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.parallel.dist import repartition_by_key
+
+    # uneven shards, one EMPTY rank — zero-length all_to_all splits must work
+    n = 0 if rank == 3 else 37 * (rank + 1)
+    batch = MessageBatch.from_dict({
+        "k": torch.arange(rank * 1000, rank * 1000 + n, dtype=torch.int64),
+        "v": torch.arange(n, dtype=torch.float32),
+        "s": [f"r{rank}-{i}" for i in range(n)],
+    })
+    out = repartition_by_key(batch, "k")
+    assert out.column("k").data.numel() == out.num_rows
+    assert len(out.column("s")) == out.num_rows
+    results[rank] = out.num_rows
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_repartition_four_ranks_uneven():
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_run_repartition_w4, args=(4, 29534, results), nprocs=4,
+                 join=True)
+        assert sum(results.values()) == 37 * (1 + 2 + 3)  # conservation
