@@ -108,9 +108,12 @@ def repartition_by_key(batch: MessageBatch, key_column: str
     in_splits = torch.bincount(sorted_dest, minlength=w).tolist()
     reordered = batch.take(order)
 
-    # exchange split sizes
-    in_t = torch.tensor(in_splits, dtype=torch.int64)
-    size_mat = [torch.zeros(w, dtype=torch.int64) for _ in range(w)]
+    # exchange split sizes (device tensors on nccl/RCCL, host on gloo)
+    split_dev = torch.device("cuda") if dist.get_backend() == "nccl" \
+        else torch.device("cpu")
+    in_t = torch.tensor(in_splits, dtype=torch.int64, device=split_dev)
+    size_mat = [torch.zeros(w, dtype=torch.int64, device=split_dev)
+                for _ in range(w)]
     dist.all_gather(size_mat, in_t)
     out_splits = [int(size_mat[src][rank()].item()) for src in range(w)]
 
@@ -128,9 +131,12 @@ def repartition_by_key(batch: MessageBatch, key_column: str
         kind = 1 if col.kind == "binary" else 0
         dtc = -1 if kind else DT.index(col.data.dtype)
         local += [kind, dtc, 1 if len(col) else 0]
-    lt = torch.tensor(local, dtype=torch.int64)
+    meta_dev = torch.device("cuda") if dist.get_backend() == "nccl" \
+        else torch.device("cpu")  # nccl/RCCL collectives need device tensors
+    lt = torch.tensor(local, dtype=torch.int64, device=meta_dev)
     gathered_meta = [torch.zeros_like(lt) for _ in range(w)]
     dist.all_gather(gathered_meta, lt)
+    gathered_meta = [g.cpu() for g in gathered_meta]
 
     cols: Dict[str, Column] = {}
     for ci, name in enumerate(names):
