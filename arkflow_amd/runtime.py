@@ -163,6 +163,7 @@ class RuntimeEntry:
         self.cancel: Optional[asyncio.Event] = None
         self.last_error: Optional[str] = None
         self.failure_class: Optional[FailureClass] = None
+        self.stream = None  # live Stream while running (embedding/test access)
 
     @property
     def convergence(self) -> ConvergenceState:
@@ -235,6 +236,7 @@ class RuntimeManager:
             self.events.push(stream_id, "failed", str(e))
             raise
         entry.metrics = stream.metrics
+        entry.stream = stream
         entry.cancel = asyncio.Event()
         entry.task = asyncio.ensure_future(
             self._supervise(entry, stream))

@@ -634,6 +634,78 @@ def test_fused_filter_gather_matches_slow_path(nat, dev):
     assert torch.equal(fast.column("f1").data, slow.column("f1").data)
 
 
+def test_engine_json_strings_like_on_gpu(dev):
+    """Full engine: memory input of JSON payloads → device schema decode
+    (string + nested fields) → SQL LIKE filter + group agg → memory output,
+    columns device-resident between stages."""
+    import asyncio
+    import json as _json
+    import arkflow_amd as af
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.config import EngineConfig
+
+    cfg = EngineConfig.from_dict({
+        "streams": [{
+            "id": "js-gpu",
+            "device": "cuda:0",
+            "input": {"type": "memory"},
+            "pipeline": {"thread_num": 1, "processors": [
+                {"type": "json_to_arrow",
+                 "schema": {"region": "str", "user.tier": "int",
+                            "amount": "float"}},
+                {"type": "sql",
+                 "query": "SELECT count(*) AS n, sum(amount) AS total "
+                          "FROM flow WHERE region LIKE 'eu-%' "
+                          "GROUP BY \"user.tier\""},
+            ]},
+            "output": {"type": "memory"},
+        }]
+    })
+    assert not cfg.validate()
+    eng = af.Engine(cfg)
+
+    async def main():
+        payloads = []
+        expect_n = 0
+        expect_total = 0.0
+        for i in range(50_000):
+            region = ["eu-west", "eu-north", "us-east"][i % 3]
+            doc = {"region": region, "user": {"tier": i % 4},
+                   "amount": round(i * 0.01, 2)}
+            if region.startswith("eu-"):
+                expect_n += 1
+                expect_total += doc["amount"]
+            payloads.append(_json.dumps(doc).encode())
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(eng.run_with_cancellation(cancel))
+        for _ in range(100):
+            await asyncio.sleep(0.05)
+            try:
+                stream = eng.runtime.get("js-gpu").stream
+                if stream is not None:
+                    break
+            except Exception:
+                pass
+        for c0 in range(0, len(payloads), 8192):
+            stream.input.push(MessageBatch.from_binary(
+                payloads[c0:c0 + 8192], input_name="memory"))
+        stream.input.finish()
+        for _ in range(200):  # engine stops at EOF
+            if eng.runtime.get("js-gpu").state.value == "stopped":
+                break
+            await asyncio.sleep(0.1)
+        cancel.set()
+        await asyncio.wait_for(task, 60)
+        out = stream.output
+        n = sum(int(r["n"]) for b in out.batches for r in b.to_rows())
+        total = sum(float(r["total"])
+                    for b in out.batches for r in b.to_rows())
+        assert n == expect_n, (n, expect_n)
+        assert abs(total - expect_total) / max(expect_total, 1) < 1e-4
+
+    asyncio.new_event_loop().run_until_complete(main())
+
+
 def test_engine_window_ring_on_gpu(dev):
     """Windowed engine on device: tumbling window emits zero-copy ring views
     feeding a GROUP BY — full async engine."""
