@@ -137,6 +137,47 @@ class JsonToArrowProcessor(Processor):
             cols[name] = Column("numeric", data, validity=v)
         return MessageBatch(cols)
 
+    def _decode_host_schema(self, col) -> MessageBatch:
+        """CPU fallback for the fixed-schema path: same output columns
+        (incl. dotted nested paths) as the GPU kernel."""
+        import json as _json
+        vals = {name: [] for name in self.schema}
+        for payload in col.to_pylist():
+            try:
+                doc = _json.loads(payload)
+            except Exception:  # noqa: BLE001
+                raise ProcessError("json decode error (malformed document)")
+            for name in self.schema:
+                cur = doc
+                for part in name.split("."):
+                    cur = cur.get(part) if isinstance(cur, dict) else None
+                    if cur is None:
+                        break
+                vals[name].append(cur)
+        cols = {}
+        for name, t in self.schema.items():
+            v = vals[name]
+            miss = [x is None or (t not in ("str", "string")
+                                  and isinstance(x, str)) for x in v]
+            if t in ("str", "string"):
+                c = Column.from_strings(
+                    ["" if m else x for m, x in zip(miss, v)])
+            elif t == "float":
+                c = Column.from_numeric(torch.tensor(
+                    [0.0 if m else float(x) for m, x in zip(miss, v)],
+                    dtype=torch.float64))
+            elif t == "bool":
+                c = Column.from_numeric(torch.tensor(
+                    [False if m else bool(x) for m, x in zip(miss, v)]))
+            else:
+                c = Column.from_numeric(torch.tensor(
+                    [0 if m else int(x) for m, x in zip(miss, v)],
+                    dtype=torch.int64))
+            if any(miss):
+                c.validity = torch.tensor([not m for m in miss])
+            cols[name] = c
+        return MessageBatch(cols)
+
     async def process(self, batch: MessageBatch) -> List[MessageBatch]:
         if batch.num_rows == 0:
             return []
@@ -146,6 +187,10 @@ class JsonToArrowProcessor(Processor):
                 f"json_to_arrow: no binary column {self.value_field!r}")
         if self.schema and col.data.is_cuda:
             out = self._decode_gpu(col)
+        elif self.schema:
+            out = self._decode_host_schema(col)
+            if self.device is not None:
+                out = out.to(self.device)
         else:
             out = json_payloads_to_columns(col.to_pylist(), self.projection,
                                            self.device)

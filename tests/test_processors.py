@@ -372,3 +372,24 @@ def test_vrl_statement_coalesce_and_errors():
     assert a == [("v", "coalesce(maybe, 0)")]
     with pytest.raises(ConfigError):
         ExprProcessor({"statement": "if .a > 1 { .b = 2 }"})
+
+
+def test_json_schema_host_fallback_nested(run):
+    """Fixed-schema host decode produces the same columns (incl. dotted
+    nested paths and validity) as the GPU kernel path."""
+    import json as _json
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+
+    async def main():
+        docs = [{"region": "eu", "user": {"tier": 2}, "amount": 1.5},
+                {"region": 7, "amount": 2.0},  # wrong type + missing nested
+                {"user": {"tier": 0}, "amount": 3.0}]
+        b = MessageBatch.from_binary([_json.dumps(d).encode() for d in docs])
+        proc = JsonToArrowProcessor({"schema": {
+            "region": "str", "user.tier": "int", "amount": "float"}})
+        out = (await proc.process(b))[0]
+        assert out.column("region").to_pylist() == ["eu", None, None]
+        assert out.column("user.tier").to_pylist() == [2, None, 0]
+        assert out.column("amount").to_pylist() == [1.5, 2.0, 3.0]
+
+    run(main())
