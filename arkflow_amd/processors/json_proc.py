@@ -157,8 +157,11 @@ class JsonToArrowProcessor(Processor):
         cols = {}
         for name, t in self.schema.items():
             v = vals[name]
-            miss = [x is None or (t not in ("str", "string")
-                                  and isinstance(x, str)) for x in v]
+            if t in ("str", "string"):
+                miss = [not isinstance(x, str) for x in v]
+            else:  # numeric schema: non-scalar values → absent (GPU parity)
+                miss = [x is None or isinstance(x, (str, dict, list))
+                        for x in v]
             if t in ("str", "string"):
                 c = Column.from_strings(
                     ["" if m else x for m, x in zip(miss, v)])
