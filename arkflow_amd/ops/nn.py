@@ -11,6 +11,22 @@ import torch
 from . import require_native
 
 ACT_NONE, ACT_RELU, ACT_GELU, ACT_SILU = 0, 1, 2, 3
+
+try:
+    from torch.utils.weak import WeakTensorKeyDictionary
+    _bias_bf16_cache = WeakTensorKeyDictionary()
+except ImportError:  # pragma: no cover
+    _bias_bf16_cache = {}
+
+
+def _bias_bf16(bias: torch.Tensor) -> torch.Tensor:
+    """Cached fp32→bf16 bias cast: inference weights are static, and a
+    per-call cast is a kernel that replays inside every captured graph."""
+    c = _bias_bf16_cache.get(bias)
+    if c is None:
+        c = bias.to(torch.bfloat16)
+        _bias_bf16_cache[bias] = c
+    return c
 _ACTS = {"none": ACT_NONE, "relu": ACT_RELU, "gelu": ACT_GELU,
          "silu": ACT_SILU}
 
@@ -40,15 +56,13 @@ def linear_bf16(x: torch.Tensor, weight: torch.Tensor,
         elif a == ACT_NONE and \
                 os.environ.get("ARKFLOW_PLAIN_GEMM", "blaslt") != "native":
             out = torch.nn.functional.linear(
-                x2, weight,
-                bias.to(torch.bfloat16) if bias is not None else None)
+                x2, weight, _bias_bf16(bias) if bias is not None else None)
         elif a != ACT_NONE and \
                 os.environ.get("ARKFLOW_FUSED_GEMM", "native") == "split":
             # A/B path: hipBLASLt GEMM + our activation kernel (measured
             # per-shape; see profiles r15)
             out = torch.nn.functional.linear(
-                x2, weight,
-                bias.to(torch.bfloat16) if bias is not None else None)
+                x2, weight, _bias_bf16(bias) if bias is not None else None)
             out = nat.bias_act_bf16(out, None, a)
         else:
             out = nat.gemm_bf16(x2, weight.contiguous(), bias, a)
