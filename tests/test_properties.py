@@ -98,3 +98,39 @@ def test_sql_expression_differential_vs_sqlite(expr, cmp, bound, seed):
                     for r in conn.execute(sql).fetchall())
     conn.close()
     assert ours == theirs, sql
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(
+    st.fixed_dictionaries({
+        "s": st.one_of(st.text(max_size=30), st.integers(), st.none()),
+        "n": st.one_of(st.integers(min_value=-10**12, max_value=10**12),
+                       st.floats(allow_nan=False, allow_infinity=False),
+                       st.text(max_size=3), st.none()),
+        "b": st.one_of(st.booleans(), st.none()),
+        "extra": st.dictionaries(st.text(min_size=1, max_size=5),
+                                 st.integers(), max_size=2),
+    }), min_size=1, max_size=20))
+def test_json_schema_host_decode_matches_jsonloads(docs):
+    """The fixed-schema host decoder (the GPU kernel's oracle) must agree
+    with json.loads semantics: wrong-typed / missing → null."""
+    import asyncio
+    import json as _json
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+    payloads = [_json.dumps(d).encode() for d in docs]
+    b = MessageBatch.from_binary(payloads)
+    proc = JsonToArrowProcessor(
+        {"schema": {"s": "str", "n": "float", "b": "bool"}})
+    out = asyncio.new_event_loop().run_until_complete(proc.process(b))[0]
+    s = out.column("s").to_pylist()
+    n = out.column("n").to_pylist()
+    bb = out.column("b").to_pylist()
+    for i, d in enumerate(docs):
+        exp_s = d["s"] if isinstance(d["s"], str) else None
+        got_s = s[i].decode() if isinstance(s[i], bytes) else s[i]
+        assert got_s == exp_s
+        if isinstance(d["n"], (int, float)) and not isinstance(d["n"], bool):
+            assert n[i] == pytest.approx(float(d["n"]))
+        else:
+            assert n[i] is None
+        assert bb[i] == (d["b"] if isinstance(d["b"], bool) else None)
