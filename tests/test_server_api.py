@@ -244,3 +244,24 @@ def test_stream_metrics_and_version_detail(run):
         await eng.runtime.stop_all()
 
     run(main(), timeout=30)
+
+
+def test_console_pages_render(run):
+    async def main():
+        eng = _engine()
+        app = create_app(eng)
+        async with httpx.AsyncClient(
+                transport=httpx.ASGITransport(app=app),
+                base_url="http://n") as c:
+            r = await c.get("/")
+            assert r.status_code == 200
+            assert "arkflow" in r.text.lower()
+        from arkflow_amd.server.hub import Hub, create_hub_app
+        happ = create_hub_app(Hub())
+        async with httpx.AsyncClient(
+                transport=httpx.ASGITransport(app=happ),
+                base_url="http://h") as c:
+            r = await c.get("/")
+            assert r.status_code == 200 and "console" in r.text.lower()
+
+    run(main())
