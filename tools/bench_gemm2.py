@@ -70,7 +70,11 @@ for (M, N, K) in [(8192, 2304, 768), (8192, 3072, 768), (8192, 768, 768),
     t0v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 0))
     t2v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 2))
     t5v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 5))
+    try:
+        t3v = bench(lambda: nat.gemm_bf16_variant(A, Bt, None, 0, 3))
+    except RuntimeError:
+        t3v = float("inf")
     tt = bench(lambda: A @ B)
     print(f"M{M} N{N} K{K}: 128²={fl/t0v/1e12:6.1f}TF  "
           f"8p256+swz={fl/t2v/1e12:6.1f}TF  8p128+swz={fl/t5v/1e12:6.1f}TF  "
-          f"torch={fl/tt/1e12:6.1f}TF", flush=True)
+          f"2p={fl/t3v/1e12:6.1f}TF  torch={fl/tt/1e12:6.1f}TF", flush=True)
