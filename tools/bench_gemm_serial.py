@@ -43,3 +43,27 @@ for v, name in ((0, "1282"), (3, "2p"), (5, "8p128+swz")):
     ser = bench(v, serial=True)
     print(f"{name:10s} independent={free:6.1f} TF   serialized={ser:6.1f} TF"
           f"   ratio={ser/free:.2f}", flush=True)
+
+
+# cold-weights variant: rotate 12 different Bt tensors (one per BERT layer)
+# so weights are L2-cold like the real in-context run
+Bts = [torch.randn(N, K, device=dev, dtype=torch.bfloat16) for _ in range(12)]
+As = [torch.randn(M, K, device=dev, dtype=torch.bfloat16) for _ in range(4)]
+
+
+def bench_cold(variant, iters=36):
+    def one(i):
+        nat.gemm_bf16_variant(As[i % 4], Bts[i % 12], None, 0, variant)
+    for i in range(6):
+        one(i)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(iters):
+        one(i)
+    torch.cuda.synchronize()
+    return fl / ((time.perf_counter() - t0) / iters) / 1e12
+
+
+print("-- cold weights (12 rotating Bt, 4 rotating A) --", flush=True)
+for v, name in ((0, "1282"), (3, "2p"), (5, "8p128+swz")):
+    print(f"{name:10s} cold={bench_cold(v):6.1f} TF", flush=True)
