@@ -67,3 +67,25 @@ def bench_cold(variant, iters=36):
 print("-- cold weights (12 rotating Bt, 4 rotating A) --", flush=True)
 for v, name in ((0, "1282"), (3, "2p"), (5, "8p128+swz")):
     print(f"{name:10s} cold={bench_cold(v):6.1f} TF", flush=True)
+
+
+# epilogue variant: bias + GELU (the ACTUAL fc1 configuration in-context)
+bias = torch.randn(N, device=dev)
+
+
+def bench_act(variant, iters=30):
+    def one():
+        nat.gemm_bf16_variant(A, Bt, bias, 2, variant)
+    for _ in range(5):
+        one()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        one()
+    torch.cuda.synchronize()
+    return fl / ((time.perf_counter() - t0) / iters) / 1e12
+
+
+print("-- bias + GELU epilogue (real fc1 config) --", flush=True)
+for v, name in ((0, "1282"), (3, "2p"), (5, "8p128+swz")):
+    print(f"{name:10s} gelu={bench_act(v):6.1f} TF", flush=True)
