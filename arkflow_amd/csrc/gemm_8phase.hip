@@ -37,7 +37,9 @@ DEV_INLINE float apply_act8(float x, int act) {
     case A8_RELU: return fmaxf(x, 0.f);
     case A8_GELU: {
       float c = 0.7978845608028654f * (x + 0.044715f * x * x * x);
-      return 0.5f * x * (1.f + tanhf(c));
+      c = fminf(fmaxf(c, -10.f), 10.f);  // saturate: no overflow in expf
+      float e = __expf(2.f * c);
+      return 0.5f * x * (1.f + (e - 1.f) / (e + 1.f));
     }
     case A8_SILU: return x / (1.f + __expf(-x));
     default: return x;

@@ -136,7 +136,9 @@ __global__ void bias_act_bf16_kernel(const __bf16* __restrict__ x,
       case 1: v = fmaxf(v, 0.f); break;
       case 2: {
         float c = 0.7978845608028654f * (v + 0.044715f * v * v * v);
-        v = 0.5f * v * (1.f + tanhf(c));
+        c = fminf(fmaxf(c, -10.f), 10.f);
+        float e = __expf(2.f * c);
+        v = 0.5f * v * (1.f + (e - 1.f) / (e + 1.f));
         break;
       }
       case 3: v = v / (1.f + __expf(-v)); break;
