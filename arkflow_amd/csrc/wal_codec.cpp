@@ -110,6 +110,45 @@ PyObject* encode_frames(PyObject*, PyObject* args) {
   return out;
 }
 
+// encode_frame_parts(seq, [part_buffers...]) -> one raw 'R' frame, packing
+// the scatter-gather parts in a single pass (no intermediate join copy)
+PyObject* encode_frame_parts(PyObject*, PyObject* args) {
+  unsigned long long seq;
+  PyObject* parts;
+  if (!PyArg_ParseTuple(args, "KO", &seq, &parts)) return nullptr;
+  PyObject* fast = PySequence_Fast(parts, "parts must be a sequence");
+  if (!fast) return nullptr;
+  Py_ssize_t np = PySequence_Fast_GET_SIZE(fast);
+  std::vector<Py_buffer> bufs(np);
+  size_t body = 0;
+  for (Py_ssize_t i = 0; i < np; i++) {
+    if (PyObject_GetBuffer(PySequence_Fast_GET_ITEM(fast, i), &bufs[i],
+                           PyBUF_SIMPLE) < 0) {
+      for (Py_ssize_t j = 0; j < i; j++) PyBuffer_Release(&bufs[j]);
+      Py_DECREF(fast);
+      return nullptr;
+    }
+    body += (size_t)bufs[i].len;
+  }
+  PyObject* out = PyBytes_FromStringAndSize(nullptr,
+                                            (Py_ssize_t)(12 + 1 + body + 4));
+  if (out) {
+    uint8_t* w = (uint8_t*)PyBytes_AS_STRING(out);
+    put_u64be(w, seq);
+    put_u32be(w + 8, (uint32_t)(body + 1));
+    w[12] = 'R';
+    uint8_t* p = w + 13;
+    for (Py_ssize_t i = 0; i < np; i++) {
+      std::memcpy(p, bufs[i].buf, (size_t)bufs[i].len);
+      p += bufs[i].len;
+    }
+    put_u32be(p, crc32_sb8(w + 12, body + 1, 0));
+  }
+  for (Py_ssize_t i = 0; i < np; i++) PyBuffer_Release(&bufs[i]);
+  Py_DECREF(fast);
+  return out;
+}
+
 // decode_frames(buf) -> list[(seq, tag:int, body_bytes)], torn-tail truncated
 PyObject* decode_frames(PyObject*, PyObject* args) {
   Py_buffer buf;
@@ -152,6 +191,8 @@ PyObject* crc32_py(PyObject*, PyObject* args) {
 PyMethodDef methods[] = {
     {"encode_frames", encode_frames, METH_VARARGS,
      "encode [(seq, payload)] into raw CRC frames"},
+    {"encode_frame_parts", encode_frame_parts, METH_VARARGS,
+     "encode one frame from scatter-gather parts (single-pass pack+crc)"},
     {"decode_frames", decode_frames, METH_VARARGS,
      "decode frames -> [(seq, tag, body)] with torn-tail truncation"},
     {"crc32", crc32_py, METH_VARARGS, "zlib-compatible slice-by-8 CRC32"},

@@ -57,6 +57,13 @@ def _tensor_from(data: bytes, dtype: torch.dtype, n: int) -> torch.Tensor:
 
 
 def serialize_batch(batch: MessageBatch) -> bytes:
+    return b"".join(batch_parts(batch))
+
+
+def batch_parts(batch: MessageBatch) -> List[bytes]:
+    """Scatter-gather serialization parts (zero-copy memoryviews for the
+    tensor buffers); the native codec frames them without an intermediate
+    join (csrc/wal_codec.cpp encode_frame_parts)."""
     parts: List[bytes] = [_MAGIC]
     name = (batch.input_name or "").encode()
     parts.append(struct.pack(">H", len(name)))
@@ -86,7 +93,7 @@ def serialize_batch(batch: MessageBatch) -> bytes:
             v = _tensor_bytes(col.validity)
             parts.append(struct.pack(">I", v.nbytes))
             parts.append(v)
-    return b"".join(parts)
+    return parts
 
 
 def deserialize_batch(buf: bytes) -> MessageBatch:
@@ -248,6 +255,15 @@ class LocalWalStore:
                      sync: bool = True) -> None:
         with self._lock:
             self._append_locked(entries, sync)
+
+    def append_framed(self, frames: List[bytes], sync: bool = True) -> None:
+        """Append pre-framed entries (native parts-based framing path)."""
+        with self._lock:
+            for f in frames:
+                self._f.write(f)
+            self._f.flush()
+            if sync and self.fsync:
+                os.fsync(self._f.fileno())
 
     def _append_locked(self, entries: List[Tuple[int, bytes]],
                        sync: bool) -> None:

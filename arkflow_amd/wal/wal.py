@@ -16,7 +16,7 @@ from typing import AsyncIterator, List, Optional, Tuple
 from ..config import DurabilityConfig
 from ..registry import build_component
 from ..spi import Ack
-from .store import deserialize_batch, serialize_batch
+from .store import _nwal, batch_parts, deserialize_batch, serialize_batch
 
 
 class WalAck(Ack):
@@ -118,8 +118,17 @@ class Wal:
             return
         staged, self._pending = self._pending, []
         loop = asyncio.get_running_loop()
+        fast = getattr(self.store, "append_framed", None) \
+            if _nwal is not None and not getattr(self.store, "compress",
+                                                 False) else None
 
         def work():
+            if fast is not None:
+                # single-pass native framing straight from serialization
+                # parts — no intermediate payload join
+                fast([_nwal.encode_frame_parts(seq, batch_parts(b))
+                      for seq, b in staged], True)
+                return
             entries = [(seq, serialize_batch(b)) for seq, b in staged]
             self.store.append_batch(entries, True)
 
