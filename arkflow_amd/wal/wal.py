@@ -77,10 +77,19 @@ class Wal:
         seq = self._seq
         self._issue(seq)
         if self.sync_policy == "per_entry":
-            payload = serialize_batch(batch)
             loop = asyncio.get_running_loop()
-            await loop.run_in_executor(
-                None, self.store.append_batch, [(seq, payload)], True)
+            fast = getattr(self.store, "append_framed", None) \
+                if _nwal is not None and not getattr(self.store, "compress",
+                                                     False) else None
+            if fast is not None:
+                await loop.run_in_executor(
+                    None, lambda: fast(
+                        [_nwal.encode_frame_parts(seq, batch_parts(batch))],
+                        True))
+            else:
+                payload = serialize_batch(batch)
+                await loop.run_in_executor(
+                    None, self.store.append_batch, [(seq, payload)], True)
             return seq
         # staged: serialization (incl. any D2H) happens in the flusher
         # thread, keeping the ingest loop free (reference stages + notifies,
