@@ -333,3 +333,37 @@ def test_config_schema_type_checking():
         "id": "s", "input": {"type": "generate", "batch_size": True},
         "output": {"type": "drop"}}]})
     assert any("batch_size" in e for e in cfg2.validate())
+
+
+def test_cli_run_to_eof_subprocess(tmp_path):
+    """`python -m arkflow_amd --config <file>` runs a finite stream to EOF
+    and exits 0 (the quickstart path, as a subprocess)."""
+    import subprocess
+    import sys
+    cfg = tmp_path / "c.yaml"
+    cfg.write_text("""
+streams:
+  - id: cli
+    input:
+      type: generate
+      batch_size: 8
+      count: 32
+      interval: 1ms
+      fields:
+        v: {dtype: float32}
+    pipeline:
+      processors:
+        - type: sql
+          query: "SELECT count(*) AS n FROM flow"
+    output:
+      type: stdout
+""")
+    import os
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "arkflow_amd", "--config", str(cfg)],
+        capture_output=True, text=True, timeout=120, env=env)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert '"n": 8' in r.stdout
