@@ -366,4 +366,4 @@ streams:
         [sys.executable, "-m", "arkflow_amd", "--config", str(cfg)],
         capture_output=True, text=True, timeout=120, env=env)
     assert r.returncode == 0, r.stderr[-500:]
-    assert '"n": 8' in r.stdout
+    assert r.stdout.count('{"n":8}') == 4
