@@ -214,3 +214,46 @@ def test_rollout_config_applied_via_agent(run):
         await eng.runtime.stop_all()
 
     run(main(), timeout=60)
+
+
+def test_hub_cli_subprocess_serves(tmp_path, run):
+    """`python -m arkflow_amd hub` boots a real HTTP hub (RBAC + console)."""
+    import os
+    import socket
+    import subprocess
+    import sys
+    import time as _time
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "arkflow_amd", "hub",
+         "--address", f"127.0.0.1:{port}",
+         "--store", str(tmp_path / "hub.db"),
+         "--operator-token", "tok:admin"],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        async def main():
+            async with httpx.AsyncClient(
+                    base_url=f"http://127.0.0.1:{port}") as c:
+                for _ in range(100):
+                    try:
+                        r = await c.get("/nodes",
+                                        headers={"Authorization":
+                                                 "Bearer tok"})
+                        if r.status_code == 200:
+                            break
+                    except httpx.TransportError:
+                        await asyncio.sleep(0.2)
+                assert r.status_code == 200 and r.json() == []
+                assert (await c.get("/nodes")).status_code == 403
+                assert "console" in (await c.get("/")).text.lower()
+
+        run(main(), timeout=60)
+    finally:
+        proc.terminate()
+        proc.wait(timeout=15)
