@@ -109,13 +109,14 @@ def main(argv: Optional[List[str]] = None) -> int:
         cancel = asyncio.Event()
         tasks = []
         if config.server.enabled:
+            from .server.api import serve
+            tasks.append(asyncio.ensure_future(serve(engine, cancel)))
             if config.server.hub_url:
+                # node serves its own API AND joins the fleet hub
+                # (reference arkflow-server runs both)
                 from .server.agent import agent_run
                 tasks.append(asyncio.ensure_future(
                     agent_run(engine, cancel)))
-            else:
-                from .server.api import serve
-                tasks.append(asyncio.ensure_future(serve(engine, cancel)))
         try:
             await engine.run_with_cancellation(cancel,
                                                install_signal_handlers=True)
