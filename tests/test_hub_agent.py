@@ -257,3 +257,87 @@ def test_hub_cli_subprocess_serves(tmp_path, run):
     finally:
         proc.terminate()
         proc.wait(timeout=15)
+
+
+def test_fleet_e2e_subprocess(tmp_path, run):
+    """Real-socket fleet: hub CLI + node CLI (hub_url) as subprocesses;
+    operator starts/stops a stream through the hub HTTP API."""
+    import os
+    import socket
+    import subprocess
+    import sys
+
+    def free_port():
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            return s.getsockname()[1]
+
+    hub_port, node_port = free_port(), free_port()
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    node_cfg = tmp_path / "node.yaml"
+    node_cfg.write_text(f"""
+streams:
+  - id: s1
+    input:
+      type: generate
+      batch_size: 4
+      interval: 20ms
+      fields:
+        v: {{dtype: float32}}
+    output:
+      type: drop
+server:
+  enabled: true
+  address: 127.0.0.1:{node_port}
+  hub_url: http://127.0.0.1:{hub_port}
+  node_id: nodeA
+""")
+    hub = subprocess.Popen(
+        [sys.executable, "-m", "arkflow_amd", "hub",
+         "--address", f"127.0.0.1:{hub_port}",
+         "--operator-token", "tok:admin"],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    node = subprocess.Popen(
+        [sys.executable, "-m", "arkflow_amd", "--config", str(node_cfg)],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        async def main():
+            h = {"Authorization": "Bearer tok"}
+            async with httpx.AsyncClient(
+                    base_url=f"http://127.0.0.1:{hub_port}",
+                    headers=h) as c:
+                node_row = None
+                for _ in range(150):
+                    try:
+                        rows = (await c.get("/nodes")).json()
+                        if rows and rows[0]["online"]:
+                            node_row = rows[0]
+                            break
+                    except (httpx.TransportError, Exception):
+                        pass
+                    await asyncio.sleep(0.2)
+                assert node_row and node_row["node_id"] == "nodeA"
+                r = await c.post("/nodes/nodeA/streams/s1/start")
+                iid = r.json()["intent_id"]
+                for _ in range(150):
+                    intents = (await c.get("/intents")).json()
+                    st = [i for i in intents
+                          if i["intent_id"] == iid][0]["state"]
+                    if st in ("succeeded", "failed"):
+                        break
+                    await asyncio.sleep(0.2)
+                assert st == "succeeded"
+            # node's own API reflects the running stream
+            async with httpx.AsyncClient(
+                    base_url=f"http://127.0.0.1:{node_port}") as nc:
+                s = (await nc.get("/api/v1/streams/s1")).json()
+                assert s["state"] == "running"
+
+        run(main(), timeout=90)
+    finally:
+        node.terminate()
+        hub.terminate()
+        node.wait(timeout=15)
+        hub.wait(timeout=15)
