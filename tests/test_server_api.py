@@ -1,6 +1,7 @@
 """Node HTTP API tests via ASGI transport (the reference tests its axum
 Router with tower::oneshot the same way — no network)."""
 import asyncio
+import json
 
 import httpx
 import pytest
@@ -265,3 +266,54 @@ def test_console_pages_render(run):
             assert r.status_code == 200 and "console" in r.text.lower()
 
     run(main())
+
+
+def test_sse_last_event_id_resume(run):
+    """SSE replays only events after the Last-Event-ID header (reference
+    lib.rs:1113 resume semantics). Real socket: ASGITransport buffers
+    streaming responses, so this boots uvicorn in-process."""
+    import socket
+
+    async def main():
+        import uvicorn
+        eng = _engine()
+        for sc in eng.config.streams:
+            eng.runtime.register(sc)
+        for i in range(6):
+            eng.runtime.events.push("s1", f"k{i}")
+        evs = eng.runtime.events.list()
+        cut = [e for e in evs if e.kind == "k2"][0].seq  # resume after k2
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        app = create_app(eng)
+        server = uvicorn.Server(uvicorn.Config(
+            app, host="127.0.0.1", port=port, log_level="error"))
+        stask = asyncio.ensure_future(server.serve())
+        try:
+            async with httpx.AsyncClient(
+                    base_url=f"http://127.0.0.1:{port}") as c:
+                for _ in range(100):
+                    try:
+                        if (await c.get("/health")).status_code == 200:
+                            break
+                    except httpx.TransportError:
+                        await asyncio.sleep(0.1)
+                got = []
+                async with c.stream(
+                        "GET", "/api/v1/events/stream",
+                        headers={"Last-Event-ID": str(cut)}) as r:
+                    async for line in r.aiter_lines():
+                        if line.startswith("data: "):
+                            got.append(json.loads(line[6:]))
+                        if len(got) >= 3:
+                            break
+                kinds = [g["kind"] for g in got]
+                assert kinds == ["k3", "k4", "k5"]
+                assert all(g["seq"] > cut for g in got)
+        finally:
+            server.should_exit = True
+            await asyncio.wait_for(stask, 10)
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)
