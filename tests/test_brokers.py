@@ -352,3 +352,34 @@ def test_http_auth_lockout(run):
         await inp.close()
 
     run(main(), timeout=30)
+
+
+def test_kafka_eos_abort_on_bad_batch(run):
+    """A failure while staging a transaction publishes NOTHING (all-or-
+    nothing, reference kafka.rs txn abort path)."""
+    from arkflow_amd.inputs.brokers import FakeBus, KafkaOutput
+
+    async def main():
+        FakeBus.reset("abrt")
+        out = KafkaOutput({"brokers": "memory://abrt", "topic": "t",
+                           "exactly_once": True, "key_column": "k"})
+        await out.connect()
+        good = MessageBatch.from_dict({"k": [1], "__value__": ["a"]})
+
+        class Poison:
+            """Batch whose row materialization raises mid-transaction."""
+            columns = {}
+
+            @property
+            def num_rows(self):
+                return 1
+
+        with pytest.raises(Exception):
+            await out.write_batch([good, Poison()])
+        bus = FakeBus.get("abrt")
+        assert sum(len(p) for p in bus.topics["t"]) == 0  # nothing visible
+        # a later good transaction still works
+        await out.write_batch([good])
+        assert sum(len(p) for p in bus.topics["t"]) == 1
+
+    run(main())
