@@ -431,7 +431,17 @@ class SqlExecutor:
         device = env.device
         n = env.n_rows
         if sel.group_by:
-            key_vals = [eval_expr(e, env) for e in sel.group_by]
+            # GROUP BY may name a projection alias (sqlite/DataFusion
+            # semantics): substitute the aliased expression
+            from .parser import ColumnRef as _CR
+            aliases = {alias: pe for pe, alias in sel.projections
+                       if alias is not None}
+            group_exprs = [aliases[e.name]
+                           if isinstance(e, _CR) and e.table is None
+                           and e.name in aliases and e.name not in env.columns
+                           else e
+                           for e in sel.group_by]
+            key_vals = [eval_expr(e, env) for e in group_exprs]
             keys = [v if isinstance(v, Column) else as_tensor(v, env)
                     for v in key_vals]
             single = keys[0]

@@ -34,6 +34,15 @@ QUERIES = [
     "SELECT a, sum(a) OVER (PARTITION BY k) AS s FROM flow ORDER BY a, s",
     "SELECT a FROM flow WHERE a < 10 UNION ALL SELECT a FROM flow "
     "WHERE a > 90",
+    "SELECT k, sum(a) AS s FROM flow GROUP BY k "
+    "HAVING sum(a) > 3 * count(*) ORDER BY k",
+    "SELECT a, b FROM flow ORDER BY CASE WHEN a % 2 = 0 THEN a ELSE -a END, "
+    "b LIMIT 15",
+    "SELECT a FROM flow WHERE abs(a - 50) < round(b * 10)",
+    "SELECT f.k, count(*) AS c FROM flow f JOIN dims d ON f.k = d.k "
+    "GROUP BY f.k ORDER BY f.k",
+    "SELECT upper(d.label) AS ul, min(f.a) AS lo FROM flow f "
+    "JOIN dims d ON f.k = d.k GROUP BY ul ORDER BY ul",
 ]
 
 
