@@ -384,7 +384,8 @@ def _eval_binop(e: BinaryOp, env: Env) -> Value:
             return l / r
         return torch.div(l, r, rounding_mode="trunc")
     if e.op == "%":
-        return torch.remainder(l, r)
+        # SQL %: truncated remainder (sign of the dividend — sqlite/DataFusion)
+        return torch.fmod(l, r)
     raise SqlError(f"unknown operator {e.op}")
 
 

@@ -43,6 +43,11 @@ QUERIES = [
     "GROUP BY f.k ORDER BY f.k",
     "SELECT upper(d.label) AS ul, min(f.a) AS lo FROM flow f "
     "JOIN dims d ON f.k = d.k GROUP BY ul ORDER BY ul",
+    "SELECT CAST(b * 10 AS INTEGER) AS bi, count(*) AS c FROM flow "
+    "GROUP BY bi ORDER BY bi",
+    "SELECT c % 3 AS m FROM flow WHERE c < 0 ORDER BY m, c",
+    "SELECT d.label FROM dims d WHERE d.label IN ('L1', 'L3', 'L7')",
+    "SELECT a FROM flow WHERE a + c BETWEEN 10 AND 40 ORDER BY a",
 ]
 
 
