@@ -171,7 +171,15 @@ class SqlExecutor:
                 {**env.columns, **result.columns},
                 result.num_rows, device, env.agg_results)
             # multi-key sort: stable sorts applied last-key-first
+            from .parser import Literal as _Lit
+            out_names = list(result.columns)
             for e, asc in reversed(sel.order_by):
+                if isinstance(e, _Lit) and isinstance(e.value, int) \
+                        and not isinstance(e.value, bool) \
+                        and 1 <= e.value <= len(out_names):
+                    # positional ORDER BY n → nth projection (sqlite/standard)
+                    from .parser import ColumnRef as _CRef
+                    e = _CRef(out_names[e.value - 1], None)
                 v = eval_expr(e, final_env)
                 key = v.data if isinstance(v, Column) and v.kind == "numeric" \
                     else v

@@ -48,6 +48,10 @@ QUERIES = [
     "SELECT c % 3 AS m FROM flow WHERE c < 0 ORDER BY m, c",
     "SELECT d.label FROM dims d WHERE d.label IN ('L1', 'L3', 'L7')",
     "SELECT a FROM flow WHERE a + c BETWEEN 10 AND 40 ORDER BY a",
+    "SELECT max(a) - min(a) AS spread, avg(b) AS m FROM flow",
+    "SELECT k, a, b FROM flow ORDER BY 1, 2, 3 LIMIT 10",
+    "SELECT f.a, f.b, d.k, d.label FROM flow f JOIN dims d ON f.k = d.k "
+    "WHERE f.a >= 75 ORDER BY f.a, f.b",
 ]
 
 
