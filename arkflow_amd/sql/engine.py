@@ -167,8 +167,9 @@ class SqlExecutor:
 
         # ---------------------------------------------------------- order by
         if sel.order_by:
+            cur_cols = dict(env.columns)  # permuted alongside result
             final_env = Env(
-                {**env.columns, **result.columns},
+                {**cur_cols, **result.columns},
                 result.num_rows, device, env.agg_results)
             # multi-key sort: stable sorts applied last-key-first
             from .parser import Literal as _Lit
@@ -194,9 +195,12 @@ class SqlExecutor:
                 else:
                     idx = ops.sort_indices(key, ascending=asc)
                 result = result.take(idx)
+                # take from the ALREADY-permuted columns: re-taking from the
+                # original env would drop earlier keys' permutations (bug
+                # caught by 3-key ORDER BY differential)
+                cur_cols = {k: c.take(idx) for k, c in cur_cols.items()}
                 final_env = Env(
-                    {**{k: c.take(idx) for k, c in env.columns.items()},
-                     **result.columns},
+                    {**cur_cols, **result.columns},
                     result.num_rows, device, env.agg_results)
 
         # ---------------------------------------------------- offset / limit
@@ -220,6 +224,68 @@ class SqlExecutor:
         return result
 
     # --------------------------------------------------------------- windows
+    def _running_window_agg(self, w_: FuncCall, name: str,
+                            vals: torch.Tensor, gid: torch.Tensor,
+                            env: Env) -> torch.Tensor:
+        """RANGE UNBOUNDED PRECEDING..CURRENT ROW over the partition order:
+        sort by (partition, ORDER BY keys), scan, and give equal peers the
+        frame end of their tie group."""
+        device = env.device
+        n = env.n_rows
+        perm = torch.arange(n, dtype=torch.int64, device=device)
+        sort_keys = []
+        for e, asc in reversed(w_.over.order_by):
+            v = eval_expr(e, env)
+            key = v.data if isinstance(v, Column) and v.kind == "numeric" \
+                else as_tensor(v, env)
+            sort_keys.append((key, asc))
+            perm = perm[ops.sort_indices(key[perm], ascending=asc)]
+        perm = perm[ops.sort_indices(gid[perm], ascending=True)]
+        sg = gid[perm]
+        sv = vals[perm].double()
+        ar = torch.arange(n, dtype=torch.int64, device=device)
+        part_start = torch.zeros(n, dtype=torch.bool, device=device)
+        part_start[0] = True
+        part_start[1:] = sg[1:] != sg[:-1]
+        group_start = torch.cummax(ar * part_start, 0).values
+        tie_change = part_start.clone()
+        for key, asc in sort_keys:
+            sk = key[perm]
+            tie_change[1:] |= sk[1:] != sk[:-1]
+        tie_id = torch.cumsum(tie_change.long(), 0) - 1
+        n_ties = int(tie_id[-1].item()) + 1 if n else 0
+        last = torch.zeros(max(n_ties, 1), dtype=torch.int64, device=device)
+        last.scatter_reduce_(0, tie_id, ar, "amax", include_self=False)
+        frame_end = last[tie_id]  # last peer of my tie group
+        if name in ("sum", "avg", "count"):
+            cs = torch.cumsum(sv, 0)
+            base = cs[group_start] - sv[group_start]  # sum before partition
+            run_sum = cs[frame_end] - base
+            run_cnt = (frame_end - group_start + 1).double()
+            if name == "count":
+                out_sorted = run_cnt
+            elif name == "sum":
+                out_sorted = run_sum
+            else:
+                out_sorted = run_sum / run_cnt
+        else:  # min / max: segmented scan by doubling within partitions
+            y = sv.clone()
+            dist = ar - group_start
+            shift = 1
+            while shift < n:
+                shifted = torch.empty_like(y)
+                shifted[shift:] = y[:-shift]
+                shifted[:shift] = y[:shift]  # unused (masked)
+                mask = dist >= shift
+                y = torch.where(mask,
+                                torch.maximum(y, shifted) if name == "max"
+                                else torch.minimum(y, shifted), y)
+                shift <<= 1
+            out_sorted = y[frame_end]
+        out = torch.empty(n, dtype=torch.float64, device=device)
+        out[perm] = out_sorted
+        return out
+
     def _compute_window(self, w_: FuncCall, env: Env) -> torch.Tensor:
         """Window functions over PARTITION BY (full-partition frame):
         row_number/rank/dense_rank and partition-wide aggregates."""
@@ -246,6 +312,11 @@ class SqlExecutor:
                 vals = torch.ones(n, device=device)
             else:
                 vals = as_tensor(eval_expr(w_.args[0], env), env)
+            if w_.over.order_by:
+                # running aggregate: the standard's default frame with
+                # ORDER BY is RANGE UNBOUNDED PRECEDING..CURRENT ROW —
+                # equal peers share the frame end (sqlite/DataFusion)
+                return self._running_window_agg(w_, name, vals, gid, env)
             per_group = ops.segment_reduce(
                 vals.float() if name != "count" else vals, gid.to(torch.int32),
                 g, "mean" if name == "avg" else name)

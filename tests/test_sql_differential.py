@@ -52,6 +52,13 @@ QUERIES = [
     "SELECT k, a, b FROM flow ORDER BY 1, 2, 3 LIMIT 10",
     "SELECT f.a, f.b, d.k, d.label FROM flow f JOIN dims d ON f.k = d.k "
     "WHERE f.a >= 75 ORDER BY f.a, f.b",
+    "SELECT a, sum(a) OVER (PARTITION BY k ORDER BY a) AS rs FROM flow "
+    "ORDER BY k, a, rs",
+    "SELECT a, count(*) OVER (ORDER BY a) AS rc FROM flow ORDER BY a, rc",
+    "SELECT a, max(b) OVER (PARTITION BY k ORDER BY a) AS rm FROM flow "
+    "ORDER BY k, a, rm",
+    "SELECT a, avg(a) OVER (PARTITION BY c ORDER BY a, b) AS ra FROM flow "
+    "ORDER BY c, a, b",
 ]
 
 
