@@ -371,6 +371,29 @@ def test_like_on_gpu_matches_cpu(nat, dev):
         assert gpu == cpu, sql
 
 
+def test_running_window_agg_on_gpu(nat, dev):
+    """Running window aggregate (RANGE..CURRENT ROW) on device columns
+    matches the CPU result."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+    torch.manual_seed(4)
+    n = 100_000
+    b = MessageBatch.from_dict({
+        "k": torch.randint(0, 64, (n,), dtype=torch.int64),
+        "a": torch.randint(0, 1000, (n,), dtype=torch.int64),
+        "v": torch.rand(n, dtype=torch.float32)})
+    sql = ("SELECT a, sum(v) OVER (PARTITION BY k ORDER BY a) AS rs, "
+           "max(v) OVER (PARTITION BY k ORDER BY a) AS rm "
+           "FROM flow ORDER BY k, a, rs LIMIT 500")
+    cpu = SqlExecutor(sql).execute({"flow": b})
+    gpu = SqlExecutor(sql).execute({"flow": b.to(dev)})
+    assert gpu.column("rs").data.is_cuda
+    for col in ("a", "rs", "rm"):
+        cv = torch.as_tensor(cpu.column(col).data, dtype=torch.float64)
+        gv = gpu.column(col).data.double().cpu()
+        assert torch.allclose(cv, gv, atol=1e-3), col
+
+
 def test_take_binary_matches_cpu(nat, dev):
     """Binary-column gather (take_binary kernel) vs the host path."""
     from arkflow_amd.batch import MessageBatch
