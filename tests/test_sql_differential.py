@@ -59,6 +59,11 @@ QUERIES = [
     "ORDER BY k, a, rm",
     "SELECT a, avg(a) OVER (PARTITION BY c ORDER BY a, b) AS ra FROM flow "
     "ORDER BY c, a, b",
+    "SELECT a, row_number() OVER (PARTITION BY k ORDER BY a DESC, b) AS rn "
+    "FROM flow ORDER BY k, rn",
+    "SELECT a, dense_rank() OVER (ORDER BY k, c) AS dr FROM flow "
+    "ORDER BY a, dr",
+    "SELECT a, min(a) OVER (ORDER BY b) AS lo FROM flow ORDER BY b, a",
 ]
 
 
