@@ -154,7 +154,13 @@ class SqlExecutor:
                 continue
             v = eval_expr(e, env)
             name = alias_name or expr_name(e)
-            out_cols[name] = _to_column(v, env)
+            col = _to_column(v, env)
+            if col.validity is None and not self.is_aggregate:
+                from .eval import expr_validity
+                vv = expr_validity(e, env)
+                if vv is not None and len(vv) == len(col):
+                    col = Column(col.kind, col.data, col.offsets, vv)
+            out_cols[name] = col
         result = MessageBatch(out_cols, input_name=None)
 
         # ---------------------------------------------------------- distinct
@@ -417,11 +423,16 @@ class SqlExecutor:
             if isinstance(l, ColumnRef) and isinstance(r, Literal) \
                     and isinstance(r.value, (int, float)):
                 col = env.lookup(l)
-                if col.kind == "numeric" and col.data.dtype in (
+                if col.kind == "numeric" and col.validity is None \
+                        and col.data.dtype in (
                         torch.float32, torch.int64, torch.int32):
                     return ops.filter_cmp_scalar(
                         col.data, pred.op, float(r.value)).long()
         mask = as_tensor(eval_expr(pred, env), env).bool()
+        from .eval import expr_validity
+        v = expr_validity(pred, env)
+        if v is not None:
+            mask = mask & v  # NULL predicate → row dropped (SQL three-valued)
         return ops.mask_to_indices(mask).long()
 
     # ------------------------------------------------------------------ join

@@ -277,6 +277,16 @@ def eval_expr(e, env: Env) -> Value:
                 r |= (t == as_tensor(eval_expr(item, env), env))
         return ~r if e.negated else r
     if isinstance(e, IsNull):
+        # a ColumnRef's validity lives on the Column (numeric eval returns
+        # the raw tensor) — consult it directly
+        if isinstance(e.expr, ColumnRef):
+            try:
+                col = env.lookup(e.expr)
+            except SqlError:
+                col = None
+            if isinstance(col, Column) and col.validity is not None:
+                r = ~col.validity
+                return ~r if e.negated else r
         v = eval_expr(e.expr, env)
         if isinstance(v, Column) and v.validity is not None:
             r = ~v.validity
@@ -459,16 +469,27 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
         return torch.pow(b, p)
     if name == "coalesce":
         result = None
+        missing = None  # rows still NULL so far
         for a in e.args:
+            validity = None
+            if isinstance(a, ColumnRef):
+                try:
+                    c = env.lookup(a)
+                    if isinstance(c, Column):
+                        validity = c.validity
+                except SqlError:
+                    pass
             v = as_tensor(eval_expr(a, env), env)
+            miss_here = torch.isnan(v) if v.is_floating_point() \
+                else torch.zeros(v.shape, dtype=torch.bool, device=v.device)
+            if validity is not None:
+                miss_here = miss_here | ~validity
             if result is None:
-                result = v
+                result, missing = v, miss_here
             else:
                 result, v = _promote(result, v)
-                result = torch.where(torch.isnan(result) if
-                                     result.is_floating_point() else
-                                     torch.zeros_like(result, dtype=torch.bool),
-                                     v, result)
+                result = torch.where(missing, v, result)
+                missing = missing & miss_here
         return result
     if name in ("to_int", "to_float", "to_string"):
         # VRL-style conversions (processor/expr_proc.py translate_vrl)
@@ -549,3 +570,48 @@ def collect_window_calls(e, out: list) -> None:
             collect_window_calls(v, out)
         if e.else_ is not None:
             collect_window_calls(e.else_, out)
+
+
+def expr_validity(e, env: Env):
+    """Combined validity of the columns an expression touches (strict-NULL
+    propagation for elementwise expressions): None = all valid. Expressions
+    containing null-AWARE constructs (IS NULL, coalesce, CASE) handle
+    validity themselves and return None here."""
+    cols: list = []
+    if _collect_strict_refs(e, cols) is False:
+        return None
+    v = None
+    for ref in cols:
+        try:
+            c = env.lookup(ref)
+        except SqlError:
+            continue
+        if isinstance(c, Column) and c.validity is not None:
+            v = c.validity if v is None else (v & c.validity)
+    return v
+
+
+def _collect_strict_refs(e, out: list):
+    """Gather ColumnRefs; returns False if the expr contains a null-aware
+    construct (the caller must not apply strict propagation)."""
+    if isinstance(e, ColumnRef):
+        out.append(e)
+        return True
+    if isinstance(e, (IsNull, Case)):
+        return False
+    if isinstance(e, FuncCall):
+        if e.name in ("coalesce", "ifnull", "count"):
+            return False
+        return all(_collect_strict_refs(a, out) is not False for a in e.args)
+    if isinstance(e, BinaryOp):
+        return (_collect_strict_refs(e.left, out) is not False
+                and _collect_strict_refs(e.right, out) is not False)
+    if isinstance(e, UnaryOp):
+        return _collect_strict_refs(e.operand, out)
+    if isinstance(e, Cast):
+        return _collect_strict_refs(e.expr, out)
+    if isinstance(e, (InList, Between, Like)):
+        inner = getattr(e, "expr", None)
+        if inner is not None:
+            return _collect_strict_refs(inner, out)
+    return True

@@ -241,3 +241,24 @@ def test_window_udf():
     r = q("SELECT v, frac_of_partition(v) OVER (PARTITION BY k) AS f "
           "FROM flow ORDER BY v", flow=flow)
     assert r.column("f").to_pylist() == [0.25, 0.75, 1.0]
+
+
+def test_null_propagation_strict():
+    """Strict SQL NULL semantics on validity columns: NULL predicates drop
+    rows, arithmetic over NULL yields NULL, IS NULL / coalesce see through."""
+    import torch
+    from arkflow_amd.batch import Column, MessageBatch
+    c = Column.from_numeric(torch.tensor([1.0, 2.0, 3.0, 4.0]))
+    c.validity = torch.tensor([True, False, True, True])
+    flow = MessageBatch({"v": c,
+                         "w": Column.from_numeric(
+                             torch.tensor([10.0, 20.0, 30.0, 40.0]))})
+    r = q("SELECT w FROM flow WHERE v > 0", flow=flow)
+    assert r.column("w").to_pylist() == [10.0, 30.0, 40.0]  # NULL row dropped
+    r = q("SELECT v + w AS s FROM flow", flow=flow)
+    assert r.column("s").to_pylist() == [11.0, None, 33.0, 44.0]
+    r = q("SELECT coalesce(v, 0 - 1) AS cv, w FROM flow WHERE v IS NULL",
+          flow=flow)
+    assert r.column("w").to_pylist() == [20.0]
+    r = q("SELECT w FROM flow WHERE v IS NOT NULL AND v >= 3", flow=flow)
+    assert r.column("w").to_pylist() == [30.0, 40.0]
