@@ -596,6 +596,28 @@ class SqlExecutor:
             return ops.segment_reduce(
                 torch.ones(env.n_rows, device=device), gid, g, "count")
         vals = as_tensor(arg, env)
+        from .eval import expr_validity
+        validity = expr_validity(a.args[0], env) if a.args else None
+        if validity is not None and bool((~validity).any()):
+            # SQL aggregates ignore NULL inputs
+            vf = vals.float()
+            if name in ("sum", "avg"):
+                s = ops.segment_reduce(
+                    torch.where(validity, vf, torch.zeros_like(vf)),
+                    gid, g, "sum")
+                if name == "sum":
+                    return s.to(torch.int64) \
+                        if not vals.dtype.is_floating_point else s
+                cnt = ops.segment_reduce(validity.float(), gid, g, "sum")
+                return s / cnt
+            fill = float("inf") if name == "min" else float("-inf")
+            out = ops.segment_reduce(
+                torch.where(validity, vf, torch.full_like(vf, fill)),
+                gid, g, name)
+            out = torch.where(torch.isinf(out),
+                              torch.full_like(out, float("nan")), out)
+            return out.to(torch.int64) if not vals.dtype.is_floating_point \
+                and not torch.isnan(out).any() else out
         if name == "avg":
             return ops.segment_reduce(vals, gid, g, "mean")
         out = ops.segment_reduce(vals, gid, g, name)  # sum|min|max

@@ -262,3 +262,18 @@ def test_null_propagation_strict():
     assert r.column("w").to_pylist() == [20.0]
     r = q("SELECT w FROM flow WHERE v IS NOT NULL AND v >= 3", flow=flow)
     assert r.column("w").to_pylist() == [30.0, 40.0]
+
+
+def test_aggregates_skip_nulls():
+    import torch
+    from arkflow_amd.batch import Column, MessageBatch
+    c = Column.from_numeric(torch.tensor([1.0, 100.0, 3.0, 100.0]))
+    c.validity = torch.tensor([True, False, True, False])
+    flow = MessageBatch({"v": c, "k": Column.from_numeric(
+        torch.tensor([1, 1, 2, 2]))})
+    r = q("SELECT k, sum(v) AS s, count(v) AS c, avg(v) AS m, max(v) AS hi "
+          "FROM flow GROUP BY k ORDER BY k", flow=flow)
+    assert r.column("s").to_pylist() == [1.0, 3.0]
+    assert r.column("c").to_pylist() == [1, 1]
+    assert r.column("m").to_pylist() == [1.0, 3.0]
+    assert r.column("hi").to_pylist() == [1.0, 3.0]
