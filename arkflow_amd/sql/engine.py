@@ -589,9 +589,14 @@ class SqlExecutor:
                 return ops.segment_reduce(
                     torch.ones(sub_gid.shape[0], device=device),
                     sub_gid, g, "count")
+            from .eval import expr_validity
+            validity = expr_validity(a.args[0], env)
             if isinstance(arg, Column) and arg.validity is not None:
+                validity = arg.validity if validity is None \
+                    else (validity & arg.validity)
+            if validity is not None:
                 return ops.segment_reduce(
-                    arg.validity.to(torch.float32), gid, g, "sum"
+                    validity.to(torch.float32), gid, g, "sum"
                 ).to(torch.int64)
             return ops.segment_reduce(
                 torch.ones(env.n_rows, device=device), gid, g, "count")
