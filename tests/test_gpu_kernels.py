@@ -371,6 +371,27 @@ def test_like_on_gpu_matches_cpu(nat, dev):
         assert gpu == cpu, sql
 
 
+def test_null_aware_sql_on_gpu(nat, dev):
+    """Validity-aware WHERE/aggregates on device columns."""
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+    n = 50_000
+    v = torch.rand(n)
+    val = torch.rand(n) > 0.3
+    c = Column("numeric", v, validity=val)
+    flow = MessageBatch({"v": c, "k": Column.from_numeric(
+        torch.randint(0, 16, (n,), dtype=torch.int64))})
+    sql = ("SELECT k, count(v) AS c, sum(v) AS s FROM flow "
+           "WHERE v >= 0 GROUP BY k ORDER BY k")
+    cpu = SqlExecutor(sql).execute({"flow": flow})
+    gpu = SqlExecutor(sql).execute({"flow": flow.to(dev)})
+    assert gpu.column("c").to_pylist() == cpu.column("c").to_pylist()
+    sc = torch.tensor(cpu.column("s").to_pylist())
+    sg = torch.tensor(gpu.column("s").to_pylist())
+    assert torch.allclose(sc, sg, rtol=1e-4)
+    assert sum(cpu.column("c").to_pylist()) == int(val.sum())
+
+
 def test_running_window_agg_on_gpu(nat, dev):
     """Running window aggregate (RANGE..CURRENT ROW) on device columns
     matches the CPU result."""
