@@ -532,8 +532,23 @@ class SqlExecutor:
                            else e
                            for e in sel.group_by]
             key_vals = [eval_expr(e, env) for e in group_exprs]
-            keys = [v if isinstance(v, Column) else as_tensor(v, env)
-                    for v in key_vals]
+            keys = []
+            for ge, v in zip(group_exprs, key_vals):
+                k = v if isinstance(v, Column) else as_tensor(v, env)
+                from .eval import expr_validity
+                validity = expr_validity(ge, env)
+                if validity is not None and bool((~validity).any()):
+                    # NULL keys form ONE group: normalize the value under
+                    # the null and add the validity bit to the key material
+                    kt = k.data if isinstance(k, Column) else k
+                    if kt.dtype != torch.bool and kt.dim() == 1 \
+                            and not isinstance(k, Column):
+                        k = torch.where(validity, kt,
+                                        torch.zeros_like(kt))
+                    keys.append(k)
+                    keys.append(validity)
+                    continue
+                keys.append(k)
             single = keys[0]
             if len(keys) == 1 and isinstance(single, torch.Tensor) \
                     and single.dtype in (torch.int64, torch.int32):

@@ -277,3 +277,17 @@ def test_aggregates_skip_nulls():
     assert r.column("c").to_pylist() == [1, 1]
     assert r.column("m").to_pylist() == [1.0, 3.0]
     assert r.column("hi").to_pylist() == [1.0, 3.0]
+
+
+def test_null_group_by_keys_one_group():
+    import torch
+    from arkflow_amd.batch import Column, MessageBatch
+    c = Column.from_numeric(torch.tensor([1, 7, 1, 9], dtype=torch.int64))
+    c.validity = torch.tensor([True, False, True, False])
+    flow = MessageBatch({"k": c, "v": Column.from_numeric(
+        torch.tensor([10., 20., 30., 40.]))})
+    r = q("SELECT k, count(*) AS n, sum(v) AS s FROM flow GROUP BY k "
+          "ORDER BY n, s", flow=flow)
+    # two groups: k=1 (rows 0,2) and NULL (rows 1,3 despite differing 7/9)
+    assert r.column("n").to_pylist() == [2, 2]
+    assert sorted(r.column("s").to_pylist()) == [40.0, 60.0]
