@@ -200,6 +200,13 @@ class SqlExecutor:
                         dtype=torch.int64, device=device)
                 else:
                     idx = ops.sort_indices(key, ascending=asc)
+                from .eval import expr_validity
+                kv = expr_validity(e, final_env)
+                if kv is not None and bool((~kv).any()):
+                    # NULL sorts as smallest (sqlite/standard): stable
+                    # re-sort by validity with the same direction
+                    idx = idx[ops.sort_indices(
+                        kv[idx].to(torch.int32), ascending=asc)]
                 result = result.take(idx)
                 # take from the ALREADY-permuted columns: re-taking from the
                 # original env would drop earlier keys' permutations (bug

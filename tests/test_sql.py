@@ -291,3 +291,16 @@ def test_null_group_by_keys_one_group():
     # two groups: k=1 (rows 0,2) and NULL (rows 1,3 despite differing 7/9)
     assert r.column("n").to_pylist() == [2, 2]
     assert sorted(r.column("s").to_pylist()) == [40.0, 60.0]
+
+
+def test_order_by_null_placement():
+    import torch
+    from arkflow_amd.batch import Column, MessageBatch
+    c = Column.from_numeric(torch.tensor([5., 1., 3.]))
+    c.validity = torch.tensor([True, False, True])
+    flow = MessageBatch({"v": c, "w": Column.from_numeric(
+        torch.tensor([1, 2, 3]))})
+    r = q("SELECT w FROM flow ORDER BY v", flow=flow)
+    assert r.column("w").to_pylist() == [2, 3, 1]  # NULL smallest → first
+    r = q("SELECT w FROM flow ORDER BY v DESC", flow=flow)
+    assert r.column("w").to_pylist() == [1, 3, 2]  # DESC → NULL last
