@@ -485,10 +485,45 @@ class SqlExecutor:
         lk = lk.to(torch.int64) if lk.dtype not in (
             torch.int64, torch.float32) else lk
         rk = rk.to(lk.dtype) if rk.dtype != lk.dtype else rk
+        # NULL join keys never match (SQL): drop invalid-key rows from each
+        # side before the hash join, remapping indices back afterwards. For
+        # LEFT joins, invalid-left rows re-enter unmatched (NULL right side).
+        from .eval import expr_validity
+        l_val = r_val = None
+        for a, b in key_pairs:
+            if side_of(a) == "right":
+                a, b = b, a
+            va = expr_validity(a, l_env)
+            vb = expr_validity(b, r_env)
+            if va is not None:
+                l_val = va if l_val is None else (l_val & va)
+            if vb is not None:
+                r_val = vb if r_val is None else (r_val & vb)
+        left_map = right_map = None
+        null_left = None
+        if l_val is not None and bool((~l_val).any()):
+            left_map = l_val.nonzero(as_tuple=True)[0]
+            null_left = (~l_val).nonzero(as_tuple=True)[0]
+            lk = lk[left_map]
+        if r_val is not None and bool((~r_val).any()):
+            right_map = r_val.nonzero(as_tuple=True)[0]
+            rk = rk[right_map]
         if j.kind == "left":
             l_idx, r_idx = ops.join_left(lk, rk)
         else:
             l_idx, r_idx = ops.join_inner(lk, rk)
+        if left_map is not None:
+            l_idx = left_map[l_idx.long()]
+            if j.kind == "left" and null_left.numel():
+                l_idx = torch.cat([l_idx, null_left])
+                r_idx = torch.cat([r_idx, torch.full(
+                    (null_left.numel(),), -1, dtype=r_idx.dtype,
+                    device=r_idx.device)])
+        if right_map is not None:
+            matched = r_idx >= 0
+            remapped = r_idx.clone().long()
+            remapped[matched] = right_map[r_idx[matched].long()]
+            r_idx = remapped
 
         new_cols: Dict[str, Column] = {}
         new_order: List[str] = []

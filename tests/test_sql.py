@@ -304,3 +304,20 @@ def test_order_by_null_placement():
     assert r.column("w").to_pylist() == [2, 3, 1]  # NULL smallest → first
     r = q("SELECT w FROM flow ORDER BY v DESC", flow=flow)
     assert r.column("w").to_pylist() == [1, 3, 2]  # DESC → NULL last
+
+
+def test_null_join_keys_never_match():
+    import torch
+    from arkflow_amd.batch import Column, MessageBatch
+    lk = Column.from_numeric(torch.tensor([1, 2, 3], dtype=torch.int64))
+    lk.validity = torch.tensor([True, False, True])
+    flow = MessageBatch({"k": lk, "a": Column.from_numeric(
+        torch.tensor([10, 20, 30]))})
+    dims = MessageBatch.from_dict({"k": [1, 2], "label": ["one", "two"]})
+    r = q("SELECT a FROM flow f JOIN dims d ON f.k = d.k ORDER BY a",
+          flow=flow, dims=dims)
+    assert r.column("a").to_pylist() == [10]  # NULL k=2 must not match
+    r = q("SELECT a, d.label AS lbl FROM flow f LEFT JOIN dims d "
+          "ON f.k = d.k ORDER BY a", flow=flow, dims=dims)
+    assert r.column("a").to_pylist() == [10, 20, 30]
+    assert r.column("lbl").to_pylist()[1] is None  # unmatched, not 'two'
