@@ -390,6 +390,19 @@ def test_null_aware_sql_on_gpu(nat, dev):
     sg = torch.tensor(gpu.column("s").to_pylist())
     assert torch.allclose(sc, sg, rtol=1e-4)
     assert sum(cpu.column("c").to_pylist()) == int(val.sum())
+    # null join keys + null group keys on device match CPU
+    lk = Column("numeric", torch.randint(0, 8, (n,), dtype=torch.int64),
+                validity=torch.rand(n) > 0.2)
+    flow2 = MessageBatch({"k": lk, "a": Column.from_numeric(
+        torch.arange(n, dtype=torch.int64))})
+    dims = MessageBatch.from_dict({"k": list(range(6)),
+                                   "label": [f"L{i}" for i in range(6)]})
+    sql2 = ("SELECT d.label AS l, count(*) AS c FROM flow f "
+            "JOIN dims d ON f.k = d.k GROUP BY l ORDER BY l")
+    cpu2 = SqlExecutor(sql2).execute({"flow": flow2, "dims": dims})
+    gpu2 = SqlExecutor(sql2).execute({"flow": flow2.to(dev),
+                                      "dims": dims.to(dev)})
+    assert gpu2.column("c").to_pylist() == cpu2.column("c").to_pylist()
 
 
 def test_running_window_agg_on_gpu(nat, dev):
