@@ -511,6 +511,29 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
             return t.trunc().to(torch.int64) if t.dtype.is_floating_point \
                 else t.to(torch.int64)
         return t.to(torch.float64)
+    if name in ("contains", "starts_with", "ends_with"):
+        v = eval_expr(e.args[0], env)
+        pat = eval_expr(e.args[1], env)
+        if not (isinstance(v, Column) and v.kind == "binary") \
+                or not isinstance(pat, str):
+            raise SqlError(f"{name}(column, 'literal') expects a string "
+                           "column and literal")
+        mode = {"contains": 0, "starts_with": 1, "ends_with": 2}[name]
+        if v.data.is_cuda and len(pat.encode()) <= 64:
+            from .. import ops
+            r = ops.require_native().bytes_match(
+                v.data.contiguous(), v.offsets.contiguous(),
+                pat.encode(), mode)
+        else:
+            f = {"contains": (lambda s, p: p in s),
+                 "starts_with": bytes.startswith,
+                 "ends_with": bytes.endswith}[name]
+            pb = pat.encode()
+            r = torch.tensor([bool(f(x, pb)) for x in v.to_pylist()],
+                             dtype=torch.bool, device=env.device)
+        if isinstance(v, Column) and v.validity is not None:
+            r = r & v.validity
+        return r
     if name in ("length", "char_length", "octet_length"):
         v = eval_expr(e.args[0], env)
         if isinstance(v, Column) and v.kind == "binary":

@@ -321,3 +321,14 @@ def test_null_join_keys_never_match():
           "ON f.k = d.k ORDER BY a", flow=flow, dims=dims)
     assert r.column("a").to_pylist() == [10, 20, 30]
     assert r.column("lbl").to_pylist()[1] is None  # unmatched, not 'two'
+
+
+def test_string_predicate_functions():
+    flow = MessageBatch.from_dict({"s": ["apple", "grape", "applesauce"],
+                                   "v": [1, 2, 3]})
+    r = q("SELECT v FROM flow WHERE starts_with(s, 'app')", flow=flow)
+    assert r.column("v").to_pylist() == [1, 3]
+    r = q("SELECT v FROM flow WHERE contains(s, 'rap')", flow=flow)
+    assert r.column("v").to_pylist() == [2]
+    r = q("SELECT v FROM flow WHERE ends_with(s, 'sauce')", flow=flow)
+    assert r.column("v").to_pylist() == [3]
