@@ -92,12 +92,20 @@ class Column:
             t = torch.from_numpy(arr)
         else:
             vals = list(values)
+            validity = None
+            if any(v is None for v in vals):
+                validity = torch.tensor([v is not None for v in vals])
+                fill = next((v for v in vals if v is not None), 0)
+                vals = [fill if v is None else v for v in vals]
             if vals and isinstance(vals[0], bool):
                 t = torch.tensor(vals, dtype=torch.bool)
             elif all(isinstance(v, int) for v in vals):
                 t = torch.tensor(vals, dtype=torch.int64)
             else:
                 t = torch.tensor([float(v) for v in vals], dtype=torch.float64)
+            if t.dim() != 1:
+                t = t.reshape(-1)
+            return Column("numeric", t, validity=validity)
         if t.dim() != 1:
             t = t.reshape(-1)
         return Column("numeric", t)

@@ -160,6 +160,19 @@ class SqlExecutor:
                 vv = expr_validity(e, env)
                 if vv is not None and len(vv) == len(col):
                     col = Column(col.kind, col.data, col.offsets, vv)
+            elif col.validity is None and self.is_aggregate \
+                    and isinstance(e, ColumnRef):
+                # group-key projection: the NULL group's key must stay NULL
+                # (aggregate RESULTS never inherit input validity)
+                try:
+                    src_col = env.lookup(e)
+                except SqlError:
+                    src_col = None
+                if isinstance(src_col, Column) \
+                        and src_col.validity is not None \
+                        and len(src_col.validity) == len(col):
+                    col = Column(col.kind, col.data, col.offsets,
+                                 src_col.validity)
             out_cols[name] = col
         result = MessageBatch(out_cols, input_name=None)
 

@@ -67,10 +67,12 @@ QUERIES = [
 ]
 
 
-def _random_table(rng, n):
+def _random_table(rng, n, nulls=False):
+    def maybe(v):
+        return None if nulls and rng.random() < 0.15 else v
     return {
-        "a": [rng.randrange(100) for _ in range(n)],
-        "b": [round(rng.random(), 6) for _ in range(n)],
+        "a": [maybe(rng.randrange(100)) for _ in range(n)],
+        "b": [maybe(round(rng.random(), 6)) for _ in range(n)],
         "c": [rng.randrange(-5, 6) for _ in range(n)],
         "k": [rng.randrange(8) for _ in range(n)],
     }
@@ -153,3 +155,36 @@ def test_parser_fuzz_no_crash(seed):
             pass
         except RecursionError:
             pass
+
+
+
+NULL_QUERIES = [
+    "SELECT a, b FROM flow WHERE a >= 50",
+    "SELECT a FROM flow WHERE a IS NULL AND c > 0 ORDER BY c LIMIT 5",
+    "SELECT a FROM flow WHERE a IS NOT NULL AND a < 30 ORDER BY a",
+    "SELECT k, count(a) AS c, sum(a) AS s FROM flow GROUP BY k",
+    "SELECT k, avg(b) AS m, min(a) AS lo, max(a) AS hi FROM flow GROUP BY k",
+    "SELECT coalesce(a, 0 - 1) AS ca FROM flow WHERE c = 1",
+    "SELECT a + b AS s FROM flow WHERE c >= 2",
+    "SELECT a, count(*) AS n FROM flow GROUP BY a HAVING count(*) > 2",
+    "SELECT f.a, d.label FROM flow f JOIN dims d ON f.a = d.k",
+]
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+@pytest.mark.parametrize("sql", NULL_QUERIES)
+def test_differential_nulls_vs_sqlite(seed, sql):
+    rng = random.Random(seed * 77 + zlib.crc32(sql.encode()) % 991)
+    n = rng.choice([1, 23, 150])
+    data = _random_table(rng, n, nulls=True)
+    dims = {"k": list(range(8)), "label": [f"L{i}" for i in range(8)]}
+    flow = MessageBatch.from_dict(data)
+    dims_b = MessageBatch.from_dict(dims)
+    ours_b = SqlExecutor(sql).execute({"flow": flow, "dims": dims_b})
+    ours = _normalize([tuple(r.values()) for r in ours_b.to_rows()])
+    theirs = _normalize(_sqlite_exec(data, dims, sql))
+    if "ORDER BY" not in sql:
+        ours = sorted(ours, key=repr)
+        theirs = sorted(theirs, key=repr)
+    assert ours == theirs, f"{sql}\nseed={seed} n={n}\n" \
+                           f"{ours[:6]} vs {theirs[:6]}"
