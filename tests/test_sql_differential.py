@@ -64,6 +64,11 @@ QUERIES = [
     "SELECT a, dense_rank() OVER (ORDER BY k, c) AS dr FROM flow "
     "ORDER BY a, dr",
     "SELECT a, min(a) OVER (ORDER BY b) AS lo FROM flow ORDER BY b, a",
+    "SELECT a, b FROM flow WHERE a < 20 UNION ALL SELECT b, a FROM flow "
+    "WHERE a > 90",
+    "SELECT a FROM flow ORDER BY a LIMIT 1000 OFFSET 190",
+    "SELECT a FROM flow ORDER BY a LIMIT 5 OFFSET 10000",
+    "SELECT count(*) AS n FROM flow WHERE 1 = 0",
 ]
 
 
