@@ -317,3 +317,49 @@ def test_control_plane_apply_and_rollback(run):
         await eng.runtime.stop_all()
 
     run(main(), timeout=30)
+
+
+def test_input_disconnect_reconnect(run, monkeypatch):
+    """DisconnectionError → backoff, reconnect, resume reading
+    (reference stream/mod.rs:289-306)."""
+    from arkflow_amd import stream as stream_mod
+    from arkflow_amd.errors import DisconnectionError
+    from arkflow_amd.pipeline import Pipeline
+    from arkflow_amd.spi import Input, NoopAck
+    from arkflow_amd.stream import Stream
+
+    monkeypatch.setattr(stream_mod, "RECONNECT_SECS", 0.05)
+
+    class FlakyInput(Input):
+        def __init__(self):
+            self.reads = 0
+            self.connects = 0
+
+        async def connect(self):
+            self.connects += 1
+
+        async def read(self):
+            self.reads += 1
+            if self.reads == 2:
+                raise DisconnectionError("broker gone")
+            if self.reads > 4:
+                from arkflow_amd.errors import EOFError_
+                raise EOFError_("done")
+            return _mk([float(self.reads)]), NoopAck()
+
+        async def close(self):
+            pass
+
+    async def main():
+        inp = FlakyInput()
+        out = CountingOutput()
+        sc = StreamConfig(id="r", input={"type": "memory"},
+                          output={"type": "drop"},
+                          pipeline=PipelineConfig(thread_num=1))
+        s = Stream(sc, inp, Pipeline([]), out)
+        await asyncio.wait_for(s.run(asyncio.Event()), 15)
+        assert s.metrics.input_reconnects == 1
+        assert inp.connects >= 2  # initial + reconnect
+        assert out.rows == 3  # reads 1, 3, 4 delivered
+
+    run(main())
