@@ -46,6 +46,9 @@ def main(argv: Optional[List[str]] = None) -> int:
         description="MI355X-native stream processing engine")
     sub = parser.add_subparsers(dest="command")
 
+    from . import __version__
+    parser.add_argument("--version", action="version",
+                        version=f"arkflow_amd {__version__}")
     run_p = sub.add_parser("run", help="run the engine (default)")
     for p in (parser, run_p):
         p.add_argument("--config", "-c", help="YAML/JSON/TOML config file")
