@@ -367,3 +367,15 @@ streams:
         capture_output=True, text=True, timeout=120, env=env)
     assert r.returncode == 0, r.stderr[-500:]
     assert r.stdout.count('{"n":8}') == 4
+
+
+def test_gpu_ops_fail_loudly_without_extension(monkeypatch):
+    """On a GPU box with no built extension, ops must raise — never fall
+    back to eager silently (driver checks which .so files loaded)."""
+    from arkflow_amd import ops
+    from arkflow_amd.errors import GpuExtensionMissing
+    monkeypatch.setattr(ops, "_native", None)
+    monkeypatch.setattr(ops, "_native_err", "not built (simulated)")
+    with pytest.raises(GpuExtensionMissing) as ei:
+        ops.require_native()
+    assert "build_ext" in str(ei.value)  # tells the operator how to fix it
