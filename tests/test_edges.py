@@ -379,3 +379,44 @@ def test_gpu_ops_fail_loudly_without_extension(monkeypatch):
     with pytest.raises(GpuExtensionMissing) as ei:
         ops.require_native()
     assert "build_ext" in str(ei.value)  # tells the operator how to fix it
+
+
+def test_cli_sigint_graceful_shutdown(tmp_path):
+    """SIGINT → cancel → streams stop cleanly → exit 0 (reference
+    engine/mod.rs run_with_cancellation signal handling)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+    cfg = tmp_path / "c.yaml"
+    out_path = tmp_path / "out.jsonl"
+    cfg.write_text(f"""
+streams:
+  - id: forever
+    input:
+      type: generate
+      batch_size: 4
+      interval: 10ms
+      fields:
+        v: {{dtype: float32}}
+    output:
+      type: file
+      path: {out_path}
+""")
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "arkflow_amd", "--config", str(cfg)],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.PIPE, text=True)
+    for _ in range(300):  # wait until the stream demonstrably runs
+        if out_path.exists() and out_path.stat().st_size > 0:
+            break
+        time.sleep(0.1)
+    else:
+        proc.kill()
+        raise AssertionError("stream never produced output")
+    proc.send_signal(signal.SIGINT)
+    rc = proc.wait(timeout=30)
+    assert rc == 0, proc.stderr.read()[-400:]
