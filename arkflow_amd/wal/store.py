@@ -165,8 +165,8 @@ def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:
     n = len(buf)
     while pos + 12 <= n:
         seq, ln = struct.unpack_from(">QI", buf, pos)
-        if pos + 12 + ln + 4 > n:
-            return
+        if ln == 0 or pos + 12 + ln + 4 > n:
+            return  # zero-filled (preallocated mmap) or torn tail
         payload = buf[pos + 12: pos + 12 + ln]
         (crc,) = struct.unpack_from(">I", buf, pos + 12 + ln)
         if zlib.crc32(payload) != crc:
@@ -320,6 +320,142 @@ class LocalWalStore:
             self._f.close()
         except OSError:
             pass
+
+
+class MmapWalStore(LocalWalStore):
+    """Memory-mapped log: appends are memcpy into a preallocated mapping +
+    msync — per-entry durability at ~100 µs instead of multi-ms fsync
+    (the redb-analog fast path; same frame format, cursor file, compaction
+    and recovery as the local store — a zeroed tail parses as torn).
+    """
+
+    CHUNK = 64 * 1024 * 1024
+
+    def __init__(self, path: str, stream_id: str = "stream",
+                 compress: bool = False, fsync: bool = True,
+                 chunk_bytes: int = 0):
+        import mmap as _mmap
+        self._mmap_mod = _mmap
+        if chunk_bytes:
+            self.CHUNK = int(chunk_bytes)
+        super().__init__(path, stream_id, compress=compress, fsync=fsync)
+        # find the write position: end of the last valid frame
+        self._f.flush()
+        with open(self.log_path, "rb") as f:
+            buf = f.read()
+        pos = 0
+        for seq, payload in decode_frames(buf):
+            pos += 12 + len(payload) + 1 + 4 + (0 if _nwal else 0)
+        # recompute precisely (compressed payload length may differ)
+        pos = _end_of_frames(buf)
+        self._pos = pos
+        self._f.close()
+        self._f = open(self.log_path, "r+b")  # mmap needs read-write
+        self._remap(max(self.CHUNK, pos + self.CHUNK))
+
+    def _remap(self, size: int) -> None:
+        self._f.flush()
+        os.ftruncate(self._f.fileno(), size)
+        self._map = self._mmap_mod.mmap(self._f.fileno(), size)
+        self._size = size
+
+    def _write_frames(self, blob: bytes, sync: bool) -> None:
+        with self._lock:
+            end = self._pos + len(blob)
+            if end > self._size:
+                self._map.flush()
+                self._map.close()
+                self._remap(max(self._size * 2, end + self.CHUNK))
+            self._map[self._pos:end] = blob
+            if sync and self.fsync:
+                page = 4096
+                lo = (self._pos // page) * page
+                self._map.flush(lo, ((end - lo + page - 1) // page) * page)
+            self._pos = end
+
+    def append_batch(self, entries, sync: bool = True) -> None:
+        if _nwal is not None and not self.compress:
+            self._write_frames(_nwal.encode_frames(entries), sync)
+            return
+        blob = b"".join(encode_frame(s, p, self.compress)
+                        for s, p in entries)
+        self._write_frames(blob, sync)
+
+    def append_framed(self, frames, sync: bool = True) -> None:
+        self._write_frames(b"".join(frames), sync)
+
+    def read_after(self, cursor: int):
+        self._map.flush()
+        buf = bytes(self._map[:self._pos]) if self._pos else b""
+        for seq, payload in decode_frames(buf):
+            if seq > cursor:
+                yield seq, payload
+
+    def _scan_max_seq(self) -> int:
+        mx = 0
+        try:
+            with open(self.log_path, "rb") as f:
+                buf = f.read()
+            for seq, _ in decode_frames(buf):
+                mx = max(mx, seq)
+        except OSError:
+            pass
+        return mx
+
+    def _compact_locked(self) -> None:
+        # rewrite live tail, then remap
+        keep = [(s, p) for s, p in decode_frames(bytes(self._map[:self._pos]))
+                if s > self._cursor]
+        self._map.flush()
+        self._map.close()
+        with open(self.log_path + ".tmp", "wb") as f:
+            for s, p in keep:
+                f.write(encode_frame(s, p, self.compress))
+        os.replace(self.log_path + ".tmp", self.log_path)
+        self._f.close()
+        self._f = open(self.log_path, "r+b")
+        with open(self.log_path, "rb") as f:
+            self._pos = _end_of_frames(f.read())
+        self._remap(max(self.CHUNK, self._pos + self.CHUNK))
+
+    def close(self) -> None:
+        try:
+            self._map.flush()
+            self._map.close()
+            os.ftruncate(self._f.fileno(), self._pos)
+            self._f.close()
+        except (OSError, ValueError):
+            pass
+
+
+def _end_of_frames(buf: bytes) -> int:
+    """Byte offset just past the last valid frame."""
+    pos = 0
+    n = len(buf)
+    while pos + 12 <= n:
+        seq, ln = struct.unpack_from(">QI", buf, pos)
+        if ln == 0 or pos + 12 + ln + 4 > n:
+            break
+        payload = buf[pos + 12: pos + 12 + ln]
+        (crc,) = struct.unpack_from(">I", buf, pos + 12 + ln)
+        if zlib.crc32(payload) != crc:
+            break
+        pos += 12 + ln + 4
+    return pos
+
+
+@register("wal_store", "mmap",
+          description="Memory-mapped WAL store: msync-based per-entry "
+                      "durability (~100 us vs multi-ms fsync)",
+          example={"type": "mmap", "path": "./wal"})
+def _build_mmap_store(config: dict, resource=None) -> MmapWalStore:
+    return MmapWalStore(
+        config.get("path", "./wal"),
+        stream_id=config.get("stream_id", "stream"),
+        compress=bool(config.get("compress", False)),
+        fsync=bool(config.get("fsync", True)),
+        chunk_bytes=int(config.get("chunk_bytes", 0)),
+    )
 
 
 @register("wal_store", "local",
