@@ -309,7 +309,7 @@ def test_every_component_example_builds(tmp_path):
             example = dict(md.example or {"type": name})
             example.setdefault("type", name)
             if kind == "wal_store":
-                example.setdefault("path", str(tmp_path / name))
+                example["path"] = str(tmp_path / name)  # never write ./wal
             if (kind, name) in skip_build:
                 continue
             comp = build_component(kind, example)
