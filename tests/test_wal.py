@@ -271,8 +271,8 @@ def test_concurrent_cursor_writes_race_free(tmp_path):
         st = build_component("wal_store", {**spec, "stream_id": "r"})
         st.append_batch([(i, b"x" * 100) for i in range(1, 65)], True)
         with ThreadPoolExecutor(16) as pool:
-            list(pool.map(st.write_cursor, list(range(1, 200)) * 4))
-        assert st.cursor == 199
+            list(pool.map(st.write_cursor, list(range(1, 80)) * 4))
+        assert st.cursor == 79
         st.close()
 
 
