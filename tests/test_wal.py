@@ -374,3 +374,49 @@ def test_mmap_store_roundtrip_and_durability(tmp_path, run):
         await wal2.close()
 
     run(main())
+
+
+def test_open_window_drains_on_graceful_stop(tmp_path, run):
+    """A graceful stop drains the open window (delivered + acked, nothing to
+    replay); only a hard crash (no close — covered by
+    test_stream_recovery_replay_order) leaves entries for the WAL.
+    Reference §5: window contents are volatile, the WAL holds un-acked
+    source batches."""
+    from arkflow_amd.buffers.windows import SlidingWindowBuffer
+    from arkflow_amd.pipeline import Pipeline
+    from arkflow_amd.stream import Stream
+    from tests.test_stream_engine import CountingOutput, StubInput
+
+    async def main():
+        cfg = DurabilityConfig(enabled=True, path=str(tmp_path),
+                               sync_policy="per_entry")
+        wal = Wal.open(cfg, "w")
+        buf = SlidingWindowBuffer({"window_size": 100, "slide_size": 100})
+        sc = StreamConfig(id="w", input={"type": "memory"},
+                          output={"type": "drop"},
+                          pipeline=PipelineConfig(thread_num=1))
+
+        class SlowEOFInput(StubInput):
+            async def read(self):
+                if not self.batches:
+                    await asyncio.sleep(10)  # stay open; cancel fires first
+                return await super().read()
+
+        inp = SlowEOFInput([_mk([i]) for i in range(5)])
+        out = CountingOutput()
+        s = Stream(sc, inp, Pipeline([]), out, wal=wal, buffer=buf)
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(s.run(cancel))
+        await asyncio.sleep(0.5)  # batches ingested into the open window
+        cancel.set()
+        await asyncio.wait_for(task, 15)
+        # drained on stop: one combined window batch, all rows, all acked
+        assert out.rows == 5
+        # cursor advanced → nothing replays on restart
+        wal2 = Wal.open(cfg, "w")
+        out2 = CountingOutput()
+        s2 = Stream(sc, StubInput([]), Pipeline([]), out2, wal=wal2)
+        await asyncio.wait_for(s2.run(asyncio.Event()), 15)
+        assert out2.rows == 0
+
+    run(main())
