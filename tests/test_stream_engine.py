@@ -363,3 +363,55 @@ def test_input_disconnect_reconnect(run, monkeypatch):
         assert out.rows == 3  # reads 1, 3, 4 delivered
 
     run(main())
+
+
+def test_many_streams_one_engine(run, tmp_path):
+    """Four concurrent streams (different buffers/durability) share one
+    engine: all progress, stop cleanly, no cross-talk."""
+    async def main():
+        cfg = EngineConfig.from_dict({"streams": [
+            {"id": "plain",
+             "input": {"type": "generate", "batch_size": 64,
+                       "interval": "2ms",
+                       "fields": {"v": {"dtype": "float32"}}},
+             "output": {"type": "memory"}},
+            {"id": "durable",
+             "input": {"type": "generate", "batch_size": 64,
+                       "interval": "2ms",
+                       "fields": {"v": {"dtype": "float32"}}},
+             "durability": {"enabled": True, "path": str(tmp_path),
+                            "sync_policy": "group_commit"},
+             "output": {"type": "drop"}},
+            {"id": "windowed",
+             "input": {"type": "generate", "batch_size": 64,
+                       "interval": "2ms",
+                       "fields": {"k": {"dtype": "int64", "low": 0,
+                                        "high": 8},
+                                  "v": {"dtype": "float32"}}},
+             "buffer": {"type": "tumbling_window", "interval": "100ms"},
+             "pipeline": {"processors": [
+                 {"type": "sql",
+                  "query": "SELECT k, count(*) AS n FROM flow GROUP BY k"}]},
+             "output": {"type": "drop"}},
+            {"id": "exprs",
+             "input": {"type": "generate", "batch_size": 64,
+                       "interval": "2ms",
+                       "fields": {"v": {"dtype": "float32"}}},
+             "pipeline": {"processors": [
+                 {"type": "vrl", "statement": ".v2 = .v * 2"}]},
+             "output": {"type": "drop"}},
+        ]})
+        eng = af.Engine(cfg)
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(eng.run_with_cancellation(cancel))
+        await asyncio.sleep(2.0)
+        cancel.set()
+        await asyncio.wait_for(task, 60)
+        for sid in ("plain", "durable", "windowed", "exprs"):
+            e = eng.runtime.get(sid)
+            assert e.state.value == "stopped", sid
+            assert e.metrics.input_messages > 0, sid
+            assert e.metrics.processing_errors == 0, sid
+            assert e.metrics.output_errors == 0, sid
+
+    run(main(), timeout=90)
