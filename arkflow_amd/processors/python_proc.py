@@ -21,12 +21,13 @@ from ..spi import Processor
 
 def _to_pyarrow(batch: MessageBatch):
     import pyarrow as pa
+    import torch
     arrays = {}
     for name, col in batch.columns.items():
         if col.kind == "numeric":
             t = col.data.detach().cpu()
             if t.dtype.is_floating_point and t.dtype not in (
-                    __import__("torch").float32, __import__("torch").float64):
+                    torch.float32, torch.float64):
                 t = t.float()
             arrays[name] = pa.array(t.numpy())
         else:
