@@ -69,6 +69,8 @@ QUERIES = [
     "SELECT a FROM flow ORDER BY a LIMIT 1000 OFFSET 190",
     "SELECT a FROM flow ORDER BY a LIMIT 5 OFFSET 10000",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
+    "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
+    "ORDER BY k, a DESC, rs",
 ]
 
 
