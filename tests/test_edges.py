@@ -420,3 +420,24 @@ streams:
     proc.send_signal(signal.SIGINT)
     rc = proc.wait(timeout=30)
     assert rc == 0, proc.stderr.read()[-400:]
+
+
+def test_cli_schema_and_components_json():
+    """`schema` and `components list` emit valid JSON with every kind."""
+    import json
+    import subprocess
+    import sys
+    import os
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run([sys.executable, "-m", "arkflow_amd", "schema"],
+                       capture_output=True, text=True, env=env, timeout=120)
+    schema = json.loads(r.stdout)
+    assert schema["title"] and "streams" in schema["properties"]
+    r = subprocess.run([sys.executable, "-m", "arkflow_amd", "components",
+                        "show", "input", "generate"],
+                       capture_output=True, text=True, env=env, timeout=120)
+    md = json.loads(r.stdout)
+    assert md["config_schema"]["properties"]["batch_size"]["type"] == \
+        "integer"
