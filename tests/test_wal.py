@@ -420,3 +420,15 @@ def test_open_window_drains_on_graceful_stop(tmp_path, run):
         assert out2.rows == 0
 
     run(main())
+
+
+def test_frame_format_golden_bytes():
+    """Pin the on-disk frame layout: [seq u64 BE|len u32 BE|tag|body|crc32 BE].
+    Breaking this silently would orphan existing WALs."""
+    from arkflow_amd.wal.store import decode_frames, encode_frame
+    golden = bytes.fromhex("00000000000000070000000752676f6c64656eab1b0064")
+    assert encode_frame(7, b"golden") == golden
+    assert list(decode_frames(golden)) == [(7, b"golden")]
+    nwal = pytest.importorskip("arkflow_amd._wal_native")
+    assert nwal.encode_frames([(7, b"golden")]) == golden
+    assert nwal.encode_frame_parts(7, [b"gol", b"den"]) == golden
