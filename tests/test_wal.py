@@ -432,3 +432,23 @@ def test_frame_format_golden_bytes():
     nwal = pytest.importorskip("arkflow_amd._wal_native")
     assert nwal.encode_frames([(7, b"golden")]) == golden
     assert nwal.encode_frame_parts(7, [b"gol", b"den"]) == golden
+
+
+def test_batch_payload_golden_bytes():
+    """Pin the batch serialization layout (magic, input name, per-column
+    dtype/kind headers, buffers)."""
+    import torch
+    from arkflow_amd.batch import Column
+    from arkflow_amd.wal.store import deserialize_batch, serialize_batch
+    b = MessageBatch(
+        {"v": Column.from_numeric(torch.tensor([1.0], dtype=torch.float32)),
+         "s": Column.from_strings(["x"])}, input_name="in")
+    golden = bytes.fromhex(
+        "4157414c0002696e00000002000176000000070000000000000001000000"
+        "04666c6f617433320000803f000173010000000000000000010000000100"
+        "0000107800000000000000000100000000000000")
+    assert serialize_batch(b) == golden
+    out = deserialize_batch(golden)
+    assert out.input_name == "in"
+    assert out.column("v").to_pylist() == [1.0]
+    assert out.column("s").to_strlist() == ["x"]
