@@ -415,3 +415,28 @@ def test_many_streams_one_engine(run, tmp_path):
             assert e.metrics.output_errors == 0, sid
 
     run(main(), timeout=90)
+
+
+def test_lifecycle_timeout_marks_operation(run):
+    """A lifecycle op that exceeds its timeout is recorded as TIMEOUT
+    (reference control_plane.rs op bookkeeping)."""
+    async def main():
+        cfg = EngineConfig.from_dict({"streams": [{
+            "id": "s1",
+            "input": {"type": "generate", "batch_size": 1, "interval": "50ms",
+                      "fields": {"v": {"dtype": "float32"}}},
+            "output": {"type": "drop"}}]})
+        eng = af.Engine(cfg)
+        for sc in cfg.streams:
+            eng.runtime.register(sc)
+
+        async def hang(stream_id):
+            await asyncio.sleep(60)
+
+        eng.runtime.start = hang  # simulate a stuck start
+        r = await eng.control_plane.lifecycle("s1", "start", timeout=0.1)
+        assert r["state"] == "timed_out"
+        op = eng.runtime.operations.get(r["id"])
+        assert op.state.value == "timed_out"
+
+    run(main(), timeout=30)
