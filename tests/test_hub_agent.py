@@ -341,3 +341,18 @@ server:
         hub.terminate()
         node.wait(timeout=15)
         hub.wait(timeout=15)
+
+
+def test_rollout_cancel_stops_advance(run):
+    async def main():
+        hub = Hub()
+        rid = await hub.create_rollout({"streams": []}, ["n1", "n2", "n3"])
+        await hub.step_rollout(rid)
+        r = await hub.control_rollout(rid, "cancel")
+        assert r["state"] == "cancelled"
+        r2 = await hub.step_rollout(rid)
+        assert r2["position"] == 1  # cancelled → frozen
+        with pytest.raises(Exception):
+            await hub.control_rollout(rid, "nonsense")
+
+    run(main())
