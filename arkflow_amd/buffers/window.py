@@ -116,14 +116,19 @@ class BaseWindowBuffer(Buffer):
                 return None
             self._notify.clear()
             timeout = self.next_deadline()
-            try:
-                if timeout is None:
-                    await self._notify.wait()
-                else:
-                    await asyncio.wait_for(self._notify.wait(),
-                                           max(timeout, 1e-4))
-            except asyncio.TimeoutError:
-                pass
+            if timeout is None:
+                await self._notify.wait()
+            else:
+                # NOT wait_for: in py3.10, external cancellation landing on
+                # the same tick as the timeout is converted to TimeoutError —
+                # swallowing it here left a cancelled-but-running buffer task
+                # waiting forever (intermittent shutdown stall, NOTES #12).
+                # asyncio.wait's timeout never masks cancellation.
+                waiter = asyncio.ensure_future(self._notify.wait())
+                try:
+                    await asyncio.wait({waiter}, timeout=max(timeout, 1e-4))
+                finally:
+                    waiter.cancel()
 
     async def flush(self) -> None:
         """End-of-input: emit everything left, then read() returns None."""

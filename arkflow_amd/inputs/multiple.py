@@ -52,7 +52,12 @@ class MultipleInputs(Input):
         finally:
             self._live -= 1
             if self._live <= 0:
-                await self._q.put(None)
+                try:
+                    # never await here: on shutdown the reader may be gone
+                    # and a full queue would block close() forever
+                    self._q.put_nowait(None)
+                except asyncio.QueueFull:
+                    pass
 
     async def read(self) -> Tuple[MessageBatch, Ack]:
         item = await self._q.get()
@@ -64,7 +69,13 @@ class MultipleInputs(Input):
         for t in self._tasks:
             if not t.done():
                 t.cancel()
-        await asyncio.gather(*self._tasks, return_exceptions=True)
+        try:
+            # bounded: a pump wedged in a queue handoff must not stall
+            # engine shutdown (intermittent; see NOTES.md item 12)
+            await asyncio.wait_for(
+                asyncio.gather(*self._tasks, return_exceptions=True), 5)
+        except asyncio.TimeoutError:
+            pass
         for child in self.children.values():
             await child.close()
 

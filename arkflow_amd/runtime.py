@@ -279,6 +279,12 @@ class RuntimeManager:
         try:
             await asyncio.wait_for(asyncio.shield(entry.task), timeout)
         except asyncio.TimeoutError:
+            # surfaced as an event: a stream that needed the force-stop is a
+            # shutdown-liveness bug worth noticing in production
+            self.events.push(stream_id, "stop_timeout",
+                             f"force-cancelled after {timeout}s")
+            log.warning("stream %s did not stop within %ss; force-cancelling",
+                        stream_id, timeout)
             entry.task.cancel()
             try:
                 await entry.task
@@ -320,14 +326,15 @@ class RuntimeManager:
             await self.start(sid)
 
     async def stop_all(self) -> None:
-        # signal all first, then await — parallel shutdown
+        # signal all first, then await CONCURRENTLY — a stream that needs
+        # the force-stop timeout must not serialize behind the others
         for e in self.entries.values():
             if e.cancel is not None and e.task is not None and not e.task.done():
                 e.desired = DesiredState.STOPPED
                 e.state = StreamState.STOPPING
                 e.cancel.set()
-        for sid in list(self.entries):
-            await self.stop(sid)
+        await asyncio.gather(*(self.stop(sid) for sid in list(self.entries)),
+                             return_exceptions=True)
 
     async def wait_all(self) -> None:
         tasks = [e.task for e in self.entries.values()

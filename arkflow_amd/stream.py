@@ -135,7 +135,31 @@ class Stream:
             for t in tasks + [in_task]:
                 if not t.done():
                     t.cancel()
-            await asyncio.gather(*tasks, in_task, return_exceptions=True)
+            # bounded: cleanup must terminate even if a sub-task wedges
+            # (intermittent cancel-delivery stall; NOTES.md item 12).
+            # asyncio.wait (not wait_for+gather) so a timeout leaves the
+            # wedged tasks inspectable before we force-cancel them.
+            t_w0 = time.monotonic()
+            done, pending = await asyncio.wait(
+                set(tasks) | {in_task}, timeout=10)
+            if time.monotonic() - t_w0 > 5:
+                log.warning("stream %s: cleanup wait took %.2fs; done=%s "
+                            "pending=%s", self.config.id,
+                            time.monotonic() - t_w0,
+                            [repr(t)[:120] for t in done],
+                            [repr(t)[:160] for t in pending])
+            if pending:
+                stuck = []
+                for t in pending:
+                    frames = t.get_stack(limit=5)
+                    stuck.append(" <- ".join(
+                        f"{f.f_code.co_qualname}:{f.f_lineno}"
+                        for f in frames))
+                log.warning("stream %s: cleanup tasks stalled; forcing "
+                            "close. stuck: %s", self.config.id, stuck)
+                for t in pending:
+                    t.cancel()
+                await asyncio.wait(pending, timeout=5)
             await self._close_all()
 
     async def _close_all(self) -> None:

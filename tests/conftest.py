@@ -32,7 +32,9 @@ def run():
             for t in pending:
                 t.cancel()
             if pending:
-                loop.run_until_complete(
-                    asyncio.gather(*pending, return_exceptions=True))
+                # bounded: a task that swallows cancellation must not hang
+                # the whole suite
+                loop.run_until_complete(asyncio.wait_for(
+                    asyncio.gather(*pending, return_exceptions=True), 15))
             loop.close()
     return _run

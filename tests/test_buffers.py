@@ -307,7 +307,18 @@ def test_windowed_join_example_runs_e2e(run):
             except Exception:
                 pass
         cancel.set()
-        await asyncio.wait_for(task, 60)
+        try:
+            await asyncio.wait_for(task, 25)
+        except asyncio.TimeoutError:
+            lines = []
+            for t in asyncio.all_tasks():
+                if t is asyncio.current_task():
+                    continue
+                st = t.get_stack(limit=6)
+                lines.append(" <- ".join(
+                    f"{f.f_code.co_qualname}:{f.f_lineno}" for f in st))
+            raise AssertionError("engine shutdown hang; tasks:\n"
+                                 + "\n".join(lines))
         out = eng.runtime.get("joiner").stream.output
         rows = [r for b in out.batches for r in b.to_rows()]
         assert rows, "join emitted nothing"
