@@ -1,0 +1,25 @@
+"""Small asyncio utilities.
+
+`event_wait` exists because CPython 3.10's `asyncio.wait_for` converts an
+external cancellation that lands on the same tick as its timeout into
+TimeoutError. Loops that catch TimeoutError around
+`wait_for(event.wait(), t)` therefore swallow cancellation and keep running
+(this produced an intermittent engine-shutdown stall; NOTES.md #12).
+`asyncio.wait`'s timeout never masks cancellation.
+"""
+from __future__ import annotations
+
+import asyncio
+
+
+async def event_wait(ev: asyncio.Event, timeout: float) -> bool:
+    """Wait for `ev` up to `timeout` seconds; True if it was set.
+    Cancellation always propagates (unlike wait_for on py3.10)."""
+    if ev.is_set():
+        return True
+    waiter = asyncio.ensure_future(ev.wait())
+    try:
+        await asyncio.wait({waiter}, timeout=timeout)
+        return waiter.done() and not waiter.cancelled()
+    finally:
+        waiter.cancel()

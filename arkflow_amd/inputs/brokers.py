@@ -179,11 +179,9 @@ class KafkaInput(Input):
                         batch = apply_codec(batch, self.codec)
                     return batch, KafkaAck(self.bus, self.group, topic, p, off)
             self.bus.notify.clear()
-            try:
-                await asyncio.wait_for(self.bus.notify.wait(), 0.5)
-            except asyncio.TimeoutError:
-                if self._closed:
-                    raise EOFError_("kafka input closed") from None
+            from ..aio import event_wait
+            if not await event_wait(self.bus.notify, 0.5) and self._closed:
+                raise EOFError_("kafka input closed")
 
     async def close(self) -> None:
         self._closed = True
@@ -395,10 +393,8 @@ class RedisInput(_RedisPubSubIn):
                     batch = apply_codec(batch, self.codec)
                 return batch, NoopAck()
             self.bus.notify.clear()
-            try:
-                await asyncio.wait_for(self.bus.notify.wait(), 0.5)
-            except asyncio.TimeoutError:
-                pass
+            from ..aio import event_wait
+            await event_wait(self.bus.notify, 0.5)
 
 
 # ---- registrations ------------------------------------------------------------

@@ -118,10 +118,8 @@ class Agent:
                 await self._poll_and_execute()
             except Exception:  # noqa: BLE001
                 log.exception("agent tick failed")
-            try:
-                await asyncio.wait_for(cancel.wait(), self.poll_interval)
-            except asyncio.TimeoutError:
-                pass
+            from ..aio import event_wait
+            await event_wait(cancel, self.poll_interval)
 
 
 async def agent_run(engine, cancel: asyncio.Event) -> None:

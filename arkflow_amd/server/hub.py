@@ -285,7 +285,5 @@ async def hub_background(hub: Hub, cancel: asyncio.Event,
         if time.monotonic() - last_sweep >= sweep_interval:
             await hub.sweep()
             last_sweep = time.monotonic()
-        try:
-            await asyncio.wait_for(cancel.wait(), poll_interval)
-        except asyncio.TimeoutError:
-            pass
+        from ..aio import event_wait
+        await event_wait(cancel, poll_interval)

@@ -441,3 +441,27 @@ def test_cli_schema_and_components_json():
     md = json.loads(r.stdout)
     assert md["config_schema"]["properties"]["batch_size"]["type"] == \
         "integer"
+
+
+def test_event_wait_never_masks_cancellation(run):
+    """aio.event_wait: cancellation wins even when it races the timeout
+    (the wait_for pitfall behind the shutdown-stall fix)."""
+    from arkflow_amd.aio import event_wait
+
+    async def main():
+        ev = asyncio.Event()
+
+        async def loop_like_buffer():
+            while True:
+                await event_wait(ev, 0.01)  # timeout fires constantly
+
+        t = asyncio.ensure_future(loop_like_buffer())
+        for _ in range(20):  # hammer the race window
+            await asyncio.sleep(0.0101)
+        t.cancel()
+        with pytest.raises(asyncio.CancelledError):
+            await asyncio.wait_for(t, 2)  # must die promptly, never wedge
+        ev.set()
+        assert await event_wait(ev, 0.1) is True
+
+    run(main())
