@@ -23,3 +23,16 @@ async def event_wait(ev: asyncio.Event, timeout: float) -> bool:
         return waiter.done() and not waiter.cancelled()
     finally:
         waiter.cancel()
+
+
+async def queue_get(q: "asyncio.Queue", timeout: float):
+    """Get from `q` with a timeout; returns (True, item) or (False, None).
+    Cancellation always propagates (same rationale as event_wait)."""
+    getter = asyncio.ensure_future(q.get())
+    try:
+        await asyncio.wait({getter}, timeout=timeout)
+        if getter.done() and not getter.cancelled():
+            return True, getter.result()
+        return False, None
+    finally:
+        getter.cancel()

@@ -120,14 +120,15 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
                 yield f"id: {ev.seq}\ndata: {json.dumps(ev.to_dict())}\n\n"
             q = engine.runtime.events.subscribe()
             try:
+                from ..aio import queue_get
                 while True:
                     if await request.is_disconnected():
                         return
-                    try:
-                        ev = await asyncio.wait_for(q.get(), timeout=5.0)
+                    got, ev = await queue_get(q, 5.0)
+                    if got:
                         yield (f"id: {ev.seq}\n"
                                f"data: {json.dumps(ev.to_dict())}\n\n")
-                    except asyncio.TimeoutError:
+                    else:
                         yield ": keepalive\n\n"
             finally:
                 engine.runtime.events.unsubscribe(q)

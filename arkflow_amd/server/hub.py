@@ -229,11 +229,12 @@ def create_hub_app(hub: Hub) -> FastAPI:
             q: asyncio.Queue = asyncio.Queue(maxsize=256)
             hub._event_subs.append(q)
             try:
+                from ..aio import queue_get
                 while not await request.is_disconnected():
-                    try:
-                        ev = await asyncio.wait_for(q.get(), timeout=5.0)
+                    got, ev = await queue_get(q, 5.0)
+                    if got:
                         yield f"data: {json.dumps(ev)}\n\n"
-                    except asyncio.TimeoutError:
+                    else:
                         yield ": keepalive\n\n"
             finally:
                 hub._event_subs.remove(q)
