@@ -71,6 +71,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     hub_p.add_argument("--lease-ttl", type=float, default=15.0)
     hub_p.add_argument("--operator-token", action="append", default=[],
                        help="token:role (role in admin|operator|viewer)")
+    hub_p.add_argument("--registration-token", default=None,
+                       help="shared secret agents must present to register")
 
     args = parser.parse_args(argv)
 
@@ -151,7 +153,8 @@ def _run_hub(args) -> int:
         token, _, role = spec.partition(":")
         tokens[token] = role or "operator"
     hub = Hub(HubStore(args.store), lease_ttl=args.lease_ttl,
-              operator_tokens=tokens)
+              operator_tokens=tokens,
+              registration_token=args.registration_token)
     app = create_hub_app(hub)
     host, _, port = args.address.partition(":")
 

@@ -94,7 +94,8 @@ class HttpInput(Input):
                 if self.lockout.locked(client):
                     return web.Response(status=429, text="locked out")
                 auth = request.headers.get("Authorization", "")
-                if auth != f"Bearer {self.token}":
+                import hmac
+                if not hmac.compare_digest(auth, f"Bearer {self.token}"):
                     self.lockout.failure(client)
                     return web.Response(status=401, text="unauthorized")
                 self.lockout.success(client)

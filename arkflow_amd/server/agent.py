@@ -23,10 +23,12 @@ class Agent:
                  heartbeat_interval: float = 5.0,
                  report_interval: float = 5.0,
                  poll_interval: float = 1.0,
+                 registration_token: Optional[str] = None,
                  transport: Optional[httpx.AsyncBaseTransport] = None):
         self.engine = engine
         self.hub_url = hub_url.rstrip("/")
         self.node_id = node_id or socket.gethostname()
+        self.registration_token = registration_token
         self.heartbeat_interval = heartbeat_interval
         self.report_interval = report_interval
         self.poll_interval = poll_interval
@@ -47,8 +49,13 @@ class Agent:
         while True:
             try:
                 async with self._client() as c:
+                    headers = {}
+                    if self.registration_token:
+                        headers["x-registration-token"] = \
+                            self.registration_token
                     r = await c.post("/agent/register",
-                                     json={"node_id": self.node_id})
+                                     json={"node_id": self.node_id},
+                                     headers=headers)
                     r.raise_for_status()
                     self.token = r.json()["node_token"]
                     return
@@ -124,5 +131,6 @@ class Agent:
 
 async def agent_run(engine, cancel: asyncio.Event) -> None:
     cfg = engine.config.server
-    agent = Agent(engine, cfg.hub_url, cfg.node_id)
+    agent = Agent(engine, cfg.hub_url, cfg.node_id,
+                  registration_token=cfg.node_token)
     await agent.run(cancel)

@@ -36,8 +36,9 @@ def create_app(engine, server_config: Optional[ServerConfig] = None) -> FastAPI:
 
     def auth(request: Request):
         if cfg.token:
+            import hmac
             header = request.headers.get("authorization", "")
-            if header != f"Bearer {cfg.token}":
+            if not hmac.compare_digest(header, f"Bearer {cfg.token}"):
                 raise HTTPException(401, "unauthorized")
 
     # ---- health (unauthenticated, reference health endpoints) --------------

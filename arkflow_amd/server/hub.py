@@ -11,6 +11,7 @@ token RBAC (api_contract.rs:9-30).
 from __future__ import annotations
 
 import asyncio
+import hmac
 import json
 import secrets
 import time
@@ -25,20 +26,47 @@ ROLES = {"admin": {"read", "write", "rollout", "admin"},
          "operator": {"read", "write", "rollout"},
          "viewer": {"read"}}
 
+# valid rollout state transitions (reference hub.rs:1472-2052 rejects e.g.
+# resume-after-cancel); action → (allowed current states, next state)
+_ROLLOUT_ACTIONS = {
+    "pause": ({"running"}, "paused"),
+    "resume": ({"paused"}, "running"),
+    "cancel": ({"running", "paused"}, "cancelled"),
+    "rollback": ({"running", "paused", "failed", "cancelled"},
+                 "rolled_back"),
+}
+
 
 class Hub:
     def __init__(self, store: Optional[HubStore] = None,
                  lease_ttl: float = 15.0,
-                 operator_tokens: Optional[Dict[str, str]] = None):
-        """operator_tokens: token → role (reference RBAC roles/scopes)."""
+                 operator_tokens: Optional[Dict[str, str]] = None,
+                 registration_token: Optional[str] = None):
+        """operator_tokens: token → role (reference RBAC roles/scopes).
+        registration_token: shared secret an agent must present to register
+        (reference hub.rs:558 checks node_token with ct_eq at register)."""
         self.store = store or HubStore()
         self.lease_ttl = lease_ttl
         self.operator_tokens = operator_tokens or {}
+        self.registration_token = registration_token
         self._event_subs: List[asyncio.Queue] = []
         self._event_seq = 0
 
     # ---- agent protocol -------------------------------------------------------
-    async def register(self, node_id: str) -> dict:
+    async def register(self, node_id: str, reg_token: str = "",
+                       current_token: str = "") -> dict:
+        if self.registration_token is not None and not hmac.compare_digest(
+                reg_token, self.registration_token):
+            raise HTTPException(401, "bad registration token")
+        # a live-leased node can only be re-registered (token rotated) by a
+        # caller presenting the node's current token — blocks identity
+        # hijack of an online node
+        for n in await self.store.nodes():
+            if n["node_id"] == node_id and n["online"]:
+                expect = await self.store.node_token(node_id)
+                if expect and not hmac.compare_digest(current_token, expect):
+                    raise HTTPException(
+                        409, "node is live; present current token to rotate")
         token = secrets.token_hex(16)
         await self.store.upsert_node(node_id, token, self.lease_ttl)
         await self._emit(node_id, "node_registered", {})
@@ -46,7 +74,7 @@ class Hub:
 
     async def check_node(self, node_id: str, token: str) -> None:
         expect = await self.store.node_token(node_id)
-        if expect is None or expect != token:
+        if expect is None or not hmac.compare_digest(expect, token):
             raise HTTPException(401, "bad node token")
 
     async def heartbeat(self, node_id: str) -> dict:
@@ -81,20 +109,23 @@ class Hub:
 
     # ---- reconciliation ---------------------------------------------------------
     async def reconcile_once(self) -> int:
-        """intents → outbox → attempts (agent commands) (hub.rs:468-520)."""
+        """intents → outbox → attempts (agent commands) (hub.rs:468-520).
+
+        Outbox rows are claimed ONLY for online nodes — intents for offline
+        nodes stay claimable and dispatch when the node's lease returns;
+        expired attempts re-enqueue their intents (r1 correctness holes)."""
         await self.store.expire_attempts()
-        claimed = await self.store.claim_outbox()
+        online = [n["node_id"] for n in await self.store.nodes()
+                  if n["online"]]
+        claimed = await self.store.claim_outbox(online)
         dispatched = 0
-        online = {n["node_id"] for n in await self.store.nodes()
-                  if n["online"]}
         for row in claimed:
-            if row["node_id"] not in online:
-                continue  # node offline — intent stays dispatched-pending
             await self.store.create_attempt(
                 row["intent_id"], row["node_id"],
                 {"kind": "lifecycle", "stream_id": row["stream_id"],
                  "op": row["op"]})
             dispatched += 1
+        await self.advance_rollouts()
         return dispatched
 
     async def sweep(self) -> None:
@@ -104,33 +135,84 @@ class Hub:
 
     # ---- rollouts ----------------------------------------------------------------
     async def create_rollout(self, config: dict, nodes: List[str],
+                             prev_config: Optional[dict] = None,
                              actor: str = "operator") -> str:
-        rid = await self.store.create_rollout(config, nodes)
+        rid = await self.store.create_rollout(config, nodes, prev_config)
         await self.store.audit(actor, "rollout:create", rid)
         return rid
 
-    async def step_rollout(self, rid: str) -> Optional[dict]:
-        """Advance one node per step (reference staged rollouts)."""
+    async def advance_rollouts(self) -> None:
+        """Auto-advance running rollouts stage by stage with health gating
+        (reference hub.rs:1472-2052 staged auto-advance): the next node's
+        apply is dispatched only after the previous stage's attempt
+        SUCCEEDED and the next node is online; a failed or expired stage
+        fails the rollout."""
+        for r in await self.store.rollouts():
+            if r["state"] == "running":
+                await self._advance_one(r["rollout_id"])
+
+    async def _advance_one(self, rid: str) -> Optional[dict]:
         r = await self.store.get_rollout(rid)
         if r is None or r["state"] != "running":
             return r
         pos = r["position"]
+        if pos > 0:  # gate on the previous stage's attempt outcome
+            prev = await self.store.attempt_for_intent(
+                f"rollout-{rid}-{pos - 1}")
+            if prev is None or prev["state"] == "pending":
+                if prev is not None and prev["expires_at"] < time.time():
+                    await self.store.update_rollout(rid, state="failed")
+                    await self._emit(prev["node_id"], "rollout_failed",
+                                     {"rollout_id": rid, "stage": pos - 1,
+                                      "reason": "attempt expired"})
+                return await self.store.get_rollout(rid)
+            if prev["state"] in ("failed", "expired"):
+                await self.store.update_rollout(rid, state="failed")
+                await self._emit(prev["node_id"], "rollout_failed",
+                                 {"rollout_id": rid, "stage": pos - 1,
+                                  "reason": prev.get("result") or "failed"})
+                return await self.store.get_rollout(rid)
         if pos >= len(r["nodes"]):
             await self.store.update_rollout(rid, state="succeeded")
             return await self.store.get_rollout(rid)
         node = r["nodes"][pos]
+        online = {n["node_id"] for n in await self.store.nodes()
+                  if n["online"]}
+        if node not in online:
+            return r  # wait for the node's lease; do not skip the stage
         await self.store.create_attempt(
             f"rollout-{rid}-{pos}", node,
             {"kind": "apply_config", "config": r["config"]})
         await self.store.update_rollout(rid, position=pos + 1)
         return await self.store.get_rollout(rid)
 
+    async def step_rollout(self, rid: str) -> Optional[dict]:
+        """Manual single advance (kept for operators; same gating)."""
+        return await self._advance_one(rid)
+
     async def control_rollout(self, rid: str, action: str) -> Optional[dict]:
-        states = {"pause": "paused", "resume": "running",
-                  "cancel": "cancelled", "rollback": "rolled_back"}
-        if action not in states:
+        if action not in _ROLLOUT_ACTIONS:
             raise HTTPException(400, f"unknown rollout action {action}")
-        await self.store.update_rollout(rid, state=states[action])
+        r = await self.store.get_rollout(rid)
+        if r is None:
+            return None
+        allowed, next_state = _ROLLOUT_ACTIONS[action]
+        if r["state"] not in allowed:
+            raise HTTPException(
+                409, f"cannot {action} a rollout in state {r['state']}")
+        if action == "rollback":
+            # actually apply the previous config to every node already
+            # touched, newest first (reference rollback enqueues the
+            # reverse applies, not just a state flip)
+            prev_cfg = r.get("prev_config")
+            if prev_cfg is None:
+                raise HTTPException(
+                    409, "rollout has no prev_config to roll back to")
+            for pos in range(min(r["position"], len(r["nodes"])) - 1, -1, -1):
+                await self.store.create_attempt(
+                    f"rollback-{rid}-{pos}", r["nodes"][pos],
+                    {"kind": "apply_config", "config": prev_cfg})
+        await self.store.update_rollout(rid, state=next_state)
         return await self.store.get_rollout(rid)
 
     # ---- events -------------------------------------------------------------------
@@ -155,7 +237,10 @@ def create_hub_app(hub: Hub) -> FastAPI:
                 return "anonymous"
             tok = request.headers.get("authorization", "").removeprefix(
                 "Bearer ").strip()
-            role = hub.operator_tokens.get(tok)
+            role = None  # constant-time scan (reference uses ct_eq)
+            for known, r in hub.operator_tokens.items():
+                if hmac.compare_digest(known, tok):
+                    role = r
             if role is None or scope not in ROLES.get(role, set()):
                 raise HTTPException(403, "forbidden")
             return role
@@ -167,11 +252,14 @@ def create_hub_app(hub: Hub) -> FastAPI:
 
     # ---- agent API (hub.rs agent protocol) ------------------------------------
     @app.post("/agent/register")
-    async def register(body: dict):
+    async def register(body: dict, request: Request):
         node_id = body.get("node_id")
         if not node_id:
             raise HTTPException(400, "node_id required")
-        return await hub.register(node_id)
+        return await hub.register(
+            node_id,
+            reg_token=request.headers.get("x-registration-token", ""),
+            current_token=request.headers.get("x-node-token", ""))
 
     @app.post("/agent/{node_id}/heartbeat")
     async def heartbeat(node_id: str, request: Request):
@@ -248,7 +336,8 @@ def create_hub_app(hub: Hub) -> FastAPI:
     @app.post("/rollouts", dependencies=[Depends(operator("rollout"))])
     async def create_rollout(body: dict):
         rid = await hub.create_rollout(body.get("config") or {},
-                                       body.get("nodes") or [])
+                                       body.get("nodes") or [],
+                                       prev_config=body.get("prev_config"))
         return {"rollout_id": rid}
 
     @app.get("/rollouts", dependencies=[Depends(operator("read"))])

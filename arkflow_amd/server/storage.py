@@ -26,7 +26,8 @@ CREATE TABLE IF NOT EXISTS desired (
 );
 CREATE TABLE IF NOT EXISTS intents (
   intent_id TEXT PRIMARY KEY, node_id TEXT, stream_id TEXT, op TEXT,
-  state TEXT, created_at REAL, updated_at REAL, error TEXT
+  state TEXT, created_at REAL, updated_at REAL, error TEXT,
+  attempts_made INTEGER DEFAULT 0
 );
 CREATE TABLE IF NOT EXISTS outbox (
   outbox_id INTEGER PRIMARY KEY AUTOINCREMENT, intent_id TEXT, node_id TEXT,
@@ -46,9 +47,16 @@ CREATE TABLE IF NOT EXISTS audit (
 );
 CREATE TABLE IF NOT EXISTS rollouts (
   rollout_id TEXT PRIMARY KEY, state TEXT, config TEXT, nodes TEXT,
-  position INTEGER, created_at REAL, updated_at REAL
+  position INTEGER, created_at REAL, updated_at REAL,
+  prev_config TEXT DEFAULT NULL
 );
 """
+
+# pre-upgrade databases lack these columns; applied best-effort on open
+_MIGRATIONS = [
+    "ALTER TABLE intents ADD COLUMN attempts_made INTEGER DEFAULT 0",
+    "ALTER TABLE rollouts ADD COLUMN prev_config TEXT DEFAULT NULL",
+]
 
 
 class HubStore:
@@ -56,6 +64,11 @@ class HubStore:
         self._db = sqlite3.connect(path, check_same_thread=False)
         self._db.row_factory = sqlite3.Row
         self._db.executescript(_SCHEMA)
+        for mig in _MIGRATIONS:
+            try:
+                self._db.execute(mig)
+            except sqlite3.OperationalError:
+                pass  # column already exists
         self._lock = asyncio.Lock()
 
     async def _run(self, fn):
@@ -136,7 +149,7 @@ class HubStore:
         def go():
             now = time.time()
             self._db.execute(
-                "INSERT INTO intents VALUES (?,?,?,?,?,?,?,NULL)",
+                "INSERT INTO intents VALUES (?,?,?,?,?,?,?,NULL,0)",
                 (intent_id, node_id, stream_id, op, "pending", now, now))
             self._db.execute(
                 "INSERT INTO outbox(intent_id, node_id) VALUES (?,?)",
@@ -145,12 +158,21 @@ class HubStore:
         await self._run(go)
         return intent_id
 
-    async def claim_outbox(self, limit: int = 16) -> List[dict]:
+    async def claim_outbox(self, online: List[str],
+                           limit: int = 16) -> List[dict]:
+        """Claim dispatchable outbox rows — ONLY for nodes currently online
+        with a valid lease. Rows for offline nodes stay claimable so the
+        intent executes when the node returns (reference hub.rs:468-520
+        keeps outbox rows pending until the node holds a valid lease)."""
         def go():
+            if not online:
+                return []
+            marks = ",".join("?" for _ in online)
             rows = self._db.execute(
                 "SELECT o.outbox_id, o.intent_id, o.node_id, i.stream_id, "
-                "i.op FROM outbox o JOIN intents i USING (intent_id) "
-                "WHERE o.claimed=0 LIMIT ?", (limit,)).fetchall()
+                f"i.op FROM outbox o JOIN intents i USING (intent_id) "
+                f"WHERE o.claimed=0 AND o.node_id IN ({marks}) LIMIT ?",
+                (*online, limit)).fetchall()
             for r in rows:
                 self._db.execute(
                     "UPDATE outbox SET claimed=1 WHERE outbox_id=?",
@@ -170,8 +192,9 @@ class HubStore:
                 (attempt_id, intent_id, node_id, json.dumps(command),
                  "pending", now, now + ttl))
             self._db.execute(
-                "UPDATE intents SET state='dispatched', updated_at=? "
-                "WHERE intent_id=?", (now, intent_id))
+                "UPDATE intents SET state='dispatched', updated_at=?, "
+                "attempts_made=attempts_made+1 WHERE intent_id=?",
+                (now, intent_id))
             self._db.commit()
         await self._run(go)
         return attempt_id
@@ -207,14 +230,50 @@ class HubStore:
             return r["intent_id"]
         return await self._run(go)
 
-    async def expire_attempts(self) -> int:
+    async def expire_attempts(self, max_retries: int = 5) -> int:
+        """Expire timed-out pending attempts and RE-ENQUEUE their intents
+        (an outbox row becomes claimable again) until the retry budget runs
+        out, then fail the intent. Regression for the r1 hole where expired
+        attempts left intents stuck at 'dispatched' forever (reference
+        hub.rs:411 attempt expiry feeds reconciliation)."""
         def go():
             now = time.time()
-            cur = self._db.execute(
-                "UPDATE attempts SET state='expired' WHERE state='pending' "
-                "AND expires_at < ?", (now,))
+            rows = self._db.execute(
+                "SELECT attempt_id, intent_id FROM attempts WHERE "
+                "state='pending' AND expires_at < ?", (now,)).fetchall()
+            for r in rows:
+                self._db.execute(
+                    "UPDATE attempts SET state='expired' WHERE attempt_id=?",
+                    (r["attempt_id"],))
+                it = self._db.execute(
+                    "SELECT node_id, attempts_made FROM intents WHERE "
+                    "intent_id=?", (r["intent_id"],)).fetchone()
+                if it is None:
+                    continue  # rollout attempt — handled by advance_rollouts
+                if it["attempts_made"] >= max_retries:
+                    self._db.execute(
+                        "UPDATE intents SET state='failed', updated_at=?, "
+                        "error='attempt retries exhausted' WHERE intent_id=?",
+                        (now, r["intent_id"]))
+                else:
+                    self._db.execute(
+                        "UPDATE intents SET state='pending', updated_at=? "
+                        "WHERE intent_id=?", (now, r["intent_id"]))
+                    self._db.execute(
+                        "INSERT INTO outbox(intent_id, node_id) VALUES (?,?)",
+                        (r["intent_id"], it["node_id"]))
             self._db.commit()
-            return cur.rowcount
+            return len(rows)
+        return await self._run(go)
+
+    async def attempt_for_intent(self, intent_id: str) -> Optional[dict]:
+        """Latest attempt row for an intent (rollout stages use synthetic
+        intent ids, so this is how rollout health gating reads results)."""
+        def go():
+            r = self._db.execute(
+                "SELECT * FROM attempts WHERE intent_id=? "
+                "ORDER BY created_at DESC LIMIT 1", (intent_id,)).fetchone()
+            return dict(r) if r else None
         return await self._run(go)
 
     async def intents(self, limit: int = 100) -> List[dict]:
@@ -261,15 +320,18 @@ class HubStore:
         return await self._run(go)
 
     # ---- rollouts --------------------------------------------------------------
-    async def create_rollout(self, config: dict, nodes: List[str]) -> str:
+    async def create_rollout(self, config: dict, nodes: List[str],
+                             prev_config: Optional[dict] = None) -> str:
         rid = uuid.uuid4().hex[:12]
 
         def go():
             now = time.time()
             self._db.execute(
-                "INSERT INTO rollouts VALUES (?,?,?,?,?,?,?)",
+                "INSERT INTO rollouts VALUES (?,?,?,?,?,?,?,?)",
                 (rid, "running", json.dumps(config), json.dumps(nodes), 0,
-                 now, now))
+                 now, now,
+                 json.dumps(prev_config) if prev_config is not None
+                 else None))
             self._db.commit()
         await self._run(go)
         return rid
@@ -283,6 +345,8 @@ class HubStore:
             d = dict(r)
             d["config"] = json.loads(d["config"])
             d["nodes"] = json.loads(d["nodes"])
+            d["prev_config"] = json.loads(d["prev_config"]) \
+                if d.get("prev_config") else None
             return d
         return await self._run(go)
 

@@ -133,8 +133,8 @@ def deserialize_batch(buf: bytes) -> MessageBatch:
         if has_validity:
             (vlen,) = struct.unpack_from(">I", buf, pos)
             pos += 4
-            col.validity = _tensor_from(buf[pos:pos + vlen], torch.bool, -1)[
-                : len(col)]
+            col.validity = _tensor_from(buf[pos:pos + vlen], torch.bool,
+                                        len(col))[: len(col)]
             pos += vlen
         cols[cname] = col
     return MessageBatch(cols, input_name)
@@ -223,10 +223,15 @@ class LocalWalStore:
                     os.fsync(f.fileno())
             os.replace(tmp, self.cursor_path)
             try:
-                if self._f.tell() >= self.compact_bytes:
+                if self._log_size() >= self.compact_bytes:
                     self._compact_locked()
             except OSError:
                 pass
+
+    def _log_size(self) -> int:
+        """Current log size for the online-compaction trigger (mmap stores
+        append through the mapping, so file.tell() would stay 0 there)."""
+        return self._f.tell()
 
     def _compact_locked(self) -> None:
         """Rewrite the log keeping only entries past the cursor; called with
@@ -354,6 +359,12 @@ class MmapWalStore(LocalWalStore):
         self._remap(max(self.CHUNK, pos + self.CHUNK))
 
     def _remap(self, size: int) -> None:
+        # page-align the mapping: a non-aligned size (possible after
+        # recovering existing data) makes the rounded-up flush range in
+        # _write_frames fall outside the map and raise "flush values out
+        # of range" exactly on the first append after a restart
+        page = self._mmap_mod.PAGESIZE
+        size = ((size + page - 1) // page) * page
         self._f.flush()
         os.ftruncate(self._f.fileno(), size)
         self._map = self._mmap_mod.mmap(self._f.fileno(), size)
@@ -368,9 +379,10 @@ class MmapWalStore(LocalWalStore):
                 self._remap(max(self._size * 2, end + self.CHUNK))
             self._map[self._pos:end] = blob
             if sync and self.fsync:
-                page = 4096
+                page = self._mmap_mod.PAGESIZE
                 lo = (self._pos // page) * page
-                self._map.flush(lo, ((end - lo + page - 1) // page) * page)
+                ln = ((end - lo + page - 1) // page) * page
+                self._map.flush(lo, min(ln, self._size - lo))
             self._pos = end
 
     def append_batch(self, entries, sync: bool = True) -> None:
@@ -383,6 +395,9 @@ class MmapWalStore(LocalWalStore):
 
     def append_framed(self, frames, sync: bool = True) -> None:
         self._write_frames(b"".join(frames), sync)
+
+    def _log_size(self) -> int:
+        return self._pos
 
     def read_after(self, cursor: int):
         self._map.flush()

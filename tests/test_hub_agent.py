@@ -136,9 +136,18 @@ def test_rbac_roles(run):
     run(main())
 
 
+async def _ok_stage(hub, rid, pos):
+    """Complete rollout stage `pos` successfully (agent stand-in)."""
+    att = await hub.store.attempt_for_intent(f"rollout-{rid}-{pos}")
+    assert att is not None, f"stage {pos} not dispatched"
+    await hub.store.command_result(att["attempt_id"], True)
+
+
 def test_rollout_lifecycle(run):
     async def main():
         hub = Hub()
+        for n in ("n1", "n2"):
+            await hub.store.upsert_node(n, "tok", 5.0)
         rid = await hub.create_rollout({"streams": []}, ["n1", "n2"])
         r = await hub.step_rollout(rid)
         assert r["position"] == 1
@@ -147,8 +156,13 @@ def test_rollout_lifecycle(run):
         r2 = await hub.step_rollout(rid)  # paused → no advance
         assert r2["position"] == 1
         r = await hub.control_rollout(rid, "resume")
+        # health gate: stage 0 not finished yet → no advance
+        r = await hub.step_rollout(rid)
+        assert r["position"] == 1
+        await _ok_stage(hub, rid, 0)
         r = await hub.step_rollout(rid)
         assert r["position"] == 2
+        await _ok_stage(hub, rid, 1)
         r = await hub.step_rollout(rid)
         assert r["state"] == "succeeded"
 
@@ -207,7 +221,13 @@ def test_rollout_config_applied_via_agent(run):
             await asyncio.sleep(0.05)
         assert "rolled" in eng.runtime.entries
         assert eng.runtime.get("rolled").state.value == "running"
-        r = await hub.step_rollout(rid)
+        # the agent posts the attempt result asynchronously; the health
+        # gate only lets the rollout finish once it lands
+        for _ in range(100):
+            r = await hub.step_rollout(rid)
+            if r["state"] == "succeeded":
+                break
+            await asyncio.sleep(0.05)
         assert r["state"] == "succeeded"
         cancel.set()
         await asyncio.gather(agent_task, return_exceptions=True)
@@ -346,6 +366,8 @@ server:
 def test_rollout_cancel_stops_advance(run):
     async def main():
         hub = Hub()
+        for n in ("n1", "n2", "n3"):
+            await hub.store.upsert_node(n, "tok", 5.0)
         rid = await hub.create_rollout({"streams": []}, ["n1", "n2", "n3"])
         await hub.step_rollout(rid)
         r = await hub.control_rollout(rid, "cancel")
@@ -354,5 +376,163 @@ def test_rollout_cancel_stops_advance(run):
         assert r2["position"] == 1  # cancelled → frozen
         with pytest.raises(Exception):
             await hub.control_rollout(rid, "nonsense")
+        # regression: resume after cancel must be REJECTED (hub.rs rollout
+        # state machine; r1 accepted any transition)
+        with pytest.raises(Exception):
+            await hub.control_rollout(rid, "resume")
+
+    run(main())
+
+
+def test_rollout_auto_advances_with_health_gate(run):
+    """Stages advance automatically from the reconcile loop once the prior
+    stage's attempt succeeds — no operator /step needed (hub.rs:1472+)."""
+    async def main():
+        hub = Hub()
+        for n in ("n1", "n2"):
+            await hub.store.upsert_node(n, "tok", 5.0)
+        rid = await hub.create_rollout({"streams": []}, ["n1", "n2"])
+        await hub.reconcile_once()  # dispatches stage 0
+        r = await hub.store.get_rollout(rid)
+        assert r["position"] == 1
+        await hub.reconcile_once()  # stage 0 pending → gate holds
+        assert (await hub.store.get_rollout(rid))["position"] == 1
+        await _ok_stage(hub, rid, 0)
+        await hub.reconcile_once()  # dispatches stage 1
+        assert (await hub.store.get_rollout(rid))["position"] == 2
+        await _ok_stage(hub, rid, 1)
+        await hub.reconcile_once()
+        assert (await hub.store.get_rollout(rid))["state"] == "succeeded"
+
+    run(main())
+
+
+def test_rollout_failed_stage_fails_rollout(run):
+    async def main():
+        hub = Hub()
+        for n in ("n1", "n2"):
+            await hub.store.upsert_node(n, "tok", 5.0)
+        rid = await hub.create_rollout({"streams": []}, ["n1", "n2"])
+        await hub.reconcile_once()
+        att = await hub.store.attempt_for_intent(f"rollout-{rid}-0")
+        await hub.store.command_result(att["attempt_id"], False, "boom")
+        await hub.reconcile_once()
+        assert (await hub.store.get_rollout(rid))["state"] == "failed"
+        # stage 1 never dispatched
+        assert await hub.store.attempt_for_intent(f"rollout-{rid}-1") is None
+
+    run(main())
+
+
+def test_rollout_rollback_applies_prev_config(run):
+    """Rollback enqueues apply_config(prev_config) for every node already
+    touched (r1 only flipped a state string)."""
+    async def main():
+        hub = Hub()
+        for n in ("n1", "n2"):
+            await hub.store.upsert_node(n, "tok", 5.0)
+        prev = {"streams": [{"id": "old"}]}
+        rid = await hub.create_rollout({"streams": []}, ["n1", "n2"],
+                                       prev_config=prev)
+        await hub.reconcile_once()
+        await _ok_stage(hub, rid, 0)
+        await hub.reconcile_once()  # n2 dispatched; position == 2
+        r = await hub.control_rollout(rid, "rollback")
+        assert r["state"] == "rolled_back"
+        for pos, node in ((1, "n2"), (0, "n1")):
+            att = await hub.store.attempt_for_intent(f"rollback-{rid}-{pos}")
+            assert att is not None and att["node_id"] == node
+            import json as _json
+            assert _json.loads(att["command"])["config"] == prev
+
+    run(main())
+
+
+def test_offline_node_intent_executes_on_return(run):
+    """Regression (VERDICT weak #4): intents enqueued while the node is
+    offline must stay claimable and dispatch when the node comes back."""
+    async def main():
+        hub = Hub(lease_ttl=0.1)
+        await hub.store.upsert_node("n1", "tok", 0.1)
+        await asyncio.sleep(0.15)
+        await hub.sweep()  # lease expired → offline
+        nodes = await hub.store.nodes()
+        assert nodes[0]["online"] == 0
+        intent_id = await hub.enqueue_intent("n1", "s1", "start")
+        assert await hub.reconcile_once() == 0  # offline: nothing dispatched
+        assert await hub.reconcile_once() == 0  # still claimable, not lost
+        await hub.store.heartbeat("n1", 5.0)  # node returns
+        assert await hub.reconcile_once() == 1
+        intents = await hub.store.intents()
+        it = [i for i in intents if i["intent_id"] == intent_id][0]
+        assert it["state"] == "dispatched"
+
+    run(main())
+
+
+def test_expired_attempt_reenqueues_intent(run):
+    """Regression (VERDICT weak #4): an expired attempt re-enqueues the
+    intent for retry instead of leaving it stuck at 'dispatched'."""
+    async def main():
+        hub = Hub()
+        await hub.store.upsert_node("n1", "tok", 5.0)
+        intent_id = await hub.enqueue_intent("n1", "s1", "start")
+        assert await hub.reconcile_once() == 1
+        att = await hub.store.attempt_for_intent(intent_id)
+        # force-expire the attempt
+        def expire():
+            hub.store._db.execute(
+                "UPDATE attempts SET expires_at=0 WHERE attempt_id=?",
+                (att["attempt_id"],))
+            hub.store._db.commit()
+        expire()
+        assert await hub.reconcile_once() == 1  # re-dispatched
+        att2 = await hub.store.attempt_for_intent(intent_id)
+        assert att2["attempt_id"] != att["attempt_id"]
+        # retry budget: exhausting retries fails the intent
+        for _ in range(6):
+            a = await hub.store.attempt_for_intent(intent_id)
+            def ex():
+                hub.store._db.execute(
+                    "UPDATE attempts SET expires_at=0 WHERE attempt_id=?",
+                    (a["attempt_id"],))
+                hub.store._db.commit()
+            ex()
+            await hub.reconcile_once()
+        it = [i for i in await hub.store.intents()
+              if i["intent_id"] == intent_id][0]
+        assert it["state"] == "failed"
+        assert "exhausted" in (it["error"] or "")
+
+    run(main())
+
+
+def test_register_requires_token_and_protects_live_nodes(run):
+    """Advisor r1: /agent/register was unauthenticated and allowed identity
+    hijack of a live node."""
+    async def main():
+        hub = Hub(registration_token="sekrit", lease_ttl=5.0)
+        app = create_hub_app(hub)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://hub") as c:
+            r = await c.post("/agent/register", json={"node_id": "n1"})
+            assert r.status_code == 401
+            r = await c.post("/agent/register", json={"node_id": "n1"},
+                             headers={"x-registration-token": "wrong"})
+            assert r.status_code == 401
+            r = await c.post("/agent/register", json={"node_id": "n1"},
+                             headers={"x-registration-token": "sekrit"})
+            assert r.status_code == 200
+            tok = r.json()["node_token"]
+            # live node: re-register without the current token → rejected
+            r = await c.post("/agent/register", json={"node_id": "n1"},
+                             headers={"x-registration-token": "sekrit"})
+            assert r.status_code == 409
+            # with the current token → rotation allowed
+            r = await c.post("/agent/register", json={"node_id": "n1"},
+                             headers={"x-registration-token": "sekrit",
+                                      "x-node-token": tok})
+            assert r.status_code == 200
 
     run(main())

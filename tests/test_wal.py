@@ -452,3 +452,47 @@ def test_batch_payload_golden_bytes():
     assert out.input_name == "in"
     assert out.column("v").to_pylist() == [1.0]
     assert out.column("s").to_strlist() == ["x"]
+
+
+def test_validity_mask_survives_wal_roundtrip():
+    """Regression: deserialize_batch used to slice validity with [:-1],
+    losing the last element of every replayed NULL mask (advisor r1)."""
+    import torch
+    from arkflow_amd.batch import Column
+    b = MessageBatch.from_dict({"v": [1, 2, 3], "s": ["a", "b", "c"]})
+    b.columns["v"].validity = torch.tensor([True, False, True])
+    b.columns["s"].validity = torch.tensor([False, True, True])
+    r = deserialize_batch(serialize_batch(b))
+    assert r.column("v").validity.tolist() == [True, False, True]
+    assert r.column("s").validity.tolist() == [False, True, True]
+
+
+def test_mmap_store_append_after_reopen(tmp_path):
+    """Regression: reopening with recovered data left the mapping size
+    non-page-aligned, so the first append's rounded-up msync range fell
+    outside the map and raised ValueError (advisor r1)."""
+    from arkflow_amd.wal.store import MmapWalStore
+    st = MmapWalStore(str(tmp_path), stream_id="m1", chunk_bytes=65536)
+    st.append_batch([(1, b"x" * 100), (2, b"y" * 100)], sync=True)
+    st.close()
+    st2 = MmapWalStore(str(tmp_path), stream_id="m1", chunk_bytes=65536)
+    # the close() truncated to _pos (non-aligned); this append must not raise
+    st2.append_batch([(3, b"z" * 100)], sync=True)
+    assert [s for s, _ in st2.read_after(0)] == [1, 2, 3]
+    st2.close()
+
+
+def test_mmap_store_online_compaction(tmp_path):
+    """Regression: the compaction trigger gated on file.tell() which stays 0
+    for mmap appends, so the acked prefix grew unbounded (advisor r1)."""
+    from arkflow_amd.wal.store import MmapWalStore
+    st = MmapWalStore(str(tmp_path), stream_id="m2", chunk_bytes=65536)
+    st.compact_bytes = 4096
+    payload = b"p" * 1024
+    for seq in range(1, 9):
+        st.append_batch([(seq, payload)], sync=True)
+    assert st._log_size() > st.compact_bytes
+    st.write_cursor(6)  # should trigger online compaction
+    assert st._log_size() < 4096  # only seqs 7,8 remain
+    assert [s for s, _ in st.read_after(0)] == [7, 8]
+    st.close()
