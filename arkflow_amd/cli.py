@@ -113,7 +113,9 @@ def main(argv: Optional[List[str]] = None) -> int:
     async def run_all():
         cancel = asyncio.Event()
         tasks = []
-        if config.server.enabled:
+        # rank 0 owns HTTP/metrics in a torchrun launch; peers would
+        # collide on the bind address
+        if config.server.enabled and engine.rank == 0:
             from .server.api import serve
             tasks.append(asyncio.ensure_future(serve(engine, cancel)))
             if config.server.hub_url:

@@ -76,6 +76,23 @@ class ServerConfig:
     config_store: Optional[str] = None  # persist config versions for rollback
 
 
+def _interpolate_env(text: str) -> str:
+    """``${VAR}`` / ``${VAR:-default}`` substitution in config files — lets
+    one YAML serve every rank of a torchrun launch (per-rank output paths,
+    ports). Unset vars without a default are left untouched."""
+    import os
+    import re
+
+    def sub(m):
+        name, sep, default = m.group(1).partition(":-")
+        val = os.environ.get(name)
+        if val is not None:
+            return val
+        return default if sep else m.group(0)
+
+    return re.sub(r"\$\{([A-Za-z_][A-Za-z0-9_]*(?::-[^}]*)?)\}", sub, text)
+
+
 @dataclass
 class EngineConfig:
     streams: List[StreamConfig] = field(default_factory=list)
@@ -87,6 +104,7 @@ class EngineConfig:
     def from_file(path: str) -> "EngineConfig":
         with open(path, "r") as f:
             text = f.read()
+        text = _interpolate_env(text)
         if path.endswith((".yaml", ".yml")):
             raw = yaml.safe_load(text)
         elif path.endswith(".json"):

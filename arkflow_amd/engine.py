@@ -15,6 +15,18 @@ log = logging.getLogger("arkflow_amd.engine")
 
 class Engine:
     def __init__(self, config: EngineConfig):
+        # Join the torchrun process group FIRST (idempotent, env-gated):
+        # maps LOCAL_RANK→cuda:N as the default device before any stream is
+        # built, so `torchrun --nproc-per-node N python -m arkflow_amd
+        # --config …` runs one engine shard per GPU with RCCL repartition
+        # live (ranks used to start independent and non-sharding)
+        from .parallel import dist as afdist
+        afdist.init_from_env()
+        self.rank = afdist.rank()
+        self.world_size = afdist.world_size()
+        if self.rank != 0:
+            log.info("rank %d/%d: HTTP/metrics served by rank 0 only",
+                     self.rank, self.world_size)
         self.config = config
         self.runtime = RuntimeManager()
         self.ready = False

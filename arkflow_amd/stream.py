@@ -345,8 +345,13 @@ def build_stream(config: StreamConfig) -> Stream:
     )
     wal = None
     if config.durability and config.durability.enabled:
+        from .parallel import dist as afdist
         from .wal.wal import Wal
-        wal = Wal.open(config.durability, stream_id=config.id)
+        # per-rank WAL identity: N engine shards sharing one config must not
+        # interleave frames in one log file
+        sid = config.id if afdist.world_size() <= 1 \
+            else f"{config.id}.rank{afdist.rank()}"
+        wal = Wal.open(config.durability, stream_id=sid)
     return Stream(
         config, input_, Pipeline(processors), output, error_output,
         buffer, wal, temporaries,

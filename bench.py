@@ -152,8 +152,8 @@ async def run_steps(gen, pipeline, n_steps, workers=1):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--batch-size", type=int, default=8192)
     p.add_argument("--model", choices=["mlp", "bert", "sqlagg", "proto_mlp"],
                    default="mlp")
