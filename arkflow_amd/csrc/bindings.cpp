@@ -64,6 +64,9 @@ void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
                          hipStream_t);
 void launch_gather_multi(int, const void**, void**, const int*,
                          const int32_t*, int64_t, hipStream_t);
+void launch_gather_multi_dyn(int, const void**, void**, const int*,
+                             const int32_t*, const int32_t*, int64_t,
+                             hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
                        hipStream_t);
 void launch_bytes_match(const uint8_t*, const int64_t*, int64_t,
@@ -618,6 +621,67 @@ std::tuple<std::vector<torch::Tensor>, int64_t> fused_filter_gather(
   return {outs, total};
 }
 
+// hipGraph-capturable variant of fused_filter_gather: writes compacted rows
+// into PREALLOCATED padded outputs and the row count into a device int32 —
+// zero host syncs, so generate→filter→infer captures as ONE graph
+// (the flagship step was host-dispatch-bound at batch 8192).
+void filter_gather_capture(std::vector<torch::Tensor> cols,
+                           int64_t filter_idx, int64_t op, double scalar,
+                           std::vector<torch::Tensor> outs,
+                           torch::Tensor count_out) {
+  TORCH_CHECK(filter_idx >= 0 && filter_idx < (int64_t)cols.size());
+  TORCH_CHECK(outs.size() == cols.size(), "outs must match cols");
+  TORCH_CHECK(count_out.scalar_type() == torch::kInt32 &&
+              count_out.numel() == 1, "count_out must be int32[1] on device");
+  auto col = cols[filter_idx];
+  check_cuda(col, "filter col");
+  int64_t n = col.numel();
+  auto st = cur_stream();
+  TORCH_CHECK(n > 0, "capture path requires a non-empty static batch");
+  int nblocks = filter_grid(n);
+  auto counts = torch::empty({nblocks}, col.options().dtype(torch::kInt32));
+  bool is_f32 = col.scalar_type() == torch::kFloat32;
+  TORCH_CHECK(is_f32 || col.scalar_type() == torch::kInt64,
+              "capture filter col must be f32 or i64");
+  if (is_f32)
+    launch_filter_count_f32(col.data_ptr<float>(), n, (int)op, (float)scalar,
+                            counts.data_ptr<int32_t>(), st);
+  else
+    launch_filter_count_i64(col.data_ptr<int64_t>(), n, (int)op,
+                            (int64_t)scalar, counts.data_ptr<int32_t>(), st);
+  auto [offs, total_t] = exscan(counts);
+  count_out.copy_(total_t, /*non_blocking=*/true);
+  auto idx = torch::empty({n}, col.options().dtype(torch::kInt32));
+  if (is_f32)
+    launch_filter_scatter_f32(col.data_ptr<float>(), n, (int)op,
+                              (float)scalar, offs.data_ptr<int32_t>(),
+                              idx.data_ptr<int32_t>(), st);
+  else
+    launch_filter_scatter_i64(col.data_ptr<int64_t>(), n, (int)op,
+                              (int64_t)scalar, offs.data_ptr<int32_t>(),
+                              idx.data_ptr<int32_t>(), st);
+  std::vector<const void*> src;
+  std::vector<void*> dst;
+  std::vector<int> esz;
+  auto flush = [&]() {
+    if (!src.empty())
+      launch_gather_multi_dyn((int)src.size(), src.data(), dst.data(),
+                              esz.data(), idx.data_ptr<int32_t>(),
+                              count_out.data_ptr<int32_t>(), n, st);
+    src.clear(); dst.clear(); esz.clear();
+  };
+  for (size_t c = 0; c < cols.size(); ++c) {
+    TORCH_CHECK(outs[c].numel() >= n &&
+                outs[c].element_size() == cols[c].element_size(),
+                "out buffer too small or wrong dtype");
+    src.push_back(cols[c].data_ptr());
+    dst.push_back(outs[c].data_ptr());
+    esz.push_back((int)cols[c].element_size());
+    if ((int)src.size() == 32) flush();
+  }
+  flush();
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
            std::vector<std::tuple<torch::Tensor, torch::Tensor>>>
 json_decode(torch::Tensor data, torch::Tensor offsets,
@@ -862,5 +926,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bytes_hash", &bytes_hash);
   m.def("json_decode", &json_decode);
   m.def("fused_filter_gather", &fused_filter_gather);
+  m.def("filter_gather_capture", &filter_gather_capture);
   m.def("radix_argsort", &radix_argsort, py::arg("keys"), py::arg("descending") = false);
 }

@@ -253,6 +253,55 @@ extern "C" void launch_gather_multi(int ncols, const void** src, void** dst,
   gather_multi_kernel<<<grid, 256, 0, st>>>(spec, idx, m);
 }
 
+// ---- dyn-count multi-gather: hipGraph-capturable compaction tail ------------
+// Same as gather_multi_kernel but the row count lives in DEVICE memory
+// (written by the scan), so the whole filter→gather chain captures into a
+// hipGraph with no host sync. Launched with a grid sized for the batch cap;
+// threads past *count_dev exit.
+__global__ void gather_multi_dyn_kernel(GatherSpec spec,
+                                        const int32_t* __restrict__ idx,
+                                        const int32_t* __restrict__ count_dev,
+                                        int64_t cap) {
+  int64_t m = (int64_t)*count_dev;
+  int64_t total = (int64_t)spec.ncols * m;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int c = (int)(i / m);
+    int64_t j = i - (int64_t)c * m;
+    int64_t s = idx[j];
+    switch (spec.esz[c]) {
+      case 1: ((uint8_t*)spec.dst[c])[j] = ((const uint8_t*)spec.src[c])[s];
+              break;
+      case 2: ((uint16_t*)spec.dst[c])[j] = ((const uint16_t*)spec.src[c])[s];
+              break;
+      case 4: ((uint32_t*)spec.dst[c])[j] = ((const uint32_t*)spec.src[c])[s];
+              break;
+      default: ((uint64_t*)spec.dst[c])[j] = ((const uint64_t*)spec.src[c])[s];
+               break;
+    }
+  }
+}
+
+extern "C" void launch_gather_multi_dyn(int ncols, const void** src,
+                                        void** dst, const int* esz,
+                                        const int32_t* idx,
+                                        const int32_t* count_dev, int64_t cap,
+                                        hipStream_t st) {
+  GatherSpec spec{};
+  spec.ncols = ncols;
+  for (int c = 0; c < ncols; ++c) {
+    spec.src[c] = src[c];
+    spec.dst[c] = dst[c];
+    spec.esz[c] = esz[c];
+  }
+  int64_t total = (int64_t)ncols * cap;
+  int grid = (int)((total + 255) / 256);
+  if (grid > 4096) grid = 4096;
+  if (grid < 1) return;
+  gather_multi_dyn_kernel<<<grid, 256, 0, st>>>(spec, idx, count_dev, cap);
+}
+
 // ---- per-row bytes hash (binary columns: group-by / repartition keys) -------
 // 64-bit FNV-1a over each row's byte slice; one thread per row. Used to
 // dictionary-encode string keys on-device (collision probability ~n²/2^65 —

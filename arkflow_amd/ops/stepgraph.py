@@ -1,0 +1,151 @@
+"""Whole-step hipGraph capture for the flagship generate→filter→infer path.
+
+The r1 flagship bench was host-dispatch-bound at the default batch (8192
+rows: ~0.15 ms GPU inside a ~0.22 ms step — VERDICT weak #3/#5). The fix is
+to capture the ENTIRE step — synthetic generation, fused filter+compact,
+feature stack, MLP scoring — as one hipGraph and replay it with a single
+launch per step. Everything inside is shape-static:
+
+  rand/randint (graph-safe RNG)  →  filter_gather_capture (device count,
+  padded outputs, no host sync)  →  stack/pad → MLP GEMM chain
+
+The only host interaction per step is ONE int32 readback (the surviving row
+count) used to slice zero-copy views out of the static output buffers.
+
+The returned batch aliases the graph's static buffers and is valid until the
+next ``step()`` — the same contract as the window ring's zero-copy slices
+(buffers/ring.py). Downstream stages that retain batches must copy.
+
+Reference analog: the hot loop the reference runs as compiled Rust end to
+end (stream/mod.rs:370-444 + DataFusion physical operators); here the whole
+step becomes one device-side graph so Python dispatch cost is O(1) per step.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Tuple
+
+import torch
+
+from ..batch import Column, MessageBatch
+from . import require_native
+
+_OPS = {"<": 0, "<=": 1, ">": 2, ">=": 3, "==": 4, "!=": 5}
+
+
+class FusedGenerateFilterInfer:
+    """generate(fields) → WHERE col OP scalar → mlp(score) as one hipGraph.
+
+    fields: ordered {name: {dtype, low, high}} — float32 fields are
+    generated with one fused rand+scale launch; int64 fields via randint.
+    """
+
+    def __init__(self, fields: Dict[str, dict], batch_size: int,
+                 filter_col: str, op: str, scalar: float, mlp,
+                 device: torch.device, seed: int = 0x5EED):
+        self.device = torch.device(device)
+        if self.device.type != "cuda":
+            raise RuntimeError("FusedGenerateFilterInfer requires a GPU")
+        self.n = int(batch_size)
+        self.fields = dict(fields)
+        self.filter_col = filter_col
+        self.op = _OPS[op]
+        self.scalar = float(scalar)
+        self.mlp = mlp
+        self.nat = require_native()
+        torch.cuda.manual_seed(seed)
+
+        names = list(fields)
+        self.float_names = [f for f in names
+                            if str(fields[f].get("dtype", "float32"))
+                            not in ("int32", "int64")]
+        self.int_names = [f for f in names if f not in self.float_names]
+        self._scale = torch.tensor(
+            [[float(fields[f].get("high", 100.0)) -
+              float(fields[f].get("low", 0.0))] for f in self.float_names],
+            device=self.device, dtype=torch.float32)
+        self._offset = torch.tensor(
+            [[float(fields[f].get("low", 0.0))] for f in self.float_names],
+            device=self.device, dtype=torch.float32)
+
+        # static output buffers (padded to batch cap) + device count
+        self.outs: Dict[str, torch.Tensor] = {}
+        for f in self.float_names:
+            self.outs[f] = torch.zeros(self.n, device=self.device,
+                                       dtype=torch.float32)
+        for f in self.int_names:
+            self.outs[f] = torch.zeros(self.n, device=self.device,
+                                       dtype=torch.int64)
+        self.count = torch.zeros(1, device=self.device, dtype=torch.int32)
+        self._graph = None
+        self._scores = None
+
+    # ------------------------------------------------------------------ body
+    def _body(self) -> torch.Tensor:
+        n = self.n
+        nf = len(self.float_names)
+        block = torch.rand((nf, n), device=self.device, dtype=torch.float32)
+        block = torch.addcmul(self._offset, block, self._scale)
+        cols: Dict[str, torch.Tensor] = {
+            f: block[i] for i, f in enumerate(self.float_names)}
+        for f in self.int_names:
+            spec = self.fields[f]
+            lo = int(float(spec.get("low", 0.0)))
+            hi = max(int(float(spec.get("high", 100.0))), lo + 1)
+            cols[f] = torch.randint(lo, hi, (n,), device=self.device,
+                                    dtype=torch.int64)
+        ordered = [cols[f].contiguous() for f in self.fields]
+        fidx = list(self.fields).index(self.filter_col)
+        self.nat.filter_gather_capture(
+            ordered, fidx, self.op, self.scalar,
+            [self.outs[f] for f in self.fields], self.count)
+        # score the padded buffer: garbage rows past `count` are sliced off
+        feats = torch.stack([self.outs[f] for f in self.float_names], dim=1)
+        pad = self.mlp.dims[0] - nf
+        h = feats.to(torch.bfloat16)
+        if pad:
+            h = torch.nn.functional.pad(h, (0, pad))
+        return self.mlp._net(h)
+
+    def capture(self) -> None:
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):  # warm up allocator + kernels
+                self._body()
+        torch.cuda.current_stream().wait_stream(s)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._scores = self._body()
+
+    # ------------------------------------------------------------------ step
+    def step(self) -> Tuple[MessageBatch, int]:
+        """One replay. Returns (batch_of_views, kept_rows); the batch is
+        valid until the next step()."""
+        if self._graph is None:
+            self.capture()
+        self._graph.replay()
+        kept = int(self.count.item())
+        cols = {f: Column("numeric", self.outs[f][:kept])
+                for f in self.fields}
+        cols["score"] = Column("numeric", self._scores[:kept])
+        return MessageBatch(cols, input_name="generate"), kept
+
+
+class FusedStepSource:
+    """Input-SPI facade over FusedGenerateFilterInfer: `read()` yields the
+    fully processed batch, so a Stream/bench drives the fused step through
+    the normal (input, pipeline) shape with an empty pipeline."""
+
+    def __init__(self, fused: FusedGenerateFilterInfer):
+        self.fused = fused
+
+    async def read(self):
+        from ..spi import NoopAck
+        batch, _ = self.fused.step()
+        return batch, NoopAck()
+
+    async def connect(self):  # pragma: no cover - trivial
+        pass
+
+    async def close(self):  # pragma: no cover - trivial
+        pass

@@ -98,6 +98,20 @@ def build_pipeline(args, device):
             "hidden": [args.hidden, args.hidden], "device": str(device),
         })
         return _ProtoGen(), Pipeline([decode, infer])
+    if device.type == "cuda" and not args.no_stepgraph:
+        # whole-step hipGraph: generate+filter+MLP replay as ONE graph —
+        # kills the host-dispatch ceiling at small batches.
+        # Same work, same outputs as the eager path (numerics-tested in
+        # tests/test_gpu_kernels.py).
+        from arkflow_amd.models.mlp import MlpAnomalyDetector
+        from arkflow_amd.ops.stepgraph import (
+            FusedGenerateFilterInfer, FusedStepSource)
+        mlp = MlpAnomalyDetector(n_features, [args.hidden, args.hidden],
+                                 device, 1234)
+        fused = FusedGenerateFilterInfer(
+            fields, args.batch_size, "f0", ">=", 0.2, mlp, device,
+            seed=7 + args.rank)
+        return FusedStepSource(fused), Pipeline([])
     infer = InferenceProcessor({
         "model": "mlp_anomaly",
         "columns": [f"f{i}" for i in range(n_features)],
@@ -161,6 +175,8 @@ def main():
     p.add_argument("--hidden", type=int, default=256)
     p.add_argument("--workers", type=int, default=1,
                    help="concurrent in-flight steps (engine thread_num analog)")
+    p.add_argument("--no-stepgraph", action="store_true",
+                   help="disable the whole-step hipGraph for --model mlp")
     args = p.parse_args()
 
     args.rank = int(os.environ.get("RANK", 0))

@@ -837,3 +837,63 @@ def test_wal_replay_restores_device(dev, tmp_path):
         assert seen_devices == ["cuda"]
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_filter_gather_capture_matches_sync_path(nat, dev):
+    """Capture-friendly filter (device count, padded outs) vs the syncing
+    fused_filter_gather on the same inputs."""
+    torch.manual_seed(11)
+    n = 5000
+    f0 = torch.rand(n, device=dev)
+    f1 = torch.randn(n, device=dev)
+    key = torch.randint(0, 100, (n,), device=dev)
+    cols = [f0, f1.contiguous(), key]
+    outs = [torch.zeros(n, device=dev, dtype=c.dtype) for c in cols]
+    count = torch.zeros(1, device=dev, dtype=torch.int32)
+    nat.filter_gather_capture(cols, 0, 3, 0.2, outs, count)  # f0 >= 0.2
+    ref_outs, ref_total = nat.fused_filter_gather(cols, 0, 3, 0.2)
+    kept = int(count.item())
+    assert kept == ref_total
+    for o, r in zip(outs, ref_outs):
+        assert torch.equal(o[:kept], r)
+
+
+def test_fused_stepgraph_whole_pipeline(dev):
+    """The whole-step hipGraph (generate→filter→MLP) produces rows that
+    (a) all satisfy the predicate, (b) carry scores matching an fp32
+    reference MLP on the emitted features, and (c) re-randomize each
+    replay (graph-safe RNG advances)."""
+    from arkflow_amd.models.mlp import MlpAnomalyDetector
+    from arkflow_amd.ops.stepgraph import FusedGenerateFilterInfer
+
+    nfeat = 16
+    fields = {f"f{i}": {"dtype": "float32", "low": 0.0, "high": 1.0}
+              for i in range(nfeat)}
+    fields["key"] = {"dtype": "int64", "low": 0, "high": 1024}
+    mlp = MlpAnomalyDetector(nfeat, [64, 64], dev, 99)
+    fused = FusedGenerateFilterInfer(fields, 8192, "f0", ">=", 0.2,
+                                     mlp, dev, seed=5)
+    batch1, kept1 = fused.step()
+    assert 0 < kept1 < 8192
+    assert bool((batch1.column("f0").data >= 0.2).all())
+    # selectivity ≈ 0.8 for uniform [0,1)
+    assert abs(kept1 / 8192 - 0.8) < 0.05
+    feats = torch.stack([batch1.column(f"f{i}").data.float()
+                         for i in range(nfeat)], dim=1)
+    # fp32 reference of the same MLP on the emitted rows
+    h = feats
+    import torch.nn.functional as F
+    for i, (w, b) in enumerate(zip(mlp.weights, mlp.biases)):
+        if h.shape[1] < w.shape[1]:
+            h = F.pad(h, (0, w.shape[1] - h.shape[1]))
+        h = F.linear(h, w.float(), b.float())
+        if i < len(mlp.weights) - 1:
+            h = torch.relu(h)
+    ref = h.reshape(-1)
+    got = batch1.column("score").data.float()
+    assert torch.allclose(got, ref, atol=0.05, rtol=0.05)
+    f0_a = batch1.column("f0").data.clone()
+    batch2, kept2 = fused.step()
+    f0_b = batch2.column("f0").data
+    m = min(len(f0_a), len(f0_b))
+    assert not torch.equal(f0_a[:m], f0_b[:m]), "RNG did not advance"
