@@ -67,6 +67,16 @@ void launch_gather_multi(int, const void**, void**, const int*,
 void launch_gather_multi_dyn(int, const void**, void**, const int*,
                              const int32_t*, const int32_t*, int64_t,
                              hipStream_t);
+void launch_gemm_bf16_skinny(const void*, const void*, const float*, void*,
+                             int, int, int, int, hipStream_t);
+void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
+                       int, int64_t, int64_t, unsigned long long*,
+                       hipStream_t);
+void launch_scan_counts(const int32_t*, int, int32_t*, int32_t*,
+                        hipStream_t);
+void launch_featpack(const float**, int, int, int64_t, void*, hipStream_t);
+void launch_gemv_bf16_f32(const void*, const void*, float, int64_t, int,
+                          float*, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
                        hipStream_t);
 void launch_bytes_match(const uint8_t*, const int64_t*, int64_t,
@@ -342,10 +352,81 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
                                      C.data_ptr(), (int)M, (int)N, (int)K,
                                      (int)act, swz_code, cur_stream())
                : -1;
-  if (rc != 0)
-    launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
-                     (int)M, (int)N, (int)K, (int)act, cur_stream());
+  if (rc != 0) {
+    // narrow outputs: the 128² grid can't fill 256 CUs (e.g. [8192,256] →
+    // 128 blocks); the 64² tile quadruples the grid (profiles r2)
+    int64_t t128grid = ((M + 127) / 128) * ((N + 127) / 128);
+    if (t128grid < 208)
+      launch_gemm_bf16_skinny(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                              C.data_ptr(), (int)M, (int)N, (int)K,
+                              (int)act, cur_stream());
+    else
+      launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
+                       (int)M, (int)N, (int)K, (int)act, cur_stream());
+  }
   return C;
+}
+
+// scores = x[M,K]·w[K] + bias → f32 (the MLP head: N=1 GEMM + downcast in
+// one launch instead of a hipBLASLt GEMV + copy kernel)
+torch::Tensor gemv_bf16_f32(torch::Tensor x, torch::Tensor w, double bias) {
+  check_cuda(x, "x");
+  check_cuda(w, "w");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16, "bf16 required");
+  TORCH_CHECK(x.dim() == 2 && w.numel() == x.size(1), "x [M,K], w [K]");
+  auto out = torch::empty({x.size(0)}, x.options().dtype(torch::kFloat32));
+  launch_gemv_bf16_f32(x.data_ptr(), w.data_ptr(), (float)bias, x.size(0),
+                       (int)x.size(1), out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+// one fused launch generating every float column + the int64 key column of
+// a synthetic batch; ctr[0] advances on-device so graph replays
+// re-randomize (ops/stepgraph.py)
+void gen_fields(torch::Tensor block, c10::optional<torch::Tensor> key,
+                std::vector<double> lo, std::vector<double> width,
+                int64_t key_lo, int64_t key_range, torch::Tensor ctr) {
+  check_cuda(block, "block");
+  TORCH_CHECK(block.dim() == 2 && block.scalar_type() == torch::kFloat32,
+              "block must be [nf, n] f32");
+  TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.numel() == 2,
+              "ctr must be int64[2] on device");
+  int nf = (int)block.size(0);
+  int64_t n = block.size(1);
+  TORCH_CHECK((int)lo.size() == nf && (int)width.size() == nf);
+  std::vector<float> lof(nf), wf(nf);
+  for (int f = 0; f < nf; ++f) { lof[f] = (float)lo[f]; wf[f] = (float)width[f]; }
+  int64_t* key_ptr = nullptr;
+  if (key.has_value() && key->defined()) {
+    check_cuda(*key, "key");
+    TORCH_CHECK(key->numel() == n && key->scalar_type() == torch::kInt64);
+    key_ptr = key->data_ptr<int64_t>();
+  } else {
+    key_range = 0;
+  }
+  launch_gen_fields(block.data_ptr<float>(), key_ptr, n, lof.data(),
+                    wf.data(), nf, key_lo, key_range,
+                    (unsigned long long*)ctr.data_ptr<int64_t>(),
+                    cur_stream());
+}
+
+// gathered f32 feature columns → [n, kpad] bf16 MFMA operand in one launch
+void featpack(std::vector<torch::Tensor> srcs, torch::Tensor out) {
+  TORCH_CHECK(!srcs.empty() && srcs.size() <= 32, "1..32 feature columns");
+  check_cuda(out, "out");
+  TORCH_CHECK(out.dim() == 2 && out.scalar_type() == torch::kBFloat16,
+              "out must be [n, kpad] bf16");
+  int64_t n = out.size(0);
+  int kpad = (int)out.size(1);
+  std::vector<const float*> ptrs;
+  for (auto& s : srcs) {
+    check_cuda(s, "src");
+    TORCH_CHECK(s.numel() >= n && s.scalar_type() == torch::kFloat32);
+    ptrs.push_back(s.data_ptr<float>());
+  }
+  launch_featpack(ptrs.data(), (int)ptrs.size(), kpad, n, out.data_ptr(),
+                  cur_stream());
 }
 
 torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
@@ -370,6 +451,10 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
                                  C.data_ptr(), (int)M, (int)N, (int)K,
                                  (int)act, cur_stream());
     TORCH_CHECK(rc == 0, "shape not supported by 2-phase kernel");
+  } else if (variant == 6) {
+    launch_gemm_bf16_skinny(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                            C.data_ptr(), (int)M, (int)N, (int)K, (int)act,
+                            cur_stream());
   } else {
     // 8-phase variants: 1=BM256, 2=BM256+swz, 4=BM128, 5=BM128+swz
     int swz_code = variant == 2 ? 1 : variant == 4 ? 2 : variant == 5 ? 3 : 0;
@@ -649,8 +734,19 @@ void filter_gather_capture(std::vector<torch::Tensor> cols,
   else
     launch_filter_count_i64(col.data_ptr<int64_t>(), n, (int)op,
                             (int64_t)scalar, counts.data_ptr<int32_t>(), st);
-  auto [offs, total_t] = exscan(counts);
-  count_out.copy_(total_t, /*non_blocking=*/true);
+  torch::Tensor offs;
+  if (nblocks <= 1024) {
+    // single-block scan writes offsets AND the device total in one launch
+    // (replaces torch cumsum + sub + copy_ — 3 kernels → 1 in the graph)
+    offs = torch::empty({nblocks}, col.options().dtype(torch::kInt32));
+    launch_scan_counts(counts.data_ptr<int32_t>(), nblocks,
+                       offs.data_ptr<int32_t>(),
+                       count_out.data_ptr<int32_t>(), st);
+  } else {
+    auto [offs2, total_t] = exscan(counts);
+    offs = offs2;
+    count_out.copy_(total_t, /*non_blocking=*/true);
+  }
   auto idx = torch::empty({n}, col.options().dtype(torch::kInt32));
   if (is_f32)
     launch_filter_scatter_f32(col.data_ptr<float>(), n, (int)op,
@@ -927,5 +1023,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("json_decode", &json_decode);
   m.def("fused_filter_gather", &fused_filter_gather);
   m.def("filter_gather_capture", &filter_gather_capture);
+  m.def("gemv_bf16_f32", &gemv_bf16_f32);
+  m.def("gen_fields", &gen_fields);
+  m.def("featpack", &featpack);
   m.def("radix_argsort", &radix_argsort, py::arg("keys"), py::arg("descending") = false);
 }

@@ -142,6 +142,133 @@ void gemm_bf16_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- skinny 64×64 tile -------------------------------------------------------
+// For narrow outputs (N ≤ a few hundred: the MLP anomaly scorer's 32→256→256
+// chain) the 128² tile yields only tiles_m×⌈N/128⌉ workgroups — 128 blocks at
+// [8192,256], half the 256 CUs idle. The 64² tile quadruples the grid so the
+// chip fills; LDS is 8 KiB → high occupancy hides the short K loop.
+// Same wave structure as the 128² kernel scaled down: 4 waves in 2×2, each
+// owning a 32×32 quadrant = 2×2 mfma_f32_16x16x32_bf16 fragments.
+#define SBM 64
+#define SBN 64
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 4)
+void gemm_bf16_skinny_kernel(const __bf16* __restrict__ A,   // [M,K]
+                             const __bf16* __restrict__ Bt,  // [N,K]
+                             const float* __restrict__ bias, // [N] or null
+                             __bf16* __restrict__ C,         // [M,N]
+                             int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {  // XCD-aware bijective swizzle (guide §5)
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * SBM, col0 = bn * SBN;
+
+  __shared__ __bf16 Asm[SBM * BK];  // [64][32] linear, 4 KiB
+  __shared__ __bf16 Bsm[SBN * BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;        // 4 waves: 2×2 over the 64×64 tile
+  const int wm = (wid >> 1) * 32;
+  const int wn = (wid & 1) * 32;
+
+  f32x4 acc[2][2] = {};
+
+  // staging: the 4 KiB operand tile = 4 waves × 1 KiB global_load_lds
+  const int lin0 = wid * 1024 + lane * 16;
+  const int trow = lin0 >> 6;   // 64 B per row (BK=32 bf16)
+  const int tcol = lin0 & 63;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    int ga_row = row0 + trow;
+    ga_row = ga_row < M ? ga_row : M - 1;
+    const char* a_src = (const char*)(A + (int64_t)ga_row * K + k0) + tcol;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)a_src,
+        (__attribute__((address_space(3))) uint32_t*)((char*)Asm + lin0),
+        16, 0, 0);
+    int gb_row = col0 + trow;
+    gb_row = gb_row < N ? gb_row : N - 1;
+    const char* b_src = (const char*)(Bt + (int64_t)gb_row * K + k0) + tcol;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)b_src,
+        (__attribute__((address_space(3))) uint32_t*)((char*)Bsm + lin0),
+        16, 0, 0);
+    __syncthreads();
+
+    bf16x8 a_frag[2], b_frag[2];
+    const int fr = lane & 15;
+    const int fk = (lane >> 4) * 8;
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+      a_frag[m] = *(const bf16x8*)&Asm[(wm + m * 16 + fr) * BK + fk];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+      b_frag[n] = *(const bf16x8*)&Bsm[(wn + n * 16 + fr) * BK + fk];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    __syncthreads();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" void launch_gemm_bf16_skinny(const void* A, const void* Bt,
+                                        const float* bias, void* C, int M,
+                                        int N, int K, int act,
+                                        hipStream_t st) {
+  int tiles_m = (M + SBM - 1) / SBM;
+  int tiles_n = (N + SBN - 1) / SBN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define SDISPATCH(ACT)                                                       \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_skinny_kernel<ACT, true><<<grid, block, 0, st>>>(            \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_skinny_kernel<ACT, false><<<grid, block, 0, st>>>(           \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: SDISPATCH(ACT_RELU); break;
+    case ACT_GELU: SDISPATCH(ACT_GELU); break;
+    case ACT_SILU: SDISPATCH(ACT_SILU); break;
+    default: SDISPATCH(ACT_NONE); break;
+  }
+#undef SDISPATCH
+}
+
 // ---- 2-phase double-buffered variant -----------------------------------------
 // T3-minimal prefetch (guide §5.5): stage K-tile t+1 into the other LDS
 // buffer while computing tile t; ONE vmcnt(0)+barrier per K-tile AFTER the

@@ -32,6 +32,7 @@ class MlpAnomalyDetector:
         self.dims = dims
         self.use_graph = True
         self._graphs = {}  # padded-cap → (graph, static_in, static_out)
+        self._head_bias = None  # scalar bias of the N=1 scoring head
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """x: [n, in_features] float → anomaly score [n] float32.
@@ -52,7 +53,17 @@ class MlpAnomalyDetector:
 
     def _net(self, h: torch.Tensor) -> torch.Tensor:
         for i, (w, b) in enumerate(zip(self.weights, self.biases)):
-            act = "relu" if i < len(self.weights) - 1 else "none"
+            last = i == len(self.weights) - 1
+            if last and h.is_cuda and w.shape[0] == 1:
+                # scoring head: one fused GEMV+downcast launch instead of a
+                # hipBLASLt N=1 GEMM + bf16→f32 copy
+                from ..ops import require_native
+                if self._head_bias is None:
+                    self._head_bias = float(b[0].item())
+                return require_native().gemv_bf16_f32(
+                    h.contiguous(), w.reshape(-1).contiguous(),
+                    self._head_bias)
+            act = "relu" if not last else "none"
             h = opsnn.linear_bf16(h, w, b, act=act)
         return h.reshape(-1).to(torch.float32)
 

@@ -67,7 +67,19 @@ class FusedGenerateFilterInfer:
             [[float(fields[f].get("low", 0.0))] for f in self.float_names],
             device=self.device, dtype=torch.float32)
 
-        # static output buffers (padded to batch cap) + device count
+        if len(self.int_names) > 1:
+            raise ValueError("fused generate supports at most one int64 "
+                             "key column")
+
+        # static buffers: generated block, key, compacted outputs, packed
+        # MFMA operand, device row count, RNG replay counter
+        nf = len(self.float_names)
+        self.block = torch.zeros((nf, self.n), device=self.device,
+                                 dtype=torch.float32)
+        self.key = torch.zeros(self.n, device=self.device,
+                               dtype=torch.int64) if self.int_names else None
+        self.ctr = torch.zeros(2, device=self.device, dtype=torch.int64)
+        self.ctr[0] = int(seed) & 0x7FFFFFFF
         self.outs: Dict[str, torch.Tensor] = {}
         for f in self.float_names:
             self.outs[f] = torch.zeros(self.n, device=self.device,
@@ -75,36 +87,45 @@ class FusedGenerateFilterInfer:
         for f in self.int_names:
             self.outs[f] = torch.zeros(self.n, device=self.device,
                                        dtype=torch.int64)
+        self.feats = torch.zeros((self.n, mlp.dims[0]), device=self.device,
+                                 dtype=torch.bfloat16)
         self.count = torch.zeros(1, device=self.device, dtype=torch.int32)
+        self._lo = [float(fields[f].get("low", 0.0))
+                    for f in self.float_names]
+        self._width = [float(fields[f].get("high", 100.0)) -
+                       float(fields[f].get("low", 0.0))
+                       for f in self.float_names]
+        if self.int_names:
+            spec = fields[self.int_names[0]]
+            self._key_lo = int(float(spec.get("low", 0.0)))
+            hi = int(float(spec.get("high", 100.0)))
+            self._key_range = max(hi - self._key_lo, 1)
+        else:
+            self._key_lo = self._key_range = 0
         self._graph = None
         self._scores = None
 
     # ------------------------------------------------------------------ body
     def _body(self) -> torch.Tensor:
-        n = self.n
-        nf = len(self.float_names)
-        block = torch.rand((nf, n), device=self.device, dtype=torch.float32)
-        block = torch.addcmul(self._offset, block, self._scale)
+        """One step as a pure kernel chain (9 launches):
+        gen_fields → filter count → scan → scatter → gather → featpack →
+        GEMM(relu) → GEMM(relu) → GEMV+f32. No host syncs, no torch
+        elementwise tail — every stage is a stepfused.hip/gemm kernel."""
         cols: Dict[str, torch.Tensor] = {
-            f: block[i] for i, f in enumerate(self.float_names)}
-        for f in self.int_names:
-            spec = self.fields[f]
-            lo = int(float(spec.get("low", 0.0)))
-            hi = max(int(float(spec.get("high", 100.0))), lo + 1)
-            cols[f] = torch.randint(lo, hi, (n,), device=self.device,
-                                    dtype=torch.int64)
-        ordered = [cols[f].contiguous() for f in self.fields]
+            f: self.block[i] for i, f in enumerate(self.float_names)}
+        if self.key is not None:
+            cols[self.int_names[0]] = self.key
+        self.nat.gen_fields(self.block, self.key, self._lo, self._width,
+                            self._key_lo, self._key_range, self.ctr)
+        ordered = [cols[f] for f in self.fields]
         fidx = list(self.fields).index(self.filter_col)
         self.nat.filter_gather_capture(
             ordered, fidx, self.op, self.scalar,
             [self.outs[f] for f in self.fields], self.count)
-        # score the padded buffer: garbage rows past `count` are sliced off
-        feats = torch.stack([self.outs[f] for f in self.float_names], dim=1)
-        pad = self.mlp.dims[0] - nf
-        h = feats.to(torch.bfloat16)
-        if pad:
-            h = torch.nn.functional.pad(h, (0, pad))
-        return self.mlp._net(h)
+        # garbage rows past `count` are scored too, then sliced off
+        self.nat.featpack([self.outs[f] for f in self.float_names],
+                          self.feats)
+        return self.mlp._net(self.feats)
 
     def capture(self) -> None:
         s = torch.cuda.Stream()

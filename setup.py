@@ -19,7 +19,8 @@ CSRC = os.path.join(ROOT, "arkflow_amd", "csrc")
 sources = [
     os.path.join(CSRC, f)
     for f in ("bindings.cpp", "filter.hip", "hash_agg.hip", "hash_join.hip",
-              "gemm_bf16.hip", "gemm_8phase.hip", "rowops.hip", "attention.hip", "proto_decode.hip", "radix_sort.hip", "json_decode.hip")
+              "gemm_bf16.hip", "gemm_8phase.hip", "rowops.hip", "attention.hip", "proto_decode.hip", "radix_sort.hip", "json_decode.hip",
+              "stepfused.hip")
 ]
 
 setup(
