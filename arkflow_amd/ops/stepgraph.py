@@ -155,15 +155,46 @@ class FusedGenerateFilterInfer:
 class FusedStepSource:
     """Input-SPI facade over FusedGenerateFilterInfer: `read()` yields the
     fully processed batch, so a Stream/bench drives the fused step through
-    the normal (input, pipeline) shape with an empty pipeline."""
+    the normal (input, pipeline) shape with an empty pipeline.
 
-    def __init__(self, fused: FusedGenerateFilterInfer):
-        self.fused = fused
+    With ``ninstances > 1`` the source round-robins over N independent
+    graph instances, each replaying on its OWN HIP stream, and waits for
+    the row-count readback in an executor thread — so with N concurrent
+    ``read()`` calls (engine thread_num / bench --workers) step k+1's
+    kernels run while step k's count drains. Per-step latency is unchanged;
+    throughput approaches pure GPU rate. Instance buffers are reused only
+    on that instance's next turn, after its consumer finished."""
+
+    def __init__(self, fused: FusedGenerateFilterInfer,
+                 ninstances: int = 1, make_instance=None):
+        import asyncio
+        self.insts = [fused]
+        self.streams = [torch.cuda.Stream()]
+        self.locks = [asyncio.Lock()]
+        for _ in range(ninstances - 1):
+            self.insts.append(make_instance() if make_instance else fused)
+            self.streams.append(torch.cuda.Stream())
+            self.locks.append(asyncio.Lock())
+        self._i = 0
 
     async def read(self):
+        import asyncio
         from ..spi import NoopAck
-        batch, _ = self.fused.step()
-        return batch, NoopAck()
+        i = self._i
+        self._i = (i + 1) % len(self.insts)
+        inst, stream = self.insts[i], self.streams[i]
+        async with self.locks[i]:
+            loop = asyncio.get_running_loop()
+
+            def run():
+                with torch.cuda.stream(stream):
+                    batch, _ = inst.step()
+                return batch
+
+            if len(self.insts) == 1:
+                return run(), NoopAck()
+            batch = await loop.run_in_executor(None, run)
+            return batch, NoopAck()
 
     async def connect(self):  # pragma: no cover - trivial
         pass

@@ -108,10 +108,18 @@ def build_pipeline(args, device):
             FusedGenerateFilterInfer, FusedStepSource)
         mlp = MlpAnomalyDetector(n_features, [args.hidden, args.hidden],
                                  device, 1234)
-        fused = FusedGenerateFilterInfer(
-            fields, args.batch_size, "f0", ">=", 0.2, mlp, device,
-            seed=7 + args.rank)
-        return FusedStepSource(fused), Pipeline([])
+
+        def make(seed_off=0):
+            return FusedGenerateFilterInfer(
+                fields, args.batch_size, "f0", ">=", 0.2, mlp, device,
+                seed=7 + args.rank + seed_off * 1000)
+
+        if args.workers == 1:
+            args.workers = 2  # pipeline two graph instances by default
+        seeds = iter(range(1, 64))
+        src = FusedStepSource(make(0), ninstances=args.workers,
+                              make_instance=lambda: make(next(seeds)))
+        return src, Pipeline([])
     infer = InferenceProcessor({
         "model": "mlp_anomaly",
         "columns": [f"f{i}" for i in range(n_features)],
