@@ -176,6 +176,13 @@ class FusedStepSource:
             self.streams.append(torch.cuda.Stream())
             self.locks.append(asyncio.Lock())
         self._i = 0
+        # capture serially up front: concurrent first-use capture from two
+        # executor threads crashes the runtime
+        for inst, stream in zip(self.insts, self.streams):
+            if inst._graph is None:
+                with torch.cuda.stream(stream):
+                    inst.capture()
+                stream.synchronize()
 
     async def read(self):
         import asyncio
