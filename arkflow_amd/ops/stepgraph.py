@@ -157,51 +157,55 @@ class FusedStepSource:
     fully processed batch, so a Stream/bench drives the fused step through
     the normal (input, pipeline) shape with an empty pipeline.
 
-    With ``ninstances > 1`` the source round-robins over N independent
-    graph instances, each replaying on its OWN HIP stream, and waits for
-    the row-count readback in an executor thread — so with N concurrent
-    ``read()`` calls (engine thread_num / bench --workers) step k+1's
-    kernels run while step k's count drains. Per-step latency is unchanged;
-    throughput approaches pure GPU rate. Instance buffers are reused only
-    on that instance's next turn, after its consumer finished."""
+    With two instances the source software-pipelines in ONE thread: each
+    ``read()`` first launches the NEXT step's replay on the other
+    instance's stream, then drains the PREVIOUS step's count readback —
+    so step k+1's kernels run while step k's result is consumed. An
+    instance's static buffers are re-replayed only one full read later,
+    after the sequential engine loop has processed its batch; callers that
+    retain batches across steps must copy (the ring-buffer contract)."""
 
     def __init__(self, fused: FusedGenerateFilterInfer,
                  ninstances: int = 1, make_instance=None):
-        import asyncio
         self.insts = [fused]
         self.streams = [torch.cuda.Stream()]
-        self.locks = [asyncio.Lock()]
-        for _ in range(ninstances - 1):
+        for _ in range(max(ninstances, 1) - 1):
             self.insts.append(make_instance() if make_instance else fused)
             self.streams.append(torch.cuda.Stream())
-            self.locks.append(asyncio.Lock())
         self._i = 0
-        # capture serially up front: concurrent first-use capture from two
-        # executor threads crashes the runtime
+        self._pending = None  # index replayed but not yet consumed
         for inst, stream in zip(self.insts, self.streams):
             if inst._graph is None:
                 with torch.cuda.stream(stream):
                     inst.capture()
                 stream.synchronize()
 
+    def _replay(self, i: int) -> None:
+        with torch.cuda.stream(self.streams[i]):
+            self.insts[i]._graph.replay()
+
+    def _consume(self, i: int):
+        with torch.cuda.stream(self.streams[i]):
+            inst = self.insts[i]
+            kept = int(inst.count.item())
+            cols = {f: Column("numeric", inst.outs[f][:kept])
+                    for f in inst.fields}
+            cols["score"] = Column("numeric", inst._scores[:kept])
+            return MessageBatch(cols, input_name="generate")
+
     async def read(self):
-        import asyncio
         from ..spi import NoopAck
-        i = self._i
-        self._i = (i + 1) % len(self.insts)
-        inst, stream = self.insts[i], self.streams[i]
-        async with self.locks[i]:
-            loop = asyncio.get_running_loop()
-
-            def run():
-                with torch.cuda.stream(stream):
-                    batch, _ = inst.step()
-                return batch
-
-            if len(self.insts) == 1:
-                return run(), NoopAck()
-            batch = await loop.run_in_executor(None, run)
+        if len(self.insts) == 1:
+            batch, _ = self.insts[0].step()
             return batch, NoopAck()
+        if self._pending is None:  # prime the pipeline
+            self._replay(0)
+            self._pending = 0
+        nxt = 1 - self._pending
+        self._replay(nxt)  # overlaps with draining the pending step
+        batch = self._consume(self._pending)
+        self._pending = nxt
+        return batch, NoopAck()
 
     async def connect(self):  # pragma: no cover - trivial
         pass

@@ -114,11 +114,10 @@ def build_pipeline(args, device):
                 fields, args.batch_size, "f0", ">=", 0.2, mlp, device,
                 seed=7 + args.rank + seed_off * 1000)
 
-        if args.workers == 1:
-            args.workers = 2  # pipeline two graph instances by default
-        seeds = iter(range(1, 64))
-        src = FusedStepSource(make(0), ninstances=args.workers,
-                              make_instance=lambda: make(next(seeds)))
+        # two graph instances software-pipeline inside the source; the
+        # engine loop itself stays sequential (workers=1)
+        src = FusedStepSource(make(0), ninstances=2,
+                              make_instance=lambda: make(1))
         return src, Pipeline([])
     infer = InferenceProcessor({
         "model": "mlp_anomaly",
