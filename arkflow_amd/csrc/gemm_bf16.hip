@@ -142,6 +142,142 @@ void gemm_bf16_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- BK=64 128² tile ---------------------------------------------------------
+// Same 128×128 tile as gemm_bf16_kernel but a 64-deep K slice per stage:
+// halves the K-loop trip count (and the __syncthreads per K) for the
+// K=768-class BERT shapes where the BK=32 kernel leaves ~300 TF on the
+// table vs hipBLASLt (profiles r06/r17). LDS 16 KiB/operand (32 KiB total
+// → still 2 blocks/CU); staging is 4 global_load_lds iterations per wave;
+// the MFMA loop gains an inner ks∈{0,1} step reading fragments at
+// [row*64 + ks*32 + fk].
+#define BK2 64
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 2)
+void gemm_bf16_k64_kernel(const __bf16* __restrict__ A,   // [M,K]
+                          const __bf16* __restrict__ Bt,  // [N,K]
+                          const float* __restrict__ bias, // [N] or null
+                          __bf16* __restrict__ C,         // [M,N]
+                          int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * BM, col0 = bn * BN;
+
+  __shared__ __bf16 Asm[BM * BK2];  // [128][64] linear, 16 KiB
+  __shared__ __bf16 Bsm[BN * BK2];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+
+  f32x4 acc[4][4] = {};
+
+  const int lin0 = wid * 1024 + lane * 16;
+
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int lin = lin0 + it * 4096;          // byte offset in the 16 KiB tile
+      int trow = lin >> 7;                 // 128 B per row (64 bf16)
+      int tcol = lin & 127;
+      int ga_row = row0 + trow;
+      ga_row = ga_row < M ? ga_row : M - 1;
+      const char* a_src = (const char*)(A + (int64_t)ga_row * K + k0) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)a_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Asm + lin),
+          16, 0, 0);
+      int gb_row = col0 + trow;
+      gb_row = gb_row < N ? gb_row : N - 1;
+      const char* b_src = (const char*)(Bt + (int64_t)gb_row * K + k0) + tcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)b_src,
+          (__attribute__((address_space(3))) uint32_t*)((char*)Bsm + lin),
+          16, 0, 0);
+    }
+    __syncthreads();
+
+    const int fr = lane & 15;
+    const int fk = (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        a_frag[m] =
+            *(const bf16x8*)&Asm[(wm + m * 16 + fr) * BK2 + ks * 32 + fk];
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        b_frag[n] =
+            *(const bf16x8*)&Bsm[(wn + n * 16 + fr) * BK2 + ks * 32 + fk];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" int launch_gemm_bf16_k64(const void* A, const void* Bt,
+                                    const float* bias, void* C, int M, int N,
+                                    int K, int act, hipStream_t st) {
+  if (K % BK2 != 0) return -1;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define KDISPATCH(ACT)                                                       \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_k64_kernel<ACT, true><<<grid, block, 0, st>>>(               \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_k64_kernel<ACT, false><<<grid, block, 0, st>>>(              \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: KDISPATCH(ACT_RELU); break;
+    case ACT_GELU: KDISPATCH(ACT_GELU); break;
+    case ACT_SILU: KDISPATCH(ACT_SILU); break;
+    default: KDISPATCH(ACT_NONE); break;
+  }
+#undef KDISPATCH
+  return 0;
+}
+
 // ---- skinny 64×64 tile -------------------------------------------------------
 // For narrow outputs (N ≤ a few hundred: the MLP anomaly scorer's 32→256→256
 // chain) the 128² tile yields only tiles_m×⌈N/128⌉ workgroups — 128 blocks at
