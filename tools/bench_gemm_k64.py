@@ -1,5 +1,9 @@
 """BK=64 128-tile vs BK=32 vs hipBLASLt at BERT shapes."""
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
