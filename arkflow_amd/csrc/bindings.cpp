@@ -358,13 +358,25 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
     // narrow outputs: the 128² grid can't fill 256 CUs (e.g. [8192,256] →
     // 128 blocks); the 64² tile quadruples the grid (profiles r2)
     int64_t t128grid = ((M + 127) / 128) * ((N + 127) / 128);
-    if (t128grid < 208)
+    if (t128grid < 208) {
       launch_gemm_bf16_skinny(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                               C.data_ptr(), (int)M, (int)N, (int)K,
                               (int)act, cur_stream());
-    else
-      launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr, C.data_ptr(),
-                       (int)M, (int)N, (int)K, (int)act, cur_stream());
+    } else {
+      // K≤1024 multiples of 64 (BERT fc1/QKV class): the BK=64 slice wins
+      // ~10-15% over BK=32 (profiles r2 bench_gemm_k64); at deep K (fc2
+      // K=3072) BK=32 stays ahead
+      int rc2 = (K % 64 == 0 && K <= 1024)
+                    ? launch_gemm_bf16_k64(A.data_ptr(), Bt.data_ptr(),
+                                           bias_ptr, C.data_ptr(), (int)M,
+                                           (int)N, (int)K, (int)act,
+                                           cur_stream())
+                    : -1;
+      if (rc2 != 0)
+        launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                         C.data_ptr(), (int)M, (int)N, (int)K, (int)act,
+                         cur_stream());
+    }
   }
   return C;
 }
