@@ -88,6 +88,20 @@ void launch_scan_partials(const int32_t*, int64_t, int64_t*, hipStream_t);
 void launch_scan_write(const int32_t*, int64_t, const int64_t*, int64_t*,
                        hipStream_t);
 int radix_sort_nblocks(int64_t);
+int onesweep_nblocks(int64_t);
+void launch_onesweep_hist_u32(const uint32_t*, int64_t, int32_t*,
+                              hipStream_t);
+void launch_onesweep_hist_u64(const uint64_t*, int64_t, int32_t*,
+                              hipStream_t);
+void launch_onesweep_scan(int32_t*, int, hipStream_t);
+void launch_onesweep_pass_u32(const uint32_t*, const int32_t*, uint32_t*,
+                              int32_t*, int64_t, int, const int32_t*,
+                              uint32_t*, int32_t*, hipStream_t);
+void launch_onesweep_pass_u64(const uint64_t*, const int32_t*, uint64_t*,
+                              int32_t*, int64_t, int, const int32_t*,
+                              uint32_t*, int32_t*, hipStream_t);
+void launch_i32_to_ordered(const int32_t*, uint32_t*, int64_t, int,
+                           hipStream_t);
 void launch_radix_hist_u32(const uint32_t*, const int32_t*, int64_t, int,
                            int32_t*, int, hipStream_t);
 void launch_radix_scatter_u32(const uint32_t*, const int32_t*, int64_t, int,
@@ -600,57 +614,72 @@ std::vector<torch::Tensor> gather_columns(std::vector<torch::Tensor> cols,
 }
 
 torch::Tensor radix_argsort(torch::Tensor keys, bool descending) {
+  // Onesweep: 8-bit digits + decoupled lookback, keys ping-ponged alongside
+  // the permutation (radix_sort.hip onesweep_* kernels). f32/i32 → 4 chained
+  // passes, i64 → 8. The 4-bit two-kernel LSD this replaces measured ~6× off
+  // rocPRIM at 10M keys (profiles r04 / VERDICT weak #7).
   check_cuda(keys, "keys");
   int64_t n = keys.numel();
   auto opts32 = keys.options().dtype(torch::kInt32);
   if (n == 0) return torch::empty({0}, opts32);
+  TORCH_CHECK(n < (int64_t)1 << 30, "radix_argsort caps at 2^30 rows");
   auto st = cur_stream();
   bool wide;  // 64-bit key path
-  torch::Tensor ordered;
+  torch::Tensor ka;
   if (keys.scalar_type() == torch::kFloat32) {
     wide = false;
-    ordered = torch::empty({n}, opts32);  // uint32 bits in an int32 tensor
+    ka = torch::empty({n}, opts32);  // uint32 bits in an int32 tensor
     launch_f32_to_ordered(keys.data_ptr<float>(),
-                          (uint32_t*)ordered.data_ptr<int32_t>(), n,
+                          (uint32_t*)ka.data_ptr<int32_t>(), n,
                           descending ? 1 : 0, st);
-  } else if (keys.scalar_type() == torch::kInt64 ||
-             keys.scalar_type() == torch::kInt32) {
+  } else if (keys.scalar_type() == torch::kInt32) {
+    wide = false;
+    ka = torch::empty({n}, opts32);
+    launch_i32_to_ordered(keys.data_ptr<int32_t>(),
+                          (uint32_t*)ka.data_ptr<int32_t>(), n,
+                          descending ? 1 : 0, st);
+  } else if (keys.scalar_type() == torch::kInt64) {
     wide = true;
-    auto k64 = keys.scalar_type() == torch::kInt64 ? keys
-                                                   : keys.to(torch::kInt64);
-    ordered = torch::empty({n}, keys.options().dtype(torch::kInt64));
-    launch_i64_to_ordered(k64.data_ptr<int64_t>(),
-                          (uint64_t*)ordered.data_ptr<int64_t>(), n,
+    ka = torch::empty({n}, keys.options().dtype(torch::kInt64));
+    launch_i64_to_ordered(keys.data_ptr<int64_t>(),
+                          (uint64_t*)ka.data_ptr<int64_t>(), n,
                           descending ? 1 : 0, st);
   } else {
     TORCH_CHECK(false, "radix_argsort supports f32/i64/i32 keys");
   }
-  int nblocks = radix_sort_nblocks(n);
+  int npasses = wide ? 8 : 4;
+  int nblocks = onesweep_nblocks(n);
+  auto hist = torch::zeros({(int64_t)npasses * 256}, opts32);
+  if (wide)
+    launch_onesweep_hist_u64((const uint64_t*)ka.data_ptr<int64_t>(), n,
+                             hist.data_ptr<int32_t>(), st);
+  else
+    launch_onesweep_hist_u32((const uint32_t*)ka.data_ptr<int32_t>(), n,
+                             hist.data_ptr<int32_t>(), st);
+  launch_onesweep_scan(hist.data_ptr<int32_t>(), npasses, st);
+  auto status = torch::zeros({(int64_t)npasses * nblocks * 256}, opts32);
+  auto tickets = torch::zeros({npasses}, opts32);
+  auto kb = torch::empty_like(ka);
   auto idx_a = torch::arange(n, opts32);
   auto idx_b = torch::empty({n}, opts32);
-  auto hist = torch::empty({(int64_t)16 * nblocks}, opts32);
-  int total_bits = wide ? 64 : 32;
-  for (int shift = 0; shift < total_bits; shift += 4) {
+  for (int p = 0; p < npasses; ++p) {
+    auto gs = hist.data_ptr<int32_t>() + (int64_t)p * 256;
+    auto stt = (uint32_t*)status.data_ptr<int32_t>() +
+               (int64_t)p * nblocks * 256;
+    auto tkt = tickets.data_ptr<int32_t>() + p;
     if (wide)
-      launch_radix_hist_u64((const uint64_t*)ordered.data_ptr<int64_t>(),
-                            idx_a.data_ptr<int32_t>(), n, shift,
-                            hist.data_ptr<int32_t>(), nblocks, st);
+      launch_onesweep_pass_u64((const uint64_t*)ka.data_ptr<int64_t>(),
+                               idx_a.data_ptr<int32_t>(),
+                               (uint64_t*)kb.data_ptr<int64_t>(),
+                               idx_b.data_ptr<int32_t>(), n, 8 * p, gs, stt,
+                               tkt, st);
     else
-      launch_radix_hist_u32((const uint32_t*)ordered.data_ptr<int32_t>(),
-                            idx_a.data_ptr<int32_t>(), n, shift,
-                            hist.data_ptr<int32_t>(), nblocks, st);
-    auto inc = hist.cumsum(0, torch::kInt32);
-    auto offs = (inc - hist).contiguous();
-    if (wide)
-      launch_radix_scatter_u64((const uint64_t*)ordered.data_ptr<int64_t>(),
-                               idx_a.data_ptr<int32_t>(), n, shift,
-                               offs.data_ptr<int32_t>(), nblocks,
-                               idx_b.data_ptr<int32_t>(), st);
-    else
-      launch_radix_scatter_u32((const uint32_t*)ordered.data_ptr<int32_t>(),
-                               idx_a.data_ptr<int32_t>(), n, shift,
-                               offs.data_ptr<int32_t>(), nblocks,
-                               idx_b.data_ptr<int32_t>(), st);
+      launch_onesweep_pass_u32((const uint32_t*)ka.data_ptr<int32_t>(),
+                               idx_a.data_ptr<int32_t>(),
+                               (uint32_t*)kb.data_ptr<int32_t>(),
+                               idx_b.data_ptr<int32_t>(), n, 8 * p, gs, stt,
+                               tkt, st);
+    std::swap(ka, kb);
     std::swap(idx_a, idx_b);
   }
   return idx_a;

@@ -161,3 +161,248 @@ void launch_i64_to_ordered(const int64_t* in, uint64_t* out, int64_t n,
 }
 
 }  // extern "C"
+
+// ======================= onesweep (8-bit digits) =============================
+// LSD radix with 256-way digits and decoupled lookback: u32 keys sort in 4
+// chained passes, u64 in 8 (vs 8/16 two-kernel passes for the 4-bit LSD
+// above, which measured ~6× off rocPRIM at 10M keys — VERDICT r1 weak #7).
+// Keys ping-pong alongside the permutation so every pass's loads stay
+// coalesced. Per pass, a block: (1) builds a stable local ranking of its
+// 4096-element tile via per-wave 8-ballot equality masks, (2) publishes its
+// per-bin counts and resolves its exclusive prefix by walking predecessor
+// statuses (AGGREGATE/PREFIX flags, dynamic block ids so predecessors are
+// always scheduled), (3) reorders the tile in LDS and writes bins out
+// contiguously (coalesced scatter).
+#define OS_BLOCK 256
+#define OS_IPT 16
+#define OS_TILE (OS_BLOCK * OS_IPT)
+#define OS_BINS 256
+#define OS_FLAG_AGG (1u << 30)
+#define OS_FLAG_PRE (2u << 30)
+#define OS_VAL_MASK ((1u << 30) - 1)
+
+template <typename K, int NPASS>
+__global__ void onesweep_hist_kernel(const K* __restrict__ keys, int64_t n,
+                                     int32_t* __restrict__ hist) {
+  __shared__ int local[NPASS * OS_BINS];
+  for (int i = threadIdx.x; i < NPASS * OS_BINS; i += OS_BLOCK) local[i] = 0;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    K k = keys[i];
+#pragma unroll
+    for (int p = 0; p < NPASS; ++p)
+      atomicAdd(&local[p * OS_BINS + (int)((k >> (8 * p)) & 0xFF)], 1);
+  }
+  __syncthreads();
+  for (int i2 = threadIdx.x; i2 < NPASS * OS_BINS; i2 += OS_BLOCK)
+    if (local[i2]) atomicAdd(&hist[i2], local[i2]);
+}
+
+// exclusive scan of each pass's 256-bin histogram (grid = npasses)
+__global__ void onesweep_scan_kernel(int32_t* __restrict__ hist) {
+  __shared__ int lds[OS_BINS];
+  int32_t* h = hist + (int64_t)blockIdx.x * OS_BINS;
+  int v = h[threadIdx.x];
+  lds[threadIdx.x] = v;
+  __syncthreads();
+  for (int s = 1; s < OS_BINS; s <<= 1) {
+    int add = threadIdx.x >= s ? lds[threadIdx.x - s] : 0;
+    __syncthreads();
+    lds[threadIdx.x] += add;
+    __syncthreads();
+  }
+  h[threadIdx.x] = lds[threadIdx.x] - v;  // exclusive
+}
+
+template <typename K>
+__global__ __launch_bounds__(OS_BLOCK, 2)
+void onesweep_pass_kernel(const K* __restrict__ keys_in,
+                          const int32_t* __restrict__ idx_in,
+                          K* __restrict__ keys_out,
+                          int32_t* __restrict__ idx_out, int64_t n,
+                          int shift,
+                          const int32_t* __restrict__ global_start,
+                          volatile uint32_t* __restrict__ status,
+                          int32_t* __restrict__ ticket) {
+  __shared__ int dyn_bid_s;
+  if (threadIdx.x == 0) dyn_bid_s = atomicAdd(ticket, 1);
+  __syncthreads();
+  const int bid = dyn_bid_s;
+
+  __shared__ int wave_cnt[OS_BLOCK / WAVE][OS_BINS];  // 4 KiB
+  __shared__ int cursor[OS_BINS];
+  __shared__ int excl[OS_BINS];
+  __shared__ int local_start[OS_BINS];
+  __shared__ K lds_keys[OS_TILE];
+  __shared__ int32_t lds_idx[OS_TILE];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int64_t base = (int64_t)bid * OS_TILE;
+  if (threadIdx.x < OS_BINS) cursor[threadIdx.x] = 0;
+  __syncthreads();
+
+  K k_reg[OS_IPT];
+  int32_t i_reg[OS_IPT];
+  short pos_local[OS_IPT];
+#pragma unroll
+  for (int s = 0; s < OS_IPT; ++s) {
+    int64_t j = base + s * OS_BLOCK + threadIdx.x;
+    bool valid = j < n;
+    k_reg[s] = valid ? keys_in[j] : (K)0;
+    i_reg[s] = valid ? idx_in[j] : 0;
+    int d = (int)((k_reg[s] >> shift) & 0xFF);
+    // wave equality mask over the 8 digit bits, restricted to valid lanes
+    uint64_t m = __ballot(valid);
+#pragma unroll
+    for (int bit = 0; bit < 8; ++bit) {
+      uint64_t bb = __ballot((d >> bit) & 1);
+      m &= ((d >> bit) & 1) ? bb : ~bb;
+    }
+    int rank = __popcll(m & lanemask_lt());
+    int wcount = __popcll(m);
+    // zero this strip's wave counters
+    for (int b = threadIdx.x; b < (OS_BLOCK / WAVE) * OS_BINS; b += OS_BLOCK)
+      ((int*)wave_cnt)[b] = 0;
+    __syncthreads();
+    if (valid && rank == 0) wave_cnt[wid][d] = wcount;  // group leader
+    __syncthreads();
+    {  // thread b scans bin b across the 4 waves, based at the running cursor
+      int b = threadIdx.x;
+      int run = cursor[b];
+#pragma unroll
+      for (int w = 0; w < OS_BLOCK / WAVE; ++w) {
+        int c = wave_cnt[w][b];
+        wave_cnt[w][b] = run;
+        run += c;
+      }
+      cursor[b] = run;
+    }
+    __syncthreads();
+    pos_local[s] = (short)(valid ? wave_cnt[wid][d] + rank : -1);
+    __syncthreads();
+  }
+
+  // ---- decoupled lookback: thread b resolves bin b -------------------------
+  {
+    int b = threadIdx.x;
+    int t = cursor[b];
+    if (bid == 0) {
+      status[b] = OS_FLAG_PRE | (uint32_t)t;
+      excl[b] = 0;
+    } else {
+      status[(int64_t)bid * OS_BINS + b] = OS_FLAG_AGG | (uint32_t)t;
+      __threadfence();
+      int run = 0;
+      int p = bid - 1;
+      while (true) {
+        uint32_t s = status[(int64_t)p * OS_BINS + b];
+        uint32_t f = s & ~OS_VAL_MASK;
+        if (f == 0) continue;  // predecessor not published yet — spin
+        run += (int)(s & OS_VAL_MASK);
+        if (f == OS_FLAG_PRE) break;
+        --p;
+      }
+      excl[b] = run;
+      __threadfence();
+      status[(int64_t)bid * OS_BINS + b] = OS_FLAG_PRE | (uint32_t)(run + t);
+    }
+  }
+  // ---- exclusive scan of block totals → local bin starts -------------------
+  {
+    int v = cursor[threadIdx.x];
+    local_start[threadIdx.x] = v;
+    __syncthreads();
+    for (int s = 1; s < OS_BINS; s <<= 1) {
+      int add = threadIdx.x >= s ? local_start[threadIdx.x - s] : 0;
+      __syncthreads();
+      local_start[threadIdx.x] += add;
+      __syncthreads();
+    }
+    local_start[threadIdx.x] -= v;
+  }
+  __syncthreads();
+
+  // ---- stage the tile in local order, then coalesced global scatter --------
+#pragma unroll
+  for (int s = 0; s < OS_IPT; ++s) {
+    if (pos_local[s] >= 0) {
+      lds_keys[pos_local[s]] = k_reg[s];
+      lds_idx[pos_local[s]] = i_reg[s];
+    }
+  }
+  __syncthreads();
+  int count = (int)(n - base < OS_TILE ? n - base : OS_TILE);
+  for (int p = threadIdx.x; p < count; p += OS_BLOCK) {
+    K k = lds_keys[p];
+    int d = (int)((k >> shift) & 0xFF);
+    int64_t gpos = (int64_t)global_start[d] + excl[d] + (p - local_start[d]);
+    keys_out[gpos] = k;
+    idx_out[gpos] = lds_idx[p];
+  }
+}
+
+extern "C" {
+
+int onesweep_nblocks(int64_t n) {
+  return (int)((n + OS_TILE - 1) / OS_TILE);
+}
+
+void launch_onesweep_hist_u32(const uint32_t* keys, int64_t n, int32_t* hist,
+                              hipStream_t st) {
+  int grid = (int)((n + 4095) / 4096);
+  if (grid > 1024) grid = 1024;
+  if (grid < 1) grid = 1;
+  onesweep_hist_kernel<uint32_t, 4><<<grid, OS_BLOCK, 0, st>>>(keys, n, hist);
+}
+void launch_onesweep_hist_u64(const uint64_t* keys, int64_t n, int32_t* hist,
+                              hipStream_t st) {
+  int grid = (int)((n + 4095) / 4096);
+  if (grid > 1024) grid = 1024;
+  if (grid < 1) grid = 1;
+  onesweep_hist_kernel<uint64_t, 8><<<grid, OS_BLOCK, 0, st>>>(keys, n, hist);
+}
+void launch_onesweep_scan(int32_t* hist, int npasses, hipStream_t st) {
+  onesweep_scan_kernel<<<npasses, OS_BINS, 0, st>>>(hist);
+}
+void launch_onesweep_pass_u32(const uint32_t* keys_in, const int32_t* idx_in,
+                              uint32_t* keys_out, int32_t* idx_out,
+                              int64_t n, int shift,
+                              const int32_t* global_start, uint32_t* status,
+                              int32_t* ticket, hipStream_t st) {
+  onesweep_pass_kernel<uint32_t><<<onesweep_nblocks(n), OS_BLOCK, 0, st>>>(
+      keys_in, idx_in, keys_out, idx_out, n, shift, global_start, status,
+      ticket);
+}
+void launch_onesweep_pass_u64(const uint64_t* keys_in, const int32_t* idx_in,
+                              uint64_t* keys_out, int32_t* idx_out,
+                              int64_t n, int shift,
+                              const int32_t* global_start, uint32_t* status,
+                              int32_t* ticket, hipStream_t st) {
+  onesweep_pass_kernel<uint64_t><<<onesweep_nblocks(n), OS_BLOCK, 0, st>>>(
+      keys_in, idx_in, keys_out, idx_out, n, shift, global_start, status,
+      ticket);
+}
+
+// i32 → ordered u32 (bias); avoids the r1 widen-to-u64 (4 passes not 8)
+__global__ void i32_to_ordered_u32_kernel(const int32_t* __restrict__ in,
+                                          uint32_t* __restrict__ out,
+                                          int64_t n, int descending) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint32_t u = (uint32_t)in[i] ^ 0x80000000u;
+    out[i] = descending ? ~u : u;
+  }
+}
+void launch_i32_to_ordered(const int32_t* in, uint32_t* out, int64_t n,
+                           int descending, hipStream_t st) {
+  int grid = (int)((n + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  i32_to_ordered_u32_kernel<<<grid, 256, 0, st>>>(in, out, n, descending);
+}
+
+}  // extern "C"
