@@ -224,7 +224,7 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
                           int32_t* __restrict__ idx_out, int64_t n,
                           int shift,
                           const int32_t* __restrict__ global_start,
-                          volatile uint32_t* __restrict__ status,
+                          uint32_t* __restrict__ status,
                           int32_t* __restrict__ ticket) {
   __shared__ int dyn_bid_s;
   if (threadIdx.x == 0) dyn_bid_s = atomicAdd(ticket, 1);
@@ -286,19 +286,25 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
   }
 
   // ---- decoupled lookback: thread b resolves bin b -------------------------
+  // agent-scope atomics: the 8 XCDs have private L2s, so plain volatile
+  // loads/stores are NOT cross-block coherent on this part
   {
     int b = threadIdx.x;
     int t = cursor[b];
     if (bid == 0) {
-      status[b] = OS_FLAG_PRE | (uint32_t)t;
+      __hip_atomic_store(&status[b], OS_FLAG_PRE | (uint32_t)t,
+                         __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
       excl[b] = 0;
     } else {
-      status[(int64_t)bid * OS_BINS + b] = OS_FLAG_AGG | (uint32_t)t;
-      __threadfence();
+      __hip_atomic_store(&status[(int64_t)bid * OS_BINS + b],
+                         OS_FLAG_AGG | (uint32_t)t, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
       int run = 0;
       int p = bid - 1;
       while (true) {
-        uint32_t s = status[(int64_t)p * OS_BINS + b];
+        uint32_t s = __hip_atomic_load(&status[(int64_t)p * OS_BINS + b],
+                                       __ATOMIC_ACQUIRE,
+                                       __HIP_MEMORY_SCOPE_AGENT);
         uint32_t f = s & ~OS_VAL_MASK;
         if (f == 0) continue;  // predecessor not published yet — spin
         run += (int)(s & OS_VAL_MASK);
@@ -306,8 +312,9 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
         --p;
       }
       excl[b] = run;
-      __threadfence();
-      status[(int64_t)bid * OS_BINS + b] = OS_FLAG_PRE | (uint32_t)(run + t);
+      __hip_atomic_store(&status[(int64_t)bid * OS_BINS + b],
+                         OS_FLAG_PRE | (uint32_t)(run + t),
+                         __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
     }
   }
   // ---- exclusive scan of block totals → local bin starts -------------------
