@@ -333,11 +333,15 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
   __syncthreads();
 
   // ---- stage the tile in local order, then coalesced global scatter --------
+  // pos_local is the element's stable rank WITHIN its bin; the tile
+  // position is local_start[digit] + rank
 #pragma unroll
   for (int s = 0; s < OS_IPT; ++s) {
     if (pos_local[s] >= 0) {
-      lds_keys[pos_local[s]] = k_reg[s];
-      lds_idx[pos_local[s]] = i_reg[s];
+      int d = (int)((k_reg[s] >> shift) & 0xFF);
+      int p = local_start[d] + pos_local[s];
+      lds_keys[p] = k_reg[s];
+      lds_idx[p] = i_reg[s];
     }
   }
   __syncthreads();
