@@ -121,6 +121,9 @@ void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
 void launch_json_copy_strings(const uint8_t*, const int64_t*, const int64_t*,
                               const uint8_t*, int64_t, int64_t, uint8_t*,
                               hipStream_t);
+void launch_json_copy_strings_wave(const uint8_t*, const int64_t*,
+                                   const int64_t*, const uint8_t*, int64_t,
+                                   int64_t, uint8_t*, hipStream_t);
 }
 
 namespace {
@@ -912,11 +915,20 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
       int frow = 0;
       for (int f = 0; f < nf; ++f)
         if (kd[f] == 2 && sl[f] == (int)s) frow = f;
-      if (tot[s] > 0)
-        launch_json_copy_strings(
-            data.data_ptr<uint8_t>(), str_start[s].data_ptr<int64_t>(),
-            offs[s].data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(), n,
-            data.numel(), out.data_ptr<uint8_t>(), cur_stream());
+      if (tot[s] > 0) {
+        // wave-per-doc for long strings (coalesced fast path); the
+        // thread-per-doc kernel stays ahead when strings are tiny
+        if (tot[s] / n >= 32)
+          launch_json_copy_strings_wave(
+              data.data_ptr<uint8_t>(), str_start[s].data_ptr<int64_t>(),
+              offs[s].data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(),
+              n, data.numel(), out.data_ptr<uint8_t>(), cur_stream());
+        else
+          launch_json_copy_strings(
+              data.data_ptr<uint8_t>(), str_start[s].data_ptr<int64_t>(),
+              offs[s].data_ptr<int64_t>(), found[frow].data_ptr<uint8_t>(),
+              n, data.numel(), out.data_ptr<uint8_t>(), cur_stream());
+      }
       strings.emplace_back(out.slice(0, 0, tot[s]), offs[s]);
     }
   }

@@ -57,3 +57,17 @@ payloads = [encode_message(
 pbatch = MessageBatch.from_binary(payloads).to(dev)
 run("proto scalars+strings", ProtobufToArrowProcessor({"proto": proto}, None),
     pbatch)
+
+# long-string case (wave-per-doc copy path; VERDICT #9 target ≥300 GB/s)
+N_LONG = 65_536
+long_payloads = [json.dumps({
+    "id": i,
+    "body": ("lorem ipsum dolor sit amet " * 10) + str(i),
+}).encode() for i in range(N_LONG)]
+lbatch = MessageBatch.from_binary(long_payloads).to(dev)
+_saveN = N
+N = N_LONG
+run("json long strings (~280B body)",
+    JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
+    lbatch)
+N = _saveN
