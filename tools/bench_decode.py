@@ -3,7 +3,11 @@ string schemas). Run on a GPU box:  python tools/bench_decode.py
 """
 import asyncio
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
