@@ -93,13 +93,60 @@ class JsonToArrowProcessor(Processor):
                     from ..errors import ConfigError
                     raise ConfigError(f"json schema type {t!r} not supported "
                                       "(float|int|bool|str)")
+        # schemaless: infer from the FIRST record (reference
+        # component/json.rs:27 infer_json_schema(..., Some(1))) and cache —
+        # repeated batches then hit the same GPU decode kernel as the
+        # fixed-schema path instead of the host pyarrow parser. False =
+        # inference found GPU-unsupported shapes (arrays, >1 nesting level).
+        self._inferred = None
 
-    def _decode_gpu(self, col) -> MessageBatch:
+    def _infer_schema(self, payload: bytes):
+        """First-record schema: scalars + one dict level (dotted names)."""
+        try:
+            doc = json.loads(payload)
+        except Exception:  # noqa: BLE001
+            return False
+        if not isinstance(doc, dict) or not doc:
+            return False
+        schema = {}
+
+        def typ(v):
+            if isinstance(v, bool):
+                return "bool"
+            if isinstance(v, int):
+                return "int"
+            if isinstance(v, float):
+                return "float"
+            if isinstance(v, str):
+                return "str"
+            return None
+
+        for k, v in doc.items():
+            if "." in k:
+                return False  # literal dots collide with nested paths
+            if isinstance(v, dict):
+                for k2, v2 in v.items():
+                    t = typ(v2)
+                    if t is None or "." in k2:
+                        return False
+                    schema[f"{k}.{k2}"] = t
+                continue
+            t = typ(v)
+            if t is None:
+                return False
+            schema[k] = t
+        if self.projection:
+            schema = {k: t for k, t in schema.items()
+                      if k in self.projection}
+        return schema or False
+
+    def _decode_gpu(self, col, schema=None) -> MessageBatch:
+        schema = schema or self.schema
         from .. import ops
         nat = ops.require_native()
         names, kinds, slot = [], [], []
         fcols, icols, scols = [], [], []
-        for name, t in self.schema.items():
+        for name, t in schema.items():
             names.append(name)
             if t == "float":
                 kinds.append(1)
@@ -123,11 +170,11 @@ class JsonToArrowProcessor(Processor):
         all_valid = fbool.all(dim=1).cpu()  # one sync for every field
         for f, name in enumerate(names):
             v = None if bool(all_valid[f]) else fbool[f]
-            if self.schema[name] == "float":
+            if schema[name] == "float":
                 data = out_f[fcols.index(name)].contiguous()
-            elif self.schema[name] == "bool":
+            elif schema[name] == "bool":
                 data = out_i[icols.index(name)].to(torch.bool)
-            elif self.schema[name] in ("str", "string"):
+            elif schema[name] in ("str", "string"):
                 sdata, soffs = strings[scols.index(name)]
                 cols[name] = Column("binary", sdata.contiguous(),
                                     soffs.contiguous(), validity=v)
@@ -137,17 +184,18 @@ class JsonToArrowProcessor(Processor):
             cols[name] = Column("numeric", data, validity=v)
         return MessageBatch(cols)
 
-    def _decode_host_schema(self, col) -> MessageBatch:
+    def _decode_host_schema(self, col, schema=None) -> MessageBatch:
         """CPU fallback for the fixed-schema path: same output columns
         (incl. dotted nested paths) as the GPU kernel."""
+        schema = schema or self.schema
         import json as _json
-        vals = {name: [] for name in self.schema}
+        vals = {name: [] for name in schema}
         for payload in col.to_pylist():
             try:
                 doc = _json.loads(payload)
             except Exception:  # noqa: BLE001
                 raise ProcessError("json decode error (malformed document)")
-            for name in self.schema:
+            for name in schema:
                 cur = doc
                 for part in name.split("."):
                     cur = cur.get(part) if isinstance(cur, dict) else None
@@ -155,7 +203,7 @@ class JsonToArrowProcessor(Processor):
                         break
                 vals[name].append(cur)
         cols = {}
-        for name, t in self.schema.items():
+        for name, t in schema.items():
             v = vals[name]
             if t in ("str", "string"):
                 miss = [not isinstance(x, str) for x in v]
@@ -188,10 +236,19 @@ class JsonToArrowProcessor(Processor):
         if col is None or col.kind != "binary":
             raise ProcessError(
                 f"json_to_arrow: no binary column {self.value_field!r}")
-        if self.schema and col.data.is_cuda:
-            out = self._decode_gpu(col)
-        elif self.schema:
-            out = self._decode_host_schema(col)
+        schema = self.schema
+        if schema is None:
+            if self._inferred is None:
+                lo = int(col.offsets[0].item())
+                hi = int(col.offsets[1].item())
+                first = bytes(col.data[lo:hi].cpu().numpy().tobytes())
+                self._inferred = self._infer_schema(first)
+            if self._inferred:
+                schema = self._inferred
+        if schema and col.data.is_cuda:
+            out = self._decode_gpu(col, schema)
+        elif schema:
+            out = self._decode_host_schema(col, schema)
             if self.device is not None:
                 out = out.to(self.device)
         else:

@@ -395,3 +395,37 @@ def test_json_schema_host_fallback_nested(run):
         assert out.column("amount").to_pylist() == [1.5, 2.0, 3.0]
 
     run(main())
+
+
+def test_json_inferred_schema_matches_fixed(run):
+    """Schemaless json_to_arrow infers from the FIRST record (reference
+    component/json.rs:27) and takes the same typed decode path as a
+    configured schema — including one level of nesting and validity for
+    fields missing in later records."""
+    import json as _json
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+
+    docs = [
+        {"a": 1, "b": 0.5, "ok": True, "s": "x", "u": {"id": 7}},
+        {"a": 2, "b": 1.5, "ok": False, "s": "yy", "u": {"id": 8}},
+        {"a": 3, "b": 2.5, "s": "z", "u": {"id": 9}},  # ok missing
+    ]
+    payloads = [_json.dumps(d).encode() for d in docs]
+    batch = MessageBatch.from_binary(payloads)
+    proc = JsonToArrowProcessor({}, None)
+    out = run(proc.process(batch))[0]
+    assert proc._inferred == {"a": "int", "b": "float", "ok": "bool",
+                              "s": "str", "u.id": "int"}
+    assert out.column("a").to_pylist() == [1, 2, 3]
+    assert out.column("u.id").to_pylist() == [7, 8, 9]
+    assert out.column("s").to_strlist() == ["x", "yy", "z"]
+    assert out.column("ok").validity.tolist() == [True, True, False]
+    # unsupported first record (array value) → host pyarrow path
+    proc2 = JsonToArrowProcessor({}, None)
+    b2 = MessageBatch.from_binary(
+        [_json.dumps({"xs": [1, 2]}).encode()])
+    out2 = run(proc2.process(b2))[0]
+    assert proc2._inferred is False
+    assert out2.num_rows == 1
