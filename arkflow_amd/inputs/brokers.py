@@ -9,9 +9,11 @@ has two drivers:
     (partitions, offsets, consumer groups, transactions) so delivery
     guarantees are TESTABLE offline, the way the reference gates its
     testcontainers suites.
-  - real driver — activated when the matching client library is importable
-    (confluent_kafka / aiokafka / paho-mqtt / nats-py / pulsar-client /
-    redis); connect() raises ConnectionError_ with a clear message otherwise.
+  - real driver — KAFKA ONLY: a full confluent_kafka consumer/transactional-
+    producer implementation (inputs/kafka_real.py), exercised by the same
+    contract test bodies when ``KAFKA_BOOTSTRAP`` is set. The mqtt / nats /
+    pulsar / redis real drivers are NOT implemented — their connect() raises
+    ConnectionError_ and only the fake bus carries their semantics.
 
 Kafka semantics mirrored from the reference:
   input: per-message read, ``__meta_*`` metadata columns, ack = commit offset
@@ -121,7 +123,9 @@ class KafkaInput(Input):
             else "real")
         from ..codecs.helper import build_codec
         self.codec = build_codec(config, resource)
+        self.client_config = config.get("client_config") or {}
         self.bus: Optional[FakeBus] = None
+        self._real = None
         self._closed = False
         # in-memory read positions (per consumer instance): a read advances
         # the position; only ack() commits the offset durably. A NEW consumer
@@ -138,16 +142,17 @@ class KafkaInput(Input):
                     self._positions[(t, p)] = self.bus.commits.get(
                         (self.group, t, p), 0)
             return
-        try:
-            import confluent_kafka  # type: ignore  # noqa: F401
-        except ImportError as e:
-            raise ConnectionError_(
-                "no kafka client library in this environment; "
-                "use driver: memory for the in-process bus") from e
+        from .kafka_real import RealKafkaConsumer
+        self._real = RealKafkaConsumer(
+            [str(b) for b in self.brokers], list(self.topics), self.group,
+            self.client_config)
+        self._real.connect()
 
     async def read(self) -> Tuple[MessageBatch, Ack]:
         if self._closed:
             raise EOFError_("kafka input closed")
+        if self.driver != "memory":
+            return await self._real.read(self.codec)
         while True:
             for topic in self.topics:
                 item = None
@@ -187,6 +192,8 @@ class KafkaInput(Input):
         self._closed = True
         if self.bus:
             self.bus.notify.set()
+        if self._real is not None:
+            self._real.close()
 
 
 class KafkaOutput(Output):
@@ -205,10 +212,14 @@ class KafkaOutput(Output):
         self.key_column = config.get("key_column")
         self.value_column = config.get("value_column", "__value__")
         self.exactly_once = bool(config.get("exactly_once", False))
+        self.transactional_id = config.get("transactional_id")
+        self.compression = config.get("compression")
+        self.client_config = config.get("client_config") or {}
         self.driver = config.get("driver") or (
             "memory" if str(self.brokers[0]).startswith("memory://")
             else "real")
         self.bus: Optional[FakeBus] = None
+        self._real = None
 
     async def connect(self) -> None:
         if self.driver == "memory":
@@ -216,9 +227,12 @@ class KafkaOutput(Output):
             self.bus = FakeBus.get(name)
             self.bus.ensure_topic(self.topic_expr)
             return
-        raise ConnectionError_(
-            "no kafka client library in this environment; "
-            "use driver: memory for the in-process bus")
+        from .kafka_real import RealKafkaProducer
+        self._real = RealKafkaProducer(
+            [str(b) for b in self.brokers], self.exactly_once,
+            transactional_id=self.transactional_id,
+            compression=self.compression, config=self.client_config)
+        self._real.connect()
 
     def _rows(self, batch: MessageBatch) -> List[Tuple[Optional[bytes], bytes]]:
         col = batch.columns.get(self.value_column)
@@ -234,6 +248,12 @@ class KafkaOutput(Output):
         return list(zip(keys, values))
 
     async def write(self, batch: MessageBatch) -> None:
+        if self._real is not None:
+            loop = asyncio.get_running_loop()
+            rows = self._rows(batch)
+            await loop.run_in_executor(
+                None, self._real.write_plain, self.topic_expr, rows)
+            return
         for key, value in self._rows(batch):
             self.bus.produce(self.topic_expr, key, value)
 
@@ -246,9 +266,20 @@ class KafkaOutput(Output):
         staged = []
         for b in batches:
             staged.extend(self._rows(b))
+        if self._real is not None:
+            # real driver: one begin/produce-all/commit transaction with
+            # fencing-aware error mapping (kafka_real.write_txn)
+            loop = asyncio.get_running_loop()
+            await loop.run_in_executor(
+                None, self._real.write_txn, self.topic_expr, staged)
+            return
         # commit point — a failure above leaves the log untouched
         for key, value in staged:
             self.bus.produce(self.topic_expr, key, value)
+
+    async def close(self) -> None:
+        if self._real is not None:
+            self._real.close()
 
 
 # ------------------------------------------------------------- pub/sub family
