@@ -4,9 +4,13 @@ Mirrors reference crates/arkflow-plugin/src/wal/s3.rs (2,203 LoC, the
 reference's most perf-engineered subsystem): in-memory active segment sealed
 on max_entries/max_bytes/flush_interval, parallel PUT workers, batched
 manifest, CRC torn-tail truncation, recovery = manifest ∪ LIST, sealed
-segments ≤ cursor reclaimed. The "object store" here is a directory (same
-interface an S3/object client would implement — no network in this env);
-parallel PUTs use a thread pool like the reference's 1-8 PUT workers.
+segments ≤ cursor reclaimed. The object store is pluggable
+(wal/object_store.py): a directory driver for CI and an S3-compatible SigV4
+driver (MinIO-tested when ``MINIO_ENDPOINT`` is set, like the reference's
+gated minio_integration suite). The manifest is written compare-and-swap
+with precondition retries (manifest.rs PutMode; s3.rs:939
+MANIFEST_WRITE_MAX_RETRIES=8); parallel PUTs use a thread pool like the
+reference's 1-8 PUT workers.
 """
 from __future__ import annotations
 
@@ -18,7 +22,10 @@ from concurrent.futures import ThreadPoolExecutor
 from typing import Iterator, List, Tuple
 
 from ..registry import register
+from .object_store import DirObjectStore, PreconditionFailed, S3ObjectStore
 from .store import _nwal, decode_frames, encode_frame
+
+MANIFEST_WRITE_MAX_RETRIES = 8  # reference s3.rs:939
 
 
 # segment sealing presets (reference wal/config.rs:31 SegmentStrategy)
@@ -38,14 +45,15 @@ class SegmentWalStore:
     def __init__(self, path: str, stream_id: str = "stream",
                  max_entries: int = 1024, max_bytes: int = 4 << 20,
                  flush_interval_secs: float = 0.5, put_workers: int = 2,
-                 compress: bool = False):
+                 compress: bool = False, store=None):
         self.dir = os.path.join(path, stream_id)
-        os.makedirs(self.dir, exist_ok=True)
+        self.store = store if store is not None \
+            else DirObjectStore(self.dir)
         self.max_entries = max_entries
         self.max_bytes = max_bytes
         self.flush_interval = flush_interval_secs
         self.compress = compress
-        self.manifest_path = os.path.join(self.dir, "manifest.json")
+        self._manifest_etag = None
         self._active: List[Tuple[int, bytes]] = []
         self._active_bytes = 0
         self._seg_counter = 0
@@ -61,20 +69,63 @@ class SegmentWalStore:
 
     # ---------------------------------------------------------------- manifest
     def _load_manifest(self) -> None:
+        got = self.store.get("manifest.json")
+        if got is None:
+            return
+        data, etag = got
         try:
-            with open(self.manifest_path) as f:
-                self._manifest = json.load(f)
+            self._manifest = json.loads(data)
+            self._manifest_etag = etag
             self._seg_counter = self._manifest.get("next_seg", 0)
-        except (OSError, json.JSONDecodeError):
+        except json.JSONDecodeError:
             pass
 
     def _store_manifest(self) -> None:
-        self._manifest["next_seg"] = self._seg_counter
-        self._manifest["cursor"] = self._cursor
-        tmp = self.manifest_path + ".tmp"
-        with open(tmp, "w") as f:
-            json.dump(self._manifest, f)
-        os.replace(tmp, self.manifest_path)
+        """Compare-and-swap write with precondition retries: on a lost race
+        (another writer/process advanced the manifest) reload, MERGE — union
+        of segments, max cursor / next_seg — and retry (manifest.rs
+        PutMode)."""
+        for attempt in range(MANIFEST_WRITE_MAX_RETRIES):
+            self._manifest["next_seg"] = self._seg_counter
+            self._manifest["cursor"] = self._cursor
+            body = json.dumps(self._manifest).encode()
+            try:
+                if self._manifest_etag is None:
+                    self._manifest_etag = self.store.put(
+                        "manifest.json", body, if_none_match=True)
+                else:
+                    self._manifest_etag = self.store.put(
+                        "manifest.json", body, if_match=self._manifest_etag)
+                return
+            except PreconditionFailed:
+                got = self.store.get("manifest.json")
+                if got is None:
+                    self._manifest_etag = None
+                    continue
+                data, etag = got
+                try:
+                    theirs = json.loads(data)
+                except json.JSONDecodeError:
+                    theirs = {}
+                mine = self._manifest
+                names = {s["name"] for s in mine.get("segments", [])}
+                merged = list(mine.get("segments", []))
+                for seg in theirs.get("segments", []):
+                    if seg["name"] not in names:
+                        merged.append(seg)
+                merged.sort(key=lambda s: s["name"])
+                self._manifest = {
+                    **theirs,
+                    "segments": merged,
+                    "cursor": max(int(theirs.get("cursor", 0)),
+                                  self._cursor),
+                    "next_seg": max(int(theirs.get("next_seg", 0)),
+                                    self._seg_counter),
+                }
+                self._cursor = self._manifest["cursor"]
+                self._seg_counter = self._manifest["next_seg"]
+                self._manifest_etag = etag
+        raise OSError("manifest CAS retries exhausted")
 
     # ------------------------------------------------------------------ append
     def append_batch(self, entries: List[Tuple[int, bytes]],
@@ -114,11 +165,7 @@ class SegmentWalStore:
         self._store_manifest()
 
     def _put_segment(self, name: str, data: bytes) -> None:
-        p = os.path.join(self.dir, name)
-        with open(p + ".tmp", "wb") as f:
-            f.write(data)
-            os.fsync(f.fileno())
-        os.replace(p + ".tmp", p)
+        self.store.put(name, data)
 
     def wait_puts(self) -> None:
         puts, self._pending_puts = self._pending_puts, []
@@ -132,16 +179,12 @@ class SegmentWalStore:
         self.wait_puts()
         # recovery = manifest ∪ LIST (reference s3.rs:680+)
         names = {s["name"] for s in self._manifest.get("segments", [])}
-        names |= {f for f in os.listdir(self.dir)
-                  if f.startswith("seg-") and f.endswith(".wal")}
+        names |= {f for f in self.store.list("seg-") if f.endswith(".wal")}
         for name in sorted(names):
-            p = os.path.join(self.dir, name)
-            try:
-                with open(p, "rb") as f:
-                    buf = f.read()
-            except OSError:
+            got = self.store.get(name)
+            if got is None:
                 continue
-            for seq, payload in decode_frames(buf):
+            for seq, payload in decode_frames(got[0]):
                 if seq > cursor:
                     yield seq, payload
 
@@ -169,10 +212,7 @@ class SegmentWalStore:
         keep = []
         for seg in self._manifest.get("segments", []):
             if seg["last"] <= self._cursor:
-                try:
-                    os.remove(os.path.join(self.dir, seg["name"]))
-                except OSError:
-                    pass
+                self.store.delete(seg["name"])
             else:
                 keep.append(seg)
         self._manifest["segments"] = keep
@@ -199,6 +239,20 @@ def _build_segment_store(config: dict, resource=None) -> SegmentWalStore:
                    if k in ("max_entries", "max_bytes",
                             "flush_interval_secs")})
     config = {**config, **preset}
+    store = None
+    os_cfg = config.get("object_store") or {}
+    if os_cfg.get("type") == "s3" or config.get("backend") == "s3":
+        stream_id = config.get("stream_id", "stream")
+        store = S3ObjectStore(
+            endpoint=os_cfg.get("endpoint")
+            or os.environ.get("MINIO_ENDPOINT", ""),
+            bucket=os_cfg.get("bucket", "arkflow-wal"),
+            prefix=os_cfg.get("prefix", stream_id),
+            access_key=os_cfg.get("access_key"),
+            secret_key=os_cfg.get("secret_key"),
+            region=os_cfg.get("region", "us-east-1"),
+        )
+        store.ensure_bucket()
     return SegmentWalStore(
         config.get("path", "./wal"),
         stream_id=config.get("stream_id", "stream"),
@@ -207,4 +261,5 @@ def _build_segment_store(config: dict, resource=None) -> SegmentWalStore:
         flush_interval_secs=float(config.get("flush_interval_secs", 0.5)),
         put_workers=int(config.get("put_workers", 2)),
         compress=bool(config.get("compress", False)),
+        store=store,
     )
