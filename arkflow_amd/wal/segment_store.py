@@ -62,6 +62,8 @@ class SegmentWalStore:
         self._pool = ThreadPoolExecutor(max_workers=max(1, put_workers))
         self._pending_puts = []
         self._last_seal = time.monotonic()
+        import uuid
+        self._nonce = uuid.uuid4().hex[:6]
         self._manifest = {"watermark": 0, "segments": [], "cursor": 0}
         self._load_manifest()
         self._cursor = int(self._manifest.get("cursor", 0))
@@ -150,7 +152,9 @@ class SegmentWalStore:
         self._active = []
         self._active_bytes = 0
         self._last_seal = time.monotonic()
-        name = f"seg-{self._seg_counter:08d}.wal"
+        # writer nonce keeps names unique when several writers share one
+        # object-store prefix (their manifests merge via CAS)
+        name = f"seg-{self._seg_counter:08d}-{self._nonce}.wal"
         self._seg_counter += 1
         first, last = entries[0][0], entries[-1][0]
         if _nwal is not None and not self.compress:
@@ -180,13 +184,16 @@ class SegmentWalStore:
         # recovery = manifest ∪ LIST (reference s3.rs:680+)
         names = {s["name"] for s in self._manifest.get("segments", [])}
         names |= {f for f in self.store.list("seg-") if f.endswith(".wal")}
+        entries = []
         for name in sorted(names):
             got = self.store.get(name)
             if got is None:
                 continue
             for seq, payload in decode_frames(got[0]):
                 if seq > cursor:
-                    yield seq, payload
+                    entries.append((seq, payload))
+        entries.sort(key=lambda e: e[0])  # replay in ascending seq order
+        yield from entries
 
     def _recover_max_seq(self) -> int:
         mx = 0
