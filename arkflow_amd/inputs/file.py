@@ -2,9 +2,11 @@
 
 Mirrors reference crates/arkflow-plugin/src/input/file.rs (:46-90): format
 selection, streaming record batches (8192 rows), optional SQL query applied
-to each batch. Object-store URLs / Ballista offload are out of scope in this
-offline environment (RCCL sharding replaces remote offload, SURVEY §2.9);
-local paths and globs are supported.
+to each batch, and remote sources: ``http(s)://`` URLs (requests) and
+``s3://bucket/key`` objects (the SigV4 client from wal/object_store.py,
+endpoint from config or MINIO_ENDPOINT/S3_ENDPOINT). Ballista offload is
+deliberately descoped (RCCL sharding replaces remote offload, SURVEY §2.9);
+local paths support globs.
 """
 from __future__ import annotations
 
@@ -52,7 +54,9 @@ class FileInput(Input):
         path = config.get("path")
         if not path:
             raise ConfigError("file input requires 'path'")
-        self.paths = sorted(globmod.glob(path)) if any(
+        self.endpoint = config.get("endpoint")  # s3:// endpoint override
+        remote = path.startswith(("http://", "https://", "s3://"))
+        self.paths = sorted(globmod.glob(path)) if not remote and any(
             c in path for c in "*?[") else [path]
         self.format = config.get("format")  # csv|json|parquet|arrow (by ext)
         self.batch_size = int(config.get("batch_size", DEFAULT_RECORD_BATCH))
@@ -68,26 +72,56 @@ class FileInput(Input):
     def _fmt(self, path: str) -> str:
         if self.format:
             return self.format
-        ext = os.path.splitext(path)[1].lower().lstrip(".")
+        ext = os.path.splitext(path.split("?")[0])[1].lower().lstrip(".")
         return {"jsonl": "json", "ndjson": "json", "pq": "parquet",
                 "feather": "arrow", "ipc": "arrow"}.get(ext, ext or "csv")
+
+    def _fetch_remote(self, path: str):
+        """http(s):// and s3://bucket/key sources → pyarrow buffer
+        (reference input/file.rs:46-90 object-store URLs)."""
+        import pyarrow as pa
+        if path.startswith(("http://", "https://")):
+            import requests
+            r = requests.get(path, timeout=60)
+            r.raise_for_status()
+            return pa.BufferReader(r.content)
+        import os as _os
+        rest = path[len("s3://"):]
+        bucket, _, key = rest.partition("/")
+        if not bucket or not key:
+            raise ConfigError(f"bad s3 url {path!r} (s3://bucket/key)")
+        from ..wal.object_store import S3ObjectStore
+        ep = self.endpoint or _os.environ.get("MINIO_ENDPOINT") \
+            or _os.environ.get("S3_ENDPOINT")
+        if not ep:
+            raise ConfigError("s3:// paths need 'endpoint' or MINIO_ENDPOINT")
+        got = S3ObjectStore(ep, bucket).get(key)
+        if got is None:
+            raise ConfigError(f"s3 object not found: {path}")
+        return pa.BufferReader(got[0])
 
     def _load_all(self) -> None:
         import pyarrow as pa
         for path in self.paths:
             fmt = self._fmt(path)
+            src_buf = None
+            if path.startswith(("http://", "https://", "s3://")):
+                src_buf = self._fetch_remote(path)
             if fmt == "csv":
                 import pyarrow.csv as pacsv
-                table = pacsv.read_csv(path)
+                table = pacsv.read_csv(src_buf or path)
             elif fmt == "json":
                 import pyarrow.json as pajson
-                table = pajson.read_json(path)
+                table = pajson.read_json(src_buf or path)
             elif fmt == "parquet":
                 import pyarrow.parquet as pq
-                table = pq.read_table(path)
+                table = pq.read_table(src_buf or path)
             elif fmt == "arrow":
-                with pa.memory_map(path) as src:
-                    table = pa.ipc.open_file(src).read_all()
+                if src_buf is not None:
+                    table = pa.ipc.open_file(src_buf).read_all()
+                else:
+                    with pa.memory_map(path) as src:
+                        table = pa.ipc.open_file(src).read_all()
             else:
                 raise ConfigError(f"unknown file format {fmt!r}")
             for start in range(0, table.num_rows, self.batch_size):

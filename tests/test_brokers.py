@@ -383,3 +383,36 @@ def test_kafka_eos_abort_on_bad_batch(run):
         assert sum(len(p) for p in bus.topics["t"]) == 1
 
     run(main())
+
+
+def test_file_input_http_url(tmp_path, run):
+    """Remote file source over HTTP (reference input/file.rs:46-90 URL
+    reads) — served from an in-process loopback server."""
+    import threading
+    from http.server import HTTPServer, SimpleHTTPRequestHandler
+
+    (tmp_path / "data.csv").write_text("a,b\n1,x\n2,y\n3,z\n")
+
+    class H(SimpleHTTPRequestHandler):
+        def __init__(self, *a, **k):
+            super().__init__(*a, directory=str(tmp_path), **k)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), H)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        from arkflow_amd.inputs.file import FileInput
+        inp = FileInput({
+            "path": f"http://127.0.0.1:{srv.server_port}/data.csv"})
+
+        async def main():
+            batch, _ = await inp.read()
+            assert batch.column("a").to_pylist() == [1, 2, 3]
+            assert batch.column("b").to_strlist() == ["x", "y", "z"]
+
+        run(main())
+    finally:
+        srv.shutdown()
