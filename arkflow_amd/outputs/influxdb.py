@@ -114,31 +114,70 @@ def _build_influx(config, resource=None):
     return InfluxDbOutput(config, resource)
 
 
+class _FakeMongoStore:
+    """Process-global fake document store (driver: memory) so the
+    row→document path is testable offline, like the broker fake bus."""
+
+    collections = {}  # (db, coll) → list of docs
+
+    @classmethod
+    def get(cls, db: str, coll: str):
+        return cls.collections.setdefault((db, coll), [])
+
+    @classmethod
+    def reset(cls):
+        cls.collections.clear()
+
+
 class MongoDbOutput(Output):
-    """`mongodb` output (reference output/mongodb.rs) — document inserts;
-    requires a mongo client library (absent offline → connect() raises)."""
+    """`mongodb` output (reference output/mongodb.rs) — document inserts.
+
+    drivers: ``memory`` (url memory://…, in-process fake store), real via
+    motor (async) or pymongo (executor thread) when importable."""
 
     def __init__(self, config: dict, resource=None):
         self.url = config.get("url", "mongodb://127.0.0.1:27017")
         self.database = config.get("database", "arkflow")
         self.collection = config.get("collection", "events")
+        self.driver = config.get("driver") or (
+            "memory" if self.url.startswith("memory://") else "real")
         self._coll = None
+        self._fake = None
+        self._pym = None
 
     async def connect(self) -> None:
+        if self.driver == "memory":
+            self._fake = _FakeMongoStore.get(self.database, self.collection)
+            return
         try:
             import motor.motor_asyncio as motor  # type: ignore
+            client = motor.AsyncIOMotorClient(self.url)
+            self._coll = client[self.database][self.collection]
+            return
+        except ImportError:
+            pass
+        try:
+            import pymongo  # type: ignore
+            client = pymongo.MongoClient(self.url)
+            self._pym = client[self.database][self.collection]
         except ImportError:
             from ..errors import ConnectionError_
             raise ConnectionError_(
-                "no mongodb client library in this environment") from None
-        client = motor.AsyncIOMotorClient(self.url)
-        self._coll = client[self.database][self.collection]
+                "no mongodb client library (motor/pymongo); use "
+                "driver: memory for the in-process store") from None
 
     async def write(self, batch: MessageBatch) -> None:
         docs = []
         for r in batch.to_rows():
             docs.append({k: _dec(v) for k, v in r.items()})
-        await self._coll.insert_many(docs)
+        if self._fake is not None:
+            self._fake.extend(docs)
+        elif self._pym is not None:
+            import asyncio
+            await asyncio.get_running_loop().run_in_executor(
+                None, self._pym.insert_many, docs)
+        else:
+            await self._coll.insert_many(docs)
 
 
 @register("output", "mongodb",

@@ -416,3 +416,25 @@ def test_file_input_http_url(tmp_path, run):
         run(main())
     finally:
         srv.shutdown()
+
+
+def test_mongodb_output_memory_driver(run):
+    """mongodb output (reference output/mongodb.rs): document conversion +
+    insert against the in-process fake store."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.outputs.influxdb import MongoDbOutput, _FakeMongoStore
+
+    _FakeMongoStore.reset()
+    out = MongoDbOutput({"url": "memory://", "database": "d",
+                         "collection": "c"})
+
+    async def main():
+        await out.connect()
+        await out.write(MessageBatch.from_dict(
+            {"a": [1, 2], "s": ["x", "y"]}))
+
+    run(main())
+    docs = _FakeMongoStore.get("d", "c")
+    assert len(docs) == 2
+    assert docs[0]["a"] == 1 and docs[0]["s"] in ("x", b"x")
+    _FakeMongoStore.reset()
