@@ -317,3 +317,66 @@ def test_sse_last_event_id_resume(run):
         await eng.runtime.stop_all()
 
     run(main(), timeout=60)
+
+
+def test_console_spa_api_contract(run):
+    """Every endpoint the console SPA calls must exist and answer — the
+    build-free analog of the reference console's vitest suites."""
+    async def main():
+        eng = _engine()
+        for sc in eng.config.streams:
+            eng.runtime.register(sc)
+        await eng.runtime.start("s1")
+        eng.ready = True
+        app = create_app(eng)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://n") as c:
+            page = (await c.get("/")).text
+            # the SPA covers the reference console's feature tabs
+            for feature in ("overview", "runtime", "configuration",
+                            "components", "events", "rollouts", "settings"):
+                assert feature in page
+            # endpoints the SPA fetches
+            for path in ("/api/v1/system/status", "/api/v1/streams",
+                         "/api/v1/streams/s1", "/api/v1/streams/s1/metrics",
+                         "/api/v1/operations", "/api/v1/components",
+                         "/api/v1/configuration",
+                         "/api/v1/configuration/versions",
+                         "/api/v1/events?limit=5"):
+                r = await c.get(path)
+                assert r.status_code == 200, path
+            # lifecycle op buttons
+            r = await c.post("/api/v1/streams/s1/restart")
+            assert r.status_code == 200
+            # config validate round-trip with the served config
+            cfg = (await c.get("/api/v1/configuration")).json()
+            r = await c.post("/api/v1/configuration/validate", json=cfg)
+            assert r.status_code == 200
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)
+
+
+def test_console_hub_views(run):
+    """Hub-side console views: /nodes, /intents, /rollouts + actions."""
+    async def main():
+        from arkflow_amd.server.hub import Hub, create_hub_app
+        hub = Hub()
+        await hub.store.upsert_node("n1", "tok", 5.0)
+        app = create_hub_app(hub)
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://h") as c:
+            assert "rollouts" in (await c.get("/")).text
+            assert (await c.get("/nodes")).status_code == 200
+            assert (await c.get("/intents")).status_code == 200
+            r = await c.post("/rollouts",
+                             json={"config": {}, "nodes": ["n1"]})
+            rid = r.json()["rollout_id"]
+            assert (await c.get("/rollouts")).status_code == 200
+            assert (await c.post(f"/rollouts/{rid}/pause")).status_code == 200
+            assert (await c.post(f"/rollouts/{rid}/resume")).status_code \
+                == 200
+
+    run(main())
