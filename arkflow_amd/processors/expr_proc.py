@@ -119,10 +119,115 @@ def _build_expr(config: dict, resource=None) -> ExprProcessor:
     return ExprProcessor(config, resource)
 
 
-# `vrl` alias so reference configs with a vrl processor map onto the columnar
-# expression engine (VRL programs must be rewritten as assignments).
+class VrlProcessor(Processor):
+    """Real VRL programs, row-wise (reference processor/vrl.rs semantics:
+    per-event remap with type-preserving conversion both directions).
+
+    ``on_error``: what to do when an uncaught error / `fn!()` abort hits a
+    row — ``keep`` (default: emit the original row unchanged), ``skip``
+    (drop the row), ``fail`` (fail the batch → error_output path).
+    """
+
+    def __init__(self, config: dict, resource=None):
+        from .vrl_lang import VrlProgram
+        source = config.get("statement") or config.get("source") \
+            or config.get("program")
+        if not source:
+            raise ConfigError("vrl processor requires 'source'")
+        self.program = VrlProgram(source)
+        self.on_error = config.get("on_error", "keep")
+        if self.on_error not in ("keep", "skip", "fail"):
+            raise ConfigError("vrl on_error must be keep|skip|fail")
+
+    async def process(self, batch: MessageBatch) -> List[MessageBatch]:
+        from ..errors import ProcessError
+        from .vrl_lang import VrlAbort, VrlError
+        if batch.num_rows == 0:
+            return []
+        rows = batch.to_rows()
+        out_rows = []
+        for row in rows:
+            # bytes → str for string columns (VRL values are JSON-like)
+            ev = {k: (v.decode("utf-8", "replace")
+                      if isinstance(v, (bytes, bytearray)) else v)
+                  for k, v in row.items()}
+            try:
+                out_rows.append(self.program.remap(dict(ev)))
+            except (VrlError, VrlAbort) as e:
+                if self.on_error == "fail":
+                    raise ProcessError(f"vrl: {e}") from e
+                if self.on_error == "keep":
+                    out_rows.append(ev)
+        if not out_rows:
+            return []
+        return [_rows_to_batch(out_rows, batch.input_name, batch.device)]
+
+
+def _rows_to_batch(rows: List[dict], input_name, device) -> MessageBatch:
+    """Union of keys → typed columns; nested values JSON-encode; missing
+    values carry a validity mask (type-preserving conversion back)."""
+    import json as _json
+
+    import torch
+
+    names: List[str] = []
+    for r in rows:
+        for k in r:
+            if k not in names:
+                names.append(k)
+    cols: Dict[str, Column] = {}
+    for name in names:
+        vals = [r.get(name) for r in rows]
+        present = [v is not None for v in vals]
+        non_null = [v for v in vals if v is not None]
+        if non_null and all(isinstance(v, bool) for v in non_null):
+            c = Column("numeric", torch.tensor(
+                [bool(v) if v is not None else False for v in vals]))
+        elif non_null and all(isinstance(v, int) and not isinstance(v, bool)
+                              for v in non_null):
+            c = Column("numeric", torch.tensor(
+                [int(v) if v is not None else 0 for v in vals],
+                dtype=torch.int64))
+        elif non_null and all(isinstance(v, (int, float))
+                              and not isinstance(v, bool)
+                              for v in non_null):
+            c = Column("numeric", torch.tensor(
+                [float(v) if v is not None else 0.0 for v in vals],
+                dtype=torch.float64))
+        elif non_null and all(isinstance(v, str) for v in non_null):
+            c = Column.from_strings([v if v is not None else ""
+                                     for v in vals])
+        else:  # mixed / nested → JSON encoding
+            c = Column.from_strings(
+                [_json.dumps(v, separators=(",", ":"))
+                 if v is not None else "" for v in vals])
+        if not all(present):
+            c.validity = torch.tensor(present)
+        cols[name] = c
+    b = MessageBatch(cols, input_name)
+    if device is not None and device.type != "cpu":
+        b = b.to(device)
+    return b
+
+
 @register("processor", "vrl",
-          description="Alias of `expr` — columnar VRL-equivalent remapping",
-          example={"type": "vrl", "assignments": {"v2": "value * 2"}})
-def _build_vrl(config: dict, resource=None) -> ExprProcessor:
-    return ExprProcessor(config, resource)
+          description="VRL remap: flat '.col = expr' programs run columnar "
+                      "(tensor ops); full VRL (nested paths, if/else, "
+                      "fallible functions, ??) runs the row-wise interpreter",
+          example={"type": "vrl", "source": '.v2 = .value * 2'})
+def _build_vrl(config: dict, resource=None):
+    source = config.get("statement") or config.get("source") \
+        or config.get("program")
+    if config.get("assignments") or config.get("drop"):
+        return ExprProcessor(config, resource)
+    if source:
+        try:
+            # flat assignment/del subset → columnar fast path (any parse
+            # failure — unsupported statement shape, VRL-only functions —
+            # falls through to the interpreter)
+            probe = dict(config)
+            probe["statement"] = source
+            return ExprProcessor(probe, resource)
+        except Exception:  # noqa: BLE001
+            pass
+    return VrlProcessor(config, resource)
