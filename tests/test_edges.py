@@ -225,10 +225,10 @@ def test_hub_storage_command_lifecycle(run):
         st = HubStore()
         await st.upsert_node("n1", "tok", 10)
         iid = await st.enqueue_intent("n1", "s1", "start")
-        rows = await st.claim_outbox()
+        rows = await st.claim_outbox(["n1"])
         assert rows[0]["intent_id"] == iid
         # second claim returns nothing (claimed=1)
-        assert await st.claim_outbox() == []
+        assert await st.claim_outbox(["n1"]) == []
         aid = await st.create_attempt(iid, "n1", {"kind": "lifecycle"})
         cmds = await st.pending_commands("n1")
         assert cmds[0]["attempt_id"] == aid
