@@ -3,8 +3,11 @@ ring) → sql(agg) with a segmented WAL, on GPU. Reports sustained rows/s and
 error counters. Usage: python tools/soak_durable.py [seconds]
 """
 import asyncio
+import os
 import sys
 import tempfile
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import arkflow_amd as af
 from arkflow_amd.config import EngineConfig
