@@ -249,24 +249,79 @@ DEV_INLINE int64_t unescape_copy(const uint8_t* __restrict__ d, int64_t p,
   return w;
 }
 
-__global__ void json_decode_kernel(const uint8_t* __restrict__ data,
-                                   const int64_t* __restrict__ offsets,
-                                   int64_t n_docs, JsonSpec spec,
-                                   double* __restrict__ out_f64,  // [nd][n]
-                                   int64_t* __restrict__ out_i64,
-                                   int64_t* __restrict__ str_start,  // [ns][n]
-                                   int32_t* __restrict__ str_ulen,   // [ns][n]
-                                   uint8_t* __restrict__ found,  // [nf][n]
-                                   int32_t* __restrict__ err) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n_docs; i += stride) {
-    int64_t p = offsets[i];
-    const int64_t end = offsets[i + 1];
+DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
+                               int64_t p, int64_t end, int64_t boff,
+                               int64_t i, int64_t n_docs,
+                               const JsonSpec& spec,
+                               double* __restrict__ out_f64,
+                               int64_t* __restrict__ out_i64,
+                               int64_t* __restrict__ str_start,
+                               int32_t* __restrict__ str_ulen,
+                               uint8_t* __restrict__ found,
+                               int32_t* __restrict__ err);
+
+// LDS doc cache: each 128-doc tile's byte range is CONTIGUOUS in the binary
+// column, so the block stages it with coalesced loads and threads then walk
+// their documents out of LDS — the thread-per-doc global byte walk was
+// latency-bound (each dependent 8 B load pays HBM latency with only a few
+// waves per CU to hide it). Tiles larger than the cache parse from global.
+#define JSON_BLOCK 128
+#define JSON_LDS_BYTES (48 * 1024)
+
+__global__ __launch_bounds__(JSON_BLOCK)
+void json_decode_kernel(const uint8_t* __restrict__ data,
+                        const int64_t* __restrict__ offsets,
+                        int64_t n_docs, JsonSpec spec,
+                        double* __restrict__ out_f64,  // [nd][n]
+                        int64_t* __restrict__ out_i64,
+                        int64_t* __restrict__ str_start,  // [ns][n]
+                        int32_t* __restrict__ str_ulen,   // [ns][n]
+                        uint8_t* __restrict__ found,  // [nf][n]
+                        int32_t* __restrict__ err) {
+  __shared__ uint8_t cache[JSON_LDS_BYTES];
+  for (int64_t tile = (int64_t)blockIdx.x * JSON_BLOCK; tile < n_docs;
+       tile += (int64_t)gridDim.x * JSON_BLOCK) {
+    int64_t tile_hi = tile + JSON_BLOCK < n_docs ? tile + JSON_BLOCK
+                                                 : n_docs;
+    const int64_t r0 = offsets[tile];
+    const int64_t r1 = offsets[tile_hi];
+    const uint8_t* dv = data;
+    int64_t boff = 0;
+    if (r1 - r0 <= JSON_LDS_BYTES) {
+      const int64_t len = r1 - r0;
+      int64_t b = (int64_t)threadIdx.x * 8;
+      for (; b + 8 <= len; b += (int64_t)JSON_BLOCK * 8)
+        __builtin_memcpy(cache + b, data + r0 + b, 8);
+      if (b < len)
+        for (; b < len; ++b) cache[b] = data[r0 + b];
+      __syncthreads();
+      dv = cache;
+      boff = r0;
+    }
+    int64_t i = tile + threadIdx.x;
+    if (i < tile_hi)
+      json_parse_doc(dv, offsets[i] - boff, offsets[i + 1] - boff, boff, i,
+                     n_docs, spec, out_f64, out_i64, str_start, str_ulen,
+                     found, err);
+    __syncthreads();  // cache reused by the next tile
+  }
+}
+
+DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
+                               int64_t p, int64_t end, int64_t boff,
+                               int64_t i, int64_t n_docs,
+                               const JsonSpec& spec,
+                               double* __restrict__ out_f64,
+                               int64_t* __restrict__ out_i64,
+                               int64_t* __restrict__ str_start,
+                               int32_t* __restrict__ str_ulen,
+                               uint8_t* __restrict__ found,
+                               int32_t* __restrict__ err) {
+  {
     while (p < end && is_ws(data[p])) ++p;
     if (p >= end || data[p] != '{') {
       if (p < end) err[0] = 1;
-      continue;
+      return;
     }
     ++p;
     int ctx = -1;  // current parent-object context (-1 = top level)
@@ -337,7 +392,7 @@ __global__ void json_decode_kernel(const uint8_t* __restrict__ data,
         int64_t s0 = p + 1;
         p = scan_string(data, s0, end, &ulen);
         if (p < end) ++p;  // closing quote
-        str_start[(int64_t)spec.slot[fi] * n_docs + i] = s0;
+        str_start[(int64_t)spec.slot[fi] * n_docs + i] = s0 + boff;
         str_ulen[(int64_t)spec.slot[fi] * n_docs + i] = ulen;
         found[(int64_t)fi * n_docs + i] = 1;
         continue;
@@ -399,10 +454,11 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
     for (int c = 0; c < parent_len[q] && c < JSON_MAX_NAME; ++c)
       spec.parents[q][c] = parents[q * JSON_MAX_NAME + c];
   }
-  int grid = (int)((n_docs + 255) / 256);
-  if (grid > 2048) grid = 2048;
+  int grid = (int)((n_docs + JSON_BLOCK - 1) / JSON_BLOCK);
+  if (grid > 4096) grid = 4096;
   if (grid < 1) return;
-  json_decode_kernel<<<grid, 256, 0, st>>>(data, offsets, n_docs, spec,
+  json_decode_kernel<<<grid, JSON_BLOCK, 0, st>>>(data, offsets, n_docs,
+                                                  spec,
                                            out_f64, out_i64, str_start,
                                            str_ulen, found, err);
 }
