@@ -160,14 +160,16 @@ class JsonToArrowProcessor(Processor):
                 kinds.append(0)
                 slot.append(len(icols))
                 icols.append(name)
-        out_f, out_i, found, err, strings = nat.json_decode(
+        out_f, out_i, found, err, strings, summary = nat.json_decode(
             col.data, col.offsets, names, kinds, slot,
             len(icols), len(fcols), len(scols))
-        if int(err.item()) != 0:
+        # summary = [err, all_valid×nf, string totals×ns] in ONE readback
+        summary = summary.tolist()
+        if summary[0] != 0:
             raise ProcessError("json decode error (malformed document)")
+        all_valid = summary[1:1 + max(len(names), 1)]
         cols = {}
         fbool = found.to(torch.bool)
-        all_valid = fbool.all(dim=1).cpu()  # one sync for every field
         for f, name in enumerate(names):
             v = None if bool(all_valid[f]) else fbool[f]
             if schema[name] == "float":
