@@ -85,6 +85,8 @@ void launch_bytes_match(const uint8_t*, const int64_t*, int64_t,
                         const uint8_t*, int, int, bool*, hipStream_t);
 int scan_grid(int64_t n);
 void launch_scan_partials(const int32_t*, int64_t, int64_t*, hipStream_t);
+void launch_scan_boffs64(const int64_t*, int, int64_t*, int64_t*,
+                         hipStream_t);
 void launch_scan_write(const int32_t*, int64_t, const int64_t*, int64_t*,
                        hipStream_t);
 int radix_sort_nblocks(int64_t);
@@ -117,7 +119,7 @@ void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
                         const char*, const int*, const int*, const int*,
                         const int*, int, const char*, const int*,
                         double*, int64_t*, int64_t*, int32_t*, uint8_t*,
-                        int32_t*, hipStream_t);
+                        int32_t*, int32_t*, hipStream_t);
 void launch_json_copy_strings(const uint8_t*, const int64_t*, const int64_t*,
                               const uint8_t*, int64_t, int64_t, uint8_t*,
                               hipStream_t);
@@ -155,11 +157,20 @@ torch::Tensor exclusive_offsets(torch::Tensor lens) {
   auto partials = torch::empty({grid}, lens.options().dtype(torch::kInt64));
   launch_scan_partials(lens.data_ptr<int32_t>(), n,
                        partials.data_ptr<int64_t>(), st);
-  auto inc = partials.cumsum(0);
-  auto boffs = inc - partials;
+  torch::Tensor boffs;
+  if (grid <= 1024) {
+    // single-block scan writes offsets AND the total slot in one launch
+    boffs = torch::empty({grid}, partials.options());
+    launch_scan_boffs64(partials.data_ptr<int64_t>(), grid,
+                        boffs.data_ptr<int64_t>(),
+                        out.data_ptr<int64_t>() + n, st);
+  } else {
+    auto inc = partials.cumsum(0);
+    boffs = inc - partials;
+    out.narrow(0, n, 1).copy_(inc.narrow(0, grid - 1, 1));
+  }
   launch_scan_write(lens.data_ptr<int32_t>(), n, boffs.data_ptr<int64_t>(),
                     out.data_ptr<int64_t>(), st);
-  out.narrow(0, n, 1).copy_(inc.narrow(0, grid - 1, 1));
   return out;
 }
 
@@ -888,6 +899,8 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   auto found = torch::zeros({std::max(nf, 1), std::max<int64_t>(n, 1)},
                             data.options().dtype(torch::kUInt8));
   auto err = torch::zeros({1}, data.options().dtype(torch::kInt32));
+  auto found_count = torch::zeros({std::max(nf, 1)},
+                                  data.options().dtype(torch::kInt32));
   if (n > 0)
     launch_json_decode(data.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(),
                        n, nf, packed.data(), nl.data(), kd.data(), sl.data(),
@@ -896,7 +909,7 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
                        str_start.data_ptr<int64_t>(),
                        str_ulen.data_ptr<int32_t>(),
                        found.data_ptr<uint8_t>(), err.data_ptr<int32_t>(),
-                       cur_stream());
+                       found_count.data_ptr<int32_t>(), cur_stream());
   // string copy-out: per field, offsets = exclusive cumsum of unescaped
   // lengths; one host sync for ALL totals at once, then one copy kernel per
   // string field (ulen of absent docs is 0 → empty strings, validity=found)
@@ -906,13 +919,13 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
   // stream drains per batch and floored small-batch decode at ~0.2 ms
   std::vector<torch::Tensor> sumv;
   sumv.push_back(err.to(torch::kInt64));
-  if (n > 0 && nf > 0)
-    sumv.push_back(std::get<0>(found.slice(0, 0, nf).min(1))
-                       .to(torch::kInt64));
-  else
-    sumv.push_back(torch::ones({std::max(nf, 1)},
-                               torch::dtype(torch::kInt64)
-                                   .device(data.device())));
+  // all_valid[f] ⇔ found_count[f] == n (counts accumulated in-kernel —
+  // a found.min(dim=1) reduce cost ~40 µs/batch)
+  sumv.push_back((found_count.to(torch::kInt64) ==
+                  torch::full({std::max(nf, 1)}, n,
+                              torch::dtype(torch::kInt64)
+                                  .device(data.device())))
+                     .to(torch::kInt64));
   std::vector<torch::Tensor> offs(std::max<int64_t>(n_str, 0));
   if (n_str > 0 && n > 0) {
     for (int64_t s = 0; s < n_str; ++s) {

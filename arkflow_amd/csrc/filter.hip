@@ -394,6 +394,33 @@ __global__ void scan_write_kernel(const int32_t* __restrict__ in, int64_t n,
   }
 }
 
+// single-block exclusive scan of ≤1024 int64 block sums → block offsets +
+// grand total (replaces the torch cumsum + sub + copy chain between the
+// two scan passes — 3-4 launches → 1)
+__global__ void scan_boffs64_kernel(const int64_t* __restrict__ partials,
+                                    int nb, int64_t* __restrict__ boffs,
+                                    int64_t* __restrict__ total_out) {
+  __shared__ int64_t lds[1024];
+  int tid = threadIdx.x;
+  int64_t v = tid < nb ? partials[tid] : 0;
+  lds[tid] = v;
+  __syncthreads();
+  for (int s = 1; s < 1024; s <<= 1) {
+    int64_t add = tid >= s ? lds[tid - s] : 0;
+    __syncthreads();
+    lds[tid] += add;
+    __syncthreads();
+  }
+  if (tid < nb) boffs[tid] = lds[tid] - v;
+  if (tid == nb - 1) *total_out = lds[tid];
+}
+
+extern "C" void launch_scan_boffs64(const int64_t* partials, int nb,
+                                    int64_t* boffs, int64_t* total_out,
+                                    hipStream_t st) {
+  scan_boffs64_kernel<<<1, 1024, 0, st>>>(partials, nb, boffs, total_out);
+}
+
 extern "C" int scan_grid(int64_t n) {
   return (int)((n + SCAN_TILE - 1) / SCAN_TILE);
 }

@@ -258,7 +258,8 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
                                int64_t* __restrict__ str_start,
                                int32_t* __restrict__ str_ulen,
                                uint8_t* __restrict__ found,
-                               int32_t* __restrict__ err);
+                               int32_t* __restrict__ err,
+                               int* __restrict__ fcnt);
 
 // LDS doc cache: each 128-doc tile's byte range is CONTIGUOUS in the binary
 // column, so the block stages it with coalesced loads and threads then walk
@@ -277,14 +278,17 @@ void json_decode_kernel(const uint8_t* __restrict__ data,
                         int64_t* __restrict__ str_start,  // [ns][n]
                         int32_t* __restrict__ str_ulen,   // [ns][n]
                         uint8_t* __restrict__ found,  // [nf][n]
-                        int32_t* __restrict__ err) {
+                        int32_t* __restrict__ err,
+                        int32_t* __restrict__ found_count) {  // [nf]
   __shared__ uint8_t cache[JSON_LDS_BYTES];
+  __shared__ int fcnt[JSON_MAX_FIELDS];
   for (int64_t tile = (int64_t)blockIdx.x * JSON_BLOCK; tile < n_docs;
        tile += (int64_t)gridDim.x * JSON_BLOCK) {
     int64_t tile_hi = tile + JSON_BLOCK < n_docs ? tile + JSON_BLOCK
                                                  : n_docs;
     const int64_t r0 = offsets[tile];
     const int64_t r1 = offsets[tile_hi];
+    if (threadIdx.x < JSON_MAX_FIELDS) fcnt[threadIdx.x] = 0;
     const uint8_t* dv = data;
     int64_t boff = 0;
     if (r1 - r0 <= JSON_LDS_BYTES) {
@@ -294,16 +298,18 @@ void json_decode_kernel(const uint8_t* __restrict__ data,
         __builtin_memcpy(cache + b, data + r0 + b, 8);
       if (b < len)
         for (; b < len; ++b) cache[b] = data[r0 + b];
-      __syncthreads();
       dv = cache;
       boff = r0;
     }
+    __syncthreads();  // cache staged + fcnt zeroed
     int64_t i = tile + threadIdx.x;
     if (i < tile_hi)
       json_parse_doc(dv, offsets[i] - boff, offsets[i + 1] - boff, boff, i,
                      n_docs, spec, out_f64, out_i64, str_start, str_ulen,
-                     found, err);
+                     found, err, fcnt);
     __syncthreads();  // cache reused by the next tile
+    if (threadIdx.x < JSON_MAX_FIELDS && fcnt[threadIdx.x])
+      atomicAdd(&found_count[threadIdx.x], fcnt[threadIdx.x]);
   }
 }
 
@@ -316,7 +322,8 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
                                int64_t* __restrict__ str_start,
                                int32_t* __restrict__ str_ulen,
                                uint8_t* __restrict__ found,
-                               int32_t* __restrict__ err) {
+                               int32_t* __restrict__ err,
+                               int* __restrict__ fcnt) {
   {
     while (p < end && is_ws(data[p])) ++p;
     if (p >= end || data[p] != '{') {
@@ -395,6 +402,7 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
         str_start[(int64_t)spec.slot[fi] * n_docs + i] = s0 + boff;
         str_ulen[(int64_t)spec.slot[fi] * n_docs + i] = ulen;
         found[(int64_t)fi * n_docs + i] = 1;
+        atomicAdd(&fcnt[fi], 1);
         continue;
       }
       double v = 0.0;
@@ -409,6 +417,7 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
       else
         out_i64[(int64_t)spec.slot[fi] * n_docs + i] = (int64_t)v;
       found[(int64_t)fi * n_docs + i] = 1;
+      atomicAdd(&fcnt[fi], 1);
     }
   }
 }
@@ -437,7 +446,8 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
                                    double* out_f64,
                                    int64_t* out_i64, int64_t* str_start,
                                    int32_t* str_ulen, uint8_t* found,
-                                   int32_t* err, hipStream_t st) {
+                                   int32_t* err, int32_t* found_count,
+                                   hipStream_t st) {
   JsonSpec spec{};
   spec.nf = nf > JSON_MAX_FIELDS ? JSON_MAX_FIELDS : nf;
   for (int f = 0; f < spec.nf; ++f) {
@@ -460,7 +470,8 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
   json_decode_kernel<<<grid, JSON_BLOCK, 0, st>>>(data, offsets, n_docs,
                                                   spec,
                                            out_f64, out_i64, str_start,
-                                           str_ulen, found, err);
+                                           str_ulen, found, err,
+                                           found_count);
 }
 
 // wave-per-doc copy: the thread-per-doc kernel above is divergence-bound at
