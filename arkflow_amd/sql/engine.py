@@ -193,7 +193,9 @@ class SqlExecutor:
             # multi-key sort: stable sorts applied last-key-first
             from .parser import Literal as _Lit
             out_names = list(result.columns)
-            for e, asc in reversed(sel.order_by):
+            for key_spec in reversed(sel.order_by):
+                e, asc = key_spec[0], key_spec[1]
+                nulls_first = key_spec[2] if len(key_spec) > 2 else None
                 if isinstance(e, _Lit) and isinstance(e.value, int) \
                         and not isinstance(e.value, bool) \
                         and 1 <= e.value <= len(out_names):
@@ -216,10 +218,14 @@ class SqlExecutor:
                 from .eval import expr_validity
                 kv = expr_validity(e, final_env)
                 if kv is not None and bool((~kv).any()):
-                    # NULL sorts as smallest (sqlite/standard): stable
-                    # re-sort by validity with the same direction
+                    # default: NULL sorts as smallest (sqlite/standard);
+                    # explicit NULLS FIRST/LAST overrides placement
+                    if nulls_first is None:
+                        null_asc = asc
+                    else:
+                        null_asc = nulls_first  # FIRST ⇔ invalid(0) first
                     idx = idx[ops.sort_indices(
-                        kv[idx].to(torch.int32), ascending=asc)]
+                        kv[idx].to(torch.int32), ascending=null_asc)]
                 result = result.take(idx)
                 # take from the ALREADY-permuted columns: re-taking from the
                 # original env would drop earlier keys' permutations (bug

@@ -144,7 +144,7 @@ _KEYWORDS = {
     "case", "when", "then", "else", "end", "cast", "join", "inner", "left",
     "right", "full", "outer", "on", "asc", "desc", "true", "false",
     "offset", "union", "all", "over", "partition", "row_number", "rank",
-    "dense_rank",
+    "dense_rank", "nulls", "first", "last",
     # rejected verbs (detected for a clear error)
     "insert", "update", "delete", "create", "drop", "alter", "truncate",
 }
@@ -322,7 +322,14 @@ class Parser:
                     asc = False
                 else:
                     self.accept("kw", "asc")
-                sel.order_by.append((e, asc))
+                nulls_first = None  # default: NULL sorts smallest
+                if self.accept("kw", "nulls"):
+                    if self.accept("kw", "first"):
+                        nulls_first = True
+                    else:
+                        self.expect("kw", "last")
+                        nulls_first = False
+                sel.order_by.append((e, asc, nulls_first))
                 if not self.accept("op", ","):
                     break
         if self.accept("kw", "limit"):

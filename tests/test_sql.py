@@ -332,3 +332,36 @@ def test_string_predicate_functions():
     assert r.column("v").to_pylist() == [2]
     r = q("SELECT v FROM flow WHERE ends_with(s, 'sauce')", flow=flow)
     assert r.column("v").to_pylist() == [3]
+
+
+def test_order_by_nulls_first_last():
+    """Explicit NULLS FIRST/LAST placement (default stays NULL-smallest)."""
+    import torch
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+
+    b = MessageBatch.from_dict({"v": [3.0, 1.0, 2.0, 9.0]})
+    b.columns["v"].validity = torch.tensor([True, True, False, True])
+
+    def vals(q):
+        out = SqlExecutor(q).execute({"flow": b})
+        col = out.column("v")
+        vv = col.to_pylist()
+        if col.validity is not None:
+            return [None if not ok else x
+                    for x, ok in zip(vv, col.validity.tolist())]
+        return vv
+
+    assert vals("SELECT v FROM flow ORDER BY v NULLS FIRST") == \
+        [None, 1.0, 3.0, 9.0]
+    assert vals("SELECT v FROM flow ORDER BY v NULLS LAST") == \
+        [1.0, 3.0, 9.0, None]
+    assert vals("SELECT v FROM flow ORDER BY v DESC NULLS FIRST") == \
+        [None, 9.0, 3.0, 1.0]
+    assert vals("SELECT v FROM flow ORDER BY v DESC NULLS LAST") == \
+        [9.0, 3.0, 1.0, None]
+    # default: NULL smallest (ASC → first, DESC → last)
+    assert vals("SELECT v FROM flow ORDER BY v") == [None, 1.0, 3.0, 9.0]
+    assert vals("SELECT v FROM flow ORDER BY v DESC") == \
+        [9.0, 3.0, 1.0, None]
