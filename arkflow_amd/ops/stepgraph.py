@@ -166,9 +166,11 @@ class FusedStepSource:
     retain batches across steps must copy (the ring-buffer contract)."""
 
     def __init__(self, fused: FusedGenerateFilterInfer,
-                 ninstances: int = 1, make_instance=None):
+                 ninstances: int = 1, make_instance=None,
+                 clone: bool = False):
         self.insts = [fused]
         self.streams = [torch.cuda.Stream()]
+        self.clone = clone  # engine mode: queues/buffers retain batches
         for _ in range(max(ninstances, 1) - 1):
             self.insts.append(make_instance() if make_instance else fused)
             self.streams.append(torch.cuda.Stream())
@@ -191,13 +193,18 @@ class FusedStepSource:
             cols = {f: Column("numeric", inst.outs[f][:kept])
                     for f in inst.fields}
             cols["score"] = Column("numeric", inst._scores[:kept])
+            if self.clone:
+                # engine queues / window buffers retain batches past the
+                # next replay — detach from the static graph buffers
+                cols = {k: Column("numeric", c.data.clone())
+                        for k, c in cols.items()}
             return MessageBatch(cols, input_name="generate")
 
     async def read(self):
         from ..spi import NoopAck
         if len(self.insts) == 1:
-            batch, _ = self.insts[0].step()
-            return batch, NoopAck()
+            self._replay(0)
+            return self._consume(0), NoopAck()
         if self._pending is None:  # prime the pipeline
             self._replay(0)
             self._pending = 0

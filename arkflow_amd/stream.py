@@ -352,10 +352,83 @@ def build_stream(config: StreamConfig) -> Stream:
         sid = config.id if afdist.world_size() <= 1 \
             else f"{config.id}.rank{afdist.rank()}"
         wal = Wal.open(config.durability, stream_id=sid)
+
+    # whole-step hipGraph fusion: a generate → sql(simple filter) →
+    # mlp-inference chain on GPU replays as ONE graph per step
+    # (ops/stepgraph.py — the same path bench.py's flagship uses)
+    if wal is None and fusable_chain(config, input_, processors, resource):
+        from .models.mlp import MlpAnomalyDetector
+        from .ops.stepgraph import FusedGenerateFilterInfer, FusedStepSource
+        sqlp, infp = processors
+        col, opi, scalar = sqlp._fast_filter
+        op = {0: "<", 1: "<=", 2: ">", 3: ">=", 4: "==", 5: "!="}[opi]
+        float_fields = [f for f, spec in input_.fields.items()
+                        if str(spec.get("dtype", "float32"))
+                        not in ("int32", "int64")]
+        mlp = MlpAnomalyDetector(len(float_fields), infp._mlp_hidden,
+                                 resource.device, infp.seed)
+        fused = FusedGenerateFilterInfer(
+            input_.fields, input_.batch_size, col, op, scalar, mlp,
+            resource.device, seed=input_.seed)
+        log.info("stream %s: fused generate→filter→mlp into one hipGraph",
+                 config.id)
+        input_ = FusedStepSource(fused, clone=True)
+        processors = []
+
     return Stream(
         config, input_, Pipeline(processors), output, error_output,
         buffer, wal, temporaries,
     )
+
+
+def fusable_chain(config, input_, processors, resource) -> bool:
+    """True when the stream is exactly the fusable hot chain:
+    GPU device; generate input (typed fields, no count/interval, ≤1 int64
+    key, f32 floats); pipeline = [sql simple scalar filter on a float
+    field, mlp_anomaly inference over the float fields with the default
+    score column]. ``fuse: false`` in the stream config opts out."""
+    import torch
+
+    from .inputs.generate import GenerateInput
+    from .processors.inference import InferenceProcessor
+    from .processors.sql import SqlProcessor
+
+    if config.input.get("fuse") is False or \
+            getattr(resource, "device", torch.device("cpu")).type != "cuda":
+        return False
+    if not isinstance(input_, GenerateInput) or not input_.fields:
+        return False
+    if input_.count is not None or input_.interval_secs > 0:
+        return False
+    floats, ints = [], []
+    for f, spec in input_.fields.items():
+        dt = str(spec.get("dtype", "float32"))
+        if dt == "float32":
+            floats.append(f)
+        elif dt == "int64":
+            ints.append(f)
+        else:
+            return False
+    if not floats or len(ints) > 1:
+        return False
+    if len(processors) != 2:
+        return False
+    sqlp, infp = processors
+    if not isinstance(sqlp, SqlProcessor) or sqlp._fast_filter is None:
+        return False
+    if sqlp._fast_filter[0] not in floats:
+        return False
+    if not isinstance(infp, InferenceProcessor) or \
+            infp.model_name not in ("mlp", "mlp_anomaly"):
+        return False
+    if infp.output_column != "score":
+        return False
+    if infp.in_features and infp.in_features != len(floats):
+        return False
+    cols = infp.columns
+    if cols is not None and list(cols) != floats:
+        return False
+    return True
 
 
 def _resolve_device(device: Optional[str]):

@@ -918,3 +918,49 @@ def test_json_inferred_schema_decodes_on_gpu(dev):
     assert out.column("a").data.is_cuda  # decoded on device, stayed there
     assert out.column("a").to_pylist() == list(range(500))
     assert out.column("s").to_strlist()[:3] == ["v0", "v1", "v2"]
+
+
+def test_engine_stream_fuses_hot_chain(dev):
+    """A YAML-shaped generate→sql(filter)→mlp stream on GPU builds with the
+    whole-step hipGraph source and produces filtered, scored batches."""
+    import asyncio
+
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+    from arkflow_amd.ops.stepgraph import FusedStepSource
+    from arkflow_amd.outputs.basic import MemoryOutput
+    from arkflow_amd.stream import build_stream
+
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "fused",
+        "input": {"type": "generate", "batch_size": 4096, "interval": "0ms",
+                  "fields": {
+                      **{f"f{i}": {"dtype": "float32"} for i in range(8)},
+                      "key": {"dtype": "int64", "low": 0, "high": 100}}},
+        "pipeline": {"thread_num": 1, "processors": [
+            {"type": "sql", "query": "SELECT * FROM flow WHERE f0 >= 0.3"},
+            {"type": "inference", "model": "mlp_anomaly",
+             "columns": [f"f{i}" for i in range(8)]},
+        ]},
+        "output": {"type": "memory"},
+    }]})
+    stream = build_stream(cfg.streams[0])
+    assert isinstance(stream.input, FusedStepSource), "fusion did not engage"
+
+    async def run_briefly():
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(stream.run(cancel))
+        await asyncio.sleep(1.0)
+        cancel.set()
+        await asyncio.wait_for(task, 30)
+
+    asyncio.new_event_loop().run_until_complete(run_briefly())
+    out = stream.output
+    assert isinstance(out, MemoryOutput) and out.batches
+    b = out.batches[0]
+    assert "score" in b.columns and "key" in b.columns
+    assert bool((b.column("f0").data >= 0.3).all())
+    assert 0 < b.num_rows <= 4096
+    # cloned outputs: retained batches stay intact across later steps
+    first_f0 = b.column("f0").data.clone()
+    assert torch.equal(first_f0, out.batches[0].column("f0").data)

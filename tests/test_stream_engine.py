@@ -440,3 +440,47 @@ def test_lifecycle_timeout_marks_operation(run):
         assert op.state.value == "timed_out"
 
     run(main(), timeout=30)
+
+
+def test_fusable_chain_detection():
+    """Whole-step graph fusion triggers exactly on the generate→filter→mlp
+    chain (GPU device) and nothing else."""
+    import torch
+
+    from arkflow_amd.inputs.generate import GenerateInput
+    from arkflow_amd.processors.inference import InferenceProcessor
+    from arkflow_amd.processors.sql import SqlProcessor
+    from arkflow_amd.stream import fusable_chain
+
+    class R:
+        device = torch.device("cuda")
+
+    class CfgIn:
+        def __init__(self, **kw):
+            self.input = kw
+
+    fields = {f"f{i}": {"dtype": "float32"} for i in range(4)}
+    fields["k"] = {"dtype": "int64", "low": 0, "high": 10}
+    gen = GenerateInput({"batch_size": 64, "interval": "0ms",
+                         "fields": fields})
+    sql = SqlProcessor({"query": "SELECT * FROM flow WHERE f0 >= 0.5"})
+    mlp = InferenceProcessor({"model": "mlp_anomaly",
+                              "columns": [f"f{i}" for i in range(4)]})
+    cfg = CfgIn()
+    assert fusable_chain(cfg, gen, [sql, mlp], R())
+    # opt-out
+    assert not fusable_chain(CfgIn(fuse=False), gen, [sql, mlp], R())
+    # CPU device → no
+    class RC:
+        device = torch.device("cpu")
+    assert not fusable_chain(cfg, gen, [sql, mlp], RC())
+    # non-trivial sql → no
+    agg = SqlProcessor({"query": "SELECT k, count(*) c FROM flow GROUP BY k"})
+    assert not fusable_chain(cfg, gen, [agg, mlp], R())
+    # count-limited generate → no (EOF semantics differ)
+    gen2 = GenerateInput({"batch_size": 64, "interval": "0ms", "count": 100,
+                          "fields": fields})
+    assert not fusable_chain(cfg, gen2, [sql, mlp], R())
+    # mismatched inference columns → no
+    mlp2 = InferenceProcessor({"model": "mlp_anomaly", "columns": ["f0"]})
+    assert not fusable_chain(cfg, gen, [sql, mlp2], R())
