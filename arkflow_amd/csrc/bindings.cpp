@@ -1016,7 +1016,8 @@ torch::Tensor bytes_hash(torch::Tensor data, torch::Tensor offsets) {
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
-           std::vector<std::tuple<torch::Tensor, torch::Tensor>>>
+           std::vector<std::tuple<torch::Tensor, torch::Tensor>>,
+           torch::Tensor>
 proto_decode(
     torch::Tensor data, torch::Tensor offsets,
     std::vector<int64_t> fno, std::vector<int64_t> kind,
@@ -1051,16 +1052,21 @@ proto_decode(
                         str_start.data_ptr<int64_t>(),
                         str_len.data_ptr<int32_t>(), err.data_ptr<int32_t>(),
                         cur_stream());
-  // string/bytes copy-out (proto3 missing field → len 0 → empty value)
+  // string/bytes copy-out (proto3 missing field → len 0 → empty value);
+  // err + totals ride ONE host readback (JSON decode learned the same —
+  // each extra sync floored small-batch decode)
   std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
+  std::vector<torch::Tensor> sumv = {err.to(torch::kInt64)};
+  std::vector<torch::Tensor> offs(std::max<int64_t>(n_str, 0));
   if (n_str > 0 && n > 0) {
-    std::vector<torch::Tensor> offs(n_str), tot_views;
     for (int64_t s = 0; s < n_str; ++s) {
       offs[s] = exclusive_offsets(str_len[s].contiguous());
-      tot_views.push_back(offs[s].narrow(0, n, 1));
+      sumv.push_back(offs[s].narrow(0, n, 1));
     }
-    auto totals = torch::cat(tot_views).to(torch::kCPU);  // one device sync
-    auto* tot = totals.data_ptr<int64_t>();
+  }
+  auto summary = torch::cat(sumv).cpu();
+  if (n_str > 0 && n > 0) {
+    auto* tot = summary.data_ptr<int64_t>() + 1;
     for (int64_t s = 0; s < n_str; ++s) {
       auto out = torch::empty({std::max<int64_t>(tot[s], 1)},
                               data.options().dtype(torch::kUInt8));
@@ -1072,7 +1078,7 @@ proto_decode(
       strings.emplace_back(out.slice(0, 0, tot[s]), offs[s]);
     }
   }
-  return {out_i, out_f, err, strings};
+  return {out_i, out_f, err, strings, summary};
 }
 
 }  // namespace

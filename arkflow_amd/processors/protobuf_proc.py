@@ -87,10 +87,10 @@ class ProtobufToArrowProcessor(Processor):
                 isf.append(0)
                 slot.append(len(int_fields))
                 int_fields.append(name)
-        out_i, out_f, err, strings = nat.proto_decode(
+        out_i, out_f, err, strings, summary = nat.proto_decode(
             col.data, col.offsets, fno, kind, isf, slot,
             len(int_fields), len(float_fields), len(str_fields))
-        if int(err.item()) != 0:
+        if int(summary[0]) != 0:  # err rode the one consolidated readback
             raise ProcessError("protobuf decode error (malformed message)")
         cols = {}
         for i, name in enumerate(int_fields):
