@@ -367,12 +367,18 @@ def build_stream(config: StreamConfig) -> Stream:
                         not in ("int32", "int64")]
         mlp = MlpAnomalyDetector(len(float_fields), infp._mlp_hidden,
                                  resource.device, infp.seed)
-        fused = FusedGenerateFilterInfer(
-            input_.fields, input_.batch_size, col, op, scalar, mlp,
-            resource.device, seed=input_.seed)
+
+        def mk(seed):
+            return FusedGenerateFilterInfer(
+                input_.fields, input_.batch_size, col, op, scalar, mlp,
+                resource.device, seed=seed)
+
         log.info("stream %s: fused generate→filter→mlp into one hipGraph",
                  config.id)
-        input_ = FusedStepSource(fused, clone=True)
+        seeds = iter(range(1, 16))
+        input_ = FusedStepSource(
+            mk(input_.seed), ninstances=2, clone=True,
+            make_instance=lambda: mk(input_.seed + next(seeds) * 7919))
         processors = []
 
     return Stream(
