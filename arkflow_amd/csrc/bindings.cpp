@@ -835,6 +835,53 @@ void filter_gather_capture(std::vector<torch::Tensor> cols,
   flush();
 }
 
+// one-call `SELECT key, count(*), AGG(col)… FROM flow WHERE fcol OP lit
+// GROUP BY key`: fused filter → key/value gather → LDS hash group →
+// segment reductions, chained in C++ with exactly TWO host syncs (surviving
+// row count, group count). The Python-op version of this chain is
+// host-dispatch-bound at stream batch sizes (BASELINE config 2 @8192).
+// val_ops: 0=sum 1=min 2=max per value column. Returns
+// (unique_keys, group_counts_f32, [reduced…]).
+std::tuple<torch::Tensor, torch::Tensor, std::vector<torch::Tensor>>
+fused_filter_agg(torch::Tensor key, torch::Tensor filter_col, int64_t op,
+                 double scalar, std::vector<torch::Tensor> val_cols,
+                 std::vector<int64_t> val_ops) {
+  check_cuda(key, "key");
+  check_cuda(filter_col, "filter_col");
+  TORCH_CHECK(key.scalar_type() == torch::kInt64, "key must be int64");
+  TORCH_CHECK(filter_col.scalar_type() == torch::kFloat32,
+              "filter col must be f32");
+  TORCH_CHECK(val_cols.size() == val_ops.size());
+  int64_t n = key.numel();
+  auto st = cur_stream();
+  auto opts32 = key.options().dtype(torch::kInt32);
+  std::vector<torch::Tensor> cols = {key, filter_col};
+  for (auto& v : val_cols) {
+    check_cuda(v, "val col");
+    TORCH_CHECK(v.scalar_type() == torch::kFloat32, "val cols must be f32");
+    cols.push_back(v);
+  }
+  auto [outs, total] = fused_filter_gather(cols, 1, op, scalar);
+  auto gkey = outs[0];
+  if (total == 0) {
+    std::vector<torch::Tensor> empt;
+    for (size_t i = 0; i < val_cols.size(); ++i)
+      empt.push_back(torch::empty({0}, filter_col.options()));
+    return {torch::empty({0}, key.options()),
+            torch::empty({0}, filter_col.options()), empt};
+  }
+  auto [gids, uniq] = hash_group_i64(gkey);
+  int64_t g = uniq.numel();
+  // counts via the f32 LDS-buffered segment sum (exact < 2^24 per group)
+  auto ones = torch::ones({total}, filter_col.options());
+  auto counts = segment_reduce_f32(ones, gids, g, 0);
+  std::vector<torch::Tensor> reduced;
+  for (size_t i = 0; i < val_cols.size(); ++i)
+    reduced.push_back(segment_reduce_f32(outs[2 + i], gids, g,
+                                         val_ops[i]));
+  return {uniq, counts, reduced};
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
            std::vector<std::tuple<torch::Tensor, torch::Tensor>>,
            torch::Tensor>
@@ -1111,6 +1158,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bytes_hash", &bytes_hash);
   m.def("json_decode", &json_decode);
   m.def("fused_filter_gather", &fused_filter_gather);
+  m.def("fused_filter_agg", &fused_filter_agg);
   m.def("filter_gather_capture", &filter_gather_capture);
   m.def("gemv_bf16_f32", &gemv_bf16_f32);
   m.def("gen_fields", &gen_fields);

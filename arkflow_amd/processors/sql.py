@@ -38,6 +38,9 @@ class SqlProcessor(Processor):
         }
         # native one-call fast path for `SELECT * FROM flow WHERE col OP lit`
         self._fast_filter = self._detect_fast_filter()
+        # native one-call fused filter→group→agg for
+        # `SELECT key, count(*)/sum/min/max(col)… WHERE col OP lit GROUP BY key`
+        self._fast_agg = self._detect_fast_agg()
 
     def _detect_fast_filter(self):
         from ..sql.parser import BinaryOp, ColumnRef, Literal, Star
@@ -58,6 +61,111 @@ class SqlProcessor(Processor):
             opi = {"<": 0, "<=": 1, ">": 2, ">=": 3, "=": 4, "!=": 5}[w.op]
             return (w.left.name, opi, float(w.right.value))
         return None
+
+    def _detect_fast_agg(self):
+        """`SELECT key, AGG…` with a simple scalar WHERE and a single int
+        GROUP BY key → ONE native call (fused_filter_agg: filter + gather +
+        LDS hash group + segment reductions, two host syncs). The Python-op
+        chain for this shape is host-dispatch-bound at batch 8192
+        (BASELINE config 2)."""
+        from ..sql.parser import BinaryOp, ColumnRef, FuncCall, Literal
+        s = self.executor.select
+        if (s.joins or s.having or s.order_by or s.distinct
+                or s.limit is not None or s.offset or s.union_all
+                or self.temporary_specs):
+            return None
+        if len(s.group_by) != 1 or not isinstance(s.group_by[0], ColumnRef):
+            return None
+        key = s.group_by[0].name
+        # WHERE col OP literal (optional: no WHERE → keep all rows)
+        filt = None
+        if s.where is not None:
+            w = s.where
+            if not (isinstance(w, BinaryOp)
+                    and w.op in ("<", "<=", ">", ">=", "=", "!=")
+                    and isinstance(w.left, ColumnRef)
+                    and isinstance(w.right, Literal)
+                    and isinstance(w.right.value, (int, float))):
+                return None
+            opi = {"<": 0, "<=": 1, ">": 2, ">=": 3, "=": 4, "!=": 5}[w.op]
+            filt = (w.left.name, opi, float(w.right.value))
+        # projections: the key + count(*)/sum/min/max/avg(plain column)
+        plan = []  # (kind, payload, alias)
+        for e, alias in s.projections:
+            if isinstance(e, ColumnRef) and e.name == key:
+                plan.append(("key", None, alias or key))
+                continue
+            if isinstance(e, FuncCall) and not e.over and not e.distinct:
+                fn = e.name
+                if fn == "count" and (not e.args or
+                                      e.args[0].__class__.__name__ == "Star"):
+                    plan.append(("count", None, alias or "count"))
+                    continue
+                if fn in ("sum", "min", "max", "avg") and len(e.args) == 1 \
+                        and isinstance(e.args[0], ColumnRef):
+                    plan.append((fn, e.args[0].name, alias or fn))
+                    continue
+            return None
+        if not any(k in ("count", "sum", "min", "max", "avg")
+                   for k, _, _ in plan):
+            return None
+        return (key, filt, plan)
+
+    def _try_fast_agg(self, batch: MessageBatch):
+        if self._fast_agg is None or batch.device.type != "cuda":
+            return None
+        key, filt, plan = self._fast_agg
+        kc = batch.columns.get(key)
+        if kc is None or kc.kind != "numeric" or kc.validity is not None \
+                or kc.data.dtype != torch.int64:
+            return None
+        if filt is None:
+            # no WHERE: synthesize an always-true filter on the first f32 col
+            fname, opi, scalar = None, 5, float("nan")  # x != nan ⇒ all
+        else:
+            fname, opi, scalar = filt
+        val_names = sorted({p for k, p, _ in plan
+                            if k in ("sum", "min", "max", "avg")})
+        tensors = {}
+        for n in ([fname] if fname else []) + val_names:
+            c = batch.columns.get(n)
+            if c is None or c.kind != "numeric" or c.validity is not None \
+                    or c.data.dtype != torch.float32:
+                return None
+            tensors[n] = c.data
+        if fname is None:
+            if not val_names:
+                return None
+            fname = val_names[0]
+            tensors.setdefault(fname, batch.column(fname).data)
+        ops_code = {"sum": 0, "avg": 0, "min": 1, "max": 2}
+        val_list = []
+        for n in val_names:
+            # one reduction per (col, op) pair used by the plan
+            for k, p, _ in plan:
+                if p == n and k in ("sum", "min", "max", "avg"):
+                    val_list.append((n, "sum" if k == "avg" else k))
+        val_list = list(dict.fromkeys(val_list))
+        from .. import ops as _ops
+        nat = _ops.require_native()
+        uniq, counts, reduced = nat.fused_filter_agg(
+            kc.data, tensors[fname], opi, scalar,
+            [tensors[n] for n, _ in val_list],
+            [ops_code[o] for _, o in val_list])
+        rmap = {pair: t for pair, t in zip(val_list, reduced)}
+        from ..batch import Column
+        cols = {}
+        for k, p, alias in plan:
+            if k == "key":
+                cols[alias] = Column("numeric", uniq)
+            elif k == "count":
+                cols[alias] = Column("numeric", counts.to(torch.int64))
+            elif k == "avg":
+                cols[alias] = Column(
+                    "numeric", rmap[(p, "sum")].double() / counts.double())
+            else:
+                cols[alias] = Column("numeric", rmap[(p, k)])
+        return MessageBatch(cols, batch.input_name)
 
     def _try_fast_filter(self, batch: MessageBatch):
         """One C++ call: fused compare+compact+multi-gather (all columns
@@ -89,6 +197,9 @@ class SqlProcessor(Processor):
         if batch.num_rows == 0:
             return []  # ProcessResult::None (sql.rs:208-210)
         fast = self._try_fast_filter(batch)
+        if fast is not None:
+            return [fast] if fast.num_rows else []
+        fast = self._try_fast_agg(batch)
         if fast is not None:
             return [fast] if fast.num_rows else []
         tables = {self.table_name: batch}

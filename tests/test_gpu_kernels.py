@@ -964,3 +964,47 @@ def test_engine_stream_fuses_hot_chain(dev):
     # cloned outputs: retained batches stay intact across later steps
     first_f0 = b.column("f0").data.clone()
     assert torch.equal(first_f0, out.batches[0].column("f0").data)
+
+
+def test_fast_agg_matches_engine_path(dev):
+    """fused_filter_agg one-call path vs the full SQL engine on the same
+    query (BASELINE config 2 shape)."""
+    import asyncio
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.sql import SqlProcessor
+    from arkflow_amd.sql.engine import SqlExecutor
+
+    torch.manual_seed(3)
+    n = 50_000
+    batch = MessageBatch.from_dict({
+        "key": torch.randint(0, 257, (n,), dtype=torch.int64),
+        "f0": torch.rand(n),
+        "f1": torch.rand(n) * 10,
+    }).to(dev)
+    q = ("SELECT key, count(*) AS c, sum(f0) AS s, avg(f1) AS a, "
+         "max(f1) AS mx FROM flow WHERE f0 >= 0.2 GROUP BY key")
+    proc = SqlProcessor({"query": q})
+    assert proc._fast_agg is not None
+    loop = asyncio.new_event_loop()
+    fast = loop.run_until_complete(proc.process(batch))[0]
+    slow = SqlExecutor(q).execute({"flow": batch})
+    assert fast.num_rows == slow.num_rows == 257
+
+    def by_key(b):
+        order = torch.argsort(b.column("key").data)
+        return {name: b.column(name).data[order].cpu()
+                for name in b.column_names}
+
+    f, s = by_key(fast), by_key(slow)
+    assert torch.equal(f["key"], s["key"])
+    assert torch.equal(f["c"].long(), s["c"].long())
+    assert torch.allclose(f["s"].float(), s["s"].float(), rtol=1e-3)
+    assert torch.allclose(f["a"].float(), s["a"].float(), rtol=1e-3)
+    assert torch.allclose(f["mx"].float(), s["mx"].float(), rtol=1e-4)
+    # no-WHERE variant also fuses
+    q2 = "SELECT key, count(*) AS c FROM flow GROUP BY key"
+    p2 = SqlProcessor({"query": q2})
+    assert p2._fast_agg is not None
+    f2 = loop.run_until_complete(p2.process(batch))[0]
+    assert int(f2.column("c").data.sum()) == n
