@@ -302,11 +302,30 @@ class _PubSubInput(Input):
         self._q: Optional[asyncio.Queue] = None
         self._closed = False
 
+    def _make_real(self):
+        """Real client for this kind, or raise ConnectionError_.
+        redis → redis-py; mqtt → paho-mqtt; nats/pulsar have no real driver
+        (fake bus only — their client libraries are not modeled)."""
+        from urllib.parse import urlparse
+
+        from .pubsub_real import RealMqttClient, RealRedisClient
+        if self.kind == "redis":
+            return RealRedisClient(self.url, getattr(self, "mode", "pubsub"))
+        if self.kind == "mqtt":
+            u = urlparse(self.url if "://" in self.url
+                         else f"mqtt://{self.url}")
+            return RealMqttClient(u.hostname or "127.0.0.1",
+                                  u.port or 1883,
+                                  username=u.username, password=u.password)
+        raise ConnectionError_(
+            f"no {self.kind} client library in this environment; "
+            "use driver: memory")
+
     async def connect(self) -> None:
         if self.driver != "memory":
-            raise ConnectionError_(
-                f"no {self.kind} client library in this environment; "
-                "use driver: memory")
+            self._real = self._make_real()
+            self._real.connect(subscribe=self.topic)
+            return
         self.bus = FakeBus.get(self.url.removeprefix("memory://") or "default")
         self._q = asyncio.Queue(maxsize=4096)
         self.bus.subscribers[self.topic].append(self._q)
@@ -314,6 +333,13 @@ class _PubSubInput(Input):
     async def read(self) -> Tuple[MessageBatch, Ack]:
         if self._closed:
             raise EOFError_(f"{self.kind} input closed")
+        if getattr(self, "_real", None) is not None:
+            from .pubsub_real import make_batch
+            if self.kind == "redis":
+                value, off = await self._real.read(self.topic)
+                return make_batch(self.topic, value, self.codec, off)
+            topic, value = await self._real.read()
+            return make_batch(topic, value, self.codec)
         item = await self._q.get()
         if item is None:
             raise EOFError_(f"{self.kind} input closed")
@@ -333,6 +359,8 @@ class _PubSubInput(Input):
 
     async def close(self) -> None:
         self._closed = True
+        if getattr(self, "_real", None) is not None:
+            self._real.close()
         if self._q is not None:
             self._q.put_nowait(None)
             if self.bus and self._q in self.bus.subscribers.get(self.topic, []):
@@ -351,13 +379,14 @@ class _PubSubOutput(Output):
         self.driver = config.get("driver") or (
             "memory" if self.url.startswith("memory://") else "real")
         self.raw_value = bool(config.get("raw_value", True))
+        self.mode = config.get("mode", "pubsub")  # redis: pubsub|list|stream
         self.bus: Optional[FakeBus] = None
 
     async def connect(self) -> None:
         if self.driver != "memory":
-            raise ConnectionError_(
-                f"no {self.kind} client library in this environment; "
-                "use driver: memory")
+            self._real = _PubSubInput._make_real(self)
+            self._real.connect()
+            return
         self.bus = FakeBus.get(self.url.removeprefix("memory://") or "default")
         self.bus.ensure_topic(self.topic)
 
@@ -367,8 +396,18 @@ class _PubSubOutput(Output):
             values = batch.binary_values()
         else:
             values = batch.to_json_lines()
+        if getattr(self, "_real", None) is not None:
+            loop = asyncio.get_running_loop()
+            for v in values:
+                await loop.run_in_executor(
+                    None, self._real.produce, self.topic, v)
+            return
         for v in values:
             self.bus.produce(self.topic, None, v)
+
+    async def close(self) -> None:
+        if getattr(self, "_real", None) is not None:
+            self._real.close()
 
 
 def _mk_pubsub(kind: str):

@@ -1,0 +1,186 @@
+"""Real redis / MQTT drivers behind the pub-sub family's driver switch.
+
+Like inputs/kafka_real.py: complete client implementations against the
+public redis-py and paho-mqtt APIs, active when the library is importable
+and the config names a real endpoint; the fake bus keeps carrying the
+semantics offline. Env-gated tests: REDIS_URL / MQTT_HOST
+(tests/test_brokers.py), the way the reference gates its testcontainers
+suites.
+
+Reference mapping:
+  redis (input/redis.rs): pubsub / list (BRPOP) / stream (XREAD) modes;
+  output publishes / RPUSHes / XADDs.
+  mqtt (input/mqtt.rs, rumqttc): subscriber with QoS, publisher.
+"""
+from __future__ import annotations
+
+import asyncio
+import queue as _queue
+import time
+from typing import Optional, Tuple
+
+from ..batch import Column, MessageBatch
+from ..errors import ConnectionError_, EOFError_
+from ..spi import Ack, NoopAck
+
+
+# ------------------------------------------------------------------- redis
+class RealRedisClient:
+    """redis-py driver: pubsub / list / stream consume + produce."""
+
+    def __init__(self, url: str, mode: str = "pubsub"):
+        self.url = url
+        self.mode = mode
+        self.client = None
+        self._pubsub = None
+        self._stream_last = "$"
+        self._closed = False
+
+    def connect(self, subscribe: Optional[str] = None) -> None:
+        try:
+            import redis  # type: ignore
+        except ImportError as e:
+            raise ConnectionError_(
+                "redis real driver requires redis-py; use driver: memory"
+            ) from e
+        try:
+            self.client = redis.Redis.from_url(self.url)
+            self.client.ping()
+            if subscribe is not None and self.mode == "pubsub":
+                self._pubsub = self.client.pubsub(
+                    ignore_subscribe_messages=True)
+                self._pubsub.subscribe(subscribe)
+        except Exception as e:  # noqa: BLE001
+            raise ConnectionError_(f"redis connect failed: {e}") from e
+
+    async def read(self, topic: str) -> Tuple[bytes, Optional[int]]:
+        """Returns (payload, offset-or-None); blocks via executor."""
+        loop = asyncio.get_running_loop()
+        while not self._closed:
+            if self.mode == "pubsub":
+                msg = await loop.run_in_executor(
+                    None, lambda: self._pubsub.get_message(timeout=0.25))
+                if msg and msg.get("type") == "message":
+                    return msg["data"], None
+            elif self.mode == "list":
+                item = await loop.run_in_executor(
+                    None, lambda: self.client.blpop(topic, timeout=1))
+                if item is not None:
+                    return item[1], None
+            else:  # stream
+                got = await loop.run_in_executor(
+                    None, lambda: self.client.xread(
+                        {topic: self._stream_last}, count=1, block=250))
+                if got:
+                    _, entries = got[0]
+                    eid, fields = entries[0]
+                    self._stream_last = eid
+                    payload = fields.get(b"value") or next(
+                        iter(fields.values()), b"")
+                    return payload, None
+        raise EOFError_("redis input closed")
+
+    def produce(self, topic: str, payload: bytes) -> None:
+        if self.mode == "pubsub":
+            self.client.publish(topic, payload)
+        elif self.mode == "list":
+            self.client.rpush(topic, payload)
+        else:
+            self.client.xadd(topic, {"value": payload})
+
+    def close(self) -> None:
+        self._closed = True
+        try:
+            if self._pubsub is not None:
+                self._pubsub.close()
+            if self.client is not None:
+                self.client.close()
+        except Exception:  # noqa: BLE001
+            pass
+
+
+# -------------------------------------------------------------------- mqtt
+class RealMqttClient:
+    """paho-mqtt driver: background network loop feeding a local queue."""
+
+    def __init__(self, host: str, port: int = 1883, qos: int = 1,
+                 client_id: str = "", username: Optional[str] = None,
+                 password: Optional[str] = None):
+        self.host, self.port, self.qos = host, port, qos
+        self.client_id = client_id
+        self.username, self.password = username, password
+        self.client = None
+        self._q: _queue.Queue = _queue.Queue(maxsize=4096)
+        self._closed = False
+
+    def connect(self, subscribe: Optional[str] = None) -> None:
+        try:
+            import paho.mqtt.client as mqtt  # type: ignore
+        except ImportError as e:
+            raise ConnectionError_(
+                "mqtt real driver requires paho-mqtt; use driver: memory"
+            ) from e
+        try:
+            try:  # paho 2.x requires an API version; 1.x has no such arg
+                self.client = mqtt.Client(
+                    mqtt.CallbackAPIVersion.VERSION1,
+                    client_id=self.client_id)
+            except AttributeError:
+                self.client = mqtt.Client(client_id=self.client_id)
+            if self.username:
+                self.client.username_pw_set(self.username, self.password)
+
+            def on_message(cli, userdata, msg):
+                try:
+                    self._q.put_nowait((msg.topic, msg.payload))
+                except _queue.Full:
+                    pass
+
+            self.client.on_message = on_message
+            self.client.connect(self.host, self.port, keepalive=30)
+            if subscribe is not None:
+                self.client.subscribe(subscribe, qos=self.qos)
+            self.client.loop_start()
+        except ConnectionError_:
+            raise
+        except Exception as e:  # noqa: BLE001
+            raise ConnectionError_(f"mqtt connect failed: {e}") from e
+
+    async def read(self) -> Tuple[str, bytes]:
+        loop = asyncio.get_running_loop()
+        while not self._closed:
+            try:
+                return await loop.run_in_executor(
+                    None, lambda: self._q.get(timeout=0.25))
+            except _queue.Empty:
+                continue
+        raise EOFError_("mqtt input closed")
+
+    def produce(self, topic: str, payload: bytes) -> None:
+        info = self.client.publish(topic, payload, qos=self.qos)
+        info.wait_for_publish(timeout=10.0)
+
+    def close(self) -> None:
+        self._closed = True
+        if self.client is not None:
+            try:
+                self.client.loop_stop()
+                self.client.disconnect()
+            except Exception:  # noqa: BLE001
+                pass
+
+
+def make_batch(topic: str, payload: bytes, codec=None,
+               offset: Optional[int] = None) -> Tuple[MessageBatch, Ack]:
+    cols = {
+        "__value__": Column.from_bytes([payload]),
+        "__meta_source": Column.from_strings([topic]),
+        "__meta_timestamp": Column.from_numeric([time.time()]),
+    }
+    if offset is not None:
+        cols["__meta_offset"] = Column.from_numeric([offset])
+    batch = MessageBatch(cols, input_name=topic)
+    if codec is not None:
+        from ..codecs.helper import apply_codec
+        batch = apply_codec(batch, codec)
+    return batch, NoopAck()

@@ -438,3 +438,36 @@ def test_mongodb_output_memory_driver(run):
     assert len(docs) == 2
     assert docs[0]["a"] == 1 and docs[0]["s"] in ("x", b"x")
     _FakeMongoStore.reset()
+
+
+@pytest.mark.skipif(not __import__("os").environ.get("REDIS_URL"),
+                    reason="REDIS_URL not set")
+@pytest.mark.timeout(60)
+def test_redis_real_driver_roundtrip(run):
+    """redis-py driver: list-mode produce/consume against a real server
+    (env-gated like the reference's testcontainers suites)."""
+    import os
+    import uuid
+
+    from arkflow_amd.inputs.brokers import RedisInput, RedisOutput
+    topic = f"l-{uuid.uuid4().hex[:8]}"
+    url = os.environ["REDIS_URL"]
+
+    async def main():
+        out = RedisOutput({"url": url, "topic": topic, "mode": "list",
+                           "driver": "real"})
+        await out.connect()
+        from arkflow_amd.batch import MessageBatch
+        await out.write(MessageBatch.from_binary([b"r1", b"r2"]))
+        inp = RedisInput({"url": url, "topic": topic, "mode": "list",
+                          "driver": "real"})
+        await inp.connect()
+        got = []
+        for _ in range(2):
+            b, _a = await inp.read()
+            got.extend(b.binary_values())
+        await inp.close()
+        await out.close()
+        assert got == [b"r1", b"r2"]
+
+    run(main(), timeout=50)
