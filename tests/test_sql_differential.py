@@ -32,6 +32,9 @@ QUERIES = [
     "FROM flow ORDER BY a, rn",
     "SELECT a, rank() OVER (ORDER BY k) AS rk FROM flow ORDER BY a, rk",
     "SELECT a, sum(a) OVER (PARTITION BY k) AS s FROM flow ORDER BY a, s",
+    "SELECT k, count(*) AS c, sum(b) AS s FROM flow WHERE b >= 0.2 "
+    "GROUP BY k ORDER BY k",
+    "SELECT a FROM flow ORDER BY a DESC NULLS LAST LIMIT 10",
     "SELECT a FROM flow WHERE a < 10 UNION ALL SELECT a FROM flow "
     "WHERE a > 90",
     "SELECT k, sum(a) AS s FROM flow GROUP BY k "
@@ -117,7 +120,7 @@ def _normalize(rows):
     return out
 
 
-@pytest.mark.parametrize("seed", [0, 1, 2, 3])
+@pytest.mark.parametrize("seed", [0, 1, 2, 3, 4, 5])
 @pytest.mark.parametrize("sql", QUERIES)
 def test_differential_vs_sqlite(seed, sql):
     rng = random.Random(seed * 1000 + zlib.crc32(sql.encode()) % 997)
@@ -178,7 +181,7 @@ NULL_QUERIES = [
 ]
 
 
-@pytest.mark.parametrize("seed", [0, 1, 2])
+@pytest.mark.parametrize("seed", [0, 1, 2, 3])
 @pytest.mark.parametrize("sql", NULL_QUERIES)
 def test_differential_nulls_vs_sqlite(seed, sql):
     rng = random.Random(seed * 77 + zlib.crc32(sql.encode()) % 991)
