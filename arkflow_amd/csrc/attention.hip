@@ -18,15 +18,15 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // [BH, S, D] layout; with Hd=H/ld=3*H*D the kernel reads q/k/v DIRECTLY out
 // of the [B,S,3,H,D] QKV-projection tensor (no transpose copies), and with
 // ldo=H*D writes O back in [B,S,H*D] so the next linear consumes it as-is.
-template <int S, int D>
+template <int S, int D, int PAD = 4>
 __global__ __launch_bounds__(ATTN_THREADS, 1)
 void attention_kernel(const __bf16* __restrict__ Q,
                       const __bf16* __restrict__ K,
                       const __bf16* __restrict__ V,
                       __bf16* __restrict__ O, float scale,
                       int ldq, int ldo, int Hd) {
-  constexpr int DP = D + 4;   // padded row strides (bank spread)
-  constexpr int SP = S + 4;
+  constexpr int DP = D + PAD;  // padded row strides (bank spread)
+  constexpr int SP = S + PAD;
   constexpr int QROWS = 32;   // q rows per wave
   constexpr int NWAVE = ATTN_THREADS / WAVE;
   static_assert(S == NWAVE * QROWS, "one block covers all S rows");
@@ -182,27 +182,44 @@ int launch_attention_bf16(const void* Q, const void* K, const void* V,
 
 // qkv: [B, S, 3, H, D] contiguous (the QKV linear's natural output);
 // O: [B, S, H*D]. No transpose copies on either side.
+// pad: LDS row-padding in bf16 elements (ARKFLOW_ATTN_PAD sweep — NOTES r2:
+// the +4 layout still measured ~590K bank conflicts/dispatch).
+template <int PAD>
+static int qkv_launch_pad(const void* QKV, void* O, int B, int H, int S,
+                          int D, float scale, hipStream_t st) {
+  constexpr int SS = 128, DD = 64;
+  size_t lds = (SS * (DD + PAD) + DD * (SS + PAD) + 4 * 32 * (SS + PAD)) *
+               sizeof(__bf16);
+  static bool attr_set2 = false;
+  if (!attr_set2) {
+    hipFuncSetAttribute(
+        (const void*)attention_kernel<SS, DD, PAD>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    attr_set2 = true;
+  }
+  const __bf16* base = (const __bf16*)QKV;
+  int ld = 3 * H * D;
+  attention_kernel<SS, DD, PAD><<<B * H, ATTN_THREADS, lds, st>>>(
+      base, base + (int64_t)H * D, base + (int64_t)2 * H * D, (__bf16*)O,
+      scale, ld, H * D, H);
+  return 0;
+}
+
+int launch_attention_qkv_bf16_pad(const void* QKV, void* O, int B, int H,
+                                  int S, int D, float scale, int pad,
+                                  hipStream_t st) {
+  if (!(S == 128 && D == 64)) return -1;
+  switch (pad) {
+    case 8: return qkv_launch_pad<8>(QKV, O, B, H, S, D, scale, st);
+    case 16: return qkv_launch_pad<16>(QKV, O, B, H, S, D, scale, st);
+    case 20: return qkv_launch_pad<20>(QKV, O, B, H, S, D, scale, st);
+    default: return qkv_launch_pad<4>(QKV, O, B, H, S, D, scale, st);
+  }
+}
+
 int launch_attention_qkv_bf16(const void* QKV, void* O, int B, int H, int S,
                               int D, float scale, hipStream_t st) {
-  if (S == 128 && D == 64) {
-    constexpr int SS = 128, DD = 64;
-    size_t lds = (SS * (DD + 4) + DD * (SS + 4) + 4 * 32 * (SS + 4)) *
-                 sizeof(__bf16);
-    static bool attr_set2 = false;
-    if (!attr_set2) {
-      hipFuncSetAttribute(
-          (const void*)attention_kernel<SS, DD>,
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-      attr_set2 = true;
-    }
-    const __bf16* base = (const __bf16*)QKV;
-    int ld = 3 * H * D;
-    attention_kernel<SS, DD><<<B * H, ATTN_THREADS, lds, st>>>(
-        base, base + (int64_t)H * D, base + (int64_t)2 * H * D, (__bf16*)O,
-        scale, ld, H * D, H);
-    return 0;
-  }
-  return -1;
+  return launch_attention_qkv_bf16_pad(QKV, O, B, H, S, D, scale, 4, st);
 }
 
 }  // extern "C"

@@ -56,6 +56,8 @@ int launch_attention_bf16(const void*, const void*, const void*, void*, int,
                           int, int, float, hipStream_t);
 int launch_attention_qkv_bf16(const void*, void*, int, int, int, int, float,
                               hipStream_t);
+int launch_attention_qkv_bf16_pad(const void*, void*, int, int, int, int,
+                                  float, int, hipStream_t);
 void launch_proto_copy_bytes(const uint8_t*, const int64_t*, const int64_t*,
                              int64_t, uint8_t*, hipStream_t);
 void launch_proto_decode(const uint8_t*, const int64_t*, int64_t, int,
@@ -583,15 +585,17 @@ torch::Tensor attention_bf16(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 
 // qkv [B,S,3,H,D] (the QKV linear's output reshaped) → O [B,S,H*D];
 // the kernel reads strided rows directly — no permute/contiguous copies.
-torch::Tensor attention_qkv_bf16(torch::Tensor qkv, double scale) {
+torch::Tensor attention_qkv_bf16(torch::Tensor qkv, double scale,
+                                 int64_t pad) {
   check_cuda(qkv, "qkv");
   TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16, "qkv must be bf16");
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3, "qkv must be [B,S,3,H,D]");
   int64_t B = qkv.size(0), S = qkv.size(1), H = qkv.size(3), D = qkv.size(4);
   auto out = torch::empty({B, S, H * D}, qkv.options());
-  int rc = launch_attention_qkv_bf16(qkv.data_ptr(), out.data_ptr(), (int)B,
-                                     (int)H, (int)S, (int)D, (float)scale,
-                                     cur_stream());
+  int rc = launch_attention_qkv_bf16_pad(qkv.data_ptr(), out.data_ptr(),
+                                         (int)B, (int)H, (int)S, (int)D,
+                                         (float)scale, (int)pad,
+                                         cur_stream());
   TORCH_CHECK(rc == 0, "attention_qkv_bf16: unsupported shape S=", S,
               " D=", D);
   return out;
@@ -1135,7 +1139,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("exclusive_offsets", &exclusive_offsets);
   m.def("take_binary", &take_binary);
   m.def("bytes_match", &bytes_match);
-  m.def("attention_qkv_bf16", &attention_qkv_bf16);
+  m.def("attention_qkv_bf16", &attention_qkv_bf16, py::arg("qkv"),
+        py::arg("scale"), py::arg("pad") = 4);
   m.def("mask_to_indices", &mask_to_indices);
   m.def("filter_cmp_scalar", &filter_cmp_scalar);
   m.def("gather", &gather);
