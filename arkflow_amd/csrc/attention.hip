@@ -182,8 +182,10 @@ int launch_attention_bf16(const void* Q, const void* K, const void* V,
 
 // qkv: [B, S, 3, H, D] contiguous (the QKV linear's natural output);
 // O: [B, S, H*D]. No transpose copies on either side.
-// pad: LDS row-padding in bf16 elements (ARKFLOW_ATTN_PAD sweep — NOTES r2:
-// the +4 layout still measured ~590K bank conflicts/dispatch).
+// pad: LDS row-padding in bf16 elements (pad sweep — NOTES r2: the +4
+// layout still measured ~590K bank conflicts/dispatch).
+}  // extern "C" (template below needs C++ linkage)
+
 template <int PAD>
 static int qkv_launch_pad(const void* QKV, void* O, int B, int H, int S,
                           int D, float scale, hipStream_t st) {
@@ -205,9 +207,10 @@ static int qkv_launch_pad(const void* QKV, void* O, int B, int H, int S,
   return 0;
 }
 
-int launch_attention_qkv_bf16_pad(const void* QKV, void* O, int B, int H,
-                                  int S, int D, float scale, int pad,
-                                  hipStream_t st) {
+extern "C" int launch_attention_qkv_bf16_pad(const void* QKV, void* O,
+                                             int B, int H, int S, int D,
+                                             float scale, int pad,
+                                             hipStream_t st) {
   if (!(S == 128 && D == 64)) return -1;
   switch (pad) {
     case 8: return qkv_launch_pad<8>(QKV, O, B, H, S, D, scale, st);
@@ -217,9 +220,8 @@ int launch_attention_qkv_bf16_pad(const void* QKV, void* O, int B, int H,
   }
 }
 
-int launch_attention_qkv_bf16(const void* QKV, void* O, int B, int H, int S,
-                              int D, float scale, hipStream_t st) {
+extern "C" int launch_attention_qkv_bf16(const void* QKV, void* O, int B,
+                                         int H, int S, int D, float scale,
+                                         hipStream_t st) {
   return launch_attention_qkv_bf16_pad(QKV, O, B, H, S, D, scale, 4, st);
 }
-
-}  // extern "C"
