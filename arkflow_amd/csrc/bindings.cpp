@@ -1140,7 +1140,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("take_binary", &take_binary);
   m.def("bytes_match", &bytes_match);
   m.def("attention_qkv_bf16", &attention_qkv_bf16, py::arg("qkv"),
-        py::arg("scale"), py::arg("pad") = 4);
+        py::arg("scale"), py::arg("pad") = 8);
   m.def("mask_to_indices", &mask_to_indices);
   m.def("filter_cmp_scalar", &filter_cmp_scalar);
   m.def("gather", &gather);
