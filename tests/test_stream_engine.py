@@ -484,3 +484,33 @@ def test_fusable_chain_detection():
     # mismatched inference columns → no
     mlp2 = InferenceProcessor({"model": "mlp_anomaly", "columns": ["f0"]})
     assert not fusable_chain(cfg, gen, [sql, mlp2], R())
+
+
+def test_direct_mode_passthrough(run):
+    """Single-stage streams (empty pipeline, no buffer/WAL) take the direct
+    read→write→ack loop; semantics (EOF, metrics, acks) match the graph."""
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+
+    class _Cap:
+        pass
+
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "direct",
+        "input": {"type": "generate", "batch_size": 8, "interval": "0ms",
+                  "count": 64, "fields": {"v": {"dtype": "float32"}}},
+        "output": {"type": "memory"},
+    }]})
+    from arkflow_amd.stream import build_stream
+    stream = build_stream(cfg.streams[0])
+    assert not stream.pipeline.processors
+
+    async def main():
+        import asyncio
+        cancel = asyncio.Event()
+        await asyncio.wait_for(stream.run(cancel), 30)
+
+    run(main())
+    assert stream.metrics.input_messages == 64
+    assert stream.metrics.output_messages == 64
+    assert sum(b.num_rows for b in stream.output.batches) == 64
