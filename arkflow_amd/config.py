@@ -72,6 +72,7 @@ class ServerConfig:
     hub_url: Optional[str] = None
     node_id: Optional[str] = None
     node_token: Optional[str] = None
+    node_labels: dict = field(default_factory=dict)
     lease_ttl_secs: float = 15.0
     config_store: Optional[str] = None  # persist config versions for rollback
 
@@ -147,6 +148,7 @@ class EngineConfig:
             hub_url=srv_raw.get("hub_url"),
             node_id=srv_raw.get("node_id"),
             node_token=srv_raw.get("node_token"),
+            node_labels=srv_raw.get("node_labels") or {},
             lease_ttl_secs=float(srv_raw.get("lease_ttl_secs", 15.0)),
             config_store=srv_raw.get("config_store"),
         )

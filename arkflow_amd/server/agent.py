@@ -24,11 +24,13 @@ class Agent:
                  report_interval: float = 5.0,
                  poll_interval: float = 1.0,
                  registration_token: Optional[str] = None,
+                 labels: Optional[dict] = None,
                  transport: Optional[httpx.AsyncBaseTransport] = None):
         self.engine = engine
         self.hub_url = hub_url.rstrip("/")
         self.node_id = node_id or socket.gethostname()
         self.registration_token = registration_token
+        self.labels = labels or {}
         self.heartbeat_interval = heartbeat_interval
         self.report_interval = report_interval
         self.poll_interval = poll_interval
@@ -54,7 +56,8 @@ class Agent:
                         headers["x-registration-token"] = \
                             self.registration_token
                     r = await c.post("/agent/register",
-                                     json={"node_id": self.node_id},
+                                     json={"node_id": self.node_id,
+                                           "labels": self.labels},
                                      headers=headers)
                     r.raise_for_status()
                     self.token = r.json()["node_token"]
@@ -132,5 +135,6 @@ class Agent:
 async def agent_run(engine, cancel: asyncio.Event) -> None:
     cfg = engine.config.server
     agent = Agent(engine, cfg.hub_url, cfg.node_id,
-                  registration_token=cfg.node_token)
+                  registration_token=cfg.node_token,
+                  labels=getattr(cfg, "node_labels", None))
     await agent.run(cancel)

@@ -54,7 +54,8 @@ class Hub:
 
     # ---- agent protocol -------------------------------------------------------
     async def register(self, node_id: str, reg_token: str = "",
-                       current_token: str = "") -> dict:
+                       current_token: str = "",
+                       labels: Optional[dict] = None) -> dict:
         if self.registration_token is not None and not hmac.compare_digest(
                 reg_token, self.registration_token):
             raise HTTPException(401, "bad registration token")
@@ -68,8 +69,10 @@ class Hub:
                     raise HTTPException(
                         409, "node is live; present current token to rotate")
         token = secrets.token_hex(16)
-        await self.store.upsert_node(node_id, token, self.lease_ttl)
-        await self._emit(node_id, "node_registered", {})
+        await self.store.upsert_node(node_id, token, self.lease_ttl,
+                                     labels=labels)
+        await self._emit(node_id, "node_registered",
+                         {"labels": labels or {}})
         return {"node_token": token, "lease_ttl_secs": self.lease_ttl}
 
     async def check_node(self, node_id: str, token: str) -> None:
@@ -136,9 +139,21 @@ class Hub:
     # ---- rollouts ----------------------------------------------------------------
     async def create_rollout(self, config: dict, nodes: List[str],
                              prev_config: Optional[dict] = None,
+                             selector: Optional[dict] = None,
                              actor: str = "operator") -> str:
+        """Placement: explicit node list, or a label selector resolved at
+        creation time ({"gpu": "mi355x", ...} → nodes whose labels match
+        every pair), reference-style placement constraints."""
+        if selector:
+            matched = [n["node_id"] for n in await self.store.nodes()
+                       if all(n.get("labels", {}).get(k) == v
+                              for k, v in selector.items())]
+            nodes = sorted(set(nodes) | set(matched)) if nodes else matched
+        if not nodes:
+            raise HTTPException(400, "rollout matched no nodes")
         rid = await self.store.create_rollout(config, nodes, prev_config)
-        await self.store.audit(actor, "rollout:create", rid)
+        await self.store.audit(actor, "rollout:create",
+                               f"{rid} nodes={','.join(nodes)}")
         return rid
 
     async def advance_rollouts(self) -> None:
@@ -259,7 +274,8 @@ def create_hub_app(hub: Hub) -> FastAPI:
         return await hub.register(
             node_id,
             reg_token=request.headers.get("x-registration-token", ""),
-            current_token=request.headers.get("x-node-token", ""))
+            current_token=request.headers.get("x-node-token", ""),
+            labels=body.get("labels") or None)
 
     @app.post("/agent/{node_id}/heartbeat")
     async def heartbeat(node_id: str, request: Request):
@@ -337,7 +353,8 @@ def create_hub_app(hub: Hub) -> FastAPI:
     async def create_rollout(body: dict):
         rid = await hub.create_rollout(body.get("config") or {},
                                        body.get("nodes") or [],
-                                       prev_config=body.get("prev_config"))
+                                       prev_config=body.get("prev_config"),
+                                       selector=body.get("selector"))
         return {"rollout_id": rid}
 
     @app.get("/rollouts", dependencies=[Depends(operator("read"))])

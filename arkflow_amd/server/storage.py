@@ -18,7 +18,8 @@ from typing import List, Optional
 _SCHEMA = """
 CREATE TABLE IF NOT EXISTS nodes (
   node_id TEXT PRIMARY KEY, token TEXT, registered_at REAL,
-  lease_expires REAL, last_report TEXT, online INTEGER DEFAULT 1
+  lease_expires REAL, last_report TEXT, online INTEGER DEFAULT 1,
+  labels TEXT DEFAULT NULL
 );
 CREATE TABLE IF NOT EXISTS desired (
   node_id TEXT, stream_id TEXT, state TEXT, generation INTEGER,
@@ -56,6 +57,7 @@ CREATE TABLE IF NOT EXISTS rollouts (
 _MIGRATIONS = [
     "ALTER TABLE intents ADD COLUMN attempts_made INTEGER DEFAULT 0",
     "ALTER TABLE rollouts ADD COLUMN prev_config TEXT DEFAULT NULL",
+    "ALTER TABLE nodes ADD COLUMN labels TEXT DEFAULT NULL",
 ]
 
 
@@ -77,14 +79,17 @@ class HubStore:
             return await loop.run_in_executor(None, fn)
 
     # ---- nodes -------------------------------------------------------------
-    async def upsert_node(self, node_id: str, token: str, lease_ttl: float):
+    async def upsert_node(self, node_id: str, token: str, lease_ttl: float,
+                          labels: Optional[dict] = None):
         def go():
             self._db.execute(
                 "INSERT INTO nodes(node_id, token, registered_at, "
-                "lease_expires, online) VALUES (?,?,?,?,1) "
+                "lease_expires, online, labels) VALUES (?,?,?,?,1,?) "
                 "ON CONFLICT(node_id) DO UPDATE SET token=excluded.token, "
-                "lease_expires=excluded.lease_expires, online=1",
-                (node_id, token, time.time(), time.time() + lease_ttl))
+                "lease_expires=excluded.lease_expires, online=1, "
+                "labels=COALESCE(excluded.labels, nodes.labels)",
+                (node_id, token, time.time(), time.time() + lease_ttl,
+                 json.dumps(labels) if labels else None))
             self._db.commit()
         await self._run(go)
 
@@ -128,6 +133,8 @@ class HubStore:
                 d = dict(r)
                 d["last_report"] = json.loads(d["last_report"]) \
                     if d["last_report"] else None
+                d["labels"] = json.loads(d["labels"]) if d.get("labels") \
+                    else {}
                 d.pop("token", None)
                 out.append(d)
             return out

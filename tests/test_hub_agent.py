@@ -536,3 +536,28 @@ def test_register_requires_token_and_protects_live_nodes(run):
             assert r.status_code == 200
 
     run(main())
+
+
+def test_rollout_label_selector_placement(run):
+    """Node labels + rollout placement by selector (resolved at create)."""
+    async def main():
+        hub = Hub()
+        await hub.store.upsert_node("a1", "t", 5.0,
+                                    labels={"gpu": "mi355x", "zone": "eu"})
+        await hub.store.upsert_node("a2", "t", 5.0,
+                                    labels={"gpu": "mi355x", "zone": "us"})
+        await hub.store.upsert_node("cpu1", "t", 5.0, labels={"gpu": "none"})
+        rid = await hub.create_rollout({"streams": []}, [],
+                                       selector={"gpu": "mi355x"})
+        r = await hub.store.get_rollout(rid)
+        assert r["nodes"] == ["a1", "a2"]
+        rid2 = await hub.create_rollout({"streams": []}, [],
+                                        selector={"zone": "eu"})
+        assert (await hub.store.get_rollout(rid2))["nodes"] == ["a1"]
+        with pytest.raises(Exception):
+            await hub.create_rollout({}, [], selector={"zone": "mars"})
+        nodes = await hub.store.nodes()
+        assert {n["node_id"]: n["labels"] for n in nodes}["a1"]["zone"] == \
+            "eu"
+
+    run(main())
