@@ -374,6 +374,54 @@ class SqlExecutor:
         part_start[0] = True
         part_start[1:] = sorted_gid[1:] != sorted_gid[:-1]
         group_start = torch.cummax(ar * part_start, 0).values
+        if name in ("lag", "lead", "first_value"):
+            # offset/navigation family over the sorted partition order;
+            # out-of-frame rows are NULL unless a default arg is given
+            if not w_.args or isinstance(w_.args[0], Star):
+                raise SqlError(f"{name}() requires a value argument")
+            v0 = eval_expr(w_.args[0], env)
+            vals = v0.data if isinstance(v0, Column) and v0.kind == "numeric" \
+                else as_tensor(v0, env)
+            vs = vals[perm]
+            if name == "first_value":
+                out_sorted_v = vs[group_start]
+                valid_sorted = None
+            else:
+                k = 1
+                if len(w_.args) > 1 and isinstance(w_.args[1], Literal):
+                    k = int(w_.args[1].value)
+                default = None
+                if len(w_.args) > 2:  # any scalar expr (e.g. -1)
+                    dv = eval_expr(w_.args[2], env)
+                    if isinstance(dv, Column):
+                        dv = dv.data
+                    if isinstance(dv, torch.Tensor):
+                        default = float(dv.reshape(-1)[0].item())
+                    else:
+                        default = float(dv)
+                counts = torch.bincount(sorted_gid, minlength=max(g, 1))
+                group_end = group_start + counts[sorted_gid] - 1
+                if name == "lag":
+                    src = ar - k
+                    valid_sorted = src >= group_start
+                else:
+                    src = ar + k
+                    valid_sorted = src <= group_end
+                out_sorted_v = vs[src.clamp(0, n - 1)]
+                if default is not None:
+                    out_sorted_v = torch.where(
+                        valid_sorted, out_sorted_v,
+                        torch.full_like(out_sorted_v, float(default)))
+                    valid_sorted = None
+            out_v = torch.empty_like(out_sorted_v)
+            out_v[perm] = out_sorted_v
+            if valid_sorted is None:
+                return out_v
+            out_valid = torch.empty(n, dtype=torch.bool, device=device)
+            out_valid[perm] = valid_sorted
+            col = Column("numeric", out_v)
+            col.validity = out_valid
+            return col
         out_sorted = ar - group_start + 1  # row_number
         if name in ("rank", "dense_rank"):
             tie_change = part_start.clone()

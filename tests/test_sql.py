@@ -365,3 +365,40 @@ def test_order_by_nulls_first_last():
     assert vals("SELECT v FROM flow ORDER BY v") == [None, 1.0, 3.0, 9.0]
     assert vals("SELECT v FROM flow ORDER BY v DESC") == \
         [9.0, 3.0, 1.0, None]
+
+
+def test_window_lag_lead_first_value():
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+
+    b = MessageBatch.from_dict({
+        "k": [1, 1, 1, 2, 2],
+        "v": [10.0, 20.0, 30.0, 5.0, 7.0],
+        "t": [1, 2, 3, 1, 2],
+    })
+
+    def run(q):
+        out = SqlExecutor(q).execute({"flow": b})
+        return out
+
+    out = run("SELECT t, k, lag(v) OVER (PARTITION BY k ORDER BY t) AS p "
+              "FROM flow ORDER BY k, t")
+    col = out.column("p")
+    assert col.validity.tolist() == [False, True, True, False, True]
+    vals = col.to_pylist()
+    assert vals[1:3] == [10.0, 20.0] and vals[4] == 5.0
+
+    out = run("SELECT t, k, lead(v, 1, -1) OVER (PARTITION BY k ORDER BY t) "
+              "AS nx FROM flow ORDER BY k, t")
+    assert out.column("nx").to_pylist() == [20.0, 30.0, -1.0, 7.0, -1.0]
+    assert out.column("nx").validity is None
+
+    out = run("SELECT t, k, first_value(v) OVER (PARTITION BY k ORDER BY t) "
+              "AS fv FROM flow ORDER BY k, t")
+    assert out.column("fv").to_pylist() == [10.0, 10.0, 10.0, 5.0, 5.0]
+
+    # lag by 2
+    out = run("SELECT t, k, lag(v, 2) OVER (PARTITION BY k ORDER BY t) AS p2 "
+              "FROM flow ORDER BY k, t")
+    assert out.column("p2").validity.tolist() == [False, False, True,
+                                                  False, False]
