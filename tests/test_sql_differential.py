@@ -35,6 +35,10 @@ QUERIES = [
     "SELECT k, count(*) AS c, sum(b) AS s FROM flow WHERE b >= 0.2 "
     "GROUP BY k ORDER BY k",
     "SELECT a FROM flow ORDER BY a DESC NULLS LAST LIMIT 10",
+    "SELECT a, lag(a, 1, -1) OVER (PARTITION BY k ORDER BY a) AS p "
+    "FROM flow ORDER BY a, p",
+    "SELECT a, first_value(b) OVER (PARTITION BY k ORDER BY a) AS fv "
+    "FROM flow ORDER BY a, fv",
     "SELECT a FROM flow WHERE a < 10 UNION ALL SELECT a FROM flow "
     "WHERE a > 90",
     "SELECT k, sum(a) AS s FROM flow GROUP BY k "
