@@ -79,6 +79,8 @@ void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
 void launch_scan_counts(const int32_t*, int, int32_t*, int32_t*,
                         hipStream_t);
 void launch_featpack(const float**, int, int, int64_t, void*, hipStream_t);
+void launch_featpack64(const double**, int, int, int64_t, void*,
+                       hipStream_t);
 void launch_gemv_bf16_f32(const void*, const void*, float, int64_t, int,
                           float*, hipStream_t);
 void launch_bytes_hash(const uint8_t*, const int64_t*, int64_t, int64_t*,
@@ -462,6 +464,18 @@ void featpack(std::vector<torch::Tensor> srcs, torch::Tensor out) {
               "out must be [n, kpad] bf16");
   int64_t n = out.size(0);
   int kpad = (int)out.size(1);
+  bool f64 = srcs[0].scalar_type() == torch::kFloat64;
+  if (f64) {
+    std::vector<const double*> ptrs;
+    for (auto& s : srcs) {
+      check_cuda(s, "src");
+      TORCH_CHECK(s.numel() >= n && s.scalar_type() == torch::kFloat64);
+      ptrs.push_back(s.data_ptr<double>());
+    }
+    launch_featpack64(ptrs.data(), (int)ptrs.size(), kpad, n,
+                      out.data_ptr(), cur_stream());
+    return;
+  }
   std::vector<const float*> ptrs;
   for (auto& s : srcs) {
     check_cuda(s, "src");
@@ -1073,7 +1087,7 @@ proto_decode(
     torch::Tensor data, torch::Tensor offsets,
     std::vector<int64_t> fno, std::vector<int64_t> kind,
     std::vector<int64_t> is_float, std::vector<int64_t> slot,
-    int64_t n_int, int64_t n_float, int64_t n_str) {
+    int64_t n_int, int64_t n_float, int64_t n_str, bool capture) {
   check_cuda(data, "data");
   check_cuda(offsets, "offsets");
   TORCH_CHECK(data.scalar_type() == torch::kUInt8);
@@ -1107,6 +1121,13 @@ proto_decode(
   // err + totals ride ONE host readback (JSON decode learned the same —
   // each extra sync floored small-batch decode)
   std::vector<std::tuple<torch::Tensor, torch::Tensor>> strings;
+  if (capture) {
+    // hipGraph capture: no host syncs allowed — string fields (which need
+    // a totals readback for allocation) are unsupported; the caller
+    // validates err outside the capture
+    TORCH_CHECK(n_str == 0, "capture-mode proto decode cannot have strings");
+    return {out_i, out_f, err, strings, torch::empty({0})};
+  }
   std::vector<torch::Tensor> sumv = {err.to(torch::kInt64)};
   std::vector<torch::Tensor> offs(std::max<int64_t>(n_str, 0));
   if (n_str > 0 && n > 0) {
@@ -1157,7 +1178,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = py::none(), py::arg("act") = 0);
   m.def("attention_bf16", &attention_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("scale"));
-  m.def("proto_decode", &proto_decode);
+  m.def("proto_decode", &proto_decode, py::arg("data"),
+        py::arg("offsets"), py::arg("fno"), py::arg("kind"),
+        py::arg("is_float"), py::arg("slot"), py::arg("n_int"),
+        py::arg("n_float"), py::arg("n_str"), py::arg("capture") = false);
   m.def("gather_columns", &gather_columns);
   m.def("gemm_bf16_variant", &gemm_bf16_variant);
   m.def("bytes_hash", &bytes_hash);

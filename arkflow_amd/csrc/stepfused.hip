@@ -140,6 +140,37 @@ extern "C" void launch_featpack(const float** src, int nf, int kpad,
   featpack_kernel<<<grid, 256, 0, st>>>(spec, n, (__bf16*)out);
 }
 
+// f64 source variant (protobuf floats decode to double)
+struct PackSpec64 {
+  const double* src[32];
+  int nf;
+  int kpad;
+};
+
+__global__ void featpack64_kernel(PackSpec64 spec, int64_t n,
+                                  __bf16* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    __bf16* row = out + i * spec.kpad;
+#pragma unroll 4
+    for (int f = 0; f < spec.nf; ++f) row[f] = (__bf16)(float)spec.src[f][i];
+    for (int f = spec.nf; f < spec.kpad; ++f) row[f] = (__bf16)0.f;
+  }
+}
+
+extern "C" void launch_featpack64(const double** src, int nf, int kpad,
+                                  int64_t n, void* out, hipStream_t st) {
+  PackSpec64 spec{};
+  spec.nf = nf > 32 ? 32 : nf;
+  spec.kpad = kpad;
+  for (int f = 0; f < spec.nf; ++f) spec.src[f] = src[f];
+  int grid = (int)((n + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) return;
+  featpack64_kernel<<<grid, 256, 0, st>>>(spec, n, (__bf16*)out);
+}
+
 // ---- gemv: scores = x[M,K]·w[K] + b → f32 ----------------------------------
 // one wave per 8 rows; lanes split K. K ≤ a few thousand (MLP hidden).
 __global__ void gemv_bf16_f32_kernel(const __bf16* __restrict__ x,

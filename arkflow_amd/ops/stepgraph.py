@@ -152,6 +152,68 @@ class FusedGenerateFilterInfer:
         return MessageBatch(cols, input_name="generate"), kept
 
 
+class FusedProtoMlp:
+    """kafka-shaped protobuf payloads → GPU varint decode → MLP scoring as
+    ONE hipGraph (BASELINE config 3). The payload batch is the static
+    device-resident input (the bench's kafka stand-in re-reads the same
+    wire bytes each step, as the eager path does); every replay re-runs
+    decode + featpack + the MFMA MLP. Scalar-only schemas (no string
+    fields: their output allocation needs a host readback). Decode errors
+    are validated once at capture; row count is fixed, so a step has ZERO
+    host syncs — the caller's event wait paces the stream.
+    """
+
+    def __init__(self, data: torch.Tensor, offsets: torch.Tensor,
+                 fno, kind, isf, slot, n_int: int, n_float: int,
+                 float_names, int_names, mlp, device: torch.device):
+        self.nat = require_native()
+        self.data, self.offsets = data, offsets
+        self.args = (list(fno), list(kind), list(isf), list(slot),
+                     n_int, n_float)
+        self.float_names = list(float_names)
+        self.int_names = list(int_names)
+        self.mlp = mlp
+        self.device = torch.device(device)
+        self.n = int(offsets.numel() - 1)
+        self.feats = torch.zeros((self.n, mlp.dims[0]), device=self.device,
+                                 dtype=torch.bfloat16)
+        self._graph = None
+
+    def _body(self):
+        fno, kind, isf, slot, n_int, n_float = self.args
+        out_i, out_f, err, _s, _sum = self.nat.proto_decode(
+            self.data, self.offsets, fno, kind, isf, slot, n_int, n_float,
+            0, capture=True)
+        self.nat.featpack([out_f[i] for i in range(len(self.float_names))],
+                          self.feats)
+        return out_i, out_f, err, self.mlp._net(self.feats)
+
+    def capture(self) -> None:
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                out = self._body()
+        torch.cuda.current_stream().wait_stream(s)
+        if int(out[2].item()) != 0:
+            raise RuntimeError("proto decode error in fused warmup")
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._out_i, self._out_f, self._err, self._scores = self._body()
+
+    def step(self) -> MessageBatch:
+        if self._graph is None:
+            self.capture()
+        self._graph.replay()
+        cols = {}
+        for i, name in enumerate(self.int_names):
+            cols[name] = Column("numeric", self._out_i[i])
+        for i, name in enumerate(self.float_names):
+            cols[name] = Column("numeric", self._out_f[i])
+        cols["score"] = Column("numeric", self._scores)
+        return MessageBatch(cols, input_name="kafka")
+
+
 class FusedStepSource:
     """Input-SPI facade over FusedGenerateFilterInfer: `read()` yields the
     fully processed batch, so a Stream/bench drives the fused step through

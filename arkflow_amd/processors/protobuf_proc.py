@@ -38,6 +38,37 @@ def _load_schema(config: dict) -> ProtoSchema:
     return ProtoSchema.parse(src, config.get("message"))
 
 
+def build_gpu_spec(schema):
+    """Kernel field spec from a ProtoSchema: (fno, kind, isf, slot,
+    int_fields, float_fields, str_fields) — shared by the processor and the
+    fused bench graph (ops/stepgraph.FusedProtoMlp)."""
+    fno, kind, isf, slot = [], [], [], []
+    int_fields, float_fields, str_fields = [], [], []
+    for no in sorted(schema.fields):
+        name, t = schema.fields[no]
+        fno.append(no)
+        if t in ("string", "bytes"):
+            kind.append(9)
+            isf.append(0)
+            slot.append(len(str_fields))
+            str_fields.append(name)
+        elif t in _FLOAT_TYPES:
+            kind.append(_KIND_ENUM["f64" if t == "double" else "f32"])
+            isf.append(1)
+            slot.append(len(float_fields))
+            float_fields.append(name)
+        else:
+            # uint32/uint64/fixed stay in int64 (values < 2^63 in
+            # practice); only true floats go to the f64 output
+            k = {"sint32": 1, "sint64": 1, "fixed64": 4, "sfixed64": 5,
+                 "fixed32": 6, "sfixed32": 7}.get(t, 0)
+            kind.append(k)
+            isf.append(0)
+            slot.append(len(int_fields))
+            int_fields.append(name)
+    return fno, kind, isf, slot, int_fields, float_fields, str_fields
+
+
 class ProtobufToArrowProcessor(Processor):
     def __init__(self, config: dict, resource=None):
         self.schema = _load_schema(config)
@@ -63,30 +94,8 @@ class ProtobufToArrowProcessor(Processor):
     def _decode_gpu(self, col: Column) -> MessageBatch:
         from .. import ops
         nat = ops.require_native()
-        fno, kind, isf, slot = [], [], [], []
-        int_fields, float_fields, str_fields = [], [], []
-        for no in sorted(self.schema.fields):
-            name, t = self.schema.fields[no]
-            fno.append(no)
-            if t in ("string", "bytes"):
-                kind.append(9)
-                isf.append(0)
-                slot.append(len(str_fields))
-                str_fields.append(name)
-            elif t in _FLOAT_TYPES:
-                kind.append(_KIND_ENUM["f64" if t == "double" else "f32"])
-                isf.append(1)
-                slot.append(len(float_fields))
-                float_fields.append(name)
-            else:
-                # uint32/uint64/fixed stay in int64 (values < 2^63 in
-                # practice); only true floats go to the f64 output
-                k = {"sint32": 1, "sint64": 1, "fixed64": 4, "sfixed64": 5,
-                     "fixed32": 6, "sfixed32": 7}.get(t, 0)
-                kind.append(k)
-                isf.append(0)
-                slot.append(len(int_fields))
-                int_fields.append(name)
+        (fno, kind, isf, slot,
+         int_fields, float_fields, str_fields) = build_gpu_spec(self.schema)
         out_i, out_f, err, strings, summary = nat.proto_decode(
             col.data, col.offsets, fno, kind, isf, slot,
             len(int_fields), len(float_fields), len(str_fields))

@@ -92,6 +92,30 @@ def build_pipeline(args, device):
                 from arkflow_amd.spi import NoopAck
                 return device_batch, NoopAck()
 
+        if device.type == "cuda" and not args.no_stepgraph:
+            # decode+featpack+MLP replay as one hipGraph (zero host syncs
+            # per step — row count is fixed; same kernels as the eager path,
+            # numerics-tested in tests/test_gpu_kernels.py)
+            from arkflow_amd.models.mlp import MlpAnomalyDetector
+            from arkflow_amd.ops.stepgraph import FusedProtoMlp
+            from arkflow_amd.processors.protobuf_proc import build_gpu_spec
+            fno, kind, isf, slot, int_f, float_f, str_f = \
+                build_gpu_spec(schema)
+            assert not str_f
+            col = device_batch.column("__value__")
+            mlp = MlpAnomalyDetector(len(float_f),
+                                     [args.hidden, args.hidden], device,
+                                     1234)
+            fused = FusedProtoMlp(col.data, col.offsets, fno, kind, isf,
+                                  slot, len(int_f), len(float_f), float_f,
+                                  int_f, mlp, device)
+
+            class _FusedProtoSrc:
+                async def read(self):
+                    from arkflow_amd.spi import NoopAck
+                    return fused.step(), NoopAck()
+
+            return _FusedProtoSrc(), Pipeline([])
         decode = ProtobufToArrowProcessor({"proto": proto}, None)
         infer = InferenceProcessor({
             "model": "mlp_anomaly", "columns": ["f0", "f1", "f2", "f3"],

@@ -1008,3 +1008,50 @@ def test_fast_agg_matches_engine_path(dev):
     assert p2._fast_agg is not None
     f2 = loop.run_until_complete(p2.process(batch))[0]
     assert int(f2.column("c").data.sum()) == n
+
+
+def test_fused_proto_mlp_matches_eager(dev):
+    """FusedProtoMlp graph output vs the eager decode→infer chain on the
+    same payloads + weights (BASELINE config 3)."""
+    import asyncio
+    import random
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.models.mlp import MlpAnomalyDetector
+    from arkflow_amd.ops.stepgraph import FusedProtoMlp
+    from arkflow_amd.processors.inference import InferenceProcessor
+    from arkflow_amd.processors.proto_wire import ProtoSchema, encode_message
+    from arkflow_amd.processors.protobuf_proc import (
+        ProtobufToArrowProcessor, build_gpu_spec)
+
+    proto = ("message T { double f0 = 1; double f1 = 2; double f2 = 3; "
+             "double f3 = 4; int64 key = 5; }")
+    schema = ProtoSchema.parse(proto)
+    rng = random.Random(99)
+    payloads = [encode_message(
+        {"f0": rng.random(), "f1": rng.random(), "f2": rng.random(),
+         "f3": rng.random(), "key": rng.randrange(1024)}, schema)
+        for _ in range(4096)]
+    batch = MessageBatch.from_binary(payloads, input_name="kafka").to(dev)
+    col = batch.column("__value__")
+    fno, kind, isf, slot, int_f, float_f, str_f = build_gpu_spec(schema)
+    mlp = MlpAnomalyDetector(4, [64, 64], dev, 7)
+    fused = FusedProtoMlp(col.data, col.offsets, fno, kind, isf, slot,
+                          len(int_f), len(float_f), float_f, int_f, mlp, dev)
+    out = fused.step()
+    out2 = fused.step()  # replay is stable
+
+    loop = asyncio.new_event_loop()
+    dec = ProtobufToArrowProcessor({"proto": proto}, None)
+    eager = loop.run_until_complete(dec.process(batch))[0]
+    inf = InferenceProcessor({"model": "mlp_anomaly",
+                              "columns": ["f0", "f1", "f2", "f3"],
+                              "hidden": [64, 64], "device": str(dev),
+                              "seed": 7})
+    eager = loop.run_until_complete(inf.process(eager))[0]
+    assert torch.equal(out.column("key").data, eager.column("key").data)
+    for f in ("f0", "f1", "f2", "f3"):
+        assert torch.allclose(out.column(f).data, eager.column(f).data)
+    assert torch.allclose(out.column("score").data,
+                          eager.column("score").data, atol=1e-3, rtol=1e-2)
+    assert torch.equal(out2.column("score").data, out.column("score").data)
