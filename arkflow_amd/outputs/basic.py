@@ -30,6 +30,8 @@ class StdoutOutput(Output):
 
 
 class DropOutput(Output):
+    retains = False  # discards immediately: zero-copy fused batches are safe
+
     def __init__(self, config: dict, resource=None):
         pass
 
