@@ -1055,3 +1055,26 @@ def test_fused_proto_mlp_matches_eager(dev):
     assert torch.allclose(out.column("score").data,
                           eager.column("score").data, atol=1e-3, rtol=1e-2)
     assert torch.equal(out2.column("score").data, out.column("score").data)
+
+
+def test_attention_flash_long_seq(nat, dev):
+    """Flash-tiled attention (S>128) vs fp32 reference, both entry points."""
+    torch.manual_seed(21)
+    for B, H, S, D in ((2, 4, 256, 64), (1, 2, 384, 64), (1, 1, 512, 64)):
+        scale = D ** -0.5
+        qkv = torch.randn(B, S, 3, H, D, device=dev, dtype=torch.bfloat16)
+        out = nat.attention_qkv_bf16(qkv, scale)
+        q = qkv[:, :, 0].permute(0, 2, 1, 3).float()
+        k = qkv[:, :, 1].permute(0, 2, 1, 3).float()
+        v = qkv[:, :, 2].permute(0, 2, 1, 3).float()
+        p = torch.softmax(q @ k.transpose(-1, -2) * scale, -1)
+        ref = (p @ v).permute(0, 2, 1, 3).reshape(B, S, H * D)
+        err = (out.float() - ref).abs().max().item()
+        assert err < 0.02, (S, err)
+        # plain [BH,S,D] entry
+        out2 = nat.attention_bf16(
+            q.to(torch.bfloat16).reshape(B * H, S, D).contiguous(),
+            k.to(torch.bfloat16).reshape(B * H, S, D).contiguous(),
+            v.to(torch.bfloat16).reshape(B * H, S, D).contiguous(), scale)
+        ref2 = (p @ v).reshape(B * H, S, D)
+        assert (out2.float() - ref2).abs().max().item() < 0.02, S
