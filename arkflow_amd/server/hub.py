@@ -110,6 +110,52 @@ class Hub:
                           "stream_id": stream_id})
         return intent_id
 
+    async def set_desired(self, node_id: str, stream_id: str, state: str,
+                          actor: str = "operator") -> int:
+        """Declarative path (reference get_desired in reconcile_once,
+        hub.rs:468-520): record the desired stream state; the reconcile
+        loop converges the node by enqueuing lifecycle intents whenever
+        the last report disagrees."""
+        if state not in ("running", "stopped"):
+            raise HTTPException(400, "desired state must be running|stopped")
+        gen = await self.store.set_desired(node_id, stream_id, state)
+        await self.store.audit(actor, f"desired:{state}",
+                               f"{node_id}/{stream_id}@g{gen}")
+        await self._emit(node_id, "desired_set",
+                         {"stream_id": stream_id, "state": state,
+                          "generation": gen})
+        return gen
+
+    async def _converge_desired(self, online: List[str]) -> int:
+        """Enqueue intents for any (node, stream) whose observed state in
+        the node's last report diverges from the desired state — unless an
+        intent is already in flight for it."""
+        enqueued = 0
+        inflight = {(i["node_id"], i["stream_id"])
+                    for i in await self.store.intents(limit=500)
+                    if i["state"] in ("pending", "dispatched")}
+        for n in await self.store.nodes():
+            node_id = n["node_id"]
+            if node_id not in online:
+                continue
+            report = n.get("last_report") or {}
+            observed = {s.get("id"): s.get("state")
+                        for s in report.get("streams") or []}
+            for d in await self.store.desired(node_id):
+                sid, want = d["stream_id"], d["state"]
+                have = observed.get(sid)
+                if have is None and want != "running":
+                    continue  # unknown stream, nothing to stop
+                if (want == "running") == (have == "running"):
+                    continue  # converged
+                if (node_id, sid) in inflight:
+                    continue
+                op = "start" if want == "running" else "stop"
+                await self.enqueue_intent(node_id, sid, op,
+                                          actor="reconciler")
+                enqueued += 1
+        return enqueued
+
     # ---- reconciliation ---------------------------------------------------------
     async def reconcile_once(self) -> int:
         """intents → outbox → attempts (agent commands) (hub.rs:468-520).
@@ -120,6 +166,7 @@ class Hub:
         await self.store.expire_attempts()
         online = [n["node_id"] for n in await self.store.nodes()
                   if n["online"]]
+        await self._converge_desired(online)
         claimed = await self.store.claim_outbox(online)
         dispatched = 0
         for row in claimed:
@@ -314,6 +361,18 @@ def create_hub_app(hub: Hub) -> FastAPI:
             raise HTTPException(400, "unknown op")
         intent_id = await hub.enqueue_intent(node_id, stream_id, op)
         return {"intent_id": intent_id}
+
+    @app.post("/nodes/{node_id}/streams/{stream_id}/desired",
+              dependencies=[Depends(operator("write"))])
+    async def set_desired(node_id: str, stream_id: str, body: dict):
+        gen = await hub.set_desired(node_id, stream_id,
+                                    str(body.get("state", "")))
+        return {"generation": gen}
+
+    @app.get("/nodes/{node_id}/desired",
+             dependencies=[Depends(operator("read"))])
+    async def get_desired(node_id: str):
+        return await hub.store.desired(node_id)
 
     @app.get("/intents", dependencies=[Depends(operator("read"))])
     async def intents():

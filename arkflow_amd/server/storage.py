@@ -148,6 +148,32 @@ class HubStore:
             return r["token"] if r else None
         return await self._run(go)
 
+    # ---- desired state ------------------------------------------------------
+    async def set_desired(self, node_id: str, stream_id: str, state: str
+                          ) -> int:
+        """Record the operator's desired state; bumps the generation."""
+        def go():
+            r = self._db.execute(
+                "SELECT generation FROM desired WHERE node_id=? AND "
+                "stream_id=?", (node_id, stream_id)).fetchone()
+            gen = (r["generation"] if r else 0) + 1
+            self._db.execute(
+                "INSERT INTO desired VALUES (?,?,?,?) ON CONFLICT"
+                "(node_id, stream_id) DO UPDATE SET state=excluded.state, "
+                "generation=excluded.generation",
+                (node_id, stream_id, state, gen))
+            self._db.commit()
+            return gen
+        return await self._run(go)
+
+    async def desired(self, node_id: str) -> List[dict]:
+        def go():
+            rows = self._db.execute(
+                "SELECT * FROM desired WHERE node_id=?",
+                (node_id,)).fetchall()
+            return [dict(r) for r in rows]
+        return await self._run(go)
+
     # ---- intents / outbox / attempts ----------------------------------------
     async def enqueue_intent(self, node_id: str, stream_id: str, op: str
                              ) -> str:

@@ -561,3 +561,44 @@ def test_rollout_label_selector_placement(run):
             "eu"
 
     run(main())
+
+
+def test_desired_state_reconciliation(run):
+    """Declarative desired state: the reconcile loop converges a node whose
+    report diverges (reference get_desired flow, hub.rs:468-520)."""
+    async def main():
+        hub = Hub()
+        await hub.store.upsert_node("n1", "tok", 5.0)
+        await hub.store.report("n1", {"streams": [
+            {"id": "s1", "state": "stopped"}]})
+        await hub.set_desired("n1", "s1", "running")
+        assert await hub.reconcile_once() == 1  # start intent dispatched
+        att = None
+        for i in await hub.store.intents():
+            if i["node_id"] == "n1" and i["stream_id"] == "s1":
+                att = i
+        assert att is not None and att["op"] == "start"
+        # while in flight, no duplicate intent
+        await hub.reconcile_once()
+        starts = [i for i in await hub.store.intents()
+                  if i["stream_id"] == "s1" and i["op"] == "start"]
+        assert len(starts) == 1
+        # node converges → no new intents
+        a = await hub.store.attempt_for_intent(att["intent_id"])
+        await hub.store.command_result(a["attempt_id"], True)
+        await hub.store.report("n1", {"streams": [
+            {"id": "s1", "state": "running"}]})
+        await hub.reconcile_once()
+        starts = [i for i in await hub.store.intents()
+                  if i["stream_id"] == "s1" and i["op"] == "start"]
+        assert len(starts) == 1
+        # desired stopped → stop intent
+        await hub.set_desired("n1", "s1", "stopped")
+        await hub.reconcile_once()
+        ops = [i["op"] for i in await hub.store.intents()
+               if i["stream_id"] == "s1"]
+        assert "stop" in ops
+        with pytest.raises(Exception):
+            await hub.set_desired("n1", "s1", "bogus")
+
+    run(main())
