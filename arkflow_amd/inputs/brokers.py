@@ -9,11 +9,13 @@ has two drivers:
     (partitions, offsets, consumer groups, transactions) so delivery
     guarantees are TESTABLE offline, the way the reference gates its
     testcontainers suites.
-  - real driver — KAFKA ONLY: a full confluent_kafka consumer/transactional-
-    producer implementation (inputs/kafka_real.py), exercised by the same
-    contract test bodies when ``KAFKA_BOOTSTRAP`` is set. The mqtt / nats /
-    pulsar / redis real drivers are NOT implemented — their connect() raises
-    ConnectionError_ and only the fake bus carries their semantics.
+  - real drivers — kafka: full confluent_kafka consumer/transactional-
+    producer (inputs/kafka_real.py), exercised by the same contract test
+    bodies when ``KAFKA_BOOTSTRAP`` is set; redis (redis-py: pubsub/BLPOP/
+    XREAD), mqtt (paho) and nats (nats-py core + JetStream w/ per-message
+    acks) in inputs/pubsub_real.py, env-gated the same way. pulsar has NO
+    real driver — its connect() raises ConnectionError_ and only the fake
+    bus carries its semantics.
 
 Kafka semantics mirrored from the reference:
   input: per-message read, ``__meta_*`` metadata columns, ack = commit offset
@@ -296,6 +298,7 @@ class _PubSubInput(Input):
         self.url = str(config.get("url", "memory://default"))
         self.driver = config.get("driver") or (
             "memory" if self.url.startswith("memory://") else "real")
+        self.jetstream = bool(config.get("jetstream", False))
         from ..codecs.helper import build_codec
         self.codec = build_codec(config, resource)
         self.bus: Optional[FakeBus] = None
@@ -304,13 +307,18 @@ class _PubSubInput(Input):
 
     def _make_real(self):
         """Real client for this kind, or raise ConnectionError_.
-        redis → redis-py; mqtt → paho-mqtt; nats/pulsar have no real driver
-        (fake bus only — their client libraries are not modeled)."""
+        redis → redis-py; mqtt → paho-mqtt; nats → nats-py (core +
+        JetStream). pulsar has no real driver (fake bus only)."""
         from urllib.parse import urlparse
 
-        from .pubsub_real import RealMqttClient, RealRedisClient
+        from .pubsub_real import (RealMqttClient, RealNatsClient,
+                                  RealRedisClient)
         if self.kind == "redis":
             return RealRedisClient(self.url, getattr(self, "mode", "pubsub"))
+        if self.kind == "nats":
+            return RealNatsClient(self.url,
+                                  jetstream=bool(getattr(self, "jetstream",
+                                                         False)))
         if self.kind == "mqtt":
             u = urlparse(self.url if "://" in self.url
                          else f"mqtt://{self.url}")
@@ -338,6 +346,10 @@ class _PubSubInput(Input):
             if self.kind == "redis":
                 value, off = await self._real.read(self.topic)
                 return make_batch(self.topic, value, self.codec, off)
+            if self.kind == "nats":
+                topic, value, ack = await self._real.read()
+                batch, noop = make_batch(topic, value, self.codec)
+                return batch, (ack or noop)
             topic, value = await self._real.read()
             return make_batch(topic, value, self.codec)
         item = await self._q.get()
@@ -397,6 +409,10 @@ class _PubSubOutput(Output):
         else:
             values = batch.to_json_lines()
         if getattr(self, "_real", None) is not None:
+            if self.kind == "nats":
+                for v in values:
+                    await self._real.aproduce(self.topic, v)
+                return
             loop = asyncio.get_running_loop()
             for v in values:
                 await loop.run_in_executor(
