@@ -254,8 +254,9 @@ void gemm_bf16_k64_kernel(const __bf16* __restrict__ A,   // [M,K]
 // period: the 16 fragment rows a quarter-wave reads per ds_read_b128 alias
 // onto the SAME banks (PMC r2: 4.4e5 conflicts/dispatch on the k64 kernel).
 // XOR-swizzling the 16 B chunk index by (row & 7) spreads them: chunk' =
-// chunk ^ (row & 7) — staging applies the same transform, so the layout is
-// consistent and conflicts drop to ≤2-way (rows r and r+8 share).
+// chunk ^ (row & 7). global_load_lds writes a wave's 1 KiB LINEARLY from a
+// scalar base, so swizzled staging needs register staging + per-lane
+// ds_write_b128 instead (the async direct-to-LDS path cannot express it).
 template <int ACT, bool HAS_BIAS>
 __global__ __launch_bounds__(GEMM_THREADS, 2)
 void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
@@ -286,28 +287,30 @@ void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
 
   const int lin0 = wid * 1024 + lane * 16;
 
+  typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
   for (int k0 = 0; k0 < K; k0 += BK2) {
+    // register staging: global 16 B loads → swizzled ds_write per lane
+    u32x4 va[4], vb[4];
+    int dsts[4];
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
       int lin = lin0 + it * 4096;
       int trow = lin >> 7;                 // 128 B per row
       int tcol = lin & 127;                // multiple of 16
-      int sw = ((tcol >> 4) ^ (trow & 7)) << 4;  // swizzled 16 B chunk
-      int dst = (trow << 7) + sw;
+      dsts[it] = (trow << 7) + (((tcol >> 4) ^ (trow & 7)) << 4);
       int ga_row = row0 + trow;
       ga_row = ga_row < M ? ga_row : M - 1;
-      const char* a_src = (const char*)(A + (int64_t)ga_row * K + k0) + tcol;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)a_src,
-          (__attribute__((address_space(3))) uint32_t*)((char*)Asm + dst),
-          16, 0, 0);
+      va[it] = *(const u32x4*)((const char*)(A + (int64_t)ga_row * K + k0)
+                               + tcol);
       int gb_row = col0 + trow;
       gb_row = gb_row < N ? gb_row : N - 1;
-      const char* b_src = (const char*)(Bt + (int64_t)gb_row * K + k0) + tcol;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) uint32_t*)b_src,
-          (__attribute__((address_space(3))) uint32_t*)((char*)Bsm + dst),
-          16, 0, 0);
+      vb[it] = *(const u32x4*)((const char*)(Bt + (int64_t)gb_row * K + k0)
+                               + tcol);
+    }
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      *(u32x4*)((char*)Asm + dsts[it]) = va[it];
+      *(u32x4*)((char*)Bsm + dsts[it]) = vb[it];
     }
     __syncthreads();
 
