@@ -396,14 +396,16 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor Bt,
                               C.data_ptr(), (int)M, (int)N, (int)K,
                               (int)act, cur_stream());
     } else {
-      // K≤1024 multiples of 64 (BERT fc1/QKV class): the BK=64 slice wins
-      // ~10-15% over BK=32 (profiles r2 bench_gemm_k64); at deep K (fc2
-      // K=3072) BK=32 stays ahead
-      int rc2 = (K % 64 == 0 && K <= 1024)
-                    ? launch_gemm_bf16_k64(A.data_ptr(), Bt.data_ptr(),
-                                           bias_ptr, C.data_ptr(), (int)M,
-                                           (int)N, (int)K, (int)act,
-                                           cur_stream())
+      // K%64==0 (BERT fc1/QKV/fc2 class): the swizzled BK=64 tile wins
+      // 6-21% over BK=32 at every such shape measured (profiles r2
+      // bench_gemm_k64; register staging + XOR chunk swizzle — the linear
+      // layout's 128B-period LDS alias cost ~4e5 bank conflicts/dispatch).
+      // Big squares already went to the 8-phase kernel above.
+      int rc2 = (K % 64 == 0)
+                    ? launch_gemm_bf16_k64s(A.data_ptr(), Bt.data_ptr(),
+                                            bias_ptr, C.data_ptr(), (int)M,
+                                            (int)N, (int)K, (int)act,
+                                            cur_stream())
                     : -1;
       if (rc2 != 0)
         launch_gemm_bf16(A.data_ptr(), Bt.data_ptr(), bias_ptr,
