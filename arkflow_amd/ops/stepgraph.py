@@ -22,7 +22,7 @@ step becomes one device-side graph so Python dispatch cost is O(1) per step.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Tuple
+from typing import Dict, Tuple
 
 import torch
 
