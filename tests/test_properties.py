@@ -134,3 +134,66 @@ def test_json_schema_host_decode_matches_jsonloads(docs):
         else:
             assert n[i] is None
         assert bb[i] == (d["b"] if isinstance(d["b"], bool) else None)
+
+
+# ---- VRL interpreter properties ---------------------------------------------
+_vrl_scalar = st.one_of(
+    st.integers(min_value=-10**6, max_value=10**6),
+    st.floats(allow_nan=False, allow_infinity=False, width=32),
+    st.booleans(),
+    st.text(alphabet=st.characters(blacklist_categories=("Cs",),
+                                   max_codepoint=0x2FF), max_size=12),
+)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.dictionaries(
+    st.from_regex(r"[a-z][a-z0-9_]{0,6}", fullmatch=True),
+    _vrl_scalar, min_size=1, max_size=5))
+def test_vrl_roundtrip_preserves_types(event):
+    """encode_json(parse_json(x)) round-trips the event unchanged through
+    the interpreter (reference vrl.rs type-preservation suite)."""
+    import json
+
+    from arkflow_amd.processors.vrl_lang import VrlProgram
+
+    src = '.encoded = encode_json(.orig)\n.back = parse_json!(.encoded)'
+    ev = VrlProgram(src).remap({"orig": dict(event)})
+    assert ev["back"] == event
+    assert json.loads(ev["encoded"]) == json.loads(
+        json.dumps(event))
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(max_size=80))
+def test_vrl_parser_never_crashes_unhandled(src):
+    """Arbitrary input either parses or raises ConfigError — never an
+    unhandled exception type."""
+    from arkflow_amd.errors import ConfigError
+    from arkflow_amd.processors.vrl_lang import parse_vrl
+    try:
+        parse_vrl(src)
+    except ConfigError:
+        pass
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.integers(min_value=-2**62, max_value=2**62),
+                min_size=0, max_size=300),
+       st.booleans())
+def test_sort_indices_matches_sorted(keys, desc):
+    """CPU sort_indices (torch stable argsort path) is a stable sort."""
+    import torch
+
+    from arkflow_amd import ops
+    t = torch.tensor(keys, dtype=torch.int64)
+    idx = ops.sort_indices(t, ascending=not desc)
+    got = t[idx].tolist()
+    assert got == sorted(keys, reverse=desc)
+    # stability: equal keys keep original order
+    seen = {}
+    for pos, i in enumerate(idx.tolist()):
+        k = keys[i]
+        if k in seen:
+            assert i > seen[k]
+        seen[k] = i
