@@ -362,6 +362,165 @@ void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- double-buffered swizzled BK=64 ------------------------------------------
+// Same tile/swizzle as k64s but two LDS buffers: the NEXT K-slice's global
+// loads are issued before the current slice's MFMAs so HBM latency hides
+// under compute, and the loop needs ONE __syncthreads per slice instead of
+// two. LDS 2×32 KiB — still 2 blocks/CU.
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 2)
+void gemm_bf16_k64d_kernel(const __bf16* __restrict__ A,   // [M,K]
+                           const __bf16* __restrict__ Bt,  // [N,K]
+                           const float* __restrict__ bias, // [N] or null
+                           __bf16* __restrict__ C,         // [M,N]
+                           int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * BM, col0 = bn * BN;
+
+  __shared__ __bf16 Asm[2][BM * BK2];
+  __shared__ __bf16 Bsm[2][BN * BK2];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+
+  f32x4 acc[4][4] = {};
+
+  const int lin0 = wid * 1024 + lane * 16;
+  typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
+
+  // per-lane staging geometry (constant across slices)
+  int trows[4], dsts[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    int lin = lin0 + it * 4096;
+    int trow = lin >> 7;
+    int tcol = lin & 127;
+    trows[it] = trow;
+    dsts[it] = (trow << 7) + (((tcol >> 4) ^ (trow & 7)) << 4);
+  }
+  int tcols[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) tcols[it] = (lin0 + it * 4096) & 127;
+
+  auto load_slice = [&](int k0, u32x4* va, u32x4* vb) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int ga_row = row0 + trows[it];
+      ga_row = ga_row < M ? ga_row : M - 1;
+      va[it] = *(const u32x4*)((const char*)(A + (int64_t)ga_row * K + k0)
+                               + tcols[it]);
+      int gb_row = col0 + trows[it];
+      gb_row = gb_row < N ? gb_row : N - 1;
+      vb[it] = *(const u32x4*)((const char*)(Bt + (int64_t)gb_row * K + k0)
+                               + tcols[it]);
+    }
+  };
+  auto write_slice = [&](int buf, const u32x4* va, const u32x4* vb) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      *(u32x4*)((char*)Asm[buf] + dsts[it]) = va[it];
+      *(u32x4*)((char*)Bsm[buf] + dsts[it]) = vb[it];
+    }
+  };
+
+  u32x4 va[4], vb[4];
+  load_slice(0, va, vb);
+  write_slice(0, va, vb);
+  __syncthreads();
+
+  const int fr = lane & 15;
+  const int fk8 = lane >> 4;
+  int buf = 0;
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+    const bool has_next = k0 + BK2 < K;
+    if (has_next) load_slice(k0 + BK2, va, vb);  // hides under the MFMAs
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int row = wm + m * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        a_frag[m] = *(const bf16x8*)&Asm[buf][row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int row = wn + n * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        b_frag[n] = *(const bf16x8*)&Bsm[buf][row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    }
+    if (has_next) write_slice(buf ^ 1, va, vb);
+    buf ^= 1;
+    __syncthreads();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" int launch_gemm_bf16_k64d(const void* A, const void* Bt,
+                                     const float* bias, void* C, int M,
+                                     int N, int K, int act, hipStream_t st) {
+  if (K % BK2 != 0) return -1;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define KDDISPATCH(ACT)                                                      \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_k64d_kernel<ACT, true><<<grid, block, 0, st>>>(              \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_k64d_kernel<ACT, false><<<grid, block, 0, st>>>(             \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: KDDISPATCH(ACT_RELU); break;
+    case ACT_GELU: KDDISPATCH(ACT_GELU); break;
+    case ACT_SILU: KDDISPATCH(ACT_SILU); break;
+    default: KDDISPATCH(ACT_NONE); break;
+  }
+#undef KDDISPATCH
+  return 0;
+}
+
 extern "C" int launch_gemm_bf16_k64s(const void* A, const void* Bt,
                                      const float* bias, void* C, int M,
                                      int N, int K, int act, hipStream_t st) {
