@@ -299,6 +299,7 @@ class _PubSubInput(Input):
         self.driver = config.get("driver") or (
             "memory" if self.url.startswith("memory://") else "real")
         self.jetstream = bool(config.get("jetstream", False))
+        self.subscription = config.get("subscription", "arkflow")
         from ..codecs.helper import build_codec
         self.codec = build_codec(config, resource)
         self.bus: Optional[FakeBus] = None
@@ -312,13 +313,16 @@ class _PubSubInput(Input):
         from urllib.parse import urlparse
 
         from .pubsub_real import (RealMqttClient, RealNatsClient,
-                                  RealRedisClient)
+                                  RealPulsarClient, RealRedisClient)
         if self.kind == "redis":
             return RealRedisClient(self.url, getattr(self, "mode", "pubsub"))
         if self.kind == "nats":
             return RealNatsClient(self.url,
                                   jetstream=bool(getattr(self, "jetstream",
                                                          False)))
+        if self.kind == "pulsar":
+            return RealPulsarClient(self.url,
+                                    getattr(self, "subscription", "arkflow"))
         if self.kind == "mqtt":
             u = urlparse(self.url if "://" in self.url
                          else f"mqtt://{self.url}")
@@ -346,7 +350,7 @@ class _PubSubInput(Input):
             if self.kind == "redis":
                 value, off = await self._real.read(self.topic)
                 return make_batch(self.topic, value, self.codec, off)
-            if self.kind == "nats":
+            if self.kind in ("nats", "pulsar"):
                 topic, value, ack = await self._real.read()
                 batch, noop = make_batch(topic, value, self.codec)
                 return batch, (ack or noop)

@@ -264,3 +264,78 @@ class RealNatsClient:
 
     def close(self) -> None:
         self._closed = True
+
+
+# ------------------------------------------------------------------ pulsar
+class _PulsarAck(Ack):
+    def __init__(self, consumer, msg):
+        self.consumer, self.msg = consumer, msg
+
+    async def ack(self) -> None:
+        import asyncio as _a
+        await _a.get_running_loop().run_in_executor(
+            None, self.consumer.acknowledge, self.msg)
+
+
+class RealPulsarClient:
+    """pulsar-client driver: shared-subscription consumer with per-message
+    acknowledge (the engine Ack), blocking calls on executor threads
+    (reference input/pulsar.rs + pulsar/common.rs)."""
+
+    def __init__(self, url: str, subscription: str = "arkflow"):
+        self.url = url
+        self.subscription = subscription
+        self.client = None
+        self.consumer = None
+        self._producers = {}
+        self._closed = False
+
+    def connect(self, subscribe: Optional[str] = None) -> None:
+        try:
+            import pulsar  # type: ignore
+        except ImportError as e:
+            raise ConnectionError_(
+                "pulsar real driver requires pulsar-client; use "
+                "driver: memory") from e
+        try:
+            self.client = pulsar.Client(self.url)
+            if subscribe is not None:
+                self.consumer = self.client.subscribe(
+                    subscribe, self.subscription,
+                    consumer_type=pulsar.ConsumerType.Shared)
+        except ConnectionError_:
+            raise
+        except Exception as e:  # noqa: BLE001
+            raise ConnectionError_(f"pulsar connect failed: {e}") from e
+
+    async def read(self):
+        import asyncio as _a
+        loop = _a.get_running_loop()
+        while not self._closed:
+            try:
+                msg = await loop.run_in_executor(
+                    None, lambda: self.consumer.receive(timeout_millis=250))
+            except Exception:  # noqa: BLE001  (pulsar Timeout)
+                continue
+            return (msg.topic_name(), msg.data(),
+                    _PulsarAck(self.consumer, msg))
+        raise EOFError_("pulsar input closed")
+
+    def produce(self, topic: str, payload: bytes) -> None:
+        prod = self._producers.get(topic)
+        if prod is None:
+            prod = self.client.create_producer(topic)
+            self._producers[topic] = prod
+        prod.send(payload)
+
+    def close(self) -> None:
+        self._closed = True
+        try:
+            if self.consumer is not None:
+                self.consumer.close()
+            for p in self._producers.values():
+                p.close()
+            if self.client is not None:
+                self.client.close()
+        except Exception:  # noqa: BLE001
+            pass
