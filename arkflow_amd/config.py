@@ -52,6 +52,7 @@ class StreamConfig:
     temporary: List[dict] = field(default_factory=list)
     durability: Optional[DurabilityConfig] = None
     device: Optional[str] = None  # "cuda:0" pins GPU placement; None = auto
+    dedicated_thread: bool = False  # own event loop in a thread
 
 
 @dataclass
@@ -263,4 +264,5 @@ def _parse_stream(raw: dict, index: int) -> StreamConfig:
         temporary=list(temps),
         durability=dur,
         device=raw.get("device"),
+        dedicated_thread=bool(raw.get("dedicated_thread", False)),
     )

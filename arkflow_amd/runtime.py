@@ -71,17 +71,35 @@ class EventStore:
         self._ring: deque = deque(maxlen=capacity)
         self._seq = 0
         self._subscribers: List[asyncio.Queue] = []
+        self._sub_loops: dict = {}  # id(queue) → owning loop
 
     def push(self, stream_id: str, kind: str, message: str = "") -> ControlEvent:
         self._seq += 1
         ev = ControlEvent(self._seq, stream_id, kind, message)
         self._ring.append(ev)
         for q in list(self._subscribers):
+            # dedicated-thread streams push from foreign threads; asyncio
+            # queues are loop-affine, so route through the subscriber's loop
+            loop = self._sub_loops.get(id(q))
             try:
-                q.put_nowait(ev)
-            except asyncio.QueueFull:
-                pass
+                running = asyncio.get_running_loop()
+            except RuntimeError:
+                running = None
+            try:
+                if loop is not None and loop is not running:
+                    loop.call_soon_threadsafe(self._safe_put, q, ev)
+                else:
+                    self._safe_put(q, ev)
+            except RuntimeError:
+                pass  # subscriber loop already closed
         return ev
+
+    @staticmethod
+    def _safe_put(q: asyncio.Queue, ev) -> None:
+        try:
+            q.put_nowait(ev)
+        except asyncio.QueueFull:
+            pass
 
     def list(self, after_seq: int = 0, limit: int = 100) -> List[ControlEvent]:
         return [e for e in self._ring if e.seq > after_seq][:limit]
@@ -89,11 +107,13 @@ class EventStore:
     def subscribe(self) -> asyncio.Queue:
         q: asyncio.Queue = asyncio.Queue(maxsize=EVENT_RING_CAPACITY)
         self._subscribers.append(q)
+        self._sub_loops[id(q)] = asyncio.get_event_loop()
         return q
 
     def unsubscribe(self, q: asyncio.Queue) -> None:
         if q in self._subscribers:
             self._subscribers.remove(q)
+        self._sub_loops.pop(id(q), None)
 
 
 class Operation:
@@ -161,6 +181,8 @@ class RuntimeEntry:
         self.metrics = RuntimeMetrics()
         self.task: Optional[asyncio.Task] = None
         self.cancel: Optional[asyncio.Event] = None
+        self.thread = None            # dedicated-thread mode
+        self.thread_loop = None
         self.last_error: Optional[str] = None
         self.failure_class: Optional[FailureClass] = None
         self.stream = None  # live Stream while running (embedding/test access)
@@ -237,9 +259,32 @@ class RuntimeManager:
             raise
         entry.metrics = stream.metrics
         entry.stream = stream
-        entry.cancel = asyncio.Event()
-        entry.task = asyncio.ensure_future(
-            self._supervise(entry, stream))
+        if getattr(entry.config, "dedicated_thread", False):
+            # own event loop in a thread: N streams scale past one loop's
+            # Python ceiling (kernels release the GIL); cancel crosses via
+            # call_soon_threadsafe, stop() awaits the thread join
+            import threading
+            loop = asyncio.new_event_loop()
+            entry.thread_loop = loop
+            entry.cancel = asyncio.Event()
+
+            def runner():
+                asyncio.set_event_loop(loop)
+                try:
+                    loop.run_until_complete(self._supervise(entry, stream))
+                finally:
+                    loop.close()
+
+            entry.thread = threading.Thread(
+                target=runner, name=f"stream-{stream_id}", daemon=True)
+            entry.thread.start()
+            entry.task = asyncio.ensure_future(
+                asyncio.get_running_loop().run_in_executor(
+                    None, entry.thread.join))
+        else:
+            entry.cancel = asyncio.Event()
+            entry.task = asyncio.ensure_future(
+                self._supervise(entry, stream))
         entry.state = StreamState.RUNNING
         entry.observed_generation = entry.desired_generation
         self.events.push(stream_id, "running")
@@ -275,7 +320,13 @@ class RuntimeManager:
             return
         entry.state = StreamState.STOPPING
         self.events.push(stream_id, "stopping")
-        entry.cancel.set()
+        if entry.thread_loop is not None:
+            try:
+                entry.thread_loop.call_soon_threadsafe(entry.cancel.set)
+            except RuntimeError:
+                pass  # loop already closed
+        else:
+            entry.cancel.set()
         try:
             await asyncio.wait_for(asyncio.shield(entry.task), timeout)
         except asyncio.TimeoutError:

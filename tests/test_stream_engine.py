@@ -514,3 +514,43 @@ def test_direct_mode_passthrough(run):
     assert stream.metrics.input_messages == 64
     assert stream.metrics.output_messages == 64
     assert sum(b.num_rows for b in stream.output.batches) == 64
+
+
+def test_dedicated_thread_stream_lifecycle(run):
+    """A dedicated_thread stream runs in its own event loop thread;
+    start/stop/metrics behave identically to the in-loop path."""
+    import arkflow_amd as af
+    from arkflow_amd.config import EngineConfig
+
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "th",
+        "dedicated_thread": True,
+        "input": {"type": "generate", "batch_size": 16, "interval": "5ms",
+                  "fields": {"v": {"dtype": "float32"}}},
+        "pipeline": {"thread_num": 1, "processors": [
+            {"type": "sql", "query": "SELECT * FROM flow WHERE v >= 0"}]},
+        "output": {"type": "drop"},
+    }]})
+    eng = af.Engine(cfg)
+    for sc in cfg.streams:
+        eng.runtime.register(sc)
+
+    async def main():
+        import asyncio
+        await eng.runtime.start("th")
+        entry = eng.runtime.get("th")
+        assert entry.thread is not None and entry.thread.is_alive()
+        for _ in range(100):
+            if entry.metrics.output_messages > 0:
+                break
+            await asyncio.sleep(0.05)
+        assert entry.metrics.output_messages > 0
+        await eng.runtime.stop("th")
+        assert entry.state.value == "stopped"
+        assert not entry.thread.is_alive()
+        # restartable
+        await eng.runtime.start("th")
+        await asyncio.sleep(0.2)
+        await eng.runtime.stop_all()
+
+    run(main(), timeout=60)

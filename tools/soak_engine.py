@@ -14,6 +14,7 @@ from arkflow_amd.config import EngineConfig
 
 SECONDS = float(sys.argv[1]) if len(sys.argv) > 1 else 20.0
 NSTREAMS = int(sys.argv[2]) if len(sys.argv) > 2 else 1
+THREADED = len(sys.argv) > 3 and sys.argv[3] == "threads"
 
 
 async def main():
@@ -21,6 +22,7 @@ async def main():
     for s in range(NSTREAMS):
         streams.append({
             "id": f"soak{s}",
+            "dedicated_thread": THREADED,
             "input": {"type": "generate", "batch_size": 8192,
                       "interval": "0ms",
                       "fields": {
@@ -55,7 +57,7 @@ async def main():
     fused = [type(e.stream.input).__name__
              for e in eng.runtime.entries.values()]
     await eng.runtime.stop_all()
-    print(f"streams={NSTREAMS} fused={fused[0]} "
+    print(f"streams={NSTREAMS} threaded={THREADED} fused={fused[0]} "
           f"rows_in/s={total_in/elapsed/1e6:.1f}M "
           f"rows_out/s={total_out/elapsed/1e6:.1f}M errors={errs}")
 
