@@ -43,9 +43,18 @@ async def main():
     for sc in cfg.streams:
         eng.runtime.register(sc)
     await eng.runtime.start_all()
+    import faulthandler
     import time
+    faulthandler.dump_traceback_later(SECONDS + 30, exit=True)
     t0 = time.perf_counter()
-    await asyncio.sleep(SECONDS)
+    left = SECONDS
+    while left > 0:
+        await asyncio.sleep(min(5, left))
+        left -= 5
+        tot = sum(e.stream.metrics.input_messages
+                  for e in eng.runtime.entries.values())
+        print(f"  t={time.perf_counter()-t0:.0f}s rows={tot/1e6:.1f}M",
+              flush=True)
     elapsed = time.perf_counter() - t0
     total_in = sum(e.stream.metrics.input_messages
                    for e in eng.runtime.entries.values())
@@ -56,7 +65,8 @@ async def main():
                for e in eng.runtime.entries.values())
     fused = [type(e.stream.input).__name__
              for e in eng.runtime.entries.values()]
-    await eng.runtime.stop_all()
+    print("stopping...", flush=True)
+    await asyncio.wait_for(eng.runtime.stop_all(), 60)
     print(f"streams={NSTREAMS} threaded={THREADED} fused={fused[0]} "
           f"rows_in/s={total_in/elapsed/1e6:.1f}M "
           f"rows_out/s={total_out/elapsed/1e6:.1f}M errors={errs}")
