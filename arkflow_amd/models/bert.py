@@ -93,7 +93,7 @@ class BertEncoder:
                     self._forward_eager(static_ids)
             torch.cuda.current_stream().wait_stream(s)
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
+            with torch.cuda.graph(graph, capture_error_mode="thread_local"):
                 static_logits = self._forward_eager(static_ids)
             entry = (graph, static_ids, static_logits)
             self._graphs[key] = entry

@@ -81,7 +81,7 @@ class MlpAnomalyDetector:
                     self._net(static_in)
             torch.cuda.current_stream().wait_stream(s)
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
+            with torch.cuda.graph(graph, capture_error_mode="thread_local"):
                 static_out = self._net(static_in)
             entry = (graph, static_in, static_out)
             self._graphs[cap] = entry

@@ -135,7 +135,7 @@ class FusedGenerateFilterInfer:
                 self._body()
         torch.cuda.current_stream().wait_stream(s)
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
             self._scores = self._body()
 
     # ------------------------------------------------------------------ step
@@ -198,7 +198,7 @@ class FusedProtoMlp:
         if int(out[2].item()) != 0:
             raise RuntimeError("proto decode error in fused warmup")
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
             self._out_i, self._out_f, self._err, self._scores = self._body()
 
     def step(self) -> MessageBatch:
