@@ -1,6 +1,7 @@
 """`modbus` input (reference input/modbus.rs — tokio-modbus TCP register
-reads). Offline env: a `driver: memory` register file supports tests; real
-TCP activates when pymodbus is importable."""
+reads). Offline env: a `driver: memory` register file supports tests; the
+real driver polls holding registers over TCP via pymodbus (3.x API with a
+2.x fallback), env-gated like the other broker clients."""
 from __future__ import annotations
 
 import asyncio
@@ -27,18 +28,43 @@ class ModbusInput(Input):
         self.count_regs = int(config.get("register_count", 8))
         self.interval = float(config.get("interval_secs", 1.0))
         self.count = config.get("count")
+        self.unit_id = int(config.get("unit_id", 1))  # modbus slave id
         self.driver = config.get("driver") or (
             "memory" if str(self.address).startswith("memory://") else "real")
         self._reads = 0
+        self._client = None
 
     async def connect(self) -> None:
         if self.driver == "memory":
             return
         try:
-            import pymodbus  # type: ignore  # noqa: F401
-        except ImportError as e:
-            raise ConnectionError_(
-                "no modbus client library; use driver: memory") from e
+            from pymodbus.client import ModbusTcpClient  # type: ignore
+        except ImportError:
+            try:  # pymodbus 2.x layout
+                from pymodbus.client.sync import (  # type: ignore
+                    ModbusTcpClient)
+            except ImportError as e:
+                raise ConnectionError_(
+                    "modbus real driver requires pymodbus; use "
+                    "driver: memory") from e
+        host, _, port = str(self.address).partition(":")
+        self._client = ModbusTcpClient(host, port=int(port or 502))
+        ok = await asyncio.get_running_loop().run_in_executor(
+            None, self._client.connect)
+        if not ok:
+            raise ConnectionError_(f"modbus connect failed: {self.address}")
+
+    def _poll_real(self) -> List[int]:
+        try:  # pymodbus 3.x keyword; 2.x uses `unit`
+            rr = self._client.read_holding_registers(
+                self.start, count=self.count_regs, slave=self.unit_id)
+        except TypeError:
+            rr = self._client.read_holding_registers(
+                self.start, self.count_regs, unit=self.unit_id)
+        if rr.isError():
+            from ..errors import DisconnectionError
+            raise DisconnectionError(f"modbus read error: {rr}")
+        return list(rr.registers)
 
     async def read(self) -> Tuple[MessageBatch, Ack]:
         if self.count is not None and self._reads >= int(self.count):
@@ -46,9 +72,14 @@ class ModbusInput(Input):
         if self._reads > 0:
             await asyncio.sleep(self.interval)
         self._reads += 1
-        unit = str(self.address).removeprefix("memory://")
-        regs = _MEMORY_REGISTERS.get(unit, [0] * (self.start + self.count_regs))
-        window = regs[self.start:self.start + self.count_regs]
+        if self._client is not None:
+            window = await asyncio.get_running_loop().run_in_executor(
+                None, self._poll_real)
+        else:
+            unit = str(self.address).removeprefix("memory://")
+            regs = _MEMORY_REGISTERS.get(
+                unit, [0] * (self.start + self.count_regs))
+            window = regs[self.start:self.start + self.count_regs]
         batch = MessageBatch({
             "register": Column.from_numeric(
                 list(range(self.start, self.start + len(window)))),
@@ -57,6 +88,13 @@ class ModbusInput(Input):
                 [time.time()] * len(window)),
         }, input_name="modbus")
         return batch, NoopAck()
+
+    async def close(self) -> None:
+        if self._client is not None:
+            try:
+                self._client.close()
+            except Exception:  # noqa: BLE001
+                pass
 
 
 @register("input", "modbus",
