@@ -3,8 +3,10 @@
 Mirrors reference crates/arkflow-plugin/src/{input,output}/sql.rs: input runs
 a query against MySQL/Postgres/SQLite/DuckDB and streams the result as
 batches; output does batched INSERT with optional UPSERT. SQLite is fully
-native here (stdlib); other engines activate when their client library is
-importable.
+native (stdlib); duckdb, mysql (pymysql) and postgres (psycopg2) are real
+DB-API drivers that activate when the client library is importable —
+engine-correct placeholders (%s vs ?) and upsert dialects (ON CONFLICT vs
+ON DUPLICATE KEY UPDATE) included.
 """
 from __future__ import annotations
 
@@ -15,6 +17,34 @@ from ..batch import Column, DEFAULT_RECORD_BATCH, MessageBatch
 from ..errors import ConfigError, ConnectionError_, EOFError_
 from ..registry import register
 from ..spi import Ack, Input, NoopAck, Output
+
+
+class _DbApi:
+    """Adapter giving pymysql/psycopg2 connections the conn.execute shape
+    the sqlite path uses; %s paramstyle; explicit commit."""
+
+    paramstyle = "%s"
+
+    def __init__(self, conn, dialect: str):
+        self._c = conn
+        self.dialect = dialect
+
+    def execute(self, sql, params=None):
+        cur = self._c.cursor()
+        cur.execute(sql, params or ())
+        self._c.commit()
+        return cur
+
+    def executemany(self, sql, data):
+        cur = self._c.cursor()
+        cur.executemany(sql, data)
+        return cur
+
+    def commit(self):
+        self._c.commit()
+
+    def close(self):
+        self._c.close()
 
 
 def _connect(config: dict):
@@ -30,8 +60,35 @@ def _connect(config: dict):
         except ImportError as e:
             raise ConnectionError_("duckdb not installed") from e
         return duckdb.connect(config.get("path", ":memory:"))
-    raise ConnectionError_(
-        f"no {engine} client library in this environment (sqlite/duckdb only)")
+    if engine == "mysql":
+        try:
+            import pymysql  # type: ignore
+        except ImportError as e:
+            raise ConnectionError_("mysql engine requires pymysql") from e
+        return _DbApi(pymysql.connect(
+            host=config.get("host", "127.0.0.1"),
+            port=int(config.get("port", 3306)),
+            user=config.get("user", "root"),
+            password=config.get("password", ""),
+            database=config.get("database")), "mysql")
+    if engine in ("postgres", "postgresql"):
+        try:
+            import psycopg2  # type: ignore
+        except ImportError as e:
+            raise ConnectionError_(
+                "postgres engine requires psycopg2") from e
+        if config.get("dsn"):
+            conn = psycopg2.connect(config["dsn"])
+        else:
+            conn = psycopg2.connect(
+                host=config.get("host", "127.0.0.1"),
+                port=int(config.get("port", 5432)),
+                user=config.get("user", "postgres"),
+                password=config.get("password", ""),
+                dbname=config.get("database"))
+        return _DbApi(conn, "postgres")
+    raise ConnectionError_(f"unknown sql engine {engine!r} "
+                           "(sqlite|duckdb|mysql|postgres)")
 
 
 class SqlInput(Input):
@@ -91,22 +148,38 @@ class SqlOutput(Output):
             return
         names = batch.column_names
         if self.create:
-            cols_sql = ", ".join(f'"{n}"' for n in names)
+            qd = "`" if getattr(self._conn, "dialect", "") == "mysql" \
+                else '"'
+            if getattr(self._conn, "dialect", "") in ("mysql", "postgres"):
+                cols_sql = ", ".join(f"{qd}{n}{qd} TEXT" for n in names)
+            else:
+                cols_sql = ", ".join(f"{qd}{n}{qd}" for n in names)
             if self.upsert_keys:
-                keys = ", ".join(f'"{k}"' for k in self.upsert_keys)
+                keys = ", ".join(f"{qd}{k}{qd}" for k in self.upsert_keys)
                 cols_sql += f", PRIMARY KEY ({keys})"
             self._conn.execute(
-                f'CREATE TABLE IF NOT EXISTS "{self.table}" ({cols_sql})')
+                f"CREATE TABLE IF NOT EXISTS {qd}{self.table}{qd} "
+                f"({cols_sql})")
             self.create = False
-        placeholders = ", ".join("?" for _ in names)
-        cols_sql = ", ".join(f'"{n}"' for n in names)
-        stmt = f'INSERT INTO "{self.table}" ({cols_sql}) VALUES ({placeholders})'
+        ph = getattr(self._conn, "paramstyle", "?")
+        dialect = getattr(self._conn, "dialect", "sqlite")
+        q = "`" if dialect == "mysql" else '"'
+        placeholders = ", ".join(ph for _ in names)
+        cols_sql = ", ".join(f"{q}{n}{q}" for n in names)
+        stmt = (f"INSERT INTO {q}{self.table}{q} ({cols_sql}) "
+                f"VALUES ({placeholders})")
         if self.upsert_keys:
-            keys = ", ".join(f'"{k}"' for k in self.upsert_keys)
-            updates = ", ".join(
-                f'"{n}"=excluded."{n}"' for n in names
-                if n not in self.upsert_keys)
-            stmt += f" ON CONFLICT({keys}) DO UPDATE SET {updates}"
+            if dialect == "mysql":
+                updates = ", ".join(
+                    f"{q}{n}{q}=VALUES({q}{n}{q})" for n in names
+                    if n not in self.upsert_keys)
+                stmt += f" ON DUPLICATE KEY UPDATE {updates}"
+            else:
+                keys = ", ".join(f"{q}{k}{q}" for k in self.upsert_keys)
+                updates = ", ".join(
+                    f"{q}{n}{q}=excluded.{q}{n}{q}" for n in names
+                    if n not in self.upsert_keys)
+                stmt += f" ON CONFLICT({keys}) DO UPDATE SET {updates}"
         data = []
         for r in rows:
             data.append(tuple(
@@ -122,8 +195,8 @@ class SqlOutput(Output):
 
 
 @register("input", "sql",
-          description="Query a relational DB (sqlite/duckdb native) into "
-                      "batches",
+          description="Query a relational DB (sqlite/duckdb native; "
+                      "mysql/postgres via pymysql/psycopg2) into batches",
           example={"type": "sql", "engine": "sqlite", "path": "db.sqlite",
                    "query": "SELECT * FROM t"})
 def _build_sql_in(config, resource=None):
