@@ -127,7 +127,7 @@ void launch_json_decode(const uint8_t*, const int64_t*, int64_t, int,
                         const char*, const int*, const int*, const int*,
                         const int*, int, const char*, const int*,
                         double*, int64_t*, int64_t*, int32_t*, uint8_t*,
-                        int32_t*, int32_t*, hipStream_t);
+                        int32_t*, int32_t*, int, hipStream_t);
 void launch_json_copy_strings(const uint8_t*, const int64_t*, const int64_t*,
                               const uint8_t*, int64_t, int64_t, uint8_t*,
                               hipStream_t);
@@ -990,7 +990,11 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
                        str_start.data_ptr<int64_t>(),
                        str_ulen.data_ptr<int32_t>(),
                        found.data_ptr<uint8_t>(), err.data_ptr<int32_t>(),
-                       found_count.data_ptr<int32_t>(), cur_stream());
+                       found_count.data_ptr<int32_t>(),
+                       // wave-per-doc parse pays off once docs are long
+                       // enough for 512 B/step cooperative scans
+                       (int)(n > 0 && data.numel() / n >= 192),
+                       cur_stream());
   // string copy-out: per field, offsets = exclusive cumsum of unescaped
   // lengths; one host sync for ALL totals at once, then one copy kernel per
   // string field (ulen of absent docs is 0 → empty strings, validity=found)

@@ -261,6 +261,122 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
                                int32_t* __restrict__ err,
                                int* __restrict__ fcnt);
 
+// ======================= wave-cooperative parse ==============================
+// For long documents the thread-per-doc walk is serialization-bound: even
+// with SWAR it advances 8 B per dependent step. Here ONE WAVE owns a doc
+// and scans 512 B per step: every lane SWAR-checks its own 8 B chunk for
+// '"'/'\\', a ballot finds the first hit, and the (wave-uniform) state
+// machine jumps there. Only lane 0 writes outputs. Dispatched when the
+// batch's mean document length clears a threshold (launcher below);
+// outputs are bit-identical to the thread-per-doc kernel.
+
+// first position ≥ p whose byte is '"' or '\\' (or end)
+DEV_INLINE int64_t wave_find_qb(const uint8_t* __restrict__ d, int64_t p,
+                                int64_t end, int lane) {
+  while (p < end) {
+    int64_t base = p + (int64_t)lane * 8;
+    uint64_t hits = 0;
+    if (base < end) {
+      if (base + 8 <= end) {
+        uint64_t x = load8(d + base);
+        hits = haszero64(x ^ 0x2222222222222222ull) |
+               haszero64(x ^ 0x5C5C5C5C5C5C5C5Cull);
+      } else {
+        for (int i = 0; base + i < end; ++i) {
+          uint8_t c = d[base + i];
+          if (c == '"' || c == '\\') hits |= 0x80ull << (8 * i);
+        }
+      }
+    }
+    uint64_t lane_has = __ballot(hits != 0);
+    if (lane_has) {
+      int src = __ffsll((unsigned long long)lane_has) - 1;
+      uint64_t h = __shfl((unsigned long long)hits, src, 64);
+      int off = __builtin_ctzll(h) >> 3;
+      return p + (int64_t)src * 8 + off;
+    }
+    p += 64 * 8;
+  }
+  return end;
+}
+
+// wave version of scan_string: returns closing-quote position, accumulates
+// the unescaped length exactly like scan_string (escape handling runs
+// wave-uniform — every lane computes the same values)
+DEV_INLINE int64_t wave_scan_string(const uint8_t* __restrict__ d, int64_t p,
+                                    int64_t end, int lane, int32_t* ulen) {
+  int32_t u = 0;
+  while (p < end) {
+    int64_t q = wave_find_qb(d, p, end, lane);
+    u += (int32_t)(q - p);
+    p = q;
+    if (p >= end || d[p] == '"') break;
+    // backslash escape — mirror scan_string byte for byte
+    if (p + 1 < end) {
+      uint8_t e = d[p + 1];
+      if (e == 'u' && p + 5 < end) {
+        uint32_t cp = hex4(d, p + 2);
+        p += 6;
+        if (cp >= 0xD800 && cp <= 0xDBFF && p + 5 < end && d[p] == '\\' &&
+            d[p + 1] == 'u') {
+          uint32_t lo = hex4(d, p + 2);
+          if (lo >= 0xDC00 && lo <= 0xDFFF) {
+            p += 6;
+            u += 4;
+            continue;
+          }
+        }
+        u += cp < 0x80 ? 1 : (cp < 0x800 ? 2 : 3);
+        continue;
+      }
+      p += 2;
+      u += 1;
+      continue;
+    }
+    ++p;
+    ++u;
+  }
+  *ulen = u;
+  return p;
+}
+
+// wave version of skip_value (string bodies scanned 512 B/step)
+DEV_INLINE int64_t wave_skip_value(const uint8_t* __restrict__ d, int64_t p,
+                                   int64_t end, int lane) {
+  while (p < end && is_ws(d[p])) ++p;
+  if (p >= end) return p;
+  uint8_t c = d[p];
+  if (c == '"') {
+    int32_t dummy;
+    p = wave_scan_string(d, p + 1, end, lane, &dummy);
+    return p < end ? p + 1 : p;
+  }
+  if (c == '{' || c == '[') {
+    uint8_t open = c, close = (c == '{') ? '}' : ']';
+    int depth = 0;
+    while (p < end) {
+      uint8_t x = d[p];
+      if (x == '"') {
+        int32_t dummy;
+        p = wave_scan_string(d, p + 1, end, lane, &dummy);
+        if (p < end) ++p;
+        continue;
+      }
+      if (x == open) ++depth;
+      else if (x == close) {
+        --depth;
+        if (depth == 0) return p + 1;
+      }
+      ++p;
+    }
+    return p;
+  }
+  while (p < end && d[p] != ',' && d[p] != '}' && d[p] != ']' &&
+         !is_ws(d[p]))
+    ++p;
+  return p;
+}
+
 // LDS doc cache: each 128-doc tile's byte range is CONTIGUOUS in the binary
 // column, so the block stages it with coalesced loads and threads then walk
 // their documents out of LDS — the thread-per-doc global byte walk was
@@ -422,6 +538,146 @@ DEV_INLINE void json_parse_doc(const uint8_t* __restrict__ data,
   }
 }
 
+// wave-cooperative twin of json_parse_doc: identical state machine, but
+// string scans advance 512 B/step via wave_scan_string / wave_skip_value
+// and only lane 0 writes. Every branch depends on wave-uniform values.
+DEV_INLINE void json_parse_doc_wave(const uint8_t* __restrict__ data,
+                                    int64_t p, int64_t end, int lane,
+                                    int64_t i, int64_t n_docs,
+                                    const JsonSpec& spec,
+                                    double* __restrict__ out_f64,
+                                    int64_t* __restrict__ out_i64,
+                                    int64_t* __restrict__ str_start,
+                                    int32_t* __restrict__ str_ulen,
+                                    uint8_t* __restrict__ found,
+                                    int32_t* __restrict__ err,
+                                    int* __restrict__ fcnt) {
+  while (p < end && is_ws(data[p])) ++p;
+  if (p >= end || data[p] != '{') {
+    if (p < end && lane == 0) err[0] = 1;
+    return;
+  }
+  ++p;
+  int ctx = -1;
+  for (;;) {
+    while (p < end && (is_ws(data[p]) || data[p] == ',')) ++p;
+    if (p >= end) break;
+    if (data[p] == '}') {
+      if (ctx < 0) break;
+      ctx = -1;
+      ++p;
+      continue;
+    }
+    if (data[p] != '"') {
+      if (lane == 0) err[0] = 1;
+      break;
+    }
+    int64_t k0 = ++p;
+    while (p < end && data[p] != '"') {
+      if (data[p] == '\\') ++p;
+      ++p;
+    }
+    int klen = (int)(p - k0);
+    ++p;
+    while (p < end && is_ws(data[p])) ++p;
+    if (p < end && data[p] == ':') ++p;
+    while (p < end && is_ws(data[p])) ++p;
+    int fi = -1;
+#pragma unroll
+    for (int f = 0; f < JSON_MAX_FIELDS; ++f) {
+      if (f < spec.nf && spec.name_len[f] == klen
+          && spec.parent[f] == ctx) {
+        bool eq = true;
+        for (int c = 0; c < klen; ++c)
+          if (spec.names[f][c] != (char)data[k0 + c]) { eq = false; break; }
+        if (eq) { fi = f; break; }
+      }
+    }
+    if (fi < 0) {
+      if (ctx < 0 && p < end && data[p] == '{') {
+        int pi = -1;
+        for (int q = 0; q < spec.np; ++q) {
+          if (spec.parent_len[q] == klen) {
+            bool eq = true;
+            for (int c = 0; c < klen; ++c)
+              if (spec.parents[q][c] != (char)data[k0 + c]) {
+                eq = false;
+                break;
+              }
+            if (eq) { pi = q; break; }
+          }
+        }
+        if (pi >= 0) {
+          ctx = pi;
+          ++p;
+          continue;
+        }
+      }
+      p = wave_skip_value(data, p, end, lane);
+      continue;
+    }
+    uint8_t c = p < end ? data[p] : 0;
+    if (spec.kind[fi] == 2) {
+      if (c != '"') {
+        p = wave_skip_value(data, p, end, lane);
+        continue;
+      }
+      int32_t ulen = 0;
+      int64_t s0 = p + 1;
+      p = wave_scan_string(data, s0, end, lane, &ulen);
+      if (p < end) ++p;
+      if (lane == 0) {
+        str_start[(int64_t)spec.slot[fi] * n_docs + i] = s0;
+        str_ulen[(int64_t)spec.slot[fi] * n_docs + i] = ulen;
+        found[(int64_t)fi * n_docs + i] = 1;
+        atomicAdd(&fcnt[fi], 1);
+      }
+      continue;
+    }
+    double v = 0.0;
+    bool is_int = true;
+    if (c == 't') { v = 1.0; p += 4; }
+    else if (c == 'f') { v = 0.0; p += 5; }
+    else if (c == 'n') { p += 4; continue; }
+    else if (c == '"') { p = wave_skip_value(data, p, end, lane); continue; }
+    else p = parse_number(data, p, end, &v, &is_int);
+    if (lane == 0) {
+      if (spec.kind[fi] == 1)
+        out_f64[(int64_t)spec.slot[fi] * n_docs + i] = v;
+      else
+        out_i64[(int64_t)spec.slot[fi] * n_docs + i] = (int64_t)v;
+      found[(int64_t)fi * n_docs + i] = 1;
+      atomicAdd(&fcnt[fi], 1);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void json_decode_wave_kernel(const uint8_t* __restrict__ data,
+                             const int64_t* __restrict__ offsets,
+                             int64_t n_docs, JsonSpec spec,
+                             double* __restrict__ out_f64,
+                             int64_t* __restrict__ out_i64,
+                             int64_t* __restrict__ str_start,
+                             int32_t* __restrict__ str_ulen,
+                             uint8_t* __restrict__ found,
+                             int32_t* __restrict__ err,
+                             int32_t* __restrict__ found_count) {
+  __shared__ int fcnt[JSON_MAX_FIELDS];
+  if (threadIdx.x < JSON_MAX_FIELDS) fcnt[threadIdx.x] = 0;
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  int64_t wstride = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t i = wave; i < n_docs; i += wstride)
+    json_parse_doc_wave(data, offsets[i], offsets[i + 1], lane, i, n_docs,
+                        spec, out_f64, out_i64, str_start, str_ulen, found,
+                        err, fcnt);
+  __syncthreads();
+  if (threadIdx.x < JSON_MAX_FIELDS && fcnt[threadIdx.x])
+    atomicAdd(&found_count[threadIdx.x], fcnt[threadIdx.x]);
+}
+
 // copy-out pass for ONE string field: thread per doc, unescaping into the
 // contiguous output at out_offs[i] (out_offs = exclusive cumsum of ulen)
 __global__ void json_copy_strings_kernel(const uint8_t* __restrict__ data,
@@ -447,7 +703,7 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
                                    int64_t* out_i64, int64_t* str_start,
                                    int32_t* str_ulen, uint8_t* found,
                                    int32_t* err, int32_t* found_count,
-                                   hipStream_t st) {
+                                   int use_wave, hipStream_t st) {
   JsonSpec spec{};
   spec.nf = nf > JSON_MAX_FIELDS ? JSON_MAX_FIELDS : nf;
   for (int f = 0; f < spec.nf; ++f) {
@@ -463,6 +719,17 @@ extern "C" void launch_json_decode(const uint8_t* data, const int64_t* offsets,
     spec.parent_len[q] = parent_len[q];
     for (int c = 0; c < parent_len[q] && c < JSON_MAX_NAME; ++c)
       spec.parents[q][c] = parents[q * JSON_MAX_NAME + c];
+  }
+  if (use_wave) {
+    // wave-per-doc: 4 waves/block, grid-stride
+    int grid = (int)((n_docs + 3) / 4);
+    if (grid > 8192) grid = 8192;
+    if (grid < 1) return;
+    json_decode_wave_kernel<<<grid, 256, 0, st>>>(data, offsets, n_docs,
+                                                  spec, out_f64, out_i64,
+                                                  str_start, str_ulen,
+                                                  found, err, found_count);
+    return;
   }
   int grid = (int)((n_docs + JSON_BLOCK - 1) / JSON_BLOCK);
   if (grid > 4096) grid = 4096;

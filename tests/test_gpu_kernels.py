@@ -1079,3 +1079,38 @@ def test_attention_flash_long_seq(nat, dev):
         ref2 = (p @ v)
         assert (out2.float() - ref2.reshape(out2.shape)).abs().max(
             ).item() < 0.02, S
+
+
+def test_json_decode_wave_path_matches_host(nat, dev):
+    """Long docs (avg ≥192 B) dispatch the wave-per-doc parse; outputs must
+    match the host reference byte for byte (incl. escapes + \\uXXXX)."""
+    import asyncio
+    import json as _json
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.json_proc import JsonToArrowProcessor
+
+    docs = []
+    for i in range(3000):
+        body = ("lorem ipsum dolor sit amet " * 9) + f"#{i}"
+        if i % 7 == 0:
+            body += ' quote:" backslash:\\ tab:\t unicode:é中'
+        docs.append({"id": i, "w": i * 0.25, "ok": i % 3 == 0,
+                     "body": body, "tag": f"t{i % 5}"})
+    payloads = [_json.dumps(d).encode() for d in docs]
+    assert sum(map(len, payloads)) / len(payloads) >= 192  # wave dispatch
+    schema = {"id": "int", "w": "float", "ok": "bool",
+              "body": "str", "tag": "str"}
+    loop = asyncio.new_event_loop()
+    gpu = loop.run_until_complete(JsonToArrowProcessor(
+        {"schema": schema}, None).process(
+        MessageBatch.from_binary(payloads).to(dev)))[0]
+    host = loop.run_until_complete(JsonToArrowProcessor(
+        {"schema": schema}, None).process(
+        MessageBatch.from_binary(payloads)))[0]
+    assert gpu.column("id").to_pylist() == host.column("id").to_pylist()
+    assert torch.allclose(gpu.column("w").data.cpu().float(),
+                          host.column("w").data.float())
+    assert gpu.column("ok").to_pylist() == host.column("ok").to_pylist()
+    assert gpu.column("body").to_pylist() == host.column("body").to_pylist()
+    assert gpu.column("tag").to_pylist() == host.column("tag").to_pylist()
