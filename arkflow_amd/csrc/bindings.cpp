@@ -991,9 +991,11 @@ json_decode(torch::Tensor data, torch::Tensor offsets,
                        str_ulen.data_ptr<int32_t>(),
                        found.data_ptr<uint8_t>(), err.data_ptr<int32_t>(),
                        found_count.data_ptr<int32_t>(),
-                       // wave-per-doc parse pays off once docs are long
-                       // enough for 512 B/step cooperative scans
-                       (int)(n > 0 && data.numel() / n >= 192),
+                       // wave-per-doc parse pays off only when string
+                       // scans dwarf the per-doc state machine (which the
+                       // thread-per-doc kernel amortizes 64 docs/wave):
+                       // measured crossover is in the multi-KB range
+                       (int)(n > 0 && data.numel() / n >= 2048),
                        cur_stream());
   // string copy-out: per field, offsets = exclusive cumsum of unescaped
   // lengths; one host sync for ALL totals at once, then one copy kernel per

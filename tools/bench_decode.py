@@ -62,6 +62,20 @@ pbatch = MessageBatch.from_binary(payloads).to(dev)
 run("proto scalars+strings", ProtobufToArrowProcessor({"proto": proto}, None),
     pbatch)
 
+# very-long-string case (wave-per-doc PARSE path: ~4 KB bodies)
+N_XL = 16_384
+xl_payloads = [json.dumps({
+    "id": i,
+    "body": ("abcdefgh " * 450) + str(i),
+}).encode() for i in range(N_XL)]
+xbatch = MessageBatch.from_binary(xl_payloads).to(dev)
+_saveN2 = N
+N = N_XL
+run("json 4KB strings (wave parse)",
+    JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
+    xbatch)
+N = _saveN2
+
 # long-string case (wave-per-doc copy path; VERDICT #9 target ≥300 GB/s)
 N_LONG = 65_536
 long_payloads = [json.dumps({
