@@ -1091,14 +1091,14 @@ def test_json_decode_wave_path_matches_host(nat, dev):
     from arkflow_amd.processors.json_proc import JsonToArrowProcessor
 
     docs = []
-    for i in range(3000):
-        body = ("lorem ipsum dolor sit amet " * 9) + f"#{i}"
+    for i in range(1200):
+        body = ("lorem ipsum dolor sit amet " * 100) + f"#{i}"
         if i % 7 == 0:
             body += ' quote:" backslash:\\ tab:\t unicode:é中'
         docs.append({"id": i, "w": i * 0.25, "ok": i % 3 == 0,
                      "body": body, "tag": f"t{i % 5}"})
     payloads = [_json.dumps(d).encode() for d in docs]
-    assert sum(map(len, payloads)) / len(payloads) >= 192  # wave dispatch
+    assert sum(map(len, payloads)) / len(payloads) >= 2048  # wave dispatch
     schema = {"id": "int", "w": "float", "ok": "bool",
               "body": "str", "tag": "str"}
     loop = asyncio.new_event_loop()
