@@ -78,6 +78,9 @@ N = _saveN2
 
 # long-string case (wave-per-doc copy path; VERDICT #9 target ≥300 GB/s)
 N_LONG = 65_536
+# same docs at 4× the batch: fixed per-batch costs (readback, offsets,
+# Python) amortize — reported separately so both regimes are honest
+N_LONG4 = 262_144
 long_payloads = [json.dumps({
     "id": i,
     "body": ("lorem ipsum dolor sit amet " * 10) + str(i),
@@ -88,4 +91,9 @@ N = N_LONG
 run("json long strings (~280B body)",
     JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
     lbatch)
+l4 = MessageBatch.from_binary(long_payloads * 4).to(dev)
+N = N_LONG4
+run("json long strings (262K-doc batch)",
+    JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
+    l4)
 N = _saveN
