@@ -17,19 +17,17 @@ def time_ln(rows, n, iters=200, residual=False):
     res = torch.randn(rows, n, device="cuda").to(torch.bfloat16) if residual else None
     g = torch.randn(n, device="cuda")
     b = torch.randn(n, device="cuda")
-    out = torch.empty_like(x)
-    ro = torch.empty_like(x) if residual else None
     for _ in range(20):
-        nat.layernorm_bf16(x, res, g, b, out, ro, 1e-5)
+        nat.layernorm_bf16(x, g, b, 1e-5, res)
     s, e = torch.cuda.Event(True), torch.cuda.Event(True)
     torch.cuda.synchronize()
     s.record()
     for _ in range(iters):
-        nat.layernorm_bf16(x, res, g, b, out, ro, 1e-5)
+        nat.layernorm_bf16(x, g, b, 1e-5, res)
     e.record()
     torch.cuda.synchronize()
     us = s.elapsed_time(e) * 1000 / iters
-    nb = rows * n * 2 * (2 if not residual else 4)  # read x (+res), write out (+ro)
+    nb = rows * n * 2 * (2 if not residual else 3)  # read x (+res), write out
     print(f"rows={rows:7d} n={n:5d} resid={int(residual)}  "
           f"{us:8.2f} us  {nb / us / 1e3:8.1f} GB/s")
 
