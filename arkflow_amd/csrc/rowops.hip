@@ -102,19 +102,43 @@ void layernorm_bf16_wave_kernel(const __bf16* __restrict__ x,
   const int lane = threadIdx.x & 63;
   int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   int64_t wstride = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const bool cached = nv <= 2 * WAVE;  // lane holds <=2 vectors in regs
   for (int64_t row = wave; row < rows; row += wstride) {
     const vbf16x8* xr = (const vbf16x8*)(x + row * n);
     const vbf16x8* rr =
         ADD_RESIDUAL ? (const vbf16x8*)(residual + row * n) : nullptr;
+    vbf16x8* orow = (vbf16x8*)(out + row * n);
+    vbf16x8* resrow = (ADD_RESIDUAL && resid_out)
+                          ? (vbf16x8*)(resid_out + row * n)
+                          : nullptr;
     float sum = 0.f, sq = 0.f;
-    for (int i = lane; i < nv; i += WAVE) {
-      vbf16x8 v = xr[i];
+    float c0[16];  // row cache: the (x+residual) floats, read-once
+    if (cached) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = (float)v[j];
-        if (ADD_RESIDUAL) f += (float)rr[i][j];
-        sum += f;
-        sq += f * f;
+      for (int t = 0; t < 2; ++t) {
+        int i = lane + t * WAVE;
+        if (i < nv) {
+          vbf16x8 v = xr[i];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float f = (float)v[j];
+            if (ADD_RESIDUAL) f += (float)rr[i][j];
+            c0[t * 8 + j] = f;
+            sum += f;
+            sq += f * f;
+          }
+        }
+      }
+    } else {
+      for (int i = lane; i < nv; i += WAVE) {
+        vbf16x8 v = xr[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = (float)v[j];
+          if (ADD_RESIDUAL) f += (float)rr[i][j];
+          sum += f;
+          sq += f * f;
+        }
       }
     }
     sum = wave_reduce_sum(sum);
@@ -124,23 +148,38 @@ void layernorm_bf16_wave_kernel(const __bf16* __restrict__ x,
     float mean = sum / n;
     float var = sq / n - mean * mean;
     float rstd = rsqrtf(var + eps);
-    vbf16x8* orow = (vbf16x8*)(out + row * n);
-    vbf16x8* resrow = (ADD_RESIDUAL && resid_out)
-                          ? (vbf16x8*)(resid_out + row * n)
-                          : nullptr;
-    for (int i = lane; i < nv; i += WAVE) {
-      vbf16x8 v = xr[i];
-      vbf16x8 o, rsum;
+    if (cached) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = (float)v[j];
-        if (ADD_RESIDUAL) f += (float)rr[i][j];
-        if (ADD_RESIDUAL && resrow) rsum[j] = (__bf16)f;
-        int c = i * 8 + j;
-        o[j] = (__bf16)((f - mean) * rstd * gamma[c] + beta[c]);
+      for (int t = 0; t < 2; ++t) {
+        int i = lane + t * WAVE;
+        if (i < nv) {
+          vbf16x8 o, rsum;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float f = c0[t * 8 + j];
+            if (ADD_RESIDUAL && resrow) rsum[j] = (__bf16)f;
+            int c = i * 8 + j;
+            o[j] = (__bf16)((f - mean) * rstd * gamma[c] + beta[c]);
+          }
+          orow[i] = o;
+          if (ADD_RESIDUAL && resrow) resrow[i] = rsum;
+        }
       }
-      orow[i] = o;
-      if (ADD_RESIDUAL && resrow) resrow[i] = rsum;
+    } else {
+      for (int i = lane; i < nv; i += WAVE) {
+        vbf16x8 v = xr[i];
+        vbf16x8 o, rsum;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = (float)v[j];
+          if (ADD_RESIDUAL) f += (float)rr[i][j];
+          if (ADD_RESIDUAL && resrow) rsum[j] = (__bf16)f;
+          int c = i * 8 + j;
+          o[j] = (__bf16)((f - mean) * rstd * gamma[c] + beta[c]);
+        }
+        orow[i] = o;
+        if (ADD_RESIDUAL && resrow) resrow[i] = rsum;
+      }
     }
   }
 }

@@ -42,7 +42,10 @@ for M, N, K, act in SHAPES:
         ref = torch.nn.functional.gelu(ref, approximate="tanh")
     flop = 2 * M * N * K
     rows = [("blaslt", lambda: torch.nn.functional.linear(A, Bt, bias_h))]
-    for name, var in [("bk32", 0), ("k64swz", 8), ("k64dbuf", 9)]:
+    if act != 0:
+        rows.append(("blaslt+act", lambda: nat.bias_act_bf16(
+            torch.nn.functional.linear(A, Bt, bias_h), None, act)))
+    for name, var in [("bk32", 0), ("k64swz", 8), ("8ph", 2), ("8ph128", 5)]:
         rows.append((name,
                      lambda v=var: nat.gemm_bf16_variant(A, Bt, bias, act, v)))
     out = [f"M{M} N{N} K{K} act{act}:"]
