@@ -92,7 +92,10 @@ class FusedGenerateFilterInfer:
         # count lives in host-mapped pinned memory: the filter kernel writes
         # it over PCIe, so reading it back is a stream sync + CPU load — no
         # hipMemcpyDtoH launch per step (was ~10 us of a ~100 us step)
+        import os
         try:
+            if os.environ.get("ARKFLOW_MAPPED_COUNT", "1") == "0":
+                raise RuntimeError("disabled")
             self.count_host, self.count = self.nat.mapped_int32(1)
         except RuntimeError:
             self.count_host = None
