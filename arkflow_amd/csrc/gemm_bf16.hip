@@ -271,32 +271,7 @@ void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
     int xcd = bid % 8, off = bid / 8;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
   }
-  // L2-panel tile order: the XCD swizzle above hands each XCD a contiguous
-  // logical range; laying that range out as PH×PW panels (PH M-tiles × PW
-  // N-tiles) keeps one A-slab (~1.6 MB) plus one B-panel (~0.8 MB) resident
-  // in the XCD's private 4 MB L2. Row-major order instead sweeps ALL of B
-  // per M-row (>L2), re-reading B from HBM tiles_m times — at fc1-like
-  // shapes (short K) that traffic IS the kernel time.
-  int bm, bn;
-  {
-    const int tiles_m = nwg / tiles_n;
-    const int PH = (tiles_m % 8 == 0) ? 8
-                   : (tiles_m % 4 == 0) ? 4
-                   : (tiles_m % 2 == 0) ? 2 : 1;
-    const int PW = (tiles_n % 4 == 0) ? 4
-                   : (tiles_n % 3 == 0) ? 3
-                   : (tiles_n % 2 == 0) ? 2 : 1;
-    if (nwg >= 16 && PH * PW >= 4) {
-      const int per_panel = PH * PW;
-      const int panels_n = tiles_n / PW;
-      const int p = bid / per_panel, w = bid % per_panel;
-      bm = (p / panels_n) * PH + w / PW;
-      bn = (p % panels_n) * PW + w % PW;
-    } else {
-      bm = bid / tiles_n;
-      bn = bid % tiles_n;
-    }
-  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
   const int row0 = bm * BM, col0 = bn * BN;
 
   __shared__ __bf16 Asm[BM * BK2];
