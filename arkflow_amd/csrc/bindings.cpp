@@ -75,6 +75,8 @@ int launch_gemm_bf16_k64(const void*, const void*, const float*, void*, int,
                          int, int, int, hipStream_t);
 int launch_gemm_bf16_k64s(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
+int launch_gemm_bf16_k64p(const void*, const void*, const float*, void*,
+                          int, int, int, int, hipStream_t);
 int launch_gemm_bf16_k64d(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
 void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
@@ -528,6 +530,11 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
                                    C.data_ptr(), (int)M, (int)N, (int)K,
                                    (int)act, cur_stream());
     TORCH_CHECK(rc == 0, "K must be a multiple of 64 for the BK=64 kernel");
+  } else if (variant == 10) {
+    TORCH_CHECK(launch_gemm_bf16_k64p(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                      C.data_ptr(), (int)M, (int)N, (int)K,
+                                      (int)act, cur_stream()) == 0,
+                "k64p requires K % 64 == 0");
   } else if (variant == 9) {
     int rc = launch_gemm_bf16_k64d(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                    C.data_ptr(), (int)M, (int)N, (int)K,

@@ -362,6 +362,129 @@ void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- register-prefetch swizzled BK=64 ----------------------------------------
+// k64s with the NEXT k-slice's global loads issued right after the LDS-write
+// barrier, so their (L2/MALL) latency overlaps the MFMA block. Unlike the
+// double-buffered k64d (two LDS buffers: register pressure killed it) this
+// keeps ONE LDS buffer and double-buffers only the 16 staging VGPRs.
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM_THREADS, 2)
+void gemm_bf16_k64p_kernel(const __bf16* __restrict__ A,   // [M,K]
+                           const __bf16* __restrict__ Bt,  // [N,K]
+                           const float* __restrict__ bias, // [N] or null
+                           __bf16* __restrict__ C,         // [M,N]
+                           int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * BM, col0 = bn * BN;
+
+  __shared__ __bf16 Asm[BM * BK2];
+  __shared__ __bf16 Bsm[BN * BK2];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+
+  f32x4 acc[4][4] = {};
+
+  const int lin0 = wid * 1024 + lane * 16;
+
+  typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
+  int dsts[4], arow[4], brow[4], tcols[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    int lin = lin0 + it * 4096;
+    int trow = lin >> 7;
+    int tcol = lin & 127;
+    dsts[it] = (trow << 7) + (((tcol >> 4) ^ (trow & 7)) << 4);
+    tcols[it] = tcol;
+    int ga_row = row0 + trow;
+    arow[it] = ga_row < M ? ga_row : M - 1;
+    int gb_row = col0 + trow;
+    brow[it] = gb_row < N ? gb_row : N - 1;
+  }
+  u32x4 va[4], vb[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    va[it] = *(const u32x4*)((const char*)(A + (int64_t)arow[it] * K) +
+                             tcols[it]);
+    vb[it] = *(const u32x4*)((const char*)(Bt + (int64_t)brow[it] * K) +
+                             tcols[it]);
+  }
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      *(u32x4*)((char*)Asm + dsts[it]) = va[it];
+      *(u32x4*)((char*)Bsm + dsts[it]) = vb[it];
+    }
+    __syncthreads();
+    const int k1 = k0 + BK2;
+    if (k1 < K) {  // prefetch next slice; latency hides under the MFMAs
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        va[it] = *(const u32x4*)((const char*)(A + (int64_t)arow[it] * K +
+                                               k1) + tcols[it]);
+        vb[it] = *(const u32x4*)((const char*)(Bt + (int64_t)brow[it] * K +
+                                               k1) + tcols[it]);
+      }
+    }
+
+    const int fr = lane & 15;
+    const int fk8 = lane >> 4;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int row = wm + m * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        a_frag[m] = *(const bf16x8*)&Asm[row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int row = wn + n * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        b_frag[n] = *(const bf16x8*)&Bsm[row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
 // ---- double-buffered swizzled BK=64 ------------------------------------------
 // Same tile/swizzle as k64s but two LDS buffers: the NEXT K-slice's global
 // loads are issued before the current slice's MFMAs so HBM latency hides
@@ -547,6 +670,35 @@ extern "C" int launch_gemm_bf16_k64s(const void* A, const void* Bt,
     default: KSDISPATCH(ACT_NONE); break;
   }
 #undef KSDISPATCH
+  return 0;
+}
+
+extern "C" int launch_gemm_bf16_k64p(const void* A, const void* Bt,
+                                     const float* bias, void* C, int M,
+                                     int N, int K, int act, hipStream_t st) {
+  if (K % BK2 != 0) return -1;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define KPDISPATCH(ACT)                                                      \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_k64p_kernel<ACT, true><<<grid, block, 0, st>>>(              \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_k64p_kernel<ACT, false><<<grid, block, 0, st>>>(             \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: KPDISPATCH(ACT_RELU); break;
+    case ACT_GELU: KPDISPATCH(ACT_GELU); break;
+    case ACT_SILU: KPDISPATCH(ACT_SILU); break;
+    default: KPDISPATCH(ACT_NONE); break;
+  }
+#undef KPDISPATCH
   return 0;
 }
 
