@@ -77,6 +77,8 @@ int launch_gemm_bf16_k64s(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
 int launch_gemm_bf16_k64p(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
+int launch_gemm_bf16_k64w(const void*, const void*, const float*, void*,
+                          int, int, int, int, hipStream_t);
 int launch_gemm_bf16_k64d(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
 void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
@@ -535,6 +537,11 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
                                       C.data_ptr(), (int)M, (int)N, (int)K,
                                       (int)act, cur_stream()) == 0,
                 "k64p requires K % 64 == 0");
+  } else if (variant == 11) {
+    TORCH_CHECK(launch_gemm_bf16_k64w(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                      C.data_ptr(), (int)M, (int)N, (int)K,
+                                      (int)act, cur_stream()) == 0,
+                "k64w requires K % 64 == 0");
   } else if (variant == 9) {
     int rc = launch_gemm_bf16_k64d(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                    C.data_ptr(), (int)M, (int)N, (int)K,

@@ -485,6 +485,148 @@ void gemm_bf16_k64p_kernel(const __bf16* __restrict__ A,   // [M,K]
   }
 }
 
+// ---- wide 256x256 swizzled BK=64 ---------------------------------------------
+// Same swizzled-LDS scheme as k64s with a 256x256 tile: 8 waves (each owning
+// a 64x128 fragment grid), one block per CU, 64 KB LDS. Quadruple the
+// MFLOP-per-byte of the 128x128 tile: A re-read N/256 times and B re-read
+// M/256 times, which at short-K fc shapes (fc1: M8192 N3072 K768) halves
+// the L2/MALL traffic that bounds the 128x128 kernel.
+#define WBM 256
+#define WBN 256
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(512, 1)
+void gemm_bf16_k64w_kernel(const __bf16* __restrict__ A,   // [M,K]
+                           const __bf16* __restrict__ Bt,  // [N,K]
+                           const float* __restrict__ bias, // [N] or null
+                           __bf16* __restrict__ C,         // [M,N]
+                           int M, int N, int K, int tiles_n) {
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, off = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = bid / tiles_n, bn = bid % tiles_n;
+  const int row0 = bm * WBM, col0 = bn * WBN;
+
+  __shared__ __bf16 Asm[WBM * BK2];
+  __shared__ __bf16 Bsm[WBN * BK2];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;         // 8 waves
+  const int wm = (wid >> 1) * 64;   // 4 wave-rows of 64
+  const int wn = (wid & 1) * 128;   // 2 wave-cols of 128
+
+  f32x4 acc[4][8] = {};
+
+  const int lin0 = wid * 1024 + lane * 16;
+
+  typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+    // stage 256x64 of A and of Bt: 32 KB each, 4 chunks per thread per mat
+    u32x4 va[4], vb[4];
+    int dsts[4];
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int lin = lin0 + it * 8192;
+      int trow = lin >> 7;                 // 128 B per LDS row
+      int tcol = lin & 127;
+      dsts[it] = (trow << 7) + (((tcol >> 4) ^ (trow & 7)) << 4);
+      int ga_row = row0 + trow;
+      ga_row = ga_row < M ? ga_row : M - 1;
+      va[it] = *(const u32x4*)((const char*)(A + (int64_t)ga_row * K + k0)
+                               + tcol);
+      int gb_row = col0 + trow;
+      gb_row = gb_row < N ? gb_row : N - 1;
+      vb[it] = *(const u32x4*)((const char*)(Bt + (int64_t)gb_row * K + k0)
+                               + tcol);
+    }
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      *(u32x4*)((char*)Asm + dsts[it]) = va[it];
+      *(u32x4*)((char*)Bsm + dsts[it]) = vb[it];
+    }
+    __syncthreads();
+
+    const int fr = lane & 15;
+    const int fk8 = lane >> 4;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[4], b_frag[8];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int row = wm + m * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        a_frag[m] = *(const bf16x8*)&Asm[row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int n = 0; n < 8; ++n) {
+        int row = wn + n * 16 + fr;
+        int chunk = (ks * 4 + fk8) ^ (row & 7);
+        b_frag[n] = *(const bf16x8*)&Bsm[row * BK2 + chunk * 8];
+      }
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 8; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      int col = col0 + wn + n * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      float b = HAS_BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + wm + m * 16 + c_row_base + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + b;
+        v = apply_act(v, ACT);
+        C[(int64_t)row * N + col] = (__bf16)v;
+      }
+    }
+  }
+}
+
+extern "C" int launch_gemm_bf16_k64w(const void* A, const void* Bt,
+                                     const float* bias, void* C, int M,
+                                     int N, int K, int act, hipStream_t st) {
+  if (K % BK2 != 0) return -1;
+  int tiles_m = (M + WBM - 1) / WBM;
+  int tiles_n = (N + WBN - 1) / WBN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(512);
+#define KWDISPATCH(ACT)                                                      \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_k64w_kernel<ACT, true><<<grid, block, 0, st>>>(              \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_k64w_kernel<ACT, false><<<grid, block, 0, st>>>(             \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: KWDISPATCH(ACT_RELU); break;
+    case ACT_GELU: KWDISPATCH(ACT_GELU); break;
+    case ACT_SILU: KWDISPATCH(ACT_SILU); break;
+    default: KWDISPATCH(ACT_NONE); break;
+  }
+#undef KWDISPATCH
+  return 0;
+}
+
 // ---- double-buffered swizzled BK=64 ------------------------------------------
 // Same tile/swizzle as k64s but two LDS buffers: the NEXT K-slice's global
 // loads are issued before the current slice's MFMAs so HBM latency hides
