@@ -31,6 +31,11 @@ void launch_hash_export(const int64_t*, const int32_t*, uint32_t, int64_t*,
                         hipStream_t);
 void launch_segment_reduce_f32(const float*, const int32_t*, int64_t, int, int,
                                float*, int32_t*, hipStream_t);
+void launch_hash_agg_capture(const int64_t*, const int32_t*, int64_t,
+                             int64_t*, int32_t*, uint32_t, int32_t*, int32_t*,
+                             float*, int, const float* const*, const int*,
+                             float* const*, int32_t* const*, int, int64_t*,
+                             int32_t*, hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int64_t*, int32_t*, int32_t*,
                        uint32_t, hipStream_t);
 void launch_join_probe_count(const int64_t*, int64_t, const int64_t*,
@@ -959,6 +964,74 @@ fused_filter_agg(torch::Tensor key, torch::Tensor filter_col, int64_t op,
   return {uniq, counts, reduced};
 }
 
+// Capture-safe GROUP BY: padded key/value columns + the filter's device row
+// count in, padded (g_cap) group table out, group count written to
+// gcount_out (host-mapped int32 — ONE CPU read per step, zero D2H copies).
+// Records into a hipGraph: static grids, no host syncs; all workspaces are
+// allocated here (graph-pool when called during capture). Reference analog:
+// DataFusion AggregateExec (processor/sql.rs execute_query) as a replayable
+// device-side program.
+std::tuple<torch::Tensor, torch::Tensor, std::vector<torch::Tensor>>
+hash_agg_capture(torch::Tensor keys, torch::Tensor nrow,
+                 std::vector<torch::Tensor> vals, std::vector<int64_t> ops,
+                 int64_t table_size, int64_t g_cap,
+                 c10::optional<torch::Tensor> gcount_out) {
+  check_cuda(keys, "keys");
+  check_cuda(nrow, "nrow");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(nrow.scalar_type() == torch::kInt32 && nrow.numel() == 1,
+              "nrow must be device int32[1]");
+  TORCH_CHECK((table_size & (table_size - 1)) == 0, "table_size pow2");
+  TORCH_CHECK(g_cap >= 1 && g_cap <= 2048, "g_cap must be in [1, 2048]");
+  TORCH_CHECK(vals.size() == ops.size());
+  int64_t n_cap = keys.numel();
+  auto st = cur_stream();
+  auto opts32 = keys.options().dtype(torch::kInt32);
+  auto optsf = keys.options().dtype(torch::kFloat32);
+  auto table_keys = torch::empty({table_size}, keys.options());
+  auto table_gids = torch::empty({table_size}, opts32);
+  auto counter = torch::empty({1}, opts32);
+  auto gids = torch::empty({std::max<int64_t>(n_cap, 1)}, opts32);
+  auto counts = torch::empty({g_cap}, optsf);
+  auto uniq = torch::zeros({g_cap}, keys.options());
+  int nv = (int)vals.size();
+  std::vector<torch::Tensor> red;
+  std::vector<const float*> vptr;
+  std::vector<float*> rptr;
+  std::vector<int32_t*> mptr;
+  std::vector<int> opsi;
+  std::vector<torch::Tensor> scratch_keep;
+  for (int i = 0; i < nv; ++i) {
+    check_cuda(vals[i], "val");
+    TORCH_CHECK(vals[i].scalar_type() == torch::kFloat32,
+                "val cols must be f32");
+    red.push_back(torch::empty({g_cap}, optsf));
+    vptr.push_back(vals[i].data_ptr<float>());
+    rptr.push_back(red.back().data_ptr<float>());
+    if (ops[i] != 0) {
+      scratch_keep.push_back(torch::empty({g_cap}, opts32));
+      mptr.push_back(scratch_keep.back().data_ptr<int32_t>());
+    } else {
+      mptr.push_back(nullptr);
+    }
+    opsi.push_back((int)ops[i]);
+  }
+  int32_t* gout = nullptr;
+  if (gcount_out.has_value() && gcount_out->defined()) {
+    TORCH_CHECK(gcount_out->scalar_type() == torch::kInt32 &&
+                gcount_out->numel() == 1);
+    gout = gcount_out->data_ptr<int32_t>();
+  }
+  launch_hash_agg_capture(
+      keys.data_ptr<int64_t>(), nrow.data_ptr<int32_t>(), n_cap,
+      table_keys.data_ptr<int64_t>(), table_gids.data_ptr<int32_t>(),
+      (uint32_t)table_size, counter.data_ptr<int32_t>(),
+      gids.data_ptr<int32_t>(), counts.data_ptr<float>(), (int)g_cap,
+      vptr.data(), opsi.data(), rptr.data(), mptr.data(), nv,
+      uniq.data_ptr<int64_t>(), gout, st);
+  return {uniq, counts, red};
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,
            std::vector<std::tuple<torch::Tensor, torch::Tensor>>,
            torch::Tensor>
@@ -1254,6 +1327,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("json_decode", &json_decode);
   m.def("fused_filter_gather", &fused_filter_gather);
   m.def("fused_filter_agg", &fused_filter_agg);
+  m.def("hash_agg_capture", &hash_agg_capture, py::arg("keys"),
+        py::arg("nrow"), py::arg("vals"), py::arg("ops"),
+        py::arg("table_size"), py::arg("g_cap"),
+        py::arg("gcount_out") = c10::nullopt);
   m.def("filter_gather_capture", &filter_gather_capture);
   m.def("gemv_bf16_f32", &gemv_bf16_f32);
   m.def("gen_fields", &gen_fields);

@@ -160,6 +160,152 @@ __global__ void fill_i32_kernel(int32_t* p, int32_t v, int64_t n) {
   for (; i < n; i += stride) p[i] = v;
 }
 
+__global__ void fill_f32_kernel(float* p, float v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;
+}
+
+// ---- capture-safe (device-count) variants ------------------------------------
+// Same algorithms with the row count read from a device pointer (the
+// capture-safe filter's surviving-row count), so the whole
+// filter -> group-by -> reduce chain records into one hipGraph with static
+// grids. Garbage rows past *nrow in the padded inputs are never touched.
+__global__ void hash_insert_dev_kernel(const int64_t* __restrict__ keys,
+                                       const int32_t* __restrict__ nrow,
+                                       int64_t* __restrict__ table_keys,
+                                       int32_t* __restrict__ table_gids,
+                                       uint32_t table_mask,
+                                       int32_t* __restrict__ counter) {
+  const int64_t n = *nrow;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    uint32_t h = (uint32_t)(mix64((uint64_t)k) & table_mask);
+    for (;;) {
+      int64_t cur = __hip_atomic_load(&table_keys[h], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == k) break;
+      if (cur == EMPTY_KEY) {
+        int64_t prev = (int64_t)atomicCAS(
+            (unsigned long long*)&table_keys[h], (unsigned long long)EMPTY_KEY,
+            (unsigned long long)k);
+        if (prev == EMPTY_KEY) {
+          int32_t gid = atomicAdd(counter, 1);
+          __hip_atomic_store(&table_gids[h], gid, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          break;
+        }
+        if (prev == k) break;
+      }
+      h = (h + 1) & table_mask;
+    }
+  }
+}
+
+__global__ void hash_lookup_dev_kernel(const int64_t* __restrict__ keys,
+                                       const int32_t* __restrict__ nrow,
+                                       const int64_t* __restrict__ table_keys,
+                                       const int32_t* __restrict__ table_gids,
+                                       uint32_t table_mask,
+                                       int32_t* __restrict__ gids_out) {
+  const int64_t n = *nrow;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    uint32_t h = (uint32_t)(mix64((uint64_t)k) & table_mask);
+    while (table_keys[h] != k) h = (h + 1) & table_mask;
+    gids_out[i] = table_gids[h];
+  }
+}
+
+// count + sum in one LDS-tiled pass (vals may be null: count only)
+__global__ void segment_cs_lds_dev_kernel(const float* __restrict__ vals,
+                                          const int32_t* __restrict__ gids,
+                                          const int32_t* __restrict__ nrow,
+                                          int g_cap,
+                                          float* __restrict__ counts,
+                                          float* __restrict__ sums) {
+  __shared__ float pc[LDS_GROUPS];
+  __shared__ float ps[LDS_GROUPS];
+  for (int j = threadIdx.x; j < g_cap; j += blockDim.x) {
+    pc[j] = 0.f;
+    ps[j] = 0.f;
+  }
+  __syncthreads();
+  const int64_t n = *nrow;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int g = gids[i];
+    if ((unsigned)g >= (unsigned)g_cap) continue;  // overflow guard; the
+    // Python wrapper validates g <= g_cap after every step
+    atomicAdd(&pc[g], 1.f);
+    if (vals) atomicAdd(&ps[g], vals[i]);
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < g_cap; j += blockDim.x) {
+    if (pc[j] != 0.f) {
+      if (counts) atomicAdd(&counts[j], pc[j]);
+      if (sums && vals) atomicAdd(&sums[j], ps[j]);
+    }
+  }
+}
+
+__global__ void segment_sum_lds_dev_kernel(const float* __restrict__ vals,
+                                           const int32_t* __restrict__ gids,
+                                           const int32_t* __restrict__ nrow,
+                                           int g_cap, float* __restrict__ out) {
+  __shared__ float part[LDS_GROUPS];
+  for (int j = threadIdx.x; j < g_cap; j += blockDim.x) part[j] = 0.f;
+  __syncthreads();
+  const int64_t n = *nrow;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    if ((unsigned)gids[i] < (unsigned)g_cap)
+      atomicAdd(&part[gids[i]], vals[i]);
+  __syncthreads();
+  for (int j = threadIdx.x; j < g_cap; j += blockDim.x)
+    if (part[j] != 0.f) atomicAdd(&out[j], part[j]);
+}
+
+__global__ void segment_mm_global_dev_kernel(
+    const float* __restrict__ vals, const int32_t* __restrict__ gids,
+    const int32_t* __restrict__ nrow, int op, int g_cap,
+    int32_t* __restrict__ out_flipped) {
+  const int64_t n = *nrow;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int g = gids[i];
+    if ((unsigned)g >= (unsigned)g_cap) continue;
+    uint32_t v = float_flip(vals[i]);
+    if (op == MIN) atomicMin((uint32_t*)&out_flipped[g], v);
+    else atomicMax((uint32_t*)&out_flipped[g], v);
+  }
+}
+
+// export with a bound (capture path: uniq buffer is g_cap entries)
+__global__ void hash_export_cap_kernel(const int64_t* __restrict__ table_keys,
+                                       const int32_t* __restrict__ table_gids,
+                                       uint32_t table_size, int g_cap,
+                                       int64_t* __restrict__ uniq) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < table_size; i += stride) {
+    int64_t k = table_keys[i];
+    int32_t g = k != EMPTY_KEY ? table_gids[i] : -1;
+    if (g >= 0 && g < g_cap) uniq[g] = k;
+  }
+}
+
+__global__ void copy_i32_kernel(const int32_t* src, int32_t* dst) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *dst = *src;
+}
+
 // ---- host launchers ----------------------------------------------------------
 extern "C" {
 
@@ -209,6 +355,63 @@ void launch_segment_reduce_f32(const float* vals, const int32_t* gids,
     segment_mm_global_kernel<<<grid, AGG_BLOCK, 0, st>>>(
         vals, gids, n, op, scratch_flipped);
   unflip_kernel<<<grid_for(g, 256), 256, 0, st>>>(scratch_flipped, out, g);
+}
+
+// Capture-safe whole-chain launcher: reset (static fills) -> insert/lookup
+// (device n) -> count+sum fused -> extra reductions -> export -> counter copy.
+// ops[i]: 0=sum 1=min 2=max. red_out[i] is the f32[g_cap] destination for
+// vals[i]; mm_scratch[i] only used (non-null) for min/max.
+void launch_hash_agg_capture(const int64_t* keys, const int32_t* nrow,
+                             int64_t n_cap, int64_t* table_keys,
+                             int32_t* table_gids, uint32_t table_size,
+                             int32_t* counter, int32_t* gids, float* counts,
+                             int g_cap, const float* const* vals,
+                             const int* ops, float* const* red_out,
+                             int32_t* const* mm_scratch, int nv,
+                             int64_t* uniq, int32_t* gcount_out,
+                             hipStream_t st) {
+  fill_i64_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_keys, EMPTY_KEY, table_size);
+  fill_i32_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_gids, -1, table_size);
+  fill_i32_kernel<<<1, 1, 0, st>>>(counter, 0, 1);
+  fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(counts, 0.f, g_cap);
+  int grid = grid_for(n_cap, AGG_BLOCK);
+  hash_insert_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+      keys, nrow, table_keys, table_gids, table_size - 1, counter);
+  hash_lookup_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+      keys, nrow, table_keys, table_gids, table_size - 1, gids);
+  // first sum column rides the count pass
+  int first_sum = -1;
+  for (int i = 0; i < nv && first_sum < 0; ++i)
+    if (ops[i] == SUM) first_sum = i;
+  if (first_sum >= 0)
+    fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(red_out[first_sum],
+                                                          0.f, g_cap);
+  segment_cs_lds_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+      first_sum >= 0 ? vals[first_sum] : nullptr, gids, nrow, g_cap, counts,
+      first_sum >= 0 ? red_out[first_sum] : nullptr);
+  for (int i = 0; i < nv; ++i) {
+    if (i == first_sum) continue;
+    if (ops[i] == SUM) {
+      fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(red_out[i], 0.f,
+                                                            g_cap);
+      segment_sum_lds_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+          vals[i], gids, nrow, g_cap, red_out[i]);
+    } else {
+      int32_t init = float_flip_host(ops[i] == MIN ? INFINITY : -INFINITY);
+      fill_i32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(mm_scratch[i],
+                                                            init, g_cap);
+      segment_mm_global_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
+          vals[i], gids, nrow, (int)ops[i], g_cap, mm_scratch[i]);
+      unflip_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(mm_scratch[i],
+                                                          red_out[i], g_cap);
+    }
+  }
+  hash_export_cap_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
+      table_keys, table_gids, table_size, g_cap, uniq);
+  if (gcount_out)
+    copy_i32_kernel<<<1, 1, 0, st>>>(counter, gcount_out);
 }
 
 }  // extern "C"

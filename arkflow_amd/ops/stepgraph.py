@@ -155,6 +155,16 @@ class FusedGenerateFilterInfer:
             return int(self.count_host[0])
         return int(self.count.item())
 
+    def _make_batch(self, clone: bool = False) -> MessageBatch:
+        kept = self._kept()
+        cols = {f: Column("numeric", self.outs[f][:kept])
+                for f in self.fields}
+        cols["score"] = Column("numeric", self._scores[:kept])
+        if clone:
+            cols = {k: Column("numeric", c.data.clone())
+                    for k, c in cols.items()}
+        return MessageBatch(cols, input_name="generate")
+
     # ------------------------------------------------------------------ step
     def step(self) -> Tuple[MessageBatch, int]:
         """One replay. Returns (batch_of_views, kept_rows); the batch is
@@ -162,11 +172,153 @@ class FusedGenerateFilterInfer:
         if self._graph is None:
             self.capture()
         self._graph.replay()
-        kept = self._kept()
-        cols = {f: Column("numeric", self.outs[f][:kept])
-                for f in self.fields}
-        cols["score"] = Column("numeric", self._scores[:kept])
-        return MessageBatch(cols, input_name="generate"), kept
+        batch = self._make_batch()
+        return batch, batch.num_rows
+
+
+class FusedGenerateAgg:
+    """generate(fields) → WHERE col OP scalar → GROUP BY key AGG(...) as ONE
+    hipGraph (BASELINE config 2 whole-step form). Shares the generator /
+    capture-safe filter stages with FusedGenerateFilterInfer; the group-by
+    stage is the capture-safe hash-agg chain (csrc/hash_agg.hip device-count
+    kernels). Per step: one replay + one CPU read of the host-mapped group
+    count. aggs: list of (fn, col, alias) with fn in count/sum/min/max/avg.
+    """
+
+    def __init__(self, fields: Dict[str, dict], batch_size: int,
+                 filter_col: str, op: str, scalar: float, key_col: str,
+                 aggs, device: torch.device, seed: int = 0x5EED,
+                 g_cap: int = 2048, table_size: int = 8192):
+        self.device = torch.device(device)
+        if self.device.type != "cuda":
+            raise RuntimeError("FusedGenerateAgg requires a GPU")
+        self.n = int(batch_size)
+        self.fields = dict(fields)
+        self.filter_col = filter_col
+        self.op = _OPS[op]
+        self.scalar = float(scalar)
+        self.key_col = key_col
+        self.aggs = list(aggs)
+        self.g_cap = int(g_cap)
+        self.table_size = int(table_size)
+        self.nat = require_native()
+        torch.cuda.manual_seed(seed)
+
+        names = list(fields)
+        self.float_names = [f for f in names
+                            if str(fields[f].get("dtype", "float32"))
+                            not in ("int32", "int64")]
+        self.int_names = [f for f in names if f not in self.float_names]
+        if self.int_names != [key_col]:
+            raise ValueError("fused agg needs exactly one int64 key column")
+        self.block = torch.zeros((len(self.float_names), self.n),
+                                 device=self.device, dtype=torch.float32)
+        self.key = torch.zeros(self.n, device=self.device, dtype=torch.int64)
+        self.ctr = torch.zeros(2, device=self.device, dtype=torch.int64)
+        self.ctr[0] = int(seed) & 0x7FFFFFFF
+        self.outs: Dict[str, torch.Tensor] = {}
+        for f in self.float_names:
+            self.outs[f] = torch.zeros(self.n, device=self.device,
+                                       dtype=torch.float32)
+        self.outs[key_col] = torch.zeros(self.n, device=self.device,
+                                         dtype=torch.int64)
+        import os
+        try:
+            if os.environ.get("ARKFLOW_MAPPED_COUNT", "1") == "0":
+                raise RuntimeError("disabled")
+            self.count_host, self.count = self.nat.mapped_int32(1)
+            self.gcount_host, self.gcount = self.nat.mapped_int32(1)
+        except RuntimeError:
+            self.count_host = self.gcount_host = None
+            self.count = torch.zeros(1, device=self.device, dtype=torch.int32)
+            self.gcount = torch.zeros(1, device=self.device,
+                                      dtype=torch.int32)
+        self._lo = [float(fields[f].get("low", 0.0))
+                    for f in self.float_names]
+        self._width = [float(fields[f].get("high", 100.0)) -
+                       float(fields[f].get("low", 0.0))
+                       for f in self.float_names]
+        spec = fields[key_col]
+        self._key_lo = int(float(spec.get("low", 0.0)))
+        hi = int(float(spec.get("high", 100.0)))
+        self._key_range = max(hi - self._key_lo, 1)
+        # one reduction per distinct (col, op!=count) pair
+        ops_code = {"sum": 0, "avg": 0, "min": 1, "max": 2}
+        pairs = []
+        for fn, col, _ in self.aggs:
+            if fn in ("sum", "min", "max", "avg"):
+                pairs.append((col, "sum" if fn == "avg" else fn))
+        self._val_list = list(dict.fromkeys(pairs))
+        self._val_ops = [ops_code[o] for _, o in self._val_list]
+        self._graph = None
+
+    def _body(self):
+        cols: Dict[str, torch.Tensor] = {
+            f: self.block[i] for i, f in enumerate(self.float_names)}
+        cols[self.key_col] = self.key
+        self.nat.gen_fields(self.block, self.key, self._lo, self._width,
+                            self._key_lo, self._key_range, self.ctr)
+        ordered = [cols[f] for f in self.fields]
+        fidx = list(self.fields).index(self.filter_col)
+        self.nat.filter_gather_capture(
+            ordered, fidx, self.op, self.scalar,
+            [self.outs[f] for f in self.fields], self.count)
+        uniq, counts, red = self.nat.hash_agg_capture(
+            self.outs[self.key_col], self.count,
+            [self.outs[c] for c, _ in self._val_list], self._val_ops,
+            self.table_size, self.g_cap, self.gcount)
+        return uniq, counts, red
+
+    def capture(self) -> None:
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._body()
+        torch.cuda.current_stream().wait_stream(s)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
+            self._uniq, self._counts, self._red = self._body()
+
+    def _kept(self) -> int:
+        """Group count for the last replay on the CURRENT stream."""
+        if self.gcount_host is not None:
+            torch.cuda.current_stream().synchronize()
+            g = int(self.gcount_host[0])
+        else:
+            g = int(self.gcount.item())
+        if g > self.g_cap:
+            raise RuntimeError(
+                f"group-by overflow: {g} groups > g_cap {self.g_cap}")
+        return g
+
+    def _make_batch(self, clone: bool = False) -> MessageBatch:
+        g = self._kept()
+        rmap = {pair: t for pair, t in zip(self._val_list, self._red)}
+        cols: Dict[str, Column] = {}
+        for fn, col, alias in self.aggs:
+            if fn == "key":
+                cols[alias] = Column("numeric", self._uniq[:g])
+            elif fn == "count":
+                cols[alias] = Column("numeric",
+                                     self._counts[:g].to(torch.int64))
+            elif fn == "avg":
+                cols[alias] = Column(
+                    "numeric",
+                    rmap[(col, "sum")][:g].double() /
+                    self._counts[:g].double())
+            else:
+                cols[alias] = Column("numeric", rmap[(col, fn)][:g])
+        if clone:
+            cols = {k: Column("numeric", c.data.clone())
+                    for k, c in cols.items()}
+        return MessageBatch(cols, input_name="generate")
+
+    def step(self) -> MessageBatch:
+        if self._graph is None:
+            self.capture()
+        self._graph.replay()
+        return self._make_batch()
 
 
 class FusedProtoMlp:
@@ -267,17 +419,9 @@ class FusedStepSource:
 
     def _consume(self, i: int):
         with torch.cuda.stream(self.streams[i]):
-            inst = self.insts[i]
-            kept = inst._kept()
-            cols = {f: Column("numeric", inst.outs[f][:kept])
-                    for f in inst.fields}
-            cols["score"] = Column("numeric", inst._scores[:kept])
-            if self.clone:
-                # engine queues / window buffers retain batches past the
-                # next replay — detach from the static graph buffers
-                cols = {k: Column("numeric", c.data.clone())
-                        for k, c in cols.items()}
-            return MessageBatch(cols, input_name="generate")
+            # clone=True (engine mode): queues / window buffers retain
+            # batches past the next replay — detach from the static buffers
+            return self.insts[i]._make_batch(clone=self.clone)
 
     async def read(self):
         from ..spi import NoopAck

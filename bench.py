@@ -59,6 +59,22 @@ def build_pipeline(args, device):
         return gen_tok, Pipeline([infer])
     if args.model == "sqlagg":
         # BASELINE config 2: sql filter + hash aggregate, GPU-resident columns
+        if device.type == "cuda" and not args.no_stepgraph:
+            # whole-step hipGraph: generate+filter+group-by+reduce replay as
+            # ONE graph, one CPU read (the host-mapped group count) per step
+            from arkflow_amd.ops.stepgraph import (
+                FusedGenerateAgg, FusedStepSource)
+
+            def make_agg(seed_off=0):
+                return FusedGenerateAgg(
+                    fields, args.batch_size, "f0", ">=", 0.2, "key",
+                    [("key", None, "key"), ("count", None, "c"),
+                     ("sum", "f0", "s")], device,
+                    seed=7 + args.rank + seed_off * 1000)
+
+            src = FusedStepSource(make_agg(0), ninstances=2,
+                                  make_instance=lambda: make_agg(1))
+            return src, Pipeline([])
         agg = SqlProcessor({
             "query": "SELECT key, count(*) AS c, sum(f0) AS s FROM flow "
                      "WHERE f0 >= 0.2 GROUP BY key"})
