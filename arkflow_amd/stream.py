@@ -441,6 +441,38 @@ def build_stream(config: StreamConfig) -> Stream:
             make_instance=lambda: mk(input_.seed + next(seeds) * 7919))
         processors = []
 
+    # same fusion for generate → sql(filter + GROUP BY aggregates): the
+    # capture-safe hash-agg chain replays inside the step graph
+    agg_spec = (fusable_agg_chain(config, input_, processors, resource)
+                if wal is None else None)
+    if agg_spec is not None:
+        from .ops.stepgraph import FusedGenerateAgg, FusedStepSource
+        key, filt, plan, g_cap, table_size = agg_spec
+        if filt is None:
+            f0 = next(f for f, s in input_.fields.items()
+                      if str(s.get("dtype", "float32"))
+                      not in ("int32", "int64"))
+            fcol, op, scalar = f0, "!=", float("nan")  # x != nan ⇒ keep all
+        else:
+            fcol, opi, scalar = filt
+            op = {0: "<", 1: "<=", 2: ">", 3: ">=", 4: "==", 5: "!="}[opi]
+
+        def mka(seed):
+            return FusedGenerateAgg(
+                input_.fields, input_.batch_size, fcol, op, scalar, key,
+                plan, resource.device, seed=seed, g_cap=g_cap,
+                table_size=table_size)
+
+        log.info("stream %s: fused generate→filter→group-by into one "
+                 "hipGraph", config.id)
+        clone = not (buffer is None
+                     and getattr(output, "retains", True) is False)
+        seeds = iter(range(1, 16))
+        input_ = FusedStepSource(
+            mka(input_.seed), ninstances=2, clone=clone,
+            make_instance=lambda: mka(input_.seed + next(seeds) * 7919))
+        processors = []
+
     return Stream(
         config, input_, Pipeline(processors), output, error_output,
         buffer, wal, temporaries,
@@ -495,6 +527,54 @@ def fusable_chain(config, input_, processors, resource) -> bool:
     if cols is not None and list(cols) != floats:
         return False
     return True
+
+
+def fusable_agg_chain(config, input_, processors, resource):
+    """When the stream is exactly generate → sql(simple filter + single-key
+    GROUP BY aggregates) on GPU, return (key, filt, plan, g_cap, table_size)
+    for FusedGenerateAgg; else None. The key field's configured range bounds
+    the group count (must fit the 2048-group LDS tile). ``fuse: false``
+    opts out."""
+    import torch
+
+    from .inputs.generate import GenerateInput
+    from .processors.sql import SqlProcessor
+
+    if config.input.get("fuse") is False or \
+            getattr(resource, "device", torch.device("cpu")).type != "cuda":
+        return None
+    if not isinstance(input_, GenerateInput) or not input_.fields:
+        return None
+    if input_.count is not None or input_.interval_secs > 0:
+        return None
+    if len(processors) != 1 or not isinstance(processors[0], SqlProcessor):
+        return None
+    fast = processors[0]._fast_agg
+    if fast is None:
+        return None
+    key, filt, plan = fast
+    spec = input_.fields.get(key)
+    if spec is None or str(spec.get("dtype", "")) != "int64":
+        return None
+    for f, s in input_.fields.items():
+        if f == key:
+            continue
+        if str(s.get("dtype", "float32")) != "float32":
+            return None
+    if filt is not None and filt[0] not in input_.fields:
+        return None
+    for fn, col, _alias in plan:
+        if fn in ("sum", "min", "max", "avg") and col not in input_.fields:
+            return None
+    key_range = int(float(spec.get("high", 100.0))) - \
+        int(float(spec.get("low", 0.0)))
+    if key_range < 1 or key_range > 2048:
+        return None  # group table must fit the LDS-tiled reduction
+    g_cap = min(max(key_range, 1), 2048)
+    table_size = 256
+    while table_size < 4 * g_cap:
+        table_size <<= 1
+    return key, filt, plan, g_cap, table_size
 
 
 def _resolve_device(device: Optional[str]):

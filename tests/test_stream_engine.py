@@ -554,3 +554,53 @@ def test_dedicated_thread_stream_lifecycle(run):
         await eng.runtime.stop_all()
 
     run(main(), timeout=60)
+
+
+def test_fusable_agg_chain_detection():
+    """The GROUP BY step-graph fusion triggers exactly on generate→sql
+    (simple filter + single-key aggregates) on GPU, with sane table sizing."""
+    import torch
+
+    from arkflow_amd.inputs.generate import GenerateInput
+    from arkflow_amd.processors.sql import SqlProcessor
+    from arkflow_amd.stream import fusable_agg_chain
+
+    class R:
+        device = torch.device("cuda")
+
+    class RC:
+        device = torch.device("cpu")
+
+    class CfgIn:
+        def __init__(self, **kw):
+            self.input = kw
+
+    fields = {"f0": {"dtype": "float32"}, "f1": {"dtype": "float32"},
+              "key": {"dtype": "int64", "low": 0, "high": 300}}
+    gen = GenerateInput({"batch_size": 64, "interval": "0ms",
+                         "fields": fields})
+    agg = SqlProcessor({
+        "query": "SELECT key, count(*) AS c, sum(f0) AS s, min(f1) AS m "
+                 "FROM flow WHERE f0 >= 0.2 GROUP BY key"})
+    cfg = CfgIn()
+    spec = fusable_agg_chain(cfg, gen, [agg], R())
+    assert spec is not None
+    key, filt, plan, g_cap, table_size = spec
+    assert key == "key" and filt == ("f0", 3, 0.2)
+    assert g_cap == 300 and table_size == 2048
+    assert [p[0] for p in plan] == ["key", "count", "sum", "min"]
+    # opt-outs / mismatches
+    assert fusable_agg_chain(CfgIn(fuse=False), gen, [agg], R()) is None
+    assert fusable_agg_chain(cfg, gen, [agg], RC()) is None
+    plain = SqlProcessor({"query": "SELECT * FROM flow WHERE f0 >= 0.5"})
+    assert fusable_agg_chain(cfg, gen, [plain], R()) is None
+    # key range too large for the LDS group tile → eager path
+    wide = dict(fields, key={"dtype": "int64", "low": 0, "high": 1 << 20})
+    genw = GenerateInput({"batch_size": 64, "interval": "0ms",
+                          "fields": wide})
+    assert fusable_agg_chain(cfg, genw, [agg], R()) is None
+    # no WHERE is still fusable (keep-all filter)
+    nowhere = SqlProcessor(
+        {"query": "SELECT key, count(*) AS c FROM flow GROUP BY key"})
+    spec2 = fusable_agg_chain(cfg, gen, [nowhere], R())
+    assert spec2 is not None and spec2[1] is None
