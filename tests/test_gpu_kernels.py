@@ -1186,3 +1186,53 @@ def test_fused_agg_step_source_pipelined(dev):
         total = int(c.sum().item())
         assert 0 < total <= 4096
         assert batch.column("key").data.numel() == c.numel() <= 32
+
+
+def test_engine_stream_fuses_group_by_chain(dev):
+    """A YAML-shaped generate→sql(filter + GROUP BY) stream on GPU builds
+    with the whole-step agg graph source and emits consistent group tables."""
+    import asyncio
+
+    from arkflow_amd.config import EngineConfig
+    from arkflow_amd.ops.stepgraph import FusedStepSource
+    from arkflow_amd.outputs.basic import MemoryOutput
+    from arkflow_amd.stream import build_stream
+
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "fusedagg",
+        "input": {"type": "generate", "batch_size": 4096, "interval": "0ms",
+                  "fields": {
+                      "f0": {"dtype": "float32"},
+                      "f1": {"dtype": "float32", "low": -2.0, "high": 2.0},
+                      "key": {"dtype": "int64", "low": 0, "high": 64}}},
+        "pipeline": {"thread_num": 1, "processors": [
+            {"type": "sql",
+             "query": "SELECT key, count(*) AS c, sum(f0) AS s, "
+                      "max(f1) AS mx FROM flow WHERE f0 >= 0.25 "
+                      "GROUP BY key"},
+        ]},
+        "output": {"type": "memory"},
+    }]})
+    stream = build_stream(cfg.streams[0])
+    assert isinstance(stream.input, FusedStepSource), "agg fusion missed"
+
+    async def run_briefly():
+        cancel = asyncio.Event()
+        task = asyncio.ensure_future(stream.run(cancel))
+        await asyncio.sleep(1.0)
+        cancel.set()
+        await asyncio.wait_for(task, 30)
+
+    asyncio.new_event_loop().run_until_complete(run_briefly())
+    out = stream.output
+    assert isinstance(out, MemoryOutput) and out.batches
+    for b in out.batches[:5]:
+        assert set(b.columns) == {"key", "c", "s", "mx"}
+        c = b.column("c").data
+        assert 0 < b.num_rows <= 64
+        assert 0 < int(c.sum().item()) <= 4096
+        assert bool((b.column("mx").data <= 2.0).all())
+        # sum(f0) of rows passing f0 >= 0.25 is bounded per group
+        s = b.column("s").data
+        assert bool((s >= 0.25 * c.float() - 1e-3).all())
+        assert bool((s <= 1.0 * c.float() + 1e-3).all())
