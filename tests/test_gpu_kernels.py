@@ -1202,7 +1202,7 @@ def test_engine_stream_fuses_group_by_chain(dev):
         "id": "fusedagg",
         "input": {"type": "generate", "batch_size": 4096, "interval": "0ms",
                   "fields": {
-                      "f0": {"dtype": "float32"},
+                      "f0": {"dtype": "float32", "low": 0.0, "high": 1.0},
                       "f1": {"dtype": "float32", "low": -2.0, "high": 2.0},
                       "key": {"dtype": "int64", "low": 0, "high": 64}}},
         "pipeline": {"thread_num": 1, "processors": [
