@@ -35,7 +35,7 @@ void launch_hash_agg_capture(const int64_t*, const int32_t*, int64_t,
                              int64_t*, int32_t*, uint32_t, int32_t*, int32_t*,
                              float*, int, const float* const*, const int*,
                              float* const*, int32_t* const*, int, int64_t*,
-                             int32_t*, hipStream_t);
+                             int64_t*, int32_t*, hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int64_t*, int32_t*, int32_t*,
                        uint32_t, hipStream_t);
 void launch_join_probe_count(const int64_t*, int64_t, const int64_t*,
@@ -984,6 +984,8 @@ hash_agg_capture(torch::Tensor keys, torch::Tensor nrow,
   TORCH_CHECK((table_size & (table_size - 1)) == 0, "table_size pow2");
   TORCH_CHECK(g_cap >= 1 && g_cap <= 2048, "g_cap must be in [1, 2048]");
   TORCH_CHECK(vals.size() == ops.size());
+  TORCH_CHECK(vals.size() <= 4, "hash_agg_capture supports up to 4 "
+              "reduction columns");
   int64_t n_cap = keys.numel();
   auto st = cur_stream();
   auto opts32 = keys.options().dtype(torch::kInt32);
@@ -993,6 +995,7 @@ hash_agg_capture(torch::Tensor keys, torch::Tensor nrow,
   auto counter = torch::empty({1}, opts32);
   auto gids = torch::empty({std::max<int64_t>(n_cap, 1)}, opts32);
   auto counts = torch::empty({g_cap}, optsf);
+  auto counts_i64 = torch::empty({g_cap}, keys.options());
   auto uniq = torch::zeros({g_cap}, keys.options());
   int nv = (int)vals.size();
   std::vector<torch::Tensor> red;
@@ -1028,8 +1031,8 @@ hash_agg_capture(torch::Tensor keys, torch::Tensor nrow,
       (uint32_t)table_size, counter.data_ptr<int32_t>(),
       gids.data_ptr<int32_t>(), counts.data_ptr<float>(), (int)g_cap,
       vptr.data(), opsi.data(), rptr.data(), mptr.data(), nv,
-      uniq.data_ptr<int64_t>(), gout, st);
-  return {uniq, counts, red};
+      uniq.data_ptr<int64_t>(), counts_i64.data_ptr<int64_t>(), gout, st);
+  return {uniq, counts_i64, red};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor,

@@ -302,6 +302,44 @@ __global__ void hash_export_cap_kernel(const int64_t* __restrict__ table_keys,
   }
 }
 
+// one-kernel reset of every per-step buffer (table, counter, counts, up to
+// 4 reduction buffers) — replaces 5-7 fill launches; at ~100-group tables
+// each extra launch is pure overhead
+__global__ void agg_reset_kernel(int64_t* __restrict__ tk,
+                                 int32_t* __restrict__ tg, uint32_t ts,
+                                 int32_t* __restrict__ counter,
+                                 float* __restrict__ counts, int g_cap,
+                                 float* s0, float* s1, float* s2, float* s3,
+                                 int32_t* m0, int32_t* m1, int32_t* m2,
+                                 int32_t* m3, int32_t i0, int32_t i1,
+                                 int32_t i2, int32_t i3) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  if (i == 0) *counter = 0;
+  for (int64_t j = i; j < ts; j += stride) {
+    tk[j] = EMPTY_KEY;
+    tg[j] = -1;
+  }
+  for (int64_t j = i; j < g_cap; j += stride) {
+    counts[j] = 0.f;
+    if (s0) s0[j] = 0.f;
+    if (s1) s1[j] = 0.f;
+    if (s2) s2[j] = 0.f;
+    if (s3) s3[j] = 0.f;
+    if (m0) m0[j] = i0;
+    if (m1) m1[j] = i1;
+    if (m2) m2[j] = i2;
+    if (m3) m3[j] = i3;
+  }
+}
+
+// SQL count() is int64: convert in-graph so no per-step torch cast runs
+__global__ void counts_to_i64_kernel(const float* __restrict__ counts,
+                                     int g_cap, int64_t* __restrict__ out) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < g_cap) out[i] = (int64_t)(counts[i] + 0.5f);
+}
+
 __global__ void copy_i32_kernel(const int32_t* src, int32_t* dst) {
   if (threadIdx.x == 0 && blockIdx.x == 0) *dst = *src;
 }
@@ -368,14 +406,25 @@ void launch_hash_agg_capture(const int64_t* keys, const int32_t* nrow,
                              int g_cap, const float* const* vals,
                              const int* ops, float* const* red_out,
                              int32_t* const* mm_scratch, int nv,
-                             int64_t* uniq, int32_t* gcount_out,
-                             hipStream_t st) {
-  fill_i64_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
-      table_keys, EMPTY_KEY, table_size);
-  fill_i32_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
-      table_gids, -1, table_size);
-  fill_i32_kernel<<<1, 1, 0, st>>>(counter, 0, 1);
-  fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(counts, 0.f, g_cap);
+                             int64_t* uniq, int64_t* counts_i64,
+                             int32_t* gcount_out, hipStream_t st) {
+  // one fused reset launch for every per-step buffer (nv <= 4)
+  float* sb[4] = {nullptr, nullptr, nullptr, nullptr};
+  int32_t* mb[4] = {nullptr, nullptr, nullptr, nullptr};
+  int32_t mi[4] = {0, 0, 0, 0};
+  for (int i = 0; i < nv && i < 4; ++i) {
+    if (ops[i] == SUM) {
+      sb[i] = red_out[i];
+    } else {
+      mb[i] = mm_scratch[i];
+      mi[i] = float_flip_host(ops[i] == MIN ? INFINITY : -INFINITY);
+    }
+  }
+  agg_reset_kernel<<<grid_for(table_size > g_cap ? table_size : g_cap, 256),
+                     256, 0, st>>>(table_keys, table_gids, table_size,
+                                   counter, counts, g_cap, sb[0], sb[1],
+                                   sb[2], sb[3], mb[0], mb[1], mb[2], mb[3],
+                                   mi[0], mi[1], mi[2], mi[3]);
   int grid = grid_for(n_cap, AGG_BLOCK);
   hash_insert_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
       keys, nrow, table_keys, table_gids, table_size - 1, counter);
@@ -385,23 +434,15 @@ void launch_hash_agg_capture(const int64_t* keys, const int32_t* nrow,
   int first_sum = -1;
   for (int i = 0; i < nv && first_sum < 0; ++i)
     if (ops[i] == SUM) first_sum = i;
-  if (first_sum >= 0)
-    fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(red_out[first_sum],
-                                                          0.f, g_cap);
   segment_cs_lds_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
       first_sum >= 0 ? vals[first_sum] : nullptr, gids, nrow, g_cap, counts,
       first_sum >= 0 ? red_out[first_sum] : nullptr);
   for (int i = 0; i < nv; ++i) {
     if (i == first_sum) continue;
     if (ops[i] == SUM) {
-      fill_f32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(red_out[i], 0.f,
-                                                            g_cap);
       segment_sum_lds_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
           vals[i], gids, nrow, g_cap, red_out[i]);
     } else {
-      int32_t init = float_flip_host(ops[i] == MIN ? INFINITY : -INFINITY);
-      fill_i32_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(mm_scratch[i],
-                                                            init, g_cap);
       segment_mm_global_dev_kernel<<<grid, AGG_BLOCK, 0, st>>>(
           vals[i], gids, nrow, (int)ops[i], g_cap, mm_scratch[i]);
       unflip_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(mm_scratch[i],
@@ -410,6 +451,9 @@ void launch_hash_agg_capture(const int64_t* keys, const int32_t* nrow,
   }
   hash_export_cap_kernel<<<grid_for(table_size, 256), 256, 0, st>>>(
       table_keys, table_gids, table_size, g_cap, uniq);
+  if (counts_i64)
+    counts_to_i64_kernel<<<grid_for(g_cap, 256), 256, 0, st>>>(counts, g_cap,
+                                                               counts_i64);
   if (gcount_out)
     copy_i32_kernel<<<1, 1, 0, st>>>(counter, gcount_out);
 }

@@ -300,8 +300,7 @@ class FusedGenerateAgg:
             if fn == "key":
                 cols[alias] = Column("numeric", self._uniq[:g])
             elif fn == "count":
-                cols[alias] = Column("numeric",
-                                     self._counts[:g].to(torch.int64))
+                cols[alias] = Column("numeric", self._counts[:g])
             elif fn == "avg":
                 cols[alias] = Column(
                     "numeric",
