@@ -97,3 +97,17 @@ run("json long strings (262K-doc batch)",
     JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
     l4)
 N = _saveN
+
+# even longer bodies: per-doc FSM fixed costs amortize further
+for kb, n_docs in [(8, 8192), (16, 4096), (32, 2048)]:
+    big = [json.dumps({
+        "id": i,
+        "body": ("abcdefgh " * (kb * 114)) + str(i),
+    }).encode() for i in range(n_docs)]
+    bbatch = MessageBatch.from_binary(big).to(dev)
+    _s = N
+    N = n_docs
+    run(f"json {kb}KB strings (wave parse)",
+        JsonToArrowProcessor({"schema": {"id": "int", "body": "str"}}, None),
+        bbatch)
+    N = _s
