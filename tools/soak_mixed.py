@@ -1,5 +1,6 @@
-"""Mixed-workload soak: four different stream shapes concurrently —
-fused flagship (direct mode), durable WAL + sliding window + agg,
+"""Mixed-workload soak: five different stream shapes concurrently —
+fused flagship (direct mode), fused GROUP BY graph, durable WAL +
+sliding window + agg,
 JSON-decode + filter, BERT inference — through the full engine.
 Usage: soak_mixed.py [seconds]"""
 import asyncio
@@ -33,6 +34,20 @@ async def main():
                  "SELECT * FROM flow WHERE f0 >= 0.2"},
                 {"type": "inference", "model": "mlp_anomaly",
                  "columns": [f"f{i}" for i in range(16)]}]},
+            "output": {"type": "drop"},
+        },
+        {   # fused GROUP BY step graph (whole-step hipGraph agg)
+            "id": "fusedagg",
+            "input": {"type": "generate", "batch_size": 8192,
+                      "interval": "0ms",
+                      "fields": {"f0": {"dtype": "float32", "low": 0.0,
+                                        "high": 1.0},
+                                 "key": {"dtype": "int64", "low": 0,
+                                         "high": 1024}}},
+            "pipeline": {"thread_num": 1, "processors": [
+                {"type": "sql", "query":
+                 "SELECT key, count(*) AS c, sum(f0) AS s FROM flow "
+                 "WHERE f0 >= 0.2 GROUP BY key"}]},
             "output": {"type": "drop"},
         },
         {   # durable + windowed aggregation
