@@ -89,6 +89,10 @@ int launch_gemm_bf16_k64d(const void*, const void*, const float*, void*,
 void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
                        int, int64_t, int64_t, unsigned long long*,
                        hipStream_t);
+int launch_genfiltpack(const float*, const float*, int, int64_t, int64_t,
+                       int64_t, int, int, float, float* const*, int64_t*,
+                       void*, int, int32_t*, int32_t*, unsigned long long*,
+                       unsigned long long*, hipStream_t);
 void launch_scan_counts(const int32_t*, int, int32_t*, int32_t*,
                         hipStream_t);
 void launch_featpack(const float**, int, int, int64_t, void*, hipStream_t);
@@ -469,6 +473,66 @@ void gen_fields(torch::Tensor block, c10::optional<torch::Tensor> key,
                     wf.data(), nf, key_lo, key_range,
                     (unsigned long long*)ctr.data_ptr<int64_t>(),
                     cur_stream());
+}
+
+// One-launch generate+filter+compact(+featpack): see stepfused.hip
+// genfiltpack_kernel. outs = compacted float columns; key_out optional;
+// feats optional [cap, dpad] bf16. counts_ws int32[>=256] and bar int64[1]
+// are persistent workspaces (the barrier counter is monotonic across graph
+// replays — never reset them between steps).
+void genfiltpack(std::vector<double> lo, std::vector<double> width,
+                 int64_t key_lo, int64_t key_range, int64_t n, int64_t fidx,
+                 int64_t op, double scalar,
+                 std::vector<torch::Tensor> outs,
+                 c10::optional<torch::Tensor> key_out,
+                 c10::optional<torch::Tensor> feats,
+                 torch::Tensor count_out, torch::Tensor counts_ws,
+                 torch::Tensor bar, torch::Tensor ctr) {
+  int nf = (int)outs.size();
+  TORCH_CHECK(nf >= 1 && nf <= 32, "1..32 float columns");
+  TORCH_CHECK((int)lo.size() == nf && (int)width.size() == nf);
+  TORCH_CHECK(fidx >= 0 && fidx < nf, "fidx out of range");
+  TORCH_CHECK(count_out.scalar_type() == torch::kInt32 &&
+              count_out.numel() == 1);
+  TORCH_CHECK(counts_ws.scalar_type() == torch::kInt32 &&
+              counts_ws.numel() >= (n + 1023) / 1024);
+  TORCH_CHECK(bar.scalar_type() == torch::kInt64 && bar.numel() >= 1);
+  TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.numel() == 2);
+  std::vector<float> lof(nf), wf(nf);
+  std::vector<float*> optr(nf);
+  for (int f = 0; f < nf; ++f) {
+    lof[f] = (float)lo[f];
+    wf[f] = (float)width[f];
+    check_cuda(outs[f], "out col");
+    TORCH_CHECK(outs[f].numel() >= n &&
+                outs[f].scalar_type() == torch::kFloat32);
+    optr[f] = outs[f].data_ptr<float>();
+  }
+  int64_t* kptr = nullptr;
+  if (key_out.has_value() && key_out->defined()) {
+    TORCH_CHECK(key_out->numel() >= n &&
+                key_out->scalar_type() == torch::kInt64);
+    kptr = key_out->data_ptr<int64_t>();
+  } else {
+    key_range = 0;
+  }
+  void* fptr = nullptr;
+  int dpad = 0;
+  if (feats.has_value() && feats->defined()) {
+    TORCH_CHECK(feats->dim() == 2 &&
+                feats->scalar_type() == torch::kBFloat16 &&
+                feats->size(0) >= n && feats->size(1) >= nf);
+    fptr = feats->data_ptr();
+    dpad = (int)feats->size(1);
+  }
+  int rc = launch_genfiltpack(
+      lof.data(), wf.data(), nf, key_lo, key_range, n, (int)fidx, (int)op,
+      (float)scalar, optr.data(), kptr, fptr, dpad,
+      count_out.data_ptr<int32_t>(), counts_ws.data_ptr<int32_t>(),
+      (unsigned long long*)bar.data_ptr<int64_t>(),
+      (unsigned long long*)ctr.data_ptr<int64_t>(), cur_stream());
+  TORCH_CHECK(rc == 0, "genfiltpack: batch too large for the in-kernel "
+              "barrier (max 256K rows) — use the multi-kernel chain");
 }
 
 // gathered f32 feature columns → [n, kpad] bf16 MFMA operand in one launch
@@ -1329,6 +1393,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bytes_hash", &bytes_hash);
   m.def("json_decode", &json_decode);
   m.def("fused_filter_gather", &fused_filter_gather);
+  m.def("genfiltpack", &genfiltpack, py::arg("lo"), py::arg("width"),
+        py::arg("key_lo"), py::arg("key_range"), py::arg("n"),
+        py::arg("fidx"), py::arg("op"), py::arg("scalar"), py::arg("outs"),
+        py::arg("key_out"), py::arg("feats"), py::arg("count_out"),
+        py::arg("counts_ws"), py::arg("bar"), py::arg("ctr"));
   m.def("fused_filter_agg", &fused_filter_agg);
   m.def("hash_agg_capture", &hash_agg_capture, py::arg("keys"),
         py::arg("nrow"), py::arg("vals"), py::arg("ops"),

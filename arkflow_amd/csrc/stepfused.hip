@@ -109,6 +109,165 @@ extern "C" void launch_scan_counts(const int32_t* counts, int nb,
   scan_counts_kernel<<<1, 1024, 0, st>>>(counts, nb, offs, total);
 }
 
+// ---- whole-front fusion: generate + filter + compact + featpack ------------
+// One persistent kernel replaces the gen_fields → filter count → scan →
+// scatter → gather → featpack chain (5-6 launches, each a ~4-5 us slot at
+// small grids). The RNG is a pure counter-hash, so phase C REGENERATES the
+// surviving rows' values instead of re-reading a staged block — the
+// generated batch never touches memory at all. One in-kernel grid barrier
+// (monotonic agent-scope counter; launcher caps the grid at 256 blocks so
+// all blocks are co-resident) orders the per-block counts before the
+// compacted writes. Reference analog: the whole filter stage of
+// stream/mod.rs:370-444 + DataFusion FilterExec as one device program.
+struct GfpOuts {
+  float* outs[32];
+  int64_t* key_out;
+  __bf16* feats;  // nullptr = no feature packing
+  int32_t* count_out;
+  int32_t* counts_ws;       // [grid]
+  unsigned long long* bar;  // [1] monotonic barrier generation counter
+  unsigned long long* ctr;  // [2] RNG replay counter + advance ticket
+};
+
+#define GFP_CH 1024
+
+__global__ __launch_bounds__(256)
+void genfiltpack_kernel(GenSpec spec, int64_t n, int fidx, int op,
+                        float scalar, int dpad, GfpOuts o) {
+  const uint64_t c = o.ctr[0];
+  const int b = blockIdx.x, grid = gridDim.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int64_t start = (int64_t)b * GFP_CH;
+  __shared__ uint16_t loc[GFP_CH];
+  __shared__ int wave_base[5];
+  __shared__ int lcount_sh;
+  __shared__ int base_sh;
+  if (tid == 0) lcount_sh = 0;
+  __syncthreads();
+  // phase A: predicate on the filter field (recomputed from the counter
+  // RNG), ordered in-block compaction of local row ids
+  for (int s = 0; s < GFP_CH / 256; ++s) {
+    int64_t idx = start + s * 256 + tid;
+    bool pred = false;
+    if (idx < n) {
+      uint64_t rb = (c << 32) ^ (uint64_t)idx;
+      uint64_t h = mix64(rb + ((uint64_t)(fidx + 1) << 52));
+      float v = spec.lo[fidx] + spec.width[fidx] * u01(h);
+      switch (op) {
+        case 0: pred = v < scalar; break;
+        case 1: pred = v <= scalar; break;
+        case 2: pred = v > scalar; break;
+        case 3: pred = v >= scalar; break;
+        case 4: pred = v == scalar; break;
+        default: pred = v != scalar; break;  // != NaN ⇒ keep-all
+      }
+    }
+    uint64_t ballot = __ballot(pred);
+    int rank = __popcll(ballot & lanemask_lt());
+    if (lane == 0) wave_base[wid + 1] = __popcll(ballot);
+    __syncthreads();
+    if (tid == 0) {
+      wave_base[0] = lcount_sh;
+      for (int w = 1; w <= 4; ++w) wave_base[w] += wave_base[w - 1];
+      lcount_sh = wave_base[4];
+    }
+    __syncthreads();
+    if (pred) loc[wave_base[wid] + rank] = (uint16_t)(s * 256 + tid);
+    __syncthreads();
+  }
+  const int lcount = lcount_sh;
+  if (tid == 0)
+    __hip_atomic_store(&o.counts_ws[b], lcount, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_AGENT);
+  // grid barrier: generation = floor(old/grid)+1 — monotonic, so graph
+  // replays need no reset and stragglers can never miss the flag
+  if (tid == 0) {
+    unsigned long long old = atomicAdd(o.bar, 1ull);
+    unsigned long long target =
+        (old / grid + 1) * (unsigned long long)grid;
+    while (__hip_atomic_load(o.bar, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_AGENT) < target)
+      __builtin_amdgcn_s_sleep(8);
+  }
+  __syncthreads();
+  // phase B: my output base + grand total from the per-block counts
+  if (tid == 0) {
+    int basev = 0, total = 0;
+    for (int j = 0; j < grid; ++j) {
+      int v = __hip_atomic_load(&o.counts_ws[j], __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+      if (j < b) basev += v;
+      total += v;
+    }
+    base_sh = basev;
+    if (b == 0) *o.count_out = total;
+  }
+  __syncthreads();
+  const int base_out = base_sh;
+  // phase C: regenerate ALL fields for surviving rows; write the compacted
+  // columns and the bf16 feature rows
+  for (int j = tid; j < lcount; j += 256) {
+    int64_t r = start + loc[j];
+    int64_t dst = base_out + j;
+    uint64_t rb = (c << 32) ^ (uint64_t)r;
+    for (int f = 0; f < spec.nf; ++f) {
+      uint64_t h = mix64(rb + ((uint64_t)(f + 1) << 52));
+      float v = spec.lo[f] + spec.width[f] * u01(h);
+      o.outs[f][dst] = v;
+      if (o.feats) o.feats[dst * dpad + f] = (__bf16)v;
+    }
+    if (spec.key_range > 0) {
+      uint64_t h = mix64(rb + (0x9E37ull << 48));
+      o.key_out[dst] =
+          spec.key_lo + (int64_t)(h % (uint64_t)spec.key_range);
+    }
+  }
+  // advance the replay counter once per launch (grid ticket)
+  __syncthreads();
+  if (tid == 0) {
+    unsigned long long t = atomicAdd(&o.ctr[1], 1ull);
+    if (t == (unsigned long long)grid - 1) {
+      o.ctr[1] = 0ull;
+      o.ctr[0] = c + 1ull;
+      __threadfence();
+    }
+  }
+}
+
+extern "C" int launch_genfiltpack(const float* lo, const float* width,
+                                  int nf, int64_t key_lo, int64_t key_range,
+                                  int64_t n, int fidx, int op, float scalar,
+                                  float* const* outs, int64_t* key_out,
+                                  void* feats, int dpad, int32_t* count_out,
+                                  int32_t* counts_ws,
+                                  unsigned long long* bar,
+                                  unsigned long long* ctr, hipStream_t st) {
+  int grid = (int)((n + GFP_CH - 1) / GFP_CH);
+  // > 256 blocks: co-residency no longer trivially guaranteed for the
+  // in-kernel barrier — caller falls back to the multi-kernel chain
+  if (grid < 1 || grid > 256 || nf > 32) return -1;
+  GenSpec spec{};
+  spec.nf = nf;
+  for (int f = 0; f < nf; ++f) {
+    spec.lo[f] = lo[f];
+    spec.width[f] = width[f];
+  }
+  spec.key_lo = key_lo;
+  spec.key_range = key_range;
+  GfpOuts o{};
+  for (int f = 0; f < nf; ++f) o.outs[f] = outs[f];
+  o.key_out = key_out;
+  o.feats = (__bf16*)feats;
+  o.count_out = count_out;
+  o.counts_ws = counts_ws;
+  o.bar = bar;
+  o.ctr = ctr;
+  genfiltpack_kernel<<<grid, 256, 0, st>>>(spec, n, fidx, op, scalar, dpad,
+                                           o);
+  return 0;
+}
+
 // ---- featpack: nf gathered f32 columns → [n, kpad] bf16 --------------------
 struct PackSpec {
   const float* src[32];

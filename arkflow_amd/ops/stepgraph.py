@@ -112,15 +112,29 @@ class FusedGenerateFilterInfer:
             self._key_range = max(hi - self._key_lo, 1)
         else:
             self._key_lo = self._key_range = 0
+        # genfiltpack workspaces: per-block counts + the MONOTONIC grid
+        # barrier counter (never reset — graph replays rely on it)
+        self._gfp_counts = torch.zeros(256, device=self.device,
+                                       dtype=torch.int32)
+        self._gfp_bar = torch.zeros(1, device=self.device, dtype=torch.int64)
         self._graph = None
         self._scores = None
 
-    # ------------------------------------------------------------------ body
-    def _body(self) -> torch.Tensor:
-        """One step as a pure kernel chain (9 launches):
-        gen_fields → filter count → scan → scatter → gather → featpack →
-        GEMM(relu) → GEMM(relu) → GEMV+f32. No host syncs, no torch
-        elementwise tail — every stage is a stepfused.hip/gemm kernel."""
+    def _front(self, feats) -> None:
+        """generate → filter → compact (→ featpack): ONE persistent kernel
+        when the batch fits the in-kernel grid barrier (≤256K rows), else
+        the multi-kernel chain. Survivor values are REGENERATED from the
+        counter RNG in the fused kernel — no staged block traffic."""
+        if self.n <= 256 * 1024 and len(self.float_names) <= 32 \
+                and self.filter_col in self.float_names:
+            self.nat.genfiltpack(
+                self._lo, self._width, self._key_lo, self._key_range,
+                self.n, self.float_names.index(self.filter_col), self.op,
+                self.scalar, [self.outs[f] for f in self.float_names],
+                self.outs[self.int_names[0]] if self.int_names else None,
+                feats, self.count, self._gfp_counts, self._gfp_bar,
+                self.ctr)
+            return
         cols: Dict[str, torch.Tensor] = {
             f: self.block[i] for i, f in enumerate(self.float_names)}
         if self.key is not None:
@@ -132,9 +146,18 @@ class FusedGenerateFilterInfer:
         self.nat.filter_gather_capture(
             ordered, fidx, self.op, self.scalar,
             [self.outs[f] for f in self.fields], self.count)
-        # garbage rows past `count` are scored too, then sliced off
-        self.nat.featpack([self.outs[f] for f in self.float_names],
-                          self.feats)
+        if feats is not None:
+            self.nat.featpack([self.outs[f] for f in self.float_names],
+                              feats)
+
+    # ------------------------------------------------------------------ body
+    def _body(self) -> torch.Tensor:
+        """One step as a pure kernel chain (4 launches at default batch):
+        genfiltpack (generate+filter+compact+featpack fused) →
+        GEMM(relu) → GEMM(relu) → GEMV+f32. No host syncs, no torch
+        elementwise tail — every stage is a stepfused.hip/gemm kernel.
+        Garbage rows past `count` are scored too, then sliced off."""
+        self._front(self.feats)
         return self.mlp._net(self.feats)
 
     def capture(self) -> None:
@@ -250,19 +273,16 @@ class FusedGenerateAgg:
                 pairs.append((col, "sum" if fn == "avg" else fn))
         self._val_list = list(dict.fromkeys(pairs))
         self._val_ops = [ops_code[o] for _, o in self._val_list]
+        self._gfp_counts = torch.zeros(256, device=self.device,
+                                       dtype=torch.int32)
+        self._gfp_bar = torch.zeros(1, device=self.device, dtype=torch.int64)
         self._graph = None
 
+    # shares the one-kernel generate+filter front with the infer graph
+    _front = FusedGenerateFilterInfer._front
+
     def _body(self):
-        cols: Dict[str, torch.Tensor] = {
-            f: self.block[i] for i, f in enumerate(self.float_names)}
-        cols[self.key_col] = self.key
-        self.nat.gen_fields(self.block, self.key, self._lo, self._width,
-                            self._key_lo, self._key_range, self.ctr)
-        ordered = [cols[f] for f in self.fields]
-        fidx = list(self.fields).index(self.filter_col)
-        self.nat.filter_gather_capture(
-            ordered, fidx, self.op, self.scalar,
-            [self.outs[f] for f in self.fields], self.count)
+        self._front(None)
         uniq, counts, red = self.nat.hash_agg_capture(
             self.outs[self.key_col], self.count,
             [self.outs[c] for c, _ in self._val_list], self._val_ops,
