@@ -32,6 +32,36 @@ from . import require_native
 _OPS = {"<": 0, "<=": 1, ">": 2, ">=": 3, "==": 4, "!=": 5}
 
 
+class _LazyStepBatch(MessageBatch):
+    """MessageBatch whose column views materialize on first access.
+
+    The fused step's outputs are zero-copy slices of the graph's static
+    buffers — building 18 Column objects per step costs more host time than
+    the whole GPU step. Consumers that read columns get the identical views;
+    consumers that only route/count/drop (the common sink path) skip the
+    construction entirely. Engine mode (clone=True) always materializes —
+    retained batches must detach from the static buffers before the next
+    replay, exactly as before."""
+
+    __slots__ = ("_build", "_cols", "_n")
+
+    def __init__(self, build, n, input_name=None):
+        self._build = build
+        self._cols = None
+        self._n = n
+        self.input_name = input_name
+
+    @property
+    def columns(self):
+        if self._cols is None:
+            self._cols = self._build()
+        return self._cols
+
+    @property
+    def num_rows(self):
+        return self._n
+
+
 class FusedGenerateFilterInfer:
     """generate(fields) → WHERE col OP scalar → mlp(score) as one hipGraph.
 
@@ -178,8 +208,9 @@ class FusedGenerateFilterInfer:
             return int(self.count_host[0])
         return int(self.count.item())
 
-    def _make_batch(self, clone: bool = False) -> MessageBatch:
-        kept = self._kept()
+    def _make_batch(self, clone: bool = False, rows: int = -1
+                    ) -> MessageBatch:
+        kept = self._kept() if rows < 0 else rows
         cols = {f: Column("numeric", self.outs[f][:kept])
                 for f in self.fields}
         cols["score"] = Column("numeric", self._scores[:kept])
@@ -312,8 +343,9 @@ class FusedGenerateAgg:
                 f"group-by overflow: {g} groups > g_cap {self.g_cap}")
         return g
 
-    def _make_batch(self, clone: bool = False) -> MessageBatch:
-        g = self._kept()
+    def _make_batch(self, clone: bool = False, rows: int = -1
+                    ) -> MessageBatch:
+        g = self._kept() if rows < 0 else rows
         rmap = {pair: t for pair, t in zip(self._val_list, self._red)}
         cols: Dict[str, Column] = {}
         for fn, col, alias in self.aggs:
@@ -438,9 +470,17 @@ class FusedStepSource:
 
     def _consume(self, i: int):
         with torch.cuda.stream(self.streams[i]):
-            # clone=True (engine mode): queues / window buffers retain
-            # batches past the next replay — detach from the static buffers
-            return self.insts[i]._make_batch(clone=self.clone)
+            inst = self.insts[i]
+            if self.clone:
+                # engine mode: queues / window buffers retain batches past
+                # the next replay — detach from the static buffers NOW
+                return inst._make_batch(clone=True)
+            # bench/direct mode: sync for the count, defer the (zero-copy)
+            # column views until something actually reads them
+            kept = inst._kept()
+            return _LazyStepBatch(
+                lambda k=kept: inst._make_batch(rows=k).columns, kept,
+                input_name="generate")
 
     async def read(self):
         from ..spi import NoopAck
