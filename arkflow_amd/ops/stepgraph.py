@@ -43,12 +43,13 @@ class _LazyStepBatch(MessageBatch):
     retained batches must detach from the static buffers before the next
     replay, exactly as before."""
 
-    __slots__ = ("_build", "_cols", "_n")
+    __slots__ = ("_build", "_cols", "_n", "_dev")
 
-    def __init__(self, build, n, input_name=None):
+    def __init__(self, build, n, device, input_name=None):
         self._build = build
         self._cols = None
         self._n = n
+        self._dev = device
         self.input_name = input_name
 
     @property
@@ -60,6 +61,10 @@ class _LazyStepBatch(MessageBatch):
     @property
     def num_rows(self):
         return self._n
+
+    @property
+    def device(self):  # MessageBatch.device walks columns — don't
+        return self._dev
 
 
 class FusedGenerateFilterInfer:
@@ -480,7 +485,7 @@ class FusedStepSource:
             kept = inst._kept()
             return _LazyStepBatch(
                 lambda k=kept: inst._make_batch(rows=k).columns, kept,
-                input_name="generate")
+                inst.device, input_name="generate")
 
     async def read(self):
         from ..spi import NoopAck
