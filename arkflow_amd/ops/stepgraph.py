@@ -461,8 +461,9 @@ class FusedStepSource:
         for _ in range(max(ninstances, 1) - 1):
             self.insts.append(make_instance() if make_instance else fused)
             self.streams.append(torch.cuda.Stream())
-        self._i = 0
-        self._pending = None  # index replayed but not yet consumed
+        from collections import deque
+        self._q = deque()  # replayed-but-not-consumed instance ring
+        self._next = 0
         for inst, stream in zip(self.insts, self.streams):
             if inst._graph is None:
                 with torch.cuda.stream(stream):
@@ -487,18 +488,25 @@ class FusedStepSource:
                 lambda k=kept: inst._make_batch(rows=k).columns, kept,
                 inst.device, input_name="generate")
 
+    def _launch_next(self) -> None:
+        self._replay(self._next)
+        self._q.append(self._next)
+        self._next = (self._next + 1) % len(self.insts)
+
     async def read(self):
         from ..spi import NoopAck
         if len(self.insts) == 1:
             self._replay(0)
             return self._consume(0), NoopAck()
-        if self._pending is None:  # prime the pipeline
-            self._replay(0)
-            self._pending = 0
-        nxt = 1 - self._pending
-        self._replay(nxt)  # overlaps with draining the pending step
-        batch = self._consume(self._pending)
-        self._pending = nxt
+        # keep N-1 replays in flight while consuming the oldest: the newest
+        # launch goes in BEFORE the oldest's sync so the GPU never drains.
+        # An instance is re-replayed exactly one read() after its batch was
+        # handed out — the documented buffer-validity contract.
+        while len(self._q) < len(self.insts) - 1:
+            self._launch_next()
+        i = self._q.popleft()
+        self._launch_next()
+        batch = self._consume(i)
         return batch, NoopAck()
 
     async def connect(self):  # pragma: no cover - trivial

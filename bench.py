@@ -72,8 +72,11 @@ def build_pipeline(args, device):
                      ("sum", "f0", "s")], device,
                     seed=7 + args.rank + seed_off * 1000)
 
-            src = FusedStepSource(make_agg(0), ninstances=2,
-                                  make_instance=lambda: make_agg(1))
+            seeds = iter(range(1, 16))
+            src = FusedStepSource(
+                make_agg(0),
+                ninstances=int(os.environ.get("ARKFLOW_NINST", "2")),
+                make_instance=lambda: make_agg(next(seeds)))
             return src, Pipeline([])
         agg = SqlProcessor({
             "query": "SELECT key, count(*) AS c, sum(f0) AS s FROM flow "
@@ -156,8 +159,10 @@ def build_pipeline(args, device):
 
         # two graph instances software-pipeline inside the source; the
         # engine loop itself stays sequential (workers=1)
-        src = FusedStepSource(make(0), ninstances=2,
-                              make_instance=lambda: make(1))
+        seeds = iter(range(1, 16))
+        src = FusedStepSource(
+            make(0), ninstances=int(os.environ.get("ARKFLOW_NINST", "2")),
+            make_instance=lambda: make(next(seeds)))
         return src, Pipeline([])
     infer = InferenceProcessor({
         "model": "mlp_anomaly",
