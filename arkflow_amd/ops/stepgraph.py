@@ -94,14 +94,6 @@ class FusedGenerateFilterInfer:
                             if str(fields[f].get("dtype", "float32"))
                             not in ("int32", "int64")]
         self.int_names = [f for f in names if f not in self.float_names]
-        self._scale = torch.tensor(
-            [[float(fields[f].get("high", 100.0)) -
-              float(fields[f].get("low", 0.0))] for f in self.float_names],
-            device=self.device, dtype=torch.float32)
-        self._offset = torch.tensor(
-            [[float(fields[f].get("low", 0.0))] for f in self.float_names],
-            device=self.device, dtype=torch.float32)
-
         if len(self.int_names) > 1:
             raise ValueError("fused generate supports at most one int64 "
                              "key column")
@@ -426,17 +418,21 @@ class FusedProtoMlp:
         with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
             self._out_i, self._out_f, self._err, self._scores = self._body()
 
-    def step(self) -> MessageBatch:
-        if self._graph is None:
-            self.capture()
-        self._graph.replay()
+    def _cols(self):
         cols = {}
         for i, name in enumerate(self.int_names):
             cols[name] = Column("numeric", self._out_i[i])
         for i, name in enumerate(self.float_names):
             cols[name] = Column("numeric", self._out_f[i])
         cols["score"] = Column("numeric", self._scores)
-        return MessageBatch(cols, input_name="kafka")
+        return cols
+
+    def step(self) -> MessageBatch:
+        if self._graph is None:
+            self.capture()
+        self._graph.replay()
+        return _LazyStepBatch(self._cols, self.n, self.device,
+                              input_name="kafka")
 
 
 class FusedStepSource:
