@@ -129,16 +129,17 @@ struct GfpOuts {
   unsigned long long* ctr;  // [2] RNG replay counter + advance ticket
 };
 
-#define GFP_CH 1024
+#define GFP_CH 1024  // MAX rows per block; the launcher shrinks the chunk
+                     // so small batches still spread over >=32 blocks
 
 __global__ __launch_bounds__(256)
-void genfiltpack_kernel(GenSpec spec, int64_t n, int fidx, int op,
+void genfiltpack_kernel(GenSpec spec, int64_t n, int ch, int fidx, int op,
                         float scalar, int dpad, GfpOuts o) {
   const uint64_t c = o.ctr[0];
   const int b = blockIdx.x, grid = gridDim.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
-  const int64_t start = (int64_t)b * GFP_CH;
+  const int64_t start = (int64_t)b * ch;
   __shared__ uint16_t loc[GFP_CH];
   __shared__ int wave_base[5];
   __shared__ int lcount_sh;
@@ -147,7 +148,7 @@ void genfiltpack_kernel(GenSpec spec, int64_t n, int fidx, int op,
   __syncthreads();
   // phase A: predicate on the filter field (recomputed from the counter
   // RNG), ordered in-block compaction of local row ids
-  for (int s = 0; s < GFP_CH / 256; ++s) {
+  for (int s = 0; s < ch / 256; ++s) {
     int64_t idx = start + s * 256 + tid;
     bool pred = false;
     if (idx < n) {
@@ -243,7 +244,9 @@ extern "C" int launch_genfiltpack(const float* lo, const float* width,
                                   int32_t* counts_ws,
                                   unsigned long long* bar,
                                   unsigned long long* ctr, hipStream_t st) {
-  int grid = (int)((n + GFP_CH - 1) / GFP_CH);
+  int ch = GFP_CH;
+  while (ch > 256 && (n + ch - 1) / ch < 64) ch >>= 1;
+  int grid = (int)((n + ch - 1) / ch);
   // > 256 blocks: co-residency no longer trivially guaranteed for the
   // in-kernel barrier — caller falls back to the multi-kernel chain
   if (grid < 1 || grid > 256 || nf > 32) return -1;
@@ -263,8 +266,8 @@ extern "C" int launch_genfiltpack(const float* lo, const float* width,
   o.counts_ws = counts_ws;
   o.bar = bar;
   o.ctr = ctr;
-  genfiltpack_kernel<<<grid, 256, 0, st>>>(spec, n, fidx, op, scalar, dpad,
-                                           o);
+  genfiltpack_kernel<<<grid, 256, 0, st>>>(spec, n, ch, fidx, op, scalar,
+                                           dpad, o);
   return 0;
 }
 
