@@ -198,20 +198,56 @@ class KafkaInput(Input):
             self._real.close()
 
 
+def _expr_cfg(v):
+    """Reference Expr<T> config shape (expr/mod.rs:29-210): a plain value,
+    or {"expr": "<sql expression>"} evaluated per row against the batch."""
+    if isinstance(v, dict) and "expr" in v:
+        from ..sql.parser import parse_expression
+        return ("expr", parse_expression(str(v["expr"])))
+    return ("value", None if v is None else str(v))
+
+
+def _eval_expr_rows(cfg, batch: MessageBatch) -> List[Optional[str]]:
+    kind, v = cfg
+    n = batch.num_rows
+    if kind == "value":
+        return [v] * n
+    import torch
+
+    from ..sql.eval import Env, eval_expr
+    out = eval_expr(v, Env(batch.columns, n, batch.device))
+    if isinstance(out, Column):
+        return [x.decode("utf-8", "replace")
+                if isinstance(x, (bytes, bytearray)) else str(x)
+                for x in out.to_pylist()]
+    if isinstance(out, torch.Tensor):
+        vals = out.cpu().tolist()
+        return [str(x) for x in vals]
+    return [str(out)] * n
+
+
 class KafkaOutput(Output):
     """Exactly-once: write_batch = one transaction (output/kafka.rs:348-446).
 
     The fake bus stages rows and appends atomically on commit; consumers only
-    ever see committed rows (read_committed)."""
+    ever see committed rows (read_committed). ``topic`` and ``key`` accept
+    either a constant or {"expr": "<sql>"} evaluated per row — the
+    reference's Expr<T> dynamic routing (output/kafka.rs topic/key)."""
 
     def __init__(self, config: dict, resource=None):
         self.brokers = config.get("brokers", ["memory://default"])
         if isinstance(self.brokers, str):
             self.brokers = [self.brokers]
-        self.topic_expr = config.get("topic")
-        if not self.topic_expr:
+        topic = config.get("topic")
+        if not topic:
             raise ConfigError("kafka output requires 'topic'")
+        self.topic_cfg = _expr_cfg(topic)
+        self.topic_expr = topic if isinstance(topic, str) else None
+        self.key_cfg = (_expr_cfg(config["key"])
+                        if isinstance(config.get("key"), dict) else None)
         self.key_column = config.get("key_column")
+        if self.key_column is None and isinstance(config.get("key"), str):
+            self.key_column = config["key"]
         self.value_column = config.get("value_column", "__value__")
         self.exactly_once = bool(config.get("exactly_once", False))
         self.transactional_id = config.get("transactional_id")
@@ -227,7 +263,8 @@ class KafkaOutput(Output):
         if self.driver == "memory":
             name = str(self.brokers[0]).removeprefix("memory://") or "default"
             self.bus = FakeBus.get(name)
-            self.bus.ensure_topic(self.topic_expr)
+            if self.topic_expr:  # dynamic topics materialize on first write
+                self.bus.ensure_topic(self.topic_expr)
             return
         from .kafka_real import RealKafkaProducer
         self._real = RealKafkaProducer(
@@ -236,28 +273,36 @@ class KafkaOutput(Output):
             compression=self.compression, config=self.client_config)
         self._real.connect()
 
-    def _rows(self, batch: MessageBatch) -> List[Tuple[Optional[bytes], bytes]]:
+    def _rows(self, batch: MessageBatch
+              ) -> List[Tuple[str, Optional[bytes], bytes]]:
+        """(topic, key, value) per row — topic/key evaluated per row when
+        configured as expressions."""
         col = batch.columns.get(self.value_column)
         if col is not None and col.kind == "binary":
             values = col.to_pylist()
         else:
             values = batch.to_json_lines()
+        topics = _eval_expr_rows(self.topic_cfg, batch)
         keys: List[Optional[bytes]] = [None] * len(values)
-        if self.key_column and self.key_column in batch.columns:
+        if self.key_cfg is not None:
+            keys = [None if k is None else k.encode()
+                    for k in _eval_expr_rows(self.key_cfg, batch)]
+        elif self.key_column and self.key_column in batch.columns:
             kc = batch.column(self.key_column)
             keys = [str(v).encode() if not isinstance(v, (bytes, bytearray))
                     else bytes(v) for v in kc.to_pylist()]
-        return list(zip(keys, values))
+        return list(zip(topics, keys, values))
 
     async def write(self, batch: MessageBatch) -> None:
+        rows = self._rows(batch)
         if self._real is not None:
             loop = asyncio.get_running_loop()
-            rows = self._rows(batch)
             await loop.run_in_executor(
                 None, self._real.write_plain, self.topic_expr, rows)
             return
-        for key, value in self._rows(batch):
-            self.bus.produce(self.topic_expr, key, value)
+        for topic, key, value in rows:
+            self.bus.ensure_topic(topic)
+            self.bus.produce(topic, key, value)
 
     async def write_batch(self, batches) -> None:
         if not self.exactly_once:
@@ -276,8 +321,9 @@ class KafkaOutput(Output):
                 None, self._real.write_txn, self.topic_expr, staged)
             return
         # commit point — a failure above leaves the log untouched
-        for key, value in staged:
-            self.bus.produce(self.topic_expr, key, value)
+        for topic, key, value in staged:
+            self.bus.ensure_topic(topic)
+            self.bus.produce(topic, key, value)
 
     async def close(self) -> None:
         if self._real is not None:

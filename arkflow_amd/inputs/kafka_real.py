@@ -186,22 +186,22 @@ class RealKafkaProducer:
                     f"kafka producer fenced/fatal during {where}: {kerr}")
         return ProcessError(f"kafka {where} failed: {e}")
 
-    def produce_rows(self, topic: str,
-                     rows: List[Tuple[Optional[bytes], bytes]]) -> None:
-        for key, value in rows:
-            self.producer.produce(topic, value=value, key=key)
+    def produce_rows(self, topic: str, rows) -> None:
+        """rows: (key, value) pairs for the fixed topic, or
+        (topic, key, value) triples when the topic is a per-row expr."""
+        for row in rows:
+            t, key, value = row if len(row) == 3 else (topic, *row)
+            self.producer.produce(t, value=value, key=key)
             self.producer.poll(0)
 
-    def write_plain(self, topic: str,
-                    rows: List[Tuple[Optional[bytes], bytes]]) -> None:
+    def write_plain(self, topic: str, rows) -> None:
         try:
             self.produce_rows(topic, rows)
             self.producer.flush(30.0)  # delivery before the engine acks
         except Exception as e:  # noqa: BLE001
             raise self._map_error(e, "produce")
 
-    def write_txn(self, topic: str,
-                  rows: List[Tuple[Optional[bytes], bytes]]) -> None:
+    def write_txn(self, topic: str, rows) -> None:
         ck = _import_client()
         try:
             self.producer.begin_transaction()

@@ -349,9 +349,16 @@ def _eval_binop(e: BinaryOp, env: Env) -> Value:
     l_bin = isinstance(lv, Column) and lv.kind == "binary"
     r_bin = isinstance(rv, Column) and rv.kind == "binary"
     if e.op == "||":
-        if l_bin or r_bin or isinstance(lv, str) or isinstance(rv, str):
-            ls = lv.to_strlist() if l_bin else [str(lv)] * env.n_rows
-            rs = rv.to_strlist() if r_bin else [str(rv)] * env.n_rows
+        def _strs(v, is_bin):
+            if is_bin:
+                return v.to_strlist()
+            if isinstance(v, torch.Tensor):  # numeric column → per-row str
+                return [str(int(x)) if float(x).is_integer() else str(x)
+                        for x in v.cpu().tolist()]
+            return [str(v)] * env.n_rows
+        if l_bin or r_bin or isinstance(lv, (str, torch.Tensor))                 or isinstance(rv, (str, torch.Tensor)):
+            ls = _strs(lv, l_bin)
+            rs = _strs(rv, r_bin)
             return Column.from_strings([a + b for a, b in zip(ls, rs)])
         raise SqlError("|| requires strings")
     if l_bin or r_bin or isinstance(lv, str) or isinstance(rv, str):

@@ -545,3 +545,13 @@ class Parser:
 
 def parse_sql(sql: str) -> Select:
     return Parser(sql).parse()
+
+
+def parse_expression(text: str):
+    """Parse ONE standalone SQL expression (reference expr/mod.rs Expr{expr}:
+    per-row dynamic config values like a kafka output's topic/key)."""
+    p = Parser(text)
+    e = p._expr()
+    if p.peek().kind != "end":
+        raise SqlError(f"trailing input in expression: {text!r}")
+    return e

@@ -471,3 +471,34 @@ def test_redis_real_driver_roundtrip(run):
         assert got == [b"r1", b"r2"]
 
     run(main(), timeout=50)
+
+
+def test_kafka_output_expr_topic_and_key():
+    """Reference Expr<T> config: topic/key as per-row SQL expressions route
+    each row to its computed topic with its computed key
+    (output/kafka.rs topic/key Expr handling)."""
+    import asyncio
+
+    import torch
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.inputs.brokers import FakeBus, KafkaOutput
+
+    bus = FakeBus.get("exprout")
+    out = KafkaOutput({
+        "brokers": "memory://exprout",
+        "topic": {"expr": "'events-' || (key % 2)"},
+        "key": {"expr": "key * 10"},
+    })
+    batch = MessageBatch({
+        "key": Column("numeric", torch.tensor([0, 1, 2, 3], dtype=torch.int64)),
+        "v": Column("numeric", torch.tensor([1.0, 2.0, 3.0, 4.0])),
+    })
+    loop = asyncio.new_event_loop()
+    loop.run_until_complete(out.connect())
+    loop.run_until_complete(out.write(batch))
+    even = [m for part in bus.topics["events-0"] for m in part]
+    odd = [m for part in bus.topics["events-1"] for m in part]
+    assert len(even) == 2 and len(odd) == 2
+    assert [m[0] for m in even] == [b"0", b"20"]
+    assert [m[0] for m in odd] == [b"10", b"30"]
