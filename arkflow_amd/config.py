@@ -172,6 +172,10 @@ class EngineConfig:
                     continue
                 expect = checks.get(sub.get("type"))
                 val = spec[key]
+                if sub.get("type") == "string" and isinstance(val, dict) \
+                        and "expr" in val:
+                    continue  # Expr<T>: per-row SQL expression accepted
+                    # wherever a string constant is (reference expr/mod.rs)
                 if expect and not isinstance(val, expect) \
                         or (expect is int and isinstance(val, bool)):
                     errors.append(
