@@ -152,8 +152,10 @@ class FusedGenerateFilterInfer:
         when the batch fits the in-kernel grid barrier (≤256K rows), else
         the multi-kernel chain. Survivor values are REGENERATED from the
         counter RNG in the fused kernel — no staged block traffic."""
+        import os
         if self.n <= 256 * 1024 and len(self.float_names) <= 32 \
-                and self.filter_col in self.float_names:
+                and self.filter_col in self.float_names \
+                and os.environ.get("ARKFLOW_NO_GFP") != "1":
             self.nat.genfiltpack(
                 self._lo, self._width, self._key_lo, self._key_range,
                 self.n, self.float_names.index(self.filter_col), self.op,

@@ -1236,3 +1236,37 @@ def test_engine_stream_fuses_group_by_chain(dev):
         s = b.column("s").data
         assert bool((s >= 0.25 * c.float() - 1e-3).all())
         assert bool((s <= 1.0 * c.float() + 1e-3).all())
+
+
+def test_genfiltpack_matches_multi_kernel_chain(dev, monkeypatch):
+    """The one-kernel generate+filter+compact front must produce BYTE-
+    identical outputs to the multi-kernel chain (same counter RNG, same
+    stable order) — including the packed bf16 feature rows."""
+    from arkflow_amd.models.mlp import MlpAnomalyDetector
+    from arkflow_amd.ops.stepgraph import FusedGenerateFilterInfer
+
+    fields = {f"f{i}": {"dtype": "float32", "low": -1.0, "high": 3.0}
+              for i in range(9)}
+    fields["key"] = {"dtype": "int64", "low": 5, "high": 777}
+
+    def run(no_gfp):
+        if no_gfp:
+            monkeypatch.setenv("ARKFLOW_NO_GFP", "1")
+        else:
+            monkeypatch.delenv("ARKFLOW_NO_GFP", raising=False)
+        mlp = MlpAnomalyDetector(9, [32], dev, 3)
+        fused = FusedGenerateFilterInfer(fields, 8192, "f2", "<", 1.7,
+                                         mlp, dev, seed=77)
+        outs = []
+        for _ in range(3):
+            batch, kept = fused.step()
+            outs.append({k: c.data.clone() for k, c in batch.columns.items()})
+            outs[-1]["__feats"] = fused.feats[:kept].clone()
+        return outs
+
+    a = run(False)
+    b = run(True)
+    for step, (x, y) in enumerate(zip(a, b)):
+        assert x.keys() == y.keys()
+        for k in x:
+            assert torch.equal(x[k], y[k]), f"step {step} col {k}"
