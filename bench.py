@@ -191,12 +191,17 @@ async def run_steps(gen, pipeline, n_steps, workers=1):
         for b in outs:
             rows += b.num_rows
         if batch.device.type == "cuda":
-            # event-based wait in an executor thread so OTHER in-flight
-            # steps keep dispatching while this one drains (a blocking
-            # torch.cuda.synchronize would stall the whole event loop)
             ev = torch.cuda.Event()
             ev.record()
-            await loop.run_in_executor(None, ev.synchronize)
+            if workers <= 1:
+                # serial loop: the fused source already synced its stream,
+                # so the event is (near-)complete — skip the thread hop
+                ev.synchronize()
+            else:
+                # event-based wait in an executor thread so OTHER in-flight
+                # steps keep dispatching while this one drains (a blocking
+                # torch.cuda.synchronize would stall the whole event loop)
+                await loop.run_in_executor(None, ev.synchronize)
         times.append(time.perf_counter() - t0)
         await ack.ack()
 
