@@ -84,6 +84,8 @@ int launch_gemm_bf16_k64p(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
 int launch_gemm_bf16_k64w(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
+int launch_gemm_bf16_k64s3(const void*, const void*, const float*, void*,
+                          int, int, int, int, hipStream_t);
 int launch_gemm_bf16_k64d(const void*, const void*, const float*, void*,
                           int, int, int, int, hipStream_t);
 void launch_gen_fields(float*, int64_t*, int64_t, const float*, const float*,
@@ -611,6 +613,11 @@ torch::Tensor gemm_bf16_variant(torch::Tensor A, torch::Tensor Bt,
                                       C.data_ptr(), (int)M, (int)N, (int)K,
                                       (int)act, cur_stream()) == 0,
                 "k64w requires K % 64 == 0");
+  } else if (variant == 12) {
+    TORCH_CHECK(launch_gemm_bf16_k64s3(A.data_ptr(), Bt.data_ptr(), bias_ptr,
+                                       C.data_ptr(), (int)M, (int)N, (int)K,
+                                       (int)act, cur_stream()) == 0,
+                "k64s3 requires K % 64 == 0");
   } else if (variant == 9) {
     int rc = launch_gemm_bf16_k64d(A.data_ptr(), Bt.data_ptr(), bias_ptr,
                                    C.data_ptr(), (int)M, (int)N, (int)K,

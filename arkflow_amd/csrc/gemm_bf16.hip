@@ -257,8 +257,8 @@ void gemm_bf16_k64_kernel(const __bf16* __restrict__ A,   // [M,K]
 // chunk ^ (row & 7). global_load_lds writes a wave's 1 KiB LINEARLY from a
 // scalar base, so swizzled staging needs register staging + per-lane
 // ds_write_b128 instead (the async direct-to-LDS path cannot express it).
-template <int ACT, bool HAS_BIAS>
-__global__ __launch_bounds__(GEMM_THREADS, 2)
+template <int ACT, bool HAS_BIAS, int OCC = 2>
+__global__ __launch_bounds__(GEMM_THREADS, OCC)
 void gemm_bf16_k64s_kernel(const __bf16* __restrict__ A,   // [M,K]
                            const __bf16* __restrict__ Bt,  // [N,K]
                            const float* __restrict__ bias, // [N] or null
@@ -841,6 +841,37 @@ extern "C" int launch_gemm_bf16_k64p(const void* A, const void* Bt,
     default: KPDISPATCH(ACT_NONE); break;
   }
 #undef KPDISPATCH
+  return 0;
+}
+
+// occupancy-3 A/B of the same kernel (more inter-block latency overlap if
+// the register allocator fits 3 blocks; measured via variant 12)
+extern "C" int launch_gemm_bf16_k64s3(const void* A, const void* Bt,
+                                      const float* bias, void* C, int M,
+                                      int N, int K, int act, hipStream_t st) {
+  if (K % BK2 != 0) return -1;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  dim3 block(GEMM_THREADS);
+#define KS3DISPATCH(ACT)                                                     \
+  do {                                                                       \
+    if (bias)                                                                \
+      gemm_bf16_k64s_kernel<ACT, true, 3><<<grid, block, 0, st>>>(           \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+    else                                                                     \
+      gemm_bf16_k64s_kernel<ACT, false, 3><<<grid, block, 0, st>>>(          \
+          (const __bf16*)A, (const __bf16*)Bt, bias, (__bf16*)C, M, N, K,    \
+          tiles_n);                                                          \
+  } while (0)
+  switch (act) {
+    case ACT_RELU: KS3DISPATCH(ACT_RELU); break;
+    case ACT_GELU: KS3DISPATCH(ACT_GELU); break;
+    case ACT_SILU: KS3DISPATCH(ACT_SILU); break;
+    default: KS3DISPATCH(ACT_NONE); break;
+  }
+#undef KS3DISPATCH
   return 0;
 }
 

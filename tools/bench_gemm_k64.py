@@ -45,7 +45,7 @@ for M, N, K, act in SHAPES:
     if act != 0:
         rows.append(("blaslt+act", lambda: nat.bias_act_bf16(
             torch.nn.functional.linear(A, Bt, bias_h), None, act)))
-    for name, var in [("bk32", 0), ("k64swz", 8), ("k64pre", 10), ("k64wide", 11),
+    for name, var in [("bk32", 0), ("k64swz", 8), ("k64pre", 10), ("k64wide", 11), ("k64s3", 12),
                       ("8ph", 2), ("8ph128", 5)]:
         rows.append((name,
                      lambda v=var: nat.gemm_bf16_variant(A, Bt, bias, act, v)))
