@@ -35,6 +35,9 @@ class GenerateInput(Input):
         )
         interval = config.get("interval", "0ms")
         self.interval_secs = _parse_duration(interval)
+        # interval-0 generators never wait: the engine's direct loop may
+        # skip its per-read cancellation race
+        self.nonblocking = self.interval_secs <= 0
         self.fields = config.get("fields")  # {name: {dtype, low, high}}
         self.seed = int(config.get("seed", 0x5EED))
         self.device = torch.device(config.get("device")) if config.get("device") \

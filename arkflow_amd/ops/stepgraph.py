@@ -450,6 +450,10 @@ class FusedStepSource:
     after the sequential engine loop has processed its batch; callers that
     retain batches across steps must copy (the ring-buffer contract)."""
 
+    nonblocking = True  # read() completes in one GPU step — never waits
+    # on external IO, so the engine's direct loop may skip its per-read
+    # cancellation race
+
     def __init__(self, fused: FusedGenerateFilterInfer,
                  ninstances: int = 1, make_instance=None,
                  clone: bool = False):
