@@ -942,46 +942,37 @@ void gemm_bf16_skinny_kernel(const __bf16* __restrict__ A,   // [M,K]
 
   f32x4 acc[2][2] = {};
 
-  // staging: register loads + XOR-swizzled ds_write. The linear [64][32]
-  // layout has a 64 B row stride = half a bank period: a quarter-wave's 16
-  // fragment rows land on TWO bank groups (PMC r2-23: 1.5e5 conflicts per
-  // dispatch, 4 per MFMA). chunk' = chunk ^ (row & 3) spreads them;
-  // global_load_lds writes lane-linearly from a scalar base and cannot
-  // express the swizzle (same trade as the k64s kernel).
-  const int lin0 = tid * 16;    // 256 threads × 16 B = the whole 4 KiB tile
+  // staging: the 4 KiB operand tile = 4 waves × 1 KiB global_load_lds
+  const int lin0 = wid * 1024 + lane * 16;
   const int trow = lin0 >> 6;   // 64 B per row (BK=32 bf16)
   const int tcol = lin0 & 63;
-  const int dst = (trow << 6) + (((tcol >> 4) ^ (trow & 3)) << 4);
-  typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     int ga_row = row0 + trow;
     ga_row = ga_row < M ? ga_row : M - 1;
-    u32x4 va = *(const u32x4*)((const char*)(A + (int64_t)ga_row * K + k0)
-                               + tcol);
+    const char* a_src = (const char*)(A + (int64_t)ga_row * K + k0) + tcol;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)a_src,
+        (__attribute__((address_space(3))) uint32_t*)((char*)Asm + lin0),
+        16, 0, 0);
     int gb_row = col0 + trow;
     gb_row = gb_row < N ? gb_row : N - 1;
-    u32x4 vb = *(const u32x4*)((const char*)(Bt + (int64_t)gb_row * K + k0)
-                               + tcol);
-    *(u32x4*)((char*)Asm + dst) = va;
-    *(u32x4*)((char*)Bsm + dst) = vb;
+    const char* b_src = (const char*)(Bt + (int64_t)gb_row * K + k0) + tcol;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)b_src,
+        (__attribute__((address_space(3))) uint32_t*)((char*)Bsm + lin0),
+        16, 0, 0);
     __syncthreads();
 
     bf16x8 a_frag[2], b_frag[2];
     const int fr = lane & 15;
-    const int fk8 = lane >> 4;  // 16 B chunk within the row
+    const int fk = (lane >> 4) * 8;
 #pragma unroll
-    for (int m = 0; m < 2; ++m) {
-      int row = wm + m * 16 + fr;
-      int chunk = fk8 ^ (row & 3);
-      a_frag[m] = *(const bf16x8*)&Asm[row * BK + chunk * 8];
-    }
+    for (int m = 0; m < 2; ++m)
+      a_frag[m] = *(const bf16x8*)&Asm[(wm + m * 16 + fr) * BK + fk];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      int row = wn + n * 16 + fr;
-      int chunk = fk8 ^ (row & 3);
-      b_frag[n] = *(const bf16x8*)&Bsm[row * BK + chunk * 8];
-    }
+    for (int n = 0; n < 2; ++n)
+      b_frag[n] = *(const bf16x8*)&Bsm[(wn + n * 16 + fr) * BK + fk];
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
