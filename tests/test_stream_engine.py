@@ -604,3 +604,31 @@ def test_fusable_agg_chain_detection():
         {"query": "SELECT key, count(*) AS c FROM flow GROUP BY key"})
     spec2 = fusable_agg_chain(cfg, gen, [nowhere], R())
     assert spec2 is not None and spec2[1] is None
+
+
+def test_lazy_step_batch_semantics():
+    """_LazyStepBatch materializes columns exactly once, on first access;
+    num_rows/device never trigger materialization (the fused source hands
+    these out per step — building views eagerly cost more than the GPU
+    step, profiles r2-21)."""
+    import torch
+
+    from arkflow_amd.batch import Column
+    from arkflow_amd.ops.stepgraph import _LazyStepBatch
+
+    calls = []
+
+    def build():
+        calls.append(1)
+        return {"a": Column("numeric", torch.arange(4)),
+                "b": Column("numeric", torch.ones(4))}
+
+    b = _LazyStepBatch(build, 4, torch.device("cpu"), input_name="generate")
+    assert b.num_rows == 4 and len(b) == 4
+    assert b.device.type == "cpu"
+    assert b.input_name == "generate"
+    assert not calls  # nothing materialized yet
+    assert b.column("a").data.tolist() == [0, 1, 2, 3]
+    assert calls == [1]
+    assert set(b.column_names) == {"a", "b"}
+    assert calls == [1]  # cached — built once
