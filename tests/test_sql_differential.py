@@ -124,7 +124,7 @@ def _normalize(rows):
     return out
 
 
-@pytest.mark.parametrize("seed", [0, 1, 2, 3, 4, 5])
+@pytest.mark.parametrize("seed", [0, 1, 2, 3, 4, 5, 6, 7])
 @pytest.mark.parametrize("sql", QUERIES)
 def test_differential_vs_sqlite(seed, sql):
     rng = random.Random(seed * 1000 + zlib.crc32(sql.encode()) % 997)
