@@ -45,7 +45,7 @@ batch = MessageBatch({
                                          dtype=torch.int64)),
     "v": Column("numeric", torch.rand(8192, device=dev)),
 })
-outb = afdist.repartition(batch, "k")
+outb = afdist.repartition_by_key(batch, "k")
 assert outb.num_rows == 8192 and outb.column("v").data.is_cuda
 print("RCCL probe ok: all_reduce/all_gather/all_to_all/reduce_scatter +"
       " repartition on gfx950, backend", dist.get_backend())
