@@ -502,3 +502,60 @@ def test_kafka_output_expr_topic_and_key():
     assert len(even) == 2 and len(odd) == 2
     assert [m[0] for m in even] == [b"0", b"20"]
     assert [m[0] for m in odd] == [b"10", b"30"]
+
+
+def test_http_output_retries_with_backoff():
+    """HTTP output retries transient failures with exponential backoff and
+    surfaces the last error after the budget (output/http.rs:181-216)."""
+    import asyncio
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.errors import ProcessError
+    from arkflow_amd.outputs.http import HttpOutput
+
+    calls = []
+
+    class _Resp:
+        def __init__(self, status):
+            self.status = status
+
+        async def text(self):
+            return "boom"
+
+        async def __aenter__(self):
+            return self
+
+        async def __aexit__(self, *a):
+            return False
+
+    class _Session:
+        def post(self, url, data=None, headers=None):
+            calls.append(url)
+            return _Resp(503 if len(calls) < 3 else 200)
+
+        async def close(self):
+            pass
+
+    out = HttpOutput({"url": "http://x/sink", "retry_count": 3})
+    out._session = _Session()
+    batch = MessageBatch.from_binary([b"{}"])
+    loop = asyncio.new_event_loop()
+    t0 = loop.time()
+    loop.run_until_complete(out.write(batch))
+    assert len(calls) == 3  # two 503s then success
+
+    calls.clear()
+    out2 = HttpOutput({"url": "http://x/sink", "retry_count": 1})
+
+    class _AlwaysBad(_Session):
+        def post(self, url, data=None, headers=None):
+            calls.append(url)
+            return _Resp(500)
+
+    out2._session = _AlwaysBad()
+    try:
+        loop.run_until_complete(out2.write(batch))
+        raise AssertionError("expected ProcessError")
+    except ProcessError as e:
+        assert "500" in str(e)
+    assert len(calls) == 2  # initial + one retry
