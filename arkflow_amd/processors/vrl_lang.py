@@ -386,6 +386,45 @@ _STDLIB.update({
     and isinstance(b, dict) else _err("merge: need objects"),
     "encode_json": lambda v: json.dumps(v, separators=(",", ":")),
     "now": lambda: time.time(),
+    # --- hashing / encoding (vector stdlib: md5, sha1, sha2, base64) ---
+    "md5": lambda s: __import__("hashlib").md5(
+        _s(s).encode()).hexdigest(),
+    "sha1": lambda s: __import__("hashlib").sha1(
+        _s(s).encode()).hexdigest(),
+    "sha2": lambda s: __import__("hashlib").sha256(
+        _s(s).encode()).hexdigest(),
+    "sha256": lambda s: __import__("hashlib").sha256(
+        _s(s).encode()).hexdigest(),
+    "encode_base64": lambda s: __import__("base64").b64encode(
+        s if isinstance(s, (bytes, bytearray)) else _s(s).encode()
+    ).decode(),
+    "decode_base64": lambda s: __import__("base64").b64decode(
+        _s(s)).decode("utf-8", "replace"),
+    "uuid_v4": lambda: str(__import__("uuid").uuid4()),
+    # --- numbers / strings ---
+    "parse_int": lambda s, base=10: int(_s(s), int(base)),
+    "truncate": lambda s, n, suffix=False: (
+        _s(s) if len(_s(s)) <= int(n)
+        else _s(s)[: int(n)] + ("..." if suffix else "")),
+    "strip_whitespace": lambda s: _s(s).strip(),
+    # --- timestamps (float unix-seconds representation, like now()) ---
+    "to_unix_timestamp": lambda t: float(_num(t)),
+    "from_unix_timestamp": lambda t: float(_num(t)),
+    "format_timestamp": lambda t, fmt="%Y-%m-%dT%H:%M:%SZ": time.strftime(
+        fmt.replace("%f", "{us:06d}").format(
+            us=int((float(_num(t)) % 1) * 1e6))
+        if "%f" in fmt else fmt, time.gmtime(float(_num(t)))),
+    "parse_timestamp": lambda s, fmt="%Y-%m-%dT%H:%M:%SZ":
+        __import__("calendar").timegm(time.strptime(_s(s), fmt)),
+    # --- regex ---
+    "match": lambda s, pat: __import__("re").search(_s(pat), _s(s))
+    is not None,
+    "parse_regex": lambda s, pat: (
+        lambda m: m.groupdict() if m and m.groupdict()
+        else (dict(enumerate(m.groups(), 1)) and
+              {str(i): g for i, g in enumerate(m.groups(), 1)}) if m
+        else _err("parse_regex: no match"))(
+        __import__("re").search(_s(pat), _s(s))),
     "is_null": lambda v: v is None,
     "is_string": lambda v: isinstance(v, str),
     "is_int": lambda v: isinstance(v, int) and not isinstance(v, bool),

@@ -121,3 +121,42 @@ def test_builder_routes_flat_to_columnar_and_rest_to_interpreter():
     assert isinstance(flat, ExprProcessor)
     full = _build_vrl({"source": 'if .a > 1 { .b = "x" }'})
     assert isinstance(full, VrlProcessor)
+
+
+def test_stdlib_hash_encode_time_regex():
+    """Vector-stdlib parity batch 2: hashing, base64, parse_int, truncate,
+    timestamps, regex match/capture, uuid (vrl.rs stdlib surface)."""
+    import hashlib
+
+    from arkflow_amd.processors.vrl_lang import VrlProgram
+
+    out = VrlProgram('''
+.h = sha256(.name)
+.m5 = md5(.name)
+.b = encode_base64(.name)
+.rt = decode_base64(.b)
+.n = parse_int("ff", 16)
+.t = truncate(.name, 3)
+.ts = format_timestamp(1700000000, "%Y-%m-%d")
+.back = parse_timestamp("2023-11-14", "%Y-%m-%d")
+.m = match(.name, "^al")
+.g = parse_regex(.name, "(?P<first>a.)")
+.u = uuid_v4()
+''').remap({"name": "alice"})
+    assert out["h"] == hashlib.sha256(b"alice").hexdigest()
+    assert out["m5"] == hashlib.md5(b"alice").hexdigest()
+    assert out["rt"] == "alice" and out["n"] == 255 and out["t"] == "ali"
+    assert out["ts"] == "2023-11-14" and out["back"] == 1699920000
+    assert out["m"] is True and out["g"] == {"first": "al"}
+    assert len(out["u"]) == 36
+
+
+def test_stdlib_parse_regex_abort_on_no_match():
+    from arkflow_amd.processors.vrl_lang import VrlError, VrlProgram
+
+    try:
+        VrlProgram('.g = parse_regex(.name, "(?P<x>zz)")').remap(
+            {"name": "alice"})
+        raise AssertionError("expected VrlError")
+    except VrlError:
+        pass
