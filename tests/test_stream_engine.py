@@ -632,3 +632,41 @@ def test_lazy_step_batch_semantics():
     assert calls == [1]
     assert set(b.column_names) == {"a", "b"}
     assert calls == [1]  # cached — built once
+
+
+def test_engine_dynamic_topic_routing_e2e():
+    """YAML stream with expr topic/key routes rows into per-key-computed
+    fake-bus topics through the full engine."""
+    import asyncio
+
+    from arkflow_amd.config import EngineConfig
+    from arkflow_amd.inputs.brokers import FakeBus
+    from arkflow_amd.stream import build_stream
+
+    FakeBus.reset() if hasattr(FakeBus, "reset") else None
+    cfg = EngineConfig.from_dict({"streams": [{
+        "id": "router",
+        "input": {"type": "generate", "batch_size": 64, "interval": "0ms",
+                  "count": 192,
+                  "fields": {"key": {"dtype": "int64", "low": 0, "high": 8},
+                             "v": {"dtype": "float32"}}},
+        "pipeline": {"thread_num": 1, "processors": []},
+        "output": {"type": "kafka", "brokers": ["memory://router_e2e"],
+                   "topic": {"expr": "'shard-' || (key % 2)"},
+                   "key": {"expr": "key"}},
+    }]})
+    stream = build_stream(cfg.streams[0])
+
+    async def run():
+        cancel = asyncio.Event()
+        await asyncio.wait_for(stream.run(cancel), 30)
+
+    asyncio.new_event_loop().run_until_complete(run())
+    bus = FakeBus.get("router_e2e")
+    msgs = {t: [m for part in parts for m in part]
+            for t, parts in bus.topics.items() if t.startswith("shard-")}
+    assert set(msgs) == {"shard-0", "shard-1"}
+    assert sum(len(v) for v in msgs.values()) == 192
+    for t, rows in msgs.items():
+        want = t.removeprefix("shard-")
+        assert all(int(m[0]) % 2 == int(want) for m in rows)
