@@ -276,14 +276,17 @@ class FusedGenerateAgg:
         self.outs[key_col] = torch.zeros(self.n, device=self.device,
                                          dtype=torch.int64)
         import os
+        # row count stays in DEVICE memory: the hash-agg chain re-reads it
+        # per thread (host-mapped would mean per-thread PCIe reads); only
+        # the GROUP count is host-mapped — that's the one the host reads.
+        self.count = torch.zeros(1, device=self.device, dtype=torch.int32)
+        self.count_host = None
         try:
             if os.environ.get("ARKFLOW_MAPPED_COUNT", "1") == "0":
                 raise RuntimeError("disabled")
-            self.count_host, self.count = self.nat.mapped_int32(1)
             self.gcount_host, self.gcount = self.nat.mapped_int32(1)
         except RuntimeError:
-            self.count_host = self.gcount_host = None
-            self.count = torch.zeros(1, device=self.device, dtype=torch.int32)
+            self.gcount_host = None
             self.gcount = torch.zeros(1, device=self.device,
                                       dtype=torch.int32)
         self._lo = [float(fields[f].get("low", 0.0))
