@@ -197,3 +197,63 @@ def test_sort_indices_matches_sorted(keys, desc):
         if k in seen:
             assert i > seen[k]
         seen[k] = i
+
+
+def test_sliding_window_random_configs_match_reference_model():
+    """Property: for random (window_size, slide_size) the sliding buffer's
+    emissions equal a pure-Python reference model (emit the last
+    window_size batches after every slide_size-th write), and every batch
+    leaving the window is acked exactly once."""
+    import asyncio
+    import random
+
+    import torch
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.buffers.windows import SlidingWindowBuffer
+
+    loop = asyncio.new_event_loop()
+    rng = random.Random(1234)
+
+    class _Ack:
+        def __init__(self):
+            self.n = 0
+
+        async def ack(self):
+            self.n += 1
+
+    for case in range(150):
+        w = rng.randint(1, 6)
+        sl = rng.randint(1, 6)
+        n = rng.randint(1, 25)
+        buf = SlidingWindowBuffer({"window_size": w, "slide_size": sl})
+        model, since, got, want, acks = [], 0, [], [], []
+        for i in range(n):
+            b = MessageBatch({"id": Column(
+                "numeric", torch.tensor([i], dtype=torch.int64))})
+            a = _Ack()
+            acks.append(a)
+            loop.run_until_complete(buf.write(b, a))
+            model.append(i)
+            since += 1
+            out = buf.try_emit()
+            if since >= sl:
+                since = 0
+                del model[: max(0, len(model) - w)]
+                want.append(list(model))
+                assert out is not None, (case, w, sl, i)
+                batch, ack = out
+                got.append(batch.column("id").data.tolist())
+                loop.run_until_complete(ack.ack())
+            else:
+                assert out is None, (case, w, sl, i)
+        assert got == want, (case, w, sl)
+        # drain yields the remaining tail once
+        tail = buf.drain_remaining()
+        if model and tail is not None:
+            assert tail[0].column("id").data.tolist() == model
+        # ack accounting: every batch acked at most once, and all acked
+        # after drain-ack fires
+        if tail is not None:
+            loop.run_until_complete(tail[1].ack())
+        assert all(a.n <= 1 for a in acks), (case, w, sl)
