@@ -101,7 +101,9 @@ PyObject* encode_frames(PyObject*, PyObject* args) {
       put_u32be(w + 8, (uint32_t)(blen + 1));
       w[12] = 'R';
       std::memcpy(w + 13, bufs[i].buf, blen);
-      put_u32be(w + 13 + blen, crc32_sb8(w + 12, blen + 1, 0));
+      // CRC covers the whole frame incl. the seq/len header (a flipped
+      // header byte must fail validation, not replay as a wrong seq)
+      put_u32be(w + 13 + blen, crc32_sb8(w, blen + 13, 0));
       w += 13 + blen + 4;
     }
   }
@@ -142,7 +144,7 @@ PyObject* encode_frame_parts(PyObject*, PyObject* args) {
       std::memcpy(p, bufs[i].buf, (size_t)bufs[i].len);
       p += bufs[i].len;
     }
-    put_u32be(p, crc32_sb8(w + 12, body + 1, 0));
+    put_u32be(p, crc32_sb8(w, body + 13, 0));
   }
   for (Py_ssize_t i = 0; i < np; i++) PyBuffer_Release(&bufs[i]);
   Py_DECREF(fast);
@@ -162,7 +164,7 @@ PyObject* decode_frames(PyObject*, PyObject* args) {
     uint32_t ln = get_u32be(p + pos + 8);
     if (pos + 12 + (size_t)ln + 4 > n || ln == 0) break;
     const uint8_t* payload = p + pos + 12;
-    if (crc32_sb8(payload, ln, 0) != get_u32be(payload + ln)) break;
+    if (crc32_sb8(p + pos, ln + 12, 0) != get_u32be(payload + ln)) break;
     PyObject* tup = Py_BuildValue("(KBy#)", (unsigned long long)seq,
                                   (unsigned char)payload[0],
                                   (const char*)payload + 1,

@@ -79,8 +79,11 @@ class SegmentWalStore:
             self._manifest = json.loads(data)
             self._manifest_etag = etag
             self._seg_counter = self._manifest.get("next_seg", 0)
-        except json.JSONDecodeError:
-            pass
+        except (json.JSONDecodeError, UnicodeDecodeError, ValueError):
+            # corrupt manifest: recovery falls back to the LIST side of
+            # manifest ∪ LIST (reference s3.rs recovery); keep the etag so
+            # the next CAS write repairs it in place
+            self._manifest_etag = etag
 
     def _store_manifest(self) -> None:
         """Compare-and-swap write with precondition retries: on a lost race

@@ -148,9 +148,12 @@ def encode_frame(seq: int, payload: bytes, compress: bool = False) -> bytes:
     else:
         body = payload
         tag = b"R"
-    crc = zlib.crc32(body, zlib.crc32(tag))
-    return b"".join([struct.pack(">QI", seq, len(body) + 1), tag, body,
-                     struct.pack(">I", crc)])
+    head = struct.pack(">QI", seq, len(body) + 1)
+    # CRC covers the WHOLE frame (header included): a flipped seq/len byte
+    # must fail validation, not replay as a wrong sequence number
+    # (found by tests/test_wal.py random-corruption fuzz)
+    crc = zlib.crc32(body, zlib.crc32(tag, zlib.crc32(head)))
+    return b"".join([head, tag, body, struct.pack(">I", crc)])
 
 
 def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:
@@ -169,7 +172,7 @@ def decode_frames(buf: bytes) -> Iterator[Tuple[int, bytes]]:
             return  # zero-filled (preallocated mmap) or torn tail
         payload = buf[pos + 12: pos + 12 + ln]
         (crc,) = struct.unpack_from(">I", buf, pos + 12 + ln)
-        if zlib.crc32(payload) != crc:
+        if zlib.crc32(buf[pos: pos + 12 + ln]) != crc:
             return
         if payload[:1] == b"Z":
             yield seq, zlib.decompress(payload[1:])
@@ -453,7 +456,7 @@ def _end_of_frames(buf: bytes) -> int:
             break
         payload = buf[pos + 12: pos + 12 + ln]
         (crc,) = struct.unpack_from(">I", buf, pos + 12 + ln)
-        if zlib.crc32(payload) != crc:
+        if zlib.crc32(buf[pos: pos + 12 + ln]) != crc:
             break
         pos += 12 + ln + 4
     return pos
