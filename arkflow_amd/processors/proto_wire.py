@@ -125,6 +125,8 @@ def decode_message(buf: bytes, schema: ProtoSchema) -> Dict[str, object]:
                     out[name] = v
         elif wt == 1:
             raw = buf[pos:pos + 8]
+            if len(raw) < 8:
+                raise ValueError("truncated fixed64 field")
             pos += 8
             if spec:
                 name, t = spec
@@ -137,6 +139,8 @@ def decode_message(buf: bytes, schema: ProtoSchema) -> Dict[str, object]:
         elif wt == 2:
             ln, pos = _read_varint(buf, pos)
             raw = buf[pos:pos + ln]
+            if len(raw) < ln:
+                raise ValueError("truncated length-delimited field")
             pos += ln
             if spec:
                 name, t = spec
@@ -144,6 +148,8 @@ def decode_message(buf: bytes, schema: ProtoSchema) -> Dict[str, object]:
                     if t == "string" else raw
         elif wt == 5:
             raw = buf[pos:pos + 4]
+            if len(raw) < 4:
+                raise ValueError("truncated fixed32 field")
             pos += 4
             if spec:
                 name, t = spec

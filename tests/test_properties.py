@@ -257,3 +257,23 @@ def test_sliding_window_random_configs_match_reference_model():
         if tail is not None:
             loop.run_until_complete(tail[1].ack())
         assert all(a.n <= 1 for a in acks), (case, w, sl)
+
+
+def test_proto_wire_decode_fuzz_no_crash():
+    """Random bytes through the protobuf wire decoder either raise a clean
+    error or return a partial dict — never crash or hang (the GPU kernel
+    mirrors this with per-row err flags)."""
+    import random
+
+    from arkflow_amd.processors.proto_wire import ProtoSchema, decode_message
+
+    schema = ProtoSchema.parse(
+        "message T { double a = 1; int64 b = 2; string s = 3; bool ok = 4; }")
+    rng = random.Random(7)
+    for _ in range(300):
+        blob = bytes(rng.randrange(256) for _ in range(rng.randrange(0, 40)))
+        try:
+            out = decode_message(blob, schema)
+            assert isinstance(out, dict)
+        except (ValueError, EOFError, IndexError) as e:
+            assert str(e) is not None
