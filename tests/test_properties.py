@@ -277,3 +277,28 @@ def test_proto_wire_decode_fuzz_no_crash():
             assert isinstance(out, dict)
         except (ValueError, EOFError, IndexError) as e:
             assert str(e) is not None
+
+
+def test_codec_decode_fuzz_no_crash():
+    """Random bytes through the debezium / json codecs raise clean errors
+    (or produce rows), never uncontrolled exceptions."""
+    import random
+
+    from arkflow_amd.codecs.debezium import DebeziumJsonCodec
+    from arkflow_amd.codecs.json_codec import JsonCodec
+    from arkflow_amd.errors import ProcessError
+
+    rng = random.Random(21)
+    deb = DebeziumJsonCodec({}, None)
+    js = JsonCodec({}, None)
+    ok_types = (ValueError, KeyError, TypeError, ProcessError)
+    for _ in range(150):
+        blob = bytes(rng.randrange(256) for _ in range(rng.randrange(0, 60)))
+        for codec in (deb, js):
+            try:
+                codec.decode([blob])
+            except ok_types:
+                pass
+            except Exception as e:  # noqa: BLE001
+                raise AssertionError(
+                    f"{type(codec).__name__} leaked {type(e).__name__}: {e}")
