@@ -134,12 +134,16 @@ class EngineConfig:
             seen.add(sc.id)
             streams.append(sc)
         log_raw = raw.get("logging") or {}
+        if not isinstance(log_raw, dict):
+            raise ConfigError("'logging' must be a mapping")
         logging_cfg = LoggingConfig(
             level=log_raw.get("level", "info"),
             format=log_raw.get("format", "plain"),
             file=log_raw.get("file"),
         )
         srv_raw = raw.get("health_check") or raw.get("server") or {}
+        if not isinstance(srv_raw, dict):
+            raise ConfigError("'server' must be a mapping")
         server_cfg = ServerConfig(
             enabled=bool(srv_raw.get("enabled", bool(srv_raw))),
             address=srv_raw.get("address", "127.0.0.1:8111"),
@@ -192,6 +196,9 @@ class EngineConfig:
                 if spec is None:
                     continue
                 t = spec.get("type") if isinstance(spec, dict) else None
+                if t is not None and not isinstance(t, str):
+                    errors.append(f"{where}: {kind} 'type' must be a string")
+                    continue
                 if not t:
                     errors.append(f"{where}: {kind} missing 'type'")
                 elif t not in _registry(kind).builders:
@@ -203,12 +210,21 @@ class EngineConfig:
                 if codec_spec is not None:
                     ct = codec_spec.get("type") \
                         if isinstance(codec_spec, dict) else None
+                    if ct is not None and not isinstance(ct, str):
+                        errors.append(
+                            f"{where}: codec 'type' must be a string")
+                        ct = None
+                        continue
                     if not ct:
                         errors.append(f"{where}: codec missing 'type'")
                     elif ct not in _registry("codec").builders:
                         errors.append(f"{where}: unknown codec type {ct!r}")
             for p in s.pipeline.processors:
                 t = p.get("type") if isinstance(p, dict) else None
+                if t is not None and not isinstance(t, str):
+                    errors.append(f"{where}: processor 'type' must be a "
+                                  "string")
+                    continue
                 if not t:
                     errors.append(f"{where}: processor missing 'type'")
                 elif t not in _registry("processor").builders:
@@ -233,24 +249,35 @@ def _parse_stream(raw: dict, index: int) -> StreamConfig:
     if "output" not in raw:
         raise ConfigError(f"stream {sid!r}: missing output")
     p_raw = raw.get("pipeline") or {}
-    pipeline = PipelineConfig(
-        thread_num=int(p_raw.get("thread_num", 0)),
-        processors=list(p_raw.get("processors") or []),
-    )
+    if not isinstance(p_raw, dict):
+        raise ConfigError(f"stream {sid!r}: 'pipeline' must be a mapping")
+    try:
+        pipeline = PipelineConfig(
+            thread_num=int(p_raw.get("thread_num") or 0),
+            processors=list(p_raw.get("processors") or []),
+        )
+    except (TypeError, ValueError) as e:
+        raise ConfigError(f"stream {sid!r}: bad pipeline config: {e}")
     dur = None
     d_raw = raw.get("durability")
+    if d_raw is not None and not isinstance(d_raw, dict):
+        raise ConfigError(f"stream {sid!r}: 'durability' must be a mapping")
     if d_raw:
         known = {"enabled", "path", "sync_policy", "group_window_ms",
                  "periodic_interval_ms", "backend"}
-        dur = DurabilityConfig(
-            enabled=bool(d_raw.get("enabled", True)),
-            path=str(d_raw.get("path", "./wal")),
-            sync_policy=str(d_raw.get("sync_policy", "group_commit")),
-            group_window_ms=int(d_raw.get("group_window_ms", 5)),
-            periodic_interval_ms=int(d_raw.get("periodic_interval_ms", 200)),
-            backend=str(d_raw.get("backend", "local")),
-            extra={k: v for k, v in d_raw.items() if k not in known},
-        )
+        try:
+            dur = DurabilityConfig(
+                enabled=bool(d_raw.get("enabled", True)),
+                path=str(d_raw.get("path", "./wal")),
+                sync_policy=str(d_raw.get("sync_policy", "group_commit")),
+                group_window_ms=int(d_raw.get("group_window_ms") or 5),
+                periodic_interval_ms=int(
+                    d_raw.get("periodic_interval_ms") or 200),
+                backend=str(d_raw.get("backend", "local")),
+                extra={k: v for k, v in d_raw.items() if k not in known},
+            )
+        except (TypeError, ValueError) as e:
+            raise ConfigError(f"stream {sid!r}: bad durability config: {e}")
         if dur.sync_policy not in ("per_entry", "group_commit", "periodic"):
             raise ConfigError(
                 f"stream {sid!r}: invalid sync_policy {dur.sync_policy!r}"

@@ -302,3 +302,52 @@ def test_codec_decode_fuzz_no_crash():
             except Exception as e:  # noqa: BLE001
                 raise AssertionError(
                     f"{type(codec).__name__} leaked {type(e).__name__}: {e}")
+
+
+def test_config_loader_fuzz_clean_errors():
+    """Mangled config dicts raise ConfigError (or validate with error
+    strings) — never uncontrolled exceptions (reference
+    configuration.rs:176 validate_config)."""
+    import random
+
+    from arkflow_amd.config import EngineConfig
+    from arkflow_amd.errors import ConfigError
+
+    rng = random.Random(5)
+    atoms = [None, 1, -3, 0.5, True, "", "x", [], {}, "memory", "generate",
+             {"type": None}, {"type": 7}, {"type": "nope"}, [1, 2]]
+
+    def mutate(d, depth=0):
+        if depth > 2 or not isinstance(d, dict):
+            return rng.choice(atoms)
+        out = {}
+        for k, v in d.items():
+            r = rng.random()
+            if r < 0.2:
+                continue  # drop key
+            if r < 0.4:
+                out[k] = rng.choice(atoms)
+            elif isinstance(v, dict):
+                out[k] = mutate(v, depth + 1)
+            else:
+                out[k] = v
+        return out
+
+    base = {"streams": [{
+        "id": "s", "input": {"type": "generate", "batch_size": 8,
+                             "interval": "0ms", "context": "{}"},
+        "pipeline": {"thread_num": 1, "processors": [
+            {"type": "sql", "query": "SELECT * FROM flow"}]},
+        "output": {"type": "drop"},
+    }]}
+    for _ in range(200):
+        cfg = mutate({"streams": [mutate(base["streams"][0])]})
+        try:
+            ec = EngineConfig.from_dict(cfg)
+            errs = ec.validate()
+            assert isinstance(errs, list)
+        except ConfigError:
+            pass
+        except Exception as e:  # noqa: BLE001
+            raise AssertionError(
+                f"config loader leaked {type(e).__name__}: {e}\n{cfg}")
