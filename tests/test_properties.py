@@ -351,3 +351,29 @@ def test_config_loader_fuzz_clean_errors():
         except Exception as e:  # noqa: BLE001
             raise AssertionError(
                 f"config loader leaked {type(e).__name__}: {e}\n{cfg}")
+
+
+def test_vrl_parser_fuzz_no_crash():
+    """Random token soup through the VRL parser/interpreter raises clean
+    Vrl/Config errors — never uncontrolled exceptions or hangs."""
+    import random
+
+    from arkflow_amd.errors import ConfigError
+    from arkflow_amd.processors.vrl_lang import (VrlAbort, VrlError,
+                                                 VrlProgram)
+
+    rng = random.Random(11)
+    toks = [".a", ".b.c", "=", "==", "if", "else", "{", "}", "(", ")",
+            "??", "!", ",", ";", "del", "upcase", "to_int", "1", "0.5",
+            '"x"', "true", "null", "x", "+", "-", "*", "/", "[0]", "&&",
+            "||", "abort"]
+    for _ in range(300):
+        prog = " ".join(rng.choice(toks)
+                        for _ in range(rng.randrange(1, 18)))
+        try:
+            VrlProgram(prog).remap({"a": 1, "b": {"c": "s"}})
+        except (VrlError, VrlAbort, ConfigError):
+            pass
+        except Exception as e:  # noqa: BLE001
+            raise AssertionError(
+                f"vrl leaked {type(e).__name__}: {e}\nprogram: {prog!r}")
