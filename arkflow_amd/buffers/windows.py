@@ -129,14 +129,15 @@ class SessionWindowBuffer(BaseWindowBuffer):
         super().__init__(config, resource)
         self.gap = _parse_duration(config.get("gap", "1s"))
         self._last_message: Optional[float] = None
+        self._now = time.monotonic  # injectable clock (deterministic tests)
 
     def on_write(self, batch: MessageBatch) -> None:
-        self._last_message = time.monotonic()
+        self._last_message = self._now()
 
     def try_emit(self, draining: bool = False):
         if self._last_message is None:
             return None
-        if time.monotonic() - self._last_message < self.gap:
+        if self._now() - self._last_message < self.gap:
             return None
         self._last_message = None
         return self._emit_all()
@@ -144,7 +145,7 @@ class SessionWindowBuffer(BaseWindowBuffer):
     def next_deadline(self) -> Optional[float]:
         if self._last_message is None:
             return None
-        return self._last_message + self.gap - time.monotonic()
+        return self._last_message + self.gap - self._now()
 
 
 @register("buffer", "tumbling_window",

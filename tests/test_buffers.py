@@ -326,3 +326,44 @@ def test_windowed_join_example_runs_e2e(run):
         assert eng.runtime.get("joiner").metrics.processing_errors == 0
 
     run(main(), timeout=60)
+
+
+def test_session_window_deterministic_clock(run):
+    """Session gap semantics under an injected clock: bursts separated by
+    more than `gap` close a session; activity inside the gap keeps it
+    open (session_window.rs:107-143) — no sleeps, fully deterministic."""
+    import torch
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.buffers.windows import SessionWindowBuffer
+    from arkflow_amd.spi import NoopAck
+
+    clock = [0.0]
+    buf = SessionWindowBuffer({"gap": "5s"})
+    buf._now = lambda: clock[0]
+
+    def b(i):
+        return MessageBatch({"id": Column(
+            "numeric", torch.tensor([i], dtype=torch.int64))})
+
+    # burst 1: t=0,1,2 — activity within the gap keeps the session open
+    for t, i in [(0.0, 0), (1.0, 1), (2.0, 2)]:
+        clock[0] = t
+        run(buf.write(b(i), NoopAck()))
+        assert buf.try_emit() is None
+    clock[0] = 6.9  # 4.9s after last message: still open
+    assert buf.try_emit() is None
+    clock[0] = 7.1  # gap exceeded: session closes with the whole burst
+    out = buf.try_emit()
+    assert out is not None
+    batch, _ack = out
+    assert batch.column("id").data.tolist() == [0, 1, 2]
+    # idle: nothing more to emit
+    assert buf.try_emit() is None
+    # burst 2 starts a fresh session
+    clock[0] = 100.0
+    run(buf.write(b(7), NoopAck()))
+    assert buf.try_emit() is None
+    clock[0] = 106.0
+    out2 = buf.try_emit()
+    assert out2 is not None and out2[0].column("id").data.tolist() == [7]
