@@ -29,15 +29,16 @@ class TumblingWindowBuffer(BaseWindowBuffer):
         super().__init__(config, resource)
         self.interval = _parse_duration(config.get("interval", "1s"))
         self._deadline = None
+        self._now = time.monotonic  # injectable clock (deterministic tests)
 
     def on_write(self, batch: MessageBatch) -> None:
         if self._deadline is None:
-            self._deadline = time.monotonic() + self.interval
+            self._deadline = self._now() + self.interval
 
     def try_emit(self, draining: bool = False):
-        if self._deadline is None or time.monotonic() < self._deadline:
+        if self._deadline is None or self._now() < self._deadline:
             return None
-        self._deadline = (time.monotonic() + self.interval
+        self._deadline = (self._now() + self.interval
                           if self._total_buffered() else None)
         out = self._emit_all()
         if out is None:
@@ -47,7 +48,7 @@ class TumblingWindowBuffer(BaseWindowBuffer):
     def next_deadline(self) -> Optional[float]:
         if self._deadline is None:
             return None
-        return self._deadline - time.monotonic()
+        return self._deadline - self._now()
 
 
 class SlidingWindowBuffer(BaseWindowBuffer):

@@ -367,3 +367,38 @@ def test_session_window_deterministic_clock(run):
     clock[0] = 106.0
     out2 = buf.try_emit()
     assert out2 is not None and out2[0].column("id").data.tolist() == [7]
+
+
+def test_tumbling_window_deterministic_clock(run):
+    """Tumbling intervals under an injected clock: each window emits once
+    per interval with exactly the batches that arrived inside it."""
+    import torch
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.buffers.windows import TumblingWindowBuffer
+    from arkflow_amd.spi import NoopAck
+
+    clock = [0.0]
+    buf = TumblingWindowBuffer({"interval": "10s"})
+    buf._now = lambda: clock[0]
+
+    def b(i):
+        return MessageBatch({"id": Column(
+            "numeric", torch.tensor([i], dtype=torch.int64))})
+
+    for t, i in [(0.0, 0), (3.0, 1), (9.0, 2)]:
+        clock[0] = t
+        run(buf.write(b(i), NoopAck()))
+        assert buf.try_emit() is None
+    clock[0] = 10.5
+    out = buf.try_emit()
+    assert out is not None
+    assert out[0].column("id").data.tolist() == [0, 1, 2]
+    # next window: one batch at t=12, fires after t=20.5
+    clock[0] = 12.0
+    run(buf.write(b(9), NoopAck()))
+    clock[0] = 20.0
+    assert buf.try_emit() is None
+    clock[0] = 22.6
+    out2 = buf.try_emit()
+    assert out2 is not None and out2[0].column("id").data.tolist() == [9]
