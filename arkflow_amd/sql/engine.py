@@ -592,35 +592,65 @@ class SqlExecutor:
             remapped[matched] = right_map[r_idx[matched].long()]
             r_idx = remapped
 
-        new_cols: Dict[str, Column] = {}
-        new_order: List[str] = []
-        for name in order:
-            c = columns[name].take(l_idx)
-            new_cols[name] = c
-            new_cols[f"{left_alias}.{name.split('.', 1)[-1]}"] = c
-            new_order.append(name)
-        safe_r = r_idx.clamp(min=0)
-        null_rows = r_idx < 0
-        for name, col in right.columns.items():
-            taken = col.take(safe_r)
-            if bool(null_rows.any()):
-                validity = ~null_rows
-                taken = Column(taken.kind, taken.data, taken.offsets, validity)
-            qual = f"{r_alias}.{name}"
-            new_cols[qual] = taken
-            if name not in new_cols:  # unqualified only when unambiguous
-                new_cols[name] = taken
-                new_order.append(name)
-            else:
-                new_order.append(qual)
+        def build(li, ri):
+            cols: Dict[str, Column] = {}
+            ordr: List[str] = []
+            for name in order:
+                c = columns[name].take(li)
+                cols[name] = c
+                cols[f"{left_alias}.{name.split('.', 1)[-1]}"] = c
+                ordr.append(name)
+            safe_r = ri.clamp(min=0)
+            nulls = ri < 0
+            for name, col in right.columns.items():
+                taken = col.take(safe_r)
+                if bool(nulls.any()):
+                    taken = Column(taken.kind, taken.data, taken.offsets,
+                                   ~nulls)
+                qual = f"{r_alias}.{name}"
+                cols[qual] = taken
+                if name not in cols:  # unqualified only when unambiguous
+                    cols[name] = taken
+                    ordr.append(name)
+                else:
+                    ordr.append(qual)
+            return cols, ordr
+
+        new_cols, new_order = build(l_idx, r_idx)
         if residual:
             n = int(l_idx.shape[0])
             env = Env(new_cols, n, device)
             mask = torch.ones(n, dtype=torch.bool, device=device)
             for c in residual:
                 mask &= as_tensor(eval_expr(c, env), env).bool()
-            idx = ops.mask_to_indices(mask).long()
-            new_cols = {k: c.take(idx) for k, c in new_cols.items()}
+            if j.kind == "left":
+                # LEFT JOIN semantics: a residual ON conjunct qualifies the
+                # MATCH, it does not filter rows — matches that fail it
+                # null-extend. Null-extended rows pass through untouched; a
+                # left row whose every match fails re-enters once with a
+                # NULL right side. (Found by the sqlite differential suite:
+                # anti-join idiom ... ON f.k = d.k AND d.k < 3 WHERE d.k
+                # IS NULL returned nothing.)
+                null_rows = r_idx < 0
+                matched_keep = mask & ~null_rows
+                li_long = l_idx.long()
+                had = torch.zeros(n_left, dtype=torch.bool, device=device)
+                had[li_long[~null_rows]] = True
+                kept_any = torch.zeros(n_left, dtype=torch.bool,
+                                       device=device)
+                kept_any[li_long[matched_keep]] = True
+                demoted = (had & ~kept_any).nonzero(as_tuple=True)[0]
+                keep = matched_keep | null_rows
+                l2 = torch.cat([l_idx[keep],
+                                demoted.to(l_idx.dtype)])
+                r2 = torch.cat([r_idx[keep],
+                                torch.full((demoted.numel(),), -1,
+                                           dtype=r_idx.dtype,
+                                           device=r_idx.device)])
+                new_cols, new_order = build(l2, r2)
+            else:
+                idx = ops.mask_to_indices(mask).long()
+                new_cols = {k: c.take(idx) for k, c in new_cols.items()}
         return new_cols, new_order
 
     # ------------------------------------------------------------- aggregate

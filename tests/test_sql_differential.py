@@ -75,6 +75,16 @@ QUERIES = [
     "WHERE a > 90",
     "SELECT a FROM flow ORDER BY a LIMIT 1000 OFFSET 190",
     "SELECT a FROM flow ORDER BY a LIMIT 5 OFFSET 10000",
+    "SELECT d.label FROM dims d WHERE d.label LIKE 'L%'",
+    "SELECT a FROM flow WHERE a NOT IN (1, 2, 3) AND a < 12 ORDER BY a",
+    "SELECT a, lead(a, 1, -99) OVER (PARTITION BY k ORDER BY a) AS nx "
+    "FROM flow ORDER BY a, nx",
+    "SELECT coalesce(nullif(k, 0), -1) AS kz, count(*) AS c FROM flow "
+    "GROUP BY kz ORDER BY kz",
+    "SELECT k, sum(CASE WHEN b >= 0.5 THEN 1 ELSE 0 END) AS hi "
+    "FROM flow GROUP BY k ORDER BY k",
+    "SELECT f.a FROM flow f LEFT JOIN dims d ON f.k = d.k AND d.k < 3 "
+    "WHERE d.k IS NULL ORDER BY f.a LIMIT 15",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
