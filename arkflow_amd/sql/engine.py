@@ -209,8 +209,13 @@ class SqlExecutor:
                     # binary sort key: CPU fallback
                     import numpy as np
                     vals = key.to_pylist()
+                    # NULLs (None) get a stable placeholder here; the
+                    # validity stable-sort below puts them in their
+                    # NULLS FIRST/LAST position
                     idx = torch.tensor(
-                        sorted(range(len(vals)), key=lambda i: vals[i],
+                        sorted(range(len(vals)),
+                               key=lambda i: vals[i]
+                               if vals[i] is not None else b"",
                                reverse=not asc),
                         dtype=torch.int64, device=device)
                 else:

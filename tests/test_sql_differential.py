@@ -85,6 +85,14 @@ QUERIES = [
     "FROM flow GROUP BY k ORDER BY k",
     "SELECT f.a FROM flow f LEFT JOIN dims d ON f.k = d.k AND d.k < 3 "
     "WHERE d.k IS NULL ORDER BY f.a LIMIT 15",
+    "SELECT f.a, d.label FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k AND f.a > 50 ORDER BY f.a, d.label",
+    "SELECT f.a, d.k FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k AND d.label != 'L2' ORDER BY f.a, d.k",
+    "SELECT count(*) AS c FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k AND d.k % 2 = 0 WHERE d.k IS NOT NULL",
+    "SELECT f.k, count(d.k) AS m FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k AND d.k >= 2 GROUP BY f.k ORDER BY f.k",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
