@@ -759,6 +759,26 @@ class SqlExecutor:
                 ).to(torch.int64)
             return ops.segment_reduce(
                 torch.ones(env.n_rows, device=device), gid, g, "count")
+        if isinstance(arg, Column) and arg.kind == "binary" \
+                and name in ("min", "max"):
+            # lexicographic min/max over strings: host fallback (DataFusion
+            # supports string MIN/MAX; group counts are small post-hash)
+            pv = arg.to_pylist()
+            gl = gid.detach().to("cpu").tolist()
+            best = [None] * g
+            for v, gi in zip(pv, gl):
+                if v is None:
+                    continue
+                b = best[gi]
+                if b is None or (v < b if name == "min" else v > b):
+                    best[gi] = v
+            out = Column.from_bytes([b if b is not None else b""
+                                     for b in best])
+            if any(b is None for b in best):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([b is not None for b in best],
+                                          dtype=torch.bool, device=device))
+            return out
         vals = as_tensor(arg, env)
         from .eval import expr_validity
         validity = expr_validity(a.args[0], env) if a.args else None

@@ -474,7 +474,62 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
         b = as_tensor(eval_expr(e.args[0], env), env).double()
         p = as_tensor(eval_expr(e.args[1], env), env).double()
         return torch.pow(b, p)
+    if name in ("substr", "substring"):
+        v = eval_expr(e.args[0], env)
+        def _si(x):
+            return int(x.item()) if isinstance(x, torch.Tensor) else int(x)
+        start = _si(eval_expr(e.args[1], env))
+        ln = _si(eval_expr(e.args[2], env)) if len(e.args) > 2 else None
+        def _sub(b):
+            if b is None:
+                return None
+            # SQL/sqlite 1-based start; negative counts from the end
+            if start > 0:
+                i = start - 1
+            elif start < 0:
+                i = max(len(b) + start, 0)
+            else:
+                i = 0
+            return b[i:] if ln is None else b[i:i + max(ln, 0)]
+        if isinstance(v, Column) and v.kind == "binary":
+            vals = [_sub(x) for x in v.to_pylist()]
+            out = Column.from_bytes([x if x is not None else b""
+                                     for x in vals])
+            if any(x is None for x in vals):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([x is not None for x in vals],
+                                          dtype=torch.bool,
+                                          device=env.device))
+            return out
+        if isinstance(v, str):
+            return _sub(v.encode()).decode("utf-8", "replace")
+        raise SqlError("substr() requires a string")
     if name == "coalesce":
+        args_v = [eval_expr(a, env) for a in e.args]
+        if any(isinstance(v, Column) and v.kind == "binary"
+               for v in args_v) or \
+                all(isinstance(v, (str, type(None))) for v in args_v):
+            # string coalesce: first non-NULL per row (host fallback)
+            rows = None
+            for v in args_v:
+                if isinstance(v, Column) and v.kind == "binary":
+                    cur = v.to_pylist()
+                elif isinstance(v, str):
+                    cur = [v.encode()] * env.n_rows
+                elif v is None:
+                    cur = [None] * env.n_rows
+                else:
+                    raise SqlError("coalesce: mixed string/numeric args")
+                rows = cur if rows is None else [
+                    a if a is not None else b for a, b in zip(rows, cur)]
+            out = Column.from_bytes([x if x is not None else b""
+                                     for x in rows])
+            if any(x is None for x in rows):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([x is not None for x in rows],
+                                          dtype=torch.bool,
+                                          device=env.device))
+            return out
         result = None
         missing = None  # rows still NULL so far
         for a in e.args:

@@ -93,6 +93,19 @@ QUERIES = [
     "ON f.k = d.k AND d.k % 2 = 0 WHERE d.k IS NOT NULL",
     "SELECT f.k, count(d.k) AS m FROM flow f LEFT JOIN dims d "
     "ON f.k = d.k AND d.k >= 2 GROUP BY f.k ORDER BY f.k",
+    "SELECT d.label, sum(f.a) AS s FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k GROUP BY d.label ORDER BY d.label NULLS FIRST",
+    "SELECT coalesce(d.label, 'none') AS lb, count(*) AS c FROM flow f "
+    "LEFT JOIN dims d ON f.k = d.k AND d.k > 1 GROUP BY lb ORDER BY lb",
+    "SELECT f.a, length(d.label) AS ln FROM flow f LEFT JOIN dims d "
+    "ON f.k = d.k ORDER BY f.a, ln",
+    "SELECT min(d.label) AS lo, max(d.label) AS hi FROM flow f "
+    "LEFT JOIN dims d ON f.k = d.k",
+    "SELECT a, substr(d.label, 2) AS tail FROM flow f JOIN dims d "
+    "ON f.k = d.k ORDER BY a",
+    "SELECT k, group_concat(a, ',') AS g FROM (SELECT k, a FROM flow "
+    "ORDER BY a) GROUP BY k ORDER BY k" if False else
+    "SELECT lower(d.label) AS l1 FROM dims d ORDER BY l1 DESC",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
