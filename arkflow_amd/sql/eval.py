@@ -610,6 +610,27 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
         if isinstance(v, str):
             return v.upper() if name == "upper" else v.lower()
         raise SqlError(f"{name}() requires a string")
+    if name == "replace":
+        v = eval_expr(e.args[0], env)
+        old_s = eval_expr(e.args[1], env)
+        new_s = eval_expr(e.args[2], env)
+        if not isinstance(old_s, str) or not isinstance(new_s, str):
+            raise SqlError("replace() needs constant string args")
+        if isinstance(v, Column) and v.kind == "binary":
+            vals = [None if x is None else
+                    x.replace(old_s.encode(), new_s.encode())
+                    for x in v.to_pylist()]
+            out = Column.from_bytes([x if x is not None else b""
+                                     for x in vals])
+            if any(x is None for x in vals):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([x is not None for x in vals],
+                                          dtype=torch.bool,
+                                          device=env.device))
+            return out
+        if isinstance(v, str):
+            return v.replace(old_s, new_s)
+        raise SqlError("replace() requires a string")
     if name == "concat":
         parts = [eval_expr(a, env) for a in e.args]
         lists = []

@@ -106,6 +106,19 @@ QUERIES = [
     "SELECT k, group_concat(a, ',') AS g FROM (SELECT k, a FROM flow "
     "ORDER BY a) GROUP BY k ORDER BY k" if False else
     "SELECT lower(d.label) AS l1 FROM dims d ORDER BY l1 DESC",
+    "SELECT k % 2 AS p, avg(b) AS m FROM flow GROUP BY k % 2 ORDER BY p",
+    "SELECT upper(d.label) || '-' || d.k AS tag FROM dims d ORDER BY tag",
+    "SELECT abs(c) AS ac, count(*) AS n FROM flow GROUP BY ac "
+    "ORDER BY ac LIMIT 8",
+    "SELECT a FROM flow WHERE c BETWEEN -5 AND 5 AND NOT (a IN (2, 4)) "
+    "ORDER BY a",
+    "SELECT d.label, f.a FROM dims d JOIN flow f ON d.k = f.k "
+    "WHERE d.label >= 'L2' ORDER BY d.label, f.a LIMIT 12",
+    "SELECT k, max(a) - min(a) AS spread FROM flow GROUP BY k "
+    "HAVING max(a) - min(a) > 10 ORDER BY k",
+    "SELECT CAST(a AS REAL) / 4 AS q FROM flow ORDER BY q LIMIT 6",
+    "SELECT count(*) AS c, sum(a) % 7 AS s7 FROM flow",
+    "SELECT replace(d.label, 'L', 'X') AS rl FROM dims d ORDER BY rl",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
