@@ -1270,3 +1270,23 @@ def test_genfiltpack_matches_multi_kernel_chain(dev, monkeypatch):
         assert x.keys() == y.keys()
         for k in x:
             assert torch.equal(x[k], y[k]), f"step {step} col {k}"
+
+
+def test_left_join_residual_anti_join_on_gpu(dev):
+    """LEFT JOIN residual ON conjuncts null-extend failed matches on
+    device columns — the anti-join idiom (WHERE right IS NULL) works."""
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.sql.engine import SqlExecutor
+
+    flow = MessageBatch.from_dict({
+        "a": torch.arange(20, device=dev),
+        "k": torch.arange(20, device=dev) % 5})
+    dims = MessageBatch.from_dict({
+        "k": torch.arange(3, device=dev),
+        "w": torch.arange(3, device=dev) * 10})
+    r = SqlExecutor(
+        "SELECT f.a FROM flow f LEFT JOIN dims d "
+        "ON f.k = d.k AND d.k < 2 WHERE d.k IS NULL ORDER BY f.a"
+    ).execute({"flow": flow, "dims": dims})
+    got = r.column("a").data.cpu().tolist()
+    assert got == [x for x in range(20) if x % 5 >= 2]
