@@ -242,6 +242,36 @@ def eval_expr(e, env: Env) -> Value:
         v = eval_expr(e.expr, env)
         return _eval_cast(v, e.to_type, env)
     if isinstance(e, Case):
+        branch_vals = [eval_expr(v, env) for _, v in e.whens]
+        else_v = eval_expr(e.else_, env) if e.else_ is not None else None
+        if any(isinstance(v, str) or
+               (isinstance(v, Column) and v.kind == "binary")
+               for v in list(branch_vals) + [else_v]):
+            # string-valued CASE: host row loop (sqlite/DataFusion parity)
+            def _rows(v):
+                if v is None:
+                    return [None] * env.n_rows
+                if isinstance(v, str):
+                    return [v.encode()] * env.n_rows
+                if isinstance(v, Column) and v.kind == "binary":
+                    return v.to_pylist()
+                raise SqlError("CASE mixes string and numeric branches")
+            out_rows = _rows(else_v)
+            for (cond, _), v in zip(reversed(list(e.whens)),
+                                    reversed(branch_vals)):
+                c = as_tensor(eval_expr(cond, env),
+                              env).bool().cpu().tolist()
+                vr = _rows(v)
+                out_rows = [b if m else a
+                            for a, b, m in zip(out_rows, vr, c)]
+            out = Column.from_bytes([x if x is not None else b""
+                                     for x in out_rows])
+            if any(x is None for x in out_rows):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([x is not None for x in out_rows],
+                                          dtype=torch.bool,
+                                          device=env.device))
+            return out
         result = None
         for cond, val in reversed(list(e.whens)):
             c = as_tensor(eval_expr(cond, env), env).bool()

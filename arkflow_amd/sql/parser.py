@@ -476,10 +476,17 @@ class Parser:
         if t.kind == "kw" and t.value == "null":
             return Literal(None)
         if t.kind == "kw" and t.value == "case":
+            # simple form `CASE expr WHEN v THEN r ...` desugars to the
+            # searched form with `expr = v` conditions (sqlite/standard)
+            operand = None
+            if self.peek().kind != "kw" or self.peek().value != "when":
+                operand = self._expr()
             whens = []
             else_ = None
             while self.accept("kw", "when"):
                 cond = self._expr()
+                if operand is not None:
+                    cond = BinaryOp("=", operand, cond)
                 self.expect("kw", "then")
                 whens.append((cond, self._expr()))
             if self.accept("kw", "else"):

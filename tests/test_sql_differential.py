@@ -119,6 +119,22 @@ QUERIES = [
     "SELECT CAST(a AS REAL) / 4 AS q FROM flow ORDER BY q LIMIT 6",
     "SELECT count(*) AS c, sum(a) % 7 AS s7 FROM flow",
     "SELECT replace(d.label, 'L', 'X') AS rl FROM dims d ORDER BY rl",
+    "SELECT f.a FROM flow f JOIN dims d ON f.k = d.k JOIN dims e "
+    "ON f.k = e.k ORDER BY f.a LIMIT 10",
+    "SELECT x.a AS a1, y.a AS a2 FROM flow x JOIN flow y ON x.k = y.k "
+    "AND x.a < y.a ORDER BY a1, a2 LIMIT 15",
+    "SELECT k, c, count(*) AS n FROM flow GROUP BY k, c "
+    "ORDER BY k, c LIMIT 12",
+    "SELECT k, count(*) AS n FROM flow GROUP BY k "
+    "HAVING count(*) > 2 OR min(a) < 5 ORDER BY k",
+    "SELECT k, sum(a) AS s FROM flow GROUP BY k "
+    "ORDER BY sum(a) DESC, k LIMIT 4",
+    "SELECT DISTINCT k FROM flow ORDER BY k DESC",
+    "SELECT count(DISTINCT k) AS u, count(*) AS n FROM flow",
+    "SELECT CASE k WHEN 0 THEN 'z' WHEN 1 THEN 'o' ELSE '?' END AS w "
+    "FROM flow ORDER BY a LIMIT 6",
+    "SELECT CASE WHEN a >= 50 THEN d.label ELSE 'lo' END AS w "
+    "FROM flow f JOIN dims d ON f.k = d.k ORDER BY a LIMIT 9",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
