@@ -519,17 +519,6 @@ class SqlExecutor:
                     device):
         r_alias = j.alias or j.table
         conjuncts = _split_and(j.on)
-        key_pairs = []
-        residual = []
-        for c in conjuncts:
-            if (isinstance(c, BinaryOp) and c.op == "=" and
-                    isinstance(c.left, ColumnRef) and
-                    isinstance(c.right, ColumnRef)):
-                key_pairs.append((c.left, c.right))
-            else:
-                residual.append(c)
-        if not key_pairs:
-            raise SqlError("JOIN requires at least one equality condition")
 
         def side_of(ref: ColumnRef):
             if ref.table == r_alias or (
@@ -538,14 +527,51 @@ class SqlExecutor:
                 return "right"
             return "left"
 
+        def _refs(e, acc):
+            if isinstance(e, ColumnRef):
+                acc.append(e)
+            for attr in ("left", "right", "expr", "low", "high"):
+                sub = getattr(e, attr, None)
+                if sub is not None and not isinstance(sub, str):
+                    _refs(sub, acc)
+            for sub in getattr(e, "args", []) or []:
+                _refs(sub, acc)
+            return acc
+
+        def expr_side(e):
+            sides = {side_of(r) for r in _refs(e, [])}
+            if not sides:
+                return "none"  # constant: cannot drive the hash join
+            if len(sides) > 1:
+                return "mixed"
+            return sides.pop()
+
+        # equality conjuncts whose sides each reference exactly one table
+        # become (oriented) hash keys — arbitrary EXPRESSIONS allowed
+        # (DataFusion supports e.g. ON f.k + 1 = d.k); everything else is
+        # a residual condition
+        key_pairs = []
+        residual = []
+        for c in conjuncts:
+            if isinstance(c, BinaryOp) and c.op == "=":
+                ls, rs = expr_side(c.left), expr_side(c.right)
+                if ls == "left" and rs == "right":
+                    key_pairs.append((c.left, c.right))
+                    continue
+                if ls == "right" and rs == "left":
+                    key_pairs.append((c.right, c.left))
+                    continue
+            residual.append(c)
+        if not key_pairs:
+            raise SqlError("JOIN requires at least one equality condition "
+                           "relating the two tables")
+
         l_keys, r_keys = [], []
         n_left = len(columns[order[0]])
         l_env = Env(columns, n_left, device)
         r_cols_q = {f"{r_alias}.{k}": v for k, v in right.columns.items()}
         r_env = Env({**right.columns, **r_cols_q}, right.num_rows, device)
         for a, b in key_pairs:
-            if side_of(a) == "right":
-                a, b = b, a
             l_keys.append(as_tensor(eval_expr(a, l_env), l_env))
             r_keys.append(as_tensor(eval_expr(b, r_env), r_env))
         lk, _, _ = (_encode_keys(l_keys, device) if len(l_keys) > 1
@@ -563,8 +589,6 @@ class SqlExecutor:
         from .eval import expr_validity
         l_val = r_val = None
         for a, b in key_pairs:
-            if side_of(a) == "right":
-                a, b = b, a
             va = expr_validity(a, l_env)
             vb = expr_validity(b, r_env)
             if va is not None:

@@ -135,6 +135,12 @@ QUERIES = [
     "FROM flow ORDER BY a LIMIT 6",
     "SELECT CASE WHEN a >= 50 THEN d.label ELSE 'lo' END AS w "
     "FROM flow f JOIN dims d ON f.k = d.k ORDER BY a LIMIT 9",
+    "SELECT f.a FROM flow f JOIN dims d ON f.k + 1 = d.k "
+    "ORDER BY f.a LIMIT 9",
+    "SELECT f.a FROM flow f LEFT JOIN dims d ON f.k % 3 = d.k "
+    "WHERE d.k IS NULL ORDER BY f.a LIMIT 8",
+    "SELECT f.a, d.label FROM flow f JOIN dims d ON d.k = f.k - 1 "
+    "ORDER BY f.a LIMIT 7",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
