@@ -324,6 +324,9 @@ def eval_expr(e, env: Env) -> Value:
             r = torch.isnan(v)
         else:
             r = torch.zeros(env.n_rows, dtype=torch.bool, device=env.device)
+        val = expr_validity(e.expr, env)  # strict + div-by-zero bits
+        if val is not None:
+            r = r | ~val
         return ~r if e.negated else r
     if isinstance(e, Like):
         v = eval_expr(e.expr, env)
@@ -427,12 +430,16 @@ def _eval_binop(e: BinaryOp, env: Env) -> Value:
     if e.op == "*":
         return l * r
     if e.op == "/":
+        # zero divisors are NULL (expr_validity masks them); substitute 1
+        # so the kernel never faults
+        rz = torch.where(r == 0, torch.ones_like(r), r)
         if l.dtype.is_floating_point or r.dtype.is_floating_point:
-            return l / r
-        return torch.div(l, r, rounding_mode="trunc")
+            return l / rz
+        return torch.div(l, rz, rounding_mode="trunc")
     if e.op == "%":
         # SQL %: truncated remainder (sign of the dividend — sqlite/DataFusion)
-        return torch.fmod(l, r)
+        rz = torch.where(r == 0, torch.ones_like(r), r)
+        return torch.fmod(l, rz)
     raise SqlError(f"unknown operator {e.op}")
 
 
@@ -712,9 +719,11 @@ def expr_validity(e, env: Env):
     """Combined validity of the columns an expression touches (strict-NULL
     propagation for elementwise expressions): None = all valid. Expressions
     containing null-AWARE constructs (IS NULL, coalesce, CASE) handle
-    validity themselves and return None here."""
+    validity themselves and return None here. Division/modulo by zero is
+    NULL (sqlite/DataFusion), so those divisors add value-dependent bits."""
     cols: list = []
-    if _collect_strict_refs(e, cols) is False:
+    divs: list = []
+    if _collect_strict_refs(e, cols, divs) is False:
         return None
     v = None
     for ref in cols:
@@ -724,12 +733,27 @@ def expr_validity(e, env: Env):
             continue
         if isinstance(c, Column) and c.validity is not None:
             v = c.validity if v is None else (v & c.validity)
+    for d in divs:
+        dv = eval_expr(d, env)
+        if isinstance(dv, torch.Tensor):
+            nz = dv != 0
+        elif isinstance(dv, (int, float)):
+            if dv != 0:
+                continue
+            nz = torch.zeros(env.n_rows, dtype=torch.bool,
+                             device=env.device)
+        else:
+            continue
+        v = nz if v is None else (v & nz)
     return v
 
 
-def _collect_strict_refs(e, out: list):
-    """Gather ColumnRefs; returns False if the expr contains a null-aware
-    construct (the caller must not apply strict propagation)."""
+def _collect_strict_refs(e, out: list, divs: list = None):
+    """Gather ColumnRefs (and /,% divisor exprs into *divs*); returns False
+    if the expr contains a null-aware construct (the caller must not apply
+    strict propagation)."""
+    if divs is None:
+        divs = []
     if isinstance(e, ColumnRef):
         out.append(e)
         return True
@@ -738,12 +762,15 @@ def _collect_strict_refs(e, out: list):
     if isinstance(e, FuncCall):
         if e.name in ("coalesce", "ifnull", "count"):
             return False
-        return all(_collect_strict_refs(a, out) is not False for a in e.args)
+        return all(_collect_strict_refs(a, out, divs) is not False
+                   for a in e.args)
     if isinstance(e, BinaryOp):
-        return (_collect_strict_refs(e.left, out) is not False
-                and _collect_strict_refs(e.right, out) is not False)
+        if e.op in ("/", "%"):
+            divs.append(e.right)
+        return (_collect_strict_refs(e.left, out, divs) is not False
+                and _collect_strict_refs(e.right, out, divs) is not False)
     if isinstance(e, UnaryOp):
-        return _collect_strict_refs(e.operand, out)
+        return _collect_strict_refs(e.operand, out, divs)
     if isinstance(e, Cast):
         return _collect_strict_refs(e.expr, out)
     if isinstance(e, (InList, Between, Like)):

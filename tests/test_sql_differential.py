@@ -141,6 +141,11 @@ QUERIES = [
     "WHERE d.k IS NULL ORDER BY f.a LIMIT 8",
     "SELECT f.a, d.label FROM flow f JOIN dims d ON d.k = f.k - 1 "
     "ORDER BY f.a LIMIT 7",
+    "SELECT a / 0 AS z FROM flow LIMIT 3",
+    "SELECT a % 0 AS z FROM flow ORDER BY a LIMIT 3",
+    "SELECT a / (k - k) AS z, a FROM flow ORDER BY a LIMIT 5",
+    "SELECT a FROM flow WHERE a / (a - a) IS NULL ORDER BY a LIMIT 4",
+    "SELECT a FROM flow WHERE a / 2 IS NOT NULL ORDER BY a LIMIT 4",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
