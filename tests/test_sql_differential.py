@@ -146,6 +146,19 @@ QUERIES = [
     "SELECT a / (k - k) AS z, a FROM flow ORDER BY a LIMIT 5",
     "SELECT a FROM flow WHERE a / (a - a) IS NULL ORDER BY a LIMIT 4",
     "SELECT a FROM flow WHERE a / 2 IS NOT NULL ORDER BY a LIMIT 4",
+    "SELECT a, count(*) OVER (PARTITION BY k) AS pc FROM flow "
+    "ORDER BY a, pc LIMIT 12",
+    "SELECT a, row_number() OVER (ORDER BY a DESC) AS rn FROM flow "
+    "ORDER BY a LIMIT 8",
+    "SELECT a, lag(a) OVER (ORDER BY a) AS p FROM flow ORDER BY a LIMIT 8",
+    "SELECT a, lead(b, 2) OVER (PARTITION BY k ORDER BY a) AS nb "
+    "FROM flow ORDER BY a, nb LIMIT 10",
+    "SELECT a, avg(b) OVER (PARTITION BY k) AS pb FROM flow "
+    "ORDER BY a LIMIT 6",
+    "SELECT k, a, sum(a) OVER (PARTITION BY k ORDER BY a DESC) AS rs "
+    "FROM flow ORDER BY k, a LIMIT 10",
+    "SELECT a, first_value(a) OVER (ORDER BY a DESC) AS fv FROM flow "
+    "ORDER BY a LIMIT 5",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
