@@ -154,7 +154,7 @@ QUERIES = [
     "SELECT a, lead(b, 2) OVER (PARTITION BY k ORDER BY a) AS nb "
     "FROM flow ORDER BY a, nb LIMIT 10",
     "SELECT a, avg(b) OVER (PARTITION BY k) AS pb FROM flow "
-    "ORDER BY a LIMIT 6",
+    "ORDER BY a, pb LIMIT 6",
     "SELECT k, a, sum(a) OVER (PARTITION BY k ORDER BY a DESC) AS rs "
     "FROM flow ORDER BY k, a LIMIT 10",
     "SELECT a, first_value(a) OVER (ORDER BY a DESC) AS fv FROM flow "
