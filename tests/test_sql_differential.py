@@ -103,8 +103,6 @@ QUERIES = [
     "LEFT JOIN dims d ON f.k = d.k",
     "SELECT a, substr(d.label, 2) AS tail FROM flow f JOIN dims d "
     "ON f.k = d.k ORDER BY a",
-    "SELECT k, group_concat(a, ',') AS g FROM (SELECT k, a FROM flow "
-    "ORDER BY a) GROUP BY k ORDER BY k" if False else
     "SELECT lower(d.label) AS l1 FROM dims d ORDER BY l1 DESC",
     "SELECT k % 2 AS p, avg(b) AS m FROM flow GROUP BY k % 2 ORDER BY p",
     "SELECT upper(d.label) || '-' || d.k AS tag FROM dims d ORDER BY tag",
