@@ -763,7 +763,8 @@ class SqlExecutor:
         arg = eval_expr(a.args[0], env)
         if name == "count":
             if a.distinct:
-                key = arg.data if isinstance(arg, Column) else arg
+                key = arg if isinstance(arg, Column) and arg.kind == "binary" \
+                    else (arg.data if isinstance(arg, Column) else arg)
                 pair_gid, _, _ = _encode_keys([gid, key], device)
                 # distinct (gid, val) pairs per gid
                 uniq_pair = _first_index_per_group(
@@ -806,6 +807,11 @@ class SqlExecutor:
         vals = as_tensor(arg, env)
         from .eval import expr_validity
         validity = expr_validity(a.args[0], env) if a.args else None
+        if vals.dtype.is_floating_point and bool(torch.isnan(vals).any()):
+            # NaN is the in-band NULL for value-level nulls (e.g. CASE
+            # without ELSE); aggregates must skip it like any NULL
+            nn = ~torch.isnan(vals)
+            validity = nn if validity is None else (validity & nn)
         if validity is not None and bool((~validity).any()):
             # SQL aggregates ignore NULL inputs
             vf = vals.float()

@@ -286,7 +286,19 @@ def eval_expr(e, env: Env) -> Value:
             result = torch.where(c, v, result)
         return result
     if isinstance(e, Between):
-        v = as_tensor(eval_expr(e.expr, env), env)
+        ev = eval_expr(e.expr, env)
+        if isinstance(ev, Column) and ev.kind == "binary":
+            lo_v = eval_expr(e.low, env)
+            hi_v = eval_expr(e.high, env)
+            if not isinstance(lo_v, str) or not isinstance(hi_v, str):
+                raise SqlError("string BETWEEN needs string bounds")
+            lob, hib = lo_v.encode(), hi_v.encode()
+            rows = ev.to_pylist()
+            r = torch.tensor([x is not None and lob <= x <= hib
+                              for x in rows], dtype=torch.bool,
+                             device=env.device)
+            return ~r if e.negated else r
+        v = as_tensor(ev, env)
         lo = as_tensor(eval_expr(e.low, env), env)
         hi = as_tensor(eval_expr(e.high, env), env)
         r = (v >= lo) & (v <= hi)
@@ -457,6 +469,12 @@ def _eval_cast(v: Value, to_type: str, env: Env) -> Value:
     if ty in ("varchar", "text", "string"):
         if isinstance(v, Column) and v.kind == "binary":
             return v
+        if isinstance(v, str):
+            return v
+        if isinstance(v, bool):
+            return "1" if v else "0"
+        if isinstance(v, (int, float)):
+            return str(v)
         t = as_tensor(v, env)
         vals = t.detach().cpu().tolist()
         return Column.from_strings([
@@ -466,6 +484,12 @@ def _eval_cast(v: Value, to_type: str, env: Env) -> Value:
     if ty not in _CAST_TYPES:
         raise SqlError(f"unknown cast type {to_type!r}")
     dt = _CAST_TYPES[ty]
+    if isinstance(v, str):  # constant: sqlite-style best-effort numeric
+        try:
+            f = float(v)
+        except ValueError:
+            f = 0.0
+        return f if dt.is_floating_point else int(f)
     if isinstance(v, Column) and v.kind == "binary":
         vals = v.to_strlist()
         if dt.is_floating_point:
@@ -678,8 +702,29 @@ def _eval_func(e: FuncCall, env: Env) -> Value:
                 lists.append([str(p)] * env.n_rows)
         return Column.from_strings(["".join(t) for t in zip(*lists)])
     if name == "nullif":
-        a = as_tensor(eval_expr(e.args[0], env), env).double()
-        b = as_tensor(eval_expr(e.args[1], env), env).double()
+        av = eval_expr(e.args[0], env)
+        bv = eval_expr(e.args[1], env)
+        if (isinstance(av, Column) and av.kind == "binary") \
+                or isinstance(av, str):
+            rows = av.to_pylist() if isinstance(av, Column) \
+                else [av.encode()] * env.n_rows
+            if isinstance(bv, str):
+                cmp = [bv.encode()] * env.n_rows
+            elif isinstance(bv, Column) and bv.kind == "binary":
+                cmp = bv.to_pylist()
+            else:
+                raise SqlError("nullif: mixed string/numeric args")
+            vals = [None if a2 == b2 else a2 for a2, b2 in zip(rows, cmp)]
+            out = Column.from_bytes([x if x is not None else b""
+                                     for x in vals])
+            if any(x is None for x in vals):
+                out = Column(out.kind, out.data, out.offsets,
+                             torch.tensor([x is not None for x in vals],
+                                          dtype=torch.bool,
+                                          device=env.device))
+            return out
+        a = as_tensor(av, env).double()
+        b = as_tensor(bv, env).double()
         return torch.where(a == b, torch.full_like(a, float("nan")), a)
     from .udf import scalar_udf
     udf = scalar_udf(name)

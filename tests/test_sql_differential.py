@@ -157,6 +157,16 @@ QUERIES = [
     "FROM flow ORDER BY k, a LIMIT 10",
     "SELECT a, first_value(a) OVER (ORDER BY a DESC) AS fv FROM flow "
     "ORDER BY a LIMIT 5",
+    "SELECT count(DISTINCT d.label) AS u FROM dims d",
+    "SELECT d.label FROM dims d WHERE d.label BETWEEN 'L1' AND 'L4' "
+    "ORDER BY d.label",
+    "SELECT CAST('42' AS INTEGER) AS i, CAST(7 AS TEXT) AS t "
+    "FROM flow LIMIT 1",
+    "SELECT nullif(d.label, 'L2') AS nl FROM dims d ORDER BY nl NULLS FIRST",
+    "SELECT k, min(CASE WHEN a > 50 THEN a END) AS m FROM flow "
+    "GROUP BY k ORDER BY k",
+    "SELECT f.k, count(DISTINCT d.label) AS u FROM flow f JOIN dims d "
+    "ON f.k = d.k GROUP BY f.k ORDER BY f.k",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
