@@ -758,8 +758,13 @@ class SqlExecutor:
                 return torch.zeros(g, dtype=torch.int64, device=device)
             return torch.full((g,), float("nan"), device=device)
         if name == "count" and (not a.args or isinstance(a.args[0], Star)):
-            return ops.segment_reduce(
-                torch.ones(env.n_rows, device=device), gid, g, "count")
+            ones = torch.ones(env.n_rows, device=device)
+            if a.filter is not None:
+                fm = as_tensor(eval_expr(a.filter, env), env).bool()
+                ones = torch.where(fm, ones, torch.zeros_like(ones))
+                return ops.segment_reduce(ones, gid, g, "sum"
+                                          ).to(torch.int64)
+            return ops.segment_reduce(ones, gid, g, "count")
         arg = eval_expr(a.args[0], env)
         if name == "count":
             if a.distinct:
@@ -807,6 +812,9 @@ class SqlExecutor:
         vals = as_tensor(arg, env)
         from .eval import expr_validity
         validity = expr_validity(a.args[0], env) if a.args else None
+        if a.filter is not None:
+            fm = as_tensor(eval_expr(a.filter, env), env).bool()
+            validity = fm if validity is None else (validity & fm)
         if vals.dtype.is_floating_point and bool(torch.isnan(vals).any()):
             # NaN is the in-band NULL for value-level nulls (e.g. CASE
             # without ELSE); aggregates must skip it like any NULL

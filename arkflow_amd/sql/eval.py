@@ -68,6 +68,8 @@ def expr_name(e) -> str:
     if isinstance(e, FuncCall):
         args = ", ".join(expr_name(a) for a in e.args)
         base = f"{e.name}({'DISTINCT ' if e.distinct else ''}{args})"
+        if getattr(e, "filter", None) is not None:
+            base += f" filter({expr_name(e.filter)})"
         if e.over is not None:
             part = ", ".join(expr_name(p) for p in e.over.partition_by)
             ob = ", ".join(expr_name(o) + ("" if asc else " desc")

@@ -64,6 +64,7 @@ class FuncCall:
     args: list
     distinct: bool = False
     over: "WindowSpec" = None
+    filter: object = None  # aggregate FILTER (WHERE ...) condition
 
 
 @dataclass
@@ -518,6 +519,14 @@ class Parser:
                     while self.accept("op", ","):
                         args.append(self._expr())
                 self.expect("op", ")")
+                filt = None
+                t2 = self.peek()
+                if t2.kind == "ident" and t2.value.lower() == "filter":
+                    self.next()
+                    self.expect("op", "(")
+                    self.expect("kw", "where")
+                    filt = self._expr()
+                    self.expect("op", ")")
                 over = None
                 if self.accept("kw", "over"):
                     self.expect("op", "(")
@@ -541,7 +550,7 @@ class Parser:
                                 break
                     self.expect("op", ")")
                     over = WindowSpec(part, order)
-                return FuncCall(name.lower(), args, distinct, over)
+                return FuncCall(name.lower(), args, distinct, over, filt)
             # qualified column
             if self.accept("op", "."):
                 col = self.expect("ident").value

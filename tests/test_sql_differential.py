@@ -167,6 +167,17 @@ QUERIES = [
     "GROUP BY k ORDER BY k",
     "SELECT f.k, count(DISTINCT d.label) AS u FROM flow f JOIN dims d "
     "ON f.k = d.k GROUP BY f.k ORDER BY f.k",
+    "SELECT sum(a) FILTER (WHERE b > 0.5) AS s FROM flow",
+    "SELECT k, count(*) FILTER (WHERE a > 50) AS hi, count(*) AS n "
+    "FROM flow GROUP BY k ORDER BY k",
+    "SELECT k, avg(a) FILTER (WHERE b < 0.5) AS m FROM flow "
+    "GROUP BY k ORDER BY k",
+    "SELECT d.label, d.k FROM dims d ORDER BY d.label DESC, d.k ASC",
+    "SELECT a, -(a + c) AS n, a - -c AS p FROM flow ORDER BY a LIMIT 6",
+    "SELECT a FROM flow WHERE NOT (a BETWEEN 10 AND 90) ORDER BY a LIMIT 6",
+    "SELECT d.label FROM dims d WHERE d.label NOT IN ('L0', 'L2') "
+    "ORDER BY d.label",
+    "SELECT a * 1.5 + c / 2.0 AS mix FROM flow ORDER BY mix LIMIT 7",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
