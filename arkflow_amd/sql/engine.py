@@ -827,10 +827,15 @@ class SqlExecutor:
                 s = ops.segment_reduce(
                     torch.where(validity, vf, torch.zeros_like(vf)),
                     gid, g, "sum")
+                cnt = ops.segment_reduce(validity.float(), gid, g, "sum")
                 if name == "sum":
+                    # SQL: SUM over zero qualifying rows is NULL, not 0
+                    if bool((cnt == 0).any()):
+                        return torch.where(cnt == 0,
+                                           torch.full_like(s, float("nan")),
+                                           s)
                     return s.to(torch.int64) \
                         if not vals.dtype.is_floating_point else s
-                cnt = ops.segment_reduce(validity.float(), gid, g, "sum")
                 return s / cnt
             fill = float("inf") if name == "min" else float("-inf")
             out = ops.segment_reduce(
