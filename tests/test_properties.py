@@ -377,3 +377,58 @@ def test_vrl_parser_fuzz_no_crash():
         except Exception as e:  # noqa: BLE001
             raise AssertionError(
                 f"vrl leaked {type(e).__name__}: {e}\nprogram: {prog!r}")
+
+
+def test_proto_wire_roundtrip_property():
+    """encode_message → decode_message round-trips random scalar messages
+    exactly (proto3 defaults: zero-valued fields drop and come back as
+    defaults)."""
+    import random
+
+    from arkflow_amd.processors.proto_wire import (ProtoSchema,
+                                                   decode_message,
+                                                   encode_message)
+
+    schema = ProtoSchema.parse(
+        "message T { double d = 1; float f = 2; int64 i = 3; "
+        "sint64 si = 4; bool ok = 5; string s = 6; fixed64 x = 7; "
+        "sfixed32 y = 8; }")
+    rng = random.Random(31)
+    for _ in range(200):
+        msg = {
+            "d": rng.uniform(-1e6, 1e6),
+            "f": 0.0,
+            "i": rng.randrange(-2**62, 2**62),
+            "si": rng.randrange(-2**30, 2**30),
+            "ok": rng.random() < 0.5,
+            "s": "".join(chr(rng.randrange(32, 0x300))
+                         for _ in range(rng.randrange(0, 12))),
+            "x": rng.randrange(0, 2**63),
+            "y": rng.randrange(-2**31, 2**31),
+        }
+        out = decode_message(encode_message(msg, schema), schema)
+        assert out["d"] == msg["d"] and out["i"] == msg["i"]
+        assert out["si"] == msg["si"] and out["ok"] == msg["ok"]
+        assert out["s"] == msg["s"] and out["x"] == msg["x"]
+        assert out["y"] == msg["y"]
+
+
+def test_vrl_processor_routes_interpreter_functions():
+    """Flat programs using row-wise stdlib functions (sha256 etc.) fall
+    back to the interpreter and still produce correct columns."""
+    import asyncio
+    import hashlib
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.processors.expr_proc import VrlProcessor
+
+    batch = MessageBatch({
+        "name": Column.from_strings(["alice", "bob"]),
+        "v": Column("numeric", __import__("torch").tensor([1.0, 2.0])),
+    })
+    proc = VrlProcessor({"program": '.h = sha256(.name)\n.w = .v'}, None)
+    loop = asyncio.new_event_loop()
+    out = loop.run_until_complete(proc.process(batch))[0]
+    assert out.column("h").to_strlist() == [
+        hashlib.sha256(b"alice").hexdigest(),
+        hashlib.sha256(b"bob").hexdigest()]
