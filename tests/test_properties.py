@@ -432,3 +432,47 @@ def test_vrl_processor_routes_interpreter_functions():
     assert out.column("h").to_strlist() == [
         hashlib.sha256(b"alice").hexdigest(),
         hashlib.sha256(b"bob").hexdigest()]
+
+
+def test_json_arrow_roundtrip_property():
+    """arrow_to_json → json_to_arrow round-trips random typed batches
+    (ints, floats, bools, unicode strings) column-for-column."""
+    import asyncio
+    import random
+
+    import torch
+
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.processors.json_proc import (ArrowToJsonProcessor,
+                                                  JsonToArrowProcessor)
+
+    rng = random.Random(17)
+    loop = asyncio.new_event_loop()
+    for case in range(20):
+        n = rng.randrange(1, 40)
+        batch = MessageBatch({
+            "i": Column("numeric", torch.tensor(
+                [rng.randrange(-10**9, 10**9) for _ in range(n)],
+                dtype=torch.int64)),
+            "f": Column("numeric", torch.tensor(
+                [round(rng.uniform(-100, 100), 4) for _ in range(n)],
+                dtype=torch.float64)),
+            "ok": Column("numeric", torch.tensor(
+                [rng.random() < 0.5 for _ in range(n)],
+                dtype=torch.bool)),
+            "s": Column.from_strings(
+                ["".join(chr(rng.randrange(32, 0x2500))
+                         for _ in range(rng.randrange(0, 10)))
+                 for _ in range(n)]),
+        })
+        enc = loop.run_until_complete(
+            ArrowToJsonProcessor({}, None).process(batch))[0]
+        dec = loop.run_until_complete(JsonToArrowProcessor(
+            {"schema": {"i": "int", "f": "float", "ok": "bool",
+                        "s": "str"}}, None).process(enc))[0]
+        assert dec.column("i").to_pylist() == batch.column("i").to_pylist()
+        assert dec.column("ok").to_pylist() == batch.column("ok").to_pylist()
+        assert dec.column("s").to_strlist() == batch.column("s").to_strlist()
+        got_f = dec.column("f").to_pylist()
+        exp_f = batch.column("f").to_pylist()
+        assert all(abs(a - b) < 1e-9 for a, b in zip(got_f, exp_f)), case
