@@ -211,6 +211,24 @@ def _sqlite_exec(data, dims, sql):
     return rows
 
 
+def _rows_close(a, b):
+    """Tuple-rows equal, with float elements compared at 1.5e-4 tolerance:
+    our columns store float32 where sqlite keeps the python double, so a
+    value can legitimately round across the 4-decimal boundary."""
+    if len(a) != len(b):
+        return False
+    for ra, rb in zip(a, b):
+        if len(ra) != len(rb):
+            return False
+        for x, y in zip(ra, rb):
+            if isinstance(x, float) and isinstance(y, float):
+                if abs(x - y) > 1.5e-4:
+                    return False
+            elif x != y:
+                return False
+    return True
+
+
 def _normalize(rows):
     out = []
     for r in rows:
@@ -227,7 +245,7 @@ def _normalize(rows):
     return out
 
 
-@pytest.mark.parametrize("seed", [0, 1, 2, 3, 4, 5, 6, 7])
+@pytest.mark.parametrize("seed", list(range(10)))
 @pytest.mark.parametrize("sql", QUERIES)
 def test_differential_vs_sqlite(seed, sql):
     rng = random.Random(seed * 1000 + zlib.crc32(sql.encode()) % 997)
@@ -250,8 +268,8 @@ def test_differential_vs_sqlite(seed, sql):
         # global aggregate over empty input: both return one row
         assert ours == theirs
         return
-    assert ours == theirs, f"{sql}\nseed={seed} n={n}\n{ours[:5]} vs " \
-                           f"{theirs[:5]}"
+    assert _rows_close(ours, theirs), \
+        f"{sql}\nseed={seed} n={n}\n{ours[:5]} vs {theirs[:5]}"
 
 
 @pytest.mark.parametrize("seed", range(5))
