@@ -89,10 +89,17 @@ class SqlExecutor:
     # ------------------------------------------------------------------- run
     def execute(self, tables: Dict[str, MessageBatch]) -> MessageBatch:
         sel = self.select
-        from_name = sel.from_table or DEFAULT_TABLE
-        if from_name not in tables:
-            raise SqlError(f"unknown table {from_name!r}; have {sorted(tables)}")
-        base = tables[from_name]
+        from_spec = sel.from_table or DEFAULT_TABLE
+        if isinstance(from_spec, Select):
+            # derived table: execute the subquery, use it as the base
+            base = SqlExecutor._from_select(from_spec).execute(tables)
+            from_name = sel.from_alias or "subquery"
+        else:
+            from_name = from_spec
+            if from_name not in tables:
+                raise SqlError(
+                    f"unknown table {from_name!r}; have {sorted(tables)}")
+            base = tables[from_name]
         device = base.device
 
         columns: Dict[str, Column] = dict(base.columns)
@@ -103,9 +110,14 @@ class SqlExecutor:
 
         # ------------------------------------------------------------- joins
         for j in sel.joins:
-            right = tables.get(j.table)
-            if right is None:
-                raise SqlError(f"unknown join table {j.table!r}")
+            if isinstance(j.table, Select):
+                if not j.alias:
+                    raise SqlError("a joined subquery requires an alias")
+                right = SqlExecutor._from_select(j.table).execute(tables)
+            else:
+                right = tables.get(j.table)
+                if right is None:
+                    raise SqlError(f"unknown join table {j.table!r}")
             columns, order = self._apply_join(
                 columns, order, alias, j, right, device)
 

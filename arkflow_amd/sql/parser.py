@@ -284,7 +284,12 @@ class Parser:
             sel.distinct = True
         sel.projections = self._projections()
         if self.accept("kw", "from"):
-            sel.from_table = self._table_name()
+            # derived table: FROM ( SELECT ... ) [AS] alias
+            if self.accept("op", "("):
+                sel.from_table = self._parse_select()
+                self.expect("op", ")")
+            else:
+                sel.from_table = self._table_name()
             alias = self._maybe_alias()
             sel.from_alias = alias
             while True:
@@ -296,7 +301,11 @@ class Parser:
                     kind = "left"
                 if self.accept("kw", "join"):
                     kind = kind or "inner"
-                    tname = self._table_name()
+                    if self.accept("op", "("):
+                        tname = self._parse_select()
+                        self.expect("op", ")")
+                    else:
+                        tname = self._table_name()
                     talias = self._maybe_alias()
                     self.expect("kw", "on")
                     on = self._expr()

@@ -178,6 +178,16 @@ QUERIES = [
     "SELECT d.label FROM dims d WHERE d.label NOT IN ('L0', 'L2') "
     "ORDER BY d.label",
     "SELECT a * 1.5 + c / 2.0 AS mix FROM flow ORDER BY mix LIMIT 7",
+    "SELECT s.k, s.tot FROM (SELECT k, sum(a) AS tot FROM flow "
+    "GROUP BY k) s ORDER BY s.k",
+    "SELECT avg(tot) AS m FROM (SELECT k, sum(a) AS tot FROM flow "
+    "GROUP BY k)",
+    "SELECT t.k, d.label FROM (SELECT DISTINCT k FROM flow) t "
+    "JOIN dims d ON t.k = d.k ORDER BY t.k",
+    "SELECT x.a FROM (SELECT a, b FROM flow WHERE b > 0.5) x "
+    "WHERE x.a < 50 ORDER BY x.a LIMIT 10",
+    "SELECT f.a, s.tot FROM flow f JOIN (SELECT k, count(*) AS tot "
+    "FROM flow GROUP BY k) s ON f.k = s.k ORDER BY f.a LIMIT 10",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
