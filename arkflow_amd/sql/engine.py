@@ -28,6 +28,7 @@ from .parser import (
     ColumnRef,
     FuncCall,
     Literal,
+    ScalarSubquery,
     Select,
     SqlError,
     Star,
@@ -86,6 +87,92 @@ class SqlExecutor:
             if select.union_all else None
         return ex
 
+    # -------------------------------------------------- subquery rewriting
+    def _rewrite_subqueries(self, e, tables):
+        """Per-execute rewrite of UNCORRELATED subqueries in WHERE / JOIN ON
+        into literals: scalar subqueries become their single value; IN
+        (SELECT ...) becomes a literal item list. Statements are pre-parsed
+        and re-executed per batch, so the AST itself is never mutated —
+        rewritten copies are built each execute."""
+        import dataclasses
+
+        from .parser import InList as _IL
+        if e is None:
+            return None
+        if isinstance(e, ScalarSubquery):
+            sub = SqlExecutor._from_select(e.select).execute(tables)
+            cols = list(sub.columns.values())
+            if len(cols) != 1:
+                raise SqlError("scalar subquery must return one column")
+            rows = cols[0].to_pylist()
+            if len(rows) == 0:
+                return Literal(None)
+            if len(rows) > 1:
+                raise SqlError("scalar subquery returned more than one row")
+            v = rows[0]
+            if isinstance(v, (bytes, bytearray)):
+                v = v.decode("utf-8", "replace")
+            return Literal(v)
+        if isinstance(e, _IL) and e.subquery is not None:
+            sub = SqlExecutor._from_select(e.subquery).execute(tables)
+            cols = list(sub.columns.values())
+            if len(cols) != 1:
+                raise SqlError("IN subquery must return one column")
+            items = []
+            for v in cols[0].to_pylist():
+                if v is None:
+                    continue  # NULL members never equal anything
+                if isinstance(v, (bytes, bytearray)):
+                    v = v.decode("utf-8", "replace")
+                items.append(Literal(v))
+            return _IL(self._rewrite_subqueries(e.expr, tables), items,
+                       e.negated)
+        if dataclasses.is_dataclass(e) and not isinstance(e, Select):
+            changed = False
+            kwargs = {}
+            for f in dataclasses.fields(e):
+                v = getattr(e, f.name)
+                if f.name == "whens" and isinstance(v, list):
+                    nv = [(self._rewrite_subqueries(c, tables),
+                           self._rewrite_subqueries(r, tables))
+                          for c, r in v]
+                    changed = True
+                elif isinstance(v, list) and f.name in ("args", "items"):
+                    nv = [self._rewrite_subqueries(x, tables) for x in v]
+                    changed |= any(a is not b for a, b in zip(nv, v))
+                elif dataclasses.is_dataclass(v) \
+                        and not isinstance(v, Select):
+                    nv = self._rewrite_subqueries(v, tables)
+                    changed |= nv is not v
+                else:
+                    nv = v
+                kwargs[f.name] = nv
+            return dataclasses.replace(e, **kwargs) if changed else e
+        return e
+
+    @staticmethod
+    def _contains_subquery(e) -> bool:
+        import dataclasses
+        if e is None or not dataclasses.is_dataclass(e):
+            return False
+        if isinstance(e, ScalarSubquery):
+            return True
+        from .parser import InList as _IL
+        if isinstance(e, _IL) and e.subquery is not None:
+            return True
+        if isinstance(e, Select):
+            return False
+        for f in dataclasses.fields(e):
+            v = getattr(e, f.name)
+            if isinstance(v, list):
+                for x in v:
+                    x2 = x[0] if isinstance(x, tuple) else x
+                    if SqlExecutor._contains_subquery(x2):
+                        return True
+            elif SqlExecutor._contains_subquery(v):
+                return True
+        return False
+
     # ------------------------------------------------------------------- run
     def execute(self, tables: Dict[str, MessageBatch]) -> MessageBatch:
         sel = self.select
@@ -110,6 +197,10 @@ class SqlExecutor:
 
         # ------------------------------------------------------------- joins
         for j in sel.joins:
+            if self._contains_subquery(j.on):
+                import dataclasses as _dc
+                j = _dc.replace(
+                    j, on=self._rewrite_subqueries(j.on, tables))
             if isinstance(j.table, Select):
                 if not j.alias:
                     raise SqlError("a joined subquery requires an alias")
@@ -126,11 +217,13 @@ class SqlExecutor:
 
         # ------------------------------------------------------------- where
         if sel.where is not None:
-            fused = self._try_fused_filter(sel.where, columns, env)
+            where_e = self._rewrite_subqueries(sel.where, tables) \
+                if self._contains_subquery(sel.where) else sel.where
+            fused = self._try_fused_filter(where_e, columns, env)
             if fused is not None:
                 columns, n_rows = fused
             else:
-                idx = self._filter_indices(sel.where, env)
+                idx = self._filter_indices(where_e, env)
                 columns = {k: c.take(idx) for k, c in columns.items()}
                 n_rows = int(idx.shape[0])
             env = Env(columns, n_rows, device)

@@ -97,6 +97,12 @@ class InList:
     expr: object
     items: list
     negated: bool = False
+    subquery: object = None  # IN (SELECT ...): uncorrelated subselect
+
+
+@dataclass
+class ScalarSubquery:
+    select: object  # uncorrelated (SELECT ...) producing one value
 
 
 @dataclass
@@ -430,6 +436,10 @@ class Parser:
         if t.kind == "kw" and t.value == "in":
             self.next()
             self.expect("op", "(")
+            if self.peek().kind == "kw" and self.peek().value == "select":
+                sub = self._parse_select()
+                self.expect("op", ")")
+                return InList(left, [], negated, subquery=sub)
             items = [self._expr()]
             while self.accept("op", ","):
                 items.append(self._expr())
@@ -511,6 +521,10 @@ class Parser:
             self.expect("op", ")")
             return Cast(e, ty)
         if t.kind == "op" and t.value == "(":
+            if self.peek().kind == "kw" and self.peek().value == "select":
+                sub = self._parse_select()
+                self.expect("op", ")")
+                return ScalarSubquery(sub)
             e = self._expr()
             self.expect("op", ")")
             return e

@@ -188,6 +188,13 @@ QUERIES = [
     "WHERE x.a < 50 ORDER BY x.a LIMIT 10",
     "SELECT f.a, s.tot FROM flow f JOIN (SELECT k, count(*) AS tot "
     "FROM flow GROUP BY k) s ON f.k = s.k ORDER BY f.a LIMIT 10",
+    "SELECT a FROM flow WHERE a > (SELECT avg(a) FROM flow) "
+    "ORDER BY a LIMIT 8",
+    "SELECT a FROM flow WHERE k IN (SELECT d.k FROM dims d "
+    "WHERE d.k < 3) ORDER BY a LIMIT 10",
+    "SELECT a FROM flow WHERE k NOT IN (SELECT d.k FROM dims d "
+    "WHERE d.k < 6) ORDER BY a LIMIT 10",
+    "SELECT count(*) AS c FROM flow WHERE b < (SELECT max(b) FROM flow)",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
