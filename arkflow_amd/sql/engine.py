@@ -28,6 +28,7 @@ from .parser import (
     ColumnRef,
     FuncCall,
     Literal,
+    ExistsSubquery,
     ScalarSubquery,
     Select,
     SqlError,
@@ -99,6 +100,10 @@ class SqlExecutor:
         from .parser import InList as _IL
         if e is None:
             return None
+        if isinstance(e, ExistsSubquery):
+            sub = SqlExecutor._from_select(e.select).execute(tables)
+            hit = sub.num_rows > 0
+            return Literal(not hit if e.negated else hit)
         if isinstance(e, ScalarSubquery):
             sub = SqlExecutor._from_select(e.select).execute(tables)
             cols = list(sub.columns.values())
@@ -155,7 +160,7 @@ class SqlExecutor:
         import dataclasses
         if e is None or not dataclasses.is_dataclass(e):
             return False
-        if isinstance(e, ScalarSubquery):
+        if isinstance(e, (ScalarSubquery, ExistsSubquery)):
             return True
         from .parser import InList as _IL
         if isinstance(e, _IL) and e.subquery is not None:

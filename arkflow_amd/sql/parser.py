@@ -106,6 +106,12 @@ class ScalarSubquery:
 
 
 @dataclass
+class ExistsSubquery:
+    select: object
+    negated: bool = False
+
+
+@dataclass
 class IsNull:
     expr: object
     negated: bool = False
@@ -528,6 +534,19 @@ class Parser:
             e = self._expr()
             self.expect("op", ")")
             return e
+        if t.kind == "ident" and t.value.lower() == "exists":
+            self.expect("op", "(")
+            sub = self._parse_select()
+            self.expect("op", ")")
+            return ExistsSubquery(sub)
+        if t.kind == "kw" and t.value == "not" \
+                and self.peek().kind == "ident" \
+                and self.peek().value.lower() == "exists":
+            self.next()
+            self.expect("op", "(")
+            sub = self._parse_select()
+            self.expect("op", ")")
+            return ExistsSubquery(sub, negated=True)
         if t.kind == "ident" or (t.kind == "kw" and t.value in (
                 "left", "right", "row_number", "rank", "dense_rank")):
             name = t.value

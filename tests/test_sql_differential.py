@@ -195,6 +195,10 @@ QUERIES = [
     "SELECT a FROM flow WHERE k NOT IN (SELECT d.k FROM dims d "
     "WHERE d.k < 6) ORDER BY a LIMIT 10",
     "SELECT count(*) AS c FROM flow WHERE b < (SELECT max(b) FROM flow)",
+    "SELECT count(*) AS c FROM flow WHERE EXISTS (SELECT 1 FROM dims d "
+    "WHERE d.k > 5)",
+    "SELECT count(*) AS c FROM flow WHERE NOT EXISTS (SELECT 1 "
+    "FROM dims d WHERE d.k > 99)",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
