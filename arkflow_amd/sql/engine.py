@@ -63,8 +63,8 @@ class SqlExecutor:
         for w_ in self._windows:
             wseen.setdefault(expr_name(w_), w_)
         self._windows = list(wseen.values())
-        self._union = SqlExecutor._from_select(self.select.union_all) \
-            if self.select.union_all else None
+        self._setops = [(op, SqlExecutor._from_select(sub))
+                        for op, sub in self.select.set_ops]
 
     @staticmethod
     def _from_select(select: Select) -> "SqlExecutor":
@@ -84,8 +84,8 @@ class SqlExecutor:
         ex._windows = []
         for e, _ in select.projections:
             collect_window_calls(e, ex._windows)
-        ex._union = SqlExecutor._from_select(select.union_all) \
-            if select.union_all else None
+        ex._setops = [(op, SqlExecutor._from_select(sub))
+                      for op, sub in select.set_ops]
         return ex
 
     # -------------------------------------------------- subquery rewriting
@@ -295,7 +295,9 @@ class SqlExecutor:
             result = result.take(rep)
 
         # ---------------------------------------------------------- order by
-        if sel.order_by:
+        # (compound selects: ORDER BY/LIMIT bind to the whole compound and
+        # run after the set ops below)
+        if sel.order_by and not self._setops:
             cur_cols = dict(env.columns)  # permuted alongside result
             final_env = Env(
                 {**cur_cols, **result.columns},
@@ -351,23 +353,98 @@ class SqlExecutor:
                     result.num_rows, device, env.agg_results)
 
         # ---------------------------------------------------- offset / limit
+        if not self._setops:
+            if sel.offset:
+                result = result.slice(min(sel.offset, result.num_rows),
+                                      max(result.num_rows - sel.offset, 0))
+            if sel.limit is not None and result.num_rows > sel.limit:
+                result = result.slice(0, sel.limit)
+            return result
+
+        # ------------------------------------------- set ops (compound)
+        # UNION [ALL] / INTERSECT / EXCEPT left-to-right, then the
+        # compound-level ORDER BY / OFFSET / LIMIT
+        for op, ex in self._setops:
+            result = self._apply_setop(result, op, ex.execute(tables),
+                                       device)
+        result = self._compound_order_limit(result, device)
+        return result
+
+    # ------------------------------------------------------------- set ops
+    @staticmethod
+    def _apply_setop(result: MessageBatch, op: str, other: MessageBatch,
+                     device) -> MessageBatch:
+        from ..batch import concat_batches
+        if other.column_names != result.column_names and \
+                len(other.column_names) == len(result.column_names):
+            # positional semantics: rename the right side to our names
+            other = MessageBatch(
+                dict(zip(result.column_names, other.columns.values())),
+                other.input_name)
+        if op == "union_all":
+            return concat_batches([result, other])
+        both = concat_batches([result, other])
+        gid, _, g = _encode_keys(
+            [c if c.kind == "binary" else c.data
+             for c in both.columns.values()], device)
+        nl = result.num_rows
+        if op == "union":
+            rep = _first_index_per_group(gid, g)
+            rep = rep[rep < both.num_rows].sort().values
+            return both.take(rep)
+        gl = gid[:nl]
+        first = _first_index_per_group(gl, g)
+        ar = torch.arange(nl, dtype=torch.int64, device=gl.device)
+        rep_rows = (first[gl] == ar).nonzero(as_tuple=True)[0]
+        right_has = torch.zeros(g, dtype=torch.bool, device=gl.device)
+        if both.num_rows > nl:
+            right_has[gid[nl:]] = True
+        keep = right_has[gl[rep_rows]]
+        if op == "except":
+            keep = ~keep
+        return result.take(rep_rows[keep])
+
+    def _compound_order_limit(self, result: MessageBatch, device
+                              ) -> MessageBatch:
+        """Compound-level ORDER BY / OFFSET / LIMIT: keys may reference the
+        output columns (names or positions) only — sqlite/standard."""
+        sel = self.select
+        if sel.order_by:
+            out_names = list(result.columns)
+            from .parser import Literal as _Lit
+            for key_spec in reversed(sel.order_by):
+                e, asc = key_spec[0], key_spec[1]
+                nulls_first = key_spec[2] if len(key_spec) > 2 else None
+                if isinstance(e, _Lit) and isinstance(e.value, int) \
+                        and not isinstance(e.value, bool) \
+                        and 1 <= e.value <= len(out_names):
+                    e = ColumnRef(out_names[e.value - 1], None)
+                env = Env(dict(result.columns), result.num_rows, device)
+                v = eval_expr(e, env)
+                if isinstance(v, Column) and v.kind == "binary":
+                    vals = v.to_pylist()
+                    idx = torch.tensor(
+                        sorted(range(len(vals)),
+                               key=lambda i: vals[i]
+                               if vals[i] is not None else b"",
+                               reverse=not asc),
+                        dtype=torch.int64, device=device)
+                else:
+                    idx = ops.sort_indices(as_tensor(v, env), ascending=asc)
+                from .eval import expr_validity
+                kv = expr_validity(e, env)
+                if isinstance(v, Column) and v.validity is not None:
+                    kv = v.validity if kv is None else (kv & v.validity)
+                if kv is not None and bool((~kv).any()):
+                    null_asc = asc if nulls_first is None else nulls_first
+                    idx = idx[ops.sort_indices(
+                        kv[idx].to(torch.int32), ascending=null_asc)]
+                result = result.take(idx)
         if sel.offset:
             result = result.slice(min(sel.offset, result.num_rows),
                                   max(result.num_rows - sel.offset, 0))
         if sel.limit is not None and result.num_rows > sel.limit:
             result = result.slice(0, sel.limit)
-
-        # --------------------------------------------------------- union all
-        if self._union is not None:
-            from ..batch import concat_batches
-            other = self._union.execute(tables)
-            # positional union: rename other's columns to ours
-            if other.column_names != result.column_names and \
-                    len(other.column_names) == len(result.column_names):
-                other = MessageBatch(
-                    dict(zip(result.column_names, other.columns.values())),
-                    other.input_name)
-            result = concat_batches([result, other])
         return result
 
     # --------------------------------------------------------------- windows

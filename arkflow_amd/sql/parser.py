@@ -145,7 +145,8 @@ class Select:
     order_by: List[Tuple[object, bool]] = field(default_factory=list)  # (expr, asc)
     limit: Optional[int] = None
     offset: int = 0
-    union_all: Optional["Select"] = None
+    union_all: Optional["Select"] = None  # legacy single UNION ALL chain
+    set_ops: list = field(default_factory=list)  # [("union"|"union_all"|"intersect"|"except", Select)]
 
 
 AGGREGATE_FUNCS = {"count", "sum", "avg", "min", "max"}
@@ -285,6 +286,10 @@ class Parser:
         return sel
 
     def _parse_select(self) -> Select:
+        sel = self._parse_select_core()
+        return self._parse_compound_tail(sel)
+
+    def _parse_select_core(self) -> Select:
         t = self.peek()
         if t.kind == "kw" and t.value in (
             "insert", "update", "delete", "create", "drop", "alter", "truncate"
@@ -335,6 +340,26 @@ class Parser:
                 sel.group_by.append(self._expr())
         if self.accept("kw", "having"):
             sel.having = self._expr()
+        return sel
+
+    def _parse_compound_tail(self, sel: Select) -> Select:
+        # compound selects: UNION [ALL] / INTERSECT / EXCEPT between the
+        # cores; a trailing ORDER BY / LIMIT applies to the WHOLE compound
+        # (sqlite/standard)
+        while True:
+            t = self.peek()
+            if t.kind == "kw" and t.value == "union":
+                self.next()
+                op = "union_all" if self.accept("kw", "all") else "union"
+            elif t.kind == "ident" and t.value.lower() in ("intersect",
+                                                           "except"):
+                self.next()
+                op = t.value.lower()
+            else:
+                break
+            sel.set_ops.append((op, self._parse_select_core()))
+        if len(sel.set_ops) == 1 and sel.set_ops[0][0] == "union_all":
+            sel.union_all = sel.set_ops[0][1]  # legacy field (fast paths)
         if self.accept("kw", "order"):
             self.expect("kw", "by")
             while True:
@@ -358,9 +383,6 @@ class Parser:
             sel.limit = int(self.expect("number").value)
         if self.accept("kw", "offset"):
             sel.offset = int(self.expect("number").value)
-        if self.accept("kw", "union"):
-            self.expect("kw", "all")  # UNION (distinct) unsupported; ALL only
-            sel.union_all = self._parse_select()
         return sel
 
     def _table_name(self) -> str:
@@ -370,7 +392,8 @@ class Parser:
         if self.accept("kw", "as"):
             return self.expect("ident").value
         t = self.peek()
-        if t.kind == "ident":
+        if t.kind == "ident" and t.value.lower() not in ("intersect",
+                                                         "except"):
             return self.next().value
         return None
 

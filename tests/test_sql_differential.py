@@ -199,6 +199,15 @@ QUERIES = [
     "WHERE d.k > 5)",
     "SELECT count(*) AS c FROM flow WHERE NOT EXISTS (SELECT 1 "
     "FROM dims d WHERE d.k > 99)",
+    "SELECT a FROM flow WHERE a < 10 UNION SELECT a FROM flow "
+    "WHERE a < 5 ORDER BY a",
+    "SELECT k FROM flow INTERSECT SELECT d.k FROM dims d ORDER BY k",
+    "SELECT k FROM flow EXCEPT SELECT d.k FROM dims d WHERE d.k < 4 "
+    "ORDER BY k",
+    "SELECT a FROM flow WHERE a < 10 UNION ALL SELECT a FROM flow "
+    "WHERE a > 95 ORDER BY a LIMIT 12",
+    "SELECT d.label FROM dims d UNION SELECT d.label FROM dims d "
+    "ORDER BY 1",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
