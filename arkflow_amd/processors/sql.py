@@ -46,7 +46,7 @@ class SqlProcessor(Processor):
         from ..sql.parser import BinaryOp, ColumnRef, Literal, Star
         s = self.executor.select
         if (s.joins or s.group_by or s.having or s.order_by or s.distinct
-                or s.limit is not None or s.offset or s.union_all
+                or s.limit is not None or s.offset or s.set_ops
                 or self.temporary_specs or s.where is None):
             return None
         if len(s.projections) != 1 or not isinstance(s.projections[0][0],
@@ -71,7 +71,7 @@ class SqlProcessor(Processor):
         from ..sql.parser import BinaryOp, ColumnRef, FuncCall, Literal
         s = self.executor.select
         if (s.joins or s.having or s.order_by or s.distinct
-                or s.limit is not None or s.offset or s.union_all
+                or s.limit is not None or s.offset or s.set_ops
                 or self.temporary_specs):
             return None
         if len(s.group_by) != 1 or not isinstance(s.group_by[0], ColumnRef):
