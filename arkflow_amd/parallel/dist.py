@@ -130,7 +130,8 @@ def repartition_by_key(batch: MessageBatch, key_column: str
         col = reordered.columns[name]
         kind = 1 if col.kind == "binary" else 0
         dtc = -1 if kind else DT.index(col.data.dtype)
-        local += [kind, dtc, 1 if len(col) else 0]
+        local += [kind, dtc, 1 if len(col) else 0,
+                  1 if col.validity is not None else 0]
     meta_dev = torch.device("cuda") if dist.get_backend() == "nccl" \
         else torch.device("cpu")  # nccl/RCCL collectives need device tensors
     lt = torch.tensor(local, dtype=torch.int64, device=meta_dev)
@@ -141,9 +142,18 @@ def repartition_by_key(batch: MessageBatch, key_column: str
     cols: Dict[str, Column] = {}
     for ci, name in enumerate(names):
         col = reordered.columns[name]
-        nonempty = [g for g in gathered_meta if int(g[3 * ci + 2])]
-        is_binary = any(int(g[3 * ci]) == 1 for g in nonempty) or (
+        nonempty = [g for g in gathered_meta if int(g[4 * ci + 2])]
+        is_binary = any(int(g[4 * ci]) == 1 for g in nonempty) or (
             not nonempty and col.kind == "binary")
+        # validity consensus: if ANY rank carries a validity mask, every
+        # rank exchanges one (all-True where absent) — NULL bits must
+        # survive the shuffle, and asymmetric exchanges deadlock
+        any_validity = any(int(g[4 * ci + 3]) for g in gathered_meta)
+        out_validity = None
+        if any_validity:
+            v = col.validity if col.validity is not None else \
+                torch.ones(len(col), dtype=torch.bool)
+            out_validity = _exchange_numeric(v, in_splits, out_splits)
         if is_binary:
             if col.kind != "binary":
                 if len(col) != 0:
@@ -151,12 +161,14 @@ def repartition_by_key(batch: MessageBatch, key_column: str
                                      "differs across ranks")
                 col = Column("binary", torch.empty(0, dtype=torch.uint8),
                              torch.zeros(1, dtype=torch.int64))
-            cols[name] = _exchange_binary(col, in_splits, out_splits, w)
+            out = _exchange_binary(col, in_splits, out_splits, w)
+            cols[name] = Column(out.kind, out.data, out.offsets,
+                                out_validity)
             continue
         # symmetric fold over NON-EMPTY ranks' dtypes (identical on every
         # rank) so all ranks pick the same wire dtype
-        dts = [DT[int(g[3 * ci + 1])] for g in nonempty
-               if int(g[3 * ci]) == 0]
+        dts = [DT[int(g[4 * ci + 1])] for g in nonempty
+               if int(g[4 * ci]) == 0]
         if dts:
             target = dts[0]
             for dt in dts[1:]:
@@ -169,7 +181,7 @@ def repartition_by_key(batch: MessageBatch, key_column: str
         else:
             data = col.data.to(target)
         data = _exchange_numeric(data, in_splits, out_splits)
-        cols[name] = Column("numeric", data)
+        cols[name] = Column("numeric", data, validity=out_validity)
     return MessageBatch(cols, batch.input_name)
 
 

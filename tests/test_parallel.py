@@ -227,3 +227,44 @@ streams:
         # repartition collective actually moved rows between processes
         assert bool((dest == rank).all())
     assert total == world * 1280  # nothing lost or duplicated
+
+
+def _run_repartition_validity(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from arkflow_amd.batch import Column, MessageBatch
+    from arkflow_amd.parallel.dist import repartition_by_key
+
+    n = 60
+    # NULL iff (k % 3 == 0) — a key-derived invariant every rank can check
+    # after the shuffle regardless of where rows land
+    k = torch.arange(rank * n, (rank + 1) * n, dtype=torch.int64)
+    valid = (k % 3) != 0
+    sc = Column.from_strings([f"x{int(x)}" for x in k])
+    batch = MessageBatch({
+        "k": Column("numeric", k),
+        "v": Column("numeric", k.to(torch.float32), validity=valid),
+        "s": Column("binary", sc.data, sc.offsets, valid),
+    })
+    out = repartition_by_key(batch, "k")
+    ko = out.column("k").data
+    for name in ("v", "s"):
+        c = out.column(name)
+        assert c.validity is not None, name
+        assert torch.equal(c.validity, (ko % 3) != 0), name
+    results[rank] = out.num_rows
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_repartition_preserves_validity():
+    port = 29537
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_run_repartition_validity, args=(2, port, results),
+                 nprocs=2, join=True)
+        assert results[0] + results[1] == 120
