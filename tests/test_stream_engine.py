@@ -670,3 +670,23 @@ def test_engine_dynamic_topic_routing_e2e():
     for t, rows in msgs.items():
         want = t.removeprefix("shard-")
         assert all(int(m[0]) % 2 == int(want) for m in rows)
+
+
+def test_stream_sql_with_subquery_runs(run):
+    """A pipeline sql processor using a scalar subquery executes per batch
+    (the rewrite happens per execute; pre-parsed statements are shared)."""
+    import torch
+
+    from arkflow_amd.batch import MessageBatch
+    from arkflow_amd.processors.sql import SqlProcessor
+
+    proc = SqlProcessor({
+        "query": "SELECT a FROM flow WHERE a > (SELECT avg(a) FROM flow) "
+                 "ORDER BY a"})
+    for base in (0, 100):
+        batch = MessageBatch.from_dict(
+            {"a": torch.arange(base, base + 10, dtype=torch.int64)})
+        out = run(proc.process(batch))[0]
+        # avg of base..base+9 is base+4.5 → keeps the top 5
+        assert out.column("a").data.tolist() == list(
+            range(base + 5, base + 10))
