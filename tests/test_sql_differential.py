@@ -208,6 +208,12 @@ QUERIES = [
     "WHERE a > 95 ORDER BY a LIMIT 12",
     "SELECT d.label FROM dims d UNION SELECT d.label FROM dims d "
     "ORDER BY 1",
+    "SELECT k, count(*) AS c FROM flow WHERE k IN (SELECT k FROM flow "
+    "GROUP BY k HAVING count(*) > 3) GROUP BY k ORDER BY k",
+    "SELECT a FROM (SELECT a FROM (SELECT a, b FROM flow WHERE b > 0.2) "
+    "WHERE a < 60) ORDER BY a LIMIT 8",
+    "SELECT t.m FROM (SELECT k, max(a) AS m FROM flow GROUP BY k) t "
+    "WHERE t.m > (SELECT avg(a) FROM flow) ORDER BY t.m",
     "SELECT count(*) AS n FROM flow WHERE 1 = 0",
     "SELECT a, sum(b) OVER (PARTITION BY k ORDER BY a DESC) AS rs FROM flow "
     "ORDER BY k, a DESC, rs",
